@@ -1,0 +1,404 @@
+// Flow-GNN kernels for MI355X (gfx950, CDNA4). Native HIP — no CUDA compat.
+//
+// Covers the DGL-delegated ops of the reference's flow-GNN (SURVEY.md §2.6
+// K1-K10): fused 4-way embedding gather, block-diagonal CSR segment-sum
+// (message aggregation + its CSC transpose backward), fused GRU gates,
+// gated-attention segment softmax pooling, per-graph label max.
+//
+// Regime: the whole model is tiny (N ~= 12k nodes, D = 128 at batch 256), so
+// every kernel here is memory/latency-bound, never MFMA-bound; the design
+// rules that matter are coalescing (vectorized 2-element lanes: 64-lane wave
+// covers a 128-wide row), one wave per CSR row (avg degree ~2.5, max ~500),
+// and few kernels per step (fused gates) — the GEMMs go to rocBLAS via
+// torch.matmul, which is the right tool for plain 128x384 GEMMs.
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+#include <algorithm>
+using std::min;
+
+#define WAVE 64
+
+// ---------------------------------------------------------------------------
+// dtype helpers
+// ---------------------------------------------------------------------------
+
+template <typename T> __device__ __forceinline__ float to_f(T v);
+template <> __device__ __forceinline__ float to_f<float>(float v) { return v; }
+template <> __device__ __forceinline__ float to_f<__hip_bfloat16>(__hip_bfloat16 v) {
+  return __bfloat162float(v);
+}
+
+template <typename T> __device__ __forceinline__ T from_f(float v);
+template <> __device__ __forceinline__ float from_f<float>(float v) { return v; }
+template <> __device__ __forceinline__ __hip_bfloat16 from_f<__hip_bfloat16>(float v) {
+  return __float2bfloat16(v);
+}
+
+__device__ __forceinline__ float sigmoidf_(float x) { return 1.0f / (1.0f + __expf(-x)); }
+
+// ---------------------------------------------------------------------------
+// K1: fused 4-way embedding gather-concat.
+// tables (4, V, 32) T; idx (N, 4) int64 -> out (N, 128) T.
+// One wave per node; lane l writes out[n][2l..2l+1]; feature f = l/16.
+// ---------------------------------------------------------------------------
+
+template <typename T>
+__global__ void embed4_fwd_kernel(const T* __restrict__ tables,
+                                  const long* __restrict__ idx,
+                                  T* __restrict__ out, int N, int V) {
+  const int wid = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  if (wid >= N) return;
+  const int c = lane * 2;          // column in [0,128)
+  const int f = c >> 5;            // feature index
+  const int off = c & 31;          // offset within the 32-wide embedding
+  const long row = idx[(long)wid * 4 + f];
+  const T* src = tables + ((long)f * V + row) * 32 + off;
+  T* dst = out + (long)wid * 128 + c;
+  dst[0] = src[0];
+  dst[1] = src[1];
+}
+
+template <typename T>
+__global__ void embed4_bwd_kernel(const T* __restrict__ grad_out,
+                                  const long* __restrict__ idx,
+                                  float* __restrict__ grad_tables, long total,
+                                  int V) {
+  // grid-stride over N*128 elements; atomic fp32 scatter-add.
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    const long n = i >> 7;
+    const int c = (int)(i & 127);
+    const int f = c >> 5;
+    const int off = c & 31;
+    const long row = idx[n * 4 + f];
+    atomicAdd(grad_tables + ((long)f * V + row) * 32 + off, to_f(grad_out[i]));
+  }
+}
+
+// ---------------------------------------------------------------------------
+// K2: CSR segment-sum  m[v] = sum_{e in [indptr[v], indptr[v+1])} x[indices[e]].
+// One wave per destination row; lane covers 2 columns, strided by 128 for
+// D > 128. fp32 accumulation. Used for forward (in-CSR) and backward (CSC).
+// ---------------------------------------------------------------------------
+
+template <typename T>
+__global__ void spmm_sum_kernel(const int* __restrict__ indptr,
+                                const int* __restrict__ indices,
+                                const T* __restrict__ x, T* __restrict__ out,
+                                int N, int D) {
+  const int v = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  if (v >= N) return;
+  const int e0 = indptr[v], e1 = indptr[v + 1];
+  for (int c = lane * 2; c < D; c += WAVE * 2) {
+    float a0 = 0.f, a1 = 0.f;
+    for (int e = e0; e < e1; ++e) {
+      const T* row = x + (long)indices[e] * D + c;
+      a0 += to_f(row[0]);
+      a1 += to_f(row[1]);
+    }
+    T* dst = out + (long)v * D + c;
+    dst[0] = from_f<T>(a0);
+    dst[1] = from_f<T>(a1);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// K3: fused GRU gates (torch.nn.GRUCell semantics).
+// gi, gh (N, 3H); h (N, H) -> h_new, r, z, n (N, H). Elementwise, fp32 math.
+// ---------------------------------------------------------------------------
+
+template <typename T>
+__global__ void gru_gates_fwd_kernel(const T* __restrict__ gi,
+                                     const T* __restrict__ gh,
+                                     const T* __restrict__ h,
+                                     T* __restrict__ h_new, T* __restrict__ r_o,
+                                     T* __restrict__ z_o, T* __restrict__ n_o,
+                                     long NH, int H) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < NH;
+       i += (long)gridDim.x * blockDim.x) {
+    const long row = i / H;
+    const int c = (int)(i - row * H);
+    const long b = row * 3 * H + c;
+    const float ir = to_f(gi[b]), iz = to_f(gi[b + H]), in_ = to_f(gi[b + 2 * H]);
+    const float hr = to_f(gh[b]), hz = to_f(gh[b + H]), hn = to_f(gh[b + 2 * H]);
+    const float r = sigmoidf_(ir + hr);
+    const float z = sigmoidf_(iz + hz);
+    const float n = tanhf(in_ + r * hn);
+    const float hv = to_f(h[i]);
+    h_new[i] = from_f<T>((1.f - z) * n + z * hv);
+    r_o[i] = from_f<T>(r);
+    z_o[i] = from_f<T>(z);
+    n_o[i] = from_f<T>(n);
+  }
+}
+
+template <typename T>
+__global__ void gru_gates_bwd_kernel(const T* __restrict__ grad_h_new,
+                                     const T* __restrict__ gh,
+                                     const T* __restrict__ h,
+                                     const T* __restrict__ r_i,
+                                     const T* __restrict__ z_i,
+                                     const T* __restrict__ n_i,
+                                     T* __restrict__ grad_gi,
+                                     T* __restrict__ grad_gh,
+                                     T* __restrict__ grad_h, long NH, int H) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < NH;
+       i += (long)gridDim.x * blockDim.x) {
+    const long row = i / H;
+    const int c = (int)(i - row * H);
+    const long b = row * 3 * H + c;
+    const float go = to_f(grad_h_new[i]);
+    const float r = to_f(r_i[i]), z = to_f(z_i[i]), n = to_f(n_i[i]);
+    const float hv = to_f(h[i]);
+    const float hn = to_f(gh[b + 2 * H]);
+    const float dn = go * (1.f - z);
+    const float dz = go * (hv - n);
+    const float dpn = dn * (1.f - n * n);
+    const float dr = dpn * hn;
+    const float dpr = dr * r * (1.f - r);
+    const float dpz = dz * z * (1.f - z);
+    grad_gi[b] = from_f<T>(dpr);
+    grad_gi[b + H] = from_f<T>(dpz);
+    grad_gi[b + 2 * H] = from_f<T>(dpn);
+    grad_gh[b] = from_f<T>(dpr);
+    grad_gh[b + H] = from_f<T>(dpz);
+    grad_gh[b + 2 * H] = from_f<T>(dpn * r);
+    grad_h[i] = from_f<T>(go * z);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// K5: gated-attention pooling. One block (256 threads) per graph segment.
+//   fwd: alpha = softmax(gate[seg]); out[g] = sum_v alpha_v * x_v
+//   bwd: grad_x = alpha * grad_out[g];  grad_gate = alpha*(s - <alpha,s>),
+//        s_v = <grad_out[g], x_v>.
+// Block-level reductions through one LDS scratch array.
+// ---------------------------------------------------------------------------
+
+__device__ float block_reduce(float v, float* scratch, int op /*0=max,1=sum*/) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int w = threadIdx.x / WAVE;
+  // wave reduce
+  for (int off = WAVE / 2; off > 0; off >>= 1) {
+    const float o = __shfl_down(v, off);
+    v = op == 0 ? fmaxf(v, o) : v + o;
+  }
+  if (lane == 0) scratch[w] = v;
+  __syncthreads();
+  const int nw = blockDim.x / WAVE;
+  float out = scratch[0];
+  if (threadIdx.x == 0) {
+    for (int i = 1; i < nw; ++i) out = op == 0 ? fmaxf(out, scratch[i]) : out + scratch[i];
+    scratch[0] = out;
+  }
+  __syncthreads();
+  out = scratch[0];
+  __syncthreads();
+  return out;
+}
+
+template <typename T>
+__global__ void attn_pool_fwd_kernel(const T* __restrict__ x,
+                                     const T* __restrict__ gate,
+                                     const int* __restrict__ node_offsets,
+                                     T* __restrict__ out,
+                                     float* __restrict__ alpha, int D) {
+  __shared__ float scratch[8];
+  const int g = blockIdx.x;
+  const int lo = node_offsets[g], hi = node_offsets[g + 1];
+  const int n = hi - lo;
+  if (n <= 0) return;
+  // phase A: segment max of gate
+  float m = -3.4e38f;
+  for (int v = threadIdx.x; v < n; v += blockDim.x) m = fmaxf(m, to_f(gate[lo + v]));
+  m = block_reduce(m, scratch, 0);
+  // phase B: exp/sum -> alpha
+  float s = 0.f;
+  for (int v = threadIdx.x; v < n; v += blockDim.x) {
+    const float e = __expf(to_f(gate[lo + v]) - m);
+    alpha[lo + v] = e;
+    s += e;
+  }
+  s = block_reduce(s, scratch, 1);
+  const float inv = 1.0f / s;
+  for (int v = threadIdx.x; v < n; v += blockDim.x) alpha[lo + v] *= inv;
+  __syncthreads();
+  // phase C: weighted segment sum, one output column per thread (strided)
+  for (int d = threadIdx.x; d < D; d += blockDim.x) {
+    float acc = 0.f;
+    for (int v = 0; v < n; ++v) acc += alpha[lo + v] * to_f(x[(long)(lo + v) * D + d]);
+    out[(long)g * D + d] = from_f<T>(acc);
+  }
+}
+
+template <typename T>
+__global__ void attn_pool_bwd_kernel(const T* __restrict__ grad_out,
+                                     const T* __restrict__ x,
+                                     const float* __restrict__ alpha,
+                                     const int* __restrict__ node_offsets,
+                                     T* __restrict__ grad_x,
+                                     T* __restrict__ grad_gate,
+                                     float* __restrict__ s_ws, int D) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* go = reinterpret_cast<float*>(smem);          // D floats
+  float* scratch = go + D;                             // 8 floats
+  const int g = blockIdx.x;
+  const int lo = node_offsets[g], hi = node_offsets[g + 1];
+  const int n = hi - lo;
+  if (n <= 0) return;
+  for (int d = threadIdx.x; d < D; d += blockDim.x)
+    go[d] = to_f(grad_out[(long)g * D + d]);
+  __syncthreads();
+  // pass 1: s_v = <go, x_v> (one wave per node), accumulate dot = sum alpha*s
+  const int wid = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int nw = blockDim.x / WAVE;
+  float local_dot = 0.f;
+  for (int v = wid; v < n; v += nw) {
+    float sv = 0.f;
+    for (int d = lane; d < D; d += WAVE) sv += go[d] * to_f(x[(long)(lo + v) * D + d]);
+    for (int off = WAVE / 2; off > 0; off >>= 1) sv += __shfl_down(sv, off);
+    sv = __shfl(sv, 0);
+    if (lane == 0) {
+      s_ws[lo + v] = sv;
+      local_dot += alpha[lo + v] * sv;
+    }
+  }
+  __syncthreads();
+  const float dot = block_reduce(lane == 0 ? local_dot : 0.f, scratch, 1);
+  // pass 2: outputs
+  for (int v = wid; v < n; v += nw) {
+    const float a = alpha[lo + v];
+    for (int d = lane; d < D; d += WAVE)
+      grad_x[(long)(lo + v) * D + d] = from_f<T>(a * go[d]);
+    if (lane == 0) grad_gate[lo + v] = from_f<T>(a * (s_ws[lo + v] - dot));
+  }
+}
+
+// ---------------------------------------------------------------------------
+// K7: per-graph max of node values (graph label reduction).
+// ---------------------------------------------------------------------------
+
+__global__ void segment_max_kernel(const float* __restrict__ values,
+                                   const int* __restrict__ node_offsets,
+                                   float* __restrict__ out) {
+  __shared__ float scratch[8];
+  const int g = blockIdx.x;
+  const int lo = node_offsets[g], hi = node_offsets[g + 1];
+  float m = -3.4e38f;
+  for (int v = lo + threadIdx.x; v < hi; v += blockDim.x) m = fmaxf(m, values[v]);
+  m = block_reduce(m, scratch, 0);
+  if (threadIdx.x == 0) out[g] = (hi > lo) ? m : 0.f;
+}
+
+// ---------------------------------------------------------------------------
+// Launchers (called from bindings.cpp)
+// ---------------------------------------------------------------------------
+
+template <typename T>
+void launch_embed4_fwd(const T* tables, const long* idx, T* out, int N, int V,
+                       hipStream_t stream) {
+  const int waves_per_block = 4;
+  const int block = WAVE * waves_per_block;
+  const int grid = (N + waves_per_block - 1) / waves_per_block;
+  if (grid > 0)
+    hipLaunchKernelGGL(embed4_fwd_kernel<T>, dim3(grid), dim3(block), 0, stream,
+                       tables, idx, out, N, V);
+}
+
+template <typename T>
+void launch_embed4_bwd(const T* grad_out, const long* idx, float* grad_tables,
+                       long total, int V, hipStream_t stream) {
+  const int block = 256;
+  const int grid = (int)min((total + block - 1) / block, (long)2048);
+  if (grid > 0)
+    hipLaunchKernelGGL(embed4_bwd_kernel<T>, dim3(grid), dim3(block), 0, stream,
+                       grad_out, idx, grad_tables, total, V);
+}
+
+template <typename T>
+void launch_spmm_sum(const int* indptr, const int* indices, const T* x, T* out,
+                     int N, int D, hipStream_t stream) {
+  const int waves_per_block = 4;
+  const int block = WAVE * waves_per_block;
+  const int grid = (N + waves_per_block - 1) / waves_per_block;
+  if (grid > 0)
+    hipLaunchKernelGGL(spmm_sum_kernel<T>, dim3(grid), dim3(block), 0, stream,
+                       indptr, indices, x, out, N, D);
+}
+
+template <typename T>
+void launch_gru_gates_fwd(const T* gi, const T* gh, const T* h, T* h_new, T* r,
+                          T* z, T* n, long NH, int H, hipStream_t stream) {
+  const int block = 256;
+  const int grid = (int)min((NH + block - 1) / block, (long)2048);
+  if (grid > 0)
+    hipLaunchKernelGGL(gru_gates_fwd_kernel<T>, dim3(grid), dim3(block), 0,
+                       stream, gi, gh, h, h_new, r, z, n, NH, H);
+}
+
+template <typename T>
+void launch_gru_gates_bwd(const T* grad_h_new, const T* gh, const T* h,
+                          const T* r, const T* z, const T* n, T* grad_gi,
+                          T* grad_gh, T* grad_h, long NH, int H,
+                          hipStream_t stream) {
+  const int block = 256;
+  const int grid = (int)min((NH + block - 1) / block, (long)2048);
+  if (grid > 0)
+    hipLaunchKernelGGL(gru_gates_bwd_kernel<T>, dim3(grid), dim3(block), 0,
+                       stream, grad_h_new, gh, h, r, z, n, grad_gi, grad_gh,
+                       grad_h, NH, H);
+}
+
+template <typename T>
+void launch_attn_pool_fwd(const T* x, const T* gate, const int* node_offsets,
+                          T* out, float* alpha, int B, int D,
+                          hipStream_t stream) {
+  if (B > 0)
+    hipLaunchKernelGGL(attn_pool_fwd_kernel<T>, dim3(B), dim3(256), 0, stream,
+                       x, gate, node_offsets, out, alpha, D);
+}
+
+template <typename T>
+void launch_attn_pool_bwd(const T* grad_out, const T* x, const float* alpha,
+                          const int* node_offsets, T* grad_x, T* grad_gate,
+                          float* s_ws, int B, int D, hipStream_t stream) {
+  const size_t lds = (D + 8) * sizeof(float);
+  if (B > 0)
+    hipLaunchKernelGGL(attn_pool_bwd_kernel<T>, dim3(B), dim3(256), lds, stream,
+                       grad_out, x, alpha, node_offsets, grad_x, grad_gate,
+                       s_ws, D);
+}
+
+void launch_segment_max(const float* values, const int* node_offsets,
+                        float* out, int B, hipStream_t stream) {
+  if (B > 0)
+    hipLaunchKernelGGL(segment_max_kernel, dim3(B), dim3(256), 0, stream,
+                       values, node_offsets, out);
+}
+
+// explicit instantiations
+#define INSTANTIATE(T)                                                        \
+  template void launch_embed4_fwd<T>(const T*, const long*, T*, int, int,     \
+                                     hipStream_t);                            \
+  template void launch_embed4_bwd<T>(const T*, const long*, float*, long,     \
+                                     int, hipStream_t);                       \
+  template void launch_spmm_sum<T>(const int*, const int*, const T*, T*, int, \
+                                   int, hipStream_t);                         \
+  template void launch_gru_gates_fwd<T>(const T*, const T*, const T*, T*, T*, \
+                                        T*, T*, long, int, hipStream_t);      \
+  template void launch_gru_gates_bwd<T>(const T*, const T*, const T*,         \
+                                        const T*, const T*, const T*, T*, T*, \
+                                        T*, long, int, hipStream_t);          \
+  template void launch_attn_pool_fwd<T>(const T*, const T*, const int*, T*,   \
+                                        float*, int, int, hipStream_t);       \
+  template void launch_attn_pool_bwd<T>(const T*, const T*, const float*,     \
+                                        const int*, T*, T*, float*, int, int, \
+                                        hipStream_t);
+
+INSTANTIATE(float)
+INSTANTIATE(__hip_bfloat16)

@@ -1,0 +1,101 @@
+"""Datamodule: train/val/test loaders over the graph dataset.
+
+Parity target: reference sastvd/linevd/datamodule.py:17-141
+(BigVulDatasetLineVDDataModule): exposes input_dim = limit_all + 2
+(datamodule.py:89-96), positive_weight = n_neg/n_pos of the train split
+(:99-108), per-epoch undersampled Subset loaders (:110-129), and duplicate
+split guards (:74-79). Built on torch DataLoader + our collate (no DGL
+GraphDataLoader)."""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+from torch.utils.data import DataLoader, Subset
+
+from .dataset import BigVulDatasetLineVD, collate_graphs
+from .features import parse_limits
+
+
+class BigVulDatasetLineVDDataModule:
+    def __init__(
+        self,
+        feat: str = "_ABS_DATAFLOW_datatype_all_limitall_1000_limitsubkeys_1000",
+        gtype: str = "cfg",
+        dsname: str = "bigvul",
+        undersample=None,
+        split: str = "fixed",
+        batch_size: int = 256,
+        sample_mode: bool = False,
+        train_workers: int = 0,
+        seed: int = 0,
+        n_synthetic: int = 2000,
+        graph_dir: Optional[str] = None,
+        train_includes_all: bool = False,
+    ):
+        self.feat = feat
+        self.batch_size = batch_size
+        self.train_workers = train_workers
+        common = dict(
+            feat=feat,
+            gtype=gtype,
+            dsname=dsname,
+            split=split,
+            sample_mode=sample_mode,
+            seed=seed,
+            n_synthetic=n_synthetic,
+            graph_dir=graph_dir,
+        )
+        if train_includes_all:
+            # combined-model mode (linevul_main.py:548-575): one dataset
+            # containing every example, indexed by id via get_indices.
+            self.train = BigVulDatasetLineVD(
+                partition="all", undersample=undersample, **common
+            )
+        else:
+            self.train = BigVulDatasetLineVD(
+                partition="train", undersample=undersample, **common
+            )
+        self.val = BigVulDatasetLineVD(partition="val", **common)
+        self.test = BigVulDatasetLineVD(partition="test", **common)
+        if not sample_mode and not train_includes_all:
+            tr = set(self.train.df.id)
+            va, te = set(self.val.df.id), set(self.test.df.id)
+            assert not (tr & va) and not (tr & te) and not (va & te), "split leak"
+
+    @property
+    def input_dim(self) -> int:
+        return parse_limits(self.feat).input_dim
+
+    @property
+    def positive_weight(self) -> float:
+        vc = self.train.df.vul.value_counts()
+        n_pos = int(vc.get(1, 0))
+        n_neg = int(vc.get(0, 0))
+        return (n_neg / n_pos) if n_pos else 1.0
+
+    def _loader(self, ds, shuffle: bool, epoch_subset: bool = False, generator=None):
+        dataset = ds
+        if epoch_subset and (ds.undersample is not None or ds.oversample is not None):
+            dataset = Subset(ds, ds.get_epoch_indices().tolist())
+        return DataLoader(
+            dataset,
+            batch_size=self.batch_size,
+            shuffle=shuffle,
+            num_workers=self.train_workers,
+            collate_fn=collate_graphs,
+            generator=generator,
+            drop_last=False,
+        )
+
+    def train_dataloader(self, generator: Optional[torch.Generator] = None):
+        """Reconstructed every epoch (reload_dataloaders_every_n_epochs: 1)
+        so the undersampled subset reshuffles per epoch."""
+        return self._loader(self.train, shuffle=True, epoch_subset=True, generator=generator)
+
+    def val_dataloader(self):
+        return self._loader(self.val, shuffle=False)
+
+    def test_dataloader(self):
+        return self._loader(self.test, shuffle=False)

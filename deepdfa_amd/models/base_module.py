@@ -1,0 +1,147 @@
+"""BaseModule: shared train/val/test-step behavior for graph models.
+
+Parity target: reference DDFA/code_gnn/models/base_module.py:26-383 —
+BCEWithLogits loss with optional pos_weight, graph-level label reduction
+(per-graph max of node _VULN), threshold-0.5 metric collections per split
+(plus positive-only / negative-only test clones), PR-curve export and a
+classification report at test end. Implemented Lightning-free on top of
+our trainer (deepdfa_amd/train/trainer.py); metrics are counter-based so
+DDP aggregation is one all_reduce.
+"""
+
+from __future__ import annotations
+
+from typing import Dict, Optional
+
+import torch
+from torch import nn
+
+from ..ops import segment_max
+from ..utils.metrics import BinaryStats, classification_report_dict, pr_curve
+
+
+class BaseModule(nn.Module):
+    def __init__(
+        self,
+        undersample_node_on_loss_factor=None,
+        test_every: bool = False,
+        tune_nni: bool = False,
+        positive_weight: Optional[float] = None,
+        profile: bool = False,
+        time: bool = False,
+        label_style: str = "graph",
+    ):
+        super().__init__()
+        self.class_threshold = 0.5
+        self.label_style = label_style
+        self.hparams: Dict = dict(
+            undersample_node_on_loss_factor=undersample_node_on_loss_factor,
+            test_every=test_every,
+            tune_nni=tune_nni,
+            positive_weight=positive_weight,
+            profile=profile,
+            time=time,
+            label_style=label_style,
+        )
+        self.metrics = {
+            "train": BinaryStats(self.class_threshold),
+            "val": BinaryStats(self.class_threshold),
+            "test": BinaryStats(self.class_threshold),
+            "test_1": BinaryStats(self.class_threshold),
+            "test_0": BinaryStats(self.class_threshold),
+        }
+        if positive_weight is not None:
+            self.register_buffer("pos_weight", torch.tensor([float(positive_weight)]))
+        else:
+            self.pos_weight = None
+        self._test_preds: list = []
+        self._test_labels: list = []
+
+    # -- labels --------------------------------------------------------------
+
+    def get_label(self, graph) -> torch.Tensor:
+        if self.label_style == "node":
+            label = graph.ndata["_VULN"]
+        elif self.label_style == "graph":
+            label = segment_max(graph.ndata["_VULN"].float(), graph)
+        else:
+            raise NotImplementedError(self.label_style)
+        return label.float()
+
+    # -- loss -----------------------------------------------------------------
+
+    def loss_fn(self, logits: torch.Tensor, label: torch.Tensor) -> torch.Tensor:
+        return torch.nn.functional.binary_cross_entropy_with_logits(
+            logits, label, pos_weight=self.pos_weight
+        )
+
+    # -- steps (called by the trainer) ---------------------------------------
+
+    def training_step(self, batch) -> torch.Tensor:
+        graph, extrafeats = batch if isinstance(batch, tuple) else (batch, {})
+        label = self.get_label(graph)
+        logits = self(graph, extrafeats)
+        loss = self.loss_fn(logits.float(), label)
+        with torch.no_grad():
+            self.metrics["train"].to(logits.device).update(torch.sigmoid(logits.float()), label)
+        return loss
+
+    @torch.no_grad()
+    def validation_step(self, batch) -> torch.Tensor:
+        graph, extrafeats = batch if isinstance(batch, tuple) else (batch, {})
+        label = self.get_label(graph)
+        logits = self(graph, extrafeats)
+        loss = self.loss_fn(logits.float(), label)
+        self.metrics["val"].to(logits.device).update(torch.sigmoid(logits.float()), label)
+        return loss
+
+    @torch.no_grad()
+    def test_step(self, batch) -> torch.Tensor:
+        graph, extrafeats = batch if isinstance(batch, tuple) else (batch, {})
+        label = self.get_label(graph)
+        logits = self(graph, extrafeats)
+        loss = self.loss_fn(logits.float(), label)
+        probs = torch.sigmoid(logits.float())
+        self.metrics["test"].to(logits.device).update(probs, label)
+        pos, neg = label == 1, label == 0
+        if pos.any():
+            self.metrics["test_1"].to(logits.device).update(probs[pos], label[pos])
+        if neg.any():
+            self.metrics["test_0"].to(logits.device).update(probs[neg], label[neg])
+        self._test_preds.append(probs.detach().cpu())
+        self._test_labels.append(label.detach().cpu())
+        return loss
+
+    # -- epoch end ------------------------------------------------------------
+
+    def epoch_metrics(self, split: str, reset: bool = True) -> Dict[str, float]:
+        stats = self.metrics[split]
+        stats.all_reduce()
+        out = {f"{split}_{k}": v for k, v in stats.compute().items()}
+        if reset:
+            stats.reset()
+        return out
+
+    def test_epoch_end(self, out_dir: Optional[str] = None) -> Dict:
+        """Compute test metrics + PR curve + classification report, optionally
+        exporting pr.csv (base_module.py:358-361 contract)."""
+        results = {}
+        for split in ("test", "test_1", "test_0"):
+            results.update(self.epoch_metrics(split, reset=False))
+        if self._test_preds:
+            probs = torch.cat(self._test_preds)
+            labels = torch.cat(self._test_labels)
+            ths, precs, recs = pr_curve(probs, labels)
+            results["classification_report"] = classification_report_dict(probs, labels)
+            if out_dir is not None:
+                import csv
+                import os
+
+                with open(os.path.join(out_dir, "pr.csv"), "w", newline="") as f:
+                    w = csv.writer(f)
+                    w.writerow(["threshold", "precision", "recall"])
+                    for row in zip(ths, precs, recs):
+                        w.writerow(row)
+        self._test_preds.clear()
+        self._test_labels.clear()
+        return results

@@ -1,0 +1,171 @@
+"""Autograd ops for the flow-GNN (K1-K10 of SURVEY.md §2.6).
+
+Each op is a torch.autograd.Function that dispatches to the hand-written
+HIP/CDNA4 kernel on GPU (required — no eager fallback on GPU) and to the
+fp32 torch reference on CPU. GEMMs that are plain library GEMMs (the GRU's
+input/hidden projections, the MLP) go through torch.matmul (rocBLAS /
+hipBLASLt on ROCm); everything irregular (gather, segment ops, fused gates)
+is a custom kernel.
+"""
+
+from __future__ import annotations
+
+from typing import List, Optional
+
+import torch
+
+from . import reference as ref
+from ._ext import ext_for
+
+
+class _Embed4(torch.autograd.Function):
+    """K1: fused 4-way embedding gather-concat. tables (4, V, Demb); idx (N, 4)."""
+
+    @staticmethod
+    def forward(ctx, tables: torch.Tensor, idx: torch.Tensor) -> torch.Tensor:
+        ctx.save_for_backward(idx)
+        ctx.tables_shape = tables.shape
+        ext = ext_for(tables)
+        if ext is not None:
+            return ext.embed4_fwd(tables, idx)
+        return ref.embed4_fwd(list(tables), idx)
+
+    @staticmethod
+    def backward(ctx, grad_out: torch.Tensor):
+        (idx,) = ctx.saved_tensors
+        F, V, Demb = ctx.tables_shape
+        ext = ext_for(grad_out)
+        if ext is not None:
+            grad_tables = ext.embed4_bwd(grad_out.contiguous(), idx, V, Demb)
+        else:
+            grad_tables = torch.stack(
+                ref.embed4_bwd(grad_out.float(), idx, V, Demb)
+            ).to(grad_out.dtype)
+        return grad_tables, None
+
+
+def embed4(tables: torch.Tensor, idx: torch.Tensor) -> torch.Tensor:
+    return _Embed4.apply(tables, idx)
+
+
+class _SpmmSum(torch.autograd.Function):
+    """K2 aggregation: m[v] = sum_{u->v} x[u] over the block-diagonal CSR.
+    Backward gathers through the transpose (CSC)."""
+
+    @staticmethod
+    def forward(
+        ctx,
+        x: torch.Tensor,
+        indptr: torch.Tensor,
+        indices: torch.Tensor,
+        t_indptr: torch.Tensor,
+        t_indices: torch.Tensor,
+    ) -> torch.Tensor:
+        ctx.save_for_backward(t_indptr, t_indices)
+        ext = ext_for(x)
+        if ext is not None:
+            return ext.spmm_sum(indptr, indices, x.contiguous())
+        return ref.spmm_sum(indptr, indices, x)
+
+    @staticmethod
+    def backward(ctx, grad_out: torch.Tensor):
+        t_indptr, t_indices = ctx.saved_tensors
+        ext = ext_for(grad_out)
+        if ext is not None:
+            gx = ext.spmm_sum(t_indptr, t_indices, grad_out.contiguous())
+        else:
+            gx = ref.spmm_sum(t_indptr, t_indices, grad_out)
+        return gx, None, None, None, None
+
+
+def spmm_sum(x, graph) -> torch.Tensor:
+    return _SpmmSum.apply(x, graph.indptr, graph.indices, graph.t_indptr, graph.t_indices)
+
+
+class _GruGates(torch.autograd.Function):
+    """K3 fused GRU gate elementwise (the GEMMs producing gi/gh are matmuls)."""
+
+    @staticmethod
+    def forward(ctx, gi: torch.Tensor, gh: torch.Tensor, h: torch.Tensor) -> torch.Tensor:
+        ext = ext_for(gi)
+        if ext is not None:
+            h_new, r, z, n = ext.gru_gates_fwd(gi.contiguous(), gh.contiguous(), h.contiguous())
+        else:
+            h_new, r, z, n = ref.gru_gates_fwd(gi, gh, h)
+        ctx.save_for_backward(gh, h, r, z, n)
+        return h_new
+
+    @staticmethod
+    def backward(ctx, grad_h_new: torch.Tensor):
+        gh, h, r, z, n = ctx.saved_tensors
+        ext = ext_for(grad_h_new)
+        if ext is not None:
+            grad_gi, grad_gh, grad_h = ext.gru_gates_bwd(
+                grad_h_new.contiguous(), gh, h, r, z, n
+            )
+        else:
+            grad_gi, grad_gh, grad_h = ref.gru_gates_bwd(grad_h_new, gh, h, r, z, n)
+        return grad_gi, grad_gh, grad_h
+
+
+def gru_cell(
+    a: torch.Tensor,
+    h: torch.Tensor,
+    w_ih: torch.Tensor,
+    w_hh: torch.Tensor,
+    b_ih: Optional[torch.Tensor],
+    b_hh: Optional[torch.Tensor],
+) -> torch.Tensor:
+    """torch.nn.GRUCell-compatible: a is the input (aggregated message)."""
+    gi = a @ w_ih.t()
+    gh = h @ w_hh.t()
+    if b_ih is not None:
+        gi = gi + b_ih.to(gi.dtype)
+    if b_hh is not None:
+        gh = gh + b_hh.to(gh.dtype)
+    # under bf16 autocast the matmuls emit bf16 while h may still be fp32
+    return _GruGates.apply(gi, gh, h.to(gi.dtype))
+
+
+class _AttnPool(torch.autograd.Function):
+    """K5: segment softmax over gate logits + weighted segment sum."""
+
+    @staticmethod
+    def forward(
+        ctx, x: torch.Tensor, gate_logits: torch.Tensor, node_offsets: torch.Tensor
+    ) -> torch.Tensor:
+        ext = ext_for(x)
+        if ext is not None:
+            out, alpha = ext.attn_pool_fwd(
+                x.contiguous(), gate_logits.contiguous(), node_offsets
+            )
+        else:
+            out, alpha = ref.attn_pool_fwd(x, gate_logits, node_offsets)
+        ctx.save_for_backward(x, alpha, node_offsets)
+        return out
+
+    @staticmethod
+    def backward(ctx, grad_out: torch.Tensor):
+        x, alpha, node_offsets = ctx.saved_tensors
+        ext = ext_for(grad_out)
+        if ext is not None:
+            grad_x, grad_gate = ext.attn_pool_bwd(
+                grad_out.contiguous(), x, alpha, node_offsets
+            )
+        else:
+            grad_x, grad_gate = ref.attn_pool_bwd(grad_out, x, alpha, node_offsets)
+        return grad_x, grad_gate, None
+
+
+def attn_pool(x: torch.Tensor, gate_logits: torch.Tensor, graph) -> torch.Tensor:
+    node_offsets = graph.node_offsets
+    return _AttnPool.apply(x, gate_logits, node_offsets)
+
+
+def segment_max(values: torch.Tensor, graph) -> torch.Tensor:
+    """K7 label reduction (no grad)."""
+    ext = ext_for(values) if values.is_floating_point() or values.is_cuda else None
+    if values.is_cuda:
+        ext = ext_for(values)
+        return ext.segment_max(values.contiguous(), graph.node_offsets)
+    return ref.segment_max(values, graph.node_offsets)

@@ -1,0 +1,67 @@
+"""Trainer loop tests: fit/val/test, checkpoint naming + best selection."""
+
+import glob
+import os
+
+import torch
+
+from deepdfa_amd.data import BigVulDatasetLineVDDataModule
+from deepdfa_amd.models import FlowGNNGGNNModule
+from deepdfa_amd.train import Trainer
+
+FEAT = "_ABS_DATAFLOW_datatype_all_limitall_1000_limitsubkeys_1000"
+
+
+def small_setup(tmp_path, epochs=2):
+    dm = BigVulDatasetLineVDDataModule(
+        feat=FEAT, batch_size=64, n_synthetic=300, undersample="v1.0"
+    )
+    torch.manual_seed(0)
+    model = FlowGNNGGNNModule(
+        feat=FEAT, input_dim=dm.input_dim, hidden_dim=32, n_steps=5, num_output_layers=3
+    )
+    tr = Trainer(max_epochs=epochs, default_root_dir=str(tmp_path), periodic_every=2)
+    return dm, model, tr
+
+
+def test_fit_creates_checkpoints(tmp_path):
+    dm, model, tr = small_setup(tmp_path)
+    out = tr.fit(model, dm)
+    assert len(out["history"]) == 2
+    ckpts = os.listdir(tr.ckpt_dir)
+    assert "last.ckpt" in ckpts
+    assert any(c.startswith("performance-") for c in ckpts)
+    assert any(c.startswith("periodical-") for c in ckpts)
+    best = out["best_checkpoint"]
+    assert best is not None and os.path.exists(best)
+    # only one performance ckpt kept (save_top_k=1)
+    assert len(glob.glob(os.path.join(tr.ckpt_dir, "performance-*.ckpt"))) == 1
+
+
+def test_checkpoint_roundtrip(tmp_path):
+    dm, model, tr = small_setup(tmp_path, epochs=1)
+    out = tr.fit(model, dm)
+    model2 = FlowGNNGGNNModule(
+        feat=FEAT, input_dim=dm.input_dim, hidden_dim=32, n_steps=5, num_output_layers=3
+    )
+    payload = tr.load_checkpoint(model2, os.path.join(tr.ckpt_dir, "last.ckpt"))
+    assert payload["epoch"] == 0
+    for (n1, p1), (n2, p2) in zip(model.named_parameters(), model2.named_parameters()):
+        assert n1 == n2
+        assert torch.equal(p1, p2)
+
+
+def test_test_loop_metrics(tmp_path):
+    dm, model, tr = small_setup(tmp_path, epochs=1)
+    tr.fit(model, dm)
+    res = tr.test(model, dm)
+    assert "test_f1" in res and "test_1_recall" in res
+    assert os.path.exists(os.path.join(str(tmp_path), "pr.csv"))
+    assert "classification_report" in res
+
+
+def test_loss_decreases(tmp_path):
+    dm, model, tr = small_setup(tmp_path, epochs=4)
+    out = tr.fit(model, dm)
+    losses = [h["train_loss"] for h in out["history"]]
+    assert losses[-1] < losses[0]
