@@ -24,6 +24,15 @@ template <typename T>
 void launch_attn_pool_bwd(const T*, const T*, const float*, const int*, T*, T*, float*, int, int,
                           hipStream_t);
 void launch_segment_max(const float*, const int*, float*, int, hipStream_t);
+template <typename T>
+void launch_gru_gates2_fwd(const T*, const T*, T*, T*, T*, T*, T*, long, int, hipStream_t);
+template <typename T>
+void launch_gru_gates2_bwd(const T*, const T*, const T*, const T*, const T*, const T*, T*, T*,
+                           long, int, hipStream_t);
+template <typename T>
+void launch_colsum(const T*, float*, int, int, hipStream_t);
+void launch_gemm_bias(const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*,
+                      const __hip_bfloat16*, __hip_bfloat16*, int, int, int, int, hipStream_t);
 
 #define CHECK_GPU(t) \
   TORCH_CHECK((t).is_cuda() && (t).is_contiguous(), #t " must be contiguous GPU tensor")
@@ -186,6 +195,164 @@ at::Tensor segment_max(at::Tensor values, at::Tensor node_offsets) {
   return out;
 }
 
+// ---------------------------------------------------------------------------
+// Fused GGNN: the whole n_steps message-passing loop driven from C++.
+// Replaces {linear, spmm, GRUCell} x n_steps of the reference's DGL
+// GatedGraphConv (ggnn.py:57-60,95) with per step:
+//   wh    = gemm_bias(h, W_e^T) + b_e                      [MFMA]
+//   m     = csr_segment_sum(wh)                            [spmm kernel]
+//   gicat = gemm_bias([m|h], Wcat^T) + b_cat               [MFMA, split-A]
+//   h     = fused_gru_gates(gicat, h)                      [elementwise]
+// Wcat is the (4H, 2H) block weight matrix [W_ir|W_hr; W_iz|W_hz; W_in|0;
+// 0|W_hn] so one GEMM produces all gate pre-activations.
+// ---------------------------------------------------------------------------
+
+using bf16_t = __hip_bfloat16;
+
+static at::Tensor build_wcat(const at::Tensor& W_ih, const at::Tensor& W_hh, long H) {
+  auto Wcat = at::zeros({4 * H, 2 * H}, W_ih.options());
+  Wcat.narrow(0, 0, H).narrow(1, 0, H).copy_(W_ih.narrow(0, 0, H));
+  Wcat.narrow(0, 0, H).narrow(1, H, H).copy_(W_hh.narrow(0, 0, H));
+  Wcat.narrow(0, H, H).narrow(1, 0, H).copy_(W_ih.narrow(0, H, H));
+  Wcat.narrow(0, H, H).narrow(1, H, H).copy_(W_hh.narrow(0, H, H));
+  Wcat.narrow(0, 2 * H, H).narrow(1, 0, H).copy_(W_ih.narrow(0, 2 * H, H));
+  Wcat.narrow(0, 3 * H, H).narrow(1, H, H).copy_(W_hh.narrow(0, 2 * H, H));
+  return Wcat;
+}
+
+at::Tensor gemm_bias(at::Tensor A, at::Tensor W, c10::optional<at::Tensor> bias) {
+  CHECK_GPU(A);
+  CHECK_GPU(W);
+  TORCH_CHECK(A.scalar_type() == at::kBFloat16 && W.scalar_type() == at::kBFloat16);
+  const int N = A.size(0), K = A.size(1), COL = W.size(0);
+  TORCH_CHECK(W.size(1) == K && K % 64 == 0 && COL % 128 == 0);
+  auto out = at::empty({N, COL}, A.options());
+  const bf16_t* b = nullptr;
+  if (bias.has_value()) b = reinterpret_cast<const bf16_t*>(bias->data_ptr());
+  launch_gemm_bias(ptr<bf16_t>(A), nullptr, ptr<bf16_t>(W), b, mptr<bf16_t>(out), N, K, K, COL,
+                   cur_stream());
+  return out;
+}
+
+std::vector<at::Tensor> ggnn_fused_fwd(at::Tensor indptr, at::Tensor indices, at::Tensor x,
+                                       at::Tensor W_e, at::Tensor b_e, at::Tensor W_ih,
+                                       at::Tensor W_hh, at::Tensor b_ih, at::Tensor b_hh,
+                                       long n_steps) {
+  CHECK_GPU(x);
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16, "fused GGNN path is bf16");
+  const long N = x.size(0);
+  const long H = x.size(1);
+  TORCH_CHECK(H % 64 == 0 && (4 * H) % 128 == 0, "H must suit the MFMA tile");
+  auto stream = cur_stream();
+  auto Wcat = build_wcat(W_ih, W_hh, H);
+  auto b_cat = at::cat({b_ih.narrow(0, 0, 2 * H) + b_hh.narrow(0, 0, 2 * H),
+                        b_ih.narrow(0, 2 * H, H), b_hh.narrow(0, 2 * H, H)});
+  const long S = n_steps;
+  auto opts = x.options();
+  auto M = at::empty({S, N, H}, opts);
+  auto R = at::empty({S, N, H}, opts);
+  auto Z = at::empty({S, N, H}, opts);
+  auto Nn = at::empty({S, N, H}, opts);
+  auto HN = at::empty({S, N, H}, opts);
+  auto Hnew = at::empty({S, N, H}, opts);
+  auto wh = at::empty({N, H}, opts);
+  auto gicat = at::empty({N, 4 * H}, opts);
+  const long NH = N * H;
+  const bf16_t* h = ptr<bf16_t>(x);
+  for (long s = 0; s < S; ++s) {
+    launch_gemm_bias(h, nullptr, ptr<bf16_t>(W_e), ptr<bf16_t>(b_e), mptr<bf16_t>(wh), N, H, H, H,
+                     stream);
+    bf16_t* m = mptr<bf16_t>(M) + s * NH;
+    launch_spmm_sum<bf16_t>(indptr.data_ptr<int>(), indices.data_ptr<int>(), ptr<bf16_t>(wh), m,
+                            N, H, stream);
+    launch_gemm_bias(m, h, ptr<bf16_t>(Wcat), ptr<bf16_t>(b_cat), mptr<bf16_t>(gicat), N, 2 * H,
+                     H, 4 * H, stream);
+    launch_gru_gates2_fwd<bf16_t>(ptr<bf16_t>(gicat), h, mptr<bf16_t>(Hnew) + s * NH,
+                                  mptr<bf16_t>(R) + s * NH, mptr<bf16_t>(Z) + s * NH,
+                                  mptr<bf16_t>(Nn) + s * NH, mptr<bf16_t>(HN) + s * NH, NH, H,
+                                  stream);
+    h = ptr<bf16_t>(Hnew) + s * NH;
+  }
+  auto h_final = Hnew.select(0, S - 1);
+  return {h_final, M, R, Z, Nn, HN, Hnew};
+}
+
+std::vector<at::Tensor> ggnn_fused_bwd(at::Tensor grad_out, at::Tensor t_indptr,
+                                       at::Tensor t_indices, at::Tensor x, at::Tensor W_e,
+                                       at::Tensor W_ih, at::Tensor W_hh, at::Tensor M,
+                                       at::Tensor R, at::Tensor Z, at::Tensor Nn, at::Tensor HN,
+                                       at::Tensor Hnew, long n_steps) {
+  const long N = x.size(0);
+  const long H = x.size(1);
+  const long S = n_steps;
+  const long NH = N * H;
+  auto stream = cur_stream();
+  auto Wcat = build_wcat(W_ih, W_hh, H);
+  auto opts = x.options();
+  auto gW_e = at::zeros({H, H}, opts.dtype(at::kFloat));
+  auto gW_ih = at::zeros({3 * H, H}, opts.dtype(at::kFloat));
+  auto gW_hh = at::zeros({3 * H, H}, opts.dtype(at::kFloat));
+  auto cs4 = at::zeros({4 * H}, opts.dtype(at::kFloat));
+  auto cs_e = at::zeros({H}, opts.dtype(at::kFloat));
+  auto grad_h = grad_out.contiguous().clone();
+  auto grad_gicat = at::empty({N, 4 * H}, opts);
+  auto grad_h_direct = at::empty({N, H}, opts);
+  auto grad_wh = at::empty({N, H}, opts);
+  for (long s = S - 1; s >= 0; --s) {
+    auto h_in = (s == 0) ? x : Hnew.select(0, s - 1);
+    auto m = M.select(0, s);
+    launch_gru_gates2_bwd<bf16_t>(ptr<bf16_t>(grad_h), ptr<bf16_t>(h_in),
+                                  ptr<bf16_t>(R) + s * NH, ptr<bf16_t>(Z) + s * NH,
+                                  ptr<bf16_t>(Nn) + s * NH, ptr<bf16_t>(HN) + s * NH,
+                                  mptr<bf16_t>(grad_gicat), mptr<bf16_t>(grad_h_direct), NH, H,
+                                  stream);
+    auto grad_A = at::matmul(grad_gicat, Wcat);  // (N, 2H)
+    auto grad_m = grad_A.narrow(1, 0, H).contiguous();
+    launch_spmm_sum<bf16_t>(t_indptr.data_ptr<int>(), t_indices.data_ptr<int>(),
+                            ptr<bf16_t>(grad_m), mptr<bf16_t>(grad_wh), N, H, stream);
+    // weight gradients (K = N reductions via rocBLAS, fp32 accumulation)
+    gW_ih += at::matmul(grad_gicat.narrow(1, 0, 3 * H).t(), m);
+    gW_hh.narrow(0, 0, 2 * H) += at::matmul(grad_gicat.narrow(1, 0, 2 * H).t(), h_in);
+    gW_hh.narrow(0, 2 * H, H) += at::matmul(grad_gicat.narrow(1, 3 * H, H).t(), h_in);
+    gW_e += at::matmul(grad_wh.t(), h_in);
+    launch_colsum<bf16_t>(ptr<bf16_t>(grad_gicat), cs4.data_ptr<float>(), N, 4 * H, stream);
+    launch_colsum<bf16_t>(ptr<bf16_t>(grad_wh), cs_e.data_ptr<float>(), N, H, stream);
+    // grad wrt h_in: direct z-path + gh-path (grad_A right half) + wh-path
+    grad_h = grad_h_direct + grad_A.narrow(1, H, H) + at::matmul(grad_wh, W_e);
+  }
+  auto gb_ih = at::cat({cs4.narrow(0, 0, 2 * H), cs4.narrow(0, 2 * H, H)});
+  auto gb_hh = at::cat({cs4.narrow(0, 0, 2 * H), cs4.narrow(0, 3 * H, H)});
+  return {grad_h, gW_e, cs_e, gW_ih, gW_hh, gb_ih, gb_hh};
+}
+
+std::vector<at::Tensor> gru_gates2_fwd(at::Tensor gicat, at::Tensor h) {
+  CHECK_GPU(gicat);
+  CHECK_GPU(h);
+  const long N = h.size(0);
+  const int H = h.size(1);
+  auto h_new = at::empty_like(h);
+  auto r = at::empty_like(h);
+  auto z = at::empty_like(h);
+  auto n = at::empty_like(h);
+  auto hn = at::empty_like(h);
+  dispatch_float_bf16(h, "gru_gates2_fwd", [&](auto tag) {
+    using T = decltype(tag);
+    launch_gru_gates2_fwd<T>(ptr<T>(gicat), ptr<T>(h), mptr<T>(h_new), mptr<T>(r), mptr<T>(z),
+                             mptr<T>(n), mptr<T>(hn), N * H, H, cur_stream());
+  });
+  return {h_new, r, z, n, hn};
+}
+
+at::Tensor colsum(at::Tensor x) {
+  CHECK_GPU(x);
+  auto out = at::zeros({x.size(1)}, x.options().dtype(at::kFloat));
+  dispatch_float_bf16(x, "colsum", [&](auto tag) {
+    using T = decltype(tag);
+    launch_colsum<T>(ptr<T>(x), out.data_ptr<float>(), x.size(0), x.size(1), cur_stream());
+  });
+  return out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "deepdfa_amd MI355X (gfx950) kernels";
   m.def("embed4_fwd", &embed4_fwd);
@@ -196,4 +363,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_pool_fwd", &attn_pool_fwd);
   m.def("attn_pool_bwd", &attn_pool_bwd);
   m.def("segment_max", &segment_max);
+  m.def("gemm_bias", &gemm_bias);
+  m.def("gru_gates2_fwd", &gru_gates2_fwd);
+  m.def("colsum", &colsum);
+  m.def("ggnn_fused_fwd", &ggnn_fused_fwd);
+  m.def("ggnn_fused_bwd", &ggnn_fused_bwd);
 }

@@ -66,7 +66,13 @@ __global__ void embed4_bwd_kernel(const T* __restrict__ grad_out,
                                   const long* __restrict__ idx,
                                   float* __restrict__ grad_tables, long total,
                                   int V) {
-  // grid-stride over N*128 elements; atomic fp32 scatter-add.
+  // Grid-stride over N*128 elements; fp32 scatter-add. Vocabulary rows 0
+  // ("not a definition", ~60% of nodes) and 1 (UNKNOWN) are accumulated in
+  // LDS per block and flushed once — global atomic contention on those two
+  // rows otherwise serializes the whole kernel (rocprof: 95 us -> ~10 us).
+  __shared__ float hot[2 * 128];
+  for (int i = threadIdx.x; i < 2 * 128; i += blockDim.x) hot[i] = 0.f;
+  __syncthreads();
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += (long)gridDim.x * blockDim.x) {
     const long n = i >> 7;
@@ -74,7 +80,23 @@ __global__ void embed4_bwd_kernel(const T* __restrict__ grad_out,
     const int f = c >> 5;
     const int off = c & 31;
     const long row = idx[n * 4 + f];
-    atomicAdd(grad_tables + ((long)f * V + row) * 32 + off, to_f(grad_out[i]));
+    const float g = to_f(grad_out[i]);
+    if (row < 2) {
+      atomicAdd(hot + (int)row * 128 + c, g);
+    } else {
+      atomicAdd(grad_tables + ((long)f * V + row) * 32 + off, g);
+    }
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < 2 * 128; i += blockDim.x) {
+    const float v = hot[i];
+    if (v != 0.f) {
+      const int row = i >> 7;
+      const int c = i & 127;
+      const int f = c >> 5;
+      const int off = c & 31;
+      atomicAdd(grad_tables + ((long)f * V + row) * 32 + off, v);
+    }
   }
 }
 
@@ -172,7 +194,88 @@ __global__ void gru_gates_bwd_kernel(const T* __restrict__ grad_h_new,
 }
 
 // ---------------------------------------------------------------------------
-// K5: gated-attention pooling. One block (256 threads) per graph segment.
+// K3b: fused GRU gates over the block-GEMM output ("gicat" layout).
+// gicat (N, 4H) columns: [rsum | zsum | i_n | h_n] where rsum = i_r + h_r
+// etc. come straight from one MFMA GEMM of [m|h] against the block weight
+// matrix Wcat (see bindings ggnn_fused_fwd). Saves r/z/n/h_n for backward.
+// ---------------------------------------------------------------------------
+
+template <typename T>
+__global__ void gru_gates2_fwd_kernel(const T* __restrict__ gicat,
+                                      const T* __restrict__ h,
+                                      T* __restrict__ h_new, T* __restrict__ r_o,
+                                      T* __restrict__ z_o, T* __restrict__ n_o,
+                                      T* __restrict__ hn_o, long NH, int H) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < NH;
+       i += (long)gridDim.x * blockDim.x) {
+    const long row = i / H;
+    const int c = (int)(i - row * H);
+    const long b = row * 4 * H + c;
+    const float rs = to_f(gicat[b]);
+    const float zs = to_f(gicat[b + H]);
+    const float in_ = to_f(gicat[b + 2 * H]);
+    const float hn = to_f(gicat[b + 3 * H]);
+    const float r = sigmoidf_(rs);
+    const float z = sigmoidf_(zs);
+    const float n = tanhf(in_ + r * hn);
+    const float hv = to_f(h[i]);
+    h_new[i] = from_f<T>((1.f - z) * n + z * hv);
+    r_o[i] = from_f<T>(r);
+    z_o[i] = from_f<T>(z);
+    n_o[i] = from_f<T>(n);
+    hn_o[i] = from_f<T>(hn);
+  }
+}
+
+// grad_gicat columns: [dpr | dpz | dpn | dpn*r]; grad_h_direct = go * z.
+template <typename T>
+__global__ void gru_gates2_bwd_kernel(const T* __restrict__ grad_h_new,
+                                      const T* __restrict__ h,
+                                      const T* __restrict__ r_i,
+                                      const T* __restrict__ z_i,
+                                      const T* __restrict__ n_i,
+                                      const T* __restrict__ hn_i,
+                                      T* __restrict__ grad_gicat,
+                                      T* __restrict__ grad_h, long NH, int H) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < NH;
+       i += (long)gridDim.x * blockDim.x) {
+    const long row = i / H;
+    const int c = (int)(i - row * H);
+    const long b = row * 4 * H + c;
+    const float go = to_f(grad_h_new[i]);
+    const float r = to_f(r_i[i]), z = to_f(z_i[i]), n = to_f(n_i[i]);
+    const float hn = to_f(hn_i[i]);
+    const float hv = to_f(h[i]);
+    const float dn = go * (1.f - z);
+    const float dz = go * (hv - n);
+    const float dpn = dn * (1.f - n * n);
+    const float dr = dpn * hn;
+    const float dpr = dr * r * (1.f - r);
+    const float dpz = dz * z * (1.f - z);
+    grad_gicat[b] = from_f<T>(dpr);
+    grad_gicat[b + H] = from_f<T>(dpz);
+    grad_gicat[b + 2 * H] = from_f<T>(dpn);
+    grad_gicat[b + 3 * H] = from_f<T>(dpn * r);
+    grad_h[i] = from_f<T>(go * z);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Column sum (bias gradients): out[c] += sum_rows x[r][c], fp32 accumulate.
+// Blocks tile (row-chunk, col-chunk); one atomicAdd per (block, column).
+// ---------------------------------------------------------------------------
+
+template <typename T>
+__global__ void colsum_kernel(const T* __restrict__ x, float* __restrict__ out,
+                              int N, int C, int rows_per_block) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  const int r0 = blockIdx.y * rows_per_block;
+  const int r1 = min(r0 + rows_per_block, N);
+  float acc = 0.f;
+  for (int r = r0; r < r1; ++r) acc += to_f(x[(long)r * C + c]);
+  atomicAdd(out + c, acc);
+}
 //   fwd: alpha = softmax(gate[seg]); out[g] = sum_v alpha_v * x_v
 //   bwd: grad_x = alpha * grad_out[g];  grad_gate = alpha*(s - <alpha,s>),
 //        s_v = <grad_out[g], x_v>.
@@ -314,7 +417,8 @@ template <typename T>
 void launch_embed4_bwd(const T* grad_out, const long* idx, float* grad_tables,
                        long total, int V, hipStream_t stream) {
   const int block = 256;
-  const int grid = (int)min((total + block - 1) / block, (long)2048);
+  // few blocks => few per-block LDS flushes of the hot rows
+  const int grid = (int)min((total + block - 1) / block, (long)512);
   if (grid > 0)
     hipLaunchKernelGGL(embed4_bwd_kernel<T>, dim3(grid), dim3(block), 0, stream,
                        grad_out, idx, grad_tables, total, V);
@@ -374,6 +478,39 @@ void launch_attn_pool_bwd(const T* grad_out, const T* x, const float* alpha,
                        s_ws, D);
 }
 
+template <typename T>
+void launch_gru_gates2_fwd(const T* gicat, const T* h, T* h_new, T* r, T* z,
+                           T* n, T* hn, long NH, int H, hipStream_t stream) {
+  const int block = 256;
+  const int grid = (int)min((NH + block - 1) / block, (long)2048);
+  if (grid > 0)
+    hipLaunchKernelGGL(gru_gates2_fwd_kernel<T>, dim3(grid), dim3(block), 0,
+                       stream, gicat, h, h_new, r, z, n, hn, NH, H);
+}
+
+template <typename T>
+void launch_gru_gates2_bwd(const T* grad_h_new, const T* h, const T* r,
+                           const T* z, const T* n, const T* hn, T* grad_gicat,
+                           T* grad_h, long NH, int H, hipStream_t stream) {
+  const int block = 256;
+  const int grid = (int)min((NH + block - 1) / block, (long)2048);
+  if (grid > 0)
+    hipLaunchKernelGGL(gru_gates2_bwd_kernel<T>, dim3(grid), dim3(block), 0,
+                       stream, grad_h_new, h, r, z, n, hn, grad_gicat, grad_h,
+                       NH, H);
+}
+
+template <typename T>
+void launch_colsum(const T* x, float* out, int N, int C, hipStream_t stream) {
+  const int block = 256;
+  const int colb = (C + block - 1) / block;
+  const int rows_per_block = max(256, (N + 31) / 32);
+  const int rowb = (N + rows_per_block - 1) / rows_per_block;
+  if (N > 0 && C > 0)
+    hipLaunchKernelGGL(colsum_kernel<T>, dim3(colb, rowb), dim3(block), 0,
+                       stream, x, out, N, C, rows_per_block);
+}
+
 void launch_segment_max(const float* values, const int* node_offsets,
                         float* out, int B, hipStream_t stream) {
   if (B > 0)
@@ -398,7 +535,13 @@ void launch_segment_max(const float* values, const int* node_offsets,
                                         float*, int, int, hipStream_t);       \
   template void launch_attn_pool_bwd<T>(const T*, const T*, const float*,     \
                                         const int*, T*, T*, float*, int, int, \
-                                        hipStream_t);
+                                        hipStream_t);                         \
+  template void launch_gru_gates2_fwd<T>(const T*, const T*, T*, T*, T*, T*,  \
+                                         T*, long, int, hipStream_t);         \
+  template void launch_gru_gates2_bwd<T>(const T*, const T*, const T*,        \
+                                         const T*, const T*, const T*, T*,    \
+                                         T*, long, int, hipStream_t);         \
+  template void launch_colsum<T>(const T*, float*, int, int, hipStream_t);
 
 INSTANTIATE(float)
 INSTANTIATE(__hip_bfloat16)

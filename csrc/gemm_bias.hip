@@ -1,0 +1,133 @@
+// MFMA bf16 GEMM with bias + split-A, tuned for the flow-GNN's node-state
+// GEMMs on MI355X (gfx950): out(N, COL) = [A1|A2](N, K) @ W(COL, K)^T + bias.
+//
+// Shapes in this framework: N ~ 11-12k nodes, K in {128, 256}, COL in
+// {128, 512}. hipBLASLt's heuristic picks an 84-91us kernel for these
+// shapes (rocprof, profiles/); this kernel is ~10x faster by being sized
+// for them: 64x128 tile, BK=64, v_mfma_f32_16x16x32_bf16, XOR-swizzled LDS
+// (cdna_hip_programming.md T2: byte ^= (row&7)<<4 spreads the 16-lane
+// ds_read_b128 fragment groups across bank slots).
+//
+// The split-A form reads K-columns [0,K1) from A1 and [K1,K) from A2 so the
+// fused-GRU GEMM can consume [messages | hidden] without materializing the
+// concatenation.
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+#define WAVE 64
+
+using bf16 = __hip_bfloat16;
+using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;  // 4 VGPRs
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+using uint4v = __attribute__((ext_vector_type(4))) unsigned int;
+
+// tile geometry
+#define BM 64
+#define BN 128
+#define BK 64
+#define ROWB (BK * 2)  // bytes per LDS row (64 bf16)
+
+__device__ __forceinline__ int swz(int row, int byte) {
+  return row * ROWB + (byte ^ ((row & 7) << 4));
+}
+
+// out(N, COL) = concat_K(A1, A2)(N, K) @ W(COL, K)^T + bias
+// grid = (ceil(N/BM), COL/BN), block = 256 (4 waves, 2x2 of 32x64 tiles)
+__global__ __launch_bounds__(256) void gemm_bias_kernel(
+    const bf16* __restrict__ A1, const bf16* __restrict__ A2,
+    const bf16* __restrict__ W, const bf16* __restrict__ bias,
+    bf16* __restrict__ out, int N, int K, int K1, int COL) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  char* a_lds = smem;                 // BM x ROWB = 8 KiB
+  char* b_lds = smem + BM * ROWB;     // BN x ROWB = 16 KiB
+
+  const int r0 = blockIdx.x * BM;
+  const int c0 = blockIdx.y * BN;
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wid = tid >> 6;           // 4 waves: (wm, wn) = (wid>>1, wid&1)
+  const int wm = wid >> 1;            // 2 row-waves of 32 rows
+  const int wn = wid & 1;             // 2 col-waves of 64 cols
+
+  f32x4 acc[2][4] = {};               // per wave: 32x64 = 2x4 fragments
+
+  for (int kk = 0; kk < K; kk += BK) {
+    // stage A tile: BM rows x 64 k (128 B/row); 8 threads/row, 32 rows/pass
+    {
+      const int row = tid >> 3;        // 0..31
+      const int off = (tid & 7) * 16;  // byte offset in row
+      for (int rr = row; rr < BM; rr += 32) {
+        const int gr = r0 + rr;
+        uint4v v = {};
+        if (gr < N) {
+          const int gk = kk + off / 2;  // element index in K
+          const bf16* src = (gk < K1) ? (A1 + (long)gr * K1 + gk)
+                                      : (A2 + (long)gr * (K - K1) + (gk - K1));
+          v = *reinterpret_cast<const uint4v*>(src);
+        }
+        *reinterpret_cast<uint4v*>(a_lds + swz(rr, off)) = v;
+      }
+    }
+    // stage B tile: BN rows (output cols) x 64 k
+    {
+      const int row = tid >> 3;
+      const int off = (tid & 7) * 16;
+      for (int rr = row; rr < BN; rr += 32) {
+        const bf16* src = W + (long)(c0 + rr) * K + kk + off / 2;
+        *reinterpret_cast<uint4v*>(b_lds + swz(rr, off)) =
+            *reinterpret_cast<const uint4v*>(src);
+      }
+    }
+    __syncthreads();
+
+#pragma unroll
+    for (int ks = 0; ks < BK / 32; ++ks) {
+      // fragment k-range: lane kb = lane>>4 (4 groups of 8 elems)
+      const int kbyte = ks * 64 + (lane >> 4) * 16;
+      bf16x8 a_frag[2], b_frag[4];
+#pragma unroll
+      for (int m = 0; m < 2; ++m) {
+        const int row = wm * 32 + m * 16 + (lane & 15);
+        a_frag[m] = *reinterpret_cast<const bf16x8*>(a_lds + swz(row, kbyte));
+      }
+#pragma unroll
+      for (int n = 0; n < 4; ++n) {
+        const int row = wn * 64 + n * 16 + (lane & 15);
+        b_frag[n] = *reinterpret_cast<const bf16x8*>(b_lds + swz(row, kbyte));
+      }
+#pragma unroll
+      for (int m = 0; m < 2; ++m)
+#pragma unroll
+        for (int n = 0; n < 4; ++n)
+          acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[m], b_frag[n], acc[m][n], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // epilogue: D mapping col = lane&15, row = (lane>>4)*4 + i
+#pragma unroll
+  for (int m = 0; m < 2; ++m) {
+    const int row_base = r0 + wm * 32 + m * 16 + (lane >> 4) * 4;
+#pragma unroll
+    for (int n = 0; n < 4; ++n) {
+      const int col = c0 + wn * 64 + n * 16 + (lane & 15);
+      const float b = bias ? __bfloat162float(bias[col]) : 0.0f;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int row = row_base + i;
+        if (row < N) out[(long)row * COL + col] = __float2bfloat16(acc[m][n][i] + b);
+      }
+    }
+  }
+}
+
+void launch_gemm_bias(const bf16* A1, const bf16* A2, const bf16* W,
+                      const bf16* bias, bf16* out, int N, int K, int K1,
+                      int COL, hipStream_t stream) {
+  const dim3 grid((N + BM - 1) / BM, COL / BN);
+  const size_t lds = (BM + BN) * ROWB;
+  hipLaunchKernelGGL(gemm_bias_kernel, grid, dim3(256), lds, stream, A1, A2, W,
+                     bias, out, N, K, K1, COL);
+}

@@ -127,6 +127,63 @@ def gru_cell(
     return _GruGates.apply(gi, gh, h.to(gi.dtype))
 
 
+class _GGNNFused(torch.autograd.Function):
+    """Whole n_steps GGNN loop as one autograd node, driven from C++
+    (bindings ggnn_fused_fwd/bwd): per step one MFMA gemm_bias for W_e h,
+    the CSR segment-sum, one split-A MFMA GEMM against the block weight
+    matrix Wcat producing all GRU gate pre-activations, and a fused gate
+    kernel. bf16 compute, fp32-master-friendly (weight grads come back
+    fp32)."""
+
+    @staticmethod
+    def forward(ctx, x, w_e, b_e, w_ih, w_hh, b_ih, b_hh, indptr, indices, t_indptr, t_indices, n_steps):
+        from ._ext import load_ext
+
+        ext = load_ext(required=True)
+        h_final, M, R, Z, Nn, HN, Hnew = ext.ggnn_fused_fwd(
+            indptr, indices, x, w_e, b_e, w_ih, w_hh, b_ih, b_hh, n_steps
+        )
+        ctx.save_for_backward(x, w_e, w_ih, w_hh, M, R, Z, Nn, HN, Hnew, t_indptr, t_indices)
+        ctx.n_steps = n_steps
+        return h_final
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        from ._ext import load_ext
+
+        ext = load_ext(required=True)
+        x, w_e, w_ih, w_hh, M, R, Z, Nn, HN, Hnew, t_indptr, t_indices = ctx.saved_tensors
+        grad_x, gW_e, gb_e, gW_ih, gW_hh, gb_ih, gb_hh = ext.ggnn_fused_bwd(
+            grad_out.contiguous(), t_indptr, t_indices, x, w_e, w_ih, w_hh,
+            M, R, Z, Nn, HN, Hnew, ctx.n_steps,
+        )
+        dt = x.dtype
+        return (
+            grad_x,
+            gW_e.to(dt), gb_e.to(dt), gW_ih.to(dt), gW_hh.to(dt), gb_ih.to(dt), gb_hh.to(dt),
+            None, None, None, None, None,
+        )
+
+
+def ggnn_fused(x, graph, linear: torch.nn.Linear, gru: torch.nn.GRUCell, n_steps: int):
+    """bf16 fused GGNN path (GPU only)."""
+    bf = torch.bfloat16
+    return _GGNNFused.apply(
+        x.to(bf).contiguous(),
+        linear.weight.to(bf).contiguous(),
+        linear.bias.to(bf).contiguous(),
+        gru.weight_ih.to(bf).contiguous(),
+        gru.weight_hh.to(bf).contiguous(),
+        gru.bias_ih.to(bf).contiguous(),
+        gru.bias_hh.to(bf).contiguous(),
+        graph.indptr,
+        graph.indices,
+        graph.t_indptr,
+        graph.t_indices,
+        n_steps,
+    )
+
+
 class _AttnPool(torch.autograd.Function):
     """K5: segment softmax over gate logits + weighted segment sum."""
 

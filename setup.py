@@ -21,6 +21,7 @@ ext = CUDAExtension(
     sources=[
         "csrc/bindings.hip",
         "csrc/flowgnn_kernels.hip",
+        "csrc/gemm_bias.hip",
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
