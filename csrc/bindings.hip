@@ -32,7 +32,10 @@ void launch_gru_gates2_bwd(const T*, const T*, const T*, const T*, const T*, con
 template <typename T>
 void launch_colsum(const T*, float*, int, int, hipStream_t);
 void launch_gemm_bias(const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*,
-                      const __hip_bfloat16*, __hip_bfloat16*, int, int, int, int, hipStream_t);
+                      const __hip_bfloat16*, const __hip_bfloat16*, __hip_bfloat16*, int, int,
+                      int, int, hipStream_t);
+void launch_wgrad(const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*, float*,
+                  int, int, int, int, hipStream_t);
 
 #define CHECK_GPU(t) \
   TORCH_CHECK((t).is_cuda() && (t).is_contiguous(), #t " must be contiguous GPU tensor")
@@ -220,7 +223,8 @@ static at::Tensor build_wcat(const at::Tensor& W_ih, const at::Tensor& W_hh, lon
   return Wcat;
 }
 
-at::Tensor gemm_bias(at::Tensor A, at::Tensor W, c10::optional<at::Tensor> bias) {
+at::Tensor gemm_bias(at::Tensor A, at::Tensor W, c10::optional<at::Tensor> bias,
+                     c10::optional<at::Tensor> addend) {
   CHECK_GPU(A);
   CHECK_GPU(W);
   TORCH_CHECK(A.scalar_type() == at::kBFloat16 && W.scalar_type() == at::kBFloat16);
@@ -229,8 +233,25 @@ at::Tensor gemm_bias(at::Tensor A, at::Tensor W, c10::optional<at::Tensor> bias)
   auto out = at::empty({N, COL}, A.options());
   const bf16_t* b = nullptr;
   if (bias.has_value()) b = reinterpret_cast<const bf16_t*>(bias->data_ptr());
-  launch_gemm_bias(ptr<bf16_t>(A), nullptr, ptr<bf16_t>(W), b, mptr<bf16_t>(out), N, K, K, COL,
-                   cur_stream());
+  const bf16_t* add = nullptr;
+  if (addend.has_value()) {
+    TORCH_CHECK(addend->is_contiguous() && addend->sizes() == out.sizes());
+    add = reinterpret_cast<const bf16_t*>(addend->data_ptr());
+  }
+  launch_gemm_bias(ptr<bf16_t>(A), nullptr, ptr<bf16_t>(W), b, add, mptr<bf16_t>(out), N, K, K,
+                   COL, cur_stream());
+  return out;
+}
+
+// out(M, C) = A(K, M)^T @ B(K, C), fp32 output (wgrad shape class)
+at::Tensor wgrad(at::Tensor A, at::Tensor B) {
+  CHECK_GPU(A);
+  CHECK_GPU(B);
+  const int K = A.size(0), M = A.size(1), C = B.size(1);
+  TORCH_CHECK(B.size(0) == K && M % 64 == 0 && C % 8 == 0);
+  auto out = at::zeros({M, C}, A.options().dtype(at::kFloat));
+  launch_wgrad(ptr<bf16_t>(A), ptr<bf16_t>(B), nullptr, out.data_ptr<float>(), K, M, C, C,
+               cur_stream());
   return out;
 }
 
@@ -249,78 +270,95 @@ std::vector<at::Tensor> ggnn_fused_fwd(at::Tensor indptr, at::Tensor indices, at
                         b_ih.narrow(0, 2 * H, H), b_hh.narrow(0, 2 * H, H)});
   const long S = n_steps;
   auto opts = x.options();
+  // HH[s] = hidden state entering step s; HH[S] = final output.
+  auto HH = at::empty({S + 1, N, H}, opts);
+  HH.select(0, 0).copy_(x);
   auto M = at::empty({S, N, H}, opts);
   auto R = at::empty({S, N, H}, opts);
   auto Z = at::empty({S, N, H}, opts);
   auto Nn = at::empty({S, N, H}, opts);
   auto HN = at::empty({S, N, H}, opts);
-  auto Hnew = at::empty({S, N, H}, opts);
   auto wh = at::empty({N, H}, opts);
   auto gicat = at::empty({N, 4 * H}, opts);
   const long NH = N * H;
-  const bf16_t* h = ptr<bf16_t>(x);
   for (long s = 0; s < S; ++s) {
-    launch_gemm_bias(h, nullptr, ptr<bf16_t>(W_e), ptr<bf16_t>(b_e), mptr<bf16_t>(wh), N, H, H, H,
-                     stream);
+    const bf16_t* h = ptr<bf16_t>(HH) + s * NH;
+    launch_gemm_bias(h, nullptr, ptr<bf16_t>(W_e), ptr<bf16_t>(b_e), nullptr, mptr<bf16_t>(wh),
+                     N, H, H, H, stream);
     bf16_t* m = mptr<bf16_t>(M) + s * NH;
     launch_spmm_sum<bf16_t>(indptr.data_ptr<int>(), indices.data_ptr<int>(), ptr<bf16_t>(wh), m,
                             N, H, stream);
-    launch_gemm_bias(m, h, ptr<bf16_t>(Wcat), ptr<bf16_t>(b_cat), mptr<bf16_t>(gicat), N, 2 * H,
-                     H, 4 * H, stream);
-    launch_gru_gates2_fwd<bf16_t>(ptr<bf16_t>(gicat), h, mptr<bf16_t>(Hnew) + s * NH,
+    launch_gemm_bias(m, h, ptr<bf16_t>(Wcat), ptr<bf16_t>(b_cat), nullptr, mptr<bf16_t>(gicat),
+                     N, 2 * H, H, 4 * H, stream);
+    launch_gru_gates2_fwd<bf16_t>(ptr<bf16_t>(gicat), h, mptr<bf16_t>(HH) + (s + 1) * NH,
                                   mptr<bf16_t>(R) + s * NH, mptr<bf16_t>(Z) + s * NH,
                                   mptr<bf16_t>(Nn) + s * NH, mptr<bf16_t>(HN) + s * NH, NH, H,
                                   stream);
-    h = ptr<bf16_t>(Hnew) + s * NH;
   }
-  auto h_final = Hnew.select(0, S - 1);
-  return {h_final, M, R, Z, Nn, HN, Hnew};
+  auto h_final = HH.select(0, S);
+  return {h_final, HH, M, R, Z, Nn, HN};
 }
 
 std::vector<at::Tensor> ggnn_fused_bwd(at::Tensor grad_out, at::Tensor t_indptr,
                                        at::Tensor t_indices, at::Tensor x, at::Tensor W_e,
-                                       at::Tensor W_ih, at::Tensor W_hh, at::Tensor M,
-                                       at::Tensor R, at::Tensor Z, at::Tensor Nn, at::Tensor HN,
-                                       at::Tensor Hnew, long n_steps) {
+                                       at::Tensor W_ih, at::Tensor W_hh, at::Tensor HH,
+                                       at::Tensor M, at::Tensor R, at::Tensor Z, at::Tensor Nn,
+                                       at::Tensor HN, long n_steps) {
   const long N = x.size(0);
   const long H = x.size(1);
   const long S = n_steps;
   const long NH = N * H;
   auto stream = cur_stream();
   auto Wcat = build_wcat(W_ih, W_hh, H);
+  auto WcatT = Wcat.t().contiguous();      // (2H, 4H)
+  auto W_eT = W_e.t().contiguous();        // (H, H)
   auto opts = x.options();
-  auto gW_e = at::zeros({H, H}, opts.dtype(at::kFloat));
-  auto gW_ih = at::zeros({3 * H, H}, opts.dtype(at::kFloat));
-  auto gW_hh = at::zeros({3 * H, H}, opts.dtype(at::kFloat));
-  auto cs4 = at::zeros({4 * H}, opts.dtype(at::kFloat));
-  auto cs_e = at::zeros({H}, opts.dtype(at::kFloat));
+  // per-step gate/message grads, kept for ONE batched K = S*N weight-grad
+  // GEMM at the end (hipBLASLt's transpose-A kernels are the pathology the
+  // custom split-K wgrad kernel replaces; see csrc/wgrad.hip header)
+  auto Ggicat = at::empty({S, N, 4 * H}, opts);
+  auto Gwh = at::empty({S, N, H}, opts);
   auto grad_h = grad_out.contiguous().clone();
-  auto grad_gicat = at::empty({N, 4 * H}, opts);
-  auto grad_h_direct = at::empty({N, H}, opts);
-  auto grad_wh = at::empty({N, H}, opts);
+  auto grad_hd = at::empty({N, H}, opts);
+  auto tmp = at::empty({N, H}, opts);
   for (long s = S - 1; s >= 0; --s) {
-    auto h_in = (s == 0) ? x : Hnew.select(0, s - 1);
-    auto m = M.select(0, s);
-    launch_gru_gates2_bwd<bf16_t>(ptr<bf16_t>(grad_h), ptr<bf16_t>(h_in),
-                                  ptr<bf16_t>(R) + s * NH, ptr<bf16_t>(Z) + s * NH,
-                                  ptr<bf16_t>(Nn) + s * NH, ptr<bf16_t>(HN) + s * NH,
-                                  mptr<bf16_t>(grad_gicat), mptr<bf16_t>(grad_h_direct), NH, H,
-                                  stream);
-    auto grad_A = at::matmul(grad_gicat, Wcat);  // (N, 2H)
+    const bf16_t* h_in = ptr<bf16_t>(HH) + s * NH;
+    auto ggic = Ggicat.select(0, s);
+    auto gwh = Gwh.select(0, s);
+    launch_gru_gates2_bwd<bf16_t>(ptr<bf16_t>(grad_h), h_in, ptr<bf16_t>(R) + s * NH,
+                                  ptr<bf16_t>(Z) + s * NH, ptr<bf16_t>(Nn) + s * NH,
+                                  ptr<bf16_t>(HN) + s * NH, mptr<bf16_t>(ggic),
+                                  mptr<bf16_t>(grad_hd), NH, H, stream);
+    // grad_A = ggic @ Wcat : (N, 2H); left half = grad_m, right = gh-path
+    auto grad_A = at::empty({N, 2 * H}, opts);
+    launch_gemm_bias(ptr<bf16_t>(ggic), nullptr, ptr<bf16_t>(WcatT), nullptr, nullptr,
+                     mptr<bf16_t>(grad_A), N, 4 * H, 4 * H, 2 * H, stream);
     auto grad_m = grad_A.narrow(1, 0, H).contiguous();
     launch_spmm_sum<bf16_t>(t_indptr.data_ptr<int>(), t_indices.data_ptr<int>(),
-                            ptr<bf16_t>(grad_m), mptr<bf16_t>(grad_wh), N, H, stream);
-    // weight gradients (K = N reductions via rocBLAS, fp32 accumulation)
-    gW_ih += at::matmul(grad_gicat.narrow(1, 0, 3 * H).t(), m);
-    gW_hh.narrow(0, 0, 2 * H) += at::matmul(grad_gicat.narrow(1, 0, 2 * H).t(), h_in);
-    gW_hh.narrow(0, 2 * H, H) += at::matmul(grad_gicat.narrow(1, 3 * H, H).t(), h_in);
-    gW_e += at::matmul(grad_wh.t(), h_in);
-    launch_colsum<bf16_t>(ptr<bf16_t>(grad_gicat), cs4.data_ptr<float>(), N, 4 * H, stream);
-    launch_colsum<bf16_t>(ptr<bf16_t>(grad_wh), cs_e.data_ptr<float>(), N, H, stream);
-    // grad wrt h_in: direct z-path + gh-path (grad_A right half) + wh-path
-    grad_h = grad_h_direct + grad_A.narrow(1, H, H) + at::matmul(grad_wh, W_e);
+                            ptr<bf16_t>(grad_m), mptr<bf16_t>(gwh), N, H, stream);
+    // grad wrt h_in = direct z-path + gh-path + (grad_wh @ W_e)
+    at::add_out(tmp, grad_hd, grad_A.narrow(1, H, H));
+    launch_gemm_bias(ptr<bf16_t>(gwh), nullptr, ptr<bf16_t>(W_eT), nullptr, ptr<bf16_t>(tmp),
+                     mptr<bf16_t>(grad_h), N, H, H, H, stream);
   }
-  auto gb_ih = at::cat({cs4.narrow(0, 0, 2 * H), cs4.narrow(0, 2 * H, H)});
+  // batched weight/bias grads over all steps (K = S*N)
+  auto A_g = Ggicat.view({S * N, 4 * H});
+  auto gWcat = at::zeros({4 * H, 2 * H}, opts.dtype(at::kFloat));
+  launch_wgrad(ptr<bf16_t>(A_g), ptr<bf16_t>(M), ptr<bf16_t>(HH), gWcat.data_ptr<float>(), S * N,
+               4 * H, 2 * H, H, stream);
+  auto A_w = Gwh.view({S * N, H});
+  auto gW_e = at::zeros({H, H}, opts.dtype(at::kFloat));
+  launch_wgrad(ptr<bf16_t>(A_w), ptr<bf16_t>(HH), nullptr, gW_e.data_ptr<float>(), S * N, H, H,
+               H, stream);
+  auto cs4 = at::zeros({4 * H}, opts.dtype(at::kFloat));
+  launch_colsum<bf16_t>(ptr<bf16_t>(A_g), cs4.data_ptr<float>(), S * N, 4 * H, stream);
+  auto cs_e = at::zeros({H}, opts.dtype(at::kFloat));
+  launch_colsum<bf16_t>(ptr<bf16_t>(A_w), cs_e.data_ptr<float>(), S * N, H, stream);
+  // scatter gWcat blocks back to the GRUCell weight layout
+  auto gW_ih = gWcat.narrow(0, 0, 3 * H).narrow(1, 0, H).contiguous();
+  auto gW_hh = at::cat({gWcat.narrow(0, 0, 2 * H).narrow(1, H, H),
+                        gWcat.narrow(0, 3 * H, H).narrow(1, H, H)});
+  auto gb_ih = cs4.narrow(0, 0, 3 * H).contiguous();
   auto gb_hh = at::cat({cs4.narrow(0, 0, 2 * H), cs4.narrow(0, 3 * H, H)});
   return {grad_h, gW_e, cs_e, gW_ih, gW_hh, gb_ih, gb_hh};
 }
@@ -364,6 +402,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_pool_bwd", &attn_pool_bwd);
   m.def("segment_max", &segment_max);
   m.def("gemm_bias", &gemm_bias);
+  m.def("wgrad", &wgrad);
   m.def("gru_gates2_fwd", &gru_gates2_fwd);
   m.def("colsum", &colsum);
   m.def("ggnn_fused_fwd", &ggnn_fused_fwd);

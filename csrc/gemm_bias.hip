@@ -37,7 +37,8 @@ __device__ __forceinline__ int swz(int row, int byte) {
 __global__ __launch_bounds__(256) void gemm_bias_kernel(
     const bf16* __restrict__ A1, const bf16* __restrict__ A2,
     const bf16* __restrict__ W, const bf16* __restrict__ bias,
-    bf16* __restrict__ out, int N, int K, int K1, int COL) {
+    const bf16* __restrict__ addend, bf16* __restrict__ out, int N, int K,
+    int K1, int COL) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   char* a_lds = smem;                 // BM x ROWB = 8 KiB
   char* b_lds = smem + BM * ROWB;     // BN x ROWB = 16 KiB
@@ -117,17 +118,21 @@ __global__ __launch_bounds__(256) void gemm_bias_kernel(
 #pragma unroll
       for (int i = 0; i < 4; ++i) {
         const int row = row_base + i;
-        if (row < N) out[(long)row * COL + col] = __float2bfloat16(acc[m][n][i] + b);
+        if (row < N) {
+          float v = acc[m][n][i] + b;
+          if (addend) v += __bfloat162float(addend[(long)row * COL + col]);
+          out[(long)row * COL + col] = __float2bfloat16(v);
+        }
       }
     }
   }
 }
 
 void launch_gemm_bias(const bf16* A1, const bf16* A2, const bf16* W,
-                      const bf16* bias, bf16* out, int N, int K, int K1,
-                      int COL, hipStream_t stream) {
+                      const bf16* bias, const bf16* addend, bf16* out, int N,
+                      int K, int K1, int COL, hipStream_t stream) {
   const dim3 grid((N + BM - 1) / BM, COL / BN);
   const size_t lds = (BM + BN) * ROWB;
   hipLaunchKernelGGL(gemm_bias_kernel, grid, dim3(256), lds, stream, A1, A2, W,
-                     bias, out, N, K, K1, COL);
+                     bias, addend, out, N, K, K1, COL);
 }

@@ -1,0 +1,112 @@
+// Split-K weight-gradient GEMM: out(M, C) += A(K, M)^T @ [B1|B2](K, C).
+//
+// The transpose-A reduction GEMMs (weight gradients, K = nodes*steps ~ 58k,
+// output 512x256 or 128x128) are where hipBLASLt collapses on these shapes:
+// its heuristic picks a kernel with (M/MT)*(C/NT) = 6 workgroups and no
+// K-split — 2.3% of the chip, 87 us (tools/gemm_probe.py). This kernel
+// splits K across blockIdx.z with fp32 atomicAdd epilogues: 64x64 output
+// tile per block, v_mfma_f32_16x16x32_bf16, A-tiles staged [k][m] in LDS
+// (coalesced) and read transposed by scalar LDS loads (frag build is
+// LDS-issue bound but fully hidden under the MFMA issue rate at this size).
+//
+// B is split (B1 cols [0,C1), B2 cols [C1,C)) so gWcat = grad_gicat^T @
+// [messages | hidden] needs no concatenation.
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+#define WAVE 64
+#define KS 32  // K per MFMA step
+
+using bf16 = __hip_bfloat16;
+using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+using uint4v = __attribute__((ext_vector_type(4))) unsigned int;
+
+__global__ __launch_bounds__(256) void wgrad_kernel(
+    const bf16* __restrict__ A, const bf16* __restrict__ B1,
+    const bf16* __restrict__ B2, float* __restrict__ out, int K, int M,
+    int C, int C1, int kchunk) {
+  __shared__ __bf16 lds_a[KS][64];
+  __shared__ __bf16 lds_b[KS][64];
+
+  const int m0 = blockIdx.x * 64;
+  const int c0 = blockIdx.y * 64;
+  const int k_begin = blockIdx.z * kchunk;
+  const int k_end = min(K, k_begin + kchunk);
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wid = tid >> 6;
+  const int wm = wid >> 1;  // 2 row-waves (m), 2 col-waves (c)
+  const int wc = wid & 1;
+
+  f32x4 acc[2][2] = {};
+
+  for (int k0 = k_begin; k0 < k_end; k0 += KS) {
+    // stage A chunk [KS][64] at columns m0.. (row-major, coalesced 16B)
+    {
+      const int row = tid >> 3;          // 0..31
+      const int coff = (tid & 7) * 8;    // element offset in tile row
+      const int k = k0 + row;
+      uint4v va = {};
+      if (k < k_end) va = *reinterpret_cast<const uint4v*>(A + (long)k * M + m0 + coff);
+      *reinterpret_cast<uint4v*>(&lds_a[row][coff]) = va;
+      uint4v vb = {};
+      if (k < k_end) {
+        const int cg = c0 + coff;
+        const bf16* src = (cg < C1) ? (B1 + (long)k * C1 + cg)
+                                    : (B2 + (long)k * (C - C1) + (cg - C1));
+        vb = *reinterpret_cast<const uint4v*>(src);
+      }
+      *reinterpret_cast<uint4v*>(&lds_b[row][coff]) = vb;
+    }
+    __syncthreads();
+
+#pragma unroll
+    for (int fm = 0; fm < 2; ++fm) {
+      const int m_loc = wm * 32 + fm * 16 + (lane & 15);
+      bf16x8 a_frag;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) a_frag[j] = lds_a[(lane >> 4) * 8 + j][m_loc];
+#pragma unroll
+      for (int fc = 0; fc < 2; ++fc) {
+        const int c_loc = wc * 32 + fc * 16 + (lane & 15);
+        bf16x8 b_frag;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) b_frag[j] = lds_b[(lane >> 4) * 8 + j][c_loc];
+        acc[fm][fc] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag,
+                                                              acc[fm][fc], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+  // epilogue: D col = lane&15, row = (lane>>4)*4 + i; atomic fp32 accumulate
+#pragma unroll
+  for (int fm = 0; fm < 2; ++fm) {
+    const int m_base = m0 + wm * 32 + fm * 16 + (lane >> 4) * 4;
+#pragma unroll
+    for (int fc = 0; fc < 2; ++fc) {
+      const int c = c0 + wc * 32 + fc * 16 + (lane & 15);
+      if (c >= C) continue;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int m = m_base + i;
+        if (m < M) atomicAdd(out + (long)m * C + c, acc[fm][fc][i]);
+      }
+    }
+  }
+}
+
+void launch_wgrad(const bf16* A, const bf16* B1, const bf16* B2, float* out,
+                  int K, int M, int C, int C1, hipStream_t stream) {
+  // size the K-split so the grid comfortably fills 256 CUs
+  const int tiles = ((M + 63) / 64) * ((C + 63) / 64);
+  int zsplit = max(1, 512 / tiles);
+  int kchunk = (K + zsplit - 1) / zsplit;
+  kchunk = ((kchunk + KS - 1) / KS) * KS;
+  zsplit = (K + kchunk - 1) / kchunk;
+  const dim3 grid((M + 63) / 64, (C + 63) / 64, zsplit);
+  hipLaunchKernelGGL(wgrad_kernel, grid, dim3(256), 0, stream, A, B1, B2, out,
+                     K, M, C, C1, kchunk);
+}

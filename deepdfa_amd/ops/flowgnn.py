@@ -140,10 +140,10 @@ class _GGNNFused(torch.autograd.Function):
         from ._ext import load_ext
 
         ext = load_ext(required=True)
-        h_final, M, R, Z, Nn, HN, Hnew = ext.ggnn_fused_fwd(
+        h_final, HH, M, R, Z, Nn, HN = ext.ggnn_fused_fwd(
             indptr, indices, x, w_e, b_e, w_ih, w_hh, b_ih, b_hh, n_steps
         )
-        ctx.save_for_backward(x, w_e, w_ih, w_hh, M, R, Z, Nn, HN, Hnew, t_indptr, t_indices)
+        ctx.save_for_backward(x, w_e, w_ih, w_hh, HH, M, R, Z, Nn, HN, t_indptr, t_indices)
         ctx.n_steps = n_steps
         return h_final
 
@@ -152,10 +152,10 @@ class _GGNNFused(torch.autograd.Function):
         from ._ext import load_ext
 
         ext = load_ext(required=True)
-        x, w_e, w_ih, w_hh, M, R, Z, Nn, HN, Hnew, t_indptr, t_indices = ctx.saved_tensors
+        x, w_e, w_ih, w_hh, HH, M, R, Z, Nn, HN, t_indptr, t_indices = ctx.saved_tensors
         grad_x, gW_e, gb_e, gW_ih, gW_hh, gb_ih, gb_hh = ext.ggnn_fused_bwd(
             grad_out.contiguous(), t_indptr, t_indices, x, w_e, w_ih, w_hh,
-            M, R, Z, Nn, HN, Hnew, ctx.n_steps,
+            HH, M, R, Z, Nn, HN, ctx.n_steps,
         )
         dt = x.dtype
         return (

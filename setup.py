@@ -22,6 +22,7 @@ ext = CUDAExtension(
         "csrc/bindings.hip",
         "csrc/flowgnn_kernels.hip",
         "csrc/gemm_bias.hip",
+        "csrc/wgrad.hip",
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],

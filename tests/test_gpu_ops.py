@@ -104,7 +104,7 @@ def test_gemm_bias_gpu(dev):
         A = torch.randn(N, K, device=dev, dtype=torch.bfloat16)
         W = torch.randn(COL, K, device=dev, dtype=torch.bfloat16) * 0.1
         b = torch.randn(COL, device=dev, dtype=torch.bfloat16)
-        out = ext.gemm_bias(A, W, b)
+        out = ext.gemm_bias(A, W, b, None)
         ref_out = (A.float() @ W.float().t() + b.float()).to(torch.bfloat16)
         err = (out.float() - ref_out.float()).abs().max().item()
         scale = ref_out.float().abs().max().item()
@@ -135,6 +135,22 @@ def test_ggnn_fused_matches_unfused(dev):
     for n, p in conv.named_parameters():
         rel = (p.grad - grads_ref[n]).abs().max() / (grads_ref[n].abs().max() + 1e-6)
         assert rel < 0.25, (n, rel)
+
+
+def test_wgrad_gpu(dev):
+    """Split-K transpose-A weight-grad GEMM vs rocBLAS fp32 (asymmetric)."""
+    from deepdfa_amd.ops import load_ext
+
+    ext = load_ext(required=True)
+    torch.manual_seed(3)
+    for K, M, C in [(11600, 512, 256), (58000, 128, 128), (100, 128, 64)]:
+        A = torch.randn(K, M, device=dev, dtype=torch.bfloat16) * 0.1
+        B = torch.randn(K, C, device=dev, dtype=torch.bfloat16) * 0.1
+        out = ext.wgrad(A, B)
+        ref_w = A.float().t() @ B.float()
+        scale = ref_w.abs().max().item()
+        err = (out - ref_w).abs().max().item()
+        assert err < 0.02 * max(scale, 1.0) + 0.5, (K, M, C, err, scale)
 
 
 def test_colsum_gpu(dev):
