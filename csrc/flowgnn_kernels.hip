@@ -339,11 +339,25 @@ __global__ void attn_pool_fwd_kernel(const T* __restrict__ x,
   const float inv = 1.0f / s;
   for (int v = threadIdx.x; v < n; v += blockDim.x) alpha[lo + v] *= inv;
   __syncthreads();
-  // phase C: weighted segment sum, one output column per thread (strided)
+  // phase C: weighted segment sum, one output column per thread (strided).
+  // 8 independent accumulators: a 500-node graph is otherwise a serial
+  // dependent-add chain at full memory latency (the kernel's whole cost).
   for (int d = threadIdx.x; d < D; d += blockDim.x) {
-    float acc = 0.f;
-    for (int v = 0; v < n; ++v) acc += alpha[lo + v] * to_f(x[(long)(lo + v) * D + d]);
-    out[(long)g * D + d] = from_f<T>(acc);
+    float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f, a4 = 0.f, a5 = 0.f, a6 = 0.f, a7 = 0.f;
+    int v = 0;
+    for (; v + 8 <= n; v += 8) {
+      const long base = (long)(lo + v) * D + d;
+      a0 += alpha[lo + v] * to_f(x[base]);
+      a1 += alpha[lo + v + 1] * to_f(x[base + D]);
+      a2 += alpha[lo + v + 2] * to_f(x[base + 2 * D]);
+      a3 += alpha[lo + v + 3] * to_f(x[base + 3 * D]);
+      a4 += alpha[lo + v + 4] * to_f(x[base + 4 * D]);
+      a5 += alpha[lo + v + 5] * to_f(x[base + 5 * D]);
+      a6 += alpha[lo + v + 6] * to_f(x[base + 6 * D]);
+      a7 += alpha[lo + v + 7] * to_f(x[base + 7 * D]);
+    }
+    for (; v < n; ++v) a0 += alpha[lo + v] * to_f(x[(long)(lo + v) * D + d]);
+    out[(long)g * D + d] = from_f<T>(((a0 + a1) + (a2 + a3)) + ((a4 + a5) + (a6 + a7)));
   }
 }
 
@@ -365,19 +379,30 @@ __global__ void attn_pool_bwd_kernel(const T* __restrict__ grad_out,
   for (int d = threadIdx.x; d < D; d += blockDim.x)
     go[d] = to_f(grad_out[(long)g * D + d]);
   __syncthreads();
-  // pass 1: s_v = <go, x_v> (one wave per node), accumulate dot = sum alpha*s
+  // pass 1: s_v = <go, x_v> (one wave per node pair, 2-node ILP so the
+  // dot-product latency chains of a 500-node graph overlap)
   const int wid = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
   const int nw = blockDim.x / WAVE;
   float local_dot = 0.f;
-  for (int v = wid; v < n; v += nw) {
-    float sv = 0.f;
-    for (int d = lane; d < D; d += WAVE) sv += go[d] * to_f(x[(long)(lo + v) * D + d]);
-    for (int off = WAVE / 2; off > 0; off >>= 1) sv += __shfl_down(sv, off);
-    sv = __shfl(sv, 0);
+  for (int v = wid * 2; v < n; v += nw * 2) {
+    float sv0 = 0.f, sv1 = 0.f;
+    const bool has1 = (v + 1) < n;
+    for (int d = lane; d < D; d += WAVE) {
+      sv0 += go[d] * to_f(x[(long)(lo + v) * D + d]);
+      if (has1) sv1 += go[d] * to_f(x[(long)(lo + v + 1) * D + d]);
+    }
+    for (int off = WAVE / 2; off > 0; off >>= 1) {
+      sv0 += __shfl_down(sv0, off);
+      sv1 += __shfl_down(sv1, off);
+    }
     if (lane == 0) {
-      s_ws[lo + v] = sv;
-      local_dot += alpha[lo + v] * sv;
+      s_ws[lo + v] = sv0;
+      local_dot += alpha[lo + v] * sv0;
+      if (has1) {
+        s_ws[lo + v + 1] = sv1;
+        local_dot += alpha[lo + v + 1] * sv1;
+      }
     }
   }
   __syncthreads();
@@ -482,7 +507,7 @@ void launch_attn_pool_bwd(const T* grad_out, const T* x, const float* alpha,
                           float* s_ws, int B, int D, hipStream_t stream) {
   const size_t lds = (D + 8) * sizeof(float);
   if (B > 0)
-    hipLaunchKernelGGL(attn_pool_bwd_kernel<T>, dim3(B), dim3(256), lds, stream,
+    hipLaunchKernelGGL(attn_pool_bwd_kernel<T>, dim3(B), dim3(512), lds, stream,
                        grad_out, x, alpha, node_offsets, grad_x, grad_gate,
                        s_ws, D);
 }

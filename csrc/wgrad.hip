@@ -4,10 +4,14 @@
 // output 512x256 or 128x128) are where hipBLASLt collapses on these shapes:
 // its heuristic picks a kernel with (M/MT)*(C/NT) = 6 workgroups and no
 // K-split — 2.3% of the chip, 87 us (tools/gemm_probe.py). This kernel
-// splits K across blockIdx.z with fp32 atomicAdd epilogues: 64x64 output
-// tile per block, v_mfma_f32_16x16x32_bf16, A-tiles staged [k][m] in LDS
-// (coalesced) and read transposed by scalar LDS loads (frag build is
-// LDS-issue bound but fully hidden under the MFMA issue rate at this size).
+// splits K across blockIdx.z with fp32 atomicAdd epilogues.
+//
+// Geometry: 128x128 output tile per block, 4 waves (2x2) of 64x64 each,
+// v_mfma_f32_16x16x32_bf16. A-tiles are staged [k][m] in LDS exactly as
+// they lie in memory (coalesced 16B) and the transposed fragments are built
+// with scalar LDS reads — at 16 MFMAs per 8-fragment k-step the MFMA issue
+// rate covers the scalar-read issue cost (the 64x64-wave shape is what
+// makes that ratio work; at 32x32 the kernel was LDS-issue bound).
 //
 // B is split (B1 cols [0,C1), B2 cols [C1,C)) so gWcat = grad_gicat^T @
 // [messages | hidden] needs no concatenation.
@@ -27,11 +31,11 @@ __global__ __launch_bounds__(256) void wgrad_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ B1,
     const bf16* __restrict__ B2, float* __restrict__ out, int K, int M,
     int C, int C1, int kchunk) {
-  __shared__ __bf16 lds_a[KS][64];
-  __shared__ __bf16 lds_b[KS][64];
+  __shared__ __bf16 lds_a[KS][128];
+  __shared__ __bf16 lds_b[KS][128];
 
-  const int m0 = blockIdx.x * 64;
-  const int c0 = blockIdx.y * 64;
+  const int m0 = blockIdx.x * 128;
+  const int c0 = blockIdx.y * 128;
   const int k_begin = blockIdx.z * kchunk;
   const int k_end = min(K, k_begin + kchunk);
   const int tid = threadIdx.x;
@@ -40,54 +44,60 @@ __global__ __launch_bounds__(256) void wgrad_kernel(
   const int wm = wid >> 1;  // 2 row-waves (m), 2 col-waves (c)
   const int wc = wid & 1;
 
-  f32x4 acc[2][2] = {};
+  f32x4 acc[4][4] = {};
 
   for (int k0 = k_begin; k0 < k_end; k0 += KS) {
-    // stage A chunk [KS][64] at columns m0.. (row-major, coalesced 16B)
+    // stage [KS][128] tiles (row-major, coalesced 16B; 16 threads/row)
     {
-      const int row = tid >> 3;          // 0..31
-      const int coff = (tid & 7) * 8;    // element offset in tile row
-      const int k = k0 + row;
-      uint4v va = {};
-      if (k < k_end) va = *reinterpret_cast<const uint4v*>(A + (long)k * M + m0 + coff);
-      *reinterpret_cast<uint4v*>(&lds_a[row][coff]) = va;
-      uint4v vb = {};
-      if (k < k_end) {
-        const int cg = c0 + coff;
-        const bf16* src = (cg < C1) ? (B1 + (long)k * C1 + cg)
-                                    : (B2 + (long)k * (C - C1) + (cg - C1));
-        vb = *reinterpret_cast<const uint4v*>(src);
+      const int row = tid >> 4;          // 0..15
+      const int coff = (tid & 15) * 8;   // element offset in tile row
+#pragma unroll
+      for (int rr = 0; rr < KS; rr += 16) {
+        const int k = k0 + row + rr;
+        uint4v va = {}, vb = {};
+        if (k < k_end) {
+          const int mg = m0 + coff;
+          if (mg < M) va = *reinterpret_cast<const uint4v*>(A + (long)k * M + mg);
+          const int cg = c0 + coff;
+          if (cg < C) {
+            const bf16* src = (cg < C1) ? (B1 + (long)k * C1 + cg)
+                                        : (B2 + (long)k * (C - C1) + (cg - C1));
+            vb = *reinterpret_cast<const uint4v*>(src);
+          }
+        }
+        *reinterpret_cast<uint4v*>(&lds_a[row + rr][coff]) = va;
+        *reinterpret_cast<uint4v*>(&lds_b[row + rr][coff]) = vb;
       }
-      *reinterpret_cast<uint4v*>(&lds_b[row][coff]) = vb;
     }
     __syncthreads();
 
+    bf16x8 a_frag[4], b_frag[4];
 #pragma unroll
-    for (int fm = 0; fm < 2; ++fm) {
-      const int m_loc = wm * 32 + fm * 16 + (lane & 15);
-      bf16x8 a_frag;
+    for (int f = 0; f < 4; ++f) {
+      const int m_loc = wm * 64 + f * 16 + (lane & 15);
+      const int c_loc = wc * 64 + f * 16 + (lane & 15);
 #pragma unroll
-      for (int j = 0; j < 8; ++j) a_frag[j] = lds_a[(lane >> 4) * 8 + j][m_loc];
-#pragma unroll
-      for (int fc = 0; fc < 2; ++fc) {
-        const int c_loc = wc * 32 + fc * 16 + (lane & 15);
-        bf16x8 b_frag;
-#pragma unroll
-        for (int j = 0; j < 8; ++j) b_frag[j] = lds_b[(lane >> 4) * 8 + j][c_loc];
-        acc[fm][fc] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag,
-                                                              acc[fm][fc], 0, 0, 0);
+      for (int j = 0; j < 8; ++j) {
+        a_frag[f][j] = lds_a[(lane >> 4) * 8 + j][m_loc];
+        b_frag[f][j] = lds_b[(lane >> 4) * 8 + j][c_loc];
       }
     }
+#pragma unroll
+    for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+      for (int fc = 0; fc < 4; ++fc)
+        acc[fm][fc] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag[fm], b_frag[fc],
+                                                              acc[fm][fc], 0, 0, 0);
     __syncthreads();
   }
 
   // epilogue: D col = lane&15, row = (lane>>4)*4 + i; atomic fp32 accumulate
 #pragma unroll
-  for (int fm = 0; fm < 2; ++fm) {
-    const int m_base = m0 + wm * 32 + fm * 16 + (lane >> 4) * 4;
+  for (int fm = 0; fm < 4; ++fm) {
+    const int m_base = m0 + wm * 64 + fm * 16 + (lane >> 4) * 4;
 #pragma unroll
-    for (int fc = 0; fc < 2; ++fc) {
-      const int c = c0 + wc * 32 + fc * 16 + (lane & 15);
+    for (int fc = 0; fc < 4; ++fc) {
+      const int c = c0 + wc * 64 + fc * 16 + (lane & 15);
       if (c >= C) continue;
 #pragma unroll
       for (int i = 0; i < 4; ++i) {
@@ -101,12 +111,15 @@ __global__ __launch_bounds__(256) void wgrad_kernel(
 void launch_wgrad(const bf16* A, const bf16* B1, const bf16* B2, float* out,
                   int K, int M, int C, int C1, hipStream_t stream) {
   // size the K-split so the grid comfortably fills 256 CUs
-  const int tiles = ((M + 63) / 64) * ((C + 63) / 64);
+  const int tiles = ((M + 127) / 128) * ((C + 127) / 128);
   int zsplit = max(1, 512 / tiles);
   int kchunk = (K + zsplit - 1) / zsplit;
   kchunk = ((kchunk + KS - 1) / KS) * KS;
+  // keep >= 1 KiB-ish of K per block: beyond that the fp32 atomic epilogues
+  // (zsplit adds per output word) outweigh the extra parallelism
+  if (kchunk < 1024) kchunk = min(((K + KS - 1) / KS) * KS, 1024);
   zsplit = (K + kchunk - 1) / kchunk;
-  const dim3 grid((M + 63) / 64, (C + 63) / 64, zsplit);
+  const dim3 grid((M + 127) / 128, (C + 127) / 128, zsplit);
   hipLaunchKernelGGL(wgrad_kernel, grid, dim3(256), 0, stream, A, B1, B2, out,
                      K, M, C, C1, kchunk);
 }
