@@ -36,6 +36,23 @@ void launch_gemm_bias(const __hip_bfloat16*, const __hip_bfloat16*, const __hip_
                       int, int, hipStream_t);
 void launch_wgrad(const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*, float*,
                   int, int, int, int, hipStream_t);
+template <typename T>
+void launch_layernorm_fwd(const T*, const float*, const float*, T*, float*, float*, long, int,
+                          float, hipStream_t);
+template <typename T>
+void launch_layernorm_bwd(const T*, const T*, const float*, const float*, const float*, T*, long,
+                          int, hipStream_t);
+template <typename T>
+void launch_layernorm_wgrad(const T*, const T*, const float*, const float*, float*, float*, long,
+                            int, hipStream_t);
+template <typename T>
+void launch_bias_gelu_fwd(const T*, const float*, T*, long, int, hipStream_t);
+template <typename T>
+void launch_bias_gelu_bwd(const T*, const T*, const float*, T*, long, int, hipStream_t);
+template <typename T>
+void launch_softmax_mask_fwd(const T*, const int*, T*, long, int, int, float, hipStream_t);
+template <typename T>
+void launch_softmax_mask_bwd(const T*, const T*, T*, long, int, float, hipStream_t);
 
 #define CHECK_GPU(t) \
   TORCH_CHECK((t).is_cuda() && (t).is_contiguous(), #t " must be contiguous GPU tensor")
@@ -391,6 +408,117 @@ at::Tensor colsum(at::Tensor x) {
   return out;
 }
 
+
+// ---------------------------------------------------------------------------
+// Transformer kernels (csrc/transformer_kernels.hip)
+// ---------------------------------------------------------------------------
+
+std::vector<at::Tensor> layernorm_fwd(at::Tensor x, at::Tensor gamma, at::Tensor beta,
+                                      double eps) {
+  CHECK_GPU(x);
+  TORCH_CHECK(gamma.scalar_type() == at::kFloat && beta.scalar_type() == at::kFloat);
+  const int D = x.size(-1);
+  const long N = x.numel() / D;
+  auto y = at::empty_like(x);
+  auto mean = at::empty({N}, x.options().dtype(at::kFloat));
+  auto rstd = at::empty({N}, x.options().dtype(at::kFloat));
+  dispatch_float_bf16(x, "layernorm_fwd", [&](auto tag) {
+    using T = decltype(tag);
+    launch_layernorm_fwd<T>(ptr<T>(x), gamma.data_ptr<float>(), beta.data_ptr<float>(),
+                            mptr<T>(y), mean.data_ptr<float>(), rstd.data_ptr<float>(), N, D,
+                            (float)eps, cur_stream());
+  });
+  return {y, mean, rstd};
+}
+
+at::Tensor layernorm_bwd(at::Tensor dy, at::Tensor x, at::Tensor gamma, at::Tensor mean,
+                         at::Tensor rstd) {
+  CHECK_GPU(dy);
+  const int D = x.size(-1);
+  const long N = x.numel() / D;
+  auto dx = at::empty_like(x);
+  dispatch_float_bf16(x, "layernorm_bwd", [&](auto tag) {
+    using T = decltype(tag);
+    launch_layernorm_bwd<T>(ptr<T>(dy), ptr<T>(x), gamma.data_ptr<float>(),
+                            mean.data_ptr<float>(), rstd.data_ptr<float>(), mptr<T>(dx), N, D,
+                            cur_stream());
+  });
+  return dx;
+}
+
+std::vector<at::Tensor> layernorm_wgrad(at::Tensor dy, at::Tensor x, at::Tensor mean,
+                                        at::Tensor rstd) {
+  const int D = x.size(-1);
+  const long N = x.numel() / D;
+  auto dgamma = at::zeros({D}, x.options().dtype(at::kFloat));
+  auto dbeta = at::zeros({D}, x.options().dtype(at::kFloat));
+  dispatch_float_bf16(x, "layernorm_wgrad", [&](auto tag) {
+    using T = decltype(tag);
+    launch_layernorm_wgrad<T>(ptr<T>(dy), ptr<T>(x), mean.data_ptr<float>(),
+                              rstd.data_ptr<float>(), dgamma.data_ptr<float>(),
+                              dbeta.data_ptr<float>(), N, D, cur_stream());
+  });
+  return {dgamma, dbeta};
+}
+
+at::Tensor bias_gelu_fwd(at::Tensor x, at::Tensor bias) {
+  CHECK_GPU(x);
+  TORCH_CHECK(bias.scalar_type() == at::kFloat);
+  const int D = x.size(-1);
+  auto y = at::empty_like(x);
+  dispatch_float_bf16(x, "bias_gelu_fwd", [&](auto tag) {
+    using T = decltype(tag);
+    launch_bias_gelu_fwd<T>(ptr<T>(x), bias.data_ptr<float>(), mptr<T>(y), x.numel(), D,
+                            cur_stream());
+  });
+  return y;
+}
+
+at::Tensor bias_gelu_bwd(at::Tensor dy, at::Tensor x, at::Tensor bias) {
+  CHECK_GPU(dy);
+  const int D = x.size(-1);
+  auto dx = at::empty_like(x);
+  dispatch_float_bf16(x, "bias_gelu_bwd", [&](auto tag) {
+    using T = decltype(tag);
+    launch_bias_gelu_bwd<T>(ptr<T>(dy), ptr<T>(x), bias.data_ptr<float>(), mptr<T>(dx),
+                            x.numel(), D, cur_stream());
+  });
+  return dx;
+}
+
+at::Tensor softmax_mask_fwd(at::Tensor S, c10::optional<at::Tensor> valid, double scale) {
+  CHECK_GPU(S);
+  const int L = S.size(-1);
+  const long R = S.numel() / L;
+  auto P = at::empty_like(S);
+  const int* vptr = nullptr;
+  int rows_per_batch = 1;
+  if (valid.has_value()) {
+    TORCH_CHECK(valid->scalar_type() == at::kInt);
+    vptr = valid->data_ptr<int>();
+    rows_per_batch = (int)(R / valid->size(0));
+  }
+  dispatch_float_bf16(S, "softmax_mask_fwd", [&](auto tag) {
+    using T = decltype(tag);
+    launch_softmax_mask_fwd<T>(ptr<T>(S), vptr, mptr<T>(P), R, L, rows_per_batch, (float)scale,
+                               cur_stream());
+  });
+  return P;
+}
+
+at::Tensor softmax_mask_bwd(at::Tensor dP, at::Tensor P, double scale) {
+  CHECK_GPU(dP);
+  const int L = P.size(-1);
+  const long R = P.numel() / L;
+  auto dS = at::empty_like(P);
+  dispatch_float_bf16(P, "softmax_mask_bwd", [&](auto tag) {
+    using T = decltype(tag);
+    launch_softmax_mask_bwd<T>(ptr<T>(dP), ptr<T>(P), mptr<T>(dS), R, L, (float)scale,
+                               cur_stream());
+  });
+  return dS;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "deepdfa_amd MI355X (gfx950) kernels";
   m.def("embed4_fwd", &embed4_fwd);
@@ -406,5 +534,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gru_gates2_fwd", &gru_gates2_fwd);
   m.def("colsum", &colsum);
   m.def("ggnn_fused_fwd", &ggnn_fused_fwd);
+  m.def("layernorm_fwd", &layernorm_fwd);
+  m.def("layernorm_bwd", &layernorm_bwd);
+  m.def("layernorm_wgrad", &layernorm_wgrad);
+  m.def("bias_gelu_fwd", &bias_gelu_fwd);
+  m.def("bias_gelu_bwd", &bias_gelu_bwd);
+  m.def("softmax_mask_fwd", &softmax_mask_fwd);
+  m.def("softmax_mask_bwd", &softmax_mask_bwd);
   m.def("ggnn_fused_bwd", &ggnn_fused_bwd);
 }

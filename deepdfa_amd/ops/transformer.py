@@ -1,0 +1,97 @@
+"""Autograd ops for the transformer stacks.
+
+GPU: hand-written HIP kernels (csrc/transformer_kernels.hip), bf16/fp32 IO
+with fp32 statistics and fp32 LN/bias parameters (mixed-precision master
+weights). CPU: plain torch (also the numerics oracle).
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+
+from ._ext import load_ext
+
+
+class _LayerNorm(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, eps):
+        ext = load_ext(required=True)
+        y, mean, rstd = ext.layernorm_fwd(
+            x.contiguous(), weight.float().contiguous(), bias.float().contiguous(), eps
+        )
+        ctx.save_for_backward(x, weight, mean, rstd)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = load_ext(required=True)
+        x, weight, mean, rstd = ctx.saved_tensors
+        dy = dy.contiguous()
+        dx = ext.layernorm_bwd(dy, x.contiguous(), weight.float().contiguous(), mean, rstd)
+        dgamma, dbeta = ext.layernorm_wgrad(dy, x.contiguous(), mean, rstd)
+        return dx, dgamma.to(weight.dtype), dbeta.to(weight.dtype), None
+
+
+def layer_norm(x, weight, bias, eps: float = 1e-5):
+    if x.is_cuda:
+        return _LayerNorm.apply(x, weight, bias, eps)
+    return torch.nn.functional.layer_norm(
+        x.float(), (x.shape[-1],), weight.float(), bias.float(), eps
+    ).to(x.dtype)
+
+
+class _BiasGelu(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, bias):
+        ext = load_ext(required=True)
+        ctx.save_for_backward(x, bias)
+        return ext.bias_gelu_fwd(x.contiguous(), bias.float().contiguous())
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = load_ext(required=True)
+        x, bias = ctx.saved_tensors
+        dy = dy.contiguous()
+        dx = ext.bias_gelu_bwd(dy, x.contiguous(), bias.float().contiguous())
+        dbias = ext.colsum(dx.view(-1, dx.shape[-1]))
+        return dx, dbias.to(bias.dtype)
+
+
+def bias_gelu(x, bias):
+    if x.is_cuda:
+        return _BiasGelu.apply(x, bias)
+    return torch.nn.functional.gelu(x.float() + bias.float()).to(x.dtype)
+
+
+class _MaskedSoftmax(torch.autograd.Function):
+    """P = softmax(S * scale) with key positions >= valid[b] masked out.
+    S shaped (B, H, L, L); valid (B,) int32 or None."""
+
+    @staticmethod
+    def forward(ctx, S, valid, scale):
+        ext = load_ext(required=True)
+        P = ext.softmax_mask_fwd(S.contiguous(), valid, scale)
+        ctx.save_for_backward(P)
+        ctx.scale = scale
+        return P
+
+    @staticmethod
+    def backward(ctx, dP):
+        ext = load_ext(required=True)
+        (P,) = ctx.saved_tensors
+        dS = ext.softmax_mask_bwd(dP.contiguous(), P, ctx.scale)
+        return dS, None, None
+
+
+def masked_softmax(S: torch.Tensor, valid: Optional[torch.Tensor], scale: float):
+    if S.is_cuda:
+        return _MaskedSoftmax.apply(S, valid, scale)
+    s = S.float() * scale
+    if valid is not None:
+        L = S.shape[-1]
+        mask = torch.arange(L).view(1, 1, 1, L) >= valid.view(-1, 1, 1, 1)
+        s = s.masked_fill(mask, float("-inf"))
+    p = torch.softmax(s, dim=-1)
+    return torch.nan_to_num(p, nan=0.0).to(S.dtype)

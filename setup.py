@@ -23,6 +23,7 @@ ext = CUDAExtension(
         "csrc/flowgnn_kernels.hip",
         "csrc/gemm_bias.hip",
         "csrc/wgrad.hip",
+        "csrc/transformer_kernels.hip",
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
