@@ -1,13 +1,17 @@
 #!/usr/bin/env python3
-"""Flagship benchmark: DeepDFA flow-GNN training throughput (graphs/s).
+"""Flagship benchmark (driver contract: bench.py --gpus N --steps K --warmup W).
 
-Driver contract (see repo instructions): `python bench.py --gpus N --steps K
---warmup W`, launched under torch.distributed.run for N>1 (one rank per GPU
-over RCCL). Measures the headline metric of BASELINE.json — training
-throughput of the DeepDFA flow-GNN on Big-Vul-shaped synthetic CFG batches,
-batch 256 graphs per GPU (weak scaling), bf16 compute + fp32 master Adam.
-Baseline derived from the reference's published train time: ~810 graphs/s on
-RTX 3090.
+Models (BASELINE.json configs):
+  --model ddfa          DeepDFA flow-GNN training, batch 256 graphs/GPU
+                        (default; headline metric train_graphs_per_sec,
+                        baseline ~810 graphs/s on RTX 3090)
+  --model linevul       LineVul RoBERTa-base 512-token fine-tune, b=16/GPU
+                        (train_examples_per_sec; baseline 40.6 ex/s = 150,908
+                        fns x 10 epochs / 10h19m on RTX 3090)
+  --model linevul_ddfa  combined LineVul+DeepDFA (baseline 39.3 ex/s, 10h40m)
+
+bf16 compute + fp32 master Adam/AdamW, synthetic data (no-network env),
+weak scaling (per-GPU batch fixed), RCCL DDP via deepdfa_amd.parallel.
 """
 
 import argparse
@@ -25,8 +29,122 @@ from deepdfa_amd.graph import synthetic_cfg_batch  # noqa: E402
 from deepdfa_amd.models import FlowGNNGGNNModule  # noqa: E402
 from deepdfa_amd.parallel.ddp import DDPEngine, init_distributed, world_size  # noqa: E402
 
-BASELINE_GRAPHS_PER_SEC = 810.0
 FEAT = "_ABS_DATAFLOW_datatype_all_limitall_1000_limitsubkeys_1000"
+
+
+def build_ddfa(args, rank, device, use_cuda):
+    spec = parse_limits(FEAT)
+    model = FlowGNNGGNNModule(
+        feat=FEAT, input_dim=spec.input_dim, hidden_dim=32, n_steps=5, num_output_layers=3
+    ).to(device)
+    opt_fn = lambda capturable: torch.optim.Adam(  # noqa: E731
+        model.parameters(), lr=1e-3, weight_decay=1e-2, capturable=capturable, foreach=True
+    )
+    batches = [
+        synthetic_cfg_batch(args.batch, seed=1000 * rank + i, input_dim=spec.input_dim).to(device)
+        for i in range(args.n_batches)
+    ]
+    autocast = torch.autocast(device_type="cuda", dtype=torch.bfloat16) if use_cuda else None
+
+    def make_step(opt, ddp, set_to_none):
+        def step(i):
+            g = batches[i % len(batches)]
+            label = model.get_label(g)
+            if autocast is not None:
+                with autocast:
+                    logits = model(g, {})
+            else:
+                logits = model(g, {})
+            loss = model.loss_fn(logits.float(), label)
+            opt.zero_grad(set_to_none=set_to_none)
+            loss.backward()
+            ddp.finalize()
+            opt.step()
+            return loss
+
+        return step
+
+    meta = dict(
+        metric="train_graphs_per_sec",
+        unit="graphs/s",
+        baseline=810.0,
+        per_step_items=args.batch,
+        config={
+            "model": "DeepDFA-FlowGNN-GGNN(n_steps=5,D=128)",
+            "global_batch": args.batch * world_size(),
+            "seq_len": None,
+            "avg_nodes_per_graph": 45,
+            "parallelism": f"dp{world_size()}",
+        },
+    )
+    return model, opt_fn, make_step, meta, True  # graph-capture ok
+
+
+def build_linevul(args, rank, device, use_cuda, with_ddfa: bool):
+    from deepdfa_amd.models.linevul import Model
+    from deepdfa_amd.models.roberta import RobertaConfig
+
+    cfg = RobertaConfig()  # codebert-base geometry: 12 x 768, s=512
+    fg = None
+    if with_ddfa:
+        spec = parse_limits(FEAT)
+        fg = FlowGNNGGNNModule(
+            feat=FEAT, input_dim=spec.input_dim, hidden_dim=32, n_steps=5,
+            num_output_layers=3, encoder_mode=True,
+        )
+    model = Model(config=cfg, flowgnn_encoder=fg).to(device)
+    opt_fn = lambda capturable: torch.optim.AdamW(  # noqa: E731
+        model.parameters(), lr=2e-5, capturable=capturable, foreach=True
+    )
+    b, s = args.batch, 512
+    gen = torch.Generator().manual_seed(1234 + rank)
+    batches = []
+    for i in range(args.n_batches):
+        ids = torch.randint(3, cfg.vocab_size, (b, s), generator=gen)
+        lens = torch.randint(64, s + 1, (b,), generator=gen)
+        for j in range(b):
+            ids[j, lens[j] :] = 1
+        ids[:, 0] = 0  # CLS
+        labels = torch.randint(0, 2, (b,), generator=gen)
+        item = [ids.to(device), labels.to(device)]
+        if with_ddfa:
+            item.append(synthetic_cfg_batch(b, seed=5000 * rank + i).to(device))
+        batches.append(item)
+    autocast = torch.autocast(device_type="cuda", dtype=torch.bfloat16) if use_cuda else None
+
+    def make_step(opt, ddp, set_to_none):
+        def step(i):
+            item = batches[i % len(batches)]
+            ids, labels = item[0], item[1]
+            g = item[2] if with_ddfa else None
+            if autocast is not None:
+                with autocast:
+                    loss, _prob = model(ids, labels=labels, graphs=g)
+            else:
+                loss, _prob = model(ids, labels=labels, graphs=g)
+            opt.zero_grad(set_to_none=set_to_none)
+            loss.backward()
+            ddp.finalize()
+            torch.nn.utils.clip_grad_norm_(model.parameters(), 1.0)
+            opt.step()
+            return loss
+
+        return step
+
+    name = "LineVul+DeepDFA(CodeBERT-base+FlowGNN)" if with_ddfa else "LineVul(CodeBERT-base)"
+    meta = dict(
+        metric="train_examples_per_sec",
+        unit="examples/s",
+        baseline=39.3 if with_ddfa else 40.6,
+        per_step_items=b,
+        config={
+            "model": name,
+            "global_batch": b * world_size(),
+            "seq_len": s,
+            "parallelism": f"dp{world_size()}",
+        },
+    )
+    return model, opt_fn, make_step, meta, False
 
 
 def main():
@@ -34,68 +152,44 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=50)
     ap.add_argument("--warmup", type=int, default=10)
-    ap.add_argument("--batch", type=int, default=256)
-    ap.add_argument("--n-batches", type=int, default=4, help="pre-built batch pool size")
+    ap.add_argument("--model", choices=["ddfa", "linevul", "linevul_ddfa"], default="ddfa")
+    ap.add_argument("--batch", type=int, default=None)
+    ap.add_argument("--n-batches", type=int, default=4)
     ap.add_argument("--no-graph-capture", action="store_true")
     args = ap.parse_args()
+    if args.batch is None:
+        args.batch = 256 if args.model == "ddfa" else 16
 
     rank = init_distributed()
     ws = world_size()
     use_cuda = torch.cuda.is_available()
-    device = torch.device("cuda", int(os.environ.get("LOCAL_RANK", 0))) if use_cuda else torch.device("cpu")
+    device = (
+        torch.device("cuda", int(os.environ.get("LOCAL_RANK", 0))) if use_cuda else torch.device("cpu")
+    )
     if use_cuda:
         torch.cuda.set_device(device)
-    dtype = "bf16" if use_cuda else "fp32"
-
     torch.manual_seed(0)
-    spec = parse_limits(FEAT)
-    model = FlowGNNGGNNModule(
-        feat=FEAT, input_dim=spec.input_dim, hidden_dim=32, n_steps=5, num_output_layers=3
-    ).to(device)
-    opt = torch.optim.Adam(model.parameters(), lr=1e-3, weight_decay=1e-2)
-    ddp = DDPEngine(model, bucket_cap_mb=64.0)
 
-    # pre-build per-rank batch pool (different data per rank, fixed shapes per
-    # batch: synthetic Big-Vul-shaped CFGs, ~45 nodes/graph avg)
-    batches = [
-        synthetic_cfg_batch(args.batch, seed=1000 * rank + i, input_dim=spec.input_dim).to(device)
-        for i in range(args.n_batches)
-    ]
-
-    autocast = torch.autocast(device_type="cuda", dtype=torch.bfloat16) if use_cuda else None
-    # hipGraph capture: the flow-GNN step is ~180 small kernels, so replaying
-    # a captured graph removes the host launch overhead wholesale. One graph
-    # per pre-built batch (shapes differ); eager path kept for multi-GPU.
-    use_graphs = use_cuda and not args.no_graph_capture and ws == 1
-    if use_graphs:
-        opt = torch.optim.Adam(
-            model.parameters(), lr=1e-3, weight_decay=1e-2, capturable=True, foreach=True
+    if args.model == "ddfa":
+        model, opt_fn, make_step, meta, capture_ok = build_ddfa(args, rank, device, use_cuda)
+    else:
+        model, opt_fn, make_step, meta, capture_ok = build_linevul(
+            args, rank, device, use_cuda, with_ddfa=(args.model == "linevul_ddfa")
         )
 
-    def step(i: int):
-        g = batches[i % len(batches)]
-        label = model.get_label(g)
-        if autocast is not None:
-            with autocast:
-                logits = model(g, {})
-        else:
-            logits = model(g, {})
-        loss = model.loss_fn(logits.float(), label)
-        opt.zero_grad(set_to_none=not use_graphs)
-        loss.backward()
-        ddp.finalize()
-        opt.step()
-        return loss
+    use_graphs = use_cuda and capture_ok and not args.no_graph_capture and ws == 1
+    opt = opt_fn(capturable=use_graphs)
+    ddp = DDPEngine(model, bucket_cap_mb=64.0)
+    step = make_step(opt, ddp, set_to_none=not use_graphs)
 
-    # warmup
     for i in range(args.warmup):
         step(i)
 
-    graphs = []
     if use_graphs:
         torch.cuda.synchronize()
         pool = torch.cuda.graph_pool_handle()
-        for i in range(len(batches)):
+        graphs = []
+        for i in range(args.n_batches):
             cg = torch.cuda.CUDAGraph()
             with torch.cuda.graph(cg, pool=pool):
                 step(i)
@@ -118,37 +212,29 @@ def main():
     barrier_sync()
     elapsed = time.perf_counter() - t0
 
-    # max over ranks
     t = torch.tensor([elapsed], device=device if use_cuda else "cpu")
     if ws > 1:
         torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
     elapsed = float(t.item())
 
-    total_graphs = args.batch * args.steps * ws
-    gps = total_graphs / elapsed
+    value = meta["per_step_items"] * args.steps * ws / elapsed
     if rank == 0:
         print(
             json.dumps(
                 {
-                    "metric": "train_graphs_per_sec",
-                    "value": gps,
-                    "unit": "graphs/s",
+                    "metric": meta["metric"],
+                    "value": value,
+                    "unit": meta["unit"],
                     "n_gpus": ws,
                     "steps": args.steps,
                     "warmup": args.warmup,
                     "ms_per_step": elapsed / args.steps * 1000.0,
                     "higher_is_better": True,
                     "scaling": "weak",
-                    "vs_baseline": gps / BASELINE_GRAPHS_PER_SEC,
-                    "dtype": dtype,
+                    "vs_baseline": value / meta["baseline"],
+                    "dtype": "bf16" if use_cuda else "fp32",
                     "data": "synthetic",
-                    "config": {
-                        "model": "DeepDFA-FlowGNN-GGNN(n_steps=5,D=128)",
-                        "global_batch": args.batch * ws,
-                        "seq_len": None,
-                        "avg_nodes_per_graph": 45,
-                        "parallelism": f"dp{ws}",
-                    },
+                    "config": meta["config"],
                 }
             )
         )
