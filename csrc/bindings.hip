@@ -50,9 +50,11 @@ void launch_bias_gelu_fwd(const T*, const float*, T*, long, int, hipStream_t);
 template <typename T>
 void launch_bias_gelu_bwd(const T*, const T*, const float*, T*, long, int, hipStream_t);
 template <typename T>
-void launch_softmax_mask_fwd(const T*, const int*, T*, long, int, int, float, hipStream_t);
+void launch_softmax_mask_fwd(const T*, const int*, T*, T*, long, int, int, float, float,
+                             unsigned long long, hipStream_t);
 template <typename T>
-void launch_softmax_mask_bwd(const T*, const T*, T*, long, int, float, hipStream_t);
+void launch_softmax_mask_bwd(const T*, const T*, T*, long, int, float, float,
+                             unsigned long long, hipStream_t);
 
 #define CHECK_GPU(t) \
   TORCH_CHECK((t).is_cuda() && (t).is_contiguous(), #t " must be contiguous GPU tensor")
@@ -486,11 +488,22 @@ at::Tensor bias_gelu_bwd(at::Tensor dy, at::Tensor x, at::Tensor bias) {
   return dx;
 }
 
-at::Tensor softmax_mask_fwd(at::Tensor S, c10::optional<at::Tensor> valid, double scale) {
+static void check_softmax_geom(const at::Tensor& t, int L) {
+  const int vec = t.scalar_type() == at::kBFloat16 ? 8 : 4;
+  TORCH_CHECK(L % vec == 0 && L <= 64 * vec * 4, "unsupported softmax row length ", L);
+}
+
+// returns {P (pre-dropout, saved for bwd), Pd (post-dropout, feeds P@V)};
+// Pd aliases P when dropout_p == 0
+std::vector<at::Tensor> softmax_mask_fwd(at::Tensor S, c10::optional<at::Tensor> valid,
+                                         double scale, double dropout_p, int64_t seed) {
   CHECK_GPU(S);
   const int L = S.size(-1);
   const long R = S.numel() / L;
+  check_softmax_geom(S, L);
   auto P = at::empty_like(S);
+  const bool drop = dropout_p > 0.0;
+  auto Pd = drop ? at::empty_like(S) : P;
   const int* vptr = nullptr;
   int rows_per_batch = 1;
   if (valid.has_value()) {
@@ -500,21 +513,24 @@ at::Tensor softmax_mask_fwd(at::Tensor S, c10::optional<at::Tensor> valid, doubl
   }
   dispatch_float_bf16(S, "softmax_mask_fwd", [&](auto tag) {
     using T = decltype(tag);
-    launch_softmax_mask_fwd<T>(ptr<T>(S), vptr, mptr<T>(P), R, L, rows_per_batch, (float)scale,
-                               cur_stream());
+    launch_softmax_mask_fwd<T>(ptr<T>(S), vptr, mptr<T>(P), drop ? mptr<T>(Pd) : nullptr, R, L,
+                               rows_per_batch, (float)scale, (float)dropout_p,
+                               (unsigned long long)seed, cur_stream());
   });
-  return P;
+  return {P, Pd};
 }
 
-at::Tensor softmax_mask_bwd(at::Tensor dP, at::Tensor P, double scale) {
-  CHECK_GPU(dP);
+at::Tensor softmax_mask_bwd(at::Tensor dPd, at::Tensor P, double scale, double dropout_p,
+                            int64_t seed) {
+  CHECK_GPU(dPd);
   const int L = P.size(-1);
   const long R = P.numel() / L;
+  check_softmax_geom(P, L);
   auto dS = at::empty_like(P);
   dispatch_float_bf16(P, "softmax_mask_bwd", [&](auto tag) {
     using T = decltype(tag);
-    launch_softmax_mask_bwd<T>(ptr<T>(dP), ptr<T>(P), mptr<T>(dS), R, L, (float)scale,
-                               cur_stream());
+    launch_softmax_mask_bwd<T>(ptr<T>(dPd), ptr<T>(P), mptr<T>(dS), R, L, (float)scale,
+                               (float)dropout_p, (unsigned long long)seed, cur_stream());
   });
   return dS;
 }

@@ -143,14 +143,22 @@ __device__ __forceinline__ float gelu_grad_f(float v) {
   return cdf + v * pdf;
 }
 
+// 16-B vectors per lane (8 bf16 / 4 fp32); D % VEC == 0 (host-checked)
 template <typename T>
 __global__ void bias_gelu_fwd_kernel(const T* __restrict__ x,
                                      const float* __restrict__ bias,
                                      T* __restrict__ y, long total, int D) {
-  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
-       i += (long)gridDim.x * blockDim.x) {
+  constexpr int VEC = 16 / sizeof(T);
+  const long nvec = total / VEC;
+  for (long iv = (long)blockIdx.x * blockDim.x + threadIdx.x; iv < nvec;
+       iv += (long)gridDim.x * blockDim.x) {
+    const long i = iv * VEC;
     const int d = (int)(i % D);
-    y[i] = ff<T>(gelu_f(tf(x[i]) + bias[d]));
+    T vx[VEC], vy[VEC];
+    *reinterpret_cast<ulonglong2*>(vx) = *reinterpret_cast<const ulonglong2*>(x + i);
+#pragma unroll
+    for (int u = 0; u < VEC; ++u) vy[u] = ff<T>(gelu_f(tf(vx[u]) + bias[d + u]));
+    *reinterpret_cast<ulonglong2*>(y + i) = *reinterpret_cast<ulonglong2*>(vy);
   }
 }
 
@@ -159,10 +167,19 @@ __global__ void bias_gelu_bwd_kernel(const T* __restrict__ dy,
                                      const T* __restrict__ x,
                                      const float* __restrict__ bias,
                                      T* __restrict__ dx, long total, int D) {
-  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
-       i += (long)gridDim.x * blockDim.x) {
+  constexpr int VEC = 16 / sizeof(T);
+  const long nvec = total / VEC;
+  for (long iv = (long)blockIdx.x * blockDim.x + threadIdx.x; iv < nvec;
+       iv += (long)gridDim.x * blockDim.x) {
+    const long i = iv * VEC;
     const int d = (int)(i % D);
-    dx[i] = ff<T>(tf(dy[i]) * gelu_grad_f(tf(x[i]) + bias[d]));
+    T vdy[VEC], vx[VEC], vdx[VEC];
+    *reinterpret_cast<ulonglong2*>(vdy) = *reinterpret_cast<const ulonglong2*>(dy + i);
+    *reinterpret_cast<ulonglong2*>(vx) = *reinterpret_cast<const ulonglong2*>(x + i);
+#pragma unroll
+    for (int u = 0; u < VEC; ++u)
+      vdx[u] = ff<T>(tf(vdy[u]) * gelu_grad_f(tf(vx[u]) + bias[d + u]));
+    *reinterpret_cast<ulonglong2*>(dx + i) = *reinterpret_cast<ulonglong2*>(vdx);
   }
 }
 
@@ -172,45 +189,120 @@ __global__ void bias_gelu_bwd_kernel(const T* __restrict__ dy,
 // P = softmax(S * scale + mask). One wave per row.
 // ---------------------------------------------------------------------------
 
+// Stateless per-element dropout RNG (regenerated identically in backward —
+// no mask storage). splitmix64-style hash of (seed, element index).
+__device__ __forceinline__ bool keep_mask(unsigned long long idx,
+                                          unsigned long long seed,
+                                          unsigned p24) {
+  unsigned long long x = idx * 0x9E3779B97F4A7C15ull ^ seed;
+  x ^= x >> 32;
+  x *= 0xD6E8FEB86659FD93ull;
+  x ^= x >> 32;
+  x *= 0xD6E8FEB86659FD93ull;
+  x ^= x >> 29;
+  return ((unsigned)x & 0xFFFFFFu) >= p24;
+}
+
+// 8 bf16 / 8 fp32-pair loads per lane (one row of L<=512 in registers).
 template <typename T>
 __global__ void softmax_mask_fwd_kernel(const T* __restrict__ S,
                                         const int* __restrict__ valid,
-                                        T* __restrict__ P, long R, int L,
-                                        int rows_per_batch, float scale) {
+                                        T* __restrict__ P, T* __restrict__ Pd,
+                                        long R, int L, int rows_per_batch,
+                                        float scale, float dropout_p,
+                                        unsigned long long seed) {
   const long row = (long)(blockIdx.x * (blockDim.x / WAVE)) + threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
   if (row >= R) return;
   const int vl = valid ? valid[row / rows_per_batch] : L;
   const T* sr = S + row * L;
+  constexpr int VEC = 16 / sizeof(T);  // one 16-B vector store per lane
+  const int niter = (L + WAVE * VEC - 1) / (WAVE * VEC);
+  float v[4][VEC];  // supports L <= 4*WAVE*VEC (host-checked)
   float m = -3.4e38f;
-  for (int j = lane; j < vl; j += WAVE) m = fmaxf(m, tf(sr[j]) * scale);
+  for (int it = 0; it < niter; ++it) {
+    const int j0 = (it * WAVE + lane) * VEC;
+#pragma unroll
+    for (int u = 0; u < VEC; ++u) {
+      const int j = j0 + u;
+      v[it][u] = (j < vl) ? tf(sr[j]) * scale : -3.4e38f;
+      m = fmaxf(m, v[it][u]);
+    }
+  }
   m = wave_max(m);
   float sum = 0.f;
-  for (int j = lane; j < vl; j += WAVE) sum += __expf(tf(sr[j]) * scale - m);
+  for (int it = 0; it < niter; ++it)
+#pragma unroll
+    for (int u = 0; u < VEC; ++u) {
+      v[it][u] = (v[it][u] > -3.0e38f) ? __expf(v[it][u] - m) : 0.f;
+      sum += v[it][u];
+    }
   sum = wave_sum(sum);
   const float inv = (sum > 0.f) ? 1.0f / sum : 0.f;
-  T* pr = P + row * L;
-  for (int j = lane; j < L; j += WAVE)
-    pr[j] = ff<T>(j < vl ? __expf(tf(sr[j]) * scale - m) * inv : 0.f);
+  const unsigned p24 = (unsigned)(dropout_p * 16777216.0f);
+  const float dscale = (dropout_p > 0.f) ? 1.0f / (1.0f - dropout_p) : 1.0f;
+  for (int it = 0; it < niter; ++it) {
+    const int j0 = (it * WAVE + lane) * VEC;
+    T pv[VEC], pdv[VEC];
+#pragma unroll
+    for (int u = 0; u < VEC; ++u) {
+      const float p = v[it][u] * inv;
+      pv[u] = ff<T>(p);
+      if (Pd) {
+        const bool keep = (p24 == 0) || keep_mask(row * L + j0 + u, seed, p24);
+        pdv[u] = ff<T>(keep ? p * dscale : 0.f);
+      }
+    }
+    if (j0 < L) {
+      *reinterpret_cast<ulonglong2*>(P + row * L + j0) = *reinterpret_cast<ulonglong2*>(pv);
+      if (Pd)
+        *reinterpret_cast<ulonglong2*>(Pd + row * L + j0) = *reinterpret_cast<ulonglong2*>(pdv);
+    }
+  }
 }
 
-// dS = scale * P * (dP - sum_j(dP * P))
+// dS = scale * P * (dP - sum_j(dP * P)); dP = dPd * mask / (1-p) regenerated
 template <typename T>
-__global__ void softmax_mask_bwd_kernel(const T* __restrict__ dP,
+__global__ void softmax_mask_bwd_kernel(const T* __restrict__ dPd,
                                         const T* __restrict__ P,
                                         T* __restrict__ dS, long R, int L,
-                                        float scale) {
+                                        float scale, float dropout_p,
+                                        unsigned long long seed) {
   const long row = (long)(blockIdx.x * (blockDim.x / WAVE)) + threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
   if (row >= R) return;
-  const T* dpr = dP + row * L;
+  const T* dpr = dPd + row * L;
   const T* pr = P + row * L;
+  constexpr int VEC = 16 / sizeof(T);
+  const int niter = (L + WAVE * VEC - 1) / (WAVE * VEC);
+  const unsigned p24 = (unsigned)(dropout_p * 16777216.0f);
+  const float dscale = (dropout_p > 0.f) ? 1.0f / (1.0f - dropout_p) : 1.0f;
+  float dp[4][VEC], p[4][VEC];
   float dot = 0.f;
-  for (int j = lane; j < L; j += WAVE) dot += tf(dpr[j]) * tf(pr[j]);
+  for (int it = 0; it < niter; ++it) {
+    const int j0 = (it * WAVE + lane) * VEC;
+    if (j0 < L) {
+#pragma unroll
+      for (int u = 0; u < VEC; ++u) {
+        const int j = j0 + u;
+        const bool keep = (p24 == 0) || keep_mask(row * L + j, seed, p24);
+        dp[it][u] = keep ? tf(dpr[j]) * dscale : 0.f;
+        p[it][u] = tf(pr[j]);
+        dot += dp[it][u] * p[it][u];
+      }
+    }
+  }
   dot = wave_sum(dot);
-  T* dsr = dS + row * L;
-  for (int j = lane; j < L; j += WAVE)
-    dsr[j] = ff<T>(scale * tf(pr[j]) * (tf(dpr[j]) - dot));
+  for (int it = 0; it < niter; ++it) {
+    const int j0 = (it * WAVE + lane) * VEC;
+    if (j0 < L) {
+      T out[VEC];
+#pragma unroll
+      for (int u = 0; u < VEC; ++u)
+        out[u] = ff<T>(scale * p[it][u] * (dp[it][u] - dot));
+      *reinterpret_cast<ulonglong2*>(dS + row * L + j0) = *reinterpret_cast<ulonglong2*>(out);
+    }
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -259,7 +351,8 @@ template <typename T>
 void launch_bias_gelu_fwd(const T* x, const float* bias, T* y, long total,
                           int D, hipStream_t stream) {
   const int block = 256;
-  const int grid = (int)min((total + block - 1) / block, (long)4096);
+  const long nvec = total / (16 / sizeof(T));
+  const int grid = (int)min((nvec + block - 1) / block, (long)4096);
   if (grid)
     hipLaunchKernelGGL(bias_gelu_fwd_kernel<T>, dim3(grid), dim3(block), 0,
                        stream, x, bias, y, total, D);
@@ -269,31 +362,34 @@ template <typename T>
 void launch_bias_gelu_bwd(const T* dy, const T* x, const float* bias, T* dx,
                           long total, int D, hipStream_t stream) {
   const int block = 256;
-  const int grid = (int)min((total + block - 1) / block, (long)4096);
+  const long nvec = total / (16 / sizeof(T));
+  const int grid = (int)min((nvec + block - 1) / block, (long)4096);
   if (grid)
     hipLaunchKernelGGL(bias_gelu_bwd_kernel<T>, dim3(grid), dim3(block), 0,
                        stream, dy, x, bias, dx, total, D);
 }
 
 template <typename T>
-void launch_softmax_mask_fwd(const T* S, const int* valid, T* P, long R, int L,
-                             int rows_per_batch, float scale,
+void launch_softmax_mask_fwd(const T* S, const int* valid, T* P, T* Pd, long R,
+                             int L, int rows_per_batch, float scale,
+                             float dropout_p, unsigned long long seed,
                              hipStream_t stream) {
   const int grid = (int)((R + ROWS_PER_BLOCK - 1) / ROWS_PER_BLOCK);
   if (grid)
     hipLaunchKernelGGL(softmax_mask_fwd_kernel<T>, dim3(grid),
-                       dim3(WAVE * ROWS_PER_BLOCK), 0, stream, S, valid, P, R, L,
-                       rows_per_batch, scale);
+                       dim3(WAVE * ROWS_PER_BLOCK), 0, stream, S, valid, P, Pd,
+                       R, L, rows_per_batch, scale, dropout_p, seed);
 }
 
 template <typename T>
-void launch_softmax_mask_bwd(const T* dP, const T* P, T* dS, long R, int L,
-                             float scale, hipStream_t stream) {
+void launch_softmax_mask_bwd(const T* dPd, const T* P, T* dS, long R, int L,
+                             float scale, float dropout_p,
+                             unsigned long long seed, hipStream_t stream) {
   const int grid = (int)((R + ROWS_PER_BLOCK - 1) / ROWS_PER_BLOCK);
   if (grid)
     hipLaunchKernelGGL(softmax_mask_bwd_kernel<T>, dim3(grid),
-                       dim3(WAVE * ROWS_PER_BLOCK), 0, stream, dP, P, dS, R, L,
-                       scale);
+                       dim3(WAVE * ROWS_PER_BLOCK), 0, stream, dPd, P, dS, R, L,
+                       scale, dropout_p, seed);
 }
 
 #define INSTANTIATE_TK(T)                                                    \
@@ -310,10 +406,12 @@ void launch_softmax_mask_bwd(const T* dP, const T* P, T* dS, long R, int L,
                                         int, hipStream_t);                    \
   template void launch_bias_gelu_bwd<T>(const T*, const T*, const float*, T*, \
                                         long, int, hipStream_t);              \
-  template void launch_softmax_mask_fwd<T>(const T*, const int*, T*, long,    \
-                                           int, int, float, hipStream_t);     \
+  template void launch_softmax_mask_fwd<T>(const T*, const int*, T*, T*,     \
+                                           long, int, int, float, float,      \
+                                           unsigned long long, hipStream_t);  \
   template void launch_softmax_mask_bwd<T>(const T*, const T*, T*, long, int, \
-                                           float, hipStream_t);
+                                           float, float, unsigned long long,  \
+                                           hipStream_t);
 
 INSTANTIATE_TK(float)
 INSTANTIATE_TK(__hip_bfloat16)

@@ -29,7 +29,7 @@ from typing import Optional
 import torch
 from torch import nn
 
-from ..ops.transformer import bias_gelu, layer_norm, masked_softmax
+from ..ops.transformer import bias_gelu, layer_norm, masked_softmax_dropout
 
 
 @dataclass
@@ -92,7 +92,8 @@ class RobertaSelfAttention(nn.Module):
         self.query = nn.Linear(cfg.hidden_size, cfg.hidden_size)
         self.key = nn.Linear(cfg.hidden_size, cfg.hidden_size)
         self.value = nn.Linear(cfg.hidden_size, cfg.hidden_size)
-        self.dropout = nn.Dropout(cfg.attention_probs_dropout_prob)
+        self.dropout_p = cfg.attention_probs_dropout_prob
+        self.dropout = nn.Dropout(cfg.attention_probs_dropout_prob)  # CPU path
 
     def forward(self, x, valid: Optional[torch.Tensor], output_attentions: bool = False):
         B, L, D = x.shape
@@ -103,8 +104,9 @@ class RobertaSelfAttention(nn.Module):
 
         q, k, v = split(self.query(x)), split(self.key(x)), split(self.value(x))
         scores = torch.matmul(q, k.transpose(-1, -2))
-        probs = masked_softmax(scores, valid, 1.0 / math.sqrt(d))
-        ctx = torch.matmul(self.dropout(probs), v)
+        p = self.dropout_p if self.training else 0.0
+        probs, probs_dropped = masked_softmax_dropout(scores, valid, 1.0 / math.sqrt(d), p)
+        ctx = torch.matmul(probs_dropped, v)
         out = ctx.transpose(1, 2).reshape(B, L, D)
         return (out, probs) if output_attentions else (out, None)
 
@@ -190,8 +192,8 @@ class RobertaModel(nn.Module):
             attention_mask = input_ids.ne(self.config.pad_token_id)
         valid = attention_mask.sum(dim=1).to(torch.int32)
         x = self.embeddings(input_ids)
-        if x.is_cuda:
-            x = x.to(torch.bfloat16)
+        # dtype policy: fp32 master params; under torch.autocast(bf16) the
+        # projections emit bf16 and the custom kernels follow the input dtype
         hidden, all_probs = self.encoder(x, valid, output_attentions)
         return hidden, all_probs
 

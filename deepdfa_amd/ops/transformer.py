@@ -65,33 +65,59 @@ def bias_gelu(x, bias):
     return torch.nn.functional.gelu(x.float() + bias.float()).to(x.dtype)
 
 
-class _MaskedSoftmax(torch.autograd.Function):
-    """P = softmax(S * scale) with key positions >= valid[b] masked out.
-    S shaped (B, H, L, L); valid (B,) int32 or None."""
+_seed_state = {"base": None, "ctr": 0}
+
+
+def _next_seed() -> int:
+    if _seed_state["base"] is None:
+        _seed_state["base"] = torch.initial_seed() & 0x7FFFFFFF
+    _seed_state["ctr"] += 1
+    return (_seed_state["base"] * 2654435761 + _seed_state["ctr"]) & 0x7FFFFFFFFFFFFFFF
+
+
+class _MaskedSoftmaxDropout(torch.autograd.Function):
+    """(P, Pd) = softmax(S * scale) with key positions >= valid[b] masked,
+    plus fused attention dropout (mask regenerated from the seed in
+    backward; no mask storage). Pd is the differentiable output feeding
+    P @ V; P is the pre-dropout probabilities (analysis output, marked
+    non-differentiable)."""
 
     @staticmethod
-    def forward(ctx, S, valid, scale):
+    def forward(ctx, S, valid, scale, dropout_p):
         ext = load_ext(required=True)
-        P = ext.softmax_mask_fwd(S.contiguous(), valid, scale)
+        seed = _next_seed() if dropout_p > 0 else 0
+        P, Pd = ext.softmax_mask_fwd(S.contiguous(), valid, scale, dropout_p, seed)
         ctx.save_for_backward(P)
         ctx.scale = scale
-        return P
+        ctx.dropout_p = dropout_p
+        ctx.seed = seed
+        ctx.mark_non_differentiable(P)
+        return P, Pd
 
     @staticmethod
-    def backward(ctx, dP):
+    def backward(ctx, _dP_unused, dPd):
         ext = load_ext(required=True)
         (P,) = ctx.saved_tensors
-        dS = ext.softmax_mask_bwd(dP.contiguous(), P, ctx.scale)
-        return dS, None, None
+        dS = ext.softmax_mask_bwd(dPd.contiguous(), P, ctx.scale, ctx.dropout_p, ctx.seed)
+        return dS, None, None, None
 
 
-def masked_softmax(S: torch.Tensor, valid: Optional[torch.Tensor], scale: float):
+def masked_softmax_dropout(
+    S: torch.Tensor, valid: Optional[torch.Tensor], scale: float, dropout_p: float = 0.0
+):
+    """Returns (P pre-dropout, Pd post-dropout)."""
     if S.is_cuda:
-        return _MaskedSoftmax.apply(S, valid, scale)
+        return _MaskedSoftmaxDropout.apply(S, valid, scale, dropout_p)
     s = S.float() * scale
     if valid is not None:
         L = S.shape[-1]
         mask = torch.arange(L).view(1, 1, 1, L) >= valid.view(-1, 1, 1, 1)
         s = s.masked_fill(mask, float("-inf"))
     p = torch.softmax(s, dim=-1)
-    return torch.nan_to_num(p, nan=0.0).to(S.dtype)
+    p = torch.nan_to_num(p, nan=0.0).to(S.dtype)
+    pd = torch.nn.functional.dropout(p, dropout_p) if dropout_p > 0 else p
+    return p, pd
+
+
+def masked_softmax(S: torch.Tensor, valid: Optional[torch.Tensor], scale: float):
+    return masked_softmax_dropout(S, valid, scale, 0.0)[0]

@@ -78,6 +78,29 @@ def test_masked_softmax_gpu(dev, dtype, tol):
     assert torch.allclose(S.grad.float().cpu(), S2.grad, atol=max(tol, 1e-3), rtol=5e-2)
 
 
+def test_masked_softmax_dropout_gpu(dev):
+    from deepdfa_amd.ops.transformer import masked_softmax_dropout
+
+    torch.manual_seed(5)
+    B, H, L = 2, 4, 512
+    S = torch.randn(B, H, L, L, device=dev, dtype=torch.bfloat16, requires_grad=True)
+    valid = torch.tensor([512, 300], dtype=torch.int32, device=dev)
+    P, Pd = masked_softmax_dropout(S, valid, 0.125, 0.5)
+    keep = Pd != 0
+    frac = keep[P > 1e-4].float().mean().item()
+    assert 0.4 < frac < 0.6, frac  # ~half kept
+    # kept entries scaled by 1/(1-p)
+    ratio = (Pd[keep].float() / P[keep].float().clamp_min(1e-8)).median().item()
+    assert abs(ratio - 2.0) < 0.1, ratio
+    # backward: dS must match the manual composition with the SAME mask
+    go = torch.randn_like(Pd)
+    Pd.backward(go)
+    dP = go.float() * keep.float() * 2.0
+    dot = (dP * P.float()).sum(-1, keepdim=True)
+    dS_ref = 0.125 * P.float() * (dP - dot)
+    assert torch.allclose(S.grad.float(), dS_ref, atol=2e-3, rtol=5e-2)
+
+
 def test_encoder_gpu_matches_cpu(dev):
     from deepdfa_amd.models.roberta import RobertaConfig, RobertaModel, init_roberta_weights
 
@@ -93,7 +116,7 @@ def test_encoder_gpu_matches_cpu(dev):
     with torch.no_grad():
         ref, _ = model(ids)
     model_gpu = model.to(dev)
-    with torch.no_grad():
+    with torch.no_grad(), torch.autocast(device_type="cuda", dtype=torch.bfloat16):
         out, _ = model_gpu(ids.to(dev))  # bf16 path
     mask = ids.ne(1).unsqueeze(-1)
     diff = ((out.float().cpu() - ref) * mask).abs().max().item()
