@@ -503,7 +503,9 @@ std::vector<at::Tensor> softmax_mask_fwd(at::Tensor S, c10::optional<at::Tensor>
   check_softmax_geom(S, L);
   auto P = at::empty_like(S);
   const bool drop = dropout_p > 0.0;
-  auto Pd = drop ? at::empty_like(S) : P;
+  // p == 0: Pd shares storage with P via an explicit alias so autograd can
+  // keep the two outputs' differentiability separate
+  auto Pd = drop ? at::empty_like(S) : at::alias(P);
   const int* vptr = nullptr;
   int rows_per_batch = 1;
   if (valid.has_value()) {

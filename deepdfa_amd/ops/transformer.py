@@ -120,4 +120,6 @@ def masked_softmax_dropout(
 
 
 def masked_softmax(S: torch.Tensor, valid: Optional[torch.Tensor], scale: float):
-    return masked_softmax_dropout(S, valid, scale, 0.0)[0]
+    # index 1 (the post-dropout output) is the differentiable one; with
+    # dropout 0 it aliases the probabilities
+    return masked_softmax_dropout(S, valid, scale, 0.0)[1]
