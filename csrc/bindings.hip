@@ -51,7 +51,14 @@ template <typename T>
 void launch_bias_gelu_bwd(const T*, const T*, const float*, T*, long, int, hipStream_t);
 template <typename T>
 void launch_softmax_mask_fwd(const T*, const int*, T*, T*, long, int, int, float, float,
-                             unsigned long long, hipStream_t);
+                             unsigned long long, int, int, hipStream_t);
+template <typename T>
+void launch_rmsnorm_fwd(const T*, const float*, T*, float*, long, int, float, hipStream_t);
+template <typename T>
+void launch_rmsnorm_bwd(const T*, const T*, const float*, const float*, T*, long, int,
+                        hipStream_t);
+template <typename T>
+void launch_rmsnorm_wgrad(const T*, const T*, const float*, float*, long, int, hipStream_t);
 template <typename T>
 void launch_softmax_mask_bwd(const T*, const T*, T*, long, int, float, float,
                              unsigned long long, hipStream_t);
@@ -496,7 +503,8 @@ static void check_softmax_geom(const at::Tensor& t, int L) {
 // returns {P (pre-dropout, saved for bwd), Pd (post-dropout, feeds P@V)};
 // Pd aliases P when dropout_p == 0
 std::vector<at::Tensor> softmax_mask_fwd(at::Tensor S, c10::optional<at::Tensor> valid,
-                                         double scale, double dropout_p, int64_t seed) {
+                                         double scale, double dropout_p, int64_t seed,
+                                         bool causal) {
   CHECK_GPU(S);
   const int L = S.size(-1);
   const long R = S.numel() / L;
@@ -517,7 +525,8 @@ std::vector<at::Tensor> softmax_mask_fwd(at::Tensor S, c10::optional<at::Tensor>
     using T = decltype(tag);
     launch_softmax_mask_fwd<T>(ptr<T>(S), vptr, mptr<T>(P), drop ? mptr<T>(Pd) : nullptr, R, L,
                                rows_per_batch, (float)scale, (float)dropout_p,
-                               (unsigned long long)seed, cur_stream());
+                               (unsigned long long)seed, causal ? 1 : 0, (int)S.size(-2),
+                               cur_stream());
   });
   return {P, Pd};
 }
@@ -535,6 +544,47 @@ at::Tensor softmax_mask_bwd(at::Tensor dPd, at::Tensor P, double scale, double d
                                (float)dropout_p, (unsigned long long)seed, cur_stream());
   });
   return dS;
+}
+
+
+std::vector<at::Tensor> rmsnorm_fwd(at::Tensor x, at::Tensor gamma, double eps) {
+  CHECK_GPU(x);
+  TORCH_CHECK(gamma.scalar_type() == at::kFloat);
+  const int D = x.size(-1);
+  const long N = x.numel() / D;
+  auto y = at::empty_like(x);
+  auto rstd = at::empty({N}, x.options().dtype(at::kFloat));
+  dispatch_float_bf16(x, "rmsnorm_fwd", [&](auto tag) {
+    using T = decltype(tag);
+    launch_rmsnorm_fwd<T>(ptr<T>(x), gamma.data_ptr<float>(), mptr<T>(y),
+                          rstd.data_ptr<float>(), N, D, (float)eps, cur_stream());
+  });
+  return {y, rstd};
+}
+
+at::Tensor rmsnorm_bwd(at::Tensor dy, at::Tensor x, at::Tensor gamma, at::Tensor rstd) {
+  CHECK_GPU(dy);
+  const int D = x.size(-1);
+  const long N = x.numel() / D;
+  auto dx = at::empty_like(x);
+  dispatch_float_bf16(x, "rmsnorm_bwd", [&](auto tag) {
+    using T = decltype(tag);
+    launch_rmsnorm_bwd<T>(ptr<T>(dy), ptr<T>(x), gamma.data_ptr<float>(),
+                          rstd.data_ptr<float>(), mptr<T>(dx), N, D, cur_stream());
+  });
+  return dx;
+}
+
+at::Tensor rmsnorm_wgrad(at::Tensor dy, at::Tensor x, at::Tensor rstd) {
+  const int D = x.size(-1);
+  const long N = x.numel() / D;
+  auto dgamma = at::zeros({D}, x.options().dtype(at::kFloat));
+  dispatch_float_bf16(x, "rmsnorm_wgrad", [&](auto tag) {
+    using T = decltype(tag);
+    launch_rmsnorm_wgrad<T>(ptr<T>(dy), ptr<T>(x), rstd.data_ptr<float>(),
+                            dgamma.data_ptr<float>(), N, D, cur_stream());
+  });
+  return dgamma;
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -558,6 +608,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bias_gelu_fwd", &bias_gelu_fwd);
   m.def("bias_gelu_bwd", &bias_gelu_bwd);
   m.def("softmax_mask_fwd", &softmax_mask_fwd);
+  m.def("rmsnorm_fwd", &rmsnorm_fwd);
+  m.def("rmsnorm_bwd", &rmsnorm_bwd);
+  m.def("rmsnorm_wgrad", &rmsnorm_wgrad);
   m.def("softmax_mask_bwd", &softmax_mask_bwd);
   m.def("ggnn_fused_bwd", &ggnn_fused_bwd);
 }

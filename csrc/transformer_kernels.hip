@@ -130,6 +130,75 @@ __global__ void layernorm_wgrad_kernel(const T* __restrict__ dy,
   atomicAdd(dbeta + d, db);
 }
 
+
+// ---------------------------------------------------------------------------
+// RMSNorm (T5LayerNorm): y = x * rsqrt(mean(x^2) + eps) * gamma.
+// T5 computes the variance in fp32 and casts the normalized value to the
+// weight dtype before multiplying; here: fp32 math, IO dtype T.
+// ---------------------------------------------------------------------------
+
+template <typename T>
+__global__ void rmsnorm_fwd_kernel(const T* __restrict__ x,
+                                   const float* __restrict__ gamma,
+                                   T* __restrict__ y, float* __restrict__ rstd,
+                                   long N, int D, float eps) {
+  const long row = (long)(blockIdx.x * (blockDim.x / WAVE)) + threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  if (row >= N) return;
+  const T* xr = x + row * D;
+  float s2 = 0.f;
+  for (int d = lane; d < D; d += WAVE) {
+    const float v = tf(xr[d]);
+    s2 += v * v;
+  }
+  s2 = wave_sum(s2);
+  const float rs = rsqrtf(s2 / D + eps);
+  if (lane == 0) rstd[row] = rs;
+  T* yr = y + row * D;
+  for (int d = lane; d < D; d += WAVE) yr[d] = ff<T>(tf(xr[d]) * rs * gamma[d]);
+}
+
+// dx = rs * (dyg - xhat * mean_d(dyg * xhat)); xhat = x * rs; dyg = dy*gamma
+template <typename T>
+__global__ void rmsnorm_bwd_kernel(const T* __restrict__ dy,
+                                   const T* __restrict__ x,
+                                   const float* __restrict__ gamma,
+                                   const float* __restrict__ rstd,
+                                   T* __restrict__ dx, long N, int D) {
+  const long row = (long)(blockIdx.x * (blockDim.x / WAVE)) + threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  if (row >= N) return;
+  const T* dyr = dy + row * D;
+  const T* xr = x + row * D;
+  const float rs = rstd[row];
+  float c2 = 0.f;
+  for (int d = lane; d < D; d += WAVE) {
+    const float xh = tf(xr[d]) * rs;
+    c2 += tf(dyr[d]) * gamma[d] * xh;
+  }
+  c2 = wave_sum(c2) / D;
+  T* dxr = dx + row * D;
+  for (int d = lane; d < D; d += WAVE) {
+    const float xh = tf(xr[d]) * rs;
+    dxr[d] = ff<T>(rs * (tf(dyr[d]) * gamma[d] - xh * c2));
+  }
+}
+
+template <typename T>
+__global__ void rmsnorm_wgrad_kernel(const T* __restrict__ dy,
+                                     const T* __restrict__ x,
+                                     const float* __restrict__ rstd,
+                                     float* __restrict__ dgamma, long N, int D,
+                                     int rows_per_block) {
+  const int d = blockIdx.x * blockDim.x + threadIdx.x;
+  if (d >= D) return;
+  const long r0 = (long)blockIdx.y * rows_per_block;
+  const long r1 = min((long)(r0 + rows_per_block), N);
+  float dg = 0.f;
+  for (long r = r0; r < r1; ++r) dg += tf(dy[r * D + d]) * tf(x[r * D + d]) * rstd[r];
+  atomicAdd(dgamma + d, dg);
+}
+
 // ---------------------------------------------------------------------------
 // bias + GELU (erf form, matching torch F.gelu / HF "gelu")
 // ---------------------------------------------------------------------------
@@ -210,11 +279,13 @@ __global__ void softmax_mask_fwd_kernel(const T* __restrict__ S,
                                         T* __restrict__ P, T* __restrict__ Pd,
                                         long R, int L, int rows_per_batch,
                                         float scale, float dropout_p,
-                                        unsigned long long seed) {
+                                        unsigned long long seed, int causal,
+                                        int Lq) {
   const long row = (long)(blockIdx.x * (blockDim.x / WAVE)) + threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
   if (row >= R) return;
-  const int vl = valid ? valid[row / rows_per_batch] : L;
+  int vl = valid ? valid[row / rows_per_batch] : L;
+  if (causal) vl = min(vl, (int)(row % Lq) + 1);
   const T* sr = S + row * L;
   constexpr int VEC = 16 / sizeof(T);  // one 16-B vector store per lane
   const int niter = (L + WAVE * VEC - 1) / (WAVE * VEC);
@@ -373,12 +444,12 @@ template <typename T>
 void launch_softmax_mask_fwd(const T* S, const int* valid, T* P, T* Pd, long R,
                              int L, int rows_per_batch, float scale,
                              float dropout_p, unsigned long long seed,
-                             hipStream_t stream) {
+                             int causal, int Lq, hipStream_t stream) {
   const int grid = (int)((R + ROWS_PER_BLOCK - 1) / ROWS_PER_BLOCK);
   if (grid)
     hipLaunchKernelGGL(softmax_mask_fwd_kernel<T>, dim3(grid),
                        dim3(WAVE * ROWS_PER_BLOCK), 0, stream, S, valid, P, Pd,
-                       R, L, rows_per_batch, scale, dropout_p, seed);
+                       R, L, rows_per_batch, scale, dropout_p, seed, causal, Lq);
 }
 
 template <typename T>
@@ -390,6 +461,39 @@ void launch_softmax_mask_bwd(const T* dPd, const T* P, T* dS, long R, int L,
     hipLaunchKernelGGL(softmax_mask_bwd_kernel<T>, dim3(grid),
                        dim3(WAVE * ROWS_PER_BLOCK), 0, stream, dPd, P, dS, R, L,
                        scale, dropout_p, seed);
+}
+
+template <typename T>
+void launch_rmsnorm_fwd(const T* x, const float* gamma, T* y, float* rstd,
+                        long N, int D, float eps, hipStream_t stream) {
+  const int grid = (int)((N + ROWS_PER_BLOCK - 1) / ROWS_PER_BLOCK);
+  if (grid)
+    hipLaunchKernelGGL(rmsnorm_fwd_kernel<T>, dim3(grid),
+                       dim3(WAVE * ROWS_PER_BLOCK), 0, stream, x, gamma, y,
+                       rstd, N, D, eps);
+}
+
+template <typename T>
+void launch_rmsnorm_bwd(const T* dy, const T* x, const float* gamma,
+                        const float* rstd, T* dx, long N, int D,
+                        hipStream_t stream) {
+  const int grid = (int)((N + ROWS_PER_BLOCK - 1) / ROWS_PER_BLOCK);
+  if (grid)
+    hipLaunchKernelGGL(rmsnorm_bwd_kernel<T>, dim3(grid),
+                       dim3(WAVE * ROWS_PER_BLOCK), 0, stream, dy, x, gamma,
+                       rstd, dx, N, D);
+}
+
+template <typename T>
+void launch_rmsnorm_wgrad(const T* dy, const T* x, const float* rstd,
+                          float* dgamma, long N, int D, hipStream_t stream) {
+  const int block = 256;
+  const int colb = (D + block - 1) / block;
+  const int rows_per_block = 64;
+  const int rowb = (int)((N + rows_per_block - 1) / rows_per_block);
+  if (colb && rowb)
+    hipLaunchKernelGGL(rmsnorm_wgrad_kernel<T>, dim3(colb, rowb), dim3(block),
+                       0, stream, dy, x, rstd, dgamma, N, D, rows_per_block);
 }
 
 #define INSTANTIATE_TK(T)                                                    \
@@ -408,7 +512,15 @@ void launch_softmax_mask_bwd(const T* dPd, const T* P, T* dS, long R, int L,
                                         long, int, hipStream_t);              \
   template void launch_softmax_mask_fwd<T>(const T*, const int*, T*, T*,     \
                                            long, int, int, float, float,      \
-                                           unsigned long long, hipStream_t);  \
+                                           unsigned long long, int, int,      \
+                                           hipStream_t);                      \
+  template void launch_rmsnorm_fwd<T>(const T*, const float*, T*, float*,     \
+                                      long, int, float, hipStream_t);         \
+  template void launch_rmsnorm_bwd<T>(const T*, const T*, const float*,       \
+                                      const float*, T*, long, int,            \
+                                      hipStream_t);                           \
+  template void launch_rmsnorm_wgrad<T>(const T*, const T*, const float*,     \
+                                        float*, long, int, hipStream_t);      \
   template void launch_softmax_mask_bwd<T>(const T*, const T*, T*, long, int, \
                                            float, float, unsigned long long,  \
                                            hipStream_t);

@@ -83,10 +83,10 @@ class _MaskedSoftmaxDropout(torch.autograd.Function):
     non-differentiable)."""
 
     @staticmethod
-    def forward(ctx, S, valid, scale, dropout_p):
+    def forward(ctx, S, valid, scale, dropout_p, causal=False):
         ext = load_ext(required=True)
         seed = _next_seed() if dropout_p > 0 else 0
-        P, Pd = ext.softmax_mask_fwd(S.contiguous(), valid, scale, dropout_p, seed)
+        P, Pd = ext.softmax_mask_fwd(S.contiguous(), valid, scale, dropout_p, seed, causal)
         ctx.save_for_backward(P)
         ctx.scale = scale
         ctx.dropout_p = dropout_p
@@ -99,24 +99,59 @@ class _MaskedSoftmaxDropout(torch.autograd.Function):
         ext = load_ext(required=True)
         (P,) = ctx.saved_tensors
         dS = ext.softmax_mask_bwd(dPd.contiguous(), P, ctx.scale, ctx.dropout_p, ctx.seed)
-        return dS, None, None, None
+        return dS, None, None, None, None
 
 
 def masked_softmax_dropout(
-    S: torch.Tensor, valid: Optional[torch.Tensor], scale: float, dropout_p: float = 0.0
+    S: torch.Tensor,
+    valid: Optional[torch.Tensor],
+    scale: float,
+    dropout_p: float = 0.0,
+    causal: bool = False,
 ):
     """Returns (P pre-dropout, Pd post-dropout)."""
     if S.is_cuda:
-        return _MaskedSoftmaxDropout.apply(S, valid, scale, dropout_p)
+        return _MaskedSoftmaxDropout.apply(S, valid, scale, dropout_p, causal)
     s = S.float() * scale
+    L = S.shape[-1]
     if valid is not None:
-        L = S.shape[-1]
         mask = torch.arange(L).view(1, 1, 1, L) >= valid.view(-1, 1, 1, 1)
         s = s.masked_fill(mask, float("-inf"))
+    if causal:
+        Lq = S.shape[-2]
+        cmask = torch.arange(L).view(1, L) > torch.arange(Lq).view(Lq, 1)
+        s = s.masked_fill(cmask.view(1, 1, Lq, L), float("-inf"))
     p = torch.softmax(s, dim=-1)
     p = torch.nan_to_num(p, nan=0.0).to(S.dtype)
     pd = torch.nn.functional.dropout(p, dropout_p) if dropout_p > 0 else p
     return p, pd
+
+
+class _RMSNorm(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, eps):
+        ext = load_ext(required=True)
+        y, rstd = ext.rmsnorm_fwd(x.contiguous(), weight.float().contiguous(), eps)
+        ctx.save_for_backward(x, weight, rstd)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = load_ext(required=True)
+        x, weight, rstd = ctx.saved_tensors
+        dy = dy.contiguous()
+        dx = ext.rmsnorm_bwd(dy, x.contiguous(), weight.float().contiguous(), rstd)
+        dgamma = ext.rmsnorm_wgrad(dy, x.contiguous(), rstd)
+        return dx, dgamma.to(weight.dtype), None
+
+
+def rms_norm(x, weight, eps: float = 1e-6):
+    """T5LayerNorm: no mean subtraction, no bias."""
+    if x.is_cuda:
+        return _RMSNorm.apply(x, weight, eps)
+    v = x.float()
+    y = v * torch.rsqrt(v.pow(2).mean(-1, keepdim=True) + eps)
+    return (y * weight.float()).to(x.dtype)
 
 
 def masked_softmax(S: torch.Tensor, valid: Optional[torch.Tensor], scale: float):
