@@ -258,18 +258,28 @@ __global__ void bias_gelu_bwd_kernel(const T* __restrict__ dy,
 // P = softmax(S * scale + mask). One wave per row.
 // ---------------------------------------------------------------------------
 
-// Stateless per-element dropout RNG (regenerated identically in backward —
-// no mask storage). splitmix64-style hash of (seed, element index).
+// Stateless dropout RNG (regenerated identically in backward — no mask
+// storage). One 32-bit hash yields keep-bytes for FOUR consecutive
+// elements (per-element 64-bit hashing measurably slowed the softmax).
+__device__ __forceinline__ unsigned hash4(unsigned long long quad_idx,
+                                          unsigned long long seed) {
+  unsigned h = (unsigned)quad_idx * 2654435761u + (unsigned)(quad_idx >> 32) * 40503u +
+               (unsigned)seed + (unsigned)(seed >> 32) * 97u;
+  h ^= h >> 16;
+  h *= 0x7feb352du;
+  h ^= h >> 15;
+  h *= 0x846ca68bu;
+  h ^= h >> 16;
+  return h;
+}
+
+// keep element `idx` with probability 1-p (p8 = p * 256, byte compare)
 __device__ __forceinline__ bool keep_mask(unsigned long long idx,
                                           unsigned long long seed,
-                                          unsigned p24) {
-  unsigned long long x = idx * 0x9E3779B97F4A7C15ull ^ seed;
-  x ^= x >> 32;
-  x *= 0xD6E8FEB86659FD93ull;
-  x ^= x >> 32;
-  x *= 0xD6E8FEB86659FD93ull;
-  x ^= x >> 29;
-  return ((unsigned)x & 0xFFFFFFu) >= p24;
+                                          unsigned p8) {
+  const unsigned h = hash4(idx >> 2, seed);
+  const unsigned byte = (h >> (8 * ((unsigned)idx & 3))) & 0xFF;
+  return byte >= p8;
 }
 
 // 8 bf16 / 8 fp32-pair loads per lane (one row of L<=512 in registers).
@@ -310,7 +320,7 @@ __global__ void softmax_mask_fwd_kernel(const T* __restrict__ S,
     }
   sum = wave_sum(sum);
   const float inv = (sum > 0.f) ? 1.0f / sum : 0.f;
-  const unsigned p24 = (unsigned)(dropout_p * 16777216.0f);
+  const unsigned p8 = (unsigned)(dropout_p * 256.0f);
   const float dscale = (dropout_p > 0.f) ? 1.0f / (1.0f - dropout_p) : 1.0f;
   for (int it = 0; it < niter; ++it) {
     const int j0 = (it * WAVE + lane) * VEC;
@@ -320,7 +330,7 @@ __global__ void softmax_mask_fwd_kernel(const T* __restrict__ S,
       const float p = v[it][u] * inv;
       pv[u] = ff<T>(p);
       if (Pd) {
-        const bool keep = (p24 == 0) || keep_mask(row * L + j0 + u, seed, p24);
+        const bool keep = (p8 == 0) || keep_mask(row * L + j0 + u, seed, p8);
         pdv[u] = ff<T>(keep ? p * dscale : 0.f);
       }
     }
@@ -346,7 +356,7 @@ __global__ void softmax_mask_bwd_kernel(const T* __restrict__ dPd,
   const T* pr = P + row * L;
   constexpr int VEC = 16 / sizeof(T);
   const int niter = (L + WAVE * VEC - 1) / (WAVE * VEC);
-  const unsigned p24 = (unsigned)(dropout_p * 16777216.0f);
+  const unsigned p8 = (unsigned)(dropout_p * 256.0f);
   const float dscale = (dropout_p > 0.f) ? 1.0f / (1.0f - dropout_p) : 1.0f;
   float dp[4][VEC], p[4][VEC];
   float dot = 0.f;
@@ -356,7 +366,7 @@ __global__ void softmax_mask_bwd_kernel(const T* __restrict__ dPd,
 #pragma unroll
       for (int u = 0; u < VEC; ++u) {
         const int j = j0 + u;
-        const bool keep = (p24 == 0) || keep_mask(row * L + j, seed, p24);
+        const bool keep = (p8 == 0) || keep_mask(row * L + j, seed, p8);
         dp[it][u] = keep ? tf(dpr[j]) * dscale : 0.f;
         p[it][u] = tf(pr[j]);
         dot += dp[it][u] * p[it][u];
