@@ -1,0 +1,84 @@
+"""Aux-parity tests: clipper union ops, git diff, statement labels, joern
+gating, profiling report."""
+
+import torch
+
+from deepdfa_amd.models.clipper import relu_union, simple_union, union_reduce
+from deepdfa_amd.pipeline.evaluate import get_dep_add_lines
+from deepdfa_amd.utils.git import code2diff, gitdiff
+
+
+def test_union_ops_properties():
+    """Reference clipper.py embedded-test properties: exact bitwise union on
+    {0,1}, commutative, idempotent, differentiable."""
+    a = torch.tensor([0.0, 0.0, 1.0, 1.0], requires_grad=True)
+    b = torch.tensor([0.0, 1.0, 0.0, 1.0], requires_grad=True)
+    expect = torch.tensor([0.0, 1.0, 1.0, 1.0])
+    for fn in (simple_union, relu_union):
+        out = fn(a, b)
+        assert torch.equal(out.detach(), expect)
+        assert torch.equal(fn(a, b).detach(), fn(b, a).detach())
+        assert torch.equal(fn(a, a).detach(), a.detach())
+        out.sum().backward()
+        assert a.grad is not None
+        a.grad = None
+        b.grad = None
+    xs = torch.stack([a.detach(), b.detach(), torch.zeros(4)])
+    assert torch.equal(union_reduce(xs, 0), expect)
+
+
+def test_code2diff():
+    old = "int f() {\n  int x = 1;\n  return x;\n}"
+    new = "int f() {\n  int x = 1;\n  if (x > 0) x = 2;\n  return x;\n}"
+    d = code2diff(old, new)
+    assert d["added"] == [3]
+    assert d["removed"] == []
+    assert "+  if (x > 0) x = 2;" in d["diff"]
+    assert "---" in gitdiff(old, new)
+
+
+def test_dep_add_lines():
+    from tests.test_pipeline import hand_cpg
+
+    cpg = hand_cpg()
+    # fix added line 2 (x = 1): lines using x's definition depend on it
+    dep = get_dep_add_lines(cpg, {2})
+    assert 2 in dep
+    assert 3 in dep  # y = x + 2 uses x@line2
+    # line 4 (x = y) redefines x; y depends on x@2 only through line 3
+    dep5 = get_dep_add_lines(cpg, {4})
+    assert 5 in dep5  # x < y uses x@line4
+
+
+def test_joern_gating():
+    from deepdfa_amd.pipeline.joern import joern_available, run_joern, scrub_ansi
+
+    assert scrub_ansi("\x1b[31mred\x1b[0m") == "red"
+    if not joern_available():
+        import pytest
+
+        with pytest.raises(RuntimeError):
+            run_joern("x.c", "/tmp")
+
+
+def test_report_profiling(tmp_path):
+    import json
+    import subprocess
+    import sys
+    import os
+
+    prof = tmp_path / "profiledata.jsonl"
+    with open(prof, "w") as f:
+        for _ in range(3):
+            f.write(json.dumps({"flops": 2e9, "macs": 1e9, "params": 1e6,
+                                "batch_size": 4, "time_ms": 10.0}) + "\n")
+    time_f = tmp_path / "timedata.jsonl"
+    with open(time_f, "w") as f:
+        for _ in range(3):
+            f.write(json.dumps({"batch_size": 4, "time_ms": 8.0}) + "\n")
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, os.path.join(repo, "scripts", "report_profiling.py"), str(tmp_path)],
+        capture_output=True, text=True, check=True,
+    ).stdout
+    assert "gflops" in out and "ms/example" in out

@@ -32,10 +32,12 @@ def hand_cpg():
         10: {"_label": "IDENTIFIER", "name": "x", "order": 1, "typeFullName": "int"},
         11: {"_label": "IDENTIFIER", "name": "y", "order": 2},
         12: {"_label": "CALL", "name": "<operator>.lessThan", "code": "x < y", "lineNumber": 5},
+        13: {"_label": "IDENTIFIER", "name": "x", "order": 1},
+        14: {"_label": "IDENTIFIER", "name": "y", "order": 2},
     }
     edges = [
         (1, 2, "AST"), (1, 3, "AST"), (4, 5, "AST"), (4, 6, "AST"), (6, 7, "AST"),
-        (6, 8, "AST"), (9, 10, "AST"), (9, 11, "AST"),
+        (6, 8, "AST"), (9, 10, "AST"), (9, 11, "AST"), (12, 13, "AST"), (12, 14, "AST"),
         (0, 1, "CFG"), (1, 4, "CFG"), (4, 9, "CFG"), (9, 12, "CFG"),
     ]
     return CPG(nodes, edges)
