@@ -59,6 +59,19 @@ void launch_rmsnorm_bwd(const T*, const T*, const float*, const float*, T*, long
                         hipStream_t);
 template <typename T>
 void launch_rmsnorm_wgrad(const T*, const T*, const float*, float*, long, int, hipStream_t);
+void launch_flash_fwd(const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*,
+                      const int*, const float*, __hip_bfloat16*, float*, int, int, int, float,
+                      int, unsigned, unsigned long long, hipStream_t);
+void launch_flash_dterm(const __hip_bfloat16*, const __hip_bfloat16*, float*, int, int, int,
+                        hipStream_t);
+void launch_flash_dq(const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*,
+                     const __hip_bfloat16*, const int*, const float*, const float*, const float*,
+                     __hip_bfloat16*, float*, int, int, int, float, int, unsigned,
+                     unsigned long long, hipStream_t);
+void launch_flash_dkv(const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*,
+                      const __hip_bfloat16*, const int*, const float*, const float*, const float*,
+                      __hip_bfloat16*, __hip_bfloat16*, int, int, int, float, int, unsigned,
+                      unsigned long long, hipStream_t);
 template <typename T>
 void launch_softmax_mask_bwd(const T*, const T*, T*, long, int, float, float,
                              unsigned long long, hipStream_t);
@@ -587,6 +600,76 @@ at::Tensor rmsnorm_wgrad(at::Tensor dy, at::Tensor x, at::Tensor rstd) {
   return dgamma;
 }
 
+
+// ---------------------------------------------------------------------------
+// Flash attention (csrc/flash_attn.hip): head_dim 64, (B, L, H*64) layout
+// ---------------------------------------------------------------------------
+
+static const int* opt_valid_ptr(const c10::optional<at::Tensor>& valid) {
+  if (!valid.has_value()) return nullptr;
+  TORCH_CHECK(valid->scalar_type() == at::kInt && valid->is_contiguous());
+  return valid->data_ptr<int>();
+}
+
+std::vector<at::Tensor> flash_attn_fwd(at::Tensor Q, at::Tensor K, at::Tensor V, int64_t H,
+                                       c10::optional<at::Tensor> valid,
+                                       c10::optional<at::Tensor> bias, double scale, bool causal,
+                                       double dropout_p, int64_t seed) {
+  CHECK_GPU(Q);
+  CHECK_GPU(K);
+  CHECK_GPU(V);
+  TORCH_CHECK(Q.scalar_type() == at::kBFloat16, "flash attention is bf16");
+  const int B = Q.size(0), L = Q.size(1);
+  TORCH_CHECK(Q.size(2) == H * 64, "head_dim must be 64");
+  TORCH_CHECK(L % 64 == 0, "L must be a multiple of 64");
+  auto O = at::empty_like(Q);
+  auto lse = at::empty({B, H, L}, Q.options().dtype(at::kFloat));
+  const float* bptr = nullptr;
+  if (bias.has_value()) {
+    TORCH_CHECK(bias->scalar_type() == at::kFloat && bias->is_contiguous());
+    TORCH_CHECK(bias->numel() == H * (long)L * L, "bias must be (H, L, L)");
+    bptr = bias->data_ptr<float>();
+  }
+  launch_flash_fwd(ptr<bf16_t>(Q), ptr<bf16_t>(K), ptr<bf16_t>(V), opt_valid_ptr(valid), bptr,
+                   mptr<bf16_t>(O), lse.data_ptr<float>(), B, H, L, (float)scale, causal ? 1 : 0,
+                   (unsigned)(dropout_p * 256.0), (unsigned long long)seed, cur_stream());
+  return {O, lse};
+}
+
+std::vector<at::Tensor> flash_attn_bwd(at::Tensor dO, at::Tensor Q, at::Tensor K, at::Tensor V,
+                                       at::Tensor O, at::Tensor lse, int64_t H,
+                                       c10::optional<at::Tensor> valid,
+                                       c10::optional<at::Tensor> bias, double scale, bool causal,
+                                       double dropout_p, int64_t seed, bool need_dbias) {
+  CHECK_GPU(dO);
+  const int B = Q.size(0), L = Q.size(1);
+  auto stream = cur_stream();
+  auto Dterm = at::empty({B, H, L}, Q.options().dtype(at::kFloat));
+  launch_flash_dterm(ptr<bf16_t>(dO), ptr<bf16_t>(O), Dterm.data_ptr<float>(), B, H, L, stream);
+  auto dQ = at::empty_like(Q);
+  auto dK = at::empty_like(K);
+  auto dV = at::empty_like(V);
+  const float* bptr = nullptr;
+  if (bias.has_value()) bptr = bias->data_ptr<float>();
+  at::Tensor dBias;
+  float* dbias_ptr = nullptr;
+  if (need_dbias) {
+    TORCH_CHECK(bias.has_value());
+    dBias = at::zeros({H, L, L}, Q.options().dtype(at::kFloat));
+    dbias_ptr = dBias.data_ptr<float>();
+  }
+  launch_flash_dq(ptr<bf16_t>(Q), ptr<bf16_t>(K), ptr<bf16_t>(V), ptr<bf16_t>(dO),
+                  opt_valid_ptr(valid), bptr, lse.data_ptr<float>(), Dterm.data_ptr<float>(),
+                  mptr<bf16_t>(dQ), dbias_ptr, B, H, L, (float)scale, causal ? 1 : 0,
+                  (unsigned)(dropout_p * 256.0), (unsigned long long)seed, stream);
+  launch_flash_dkv(ptr<bf16_t>(Q), ptr<bf16_t>(K), ptr<bf16_t>(V), ptr<bf16_t>(dO),
+                   opt_valid_ptr(valid), bptr, lse.data_ptr<float>(), Dterm.data_ptr<float>(),
+                   mptr<bf16_t>(dK), mptr<bf16_t>(dV), B, H, L, (float)scale, causal ? 1 : 0,
+                   (unsigned)(dropout_p * 256.0), (unsigned long long)seed, stream);
+  if (need_dbias) return {dQ, dK, dV, dBias};
+  return {dQ, dK, dV};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "deepdfa_amd MI355X (gfx950) kernels";
   m.def("embed4_fwd", &embed4_fwd);
@@ -611,6 +694,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
   m.def("rmsnorm_bwd", &rmsnorm_bwd);
   m.def("rmsnorm_wgrad", &rmsnorm_wgrad);
+  m.def("flash_attn_fwd", &flash_attn_fwd);
+  m.def("flash_attn_bwd", &flash_attn_bwd);
   m.def("softmax_mask_bwd", &softmax_mask_bwd);
   m.def("ggnn_fused_bwd", &ggnn_fused_bwd);
 }

@@ -1,0 +1,777 @@
+// Flash attention for MI355X (gfx950), head_dim = 64.
+//
+// Replaces the materialized-P attention path (QK^T bmm + softmax + P@V bmm
+// + the layout transposes around them — together ~10 ms of the 27 ms
+// LineVul step, rocprof in profiles/) with one fused kernel per direction.
+//
+// Shapes: Q/K/V/O are (B, L, H*64) — the projections' natural output — so
+// no transpose/contiguous copies exist anywhere in the attention path.
+// Supports: suffix-padding valid lengths per batch row, causal masking
+// (T5 decoder), additive position bias (T5 relative attention, fp32
+// (H, L, L)), fused stateless dropout (mask regenerated in backward),
+// softmax scale, and the log-sum-exp save for the backward pass.
+//
+// Tiling (cdna_hip_programming.md §B attention ladder, adapted to d=64):
+// one block = 4 waves = 128 query rows of one (b, h); KV tiles of 64 keys
+// staged in LDS (K natural [key][d] with XOR swizzle; V transposed to
+// [d][key] at staging so the PV B-fragments are contiguous reads).
+// v_mfma_f32_16x16x32_bf16 throughout; swapped-operand QK^T
+// (S^T = mfma(K, Q)) keeps the softmax row statistics lane-local to a
+// 16-lane column group (two shfl_xor hops per reduction).
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+#define WAVE 64
+#define TK 64    // keys per KV tile
+#define TQW 32   // query rows per wave
+#define NWAVE 4  // waves per block -> 128 q rows per block
+
+using bf16 = __hip_bfloat16;
+using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+using uint4v = __attribute__((ext_vector_type(4))) unsigned int;
+
+__device__ __forceinline__ unsigned fa_hash4(unsigned long long quad_idx,
+                                             unsigned long long seed) {
+  unsigned h = (unsigned)quad_idx * 2654435761u + (unsigned)(quad_idx >> 32) * 40503u +
+               (unsigned)seed + (unsigned)(seed >> 32) * 97u;
+  h ^= h >> 16;
+  h *= 0x7feb352du;
+  h ^= h >> 15;
+  h *= 0x846ca68bu;
+  h ^= h >> 16;
+  return h;
+}
+
+__device__ __forceinline__ bool fa_keep(unsigned long long idx,
+                                        unsigned long long seed, unsigned p8) {
+  const unsigned h = fa_hash4(idx >> 2, seed);
+  return ((h >> (8 * ((unsigned)idx & 3))) & 0xFF) >= p8;
+}
+
+// LDS addressing: 128-B rows, XOR swizzle vs 16-lane b128 fragment groups
+__device__ __forceinline__ int fa_swz(int row, int byte) {
+  return row * 128 + (byte ^ ((row & 7) << 4));
+}
+
+__global__ __launch_bounds__(256) void flash_fwd_kernel(
+    const bf16* __restrict__ Q, const bf16* __restrict__ K,
+    const bf16* __restrict__ V, const int* __restrict__ valid,
+    const float* __restrict__ bias, bf16* __restrict__ O,
+    float* __restrict__ lse, int B, int H, int L, float scale, int causal,
+    unsigned p8, unsigned long long seed) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  char* k_lds = smem;                // [TK][64] bf16 swizzled = 8 KiB
+  char* v_lds = smem + TK * 128;     // [64][TK] bf16 (transposed) = 8 KiB
+
+  const int bh = blockIdx.y;          // (b, h)
+  const int b = bh / H;
+  const int h = bh % H;
+  const int q0 = blockIdx.x * (NWAVE * TQW);
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wid = tid >> 6;
+  const int qw = q0 + wid * TQW;      // this wave's first q row
+  const long HD = (long)H * 64;
+  const int vl = valid ? valid[b] : L;
+
+  // ---- load Q fragments (B-operand layout: lane holds Q[qrow][8 d]) ----
+  // frag index [fq][ks]: qrow = qw + fq*16 + (lane&15), d = ks*32 + (lane>>4)*8
+  bf16x8 q_frag[2][2];
+#pragma unroll
+  for (int fq = 0; fq < 2; ++fq) {
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      const int qrow = qw + fq * 16 + (lane & 15);
+      const int d = ks * 32 + (lane >> 4) * 8;
+      if (qrow < L)
+        q_frag[fq][ks] = *reinterpret_cast<const bf16x8*>(
+            Q + ((long)b * L + qrow) * HD + (long)h * 64 + d);
+      else
+        q_frag[fq][ks] = bf16x8{};
+    }
+  }
+
+  // softmax state per (fq, its lane-column qrow)
+  float m_st[2] = {-3.4e38f, -3.4e38f};
+  float l_st[2] = {0.f, 0.f};
+  f32x4 o_acc[2][4] = {};  // [fq][fd]: O rows (l>>4)*4+i of fq block, col fd*16+(l&15)
+
+  const int kv_end = causal ? min(vl, q0 + NWAVE * TQW) : vl;
+  for (int kv0 = 0; kv0 < kv_end; kv0 += TK) {
+    // ---- stage K tile [key][d] (swizzled) and V tile transposed [d][key] ----
+    {
+      const int row = tid >> 3;          // 0..31
+      const int off = (tid & 7) * 16;    // byte offset (8 bf16)
+#pragma unroll
+      for (int rr = 0; rr < TK; rr += 32) {
+        const int key = kv0 + row + rr;
+        uint4v kv = {};
+        if (key < vl)
+          kv = *reinterpret_cast<const uint4v*>(
+              K + ((long)b * L + key) * HD + (long)h * 64 + off / 2);
+        *reinterpret_cast<uint4v*>(k_lds + fa_swz(row + rr, off)) = kv;
+        // V: read 8 contiguous d, scatter-transpose into v_lds[d][key]
+        bf16 vv[8] = {};
+        if (key < vl)
+          *reinterpret_cast<uint4v*>(vv) = *reinterpret_cast<const uint4v*>(
+              V + ((long)b * L + key) * HD + (long)h * 64 + off / 2);
+        const int d0 = off / 2;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          *reinterpret_cast<bf16*>(
+              v_lds + fa_swz(d0 + j, (row + rr) * 2)) = vv[j];
+        }
+      }
+    }
+    __syncthreads();
+
+    // ---- S^T = K @ Q^T : D[key][qrow] ----
+    f32x4 s_acc[4][2] = {};  // [fkey][fq]
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      bf16x8 k_frag[4];
+#pragma unroll
+      for (int fk = 0; fk < 4; ++fk) {
+        const int key = fk * 16 + (lane & 15);
+        const int kbyte = ks * 64 + (lane >> 4) * 16;
+        k_frag[fk] = *reinterpret_cast<const bf16x8*>(k_lds + fa_swz(key, kbyte));
+      }
+#pragma unroll
+      for (int fk = 0; fk < 4; ++fk)
+#pragma unroll
+        for (int fq = 0; fq < 2; ++fq)
+          s_acc[fk][fq] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              k_frag[fk], q_frag[fq][ks], s_acc[fk][fq], 0, 0, 0);
+    }
+
+    // ---- masking + bias + tile row-max (over keys, per qrow column) ----
+    const int qcol[2] = {qw + (lane & 15), qw + 16 + (lane & 15)};
+    float p[4][2][4];  // [fkey][fq][i] raw scores -> probabilities
+    float tile_max[2] = {-3.4e38f, -3.4e38f};
+#pragma unroll
+    for (int fk = 0; fk < 4; ++fk)
+#pragma unroll
+      for (int fq = 0; fq < 2; ++fq)
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          const int key = kv0 + fk * 16 + (lane >> 4) * 4 + i;
+          float s = s_acc[fk][fq][i] * scale;
+          if (bias) s += bias[((long)h * L + qcol[fq]) * L + key];
+          const bool masked = key >= vl || (causal && key > qcol[fq]) || qcol[fq] >= L;
+          s = masked ? -3.4e38f : s;
+          p[fk][fq][i] = s;
+          tile_max[fq] = fmaxf(tile_max[fq], s);
+        }
+#pragma unroll
+    for (int fq = 0; fq < 2; ++fq) {
+      tile_max[fq] = fmaxf(tile_max[fq], __shfl_xor(tile_max[fq], 16));
+      tile_max[fq] = fmaxf(tile_max[fq], __shfl_xor(tile_max[fq], 32));
+    }
+
+    // ---- online softmax update ----
+    float alpha[2], sum_p[2] = {0.f, 0.f};
+#pragma unroll
+    for (int fq = 0; fq < 2; ++fq) {
+      const float m_new = fmaxf(m_st[fq], tile_max[fq]);
+      alpha[fq] = (m_st[fq] > -3.0e38f) ? __expf(m_st[fq] - m_new) : 0.f;
+      m_st[fq] = m_new;
+#pragma unroll
+      for (int fk = 0; fk < 4; ++fk)
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          const float e = (p[fk][fq][i] > -3.0e38f && m_new > -3.0e38f)
+                              ? __expf(p[fk][fq][i] - m_new)
+                              : 0.f;
+          p[fk][fq][i] = e;
+          sum_p[fq] += e;
+        }
+      sum_p[fq] += __shfl_xor(sum_p[fq], 16);
+      sum_p[fq] += __shfl_xor(sum_p[fq], 32);
+      l_st[fq] = l_st[fq] * alpha[fq] + sum_p[fq];
+    }
+
+    // ---- dropout on P (post-softmax-numerator; scaled at epilogue) ----
+    if (p8 > 0) {
+      const float dscale = 256.0f / (256.0f - p8);
+#pragma unroll
+      for (int fk = 0; fk < 4; ++fk)
+#pragma unroll
+        for (int fq = 0; fq < 2; ++fq)
+#pragma unroll
+          for (int i = 0; i < 4; ++i) {
+            const int key = kv0 + fk * 16 + (lane >> 4) * 4 + i;
+            const unsigned long long idx =
+                ((unsigned long long)(bh)*L + qcol[fq]) * L + key;
+            p[fk][fq][i] = fa_keep(idx, seed, p8) ? p[fk][fq][i] * dscale : 0.f;
+          }
+    }
+
+    // ---- rescale O by alpha (per O-row qrow = (l>>4)*4 + i) ----
+#pragma unroll
+    for (int fq = 0; fq < 2; ++fq)
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int qrow16 = (lane >> 4) * 4 + i;
+        const float a = __shfl(alpha[fq], qrow16);
+#pragma unroll
+        for (int fd = 0; fd < 4; ++fd) o_acc[fq][fd][i] *= a;
+      }
+
+    // ---- P^T (D-layout) -> P A-fragments via lane exchange, then PV ----
+    // A-frag for PV k-step kp (keys kp*32..kp*32+31): lane needs
+    // P[qrow = l&15 + fq*16][key = kp*32 + (l>>4)*8 + j], j = 0..7.
+    // source: value (key, qrow) lives at lane ((key%16)>>2)<<4 | (qrow%16),
+    // frag fkey = key/16, reg i = key%4.
+#pragma unroll
+    for (int kp = 0; kp < 2; ++kp) {
+      bf16x8 pa[2];  // [fq]
+      const int hi_half = (lane >> 5) & 1;  // lane groups 2,3 take fkey kp*2+1
+#pragma unroll
+      for (int fq = 0; fq < 2; ++fq) {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const int key = kp * 32 + (lane >> 4) * 8 + j;
+          const int src = (((key & 15) >> 2) << 4) | (lane & 15);
+          // register indices must be compile-time (rule 20); the source
+          // fragment differs only between lane halves -> shuffle both
+          // candidate registers and select
+          const float v0 = __shfl(p[kp * 2][fq][j & 3], src);
+          const float v1 = __shfl(p[kp * 2 + 1][fq][j & 3], src);
+          pa[fq][j] = (__bf16)(hi_half ? v1 : v0);
+        }
+      }
+      bf16x8 v_frag[4];
+#pragma unroll
+      for (int fd = 0; fd < 4; ++fd) {
+        const int d = fd * 16 + (lane & 15);
+        const int keybyte = kp * 64 + (lane >> 4) * 16;
+        v_frag[fd] = *reinterpret_cast<const bf16x8*>(v_lds + fa_swz(d, keybyte));
+      }
+#pragma unroll
+      for (int fq = 0; fq < 2; ++fq)
+#pragma unroll
+        for (int fd = 0; fd < 4; ++fd)
+          o_acc[fq][fd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              pa[fq], v_frag[fd], o_acc[fq][fd], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // ---- epilogue: O /= l, write (B, L, H*64); save lse = m + log(l) ----
+#pragma unroll
+  for (int fq = 0; fq < 2; ++fq) {
+    const float inv_l = (l_st[fq] > 0.f) ? 1.0f / l_st[fq] : 0.f;
+    if ((lane >> 4) == 0 && lse) {
+      const int qrow = qw + fq * 16 + (lane & 15);
+      if (qrow < L)
+        lse[((long)bh) * L + qrow] =
+            (l_st[fq] > 0.f) ? m_st[fq] + __logf(l_st[fq]) : 0.f;
+    }
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const int qrow16 = (lane >> 4) * 4 + i;
+      const int qrow = qw + fq * 16 + qrow16;
+      const float il = __shfl(inv_l, qrow16);
+      if (qrow >= L) continue;
+#pragma unroll
+      for (int fd = 0; fd < 4; ++fd) {
+        const int d = fd * 16 + (lane & 15);
+        O[((long)b * L + qrow) * HD + (long)h * 64 + d] =
+            __float2bfloat16(o_acc[fq][fd][i] * il);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Backward. Standard FA2 two-pass split so every output is owned by exactly
+// one block (no global atomics on dQ/dK/dV):
+//   * Dterm kernel: D[b,h,q] = sum_d dO * O  (needed by both passes; equals
+//     rowsum(Pd o dPd), which the dropout mask cancels out of)
+//   * dq kernel:  grid (q-tile, bh), loops KV tiles, accumulates dQ in regs
+//   * dkv kernel: grid (kv-tile, bh), loops Q strips, accumulates dK/dV in
+//     regs, cross-wave reduce through LDS
+// T5's additive position bias gets its gradient via fp32 atomicAdd from the
+// dq pass (B adds per element).
+// ---------------------------------------------------------------------------
+
+__global__ void flash_dterm_kernel(const bf16* __restrict__ dO,
+                                   const bf16* __restrict__ O,
+                                   float* __restrict__ Dterm, int B, int H,
+                                   int L) {
+  // one wave per (b, q, h) row of 64 elements
+  const long row = (long)(blockIdx.x * (blockDim.x / WAVE)) + threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const long R = (long)B * L * H;
+  if (row >= R) return;  // row = ((b*L + q)*H + h)
+  const long base = row * 64;
+  float acc = __bfloat162float(dO[base + lane]) * __bfloat162float(O[base + lane]);
+#pragma unroll
+  for (int off = WAVE / 2; off > 0; off >>= 1) acc += __shfl_down(acc, off);
+  if (lane == 0) {
+    const long bq = row / H;
+    const int h = (int)(row % H);
+    const long b = bq / L;
+    const long q = bq % L;
+    Dterm[((b * H + h) * L) + q] = acc;
+  }
+}
+
+// recompute P^T for one (wave q-strip, kv tile): returns scores in p[4][2][4]
+// D-layout (key = fk*16 + (lane>>4)*4 + i, qrow-col = fq*16 + (lane&15)).
+// Shared by the dq and dkv kernels.
+__device__ __forceinline__ void recompute_pT(
+    const char* k_lds, const bf16x8 q_frag[2][2], const float* __restrict__ bias,
+    const float lse_w[2], int lane, int h, int L, int qw, int kv0,
+    int vl, float scale, int causal, float p[4][2][4]) {
+  f32x4 s_acc[4][2] = {};
+#pragma unroll
+  for (int ks = 0; ks < 2; ++ks) {
+    bf16x8 k_frag[4];
+#pragma unroll
+    for (int fk = 0; fk < 4; ++fk) {
+      const int key = fk * 16 + (lane & 15);
+      const int kbyte = ks * 64 + (lane >> 4) * 16;
+      k_frag[fk] = *reinterpret_cast<const bf16x8*>(k_lds + fa_swz(key, kbyte));
+    }
+#pragma unroll
+    for (int fk = 0; fk < 4; ++fk)
+#pragma unroll
+      for (int fq = 0; fq < 2; ++fq)
+        s_acc[fk][fq] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            k_frag[fk], q_frag[fq][ks], s_acc[fk][fq], 0, 0, 0);
+  }
+#pragma unroll
+  for (int fk = 0; fk < 4; ++fk)
+#pragma unroll
+    for (int fq = 0; fq < 2; ++fq) {
+      const int qcol = qw + fq * 16 + (lane & 15);
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int key = kv0 + fk * 16 + (lane >> 4) * 4 + i;
+        float s = s_acc[fk][fq][i] * scale;
+        if (bias) s += bias[((long)h * L + qcol) * L + key];
+        const bool masked = key >= vl || (causal && key > qcol) || qcol >= L;
+        p[fk][fq][i] = masked ? 0.f : __expf(s - lse_w[fq]);
+      }
+    }
+}
+
+__global__ __launch_bounds__(256) void flash_dq_kernel(
+    const bf16* __restrict__ Q, const bf16* __restrict__ K,
+    const bf16* __restrict__ V, const bf16* __restrict__ dO,
+    const int* __restrict__ valid, const float* __restrict__ bias,
+    const float* __restrict__ lse, const float* __restrict__ Dterm,
+    bf16* __restrict__ dQ, float* __restrict__ dBias, int B, int H, int L,
+    float scale, int causal, unsigned p8, unsigned long long seed) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  char* k_lds = smem;                 // [key][d] swizzled, 8 KiB
+  char* kt_lds = smem + TK * 128;     // [d][key] swizzled, 8 KiB
+  char* v_lds = smem + 2 * TK * 128;  // [key][d] swizzled, 8 KiB
+
+  const int bh = blockIdx.y;
+  const int b = bh / H;
+  const int h = bh % H;
+  const int q0 = blockIdx.x * (NWAVE * TQW);
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x >> 6;
+  const int qw = q0 + wid * TQW;
+  const long HD = (long)H * 64;
+  const int vl = valid ? valid[b] : L;
+  const float dscale = (p8 > 0) ? 256.0f / (256.0f - p8) : 1.0f;
+
+  bf16x8 q_frag[2][2], do_frag[2][2];
+#pragma unroll
+  for (int fq = 0; fq < 2; ++fq)
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      const int qrow = qw + fq * 16 + (lane & 15);
+      const int d = ks * 32 + (lane >> 4) * 8;
+      if (qrow < L) {
+        q_frag[fq][ks] = *reinterpret_cast<const bf16x8*>(
+            Q + ((long)b * L + qrow) * HD + (long)h * 64 + d);
+        do_frag[fq][ks] = *reinterpret_cast<const bf16x8*>(
+            dO + ((long)b * L + qrow) * HD + (long)h * 64 + d);
+      } else {
+        q_frag[fq][ks] = bf16x8{};
+        do_frag[fq][ks] = bf16x8{};
+      }
+    }
+  float lse_w[2], dterm_w[2];
+#pragma unroll
+  for (int fq = 0; fq < 2; ++fq) {
+    const int qrow = qw + fq * 16 + (lane & 15);
+    lse_w[fq] = (qrow < L) ? lse[(long)bh * L + qrow] : 0.f;
+    dterm_w[fq] = (qrow < L) ? Dterm[(long)bh * L + qrow] : 0.f;
+  }
+
+  f32x4 dq_acc[2][4] = {};  // [fq][fd]: rows (l>>4)*4+i, col fd*16+(l&15)
+
+  const int kv_end = causal ? min(vl, q0 + NWAVE * TQW) : vl;
+  for (int kv0 = 0; kv0 < kv_end; kv0 += TK) {
+    {  // stage K (nat + transposed) and V (nat)
+      const int row = threadIdx.x >> 3;
+      const int off = (threadIdx.x & 7) * 16;
+#pragma unroll
+      for (int rr = 0; rr < TK; rr += 32) {
+        const int key = kv0 + row + rr;
+        uint4v kv = {}, vv4 = {};
+        if (key < vl) {
+          kv = *reinterpret_cast<const uint4v*>(
+              K + ((long)b * L + key) * HD + (long)h * 64 + off / 2);
+          vv4 = *reinterpret_cast<const uint4v*>(
+              V + ((long)b * L + key) * HD + (long)h * 64 + off / 2);
+        }
+        *reinterpret_cast<uint4v*>(k_lds + fa_swz(row + rr, off)) = kv;
+        *reinterpret_cast<uint4v*>(v_lds + fa_swz(row + rr, off)) = vv4;
+        bf16 kk[8];
+        *reinterpret_cast<uint4v*>(kk) = kv;
+        const int d0 = off / 2;
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          *reinterpret_cast<bf16*>(kt_lds + fa_swz(d0 + j, (row + rr) * 2)) = kk[j];
+      }
+    }
+    __syncthreads();
+
+    float p[4][2][4];
+    recompute_pT(k_lds, q_frag, bias, lse_w, lane, h, L, qw, kv0, vl, scale,
+                 causal, p);
+
+    // dPd^T = V dO^T (same structure as S^T), then ds
+    f32x4 dp_acc[4][2] = {};
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      bf16x8 v_frag[4];
+#pragma unroll
+      for (int fk = 0; fk < 4; ++fk) {
+        const int key = fk * 16 + (lane & 15);
+        const int kbyte = ks * 64 + (lane >> 4) * 16;
+        v_frag[fk] = *reinterpret_cast<const bf16x8*>(v_lds + fa_swz(key, kbyte));
+      }
+#pragma unroll
+      for (int fk = 0; fk < 4; ++fk)
+#pragma unroll
+        for (int fq = 0; fq < 2; ++fq)
+          dp_acc[fk][fq] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              v_frag[fk], do_frag[fq][ks], dp_acc[fk][fq], 0, 0, 0);
+    }
+
+    float ds[4][2][4];  // post-bias dS (pre-scale)
+#pragma unroll
+    for (int fk = 0; fk < 4; ++fk)
+#pragma unroll
+      for (int fq = 0; fq < 2; ++fq)
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          const int key = kv0 + fk * 16 + (lane >> 4) * 4 + i;
+          const int qcol = qw + fq * 16 + (lane & 15);
+          float dp = dp_acc[fk][fq][i];
+          if (p8 > 0) {
+            const unsigned long long idx =
+                ((unsigned long long)(bh)*L + qcol) * L + key;
+            dp = fa_keep(idx, seed, p8) ? dp * dscale : 0.f;
+          }
+          ds[fk][fq][i] = p[fk][fq][i] * (dp - dterm_w[fq]);
+          if (dBias && ds[fk][fq][i] != 0.f)
+            atomicAdd(dBias + ((long)h * L + qcol) * L + key, ds[fk][fq][i]);
+        }
+
+    // dQ += scale * ds @ K : A = ds (q, key) via lane exchange, B = K^T
+#pragma unroll
+    for (int kp = 0; kp < 2; ++kp) {
+      bf16x8 dsa[2];
+      const int hi_half = (lane >> 5) & 1;
+#pragma unroll
+      for (int fq = 0; fq < 2; ++fq)
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const int key = kp * 32 + (lane >> 4) * 8 + j;
+          const int src = (((key & 15) >> 2) << 4) | (lane & 15);
+          const float v0 = __shfl(ds[kp * 2][fq][j & 3], src);
+          const float v1 = __shfl(ds[kp * 2 + 1][fq][j & 3], src);
+          dsa[fq][j] = (__bf16)(scale * (hi_half ? v1 : v0));
+        }
+      bf16x8 kt_frag[4];
+#pragma unroll
+      for (int fd = 0; fd < 4; ++fd) {
+        const int d = fd * 16 + (lane & 15);
+        const int keybyte = kp * 64 + (lane >> 4) * 16;
+        kt_frag[fd] = *reinterpret_cast<const bf16x8*>(kt_lds + fa_swz(d, keybyte));
+      }
+#pragma unroll
+      for (int fq = 0; fq < 2; ++fq)
+#pragma unroll
+        for (int fd = 0; fd < 4; ++fd)
+          dq_acc[fq][fd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              dsa[fq], kt_frag[fd], dq_acc[fq][fd], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // write dQ (B, L, H*64)
+#pragma unroll
+  for (int fq = 0; fq < 2; ++fq)
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const int qrow = qw + fq * 16 + (lane >> 4) * 4 + i;
+      if (qrow >= L) continue;
+#pragma unroll
+      for (int fd = 0; fd < 4; ++fd) {
+        const int d = fd * 16 + (lane & 15);
+        dQ[((long)b * L + qrow) * HD + (long)h * 64 + d] =
+            __float2bfloat16(dq_acc[fq][fd][i]);
+      }
+    }
+}
+
+__global__ __launch_bounds__(256) void flash_dkv_kernel(
+    const bf16* __restrict__ Q, const bf16* __restrict__ K,
+    const bf16* __restrict__ V, const bf16* __restrict__ dO,
+    const int* __restrict__ valid, const float* __restrict__ bias,
+    const float* __restrict__ lse, const float* __restrict__ Dterm,
+    bf16* __restrict__ dK, bf16* __restrict__ dV, int B, int H, int L,
+    float scale, int causal, unsigned p8, unsigned long long seed) {
+  // LDS map (per block): K nat 8K | V nat 8K | per-wave slices: pd 4K, ds 4K,
+  // doT 4K, qT 4K (16K each across 4 waves). The pd/ds/doT/qT region is
+  // reused as the fp32 cross-wave reduction buffer after the q loop.
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  char* k_lds = smem;
+  char* v_lds = smem + TK * 128;
+  char* wave_base = smem + 2 * TK * 128;
+
+  const int bh = blockIdx.y;
+  const int b = bh / H;
+  const int h = bh % H;
+  const int kv0 = blockIdx.x * TK;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x >> 6;
+  const long HD = (long)H * 64;
+  const int vl = valid ? valid[b] : L;
+  const float dscale = (p8 > 0) ? 256.0f / (256.0f - p8) : 1.0f;
+  char* pd_lds = wave_base + wid * 4096;
+  char* ds_lds = wave_base + 16384 + wid * 4096;
+  char* dot_lds = wave_base + 32768 + wid * 4096;  // [d 64][q 32] bf16
+  char* qt_lds = wave_base + 49152 + wid * 4096;
+
+  {  // stage K and V tiles (natural layout, swizzled)
+    const int row = threadIdx.x >> 3;
+    const int off = (threadIdx.x & 7) * 16;
+#pragma unroll
+    for (int rr = 0; rr < TK; rr += 32) {
+      const int key = kv0 + row + rr;
+      uint4v kv = {}, vv = {};
+      if (key < vl && kv0 < vl) {
+        kv = *reinterpret_cast<const uint4v*>(
+            K + ((long)b * L + key) * HD + (long)h * 64 + off / 2);
+        vv = *reinterpret_cast<const uint4v*>(
+            V + ((long)b * L + key) * HD + (long)h * 64 + off / 2);
+      }
+      *reinterpret_cast<uint4v*>(k_lds + fa_swz(row + rr, off)) = kv;
+      *reinterpret_cast<uint4v*>(v_lds + fa_swz(row + rr, off)) = vv;
+    }
+  }
+  __syncthreads();
+
+  f32x4 dv_acc[4][4] = {};  // [fkey][fd]
+  f32x4 dk_acc[4][4] = {};
+
+  if (kv0 < vl) {
+    const int q_begin = causal ? (kv0 / (NWAVE * TQW)) * (NWAVE * TQW) : 0;
+    for (int q0 = q_begin; q0 < L; q0 += NWAVE * TQW) {
+      const int qw = q0 + wid * TQW;
+      bf16x8 q_frag[2][2], do_frag[2][2];
+#pragma unroll
+      for (int fq = 0; fq < 2; ++fq)
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks) {
+          const int qrow = qw + fq * 16 + (lane & 15);
+          const int d = ks * 32 + (lane >> 4) * 8;
+          if (qrow < L) {
+            q_frag[fq][ks] = *reinterpret_cast<const bf16x8*>(
+                Q + ((long)b * L + qrow) * HD + (long)h * 64 + d);
+            do_frag[fq][ks] = *reinterpret_cast<const bf16x8*>(
+                dO + ((long)b * L + qrow) * HD + (long)h * 64 + d);
+          } else {
+            q_frag[fq][ks] = bf16x8{};
+            do_frag[fq][ks] = bf16x8{};
+          }
+        }
+      // stage this wave's dO^T and Q^T slices ([d][q] 64-B rows) from the
+      // already-loaded fragments: element d = ks*32 + (lane>>4)*8 + j at
+      // q_loc = fq*16 + (lane&15)
+#pragma unroll
+      for (int fq = 0; fq < 2; ++fq)
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks)
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            const int d = ks * 32 + (lane >> 4) * 8 + j;
+            const int q_loc = fq * 16 + (lane & 15);
+            *reinterpret_cast<bf16*>(dot_lds + d * 64 + q_loc * 2) =
+                (bf16)do_frag[fq][ks][j];
+            *reinterpret_cast<bf16*>(qt_lds + d * 64 + q_loc * 2) =
+                (bf16)q_frag[fq][ks][j];
+          }
+
+      float lse_w[2], dterm_w[2];
+#pragma unroll
+      for (int fq = 0; fq < 2; ++fq) {
+        const int qrow = qw + fq * 16 + (lane & 15);
+        lse_w[fq] = (qrow < L) ? lse[(long)bh * L + qrow] : 0.f;
+        dterm_w[fq] = (qrow < L) ? Dterm[(long)bh * L + qrow] : 0.f;
+      }
+
+      float p[4][2][4];
+      recompute_pT(k_lds, q_frag, bias, lse_w, lane, h, L, qw, kv0, vl, scale,
+                   causal, p);
+
+      f32x4 dp_acc[4][2] = {};
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        bf16x8 v_frag[4];
+#pragma unroll
+        for (int fk = 0; fk < 4; ++fk) {
+          const int key = fk * 16 + (lane & 15);
+          const int kbyte = ks * 64 + (lane >> 4) * 16;
+          v_frag[fk] = *reinterpret_cast<const bf16x8*>(v_lds + fa_swz(key, kbyte));
+        }
+#pragma unroll
+        for (int fk = 0; fk < 4; ++fk)
+#pragma unroll
+          for (int fq = 0; fq < 2; ++fq)
+            dp_acc[fk][fq] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                v_frag[fk], do_frag[fq][ks], dp_acc[fk][fq], 0, 0, 0);
+      }
+
+      // pd (dropout-masked P) and ds = scale * P (dP - D); bounce both
+      // through [key 64-B-row][q] LDS slices for the transposed A-frags
+#pragma unroll
+      for (int fk = 0; fk < 4; ++fk)
+#pragma unroll
+        for (int fq = 0; fq < 2; ++fq)
+#pragma unroll
+          for (int i = 0; i < 4; ++i) {
+            const int key_loc = fk * 16 + (lane >> 4) * 4 + i;
+            const int key = kv0 + key_loc;
+            const int qcol = qw + fq * 16 + (lane & 15);
+            const int q_loc = fq * 16 + (lane & 15);
+            float pd = p[fk][fq][i];
+            float dp = dp_acc[fk][fq][i];
+            if (p8 > 0) {
+              const unsigned long long idx =
+                  ((unsigned long long)(bh)*L + qcol) * L + key;
+              const bool keep = fa_keep(idx, seed, p8);
+              pd = keep ? pd * dscale : 0.f;
+              dp = keep ? dp * dscale : 0.f;
+            }
+            const float dsv = scale * p[fk][fq][i] * (dp - dterm_w[fq]);
+            *reinterpret_cast<bf16*>(pd_lds + key_loc * 64 + q_loc * 2) =
+                __float2bfloat16(pd);
+            *reinterpret_cast<bf16*>(ds_lds + key_loc * 64 + q_loc * 2) =
+                __float2bfloat16(dsv);
+          }
+
+      // dV += pd(key, q) @ dO(q, d);  dK += ds(key, q) @ Q(q, d)
+      bf16x8 pa[4], dsa[4];
+#pragma unroll
+      for (int fk = 0; fk < 4; ++fk) {
+        const int key_loc = fk * 16 + (lane & 15);
+        const int qbyte = (lane >> 4) * 16;
+        pa[fk] = *reinterpret_cast<const bf16x8*>(pd_lds + key_loc * 64 + qbyte);
+        dsa[fk] = *reinterpret_cast<const bf16x8*>(ds_lds + key_loc * 64 + qbyte);
+      }
+      bf16x8 dob[4], qb[4];
+#pragma unroll
+      for (int fd = 0; fd < 4; ++fd) {
+        const int d = fd * 16 + (lane & 15);
+        const int qbyte = (lane >> 4) * 16;
+        dob[fd] = *reinterpret_cast<const bf16x8*>(dot_lds + d * 64 + qbyte);
+        qb[fd] = *reinterpret_cast<const bf16x8*>(qt_lds + d * 64 + qbyte);
+      }
+#pragma unroll
+      for (int fk = 0; fk < 4; ++fk)
+#pragma unroll
+        for (int fd = 0; fd < 4; ++fd) {
+          dv_acc[fk][fd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              pa[fk], dob[fd], dv_acc[fk][fd], 0, 0, 0);
+          dk_acc[fk][fd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              dsa[fk], qb[fd], dk_acc[fk][fd], 0, 0, 0);
+        }
+    }
+  }
+
+  // cross-wave reduce through LDS, then one plain store per element
+  float* red = reinterpret_cast<float*>(wave_base);  // 64x64 fp32 = 16 KiB
+#define FA_REDUCE_STORE(ACC, OUT)                                             \
+  __syncthreads();                                                            \
+  for (int i = threadIdx.x; i < TK * 64; i += blockDim.x) red[i] = 0.f;       \
+  __syncthreads();                                                            \
+  _Pragma("unroll") for (int fk = 0; fk < 4; ++fk)                            \
+      _Pragma("unroll") for (int fd = 0; fd < 4; ++fd)                        \
+      _Pragma("unroll") for (int i = 0; i < 4; ++i) {                         \
+    const int key_loc = fk * 16 + (lane >> 4) * 4 + i;                        \
+    const int d = fd * 16 + (lane & 15);                                      \
+    atomicAdd(red + key_loc * 64 + d, ACC[fk][fd][i]);                        \
+  }                                                                           \
+  __syncthreads();                                                            \
+  for (int i = threadIdx.x; i < TK * 64; i += blockDim.x) {                   \
+    const int key = kv0 + i / 64;                                             \
+    const int d = i % 64;                                                     \
+    if (key < L)                                                              \
+      OUT[((long)b * L + key) * HD + (long)h * 64 + d] =                      \
+          __float2bfloat16(red[i]);                                           \
+  }
+  FA_REDUCE_STORE(dv_acc, dV)
+  FA_REDUCE_STORE(dk_acc, dK)
+#undef FA_REDUCE_STORE
+}
+
+void launch_flash_fwd(const bf16* Q, const bf16* K, const bf16* V,
+                      const int* valid, const float* bias, bf16* O, float* lse,
+                      int B, int H, int L, float scale, int causal,
+                      unsigned p8, unsigned long long seed,
+                      hipStream_t stream) {
+  const dim3 grid((L + NWAVE * TQW - 1) / (NWAVE * TQW), B * H);
+  const size_t lds = 2 * TK * 128;
+  hipLaunchKernelGGL(flash_fwd_kernel, grid, dim3(256), lds, stream, Q, K, V,
+                     valid, bias, O, lse, B, H, L, scale, causal, p8, seed);
+}
+
+void launch_flash_dterm(const bf16* dO, const bf16* O, float* Dterm, int B,
+                        int H, int L, hipStream_t stream) {
+  const long rows = (long)B * L * H;
+  const int rows_per_block = 4;
+  hipLaunchKernelGGL(flash_dterm_kernel,
+                     dim3((rows + rows_per_block - 1) / rows_per_block),
+                     dim3(WAVE * rows_per_block), 0, stream, dO, O, Dterm, B, H,
+                     L);
+}
+
+void launch_flash_dq(const bf16* Q, const bf16* K, const bf16* V,
+                     const bf16* dO, const int* valid, const float* bias,
+                     const float* lse, const float* Dterm, bf16* dQ,
+                     float* dBias, int B, int H, int L, float scale,
+                     int causal, unsigned p8, unsigned long long seed,
+                     hipStream_t stream) {
+  const dim3 grid((L + NWAVE * TQW - 1) / (NWAVE * TQW), B * H);
+  const size_t lds = 3 * TK * 128;
+  hipLaunchKernelGGL(flash_dq_kernel, grid, dim3(256), lds, stream, Q, K, V,
+                     dO, valid, bias, lse, Dterm, dQ, dBias, B, H, L, scale,
+                     causal, p8, seed);
+}
+
+void launch_flash_dkv(const bf16* Q, const bf16* K, const bf16* V,
+                      const bf16* dO, const int* valid, const float* bias,
+                      const float* lse, const float* Dterm, bf16* dK,
+                      bf16* dV, int B, int H, int L, float scale, int causal,
+                      unsigned p8, unsigned long long seed,
+                      hipStream_t stream) {
+  const dim3 grid((L + TK - 1) / TK, B * H);
+  const size_t lds = 2 * TK * 128 + 4 * 16384;  // K,V + 4 per-wave slice sets
+  hipLaunchKernelGGL(flash_dkv_kernel, grid, dim3(256), lds, stream, Q, K, V,
+                     dO, valid, bias, lse, Dterm, dK, dV, B, H, L, scale,
+                     causal, p8, seed);
+}

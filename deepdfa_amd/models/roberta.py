@@ -29,7 +29,13 @@ from typing import Optional
 import torch
 from torch import nn
 
-from ..ops.transformer import bias_gelu, layer_norm, masked_softmax_dropout
+from ..ops.transformer import (
+    bias_gelu,
+    flash_attention,
+    flash_usable,
+    layer_norm,
+    masked_softmax_dropout,
+)
 
 
 @dataclass
@@ -98,13 +104,19 @@ class RobertaSelfAttention(nn.Module):
     def forward(self, x, valid: Optional[torch.Tensor], output_attentions: bool = False):
         B, L, D = x.shape
         H, d = self.num_heads, self.head_dim
+        p = self.dropout_p if self.training else 0.0
+        q, k, v = self.query(x), self.key(x), self.value(x)
+        if d == 64 and not output_attentions and flash_usable(q, L):
+            out = flash_attention(q, k, v, H, valid=valid, scale=1.0 / math.sqrt(d),
+                                  dropout_p=p)
+            return out, None
+        # materialized path: CPU / fp32 / attention-probs output
 
         def split(t):
             return t.view(B, L, H, d).transpose(1, 2)
 
-        q, k, v = split(self.query(x)), split(self.key(x)), split(self.value(x))
+        q, k, v = split(q), split(k), split(v)
         scores = torch.matmul(q, k.transpose(-1, -2))
-        p = self.dropout_p if self.training else 0.0
         probs, probs_dropped = masked_softmax_dropout(scores, valid, 1.0 / math.sqrt(d), p)
         ctx = torch.matmul(probs_dropped, v)
         out = ctx.transpose(1, 2).reshape(B, L, D)

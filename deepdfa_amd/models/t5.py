@@ -21,7 +21,7 @@ from typing import Optional
 import torch
 from torch import nn
 
-from ..ops.transformer import masked_softmax_dropout, rms_norm
+from ..ops.transformer import flash_attention, flash_usable, masked_softmax_dropout, rms_norm
 
 
 @dataclass
@@ -116,15 +116,24 @@ class T5Attention(nn.Module):
         def split(t, L):
             return t.view(B, L, H, d).transpose(1, 2)
 
-        q = split(self.q(x), Lq)
-        k = split(self.k(src), Lk)
-        v = split(self.v(src), Lk)
+        qp = self.q(x)
+        kp = self.k(src)
+        vp = self.v(src)
+        causal = self.causal and kv is None
+        if d == 64 and flash_usable(qp, Lq, Lk):
+            bias = None
+            if position_bias is not None:
+                bias = position_bias.squeeze(0).float().contiguous()
+            out = flash_attention(qp, kp, vp, H, valid=valid, bias=bias, scale=1.0,
+                                  causal=causal, dropout_p=dropout_p)
+            return self.o(out)
+        q = split(qp, Lq)
+        k = split(kp, Lk)
+        v = split(vp, Lk)
         scores = torch.matmul(q, k.transpose(-1, -2))  # T5: no 1/sqrt(d)
         if position_bias is not None:
             scores = scores + position_bias.to(scores.dtype)
-        _, probs_dropped = masked_softmax_dropout(
-            scores, valid, 1.0, dropout_p, causal=(self.causal and kv is None)
-        )
+        _, probs_dropped = masked_softmax_dropout(scores, valid, 1.0, dropout_p, causal=causal)
         ctx = torch.matmul(probs_dropped, v)
         out = ctx.transpose(1, 2).reshape(B, Lq, H * d)
         return self.o(out)

@@ -127,6 +127,51 @@ def masked_softmax_dropout(
     return p, pd
 
 
+class _FlashAttention(torch.autograd.Function):
+    """Fused attention, head_dim 64, (B, L, H*64) layout (csrc/flash_attn.hip):
+    masked softmax + dropout + PV in one kernel, FA2-style two-pass backward.
+    `bias` is the T5 additive position bias (fp32, (H, L, L))."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, H, valid, bias, scale, causal, dropout_p):
+        ext = load_ext(required=True)
+        seed = _next_seed() if dropout_p > 0 else 0
+        O, lse = ext.flash_attn_fwd(
+            q.contiguous(), k.contiguous(), v.contiguous(), H, valid, bias,
+            scale, causal, dropout_p, seed,
+        )
+        ctx.save_for_backward(q, k, v, O, lse)
+        ctx.meta = (H, valid, bias, scale, causal, dropout_p, seed)
+        ctx.need_dbias = bias is not None and ctx.needs_input_grad[5]
+        return O
+
+    @staticmethod
+    def backward(ctx, dO):
+        ext = load_ext(required=True)
+        q, k, v, O, lse = ctx.saved_tensors
+        H, valid, bias, scale, causal, dropout_p, seed = ctx.meta
+        outs = ext.flash_attn_bwd(
+            dO.contiguous(), q.contiguous(), k.contiguous(), v.contiguous(), O, lse,
+            H, valid, bias, scale, causal, dropout_p, seed, ctx.need_dbias,
+        )
+        dbias = outs[3] if ctx.need_dbias else None
+        return outs[0], outs[1], outs[2], None, None, dbias, None, None, None
+
+
+def flash_attention(q, k, v, num_heads, valid=None, bias=None, scale=1.0,
+                    causal=False, dropout_p=0.0):
+    return _FlashAttention.apply(q, k, v, num_heads, valid, bias, scale, causal, dropout_p)
+
+
+def flash_usable(x, L, Lk=None) -> bool:
+    return (
+        x.is_cuda
+        and x.dtype == torch.bfloat16
+        and L % 64 == 0
+        and (Lk is None or Lk == L)
+    )
+
+
 class _RMSNorm(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, eps):

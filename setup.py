@@ -24,6 +24,7 @@ ext = CUDAExtension(
         "csrc/gemm_bias.hip",
         "csrc/wgrad.hip",
         "csrc/transformer_kernels.hip",
+        "csrc/flash_attn.hip",
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
