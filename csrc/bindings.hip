@@ -68,6 +68,8 @@ void launch_flash_dq(const __hip_bfloat16*, const __hip_bfloat16*, const __hip_b
                      const __hip_bfloat16*, const int*, const float*, const float*, const float*,
                      __hip_bfloat16*, float*, int, int, int, float, int, unsigned,
                      unsigned long long, hipStream_t);
+void launch_gemm2(const __hip_bfloat16*, const __hip_bfloat16*, const float*,
+                  const __hip_bfloat16*, __hip_bfloat16*, int, int, int, hipStream_t);
 void launch_flash_dkv(const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*,
                       const __hip_bfloat16*, const int*, const float*, const float*, const float*,
                       __hip_bfloat16*, __hip_bfloat16*, int, int, int, float, int, unsigned,
@@ -670,6 +672,30 @@ std::vector<at::Tensor> flash_attn_bwd(at::Tensor dO, at::Tensor Q, at::Tensor K
   return {dQ, dK, dV};
 }
 
+
+// out(N, COL) = A(N, K) @ W(COL, K)^T [+ bias] [+ addend] — transformer
+// shapes (csrc/gemm_bf16.hip). Geometry: N%128, K%64, COL%128 == 0.
+at::Tensor gemm2(at::Tensor A, at::Tensor W, c10::optional<at::Tensor> bias,
+                 c10::optional<at::Tensor> addend) {
+  CHECK_GPU(A);
+  CHECK_GPU(W);
+  TORCH_CHECK(A.scalar_type() == at::kBFloat16 && W.scalar_type() == at::kBFloat16);
+  const int N = A.size(0), K = A.size(1), COL = W.size(0);
+  TORCH_CHECK(W.size(1) == K && N % 128 == 0 && K % 64 == 0 && COL % 128 == 0,
+              "gemm2 geometry violated: ", N, "x", K, "x", COL);
+  auto out = at::empty({N, COL}, A.options());
+  const float* b = nullptr;
+  if (bias.has_value()) {
+    TORCH_CHECK(bias->scalar_type() == at::kFloat && bias->is_contiguous());
+    b = bias->data_ptr<float>();
+  }
+  const bf16_t* add = nullptr;
+  if (addend.has_value()) add = reinterpret_cast<const bf16_t*>(addend->data_ptr());
+  launch_gemm2(ptr<bf16_t>(A), ptr<bf16_t>(W), b, add, mptr<bf16_t>(out), N, K, COL,
+               cur_stream());
+  return out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "deepdfa_amd MI355X (gfx950) kernels";
   m.def("embed4_fwd", &embed4_fwd);
@@ -696,6 +722,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_wgrad", &rmsnorm_wgrad);
   m.def("flash_attn_fwd", &flash_attn_fwd);
   m.def("flash_attn_bwd", &flash_attn_bwd);
+  m.def("gemm2", &gemm2);
   m.def("softmax_mask_bwd", &softmax_mask_bwd);
   m.def("ggnn_fused_bwd", &ggnn_fused_bwd);
 }

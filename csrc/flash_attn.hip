@@ -527,6 +527,44 @@ __global__ __launch_bounds__(256) void flash_dq_kernel(
     }
 }
 
+// recompute P^T for a 32-key half-tile: p[2][2][4], keys key_off + fk*16 + ...
+__device__ __forceinline__ void recompute_pT32(
+    const char* k_lds, const bf16x8 q_frag[2][2], const float* __restrict__ bias,
+    const float lse_w[2], int lane, int h, int L, int qw, int kv0, int key_off,
+    int vl, float scale, int causal, float p[2][2][4]) {
+  f32x4 s_acc[2][2] = {};
+#pragma unroll
+  for (int ks = 0; ks < 2; ++ks) {
+    bf16x8 k_frag[2];
+#pragma unroll
+    for (int fk = 0; fk < 2; ++fk) {
+      const int key = key_off + fk * 16 + (lane & 15);
+      const int kbyte = ks * 64 + (lane >> 4) * 16;
+      k_frag[fk] = *reinterpret_cast<const bf16x8*>(k_lds + fa_swz(key, kbyte));
+    }
+#pragma unroll
+    for (int fk = 0; fk < 2; ++fk)
+#pragma unroll
+      for (int fq = 0; fq < 2; ++fq)
+        s_acc[fk][fq] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            k_frag[fk], q_frag[fq][ks], s_acc[fk][fq], 0, 0, 0);
+  }
+#pragma unroll
+  for (int fk = 0; fk < 2; ++fk)
+#pragma unroll
+    for (int fq = 0; fq < 2; ++fq) {
+      const int qcol = qw + fq * 16 + (lane & 15);
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int key = kv0 + key_off + fk * 16 + (lane >> 4) * 4 + i;
+        float sc = s_acc[fk][fq][i] * scale;
+        if (bias) sc += bias[((long)h * L + qcol) * L + key];
+        const bool masked = key >= vl || (causal && key > qcol) || qcol >= L;
+        p[fk][fq][i] = masked ? 0.f : __expf(sc - lse_w[fq]);
+      }
+    }
+}
+
 __global__ __launch_bounds__(256) void flash_dkv_kernel(
     const bf16* __restrict__ Q, const bf16* __restrict__ K,
     const bf16* __restrict__ V, const bf16* __restrict__ dO,
@@ -534,9 +572,11 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
     const float* __restrict__ lse, const float* __restrict__ Dterm,
     bf16* __restrict__ dK, bf16* __restrict__ dV, int B, int H, int L,
     float scale, int causal, unsigned p8, unsigned long long seed) {
-  // LDS map (per block): K nat 8K | V nat 8K | per-wave slices: pd 4K, ds 4K,
-  // doT 4K, qT 4K (16K each across 4 waves). The pd/ds/doT/qT region is
-  // reused as the fp32 cross-wave reduction buffer after the q loop.
+  // Wave grid 2 (key halves) x 2 (q interleave): halves the per-wave
+  // accumulator footprint (the 64-key variant needed 128 fp32 accumulators
+  // on top of ~176 VGPRs -> 1 wave/SIMD on the unified register file).
+  // LDS map: K nat 8K | V nat 8K | per-wave pd 2K, ds 2K, doT 4K, qT 4K
+  // (reused as the fp32 reduction buffer afterwards).
   extern __shared__ __attribute__((aligned(16))) char smem[];
   char* k_lds = smem;
   char* v_lds = smem + TK * 128;
@@ -548,13 +588,16 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
   const int kv0 = blockIdx.x * TK;
   const int lane = threadIdx.x & (WAVE - 1);
   const int wid = threadIdx.x >> 6;
+  const int wk = wid >> 1;   // key half: keys wk*32 .. wk*32+31
+  const int wq = wid & 1;    // q interleave
+  const int key_off = wk * 32;
   const long HD = (long)H * 64;
   const int vl = valid ? valid[b] : L;
   const float dscale = (p8 > 0) ? 256.0f / (256.0f - p8) : 1.0f;
-  char* pd_lds = wave_base + wid * 4096;
-  char* ds_lds = wave_base + 16384 + wid * 4096;
-  char* dot_lds = wave_base + 32768 + wid * 4096;  // [d 64][q 32] bf16
-  char* qt_lds = wave_base + 49152 + wid * 4096;
+  char* pd_lds = wave_base + wid * 2048;
+  char* ds_lds = wave_base + 8192 + wid * 2048;
+  char* dot_lds = wave_base + 16384 + wid * 4096;  // [d 64][q 32] bf16
+  char* qt_lds = wave_base + 32768 + wid * 4096;
 
   {  // stage K and V tiles (natural layout, swizzled)
     const int row = threadIdx.x >> 3;
@@ -575,13 +618,14 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
   }
   __syncthreads();
 
-  f32x4 dv_acc[4][4] = {};  // [fkey][fd]
-  f32x4 dk_acc[4][4] = {};
+  f32x4 dv_acc[2][4] = {};  // [fkey][fd] over this wave's 32-key half
+  f32x4 dk_acc[2][4] = {};
 
   if (kv0 < vl) {
-    const int q_begin = causal ? (kv0 / (NWAVE * TQW)) * (NWAVE * TQW) : 0;
-    for (int q0 = q_begin; q0 < L; q0 += NWAVE * TQW) {
-      const int qw = q0 + wid * TQW;
+    const int q_begin = causal ? kv0 : 0;  // kv0 is a multiple of 32
+    const int nstrips = (L - q_begin + TQW - 1) / TQW;
+    for (int strip = wq; strip < nstrips; strip += 2) {
+      const int qw = q_begin + strip * TQW;
       bf16x8 q_frag[2][2], do_frag[2][2];
 #pragma unroll
       for (int fq = 0; fq < 2; ++fq)
@@ -599,9 +643,7 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
             do_frag[fq][ks] = bf16x8{};
           }
         }
-      // stage this wave's dO^T and Q^T slices ([d][q] 64-B rows) from the
-      // already-loaded fragments: element d = ks*32 + (lane>>4)*8 + j at
-      // q_loc = fq*16 + (lane&15)
+      // stage this wave's dO^T and Q^T slices ([d][q] 64-B rows)
 #pragma unroll
       for (int fq = 0; fq < 2; ++fq)
 #pragma unroll
@@ -624,38 +666,37 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
         dterm_w[fq] = (qrow < L) ? Dterm[(long)bh * L + qrow] : 0.f;
       }
 
-      float p[4][2][4];
-      recompute_pT(k_lds, q_frag, bias, lse_w, lane, h, L, qw, kv0, vl, scale,
-                   causal, p);
+      float p[2][2][4];
+      recompute_pT32(k_lds, q_frag, bias, lse_w, lane, h, L, qw, kv0, key_off,
+                     vl, scale, causal, p);
 
-      f32x4 dp_acc[4][2] = {};
+      f32x4 dp_acc[2][2] = {};
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks) {
-        bf16x8 v_frag[4];
+        bf16x8 v_frag[2];
 #pragma unroll
-        for (int fk = 0; fk < 4; ++fk) {
-          const int key = fk * 16 + (lane & 15);
+        for (int fk = 0; fk < 2; ++fk) {
+          const int key = key_off + fk * 16 + (lane & 15);
           const int kbyte = ks * 64 + (lane >> 4) * 16;
           v_frag[fk] = *reinterpret_cast<const bf16x8*>(v_lds + fa_swz(key, kbyte));
         }
 #pragma unroll
-        for (int fk = 0; fk < 4; ++fk)
+        for (int fk = 0; fk < 2; ++fk)
 #pragma unroll
           for (int fq = 0; fq < 2; ++fq)
             dp_acc[fk][fq] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                 v_frag[fk], do_frag[fq][ks], dp_acc[fk][fq], 0, 0, 0);
       }
 
-      // pd (dropout-masked P) and ds = scale * P (dP - D); bounce both
-      // through [key 64-B-row][q] LDS slices for the transposed A-frags
+      // pd (dropout-masked P) and ds = scale * P (dP - D) -> LDS bounce
 #pragma unroll
-      for (int fk = 0; fk < 4; ++fk)
+      for (int fk = 0; fk < 2; ++fk)
 #pragma unroll
         for (int fq = 0; fq < 2; ++fq)
 #pragma unroll
           for (int i = 0; i < 4; ++i) {
             const int key_loc = fk * 16 + (lane >> 4) * 4 + i;
-            const int key = kv0 + key_loc;
+            const int key = kv0 + key_off + key_loc;
             const int qcol = qw + fq * 16 + (lane & 15);
             const int q_loc = fq * 16 + (lane & 15);
             float pd = p[fk][fq][i];
@@ -675,9 +716,9 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
           }
 
       // dV += pd(key, q) @ dO(q, d);  dK += ds(key, q) @ Q(q, d)
-      bf16x8 pa[4], dsa[4];
+      bf16x8 pa[2], dsa[2];
 #pragma unroll
-      for (int fk = 0; fk < 4; ++fk) {
+      for (int fk = 0; fk < 2; ++fk) {
         const int key_loc = fk * 16 + (lane & 15);
         const int qbyte = (lane >> 4) * 16;
         pa[fk] = *reinterpret_cast<const bf16x8*>(pd_lds + key_loc * 64 + qbyte);
@@ -692,7 +733,7 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
         qb[fd] = *reinterpret_cast<const bf16x8*>(qt_lds + d * 64 + qbyte);
       }
 #pragma unroll
-      for (int fk = 0; fk < 4; ++fk)
+      for (int fk = 0; fk < 2; ++fk)
 #pragma unroll
         for (int fd = 0; fd < 4; ++fd) {
           dv_acc[fk][fd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
@@ -703,16 +744,16 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
     }
   }
 
-  // cross-wave reduce through LDS, then one plain store per element
+  // cross-wave reduce through LDS (2 waves per key half), then plain store
   float* red = reinterpret_cast<float*>(wave_base);  // 64x64 fp32 = 16 KiB
 #define FA_REDUCE_STORE(ACC, OUT)                                             \
   __syncthreads();                                                            \
   for (int i = threadIdx.x; i < TK * 64; i += blockDim.x) red[i] = 0.f;       \
   __syncthreads();                                                            \
-  _Pragma("unroll") for (int fk = 0; fk < 4; ++fk)                            \
+  _Pragma("unroll") for (int fk = 0; fk < 2; ++fk)                            \
       _Pragma("unroll") for (int fd = 0; fd < 4; ++fd)                        \
       _Pragma("unroll") for (int i = 0; i < 4; ++i) {                         \
-    const int key_loc = fk * 16 + (lane >> 4) * 4 + i;                        \
+    const int key_loc = key_off + fk * 16 + (lane >> 4) * 4 + i;              \
     const int d = fd * 16 + (lane & 15);                                      \
     atomicAdd(red + key_loc * 64 + d, ACC[fk][fd][i]);                        \
   }                                                                           \
@@ -770,7 +811,7 @@ void launch_flash_dkv(const bf16* Q, const bf16* K, const bf16* V,
                       unsigned p8, unsigned long long seed,
                       hipStream_t stream) {
   const dim3 grid((L + TK - 1) / TK, B * H);
-  const size_t lds = 2 * TK * 128 + 4 * 16384;  // K,V + 4 per-wave slice sets
+  const size_t lds = 2 * TK * 128 + 16384 + 32768;  // K,V + pd/ds + doT/qT
   hipLaunchKernelGGL(flash_dkv_kernel, grid, dim3(256), lds, stream, Q, K, V,
                      dO, valid, bias, lse, Dterm, dK, dV, B, H, L, scale,
                      causal, p8, seed);

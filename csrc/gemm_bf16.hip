@@ -1,0 +1,134 @@
+// Transformer-shape bf16 GEMM: out(N, COL) = A(N, K) @ W(COL, K)^T [+ bias]
+// [+ addend], MI355X (gfx950).
+//
+// hipBLASLt runs the encoder projections at 173-420 TF on these shapes
+// (rocprof, profiles/): this kernel applies the guide's minimum 2-phase
+// glds pipeline (cdna_hip_programming.md §5.5 T3 recipe): 128x128 tile,
+// BK=64, double-buffered LDS filled by global_load_lds (16-B pieces),
+// XOR-swizzle applied on the SOURCE address + the read side (rule 21),
+// one vmcnt(0)+barrier per K-tile, s_setprio around the MFMA cluster.
+//
+// Geometry contract (wrapper-enforced, else fall back to rocBLAS):
+// N % 128 == 0, K % 64 == 0, COL % 128 == 0.
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+#define WAVE 64
+#define BM2 128
+#define BN2 128
+#define BK2 64
+#define ROWB2 128  // bytes per LDS row (64 bf16)
+
+using bf16 = __hip_bfloat16;
+using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+
+__device__ __forceinline__ int g2_swz(int row, int byte) {
+  return row * ROWB2 + (byte ^ ((row & 7) << 4));
+}
+
+// stage one 128x64 bf16 tile into LDS via glds: 16 KiB = 16 pieces of 1 KiB,
+// 4 pieces per wave. Lane l of piece c covers (row = 8c + l/8,
+// kbyte = (l%8)*16); the source byte offset carries the XOR swizzle.
+__device__ __forceinline__ void g2_stage(const bf16* __restrict__ src_base,
+                                         long row_stride_elems, int kk,
+                                         char* lds, int wid, int lane) {
+#pragma unroll
+  for (int c4 = 0; c4 < 4; ++c4) {
+    const int c = wid * 4 + c4;
+    const int row = 8 * c + (lane >> 3);
+    const int kbyte = (lane & 7) * 16;
+    const int src_byte = kbyte ^ ((row & 7) << 4);
+    const bf16* gsrc = src_base + row * row_stride_elems + kk + src_byte / 2;
+    __builtin_amdgcn_global_load_lds(
+        reinterpret_cast<const unsigned int*>(gsrc),
+        reinterpret_cast<unsigned int*>(lds + c * 1024 + (lane & 63) * 16), 16, 0, 0);
+  }
+}
+
+__global__ __launch_bounds__(256) void gemm2_kernel(
+    const bf16* __restrict__ A, const bf16* __restrict__ W,
+    const float* __restrict__ bias, const bf16* __restrict__ addend,
+    bf16* __restrict__ out, int N, int K, int COL) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  // double-buffered A and B tiles: 4 x 16 KiB
+  // (pointer arrays with addrspace casts are rejected as static
+  // initializers; compute the buffer base per use)
+#define A_BUF(i) (smem + (i)*32768)
+#define B_BUF(i) (smem + 16384 + (i)*32768)
+
+  const int r0 = blockIdx.x * BM2;
+  const int c0 = blockIdx.y * BN2;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x >> 6;
+  const int wm = wid >> 1;  // 2x2 waves, 64x64 tile each
+  const int wn = wid & 1;
+
+  f32x4 acc[4][4] = {};
+
+  const int nt = K / BK2;
+  int cur = 0;
+  g2_stage(A + (long)r0 * K, K, 0, A_BUF(0), wid, lane);
+  g2_stage(W + (long)c0 * K, K, 0, B_BUF(0), wid, lane);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+
+  for (int t = 0; t < nt; ++t) {
+    if (t + 1 < nt) {
+      g2_stage(A + (long)r0 * K, K, (t + 1) * BK2, A_BUF(cur ^ 1), wid, lane);
+      g2_stage(W + (long)c0 * K, K, (t + 1) * BK2, B_BUF(cur ^ 1), wid, lane);
+    }
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      const int kbyte = ks * 64 + (lane >> 4) * 16;
+      bf16x8 a_frag[4], b_frag[4];
+#pragma unroll
+      for (int f = 0; f < 4; ++f) {
+        const int arow = wm * 64 + f * 16 + (lane & 15);
+        const int brow = wn * 64 + f * 16 + (lane & 15);
+        a_frag[f] = *reinterpret_cast<const bf16x8*>(A_BUF(cur) + g2_swz(arow, kbyte));
+        b_frag[f] = *reinterpret_cast<const bf16x8*>(B_BUF(cur) + g2_swz(brow, kbyte));
+      }
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+        for (int fn = 0; fn < 4; ++fn)
+          acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[fm], b_frag[fn], acc[fm][fn], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    cur ^= 1;
+  }
+
+#undef A_BUF
+#undef B_BUF
+  // epilogue: D col = lane&15, row = (lane>>4)*4 + i
+#pragma unroll
+  for (int fm = 0; fm < 4; ++fm) {
+    const int row = r0 + wm * 64 + fm * 16 + (lane >> 4) * 4;
+#pragma unroll
+    for (int fn = 0; fn < 4; ++fn) {
+      const int col = c0 + wn * 64 + fn * 16 + (lane & 15);
+      const float b = bias ? bias[col] : 0.f;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        float v = acc[fm][fn][i] + b;
+        if (addend) v += __bfloat162float(addend[(long)(row + i) * COL + col]);
+        out[(long)(row + i) * COL + col] = __float2bfloat16(v);
+      }
+    }
+  }
+}
+
+void launch_gemm2(const bf16* A, const bf16* W, const float* bias,
+                  const bf16* addend, bf16* out, int N, int K, int COL,
+                  hipStream_t stream) {
+  const dim3 grid(N / BM2, COL / BN2);
+  hipLaunchKernelGGL(gemm2_kernel, grid, dim3(256), 65536, stream, A, W, bias,
+                     addend, out, N, K, COL);
+}

@@ -33,6 +33,7 @@ from ..ops.transformer import (
     bias_gelu,
     flash_attention,
     flash_usable,
+    fused_linear,
     layer_norm,
     masked_softmax_dropout,
 )
@@ -105,7 +106,9 @@ class RobertaSelfAttention(nn.Module):
         B, L, D = x.shape
         H, d = self.num_heads, self.head_dim
         p = self.dropout_p if self.training else 0.0
-        q, k, v = self.query(x), self.key(x), self.value(x)
+        q = fused_linear(x, self.query.weight, self.query.bias)
+        k = fused_linear(x, self.key.weight, self.key.bias)
+        v = fused_linear(x, self.value.weight, self.value.bias)
         if d == 64 and not output_attentions and flash_usable(q, L):
             out = flash_attention(q, k, v, H, valid=valid, scale=1.0 / math.sqrt(d),
                                   dropout_p=p)
@@ -131,7 +134,8 @@ class RobertaSelfOutput(nn.Module):
         self.dropout = nn.Dropout(cfg.hidden_dropout_prob)
 
     def forward(self, hidden, residual):
-        return self.LayerNorm(self.dropout(self.dense(hidden)) + residual)
+        h = fused_linear(hidden, self.dense.weight, self.dense.bias)
+        return self.LayerNorm(self.dropout(h) + residual)
 
 
 class RobertaAttention(nn.Module):
@@ -152,7 +156,7 @@ class RobertaIntermediate(nn.Module):
 
     def forward(self, x):
         # weight-only GEMM; the bias rides in the fused bias+GELU kernel
-        return bias_gelu(torch.nn.functional.linear(x, self.dense.weight), self.dense.bias)
+        return bias_gelu(fused_linear(x, self.dense.weight), self.dense.bias)
 
 
 class RobertaOutput(nn.Module):
@@ -163,7 +167,8 @@ class RobertaOutput(nn.Module):
         self.dropout = nn.Dropout(cfg.hidden_dropout_prob)
 
     def forward(self, hidden, residual):
-        return self.LayerNorm(self.dropout(self.dense(hidden)) + residual)
+        h = fused_linear(hidden, self.dense.weight, self.dense.bias)
+        return self.LayerNorm(self.dropout(h) + residual)
 
 
 class RobertaLayer(nn.Module):
@@ -205,7 +210,10 @@ class RobertaModel(nn.Module):
         valid = attention_mask.sum(dim=1).to(torch.int32)
         x = self.embeddings(input_ids)
         # dtype policy: fp32 master params; under torch.autocast(bf16) the
-        # projections emit bf16 and the custom kernels follow the input dtype
+        # compute dtype is bf16 from the first layer (the custom linear /
+        # flash kernels take bf16 activations against fp32 master weights)
+        if x.is_cuda and torch.is_autocast_enabled():
+            x = x.to(torch.bfloat16)
         hidden, all_probs = self.encoder(x, valid, output_attentions)
         return hidden, all_probs
 

@@ -21,7 +21,7 @@ from typing import Optional
 import torch
 from torch import nn
 
-from ..ops.transformer import flash_attention, flash_usable, masked_softmax_dropout, rms_norm
+from ..ops.transformer import (flash_attention, flash_usable, fused_linear, masked_softmax_dropout, rms_norm)
 
 
 @dataclass
@@ -116,9 +116,9 @@ class T5Attention(nn.Module):
         def split(t, L):
             return t.view(B, L, H, d).transpose(1, 2)
 
-        qp = self.q(x)
-        kp = self.k(src)
-        vp = self.v(src)
+        qp = fused_linear(x, self.q.weight)
+        kp = fused_linear(src, self.k.weight)
+        vp = fused_linear(src, self.v.weight)
         causal = self.causal and kv is None
         if d == 64 and flash_usable(qp, Lq, Lk):
             bias = None
@@ -126,7 +126,7 @@ class T5Attention(nn.Module):
                 bias = position_bias.squeeze(0).float().contiguous()
             out = flash_attention(qp, kp, vp, H, valid=valid, bias=bias, scale=1.0,
                                   causal=causal, dropout_p=dropout_p)
-            return self.o(out)
+            return fused_linear(out, self.o.weight)
         q = split(qp, Lq)
         k = split(kp, Lk)
         v = split(vp, Lk)
@@ -136,7 +136,7 @@ class T5Attention(nn.Module):
         _, probs_dropped = masked_softmax_dropout(scores, valid, 1.0, dropout_p, causal=causal)
         ctx = torch.matmul(probs_dropped, v)
         out = ctx.transpose(1, 2).reshape(B, Lq, H * d)
-        return self.o(out)
+        return fused_linear(out, self.o.weight)
 
 
 class T5LayerSelfAttention(nn.Module):
@@ -172,7 +172,8 @@ class T5DenseActDense(nn.Module):
         self.dropout = nn.Dropout(cfg.dropout_rate)
 
     def forward(self, x):
-        return self.wo(self.dropout(torch.relu(self.wi(x))))
+        h = torch.relu(fused_linear(x, self.wi.weight))
+        return fused_linear(self.dropout(h), self.wo.weight)
 
 
 class T5LayerFF(nn.Module):
@@ -218,6 +219,8 @@ class T5Stack(nn.Module):
 
     def forward(self, input_ids, valid, enc=None, enc_valid=None):
         x = self.dropout(self.embed_tokens(input_ids))
+        if x.is_cuda and torch.is_autocast_enabled():
+            x = x.to(torch.bfloat16)
         L = input_ids.shape[1]
         attn0 = self.block[0].layer[0].SelfAttention
         position_bias = attn0.compute_bias(L, L, input_ids.device)

@@ -127,6 +127,55 @@ def masked_softmax_dropout(
     return p, pd
 
 
+class _LinearBf16(torch.autograd.Function):
+    """Linear layer on the custom transformer-shape GEMMs: forward and
+    input-grad via gemm2 (2-phase glds MFMA), weight grad via the split-K
+    wgrad kernel (fp32 output — the master-weight grad dtype directly),
+    bias grad via colsum."""
+
+    @staticmethod
+    def forward(ctx, x, weight, bias):
+        ext = load_ext(required=True)
+        x2d = x.reshape(-1, x.shape[-1]).contiguous()
+        w16 = weight.to(torch.bfloat16).contiguous()
+        wt16 = w16.t().contiguous()
+        b32 = bias.float().contiguous() if bias is not None else None
+        out = ext.gemm2(x2d, w16, b32, None)
+        ctx.save_for_backward(x2d, w16, wt16)
+        ctx.has_bias = bias is not None
+        ctx.x_shape = x.shape
+        return out.view(*x.shape[:-1], w16.shape[0])
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = load_ext(required=True)
+        x2d, w16, wt16 = ctx.saved_tensors
+        dy2d = dy.reshape(-1, dy.shape[-1]).contiguous()
+        dx = ext.gemm2(dy2d, wt16, None, None)  # dY @ W
+        dw = ext.wgrad(dy2d, x2d)  # (COL, K) fp32
+        db = ext.colsum(dy2d) if ctx.has_bias else None
+        return dx.view(ctx.x_shape), dw, db
+
+
+def linear_usable(x, weight) -> bool:
+    n = x.numel() // x.shape[-1]
+    K, COL = weight.shape[1], weight.shape[0]
+    return (
+        x.is_cuda
+        and x.dtype == torch.bfloat16
+        and n % 128 == 0
+        and K % 64 == 0
+        and COL % 128 == 0
+    )
+
+
+def fused_linear(x, weight, bias=None):
+    """F.linear with custom MFMA kernels where the geometry allows."""
+    if linear_usable(x, weight):
+        return _LinearBf16.apply(x, weight, bias)
+    return torch.nn.functional.linear(x, weight, bias)
+
+
 class _FlashAttention(torch.autograd.Function):
     """Fused attention, head_dim 64, (B, L, H*64) layout (csrc/flash_attn.hip):
     masked softmax + dropout + PV in one kernel, FA2-style two-pass backward.

@@ -25,6 +25,7 @@ ext = CUDAExtension(
         "csrc/wgrad.hip",
         "csrc/transformer_kernels.hip",
         "csrc/flash_attn.hip",
+        "csrc/gemm_bf16.hip",
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],

@@ -148,3 +148,30 @@ def test_linevul_combined_train_step_gpu(dev):
         opt.step()
     assert torch.isfinite(loss)
     assert prob.shape == (4, 2)
+
+
+def test_fused_linear_gpu(dev):
+    """Custom transformer-shape GEMM path vs fp32 F.linear (fwd + all grads)."""
+    from deepdfa_amd.ops.transformer import fused_linear, linear_usable
+
+    torch.manual_seed(7)
+    N, K, COL = 1024, 768, 768
+    x = (torch.randn(2, N // 2, K, device=dev) * 0.5).to(torch.bfloat16).requires_grad_(True)
+    w = torch.randn(COL, K, device=dev) * 0.05
+    w.requires_grad_(True)
+    b = torch.randn(COL, device=dev, requires_grad=True)
+    assert linear_usable(x, w)
+    out = fused_linear(x, w, b)
+    x2 = x.detach().float().requires_grad_(True)
+    w2 = w.detach().clone().requires_grad_(True)
+    b2 = b.detach().clone().requires_grad_(True)
+    ref = torch.nn.functional.linear(x2, w2, b2)
+    scale = ref.abs().max().item()
+    assert (out.float() - ref).abs().max().item() < 0.03 * scale
+    go = torch.randn_like(ref)
+    out.backward(go.to(torch.bfloat16))
+    ref.backward(go)
+    assert (x.grad.float() - x2.grad).abs().max().item() < 0.05 * x2.grad.abs().max().item()
+    assert (w.grad - w2.grad).abs().max().item() < 0.05 * w2.grad.abs().max().item()
+    assert (b.grad - b2.grad).abs().max().item() < 0.05 * b2.grad.abs().max().item()
+    assert w.grad.dtype == torch.float32  # master-weight grad dtype
