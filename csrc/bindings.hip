@@ -68,6 +68,8 @@ void launch_flash_dq(const __hip_bfloat16*, const __hip_bfloat16*, const __hip_b
                      const __hip_bfloat16*, const int*, const float*, const float*, const float*,
                      __hip_bfloat16*, float*, int, int, int, float, int, unsigned,
                      unsigned long long, hipStream_t);
+void launch_wgrad2(const __hip_bfloat16*, const __hip_bfloat16*, float*, int, int, int,
+                   hipStream_t);
 void launch_gemm2(const __hip_bfloat16*, const __hip_bfloat16*, const float*,
                   const __hip_bfloat16*, __hip_bfloat16*, int, int, int, hipStream_t);
 void launch_flash_dkv(const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*,
@@ -284,12 +286,27 @@ at::Tensor gemm_bias(at::Tensor A, at::Tensor W, c10::optional<at::Tensor> bias,
   return out;
 }
 
-// out(M, C) = A(K, M)^T @ B(K, C), fp32 output (wgrad shape class)
+// out(M, C) = A(K, M)^T @ B(K, C), fp32 output (wgrad shape class).
+// 128-aligned shapes take the pipelined wgrad2 kernel; others the generic
+// split-K kernel.
 at::Tensor wgrad(at::Tensor A, at::Tensor B) {
   CHECK_GPU(A);
   CHECK_GPU(B);
   const int K = A.size(0), M = A.size(1), C = B.size(1);
   TORCH_CHECK(B.size(0) == K && M % 64 == 0 && C % 8 == 0);
+  if (M % 128 == 0 && C % 128 == 0 && K % 64 == 0) {
+    const int tiles = (M / 128) * (C / 128);
+    int zsplit = std::max(1, 512 / tiles);
+    int kchunk = (K + zsplit - 1) / zsplit;
+    kchunk = ((kchunk + 63) / 64) * 64;
+    if (kchunk < 512) kchunk = std::min(((K + 63) / 64) * 64, 512);
+    zsplit = (K + kchunk - 1) / kchunk;
+    auto out = zsplit == 1 ? at::empty({M, C}, A.options().dtype(at::kFloat))
+                           : at::zeros({M, C}, A.options().dtype(at::kFloat));
+    launch_wgrad2(ptr<bf16_t>(A), ptr<bf16_t>(B), out.data_ptr<float>(), K, M, C,
+                  cur_stream());
+    return out;
+  }
   auto out = at::zeros({M, C}, A.options().dtype(at::kFloat));
   launch_wgrad(ptr<bf16_t>(A), ptr<bf16_t>(B), nullptr, out.data_ptr<float>(), K, M, C, C,
                cur_stream());
@@ -386,7 +403,7 @@ std::vector<at::Tensor> ggnn_fused_bwd(at::Tensor grad_out, at::Tensor t_indptr,
   auto A_g = Ggicat.view({S * N, 4 * H});
   auto gWcat = at::zeros({4 * H, 2 * H}, opts.dtype(at::kFloat));
   launch_wgrad(ptr<bf16_t>(A_g), ptr<bf16_t>(M), ptr<bf16_t>(HH), gWcat.data_ptr<float>(), S * N,
-               4 * H, 2 * H, H, stream);
+               4 * H, 2 * H, H, stream);  // split-B form (no [m|h] concat)
   auto A_w = Gwh.view({S * N, H});
   auto gW_e = at::zeros({H, H}, opts.dtype(at::kFloat));
   launch_wgrad(ptr<bf16_t>(A_w), ptr<bf16_t>(HH), nullptr, gW_e.data_ptr<float>(), S * N, H, H,

@@ -1,0 +1,158 @@
+// Split-K transpose-A weight-grad GEMM, tiled like gemm2:
+//   out(M, C) += A(K, M)^T @ B(K, C)    (fp32 out, bf16 in)
+//
+// Replaces csrc/wgrad.hip for 128-aligned shapes (transformer weight grads:
+// M = out_features, C = in_features, K = tokens). The first version staged
+// [k][m] tiles naturally and built transposed fragments with scalar LDS
+// reads behind a per-32-K barrier — fully latency-serialized (105 us per
+// 768x768xK=8192 call). Here both operands are register-staged TRANSPOSED
+// into [m][k] / [c][k] XOR-swizzled images (glds cannot transpose), with
+// the T14 split: next tile's global loads issue before the current tile's
+// MFMAs, ds_writes land after the barrier. Fragments are then contiguous
+// b128 reads and the inner loop is gemm2's.
+//
+// Geometry: M % 128 == 0, C % 128 == 0, K % 64 == 0.
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+#define WAVE 64
+#define WBM 128  // out rows (M) per block
+#define WBC 128  // out cols (C) per block
+#define WBK 64   // K rows per tile
+
+using bf16 = __hip_bfloat16;
+using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+using uint4v = __attribute__((ext_vector_type(4))) unsigned int;
+
+__device__ __forceinline__ int w2_swz(int row, int byte) {
+  return row * 128 + (byte ^ ((row & 7) << 4));
+}
+
+// each thread loads 4 x 16 B of one operand tile (64 K-rows x 128 cols):
+// pass p: k = (t>>4) + 16p, col8 = (t&15)*8
+__device__ __forceinline__ void w2_load(const bf16* __restrict__ src, long ld,
+                                        int k0, int col0, int tid,
+                                        uint4v regs[4], int k_lim) {
+  const int kr = tid >> 4;
+  const int c8 = (tid & 15) * 8;
+#pragma unroll
+  for (int p = 0; p < 4; ++p) {
+    const int k = k0 + kr + 16 * p;
+    regs[p] = {};
+    if (k < k_lim)
+      regs[p] = *reinterpret_cast<const uint4v*>(src + (long)k * ld + col0 + c8);
+  }
+}
+
+// transposed write: element (k, col8 + j) -> lds[col][k*2 bytes], swizzled
+__device__ __forceinline__ void w2_write(char* lds, const uint4v regs[4],
+                                         int tid) {
+  const int kr = tid >> 4;
+  const int c8 = (tid & 15) * 8;
+#pragma unroll
+  for (int p = 0; p < 4; ++p) {
+    const int k = kr + 16 * p;
+    bf16 vals[8];
+    *reinterpret_cast<uint4v*>(vals) = regs[p];
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      *reinterpret_cast<bf16*>(lds + w2_swz(c8 + j, k * 2)) = vals[j];
+  }
+}
+
+__global__ __launch_bounds__(256) void wgrad2_kernel(
+    const bf16* __restrict__ A, const bf16* __restrict__ B,
+    float* __restrict__ out, int K, int M, int C, int kchunk, int zsplit) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+#define A_LDS(i) (smem + (i)*32768)            // [m 128][k 64] bf16, 16 KiB
+#define B_LDS(i) (smem + 16384 + (i)*32768)    // [c 128][k 64] bf16
+
+  const int m0 = blockIdx.x * WBM;
+  const int c0 = blockIdx.y * WBC;
+  const int k_begin = blockIdx.z * kchunk;
+  const int k_end = min(K, k_begin + kchunk);
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wid = tid >> 6;
+  const int wm = wid >> 1;
+  const int wc = wid & 1;
+
+  f32x4 acc[4][4] = {};
+
+  uint4v a_regs[4], b_regs[4];
+  w2_load(A, M, k_begin, m0, tid, a_regs, k_end);
+  w2_load(B, C, k_begin, c0, tid, b_regs, k_end);
+  w2_write(A_LDS(0), a_regs, tid);
+  w2_write(B_LDS(0), b_regs, tid);
+  __syncthreads();
+
+  int cur = 0;
+  for (int k0 = k_begin; k0 < k_end; k0 += WBK) {
+    const bool has_next = (k0 + WBK) < k_end;
+    if (has_next) {  // T14: issue next tile's loads before this tile's MFMAs
+      w2_load(A, M, k0 + WBK, m0, tid, a_regs, k_end);
+      w2_load(B, C, k0 + WBK, c0, tid, b_regs, k_end);
+    }
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      const int kbyte = ks * 64 + (lane >> 4) * 16;
+      bf16x8 a_frag[4], b_frag[4];
+#pragma unroll
+      for (int f = 0; f < 4; ++f) {
+        const int mrow = wm * 64 + f * 16 + (lane & 15);
+        const int crow = wc * 64 + f * 16 + (lane & 15);
+        a_frag[f] = *reinterpret_cast<const bf16x8*>(A_LDS(cur) + w2_swz(mrow, kbyte));
+        b_frag[f] = *reinterpret_cast<const bf16x8*>(B_LDS(cur) + w2_swz(crow, kbyte));
+      }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+        for (int fc = 0; fc < 4; ++fc)
+          acc[fm][fc] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[fm], b_frag[fc], acc[fm][fc], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+    }
+    __syncthreads();  // everyone done reading LDS(cur)
+    if (has_next) {
+      w2_write(A_LDS(cur ^ 1), a_regs, tid);
+      w2_write(B_LDS(cur ^ 1), b_regs, tid);
+    }
+    __syncthreads();
+    cur ^= 1;
+  }
+
+#undef A_LDS
+#undef B_LDS
+  // epilogue: D col = lane&15 (C dim), row = (lane>>4)*4 + i (M dim)
+#pragma unroll
+  for (int fm = 0; fm < 4; ++fm) {
+    const int m_base = m0 + wm * 64 + fm * 16 + (lane >> 4) * 4;
+#pragma unroll
+    for (int fc = 0; fc < 4; ++fc) {
+      const int c = c0 + wc * 64 + fc * 16 + (lane & 15);
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        if (zsplit == 1)
+          out[(long)(m_base + i) * C + c] = acc[fm][fc][i];
+        else
+          atomicAdd(out + (long)(m_base + i) * C + c, acc[fm][fc][i]);
+      }
+    }
+  }
+}
+
+void launch_wgrad2(const bf16* A, const bf16* B, float* out, int K, int M,
+                   int C, hipStream_t stream) {
+  const int tiles = (M / WBM) * (C / WBC);
+  int zsplit = max(1, 512 / tiles);
+  int kchunk = (K + zsplit - 1) / zsplit;
+  kchunk = ((kchunk + WBK - 1) / WBK) * WBK;
+  if (kchunk < 512) kchunk = min(((K + WBK - 1) / WBK) * WBK, 512);
+  zsplit = (K + kchunk - 1) / kchunk;
+  const dim3 grid(M / WBM, C / WBC, zsplit);
+  hipLaunchKernelGGL(wgrad2_kernel, grid, dim3(256), 65536, stream, A, B, out,
+                     K, M, C, kchunk, zsplit);
+}

@@ -26,6 +26,7 @@ ext = CUDAExtension(
         "csrc/transformer_kernels.hip",
         "csrc/flash_attn.hip",
         "csrc/gemm_bf16.hip",
+        "csrc/wgrad2.hip",
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
