@@ -70,6 +70,8 @@ void launch_flash_dq(const __hip_bfloat16*, const __hip_bfloat16*, const __hip_b
                      unsigned long long, hipStream_t);
 void launch_wgrad2(const __hip_bfloat16*, const __hip_bfloat16*, float*, int, int, int,
                    hipStream_t);
+void launch_adamw_fused(float*, const float*, float*, float*, long, float, float, float, float,
+                        float, int, int, hipStream_t);
 void launch_gemm2(const __hip_bfloat16*, const __hip_bfloat16*, const float*,
                   const __hip_bfloat16*, __hip_bfloat16*, int, int, int, hipStream_t);
 void launch_flash_dkv(const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*,
@@ -713,6 +715,18 @@ at::Tensor gemm2(at::Tensor A, at::Tensor W, c10::optional<at::Tensor> bias,
   return out;
 }
 
+
+void adamw_fused(at::Tensor p, at::Tensor g, at::Tensor m, at::Tensor v, double lr,
+                 double beta1, double beta2, double eps, double weight_decay, int64_t step,
+                 bool l2_mode) {
+  CHECK_GPU(p);
+  TORCH_CHECK(p.scalar_type() == at::kFloat && g.scalar_type() == at::kFloat);
+  TORCH_CHECK(p.numel() == g.numel() && p.numel() == m.numel() && p.numel() == v.numel());
+  launch_adamw_fused(p.data_ptr<float>(), g.data_ptr<float>(), m.data_ptr<float>(),
+                     v.data_ptr<float>(), p.numel(), (float)lr, (float)beta1, (float)beta2,
+                     (float)eps, (float)weight_decay, (int)step, l2_mode ? 1 : 0, cur_stream());
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "deepdfa_amd MI355X (gfx950) kernels";
   m.def("embed4_fwd", &embed4_fwd);
@@ -740,6 +754,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("flash_attn_fwd", &flash_attn_fwd);
   m.def("flash_attn_bwd", &flash_attn_bwd);
   m.def("gemm2", &gemm2);
+  m.def("adamw_fused", &adamw_fused);
   m.def("softmax_mask_bwd", &softmax_mask_bwd);
   m.def("ggnn_fused_bwd", &ggnn_fused_bwd);
 }

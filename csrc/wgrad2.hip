@@ -65,10 +65,13 @@ __device__ __forceinline__ void w2_write(char* lds, const uint4v regs[4],
 __global__ __launch_bounds__(256) void wgrad2_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ B,
     float* __restrict__ out, int K, int M, int C, int kchunk, int zsplit) {
+  // Natural [k][col] staging (coalesced 16-B LDS writes; the transposed
+  // write variant serialized on 2 banks), transposed FRAGMENT READS as
+  // scalar LDS loads (~4-way), double-buffered with the T14 load split so
+  // the HBM latency hides under the MFMAs.
   extern __shared__ __attribute__((aligned(16))) char smem[];
-#define A_LDS(i) (smem + (i)*32768)            // [m 128][k 64] bf16, 16 KiB
-#define B_LDS(i) (smem + 16384 + (i)*32768)    // [c 128][k 64] bf16
-
+#define A_LDS(i) (smem + (i)*32768)            // [k 64][m 128] bf16, 16 KiB
+#define B_LDS(i) (smem + 16384 + (i)*32768)    // [k 64][c 128] bf16
   const int m0 = blockIdx.x * WBM;
   const int c0 = blockIdx.y * WBC;
   const int k_begin = blockIdx.z * kchunk;
@@ -81,11 +84,20 @@ __global__ __launch_bounds__(256) void wgrad2_kernel(
 
   f32x4 acc[4][4] = {};
 
+  // natural-layout write: thread covers (k = tid>>4 + 16p, col8 = (tid&15)*8)
+  auto write_nat = [&](char* lds, const uint4v regs[4]) {
+    const int kr = tid >> 4;
+    const int c8b = (tid & 15) * 16;
+#pragma unroll
+    for (int p = 0; p < 4; ++p)
+      *reinterpret_cast<uint4v*>(lds + (kr + 16 * p) * 256 + c8b) = regs[p];
+  };
+
   uint4v a_regs[4], b_regs[4];
   w2_load(A, M, k_begin, m0, tid, a_regs, k_end);
   w2_load(B, C, k_begin, c0, tid, b_regs, k_end);
-  w2_write(A_LDS(0), a_regs, tid);
-  w2_write(B_LDS(0), b_regs, tid);
+  write_nat(A_LDS(0), a_regs);
+  write_nat(B_LDS(0), b_regs);
   __syncthreads();
 
   int cur = 0;
@@ -97,14 +109,19 @@ __global__ __launch_bounds__(256) void wgrad2_kernel(
     }
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
-      const int kbyte = ks * 64 + (lane >> 4) * 16;
       bf16x8 a_frag[4], b_frag[4];
 #pragma unroll
       for (int f = 0; f < 4; ++f) {
-        const int mrow = wm * 64 + f * 16 + (lane & 15);
-        const int crow = wc * 64 + f * 16 + (lane & 15);
-        a_frag[f] = *reinterpret_cast<const bf16x8*>(A_LDS(cur) + w2_swz(mrow, kbyte));
-        b_frag[f] = *reinterpret_cast<const bf16x8*>(B_LDS(cur) + w2_swz(crow, kbyte));
+        const int mloc = wm * 64 + f * 16 + (lane & 15);
+        const int cloc = wc * 64 + f * 16 + (lane & 15);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const int k = ks * 32 + (lane >> 4) * 8 + j;
+          a_frag[f][j] = *reinterpret_cast<const __bf16*>(
+              A_LDS(cur) + k * 256 + mloc * 2);
+          b_frag[f][j] = *reinterpret_cast<const __bf16*>(
+              B_LDS(cur) + k * 256 + cloc * 2);
+        }
       }
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
@@ -115,10 +132,10 @@ __global__ __launch_bounds__(256) void wgrad2_kernel(
               a_frag[fm], b_frag[fc], acc[fm][fc], 0, 0, 0);
       __builtin_amdgcn_s_setprio(0);
     }
-    __syncthreads();  // everyone done reading LDS(cur)
+    __syncthreads();
     if (has_next) {
-      w2_write(A_LDS(cur ^ 1), a_regs, tid);
-      w2_write(B_LDS(cur ^ 1), b_regs, tid);
+      write_nat(A_LDS(cur ^ 1), a_regs);
+      write_nat(B_LDS(cur ^ 1), b_regs);
     }
     __syncthreads();
     cur ^= 1;

@@ -1,0 +1,118 @@
+"""Flat fused AdamW/Adam (MI355X-native optimizer).
+
+All parameters are re-viewed into ONE contiguous fp32 master buffer at
+construction (param .data becomes a view; .grad is pre-assigned a view of a
+flat gradient buffer, which autograd accumulates into). The step is then a
+single fused HIP kernel (csrc/adamw.hip), zero_grad is one fill, the global
+grad norm is one reduction, and data-parallel gradient averaging is ONE
+flat RCCL all-reduce (`allreduce_grads`).
+
+288 GB HBM per GPU makes the flat fp32 layout free even for the largest
+model here (~224M params => ~3.6 GB of param+grad+state).
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Iterable, Optional
+
+import torch
+
+
+class FlatAdamW:
+    def __init__(
+        self,
+        params: Iterable[torch.nn.Parameter],
+        lr: float = 1e-3,
+        betas=(0.9, 0.999),
+        eps: float = 1e-8,
+        weight_decay: float = 0.0,
+        l2_mode: bool = False,  # True = classic Adam (L2 in gradient)
+    ):
+        self.params = [p for p in params if p.requires_grad]
+        assert self.params, "no trainable parameters"
+        device = self.params[0].device
+        total = sum(p.numel() for p in self.params)
+        self.flat_p = torch.empty(total, dtype=torch.float32, device=device)
+        self.flat_g = torch.zeros(total, dtype=torch.float32, device=device)
+        self.m = torch.zeros(total, dtype=torch.float32, device=device)
+        self.v = torch.zeros(total, dtype=torch.float32, device=device)
+        off = 0
+        for p in self.params:
+            n = p.numel()
+            self.flat_p[off : off + n].copy_(p.data.float().flatten())
+            p.data = self.flat_p[off : off + n].view(p.shape)
+            p.grad = self.flat_g[off : off + n].view(p.shape)
+            off += n
+        self.lr = lr
+        self.betas = betas
+        self.eps = eps
+        self.weight_decay = weight_decay
+        self.l2_mode = l2_mode
+        self.step_count = 0
+        self.param_groups = [{"params": self.params, "lr": lr}]  # LR-sched compat
+
+    def zero_grad(self, set_to_none: bool = False):
+        self.flat_g.zero_()
+
+    def grad_norm(self) -> torch.Tensor:
+        return torch.linalg.vector_norm(self.flat_g)
+
+    def clip_grad_norm_(self, max_norm: float) -> torch.Tensor:
+        norm = self.grad_norm()
+        scale = max_norm / (norm + 1e-6)
+        if float(scale) < 1.0:
+            self.flat_g.mul_(scale)
+        return norm
+
+    def allreduce_grads(self):
+        import torch.distributed as dist
+
+        if dist.is_available() and dist.is_initialized() and dist.get_world_size() > 1:
+            self.flat_g.div_(dist.get_world_size())
+            dist.all_reduce(self.flat_g)
+
+    @torch.no_grad()
+    def step(self, closure=None):
+        self.step_count += 1
+        lr = self.param_groups[0]["lr"]
+        b1, b2 = self.betas
+        if self.flat_p.is_cuda:
+            from ..ops import load_ext
+
+            ext = load_ext(required=True)
+            ext.adamw_fused(
+                self.flat_p, self.flat_g, self.m, self.v, lr, b1, b2, self.eps,
+                self.weight_decay, self.step_count, self.l2_mode,
+            )
+            return
+        # CPU reference path (same math)
+        g = self.flat_g
+        if self.l2_mode:
+            g = g + self.weight_decay * self.flat_p
+        self.m.mul_(b1).add_(g, alpha=1 - b1)
+        self.v.mul_(b2).addcmul_(g, g, value=1 - b2)
+        bc1 = 1 - b1 ** self.step_count
+        bc2 = 1 - b2 ** self.step_count
+        if not self.l2_mode:
+            self.flat_p.mul_(1 - lr * self.weight_decay)
+        denom = (self.v / bc2).sqrt_().add_(self.eps)
+        self.flat_p.addcdiv_(self.m / bc1, denom, value=-lr)
+
+    # -- minimal state-dict compat -------------------------------------------
+
+    def state_dict(self):
+        return {
+            "flat_p": self.flat_p,
+            "m": self.m,
+            "v": self.v,
+            "step_count": self.step_count,
+            "lr": self.param_groups[0]["lr"],
+        }
+
+    def load_state_dict(self, sd):
+        self.flat_p.copy_(sd["flat_p"])
+        self.m.copy_(sd["m"])
+        self.v.copy_(sd["v"])
+        self.step_count = sd["step_count"]
+        self.param_groups[0]["lr"] = sd["lr"]

@@ -27,6 +27,7 @@ ext = CUDAExtension(
         "csrc/flash_attn.hip",
         "csrc/gemm_bf16.hip",
         "csrc/wgrad2.hip",
+        "csrc/adamw.hip",
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],

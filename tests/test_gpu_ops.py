@@ -203,3 +203,26 @@ def test_train_step_bf16_gpu():
         loss.backward()
         opt.step()
     assert torch.isfinite(loss)
+
+
+def test_fused_adamw_gpu(dev):
+    """Fused flat AdamW kernel vs torch.optim.AdamW on GPU."""
+    from deepdfa_amd.parallel.optim import FlatAdamW
+
+    torch.manual_seed(0)
+    m1 = torch.nn.Linear(256, 256).to(dev)
+    m2 = torch.nn.Linear(256, 256).to(dev)
+    m2.load_state_dict(m1.state_dict())
+    o1 = FlatAdamW(m1.parameters(), lr=1e-2, weight_decay=0.05)
+    o2 = torch.optim.AdamW(m2.parameters(), lr=1e-2, weight_decay=0.05)
+    gen = torch.Generator(device="cpu").manual_seed(1)
+    for _ in range(4):
+        x = torch.randn(32, 256, generator=gen).to(dev)
+        for m, o in ((m1, o1), (m2, o2)):
+            loss = m(x).square().mean()
+            o.zero_grad()
+            loss.backward()
+            o.step()
+    p1 = torch.cat([p.detach().flatten() for p in m1.parameters()])
+    p2 = torch.cat([p.detach().flatten() for p in m2.parameters()])
+    assert torch.allclose(p1, p2, atol=1e-5), (p1 - p2).abs().max()
