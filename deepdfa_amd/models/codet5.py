@@ -62,3 +62,58 @@ class DefectModel(nn.Module):
             loss = torch.nn.functional.cross_entropy(logits, labels)
             return loss, prob
         return prob
+
+
+class RobertaClassificationHead(nn.Module):
+    """Pair-classification head (reference CodeT5/models.py clone head):
+    concat the two EOS vectors -> dense -> tanh -> out_proj."""
+
+    def __init__(self, config: T5Config):
+        super().__init__()
+        self.dense = nn.Linear(config.d_model * 2, config.d_model)
+        self.out_proj = nn.Linear(config.d_model, 2)
+
+    def forward(self, x):
+        x = x.reshape(-1, x.size(-1) * 2)
+        x = torch.tanh(self.dense(x))
+        return self.out_proj(x)
+
+
+class CloneModel(nn.Module):
+    """Clone detection (reference CodeT5/models.py:64-122): each example is
+    a PAIR of functions; both run the full seq2seq EOS pooling; the pair of
+    vectors feeds the classification head."""
+
+    def __init__(self, encoder=None, config: Optional[T5Config] = None, tokenizer=None,
+                 args=None, max_source_length: int = 512):
+        super().__init__()
+        if config is None:
+            config = T5Config()
+        self.config = config
+        self.encoder = encoder if encoder is not None else T5ForConditionalGeneration(config)
+        self.classifier = RobertaClassificationHead(config)
+        self.tokenizer = tokenizer
+        self.args = args
+        self.max_source_length = (
+            getattr(args, "max_source_length", max_source_length) if args else max_source_length
+        )
+
+    def get_t5_vec(self, source_ids):
+        attention_mask = source_ids.ne(self.config.pad_token_id)
+        hidden = self.encoder(source_ids, attention_mask=attention_mask, labels=source_ids,
+                              output_hidden_only=True)
+        eos_mask = source_ids.eq(self.config.eos_token_id)
+        if len(torch.unique(eos_mask.sum(1))) > 1:
+            raise ValueError("All examples must have the same number of <eos> tokens.")
+        b, _, d = hidden.shape
+        return hidden[eos_mask, :].view(b, -1, d)[:, -1, :]
+
+    def forward(self, source_ids, labels=None):
+        source_ids = source_ids.view(-1, self.max_source_length)
+        vec = self.get_t5_vec(source_ids)
+        logits = self.classifier(vec.float())
+        prob = torch.softmax(logits, dim=-1)
+        if labels is not None:
+            loss = torch.nn.functional.cross_entropy(logits, labels)
+            return loss, prob
+        return prob

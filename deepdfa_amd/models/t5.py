@@ -278,3 +278,64 @@ class T5ForConditionalGeneration(nn.Module):
                 logits.float().view(-1, logits.shape[-1]), labels.view(-1), ignore_index=-100
             )
         return loss, logits, dec
+
+
+    @torch.no_grad()
+    def generate(self, input_ids, attention_mask=None, max_length: int = 64,
+                 num_beams: int = 1):
+        """Greedy / beam-search generation (reference CodeT5 Beam,
+        models.py:298-408 capability). The decoder is re-run per step
+        (no KV cache yet); beams are batched through the decoder."""
+        if attention_mask is None:
+            attention_mask = input_ids.ne(self.config.pad_token_id)
+        enc_valid = attention_mask.sum(1).to(torch.int32)
+        enc = self.encoder(input_ids, enc_valid)
+        B = input_ids.shape[0]
+        device = input_ids.device
+        eos, pad, start = (self.config.eos_token_id, self.config.pad_token_id,
+                           self.config.decoder_start_token_id)
+
+        def logits_for(dec_in, enc_rep, valid_rep):
+            dec_valid = torch.full((dec_in.shape[0],), dec_in.shape[1],
+                                   dtype=torch.int32, device=device)
+            dec = self.decoder(dec_in, dec_valid, enc=enc_rep, enc_valid=valid_rep)
+            h = dec[:, -1]
+            if self.config.tie_word_embeddings:
+                h = h * (self.config.d_model ** -0.5)
+            return self.lm_head(h).float()
+
+        if num_beams <= 1:
+            seq = torch.full((B, 1), start, dtype=torch.long, device=device)
+            done = torch.zeros(B, dtype=torch.bool, device=device)
+            for _ in range(max_length - 1):
+                nxt = logits_for(seq, enc, enc_valid).argmax(-1)
+                nxt = torch.where(done, torch.full_like(nxt, pad), nxt)
+                seq = torch.cat([seq, nxt.unsqueeze(1)], dim=1)
+                done |= nxt == eos
+                if bool(done.all()):
+                    break
+            return seq
+        # beam search
+        K = num_beams
+        enc_rep = enc.repeat_interleave(K, dim=0)
+        valid_rep = enc_valid.repeat_interleave(K)
+        seq = torch.full((B * K, 1), start, dtype=torch.long, device=device)
+        beam_scores = torch.full((B, K), -1e9, device=device)
+        beam_scores[:, 0] = 0.0
+        done = torch.zeros(B * K, dtype=torch.bool, device=device)
+        for _ in range(max_length - 1):
+            logp = torch.log_softmax(logits_for(seq, enc_rep, valid_rep), -1)
+            logp = logp.masked_fill(done.unsqueeze(1), 0.0)
+            V = logp.shape[-1]
+            total = (beam_scores.view(-1, 1) + logp).view(B, K * V)
+            top_scores, top_idx = total.topk(K, dim=-1)
+            beam_idx = top_idx // V
+            tok_idx = top_idx % V
+            flat_src = (torch.arange(B, device=device).unsqueeze(1) * K + beam_idx).view(-1)
+            seq = torch.cat([seq[flat_src],
+                             tok_idx.view(-1, 1)], dim=1)
+            done = done[flat_src] | (tok_idx.view(-1) == eos)
+            beam_scores = top_scores
+            if bool(done.all()):
+                break
+        return seq.view(B, K, -1)[:, 0]
