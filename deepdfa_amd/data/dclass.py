@@ -30,14 +30,33 @@ from .. import hashstr
 def synthetic_bigvul_df(
     n: int = 2000, vuln_rate: float = 0.058, seed: int = 0
 ) -> pd.DataFrame:
-    """Big-Vul-shaped metadata table: id, vul, node count. The real dataset
-    is 188k functions with ~5.8% vulnerable (paper Table 6); tests and
-    benchmarks use a scaled-down draw with the same shape."""
+    """Big-Vul-shaped metadata table: id, vul, node count, project. The real
+    dataset is 188k functions with ~5.8% vulnerable (paper Table 6) drawn
+    from ~300 projects; tests and benchmarks use a scaled-down draw with
+    the same shape. The project column backs the cross-project split
+    (reference paper Table 7 / LineVul cross-project scripts)."""
     rng = np.random.RandomState(seed)
     ids = np.arange(n)
     vul = (rng.rand(n) < vuln_rate).astype(np.int64)
     n_nodes = np.clip(np.exp(rng.normal(3.55, 0.75, size=n)).astype(np.int64), 3, 500)
-    return pd.DataFrame({"id": ids, "vul": vul, "n_nodes": n_nodes})
+    project = rng.randint(0, max(2, n // 100), size=n)
+    return pd.DataFrame({"id": ids, "vul": vul, "n_nodes": n_nodes, "project": project})
+
+
+def ds(dsname: str = "bigvul", n: int = 2000, seed: int = 0, sample: bool = False):
+    """Dataset-family dispatch (reference sastvd/helpers/datasets.py:129-292
+    ds/bigvul/devign/mutated): synthetic stand-ins shaped per family."""
+    if sample:
+        n = 200
+    if dsname == "bigvul":
+        return synthetic_bigvul_df(n, seed=seed)
+    if dsname == "devign":  # ~27k functions, ~46% vulnerable
+        return synthetic_bigvul_df(n, vuln_rate=0.46, seed=seed + 7)
+    if dsname == "mutated":  # mutation-augmented variant
+        df = synthetic_bigvul_df(n, seed=seed)
+        df["mutated"] = (np.random.RandomState(seed + 3).rand(n) < 0.5).astype(np.int64)
+        return df
+    raise ValueError(f"unknown dsname {dsname!r}")
 
 
 def ds_partition(
@@ -48,6 +67,21 @@ def ds_partition(
     the property the reference gets from its saved split files);
     "random": seeded shuffle 80/10/10 (datasets.py:475-520 semantics)."""
     if "partition" in df.columns and split == "column":
+        return df[df.partition == partition]
+    if split == "cross_project":
+        # hold out whole projects for val/test (reference cross-project eval)
+        projects = np.sort(df.project.unique())
+        rng = np.random.RandomState(seed)
+        rng.shuffle(projects)
+        n_val = max(1, len(projects) // 10)
+        val_p = set(projects[:n_val].tolist())
+        test_p = set(projects[n_val : 2 * n_val].tolist())
+        part = np.where(
+            df.project.isin(test_p), "test", np.where(df.project.isin(val_p), "val", "train")
+        )
+        df = df.assign(partition=part)
+        if partition == "all":
+            return df
         return df[df.partition == partition]
     if split.startswith("random"):
         s = int(split.split("_")[1]) if "_" in split else seed
@@ -88,7 +122,7 @@ class BigVulDataset:
         self.undersample = undersample
         self.oversample = oversample
         if df is None:
-            df = synthetic_bigvul_df(200 if sample_mode else n_synthetic)
+            df = ds(dsname, n=n_synthetic, seed=seed, sample=sample_mode)
         if sample != -1:
             df = df.sample(sample, random_state=seed)
         if not sample_mode:

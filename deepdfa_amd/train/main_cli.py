@@ -91,6 +91,17 @@ def parse_cli(argv: Optional[List[str]] = None):
             i += 2
         else:
             i += 1
+    if args.hpo_params is None:
+        from ..utils.logging import HPOReporter
+
+        injected = HPOReporter.get_next_parameter()
+        if injected:
+            for k, v in injected.items():
+                node = cfg
+                parts = k.split(".")
+                for part in parts[:-1]:
+                    node = node.setdefault(part, {})
+                node[parts[-1]] = v
     if args.hpo_params:
         with open(args.hpo_params) as f:
             for k, v in json.load(f).items():
@@ -195,15 +206,19 @@ def main(argv: Optional[List[str]] = None) -> Dict:
         if cfg.get("ckpt_path"):
             trainer.load_checkpoint(model, cfg["ckpt_path"])
         if args.subcommand == "fit":
+            from ..utils.logging import HPOReporter
+
             opt = torch.optim.Adam(
                 [p for p in model.parameters() if p.requires_grad], **cfg["optimizer"]
             )
-            out = trainer.fit(model, dm, optimizer=opt)
+            reporter = HPOReporter(run_dir)
+            out = trainer.fit(model, dm, optimizer=opt, hpo_reporter=reporter)
             # post-fit: validate the best checkpoint (main_cli.py:167-184)
             best = out["best_checkpoint"]
             if best:
                 trainer.load_checkpoint(model, best)
                 out["best_validate"] = trainer.validate(model, dm)
+                reporter.report_final(out["best_validate"].get("val_f1", 0.0))
             return out
         if args.subcommand == "validate":
             return trainer.validate(model, dm)

@@ -98,12 +98,15 @@ class Trainer:
 
     # -- loops ----------------------------------------------------------------
 
-    def fit(self, model, datamodule, optimizer=None) -> Dict:
+    def fit(self, model, datamodule, optimizer=None, hpo_reporter=None) -> Dict:
+        from ..utils.logging import ScalarLogger
+
         torch.manual_seed(self.seed)
         model = model.to(self.device)
         if optimizer is None:
             optimizer = torch.optim.Adam(model.parameters(), lr=1e-3, weight_decay=1e-2)
         history = []
+        scalars = ScalarLogger(self.root)
         gen = torch.Generator().manual_seed(self.seed)
         for epoch in range(self.max_epochs):
             model.train()
@@ -152,7 +155,11 @@ class Trainer:
                     f"periodical-{epoch}-{self.global_step}.ckpt",
                 )
             history.append(row)
+            scalars.log({**row, "lr": optimizer.param_groups[0].get("lr", 0.0)}, self.global_step)
+            if hpo_reporter is not None and "val_f1" in row:
+                hpo_reporter.report_intermediate(row["val_f1"])
             logger.info("epoch %d: %s", epoch, json.dumps({k: round(v, 5) if isinstance(v, float) else v for k, v in row.items()}))
+        scalars.close()
         return {"history": history, "best_checkpoint": self.best_checkpoint()}
 
     @torch.no_grad()

@@ -96,3 +96,30 @@ def test_get_indices_missing_join():
 def test_sample_mode_small():
     dm = BigVulDatasetLineVDDataModule(sample_mode=True, batch_size=16)
     assert len(dm.train) == 200
+
+
+def test_dataset_families():
+    from deepdfa_amd.data.dclass import ds
+
+    bv = ds("bigvul", n=500)
+    dv = ds("devign", n=500)
+    assert dv.vul.mean() > 0.3 > bv.vul.mean()
+    mu = ds("mutated", n=100)
+    assert "mutated" in mu.columns
+    import pytest
+
+    with pytest.raises(ValueError):
+        ds("nope")
+
+
+def test_cross_project_split():
+    from deepdfa_amd.data.dclass import ds_partition, synthetic_bigvul_df
+
+    df = synthetic_bigvul_df(1000)
+    tr = ds_partition(df, "train", split="cross_project")
+    te = ds_partition(df, "test", split="cross_project")
+    va = ds_partition(df, "val", split="cross_project")
+    assert len(tr) and len(te) and len(va)
+    # whole projects held out: no project overlap between splits
+    assert not (set(tr.project) & set(te.project))
+    assert not (set(tr.project) & set(va.project))

@@ -60,3 +60,15 @@ def test_freeze_graph_transfer(tmp_path, monkeypatch):
     assert torch.equal(
         model.state_dict()["ggnn.gru.weight_ih"], payload["state_dict"]["ggnn.gru.weight_ih"]
     )
+
+
+def test_scalar_and_hpo_logging(tmp_path, monkeypatch):
+    import json
+
+    monkeypatch.chdir(os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    out, root = run(tmp_path, [])
+    scalars = [json.loads(l) for l in open(os.path.join(root, "scalars.jsonl"))]
+    assert scalars and "train_loss" in scalars[0]
+    hpo = [json.loads(l) for l in open(os.path.join(root, "hpo_metrics.jsonl"))]
+    assert any("intermediate" in r for r in hpo)
+    assert any("final" in r for r in hpo)

@@ -121,3 +121,18 @@ def test_dbize_end_to_end(tmp_path):
     assert gg.num_nodes == g.num_nodes
     # self-loops present
     assert gg.num_edges >= gg.num_nodes
+
+
+def test_preprocess_driver(tmp_path):
+    from deepdfa_amd.pipeline import preprocess
+
+    res = preprocess.main(["--out", str(tmp_path / "proc"), "--n", "30"])
+    out = res["out"]
+    assert os.path.exists(os.path.join(out, "nodes.csv"))
+    assert os.path.exists(os.path.join(out, "statement_labels.pkl"))
+    assert os.path.exists(os.path.join(out, "dataset.parquet"))
+    assert res["vocabs"]["operator"] > 0
+    import pickle
+
+    labels = pickle.load(open(os.path.join(out, "statement_labels.pkl"), "rb"))
+    assert all(isinstance(v, list) for v in labels.values())
