@@ -37,9 +37,16 @@ def build_ddfa(args, rank, device, use_cuda):
     model = FlowGNNGGNNModule(
         feat=FEAT, input_dim=spec.input_dim, hidden_dim=32, n_steps=5, num_output_layers=3
     ).to(device)
-    opt_fn = lambda capturable: torch.optim.Adam(  # noqa: E731
-        model.parameters(), lr=1e-3, weight_decay=1e-2, capturable=capturable, foreach=True
-    )
+
+    def opt_fn(capturable):
+        if capturable:
+            # torch's capturable Adam keeps step counts on-device (correct
+            # bias correction under hipGraph replay)
+            return torch.optim.Adam(model.parameters(), lr=1e-3, weight_decay=1e-2,
+                                    capturable=True, foreach=True)
+        from deepdfa_amd.parallel.optim import FlatAdamW
+
+        return FlatAdamW(model.parameters(), lr=1e-3, weight_decay=1e-2, l2_mode=True)
     batches = [
         synthetic_cfg_batch(args.batch, seed=1000 * rank + i, input_dim=spec.input_dim).to(device)
         for i in range(args.n_batches)
@@ -47,6 +54,8 @@ def build_ddfa(args, rank, device, use_cuda):
     autocast = torch.autocast(device_type="cuda", dtype=torch.bfloat16) if use_cuda else None
 
     def make_step(opt, ddp, set_to_none):
+        flat = hasattr(opt, "allreduce_grads")
+
         def step(i):
             g = batches[i % len(batches)]
             label = model.get_label(g)
@@ -56,9 +65,12 @@ def build_ddfa(args, rank, device, use_cuda):
             else:
                 logits = model(g, {})
             loss = model.loss_fn(logits.float(), label)
-            opt.zero_grad(set_to_none=set_to_none)
+            opt.zero_grad(set_to_none=set_to_none and not flat)
             loss.backward()
-            ddp.finalize()
+            if flat:
+                opt.allreduce_grads()
+            elif ddp is not None:
+                ddp.finalize()
             opt.step()
             return loss
 
@@ -251,7 +263,13 @@ def main():
     use_graphs = use_cuda and capture_ok and not args.no_graph_capture and ws == 1
     opt = opt_fn(capturable=use_graphs)
     if args.model == "ddfa":
-        ddp = DDPEngine(model, bucket_cap_mb=64.0)
+        if hasattr(opt, "flat_p"):
+            ddp = None  # flat-buffer path: broadcast once, all-reduce flat
+            if ws > 1:
+                with torch.no_grad():
+                    torch.distributed.broadcast(opt.flat_p, src=0)
+        else:
+            ddp = DDPEngine(model, bucket_cap_mb=64.0)
     else:
         ddp = None  # FlatAdamW: flat grads + one flat all-reduce; broadcast once
         if ws > 1:

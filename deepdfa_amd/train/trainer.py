@@ -44,7 +44,9 @@ class Trainer:
         grad_clip: Optional[float] = None,
         log_every_n_steps: int = 50,
         seed: int = 0,
+        precision: str = "bf16",  # "bf16" (autocast on GPU) or "fp32"
     ):
+        self.precision = precision
         self.max_epochs = max_epochs
         self.root = default_root_dir
         self.check_val_every = check_val_every_n_epoch
@@ -116,7 +118,8 @@ class Trainer:
             loss_sum = 0.0
             for batch in train_loader:
                 batch = self._to_device(batch)
-                loss = model.training_step(batch)
+                with self._autocast():
+                    loss = model.training_step(batch)
                 optimizer.zero_grad(set_to_none=True)
                 loss.backward()
                 if self.grad_clip:
@@ -169,7 +172,8 @@ class Trainer:
         loss_sum, n = 0.0, 0
         for batch in datamodule.val_dataloader():
             batch = self._to_device(batch)
-            loss = model.validation_step(batch)
+            with self._autocast():
+                loss = model.validation_step(batch)
             loss_sum += float(loss)
             n += 1
         out = {"val_loss": loss_sum / max(1, n)}
@@ -185,7 +189,8 @@ class Trainer:
         loss_sum, n = 0.0, 0
         for batch in datamodule.test_dataloader():
             batch = self._to_device(batch)
-            loss = model.test_step(batch)
+            with self._autocast():
+                loss = model.test_step(batch)
             loss_sum += float(loss)
             n += 1
         results = {"test_loss": loss_sum / max(1, n)}
@@ -193,6 +198,13 @@ class Trainer:
         return results
 
     # -- helpers --------------------------------------------------------------
+
+    def _autocast(self):
+        import contextlib
+
+        if self.precision == "bf16" and self.device.type == "cuda":
+            return torch.autocast(device_type="cuda", dtype=torch.bfloat16)
+        return contextlib.nullcontext()
 
     def _to_device(self, batch):
         if isinstance(batch, tuple) and len(batch) == 2:
