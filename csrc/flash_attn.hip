@@ -624,25 +624,33 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
   if (kv0 < vl) {
     const int q_begin = causal ? kv0 : 0;  // kv0 is a multiple of 32
     const int nstrips = (L - q_begin + TQW - 1) / TQW;
-    for (int strip = wq; strip < nstrips; strip += 2) {
-      const int qw = q_begin + strip * TQW;
-      bf16x8 q_frag[2][2], do_frag[2][2];
+    // T14 prefetch: this strip's Q/dO fragments were loaded at the end of
+    // the previous iteration (or here for the first one)
+    bf16x8 q_frag[2][2], do_frag[2][2];
+    auto load_strip = [&](int strip, bf16x8 qf[2][2], bf16x8 dof[2][2]) {
+      const int qws = q_begin + strip * TQW;
 #pragma unroll
       for (int fq = 0; fq < 2; ++fq)
 #pragma unroll
         for (int ks = 0; ks < 2; ++ks) {
-          const int qrow = qw + fq * 16 + (lane & 15);
+          const int qrow = qws + fq * 16 + (lane & 15);
           const int d = ks * 32 + (lane >> 4) * 8;
-          if (qrow < L) {
-            q_frag[fq][ks] = *reinterpret_cast<const bf16x8*>(
+          if (strip < nstrips && qrow < L) {
+            qf[fq][ks] = *reinterpret_cast<const bf16x8*>(
                 Q + ((long)b * L + qrow) * HD + (long)h * 64 + d);
-            do_frag[fq][ks] = *reinterpret_cast<const bf16x8*>(
+            dof[fq][ks] = *reinterpret_cast<const bf16x8*>(
                 dO + ((long)b * L + qrow) * HD + (long)h * 64 + d);
           } else {
-            q_frag[fq][ks] = bf16x8{};
-            do_frag[fq][ks] = bf16x8{};
+            qf[fq][ks] = bf16x8{};
+            dof[fq][ks] = bf16x8{};
           }
         }
+    };
+    load_strip(wq, q_frag, do_frag);
+    for (int strip = wq; strip < nstrips; strip += 2) {
+      const int qw = q_begin + strip * TQW;
+      bf16x8 qn_frag[2][2], don_frag[2][2];
+      load_strip(strip + 2, qn_frag, don_frag);  // next strip, in flight
       // stage this wave's dO^T and Q^T slices ([d][q] 64-B rows)
 #pragma unroll
       for (int fq = 0; fq < 2; ++fq)
@@ -740,6 +748,13 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
               pa[fk], dob[fd], dv_acc[fk][fd], 0, 0, 0);
           dk_acc[fk][fd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               dsa[fk], qb[fd], dk_acc[fk][fd], 0, 0, 0);
+        }
+#pragma unroll
+      for (int fq = 0; fq < 2; ++fq)
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks) {
+          q_frag[fq][ks] = qn_frag[fq][ks];
+          do_frag[fq][ks] = don_frag[fq][ks];
         }
     }
   }

@@ -105,22 +105,43 @@ __global__ __launch_bounds__(256) void gemm2_kernel(
     cur ^= 1;
   }
 
-#undef A_BUF
-#undef B_BUF
-  // epilogue: D col = lane&15, row = (lane>>4)*4 + i
+  // epilogue: the D fragment layout (col = lane&15, row = (lane>>4)*4+i)
+  // would need 64 x 2-B scalar stores per lane (store-issue bound, T21
+  // diagnostic); bounce the tile through the now-free LDS and store
+  // 16-B-per-lane row-major instead (8 dwordx4 stores per thread).
+  __syncthreads();
+  char* tile = smem;  // [128][128] bf16 = 32 KiB
 #pragma unroll
   for (int fm = 0; fm < 4; ++fm) {
-    const int row = r0 + wm * 64 + fm * 16 + (lane >> 4) * 4;
+    const int row = wm * 64 + fm * 16 + (lane >> 4) * 4;
 #pragma unroll
     for (int fn = 0; fn < 4; ++fn) {
-      const int col = c0 + wn * 64 + fn * 16 + (lane & 15);
-      const float b = bias ? bias[col] : 0.f;
+      const int col = wn * 64 + fn * 16 + (lane & 15);
+      const float b = bias ? bias[c0 + col] : 0.f;
 #pragma unroll
-      for (int i = 0; i < 4; ++i) {
-        float v = acc[fm][fn][i] + b;
-        if (addend) v += __bfloat162float(addend[(long)(row + i) * COL + col]);
-        out[(long)(row + i) * COL + col] = __float2bfloat16(v);
+      for (int i = 0; i < 4; ++i)
+        *reinterpret_cast<__bf16*>(tile + (row + i) * 256 + col * 2) =
+            (__bf16)(acc[fm][fn][i] + b);
+    }
+  }
+  __syncthreads();
+  {
+    const int trow = threadIdx.x >> 4;          // 16 rows per pass, 8 passes
+    const int tcol = (threadIdx.x & 15) * 8;    // 8 bf16 = 16 B per store
+#pragma unroll
+    for (int p = 0; p < 8; ++p) {
+      const int row = trow + p * 16;
+      uint4 v = *reinterpret_cast<const uint4*>(tile + row * 256 + tcol * 2);
+      if (addend) {
+        __bf16 vals[8];
+        *reinterpret_cast<uint4*>(vals) = v;
+        const bf16* add = addend + (long)(r0 + row) * COL + c0 + tcol;
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          vals[j] = (__bf16)((float)vals[j] + __bfloat162float(add[j]));
+        v = *reinterpret_cast<const uint4*>(vals);
       }
+      *reinterpret_cast<uint4*>(out + (long)(r0 + row) * COL + c0 + tcol) = v;
     }
   }
 }
