@@ -9,29 +9,48 @@
 
 #include <hip/hip_runtime.h>
 
+using f4 = __attribute__((ext_vector_type(4))) float;
+
 extern "C" __global__ void adamw_fused_kernel(
     float* __restrict__ p, const float* __restrict__ g, float* __restrict__ m,
     float* __restrict__ v, long n, float lr, float beta1, float beta2,
     float eps, float weight_decay, float bc1, float bc2, int l2_mode) {
-  const long stride = (long)gridDim.x * blockDim.x * 4;
-  for (long i0 = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 4; i0 < n;
-       i0 += stride) {
+  const long nvec = n / 4;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long iv = (long)blockIdx.x * blockDim.x + threadIdx.x; iv < nvec;
+       iv += stride) {
+    f4 pv = reinterpret_cast<f4*>(p)[iv];
+    f4 gv = reinterpret_cast<const f4*>(g)[iv];
+    f4 mv = reinterpret_cast<f4*>(m)[iv];
+    f4 vv = reinterpret_cast<f4*>(v)[iv];
 #pragma unroll
     for (int u = 0; u < 4; ++u) {
-      const long i = i0 + u;
-      if (i >= n) break;
-      float gi = g[i];
-      float pi = p[i];
+      float gi = gv[u];
+      float pi = pv[u];
       if (l2_mode) gi += weight_decay * pi;  // classic Adam L2
-      float mi = beta1 * m[i] + (1.f - beta1) * gi;
-      float vi = beta2 * v[i] + (1.f - beta2) * gi * gi;
-      m[i] = mi;
-      v[i] = vi;
-      const float mhat = mi / bc1;
-      const float vhat = vi / bc2;
+      const float mi = beta1 * mv[u] + (1.f - beta1) * gi;
+      const float vi = beta2 * vv[u] + (1.f - beta2) * gi * gi;
+      mv[u] = mi;
+      vv[u] = vi;
       if (!l2_mode) pi -= lr * weight_decay * pi;  // decoupled decay (AdamW)
-      p[i] = pi - lr * mhat / (sqrtf(vhat) + eps);
+      pv[u] = pi - lr * (mi / bc1) / (sqrtf(vi / bc2) + eps);
     }
+    reinterpret_cast<f4*>(p)[iv] = pv;
+    reinterpret_cast<f4*>(m)[iv] = mv;
+    reinterpret_cast<f4*>(v)[iv] = vv;
+  }
+  // scalar tail
+  for (long i = 4 * nvec + (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    float gi = g[i];
+    float pi = p[i];
+    if (l2_mode) gi += weight_decay * pi;
+    const float mi = beta1 * m[i] + (1.f - beta1) * gi;
+    const float vi = beta2 * v[i] + (1.f - beta2) * gi * gi;
+    m[i] = mi;
+    v[i] = vi;
+    if (!l2_mode) pi -= lr * weight_decay * pi;
+    p[i] = pi - lr * (mi / bc1) / (sqrtf(vi / bc2) + eps);
   }
 }
 

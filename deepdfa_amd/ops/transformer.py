@@ -18,20 +18,21 @@ class _LayerNorm(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, bias, eps):
         ext = load_ext(required=True)
-        y, mean, rstd = ext.layernorm_fwd(
-            x.contiguous(), weight.float().contiguous(), bias.float().contiguous(), eps
-        )
-        ctx.save_for_backward(x, weight, mean, rstd)
+        wf = weight.float().contiguous()
+        bf = bias.float().contiguous()
+        x = x.contiguous()
+        y, mean, rstd = ext.layernorm_fwd(x, wf, bf, eps)
+        ctx.save_for_backward(x, wf, mean, rstd)
         return y
 
     @staticmethod
     def backward(ctx, dy):
         ext = load_ext(required=True)
-        x, weight, mean, rstd = ctx.saved_tensors
+        x, wf, mean, rstd = ctx.saved_tensors
         dy = dy.contiguous()
-        dx = ext.layernorm_bwd(dy, x.contiguous(), weight.float().contiguous(), mean, rstd)
-        dgamma, dbeta = ext.layernorm_wgrad(dy, x.contiguous(), mean, rstd)
-        return dx, dgamma.to(weight.dtype), dbeta.to(weight.dtype), None
+        dx = ext.layernorm_bwd(dy, x, wf, mean, rstd)
+        dgamma, dbeta = ext.layernorm_wgrad(dy, x, mean, rstd)
+        return dx, dgamma, dbeta, None
 
 
 def layer_norm(x, weight, bias, eps: float = 1e-5):
@@ -46,17 +47,19 @@ class _BiasGelu(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, bias):
         ext = load_ext(required=True)
-        ctx.save_for_backward(x, bias)
-        return ext.bias_gelu_fwd(x.contiguous(), bias.float().contiguous())
+        bf = bias.float().contiguous()
+        x = x.contiguous()
+        ctx.save_for_backward(x, bf)
+        return ext.bias_gelu_fwd(x, bf)
 
     @staticmethod
     def backward(ctx, dy):
         ext = load_ext(required=True)
-        x, bias = ctx.saved_tensors
+        x, bf = ctx.saved_tensors
         dy = dy.contiguous()
-        dx = ext.bias_gelu_bwd(dy, x.contiguous(), bias.float().contiguous())
+        dx = ext.bias_gelu_bwd(dy, x, bf)
         dbias = ext.colsum(dx.view(-1, dx.shape[-1]))
-        return dx, dbias.to(bias.dtype)
+        return dx, dbias
 
 
 def bias_gelu(x, bias):
@@ -137,9 +140,19 @@ class _LinearBf16(torch.autograd.Function):
     def forward(ctx, x, weight, bias):
         ext = load_ext(required=True)
         x2d = x.reshape(-1, x.shape[-1]).contiguous()
-        w16 = weight.to(torch.bfloat16).contiguous()
-        wt16 = w16.t().contiguous()
-        b32 = bias.float().contiguous() if bias is not None else None
+        # cache the bf16 / transposed weight per parameter VERSION: the cast
+        # and transpose copies otherwise dominate the elementwise kernel
+        # count (~3 kernels x 72 linears per step)
+        cache = getattr(weight, "_dfa_cast_cache", None)
+        if cache is None or cache[0] != weight._version:
+            w16 = weight.detach().to(torch.bfloat16).contiguous()
+            wt16 = w16.t().contiguous()
+            b32 = bias.detach().float().contiguous() if bias is not None else None
+            weight._dfa_cast_cache = (weight._version, w16, wt16, b32)
+        else:
+            _, w16, wt16, b32 = cache
+            if bias is not None and b32 is None:
+                b32 = bias.detach().float().contiguous()
         out = ext.gemm2(x2d, w16, b32, None)
         ctx.save_for_backward(x2d, w16, wt16)
         ctx.has_bias = bias is not None
@@ -225,18 +238,20 @@ class _RMSNorm(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, eps):
         ext = load_ext(required=True)
-        y, rstd = ext.rmsnorm_fwd(x.contiguous(), weight.float().contiguous(), eps)
-        ctx.save_for_backward(x, weight, rstd)
+        wf = weight.float().contiguous()
+        x = x.contiguous()
+        y, rstd = ext.rmsnorm_fwd(x, wf, eps)
+        ctx.save_for_backward(x, wf, rstd)
         return y
 
     @staticmethod
     def backward(ctx, dy):
         ext = load_ext(required=True)
-        x, weight, rstd = ctx.saved_tensors
+        x, wf, rstd = ctx.saved_tensors
         dy = dy.contiguous()
-        dx = ext.rmsnorm_bwd(dy, x.contiguous(), weight.float().contiguous(), rstd)
-        dgamma = ext.rmsnorm_wgrad(dy, x.contiguous(), rstd)
-        return dx, dgamma.to(weight.dtype), None
+        dx = ext.rmsnorm_bwd(dy, x, wf, rstd)
+        dgamma = ext.rmsnorm_wgrad(dy, x, rstd)
+        return dx, dgamma, None
 
 
 def rms_norm(x, weight, eps: float = 1e-6):
