@@ -217,17 +217,32 @@ template <typename T>
 __global__ void bias_gelu_fwd_kernel(const T* __restrict__ x,
                                      const float* __restrict__ bias,
                                      T* __restrict__ y, long total, int D) {
+  // 4 independent 16-B vectors per iteration: a single in-flight load per
+  // thread left the kernel latency-parked at ~30% of HBM peak (PMC:
+  // SQ_WAIT_ANY ~ 0.9 of WAVE_CYCLES)
   constexpr int VEC = 16 / sizeof(T);
   const long nvec = total / VEC;
-  for (long iv = (long)blockIdx.x * blockDim.x + threadIdx.x; iv < nvec;
-       iv += (long)gridDim.x * blockDim.x) {
-    const long i = iv * VEC;
-    const int d = (int)(i % D);
-    T vx[VEC], vy[VEC];
-    *reinterpret_cast<ulonglong2*>(vx) = *reinterpret_cast<const ulonglong2*>(x + i);
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long iv0 = (long)blockIdx.x * blockDim.x + threadIdx.x; iv0 < nvec;
+       iv0 += stride * 4) {
+    T vx[4][VEC], vy[4][VEC];
+    long idx[4];
 #pragma unroll
-    for (int u = 0; u < VEC; ++u) vy[u] = ff<T>(gelu_f(tf(vx[u]) + bias[d + u]));
-    *reinterpret_cast<ulonglong2*>(y + i) = *reinterpret_cast<ulonglong2*>(vy);
+    for (int r = 0; r < 4; ++r) {
+      idx[r] = iv0 + r * stride;
+      if (idx[r] < nvec)
+        *reinterpret_cast<ulonglong2*>(vx[r]) =
+            *reinterpret_cast<const ulonglong2*>(x + idx[r] * VEC);
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      if (idx[r] >= nvec) continue;
+      const int d = (int)((idx[r] * VEC) % D);
+#pragma unroll
+      for (int u = 0; u < VEC; ++u) vy[r][u] = ff<T>(gelu_f(tf(vx[r][u]) + bias[d + u]));
+      *reinterpret_cast<ulonglong2*>(y + idx[r] * VEC) =
+          *reinterpret_cast<ulonglong2*>(vy[r]);
+    }
   }
 }
 
@@ -238,17 +253,31 @@ __global__ void bias_gelu_bwd_kernel(const T* __restrict__ dy,
                                      T* __restrict__ dx, long total, int D) {
   constexpr int VEC = 16 / sizeof(T);
   const long nvec = total / VEC;
-  for (long iv = (long)blockIdx.x * blockDim.x + threadIdx.x; iv < nvec;
-       iv += (long)gridDim.x * blockDim.x) {
-    const long i = iv * VEC;
-    const int d = (int)(i % D);
-    T vdy[VEC], vx[VEC], vdx[VEC];
-    *reinterpret_cast<ulonglong2*>(vdy) = *reinterpret_cast<const ulonglong2*>(dy + i);
-    *reinterpret_cast<ulonglong2*>(vx) = *reinterpret_cast<const ulonglong2*>(x + i);
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long iv0 = (long)blockIdx.x * blockDim.x + threadIdx.x; iv0 < nvec;
+       iv0 += stride * 4) {
+    T vdy[4][VEC], vx[4][VEC], vdx[4][VEC];
+    long idx[4];
 #pragma unroll
-    for (int u = 0; u < VEC; ++u)
-      vdx[u] = ff<T>(tf(vdy[u]) * gelu_grad_f(tf(vx[u]) + bias[d + u]));
-    *reinterpret_cast<ulonglong2*>(dx + i) = *reinterpret_cast<ulonglong2*>(vdx);
+    for (int r = 0; r < 4; ++r) {
+      idx[r] = iv0 + r * stride;
+      if (idx[r] < nvec) {
+        *reinterpret_cast<ulonglong2*>(vdy[r]) =
+            *reinterpret_cast<const ulonglong2*>(dy + idx[r] * VEC);
+        *reinterpret_cast<ulonglong2*>(vx[r]) =
+            *reinterpret_cast<const ulonglong2*>(x + idx[r] * VEC);
+      }
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      if (idx[r] >= nvec) continue;
+      const int d = (int)((idx[r] * VEC) % D);
+#pragma unroll
+      for (int u = 0; u < VEC; ++u)
+        vdx[r][u] = ff<T>(tf(vdy[r][u]) * gelu_grad_f(tf(vx[r][u]) + bias[d + u]));
+      *reinterpret_cast<ulonglong2*>(dx + idx[r] * VEC) =
+          *reinterpret_cast<ulonglong2*>(vdx[r]);
+    }
   }
 }
 

@@ -70,8 +70,12 @@ __global__ __launch_bounds__(256) void wgrad2_kernel(
   // scalar LDS loads (~4-way), double-buffered with the T14 load split so
   // the HBM latency hides under the MFMAs.
   extern __shared__ __attribute__((aligned(16))) char smem[];
-#define A_LDS(i) (smem + (i)*32768)            // [k 64][m 128] bf16, 16 KiB
-#define B_LDS(i) (smem + 16384 + (i)*32768)    // [k 64][c 128] bf16
+// row stride 272 B (not 256): the transposed scalar fragment reads hit the
+// same bank for every k row at 256 (SQ_LDS_BANK_CONFLICT = 44% of LDS
+// cycles); +16 B shifts each k row by 4 banks
+#define W2_ROWB 272
+#define A_LDS(i) (smem + (i)*35840)            // [k 64][m 128] bf16 padded
+#define B_LDS(i) (smem + 17408 + (i)*35840)    // [k 64][c 128] bf16 padded
   const int m0 = blockIdx.x * WBM;
   const int c0 = blockIdx.y * WBC;
   const int k_begin = blockIdx.z * kchunk;
@@ -90,7 +94,7 @@ __global__ __launch_bounds__(256) void wgrad2_kernel(
     const int c8b = (tid & 15) * 16;
 #pragma unroll
     for (int p = 0; p < 4; ++p)
-      *reinterpret_cast<uint4v*>(lds + (kr + 16 * p) * 256 + c8b) = regs[p];
+      *reinterpret_cast<uint4v*>(lds + (kr + 16 * p) * W2_ROWB + c8b) = regs[p];
   };
 
   uint4v a_regs[4], b_regs[4];
@@ -118,9 +122,9 @@ __global__ __launch_bounds__(256) void wgrad2_kernel(
         for (int j = 0; j < 8; ++j) {
           const int k = ks * 32 + (lane >> 4) * 8 + j;
           a_frag[f][j] = *reinterpret_cast<const __bf16*>(
-              A_LDS(cur) + k * 256 + mloc * 2);
+              A_LDS(cur) + k * W2_ROWB + mloc * 2);
           b_frag[f][j] = *reinterpret_cast<const __bf16*>(
-              B_LDS(cur) + k * 256 + cloc * 2);
+              B_LDS(cur) + k * W2_ROWB + cloc * 2);
         }
       }
       __builtin_amdgcn_s_setprio(1);
@@ -170,6 +174,6 @@ void launch_wgrad2(const bf16* A, const bf16* B, float* out, int K, int M,
   if (kchunk < 512) kchunk = min(((K + WBK - 1) / WBK) * WBK, 512);
   zsplit = (K + kchunk - 1) / kchunk;
   const dim3 grid(M / WBM, C / WBC, zsplit);
-  hipLaunchKernelGGL(wgrad2_kernel, grid, dim3(256), 65536, stream, A, B, out,
+  hipLaunchKernelGGL(wgrad2_kernel, grid, dim3(256), 71680, stream, A, B, out,
                      K, M, C, kchunk, zsplit);
 }
