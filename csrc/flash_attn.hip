@@ -624,33 +624,25 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
   if (kv0 < vl) {
     const int q_begin = causal ? kv0 : 0;  // kv0 is a multiple of 32
     const int nstrips = (L - q_begin + TQW - 1) / TQW;
-    // T14 prefetch: this strip's Q/dO fragments were loaded at the end of
-    // the previous iteration (or here for the first one)
     bf16x8 q_frag[2][2], do_frag[2][2];
-    auto load_strip = [&](int strip, bf16x8 qf[2][2], bf16x8 dof[2][2]) {
-      const int qws = q_begin + strip * TQW;
+    for (int strip = wq; strip < nstrips; strip += 2) {
+      const int qw = q_begin + strip * TQW;
 #pragma unroll
       for (int fq = 0; fq < 2; ++fq)
 #pragma unroll
         for (int ks = 0; ks < 2; ++ks) {
-          const int qrow = qws + fq * 16 + (lane & 15);
+          const int qrow = qw + fq * 16 + (lane & 15);
           const int d = ks * 32 + (lane >> 4) * 8;
-          if (strip < nstrips && qrow < L) {
-            qf[fq][ks] = *reinterpret_cast<const bf16x8*>(
+          if (qrow < L) {
+            q_frag[fq][ks] = *reinterpret_cast<const bf16x8*>(
                 Q + ((long)b * L + qrow) * HD + (long)h * 64 + d);
-            dof[fq][ks] = *reinterpret_cast<const bf16x8*>(
+            do_frag[fq][ks] = *reinterpret_cast<const bf16x8*>(
                 dO + ((long)b * L + qrow) * HD + (long)h * 64 + d);
           } else {
-            qf[fq][ks] = bf16x8{};
-            dof[fq][ks] = bf16x8{};
+            q_frag[fq][ks] = bf16x8{};
+            do_frag[fq][ks] = bf16x8{};
           }
         }
-    };
-    load_strip(wq, q_frag, do_frag);
-    for (int strip = wq; strip < nstrips; strip += 2) {
-      const int qw = q_begin + strip * TQW;
-      bf16x8 qn_frag[2][2], don_frag[2][2];
-      load_strip(strip + 2, qn_frag, don_frag);  // next strip, in flight
       // stage this wave's dO^T and Q^T slices ([d][q] 64-B rows)
 #pragma unroll
       for (int fq = 0; fq < 2; ++fq)
@@ -732,13 +724,26 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
         pa[fk] = *reinterpret_cast<const bf16x8*>(pd_lds + key_loc * 64 + qbyte);
         dsa[fk] = *reinterpret_cast<const bf16x8*>(ds_lds + key_loc * 64 + qbyte);
       }
+      // dO/Q B-fragments read straight from global in transposed order:
+      // the 16-B q_frag/do_frag loads above warmed exactly these L1 lines,
+      // and dropping the [d][q] LDS bounce removes ~2/3 of the kernel's
+      // LDS-issue cost (PMC: WAIT_INST_ANY 55%, LDS_IDX 39% of cycles)
       bf16x8 dob[4], qb[4];
 #pragma unroll
       for (int fd = 0; fd < 4; ++fd) {
         const int d = fd * 16 + (lane & 15);
-        const int qbyte = (lane >> 4) * 16;
-        dob[fd] = *reinterpret_cast<const bf16x8*>(dot_lds + d * 64 + qbyte);
-        qb[fd] = *reinterpret_cast<const bf16x8*>(qt_lds + d * 64 + qbyte);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const int qrow = qw + (lane >> 4) * 8 + j;
+          if (qrow < L) {
+            const long base = ((long)b * L + qrow) * HD + (long)h * 64 + d;
+            dob[fd][j] = *reinterpret_cast<const __bf16*>(dO + base);
+            qb[fd][j] = *reinterpret_cast<const __bf16*>(Q + base);
+          } else {
+            dob[fd][j] = (__bf16)0.0f;
+            qb[fd][j] = (__bf16)0.0f;
+          }
+        }
       }
 #pragma unroll
       for (int fk = 0; fk < 2; ++fk)
@@ -748,13 +753,6 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
               pa[fk], dob[fd], dv_acc[fk][fd], 0, 0, 0);
           dk_acc[fk][fd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               dsa[fk], qb[fd], dk_acc[fk][fd], 0, 0, 0);
-        }
-#pragma unroll
-      for (int fq = 0; fq < 2; ++fq)
-#pragma unroll
-        for (int ks = 0; ks < 2; ++ks) {
-          q_frag[fq][ks] = qn_frag[fq][ks];
-          do_frag[fq][ks] = don_frag[fq][ks];
         }
     }
   }
@@ -826,7 +824,7 @@ void launch_flash_dkv(const bf16* Q, const bf16* K, const bf16* V,
                       unsigned p8, unsigned long long seed,
                       hipStream_t stream) {
   const dim3 grid((L + TK - 1) / TK, B * H);
-  const size_t lds = 2 * TK * 128 + 16384 + 32768;  // K,V + pd/ds + doT/qT
+  const size_t lds = 2 * TK * 128 + 16384;  // K,V + pd/ds bounces
   hipLaunchKernelGGL(flash_dkv_kernel, grid, dim3(256), lds, stream, Q, K, V,
                      dO, valid, bias, lse, Dterm, dK, dV, B, H, L, scale,
                      causal, p8, seed);
