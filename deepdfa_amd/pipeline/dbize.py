@@ -125,3 +125,35 @@ def dbize(
             index=False,
         )
     return vocabs
+
+
+def load_graphs_from_csv(out_dir: str, feat_names=None):
+    """Graph loader over the CSV artifacts (reference
+    sastvd/linevd/graphmogrifier.py:20-97 semantics: merge nodes.csv with
+    the per-feature nodes_feat_*.csv tables, rebuild per-graph BatchedCFGs
+    from edges.csv with self-loops, attach _ABS_DATAFLOW_* and _VULN).
+    The graphs/<id>.pt fast path bakes the same ndata at dbize time; this
+    loader exists for artifact dirs that only carry the CSV tables."""
+    import glob as _glob
+
+    nodes = pd.read_csv(os.path.join(out_dir, "nodes.csv"))
+    edges = pd.read_csv(os.path.join(out_dir, "edges.csv"))
+    feats = {}
+    for path in _glob.glob(os.path.join(out_dir, "nodes_feat_*.csv")):
+        base = os.path.basename(path)[len("nodes_feat_"):-len(".csv")]
+        # name like _ABS_DATAFLOW_api_all_limitall_1000_..._fixed
+        featname = "_".join(base.split("_")[:4])  # _ABS_DATAFLOW_<subkey>
+        feats[featname] = pd.read_csv(path)
+    graphs = {}
+    for gid, nd in nodes.groupby("graph_id"):
+        nd = nd.sort_values("node_idx")
+        n = len(nd)
+        ed = edges[edges.graph_id == gid]
+        ndata = {"_VULN": torch.tensor(nd.vuln.values, dtype=torch.int64)}
+        for featname, fdf in feats.items():
+            sub = fdf[fdf.graph_id == gid].sort_values("node_idx")
+            ndata[featname] = torch.tensor(sub.feat_idx.values, dtype=torch.int64)
+        graphs[int(gid)] = BatchedCFG.from_edges(
+            n, ed.src.tolist(), ed.dst.tolist(), ndata=ndata, add_self_loops=True
+        )
+    return graphs

@@ -136,3 +136,21 @@ def test_preprocess_driver(tmp_path):
 
     labels = pickle.load(open(os.path.join(out, "statement_labels.pkl"), "rb"))
     assert all(isinstance(v, list) for v in labels.values())
+
+
+def test_load_graphs_from_csv(tmp_path):
+    from deepdfa_amd.pipeline.dbize import load_graphs_from_csv
+
+    out = str(tmp_path / "csvout")
+    cpgs = {i: synthetic_cpg(i) for i in range(6)}
+    dbize(cpgs, out, train_ids=range(4))
+    graphs = load_graphs_from_csv(out)
+    assert set(graphs) == set(range(6))
+    g = graphs[0]
+    assert "_ABS_DATAFLOW_api" in g.ndata
+    # matches the baked .pt artifact
+    from deepdfa_amd.graph import BatchedCFG
+
+    ref = BatchedCFG.load(os.path.join(out, "graphs", "0.pt"))
+    assert torch.equal(g.indices, ref.indices)
+    assert torch.equal(g.ndata["_ABS_DATAFLOW_api"], ref.ndata["_ABS_DATAFLOW_api"])

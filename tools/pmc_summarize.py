@@ -17,7 +17,10 @@ JOIN rocpd_kernel_dispatch_{sfx} k ON e.event_id = k.event_id
 JOIN rocpd_info_kernel_symbol_{sfx} s ON k.kernel_id = s.id
 GROUP BY s.display_name, i.name
 """
-rows = cur.execute(q).fetchall()
+try:
+    rows = cur.execute(q).fetchall()
+except sqlite3.OperationalError:
+    rows = []
 agg = {}
 for name, ctr, val, cnt in rows:
     agg.setdefault(name[:56], {})[ctr] = (val, cnt)
@@ -26,10 +29,11 @@ q2 = f"""SELECT s.display_name, SUM(k.end-k.start)/1000.0, COUNT(*)
 FROM rocpd_kernel_dispatch_{sfx} k JOIN rocpd_info_kernel_symbol_{sfx} s ON k.kernel_id=s.id
 GROUP BY s.display_name"""
 wall = {r[0][:56]: (r[1], r[2]) for r in cur.execute(q2).fetchall()}
-print(f"{'kernel':<56} {'calls':>5} {'us_tot':>9} | counters per call")
-for name, ctrs in sorted(agg.items(), key=lambda kv: -wall.get(kv[0], (0, 1))[0]):
+print(f"{'kernel':<56} {'calls':>5} {'us_tot':>9} {'avg_us':>8} | counters per call")
+keys = agg.keys() if agg else wall.keys()
+for name in sorted(keys, key=lambda k: -wall.get(k, (0, 1))[0]):
     w, c = wall.get(name, (0, 1))
-    line = f"{name:<56} {c:>5} {w:>9.0f} | "
-    for ctr, (val, cnt) in sorted(ctrs.items()):
+    line = f"{name:<56} {c:>5} {w:>9.0f} {w/max(1,c):>8.2f} | "
+    for ctr, (val, cnt) in sorted(agg.get(name, {}).items()):
         line += f"{ctr.replace('SQ_','')}={val/max(1,c):.3g} "
     print(line)
