@@ -79,13 +79,17 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
   // ---- load Q fragments (B-operand layout: lane holds Q[qrow][8 d]) ----
   // frag index [fq][ks]: qrow = qw + fq*16 + (lane&15), d = ks*32 + (lane>>4)*8
   bf16x8 q_frag[2][2];
+  const bool q_full = (qw + TQW) <= L;  // uniform: avoids per-load branches
 #pragma unroll
   for (int fq = 0; fq < 2; ++fq) {
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
       const int qrow = qw + fq * 16 + (lane & 15);
       const int d = ks * 32 + (lane >> 4) * 8;
-      if (qrow < L)
+      if (q_full)
+        q_frag[fq][ks] = *reinterpret_cast<const bf16x8*>(
+            Q + ((long)b * L + qrow) * HD + (long)h * 64 + d);
+      else if (qrow < L)
         q_frag[fq][ks] = *reinterpret_cast<const bf16x8*>(
             Q + ((long)b * L + qrow) * HD + (long)h * 64 + d);
       else
@@ -104,19 +108,24 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
     {
       const int row = tid >> 3;          // 0..31
       const int off = (tid & 7) * 16;    // byte offset (8 bf16)
+      const bool kv_full = (kv0 + TK) <= vl;  // uniform fast path
 #pragma unroll
       for (int rr = 0; rr < TK; rr += 32) {
         const int key = kv0 + row + rr;
         uint4v kv = {};
-        if (key < vl)
+        bf16 vv[8] = {};
+        if (kv_full) {
           kv = *reinterpret_cast<const uint4v*>(
               K + ((long)b * L + key) * HD + (long)h * 64 + off / 2);
-        *reinterpret_cast<uint4v*>(k_lds + fa_swz(row + rr, off)) = kv;
-        // V: read 8 contiguous d, scatter-transpose into v_lds[d][key]
-        bf16 vv[8] = {};
-        if (key < vl)
           *reinterpret_cast<uint4v*>(vv) = *reinterpret_cast<const uint4v*>(
               V + ((long)b * L + key) * HD + (long)h * 64 + off / 2);
+        } else if (key < vl) {
+          kv = *reinterpret_cast<const uint4v*>(
+              K + ((long)b * L + key) * HD + (long)h * 64 + off / 2);
+          *reinterpret_cast<uint4v*>(vv) = *reinterpret_cast<const uint4v*>(
+              V + ((long)b * L + key) * HD + (long)h * 64 + off / 2);
+        }
+        *reinterpret_cast<uint4v*>(k_lds + fa_swz(row + rr, off)) = kv;
         const int d0 = off / 2;
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
@@ -383,17 +392,20 @@ __global__ __launch_bounds__(256) void flash_dq_kernel(
   const float dscale = (p8 > 0) ? 256.0f / (256.0f - p8) : 1.0f;
 
   bf16x8 q_frag[2][2], do_frag[2][2];
+  const bool q_full = (qw + TQW) <= L;
 #pragma unroll
   for (int fq = 0; fq < 2; ++fq)
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
       const int qrow = qw + fq * 16 + (lane & 15);
       const int d = ks * 32 + (lane >> 4) * 8;
-      if (qrow < L) {
-        q_frag[fq][ks] = *reinterpret_cast<const bf16x8*>(
-            Q + ((long)b * L + qrow) * HD + (long)h * 64 + d);
-        do_frag[fq][ks] = *reinterpret_cast<const bf16x8*>(
-            dO + ((long)b * L + qrow) * HD + (long)h * 64 + d);
+      const long base = ((long)b * L + qrow) * HD + (long)h * 64 + d;
+      if (q_full) {
+        q_frag[fq][ks] = *reinterpret_cast<const bf16x8*>(Q + base);
+        do_frag[fq][ks] = *reinterpret_cast<const bf16x8*>(dO + base);
+      } else if (qrow < L) {
+        q_frag[fq][ks] = *reinterpret_cast<const bf16x8*>(Q + base);
+        do_frag[fq][ks] = *reinterpret_cast<const bf16x8*>(dO + base);
       } else {
         q_frag[fq][ks] = bf16x8{};
         do_frag[fq][ks] = bf16x8{};
@@ -414,11 +426,17 @@ __global__ __launch_bounds__(256) void flash_dq_kernel(
     {  // stage K (nat + transposed) and V (nat)
       const int row = threadIdx.x >> 3;
       const int off = (threadIdx.x & 7) * 16;
+      const bool kv_full = (kv0 + TK) <= vl;
 #pragma unroll
       for (int rr = 0; rr < TK; rr += 32) {
         const int key = kv0 + row + rr;
         uint4v kv = {}, vv4 = {};
-        if (key < vl) {
+        if (kv_full) {
+          kv = *reinterpret_cast<const uint4v*>(
+              K + ((long)b * L + key) * HD + (long)h * 64 + off / 2);
+          vv4 = *reinterpret_cast<const uint4v*>(
+              V + ((long)b * L + key) * HD + (long)h * 64 + off / 2);
+        } else if (key < vl) {
           kv = *reinterpret_cast<const uint4v*>(
               K + ((long)b * L + key) * HD + (long)h * 64 + off / 2);
           vv4 = *reinterpret_cast<const uint4v*>(
@@ -602,11 +620,17 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
   {  // stage K and V tiles (natural layout, swizzled)
     const int row = threadIdx.x >> 3;
     const int off = (threadIdx.x & 7) * 16;
+    const bool kv_full = (kv0 + TK) <= vl;
 #pragma unroll
     for (int rr = 0; rr < TK; rr += 32) {
       const int key = kv0 + row + rr;
       uint4v kv = {}, vv = {};
-      if (key < vl && kv0 < vl) {
+      if (kv_full) {
+        kv = *reinterpret_cast<const uint4v*>(
+            K + ((long)b * L + key) * HD + (long)h * 64 + off / 2);
+        vv = *reinterpret_cast<const uint4v*>(
+            V + ((long)b * L + key) * HD + (long)h * 64 + off / 2);
+      } else if (key < vl && kv0 < vl) {
         kv = *reinterpret_cast<const uint4v*>(
             K + ((long)b * L + key) * HD + (long)h * 64 + off / 2);
         vv = *reinterpret_cast<const uint4v*>(
@@ -627,17 +651,20 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
     bf16x8 q_frag[2][2], do_frag[2][2];
     for (int strip = wq; strip < nstrips; strip += 2) {
       const int qw = q_begin + strip * TQW;
+      const bool q_full = (qw + TQW) <= L;
 #pragma unroll
       for (int fq = 0; fq < 2; ++fq)
 #pragma unroll
         for (int ks = 0; ks < 2; ++ks) {
           const int qrow = qw + fq * 16 + (lane & 15);
           const int d = ks * 32 + (lane >> 4) * 8;
-          if (qrow < L) {
-            q_frag[fq][ks] = *reinterpret_cast<const bf16x8*>(
-                Q + ((long)b * L + qrow) * HD + (long)h * 64 + d);
-            do_frag[fq][ks] = *reinterpret_cast<const bf16x8*>(
-                dO + ((long)b * L + qrow) * HD + (long)h * 64 + d);
+          const long base = ((long)b * L + qrow) * HD + (long)h * 64 + d;
+          if (q_full) {
+            q_frag[fq][ks] = *reinterpret_cast<const bf16x8*>(Q + base);
+            do_frag[fq][ks] = *reinterpret_cast<const bf16x8*>(dO + base);
+          } else if (qrow < L) {
+            q_frag[fq][ks] = *reinterpret_cast<const bf16x8*>(Q + base);
+            do_frag[fq][ks] = *reinterpret_cast<const bf16x8*>(dO + base);
           } else {
             q_frag[fq][ks] = bf16x8{};
             do_frag[fq][ks] = bf16x8{};
@@ -729,19 +756,33 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
       // and dropping the [d][q] LDS bounce removes ~2/3 of the kernel's
       // LDS-issue cost (PMC: WAIT_INST_ANY 55%, LDS_IDX 39% of cycles)
       bf16x8 dob[4], qb[4];
+      if (q_full) {
 #pragma unroll
-      for (int fd = 0; fd < 4; ++fd) {
-        const int d = fd * 16 + (lane & 15);
+        for (int fd = 0; fd < 4; ++fd) {
+          const int d = fd * 16 + (lane & 15);
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          const int qrow = qw + (lane >> 4) * 8 + j;
-          if (qrow < L) {
-            const long base = ((long)b * L + qrow) * HD + (long)h * 64 + d;
+          for (int j = 0; j < 8; ++j) {
+            const long base =
+                ((long)b * L + qw + (lane >> 4) * 8 + j) * HD + (long)h * 64 + d;
             dob[fd][j] = *reinterpret_cast<const __bf16*>(dO + base);
             qb[fd][j] = *reinterpret_cast<const __bf16*>(Q + base);
-          } else {
-            dob[fd][j] = (__bf16)0.0f;
-            qb[fd][j] = (__bf16)0.0f;
+          }
+        }
+      } else {
+#pragma unroll
+        for (int fd = 0; fd < 4; ++fd) {
+          const int d = fd * 16 + (lane & 15);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            const int qrow = qw + (lane >> 4) * 8 + j;
+            if (qrow < L) {
+              const long base = ((long)b * L + qrow) * HD + (long)h * 64 + d;
+              dob[fd][j] = *reinterpret_cast<const __bf16*>(dO + base);
+              qb[fd][j] = *reinterpret_cast<const __bf16*>(Q + base);
+            } else {
+              dob[fd][j] = (__bf16)0.0f;
+              qb[fd][j] = (__bf16)0.0f;
+            }
           }
         }
       }

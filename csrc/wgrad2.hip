@@ -37,12 +37,21 @@ __device__ __forceinline__ void w2_load(const bf16* __restrict__ src, long ld,
                                         uint4v regs[4], int k_lim) {
   const int kr = tid >> 4;
   const int c8 = (tid & 15) * 8;
+  // a per-element bound check makes hipcc branch around each load and
+  // drain vmcnt per element (guide §5 trap (c)); hoist to one uniform test
+  if (k0 + 64 <= k_lim) {
 #pragma unroll
-  for (int p = 0; p < 4; ++p) {
-    const int k = k0 + kr + 16 * p;
-    regs[p] = {};
-    if (k < k_lim)
-      regs[p] = *reinterpret_cast<const uint4v*>(src + (long)k * ld + col0 + c8);
+    for (int p = 0; p < 4; ++p)
+      regs[p] = *reinterpret_cast<const uint4v*>(
+          src + (long)(k0 + kr + 16 * p) * ld + col0 + c8);
+  } else {
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      const int k = k0 + kr + 16 * p;
+      regs[p] = {};
+      if (k < k_lim)
+        regs[p] = *reinterpret_cast<const uint4v*>(src + (long)k * ld + col0 + c8);
+    }
   }
 }
 
