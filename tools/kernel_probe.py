@@ -5,7 +5,11 @@ rounds in one process, medians reported — the definitive per-kernel numbers
 on one box/clock. Run on a GPU box: python tools/kernel_probe.py
 """
 
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
