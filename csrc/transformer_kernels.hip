@@ -217,30 +217,38 @@ template <typename T>
 __global__ void bias_gelu_fwd_kernel(const T* __restrict__ x,
                                      const float* __restrict__ bias,
                                      T* __restrict__ y, long total, int D) {
-  // 4 independent 16-B vectors per iteration: a single in-flight load per
-  // thread left the kernel latency-parked at ~30% of HBM peak (PMC:
-  // SQ_WAIT_ANY ~ 0.9 of WAVE_CYCLES)
+  // 4 independent 16-B vectors in flight per iteration (latency), bias
+  // column tracked incrementally (one modulo at entry; a per-vector 64-bit
+  // modulo costs ~20 VALU ops on this erf-bound kernel)
   constexpr int VEC = 16 / sizeof(T);
   const long nvec = total / VEC;
   const long stride = (long)gridDim.x * blockDim.x;
-  for (long iv0 = (long)blockIdx.x * blockDim.x + threadIdx.x; iv0 < nvec;
-       iv0 += stride * 4) {
+  const int step = (int)((stride * VEC) % D);
+  long iv = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  int d0 = (int)((iv * VEC) % D);
+  for (; iv < nvec; iv += stride * 4) {
     T vx[4][VEC], vy[4][VEC];
-    long idx[4];
+    int ds[4];
+    int d = d0;
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      idx[r] = iv0 + r * stride;
-      if (idx[r] < nvec)
+      const long i = iv + r * stride;
+      ds[r] = d;
+      d += step;
+      if (d >= D) d -= D;
+      if (i < nvec)
         *reinterpret_cast<ulonglong2*>(vx[r]) =
-            *reinterpret_cast<const ulonglong2*>(x + idx[r] * VEC);
+            *reinterpret_cast<const ulonglong2*>(x + i * VEC);
     }
+    d0 = d;
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      if (idx[r] >= nvec) continue;
-      const int d = (int)((idx[r] * VEC) % D);
+      const long i = iv + r * stride;
+      if (i >= nvec) continue;
 #pragma unroll
-      for (int u = 0; u < VEC; ++u) vy[r][u] = ff<T>(gelu_f(tf(vx[r][u]) + bias[d + u]));
-      *reinterpret_cast<ulonglong2*>(y + idx[r] * VEC) =
+      for (int u = 0; u < VEC; ++u)
+        vy[r][u] = ff<T>(gelu_f(tf(vx[r][u]) + bias[ds[r] + u]));
+      *reinterpret_cast<ulonglong2*>(y + i * VEC) =
           *reinterpret_cast<ulonglong2*>(vy[r]);
     }
   }
@@ -254,28 +262,35 @@ __global__ void bias_gelu_bwd_kernel(const T* __restrict__ dy,
   constexpr int VEC = 16 / sizeof(T);
   const long nvec = total / VEC;
   const long stride = (long)gridDim.x * blockDim.x;
-  for (long iv0 = (long)blockIdx.x * blockDim.x + threadIdx.x; iv0 < nvec;
-       iv0 += stride * 4) {
+  const int step = (int)((stride * VEC) % D);
+  long iv = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  int d0 = (int)((iv * VEC) % D);
+  for (; iv < nvec; iv += stride * 4) {
     T vdy[4][VEC], vx[4][VEC], vdx[4][VEC];
-    long idx[4];
+    int ds[4];
+    int d = d0;
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      idx[r] = iv0 + r * stride;
-      if (idx[r] < nvec) {
+      const long i = iv + r * stride;
+      ds[r] = d;
+      d += step;
+      if (d >= D) d -= D;
+      if (i < nvec) {
         *reinterpret_cast<ulonglong2*>(vdy[r]) =
-            *reinterpret_cast<const ulonglong2*>(dy + idx[r] * VEC);
+            *reinterpret_cast<const ulonglong2*>(dy + i * VEC);
         *reinterpret_cast<ulonglong2*>(vx[r]) =
-            *reinterpret_cast<const ulonglong2*>(x + idx[r] * VEC);
+            *reinterpret_cast<const ulonglong2*>(x + i * VEC);
       }
     }
+    d0 = d;
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      if (idx[r] >= nvec) continue;
-      const int d = (int)((idx[r] * VEC) % D);
+      const long i = iv + r * stride;
+      if (i >= nvec) continue;
 #pragma unroll
       for (int u = 0; u < VEC; ++u)
-        vdx[r][u] = ff<T>(tf(vdy[r][u]) * gelu_grad_f(tf(vx[r][u]) + bias[d + u]));
-      *reinterpret_cast<ulonglong2*>(dx + idx[r] * VEC) =
+        vdx[r][u] = ff<T>(tf(vdy[r][u]) * gelu_grad_f(tf(vx[r][u]) + bias[ds[r] + u]));
+      *reinterpret_cast<ulonglong2*>(dx + i * VEC) =
           *reinterpret_cast<ulonglong2*>(vdx[r]);
     }
   }
