@@ -66,6 +66,13 @@ class HashTokenizer:
         return self.vocab_size
 
 
+def synthetic_corpus(n: int = 256, seed: int = 0) -> List[str]:
+    """Synthetic C-function corpus for offline tokenizer training."""
+    from .text_dataset import synthetic_func_source
+
+    return [synthetic_func_source(seed * 100000 + i) for i in range(n)]
+
+
 def train_bpe_tokenizer(corpus: List[str], vocab_size: int = 50265):
     """Train a byte-level BPE offline (HF `tokenizers`)."""
     from tokenizers import Tokenizer, models, pre_tokenizers, trainers
@@ -73,6 +80,21 @@ def train_bpe_tokenizer(corpus: List[str], vocab_size: int = 50265):
     tok = Tokenizer(models.BPE(unk_token="<unk>"))
     tok.pre_tokenizer = pre_tokenizers.ByteLevel(add_prefix_space=False)
     trainer = trainers.BpeTrainer(
+        vocab_size=vocab_size, special_tokens=["<s>", "<pad>", "</s>", "<unk>", "<mask>"]
+    )
+    tok.train_from_iterator(corpus, trainer)
+    return tok
+
+
+def train_word_level_tokenizer(corpus: List[str], vocab_size: int = 50265):
+    """Train a whitespace word-level tokenizer offline (the reference's
+    --use_word_level_tokenizer path, linevul_main.py:608-610, which loads a
+    pretrained word_level/train_word_level.json asset)."""
+    from tokenizers import Tokenizer, models, pre_tokenizers, trainers
+
+    tok = Tokenizer(models.WordLevel(unk_token="<unk>"))
+    tok.pre_tokenizer = pre_tokenizers.Whitespace()
+    trainer = trainers.WordLevelTrainer(
         vocab_size=vocab_size, special_tokens=["<s>", "<pad>", "</s>", "<unk>", "<mask>"]
     )
     tok.train_from_iterator(corpus, trainer)

@@ -209,7 +209,68 @@ def build_args(argv=None):
     p.add_argument("--train_data_file", default=None)
     p.add_argument("--eval_data_file", default=None)
     p.add_argument("--test_data_file", default=None)
+    # tokenizer selection (reference linevul_main.py:600-616): pretrained dir,
+    # word-level / BPE trained offline from the train corpus, or the
+    # deterministic hashing fallback (default in this asset-free env)
+    p.add_argument("--tokenizer_name", default=None,
+                   help="HF tokenizer directory (pretrained parity path)")
+    p.add_argument("--use_word_level_tokenizer", action="store_true")
+    p.add_argument("--use_non_pretrained_tokenizer", action="store_true",
+                   help="train a byte-level BPE from the training corpus")
     return p.parse_args(argv)
+
+
+def build_tokenizer(args, cfg, corpus=None):
+    if args.tokenizer_name:
+        from ..data.tokenization import load_pretrained_tokenizer
+
+        return _wrap_hf(load_pretrained_tokenizer(args.tokenizer_name), cfg)
+    if args.use_word_level_tokenizer or args.use_non_pretrained_tokenizer:
+        from ..data.tokenization import (
+            synthetic_corpus,
+            train_bpe_tokenizer,
+            train_word_level_tokenizer,
+        )
+
+        corpus = corpus or synthetic_corpus(256)
+        train = (
+            train_word_level_tokenizer
+            if args.use_word_level_tokenizer
+            else train_bpe_tokenizer
+        )
+        return _wrap_hf(train(corpus, vocab_size=cfg.vocab_size), cfg)
+    return HashTokenizer(vocab_size=cfg.vocab_size)
+
+
+def _wrap_hf(tok, cfg):
+    """Adapt a `tokenizers`/HF tokenizer to the driver surface
+    (encode(text, max_length) -> padded ids + cls/sep/pad ids)."""
+
+    class _Wrapped:
+        def __init__(self):
+            self.cls_token_id, self.pad_token_id, self.sep_token_id = 0, 1, 2
+            for name, attr in (("<s>", "cls_token_id"), ("<pad>", "pad_token_id"),
+                               ("</s>", "sep_token_id")):
+                tid = tok.token_to_id(name) if hasattr(tok, "token_to_id") else None
+                if tid is not None:
+                    setattr(self, attr, tid)
+            self.vocab_size = (
+                tok.get_vocab_size() if hasattr(tok, "get_vocab_size") else len(tok)
+            )
+
+        def encode(self, text, max_length=512):
+            if hasattr(tok, "encode") and hasattr(tok, "token_to_id"):
+                ids = tok.encode(text).ids
+            else:  # transformers AutoTokenizer
+                ids = tok.encode(text, add_special_tokens=False)
+            ids = [self.cls_token_id] + ids[: max_length - 2] + [self.sep_token_id]
+            ids += [self.pad_token_id] * (max_length - len(ids))
+            return ids[:max_length]
+
+        def __len__(self):
+            return self.vocab_size
+
+    return _Wrapped()
 
 
 def main(argv=None):
@@ -222,7 +283,7 @@ def main(argv=None):
 
     n_syn = 200 if args.sample else args.n_synthetic
     cfg = RobertaConfig(num_hidden_layers=args.num_layers)
-    tokenizer = HashTokenizer(vocab_size=cfg.vocab_size)
+    tokenizer = build_tokenizer(args, cfg)
     flowgnn_datamodule = None
     flowgnn_model = None
     if not args.no_flowgnn:

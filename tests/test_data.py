@@ -123,3 +123,41 @@ def test_cross_project_split():
     # whole projects held out: no project overlap between splits
     assert not (set(tr.project) & set(te.project))
     assert not (set(tr.project) & set(va.project))
+
+
+def test_trained_tokenizers_roundtrip():
+    """Offline BPE + word-level training and the driver wrapper surface
+    (reference --use_word_level_tokenizer / non-pretrained BPE paths)."""
+    from deepdfa_amd.data.tokenization import (
+        synthetic_corpus,
+        train_bpe_tokenizer,
+        train_word_level_tokenizer,
+    )
+    from deepdfa_amd.train.linevul_main import _wrap_hf
+    from deepdfa_amd.models.roberta import RobertaConfig
+
+    corpus = synthetic_corpus(32)
+    cfg = RobertaConfig(vocab_size=600)
+    for trainer in (train_bpe_tokenizer, train_word_level_tokenizer):
+        tok = _wrap_hf(trainer(corpus, vocab_size=600), cfg)
+        ids = tok.encode(corpus[0], max_length=64)
+        assert len(ids) == 64
+        assert ids[0] == tok.cls_token_id
+        assert tok.sep_token_id in ids
+        assert all(0 <= i < len(tok) for i in ids)
+        # deterministic
+        assert ids == tok.encode(corpus[0], max_length=64)
+
+
+def test_linevul_tokenizer_flag_routing():
+    from deepdfa_amd.train.linevul_main import build_args, build_tokenizer
+    from deepdfa_amd.data.tokenization import HashTokenizer
+    from deepdfa_amd.models.roberta import RobertaConfig
+
+    cfg = RobertaConfig(vocab_size=600)
+    args = build_args([])
+    assert isinstance(build_tokenizer(args, cfg), HashTokenizer)
+    args = build_args(["--use_word_level_tokenizer"])
+    tok = build_tokenizer(args, cfg)
+    assert not isinstance(tok, HashTokenizer)
+    assert len(tok.encode("int main() { return 0; }", max_length=32)) == 32
