@@ -56,7 +56,37 @@ def ds(dsname: str = "bigvul", n: int = 2000, seed: int = 0, sample: bool = Fals
         df = synthetic_bigvul_df(n, seed=seed)
         df["mutated"] = (np.random.RandomState(seed + 3).rand(n) < 0.5).astype(np.int64)
         return df
+    if dsname == "dbgbench":
+        return synthetic_dbgbench_df(max(2, n // 10), seed=seed)
     raise ValueError(f"unknown dsname {dsname!r}")
+
+
+def synthetic_dbgbench_df(n_bugs: int = 100, seed: int = 0) -> pd.DataFrame:
+    """DbgBench-shaped table (reference LineVul/unixcoder/linevul_main.py:
+    142-145 and --dbgbench_ddfa :1530-1575): each real bug appears as a
+    buggy version plus one or more developer-patched versions; the variant
+    name column `c` contains "patched" for fixed code, and the label is
+    derived as `"patched" not in c`. Used as a held-out TEST set for models
+    trained on Big-Vul."""
+    rng = np.random.RandomState(seed + 11)
+    rows = []
+    rid = 0
+    for bug in range(n_bugs):
+        n_patches = 1 + rng.randint(2)
+        variants = ["buggy"] + [f"patched-dev{i}" for i in range(n_patches)]
+        for c in variants:
+            rows.append(
+                {
+                    "id": rid,
+                    "c": f"bug{bug}.{c}",
+                    "vul": int("patched" not in c),
+                    "n_nodes": int(np.clip(np.exp(rng.normal(3.55, 0.75)), 3, 500)),
+                    "project": bug % 7,
+                    "split": "holdout",
+                }
+            )
+            rid += 1
+    return pd.DataFrame(rows)
 
 
 def ds_partition(

@@ -39,6 +39,11 @@ def tokenise(s: str) -> List[str]:
     return out
 
 
+def tokenise_lines(s: str) -> List[List[str]]:
+    """Per-line subtoken split (reference tokenise.py:23-35)."""
+    return [tokenise(line) for line in s.splitlines()]
+
+
 class HashTokenizer:
     """Deterministic vocabulary-hashing tokenizer (no assets needed)."""
 

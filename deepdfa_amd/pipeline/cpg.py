@@ -126,3 +126,31 @@ def synthetic_cpg(_id: int, n_stmts: int = 12) -> CPG:
     for s in stmt_nodes:
         edges.append((method, s, "AST"))
     return CPG(nodes, edges)
+
+
+# graph reduction by edge-type set (reference sastvd/helpers/joern.py:419-441
+# rdg): which edge types each gtype keeps
+RDG_TYPES = {
+    "reftype": {"EVAL_TYPE", "REF"},
+    "ast": {"AST"},
+    "pdg": {"REACHING_DEF", "CDG"},
+    "cfgcdg": {"CFG", "CDG"},
+    "cfg": {"CFG"},
+    "all": {"REACHING_DEF", "CDG", "AST", "EVAL_TYPE", "REF"},
+    "dataflow": {"CFG", "AST"},
+}
+
+
+def rdg(cpg: CPG, gtype: str) -> CPG:
+    """Reduce the CPG to the edge types of `gtype` (joern.py rdg parity)."""
+    keep = RDG_TYPES[gtype]
+    return CPG(dict(cpg.nodes), [e for e in cpg.edges if e[2] in keep])
+
+
+def drop_lone_nodes(cpg: CPG) -> CPG:
+    """Remove nodes with no edge connections (joern.py:485-493)."""
+    touched = set()
+    for s, d, _ in cpg.edges:
+        touched.add(s)
+        touched.add(d)
+    return CPG({i: p for i, p in cpg.nodes.items() if i in touched}, list(cpg.edges))

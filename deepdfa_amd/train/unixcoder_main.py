@@ -239,6 +239,17 @@ def main(argv=None):
                           n_synthetic=args.n_synthetic)
         for part in ("train", "val", "test")
     }
+    if args.dbgbench:
+        # DbgBench is a held-out TEST set (reference unixcoder
+        # linevul_main.py:133,142-145: split "holdout" -> test, label =
+        # "patched" not in variant name); train/val stay Big-Vul
+        from ..data.dclass import ds
+        from ..data.text_dataset import synthetic_func_source
+
+        dbg = ds("dbgbench", n=args.n_synthetic, seed=args.seed)
+        dbg = dbg.assign(func=[synthetic_func_source(i + 10**6) for i in dbg.id])
+        datasets["test"] = TextDataset(tokenizer, args, df=dbg,
+                                       block_size=args.block_size)
     results = {}
     os.makedirs(args.output_dir, exist_ok=True)
     if args.do_train:

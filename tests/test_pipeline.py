@@ -154,3 +154,25 @@ def test_load_graphs_from_csv(tmp_path):
     ref = BatchedCFG.load(os.path.join(out, "graphs", "0.pt"))
     assert torch.equal(g.indices, ref.indices)
     assert torch.equal(g.ndata["_ABS_DATAFLOW_api"], ref.ndata["_ABS_DATAFLOW_api"])
+
+
+def test_rdg_and_drop_lone_nodes():
+    from deepdfa_amd.pipeline.cpg import drop_lone_nodes, rdg, synthetic_cpg
+
+    cpg = synthetic_cpg(7)
+    cfg_only = rdg(cpg, "cfg")
+    assert cfg_only.edges and all(t == "CFG" for _, _, t in cfg_only.edges)
+    ast_only = rdg(cpg, "ast")
+    assert all(t == "AST" for _, _, t in ast_only.edges)
+    # dropping lone nodes after CFG reduction keeps only CFG-touched nodes
+    reduced = drop_lone_nodes(cfg_only)
+    touched = {x for s, d, _ in cfg_only.edges for x in (s, d)}
+    assert set(reduced.nodes) == touched
+    assert len(reduced.nodes) < len(cpg.nodes)
+
+
+def test_tokenise_lines():
+    from deepdfa_amd.data.tokenization import tokenise_lines
+
+    out = tokenise_lines("int fooBar = 1;\nreturn fooBar;")
+    assert out == [["int", "foo", "Bar", "1"], ["return", "foo", "Bar"]]

@@ -72,3 +72,21 @@ def test_eval_export_and_codet5_export(tmp_path):
     lines = [json.loads(l) for l in open(out_jsonl)]
     assert len(lines) == len(ds)
     assert {"idx", "target", "func"} <= set(lines[0])
+
+
+def test_dbgbench_family_and_holdout_eval(tmp_path):
+    from deepdfa_amd.data.dclass import ds
+
+    df = ds("dbgbench", n=200, seed=0)
+    # variants: buggy rows labeled 1, patched rows labeled 0
+    assert set(df.vul.unique()) == {0, 1}
+    assert (df[df.c.str.contains("patched")].vul == 0).all()
+    assert (df[~df.c.str.contains("patched")].vul == 1).all()
+    assert (df.split == "holdout").all()
+
+    res = uxc.main([
+        "--do_test", "--dbgbench", "--n_synthetic", "120",
+        "--num_layers", "1", "--block_size", "64",
+        "--eval_batch_size", "8", "--output_dir", str(tmp_path / "uxc"),
+    ])
+    assert "test" in res
