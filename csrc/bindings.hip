@@ -81,6 +81,7 @@ void launch_flash_dkv(const __hip_bfloat16*, const __hip_bfloat16*, const __hip_
 template <typename T>
 void launch_softmax_mask_bwd(const T*, const T*, T*, long, int, float, float,
                              unsigned long long, hipStream_t);
+void dkv_prof_fetch(unsigned long long*);
 
 #define CHECK_GPU(t) \
   TORCH_CHECK((t).is_cuda() && (t).is_contiguous(), #t " must be contiguous GPU tensor")
@@ -753,6 +754,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_wgrad", &rmsnorm_wgrad);
   m.def("flash_attn_fwd", &flash_attn_fwd);
   m.def("flash_attn_bwd", &flash_attn_bwd);
+  m.def("dkv_prof", []() {
+    // variant-9 instrumentation readout: per-segment cycle totals
+    auto t = torch::zeros({8}, torch::dtype(torch::kLong));
+    dkv_prof_fetch(reinterpret_cast<unsigned long long*>(t.data_ptr<long>()));
+    return t;
+  });
   m.def("gemm2", &gemm2);
   m.def("adamw_fused", &adamw_fused);
   m.def("softmax_mask_bwd", &softmax_mask_bwd);

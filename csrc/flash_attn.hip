@@ -583,6 +583,16 @@ __device__ __forceinline__ void recompute_pT32(
     }
 }
 
+// dO^T/Q^T bounce image (variant 2): per-wave [d 64][q 32] bf16 rows padded
+// to 72 B so the 16-B B-fragment reads spread across the 64-dword bank
+// modulus instead of 4-way conflicting at a 64-B stride.
+#define DKV_T_STRIDE 72
+#define DKV_T_BYTES (64 * DKV_T_STRIDE)
+
+// variant-9 instrumentation accumulator (cycles per kernel segment)
+__device__ unsigned long long dfa_dkv_prof[8];
+
+template <int VAR>
 __global__ __launch_bounds__(256) void flash_dkv_kernel(
     const bf16* __restrict__ Q, const bf16* __restrict__ K,
     const bf16* __restrict__ V, const bf16* __restrict__ dO,
@@ -593,8 +603,12 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
   // Wave grid 2 (key halves) x 2 (q interleave): halves the per-wave
   // accumulator footprint (the 64-key variant needed 128 fp32 accumulators
   // on top of ~176 VGPRs -> 1 wave/SIMD on the unified register file).
-  // LDS map: K nat 8K | V nat 8K | per-wave pd 2K, ds 2K, doT 4K, qT 4K
-  // (reused as the fp32 reduction buffer afterwards).
+  // LDS map: K nat 8K | V nat 8K | per-wave pd 2K, ds 2K (reused as the
+  // fp32 reduction buffer afterwards) | VAR 2 only: per-wave doT/qT images.
+  // Variants (within-probe A/B, see profiles/flash_dkv_pmc.md):
+  //   0 = direct L1 scalar dob/qb loads
+  //   2 = 0 with the transposed fragments bounced through LDS (vector reads)
+  //   3 = 0 + s_setprio(1) on the younger wave half through the MFMA loop
   extern __shared__ __attribute__((aligned(16))) char smem[];
   char* k_lds = smem;
   char* v_lds = smem + TK * 128;
@@ -614,8 +628,10 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
   const float dscale = (p8 > 0) ? 256.0f / (256.0f - p8) : 1.0f;
   char* pd_lds = wave_base + wid * 2048;
   char* ds_lds = wave_base + 8192 + wid * 2048;
-  char* dot_lds = wave_base + 16384 + wid * 4096;  // [d 64][q 32] bf16
-  char* qt_lds = wave_base + 32768 + wid * 4096;
+  char* dot_lds = wave_base + 16384 + wid * 2 * DKV_T_BYTES;
+  char* qt_lds = dot_lds + DKV_T_BYTES;
+  if (VAR == 3 && __builtin_amdgcn_readfirstlane(threadIdx.x) >= 128)
+    __builtin_amdgcn_s_setprio(1);
 
   {  // stage K and V tiles (natural layout, swizzled)
     const int row = threadIdx.x >> 3;
@@ -645,6 +661,17 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
   f32x4 dv_acc[2][4] = {};  // [fkey][fd] over this wave's 32-key half
   f32x4 dk_acc[2][4] = {};
 
+  // VAR 9: per-segment cycle instrumentation (s_memtime; guide §5.4 —
+  // ~+11% overhead, relative shares are what matters)
+  unsigned long long segt[8] = {};
+  unsigned long long seg_last = (VAR == 9) ? __builtin_readcyclecounter() : 0;
+#define SEG_MARK(i)                                          \
+  if (VAR == 9) {                                            \
+    unsigned long long now = __builtin_readcyclecounter();   \
+    segt[i] += now - seg_last;                               \
+    seg_last = now;                                          \
+  }
+
   if (kv0 < vl) {
     const int q_begin = causal ? kv0 : 0;  // kv0 is a multiple of 32
     const int nstrips = (L - q_begin + TQW - 1) / TQW;
@@ -652,6 +679,7 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
     for (int strip = wq; strip < nstrips; strip += 2) {
       const int qw = q_begin + strip * TQW;
       const bool q_full = (qw + TQW) <= L;
+      SEG_MARK(7)  // loop overhead / previous-iteration tail
 #pragma unroll
       for (int fq = 0; fq < 2; ++fq)
 #pragma unroll
@@ -670,20 +698,22 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
             do_frag[fq][ks] = bf16x8{};
           }
         }
-      // stage this wave's dO^T and Q^T slices ([d][q] 64-B rows)
+      if (VAR == 2) {
+        // stage this wave's dO^T and Q^T slices ([d][q] padded rows)
 #pragma unroll
-      for (int fq = 0; fq < 2; ++fq)
+        for (int fq = 0; fq < 2; ++fq)
 #pragma unroll
-        for (int ks = 0; ks < 2; ++ks)
+          for (int ks = 0; ks < 2; ++ks)
 #pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            const int d = ks * 32 + (lane >> 4) * 8 + j;
-            const int q_loc = fq * 16 + (lane & 15);
-            *reinterpret_cast<bf16*>(dot_lds + d * 64 + q_loc * 2) =
-                (bf16)do_frag[fq][ks][j];
-            *reinterpret_cast<bf16*>(qt_lds + d * 64 + q_loc * 2) =
-                (bf16)q_frag[fq][ks][j];
-          }
+            for (int j = 0; j < 8; ++j) {
+              const int d = ks * 32 + (lane >> 4) * 8 + j;
+              const int q_loc = fq * 16 + (lane & 15);
+              *reinterpret_cast<bf16*>(dot_lds + d * DKV_T_STRIDE + q_loc * 2) =
+                  (bf16)do_frag[fq][ks][j];
+              *reinterpret_cast<bf16*>(qt_lds + d * DKV_T_STRIDE + q_loc * 2) =
+                  (bf16)q_frag[fq][ks][j];
+            }
+      }
 
       float lse_w[2], dterm_w[2];
 #pragma unroll
@@ -692,10 +722,12 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
         lse_w[fq] = (qrow < L) ? lse[(long)bh * L + qrow] : 0.f;
         dterm_w[fq] = (qrow < L) ? Dterm[(long)bh * L + qrow] : 0.f;
       }
+      SEG_MARK(0)  // q/dO fragment + lse/dterm loads
 
       float p[2][2][4];
       recompute_pT32(k_lds, q_frag, bias, lse_w, lane, h, L, qw, kv0, key_off,
                      vl, scale, causal, p);
+      SEG_MARK(1)  // P recompute (QK^T MFMAs + exp)
 
       f32x4 dp_acc[2][2] = {};
 #pragma unroll
@@ -714,6 +746,7 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
             dp_acc[fk][fq] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                 v_frag[fk], do_frag[fq][ks], dp_acc[fk][fq], 0, 0, 0);
       }
+      SEG_MARK(2)  // dP = dO V^T MFMAs
 
       // pd (dropout-masked P) and ds = scale * P (dP - D) -> LDS bounce
 #pragma unroll
@@ -741,6 +774,7 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
             *reinterpret_cast<bf16*>(ds_lds + key_loc * 64 + q_loc * 2) =
                 __float2bfloat16(dsv);
           }
+      SEG_MARK(3)  // pd/ds compute + LDS bounce writes
 
       // dV += pd(key, q) @ dO(q, d);  dK += ds(key, q) @ Q(q, d)
       bf16x8 pa[2], dsa[2];
@@ -756,7 +790,16 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
       // and dropping the [d][q] LDS bounce removes ~2/3 of the kernel's
       // LDS-issue cost (PMC: WAIT_INST_ANY 55%, LDS_IDX 39% of cycles)
       bf16x8 dob[4], qb[4];
-      if (q_full) {
+      if (VAR == 2) {
+        // vector B-fragment reads from the padded [d][q] bounce images
+        // (per-wave buffers: the compiler orders same-wave LDS write->read)
+#pragma unroll
+        for (int fd = 0; fd < 4; ++fd) {
+          const int off = (fd * 16 + (lane & 15)) * DKV_T_STRIDE + (lane >> 4) * 16;
+          dob[fd] = *reinterpret_cast<const bf16x8*>(dot_lds + off);
+          qb[fd] = *reinterpret_cast<const bf16x8*>(qt_lds + off);
+        }
+      } else if (q_full) {
 #pragma unroll
         for (int fd = 0; fd < 4; ++fd) {
           const int d = fd * 16 + (lane & 15);
@@ -786,6 +829,7 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
           }
         }
       }
+      SEG_MARK(4)  // pa/dsa LDS reads + transposed dob/qb loads
 #pragma unroll
       for (int fk = 0; fk < 2; ++fk)
 #pragma unroll
@@ -795,7 +839,48 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
           dk_acc[fk][fd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               dsa[fk], qb[fd], dk_acc[fk][fd], 0, 0, 0);
         }
+      SEG_MARK(5)  // dV/dK MFMAs
     }
+  }
+
+  if (VAR == 4) {
+    // pair-wise reduce: the two waves of a key half (wq 0/1) hold the only
+    // partials for those 32 keys. wq=1 vector-writes its accs to a [d][key]
+    // f32 image with 272-B padded rows ((4d + key) mod 64 distinct within
+    // every b128 access -> conflict-free), wq=0 adds and stores to global.
+    // Replaces the atomic+zero-fill epilogue that was 45% of the kernel
+    // (s_memtime segment breakdown, profiles/flash_dkv_pmc.md).
+    char* img = wave_base;  // 64 * 272 B = 17 KiB (reuses the pd/ds region)
+#define FA_PAIR_REDUCE(ACC, OUT)                                              \
+  __syncthreads();                                                            \
+  if (wq == 1) {                                                              \
+    _Pragma("unroll") for (int fk = 0; fk < 2; ++fk)                          \
+        _Pragma("unroll") for (int fd = 0; fd < 4; ++fd) {                    \
+      const int d = fd * 16 + (lane & 15);                                    \
+      const int key_loc = key_off + fk * 16 + (lane >> 4) * 4;                \
+      *reinterpret_cast<f32x4*>(img + d * 272 + key_loc * 4) = ACC[fk][fd];   \
+    }                                                                         \
+  }                                                                           \
+  __syncthreads();                                                            \
+  if (wq == 0) {                                                              \
+    _Pragma("unroll") for (int fk = 0; fk < 2; ++fk)                          \
+        _Pragma("unroll") for (int fd = 0; fd < 4; ++fd) {                    \
+      const int d = fd * 16 + (lane & 15);                                    \
+      const int key_loc = key_off + fk * 16 + (lane >> 4) * 4;                \
+      const f32x4 other =                                                     \
+          *reinterpret_cast<const f32x4*>(img + d * 272 + key_loc * 4);       \
+      _Pragma("unroll") for (int i = 0; i < 4; ++i) {                         \
+        const int key = kv0 + key_loc + i;                                    \
+        if (key < L)                                                          \
+          OUT[((long)b * L + key) * HD + (long)h * 64 + d] =                  \
+              __float2bfloat16(ACC[fk][fd][i] + other[i]);                    \
+      }                                                                       \
+    }                                                                         \
+  }
+    FA_PAIR_REDUCE(dv_acc, dV)
+    FA_PAIR_REDUCE(dk_acc, dK)
+#undef FA_PAIR_REDUCE
+    return;
   }
 
   // cross-wave reduce through LDS (2 waves per key half), then plain store
@@ -822,6 +907,12 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
   FA_REDUCE_STORE(dv_acc, dV)
   FA_REDUCE_STORE(dk_acc, dK)
 #undef FA_REDUCE_STORE
+  SEG_MARK(6)  // cross-wave reduce + store epilogue
+  if (VAR == 9 && lane == 0) {
+#pragma unroll
+    for (int i = 0; i < 8; ++i) atomicAdd(&dfa_dkv_prof[i], segt[i]);
+  }
+#undef SEG_MARK
 }
 
 void launch_flash_fwd(const bf16* Q, const bf16* K, const bf16* V,
@@ -865,8 +956,38 @@ void launch_flash_dkv(const bf16* Q, const bf16* K, const bf16* V,
                       unsigned p8, unsigned long long seed,
                       hipStream_t stream) {
   const dim3 grid((L + TK - 1) / TK, B * H);
-  const size_t lds = 2 * TK * 128 + 16384;  // K,V + pd/ds bounces
-  hipLaunchKernelGGL(flash_dkv_kernel, grid, dim3(256), lds, stream, Q, K, V,
-                     dO, valid, bias, lse, Dterm, dK, dV, B, H, L, scale,
-                     causal, p8, seed);
+  size_t lds = 2 * TK * 128 + 16384;  // K,V + pd/ds bounces
+  const char* e = getenv("DFA_DKV_VARIANT");  // re-read: lets one probe
+  const int var = e ? atoi(e) : 4;            // process A/B the variants
+  if (var == 2) {
+    lds += 4 * 2 * DKV_T_BYTES;
+    hipLaunchKernelGGL(flash_dkv_kernel<2>, grid, dim3(256), lds, stream, Q, K,
+                       V, dO, valid, bias, lse, Dterm, dK, dV, B, H, L, scale,
+                       causal, p8, seed);
+  } else if (var == 4) {
+    lds = 2 * TK * 128 + 64 * 272;  // K,V + pd/ds reused as the pair image
+    hipLaunchKernelGGL(flash_dkv_kernel<4>, grid, dim3(256), lds, stream, Q, K,
+                       V, dO, valid, bias, lse, Dterm, dK, dV, B, H, L, scale,
+                       causal, p8, seed);
+  } else if (var == 3) {
+    hipLaunchKernelGGL(flash_dkv_kernel<3>, grid, dim3(256), lds, stream, Q, K,
+                       V, dO, valid, bias, lse, Dterm, dK, dV, B, H, L, scale,
+                       causal, p8, seed);
+  } else if (var == 9) {
+    hipLaunchKernelGGL(flash_dkv_kernel<9>, grid, dim3(256), lds, stream, Q, K,
+                       V, dO, valid, bias, lse, Dterm, dK, dV, B, H, L, scale,
+                       causal, p8, seed);
+  } else {
+    hipLaunchKernelGGL(flash_dkv_kernel<0>, grid, dim3(256), lds, stream, Q, K,
+                       V, dO, valid, bias, lse, Dterm, dK, dV, B, H, L, scale,
+                       causal, p8, seed);
+  }
+}
+
+void dkv_prof_fetch(unsigned long long* out) {
+  // read + zero the variant-9 per-segment cycle accumulators
+  (void)hipMemcpyFromSymbol(out, HIP_SYMBOL(dfa_dkv_prof),
+                            8 * sizeof(unsigned long long));
+  unsigned long long z[8] = {};
+  (void)hipMemcpyToSymbol(HIP_SYMBOL(dfa_dkv_prof), z, sizeof(z));
 }
