@@ -509,6 +509,8 @@ at::Tensor bias_gelu_fwd(at::Tensor x, at::Tensor bias) {
   CHECK_GPU(x);
   TORCH_CHECK(bias.scalar_type() == at::kFloat);
   const int D = x.size(-1);
+  TORCH_CHECK(D % (x.scalar_type() == at::kFloat ? 4 : 8) == 0,
+              "bias_gelu: D must be a multiple of the 16-B vector width");
   auto y = at::empty_like(x);
   dispatch_float_bf16(x, "bias_gelu_fwd", [&](auto tag) {
     using T = decltype(tag);

@@ -219,8 +219,12 @@ __global__ void bias_gelu_fwd_kernel(const T* __restrict__ x,
                                      T* __restrict__ y, long total, int D) {
   // 4 independent 16-B vectors in flight per iteration (latency), bias
   // column tracked incrementally (one modulo at entry; a per-vector 64-bit
-  // modulo costs ~20 VALU ops on this erf-bound kernel)
+  // modulo costs ~20 VALU ops), bias row staged in LDS (per-element global
+  // bias loads were 64 scalar vmem ops per iteration on a streaming kernel)
   constexpr int VEC = 16 / sizeof(T);
+  extern __shared__ float bias_lds[];
+  for (int i = threadIdx.x; i < D; i += blockDim.x) bias_lds[i] = bias[i];
+  __syncthreads();
   const long nvec = total / VEC;
   const long stride = (long)gridDim.x * blockDim.x;
   const int step = (int)((stride * VEC) % D);
@@ -245,9 +249,14 @@ __global__ void bias_gelu_fwd_kernel(const T* __restrict__ x,
     for (int r = 0; r < 4; ++r) {
       const long i = iv + r * stride;
       if (i >= nvec) continue;
+      float b[VEC];
+#pragma unroll
+      for (int u = 0; u < VEC; u += 4)
+        *reinterpret_cast<float4*>(b + u) =
+            *reinterpret_cast<const float4*>(bias_lds + ds[r] + u);
 #pragma unroll
       for (int u = 0; u < VEC; ++u)
-        vy[r][u] = ff<T>(gelu_f(tf(vx[r][u]) + bias[ds[r] + u]));
+        vy[r][u] = ff<T>(gelu_f(tf(vx[r][u]) + b[u]));
       *reinterpret_cast<ulonglong2*>(y + i * VEC) =
           *reinterpret_cast<ulonglong2*>(vy[r]);
     }
@@ -260,6 +269,9 @@ __global__ void bias_gelu_bwd_kernel(const T* __restrict__ dy,
                                      const float* __restrict__ bias,
                                      T* __restrict__ dx, long total, int D) {
   constexpr int VEC = 16 / sizeof(T);
+  extern __shared__ float bias_lds[];
+  for (int i = threadIdx.x; i < D; i += blockDim.x) bias_lds[i] = bias[i];
+  __syncthreads();
   const long nvec = total / VEC;
   const long stride = (long)gridDim.x * blockDim.x;
   const int step = (int)((stride * VEC) % D);
@@ -287,9 +299,14 @@ __global__ void bias_gelu_bwd_kernel(const T* __restrict__ dy,
     for (int r = 0; r < 4; ++r) {
       const long i = iv + r * stride;
       if (i >= nvec) continue;
+      float b[VEC];
+#pragma unroll
+      for (int u = 0; u < VEC; u += 4)
+        *reinterpret_cast<float4*>(b + u) =
+            *reinterpret_cast<const float4*>(bias_lds + ds[r] + u);
 #pragma unroll
       for (int u = 0; u < VEC; ++u)
-        vdx[r][u] = ff<T>(tf(vdy[r][u]) * gelu_grad_f(tf(vx[r][u]) + bias[ds[r] + u]));
+        vdx[r][u] = ff<T>(tf(vdy[r][u]) * gelu_grad_f(tf(vx[r][u]) + b[u]));
       *reinterpret_cast<ulonglong2*>(dx + i * VEC) =
           *reinterpret_cast<ulonglong2*>(vdx[r]);
     }
@@ -479,8 +496,8 @@ void launch_bias_gelu_fwd(const T* x, const float* bias, T* y, long total,
   const long nvec = total / (16 / sizeof(T));
   const int grid = (int)min((nvec + block - 1) / block, (long)4096);
   if (grid)
-    hipLaunchKernelGGL(bias_gelu_fwd_kernel<T>, dim3(grid), dim3(block), 0,
-                       stream, x, bias, y, total, D);
+    hipLaunchKernelGGL(bias_gelu_fwd_kernel<T>, dim3(grid), dim3(block),
+                       D * sizeof(float), stream, x, bias, y, total, D);
 }
 
 template <typename T>
@@ -490,8 +507,8 @@ void launch_bias_gelu_bwd(const T* dy, const T* x, const float* bias, T* dx,
   const long nvec = total / (16 / sizeof(T));
   const int grid = (int)min((nvec + block - 1) / block, (long)4096);
   if (grid)
-    hipLaunchKernelGGL(bias_gelu_bwd_kernel<T>, dim3(grid), dim3(block), 0,
-                       stream, dy, x, bias, dx, total, D);
+    hipLaunchKernelGGL(bias_gelu_bwd_kernel<T>, dim3(grid), dim3(block),
+                       D * sizeof(float), stream, dy, x, bias, dx, total, D);
 }
 
 template <typename T>
