@@ -310,16 +310,26 @@ __global__ void flash_dterm_kernel(const bf16* __restrict__ dO,
                                    const bf16* __restrict__ O,
                                    float* __restrict__ Dterm, int B, int H,
                                    int L) {
-  // one wave per (b, q, h) row of 64 elements
-  const long row = (long)(blockIdx.x * (blockDim.x / WAVE)) + threadIdx.x / WAVE;
+  // 8 lanes per 64-element (b, q, h) row via 16-B loads -> 8 rows per wave
+  // (the 1-row-per-wave form was ~17 instructions per 256 B of traffic and
+  // ran 5x off the copy roofline)
   const int lane = threadIdx.x & (WAVE - 1);
+  const int sub = lane >> 3;  // row slot within the wave
+  const int sl = lane & 7;    // lane within the row
   const long R = (long)B * L * H;
+  const long row =
+      ((long)blockIdx.x * (blockDim.x / WAVE) + threadIdx.x / WAVE) * 8 + sub;
   if (row >= R) return;  // row = ((b*L + q)*H + h)
-  const long base = row * 64;
-  float acc = __bfloat162float(dO[base + lane]) * __bfloat162float(O[base + lane]);
+  const long base = row * 64 + sl * 8;
+  const bf16x8 a = *reinterpret_cast<const bf16x8*>(dO + base);
+  const bf16x8 o = *reinterpret_cast<const bf16x8*>(O + base);
+  float acc = 0.f;
 #pragma unroll
-  for (int off = WAVE / 2; off > 0; off >>= 1) acc += __shfl_down(acc, off);
-  if (lane == 0) {
+  for (int j = 0; j < 8; ++j) acc += (float)a[j] * (float)o[j];
+  acc += __shfl_down(acc, 4);
+  acc += __shfl_down(acc, 2);
+  acc += __shfl_down(acc, 1);
+  if (sl == 0) {
     const long bq = row / H;
     const int h = (int)(row % H);
     const long b = bq / L;
@@ -929,11 +939,10 @@ void launch_flash_fwd(const bf16* Q, const bf16* K, const bf16* V,
 void launch_flash_dterm(const bf16* dO, const bf16* O, float* Dterm, int B,
                         int H, int L, hipStream_t stream) {
   const long rows = (long)B * L * H;
-  const int rows_per_block = 4;
+  const int rows_per_block = 32;  // 4 waves x 8 rows
   hipLaunchKernelGGL(flash_dterm_kernel,
                      dim3((rows + rows_per_block - 1) / rows_per_block),
-                     dim3(WAVE * rows_per_block), 0, stream, dO, O, Dterm, B, H,
-                     L);
+                     dim3(256), 0, stream, dO, O, Dterm, B, H, L);
 }
 
 void launch_flash_dq(const bf16* Q, const bf16* K, const bf16* V,
