@@ -79,10 +79,15 @@ __global__ __launch_bounds__(256) void wgrad2_kernel(
   // scalar LDS loads (~4-way), double-buffered with the T14 load split so
   // the HBM latency hides under the MFMAs.
   extern __shared__ __attribute__((aligned(16))) char smem[];
-// row stride 272 B (not 256): the transposed scalar fragment reads hit the
-// same bank for every k row at 256 (SQ_LDS_BANK_CONFLICT = 44% of LDS
-// cycles); +16 B shifts each k row by 4 banks
+// row stride 272 B + k-group XOR: at any 16-B-multiple stride the scalar
+// transposed fragment reads put all four (lane>>4) k-groups (8 k apart,
+// 8*stride = 0 mod 32 dwords) on the SAME 8 banks -> 4-way conflict
+// (SQ_LDS_BANK_CONFLICT = 70% of ACTIVE_INST). XORing the column byte
+// with ((k>>3)&3)<<5 is a bijection inside the 256-B row, keeps the 16-B
+// write alignment (bits 5-6 only), and maps the four k-groups onto the
+// four disjoint 8-dword bank blocks -> full 32-bank coverage.
 #define W2_ROWB 272
+#define W2_ADDR(k, colbyte) ((k)*W2_ROWB + ((colbyte) ^ ((((k) >> 3) & 3) << 5)))
 #define A_LDS(i) (smem + (i)*35840)            // [k 64][m 128] bf16 padded
 #define B_LDS(i) (smem + 17408 + (i)*35840)    // [k 64][c 128] bf16 padded
   const int m0 = blockIdx.x * WBM;
@@ -103,7 +108,7 @@ __global__ __launch_bounds__(256) void wgrad2_kernel(
     const int c8b = (tid & 15) * 16;
 #pragma unroll
     for (int p = 0; p < 4; ++p)
-      *reinterpret_cast<uint4v*>(lds + (kr + 16 * p) * W2_ROWB + c8b) = regs[p];
+      *reinterpret_cast<uint4v*>(lds + W2_ADDR(kr + 16 * p, c8b)) = regs[p];
   };
 
   uint4v a_regs[4], b_regs[4];
@@ -131,9 +136,9 @@ __global__ __launch_bounds__(256) void wgrad2_kernel(
         for (int j = 0; j < 8; ++j) {
           const int k = ks * 32 + (lane >> 4) * 8 + j;
           a_frag[f][j] = *reinterpret_cast<const __bf16*>(
-              A_LDS(cur) + k * W2_ROWB + mloc * 2);
+              A_LDS(cur) + W2_ADDR(k, mloc * 2));
           b_frag[f][j] = *reinterpret_cast<const __bf16*>(
-              B_LDS(cur) + k * W2_ROWB + cloc * 2);
+              B_LDS(cur) + W2_ADDR(k, cloc * 2));
         }
       }
       __builtin_amdgcn_s_setprio(1);
