@@ -55,6 +55,16 @@ __device__ __forceinline__ int fa_swz(int row, int byte) {
   return row * 128 + (byte ^ ((row & 7) << 4));
 }
 
+// Swizzle for the TRANSPOSED ([d][key]) images. Their scalar staging
+// writes have d = (lane&7)*8 + j, i.e. d&7 is constant per instruction —
+// with the fa_swz map all 64 lanes land in a 4-dword window (16-way
+// conflict). XOR on d>>3 instead: the staging writes then spread over all
+// 32 banks, and the b128 fragment reads (d = fd*16 + (lane&15)) stay at
+// their previous ~2-way worst case.
+__device__ __forceinline__ int fa_swzT(int row, int byte) {
+  return row * 128 + (byte ^ (((row >> 3) & 7) << 4));
+}
+
 __global__ __launch_bounds__(256) void flash_fwd_kernel(
     const bf16* __restrict__ Q, const bf16* __restrict__ K,
     const bf16* __restrict__ V, const int* __restrict__ valid,
@@ -130,7 +140,7 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           *reinterpret_cast<bf16*>(
-              v_lds + fa_swz(d0 + j, (row + rr) * 2)) = vv[j];
+              v_lds + fa_swzT(d0 + j, (row + rr) * 2)) = vv[j];
         }
       }
     }
@@ -256,7 +266,7 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
       for (int fd = 0; fd < 4; ++fd) {
         const int d = fd * 16 + (lane & 15);
         const int keybyte = kp * 64 + (lane >> 4) * 16;
-        v_frag[fd] = *reinterpret_cast<const bf16x8*>(v_lds + fa_swz(d, keybyte));
+        v_frag[fd] = *reinterpret_cast<const bf16x8*>(v_lds + fa_swzT(d, keybyte));
       }
 #pragma unroll
       for (int fq = 0; fq < 2; ++fq)
@@ -459,7 +469,7 @@ __global__ __launch_bounds__(256) void flash_dq_kernel(
         const int d0 = off / 2;
 #pragma unroll
         for (int j = 0; j < 8; ++j)
-          *reinterpret_cast<bf16*>(kt_lds + fa_swz(d0 + j, (row + rr) * 2)) = kk[j];
+          *reinterpret_cast<bf16*>(kt_lds + fa_swzT(d0 + j, (row + rr) * 2)) = kk[j];
       }
     }
     __syncthreads();
@@ -527,7 +537,7 @@ __global__ __launch_bounds__(256) void flash_dq_kernel(
       for (int fd = 0; fd < 4; ++fd) {
         const int d = fd * 16 + (lane & 15);
         const int keybyte = kp * 64 + (lane >> 4) * 16;
-        kt_frag[fd] = *reinterpret_cast<const bf16x8*>(kt_lds + fa_swz(d, keybyte));
+        kt_frag[fd] = *reinterpret_cast<const bf16x8*>(kt_lds + fa_swzT(d, keybyte));
       }
 #pragma unroll
       for (int fq = 0; fq < 2; ++fq)
@@ -636,9 +646,12 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
   const long HD = (long)H * 64;
   const int vl = valid ? valid[b] : L;
   const float dscale = (p8 > 0) ? 256.0f / (256.0f - p8) : 1.0f;
-  char* pd_lds = wave_base + wid * 2048;
-  char* ds_lds = wave_base + 8192 + wid * 2048;
-  char* dot_lds = wave_base + 16384 + wid * 2 * DKV_T_BYTES;
+  // pd/ds rows padded to 72 B: at 64 B the scalar writes put all four
+  // (lane>>4) key groups (4 rows * 16 dwords = 0 mod 32) on the same 8
+  // banks; 72 B (18 dwords) staggers the groups by 8 banks each
+  char* pd_lds = wave_base + wid * 2304;
+  char* ds_lds = wave_base + 9216 + wid * 2304;
+  char* dot_lds = wave_base + 18432 + wid * 2 * DKV_T_BYTES;
   char* qt_lds = dot_lds + DKV_T_BYTES;
   if (VAR == 3 && __builtin_amdgcn_readfirstlane(threadIdx.x) >= 128)
     __builtin_amdgcn_s_setprio(1);
@@ -708,7 +721,7 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
             do_frag[fq][ks] = bf16x8{};
           }
         }
-      if (VAR == 2) {
+      if (VAR == 2 || VAR == 5) {
         // stage this wave's dO^T and Q^T slices ([d][q] padded rows)
 #pragma unroll
         for (int fq = 0; fq < 2; ++fq)
@@ -779,9 +792,9 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
               dp = keep ? dp * dscale : 0.f;
             }
             const float dsv = scale * p[fk][fq][i] * (dp - dterm_w[fq]);
-            *reinterpret_cast<bf16*>(pd_lds + key_loc * 64 + q_loc * 2) =
+            *reinterpret_cast<bf16*>(pd_lds + key_loc * 72 + q_loc * 2) =
                 __float2bfloat16(pd);
-            *reinterpret_cast<bf16*>(ds_lds + key_loc * 64 + q_loc * 2) =
+            *reinterpret_cast<bf16*>(ds_lds + key_loc * 72 + q_loc * 2) =
                 __float2bfloat16(dsv);
           }
       SEG_MARK(3)  // pd/ds compute + LDS bounce writes
@@ -792,15 +805,15 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
       for (int fk = 0; fk < 2; ++fk) {
         const int key_loc = fk * 16 + (lane & 15);
         const int qbyte = (lane >> 4) * 16;
-        pa[fk] = *reinterpret_cast<const bf16x8*>(pd_lds + key_loc * 64 + qbyte);
-        dsa[fk] = *reinterpret_cast<const bf16x8*>(ds_lds + key_loc * 64 + qbyte);
+        pa[fk] = *reinterpret_cast<const bf16x8*>(pd_lds + key_loc * 72 + qbyte);
+        dsa[fk] = *reinterpret_cast<const bf16x8*>(ds_lds + key_loc * 72 + qbyte);
       }
       // dO/Q B-fragments read straight from global in transposed order:
       // the 16-B q_frag/do_frag loads above warmed exactly these L1 lines,
       // and dropping the [d][q] LDS bounce removes ~2/3 of the kernel's
       // LDS-issue cost (PMC: WAIT_INST_ANY 55%, LDS_IDX 39% of cycles)
       bf16x8 dob[4], qb[4];
-      if (VAR == 2) {
+      if (VAR == 2 || VAR == 5) {
         // vector B-fragment reads from the padded [d][q] bounce images
         // (per-wave buffers: the compiler orders same-wave LDS write->read)
 #pragma unroll
@@ -853,7 +866,7 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
     }
   }
 
-  if (VAR == 4) {
+  if (VAR == 4 || VAR == 5) {
     // pair-wise reduce: the two waves of a key half (wq 0/1) hold the only
     // partials for those 32 keys. wq=1 vector-writes its accs to a [d][key]
     // f32 image with 272-B padded rows ((4d + key) mod 64 distinct within
@@ -965,7 +978,7 @@ void launch_flash_dkv(const bf16* Q, const bf16* K, const bf16* V,
                       unsigned p8, unsigned long long seed,
                       hipStream_t stream) {
   const dim3 grid((L + TK - 1) / TK, B * H);
-  size_t lds = 2 * TK * 128 + 16384;  // K,V + pd/ds bounces
+  size_t lds = 2 * TK * 128 + 18432;  // K,V + pd/ds bounces (72-B rows)
   const char* e = getenv("DFA_DKV_VARIANT");  // re-read: lets one probe
   const int var = e ? atoi(e) : 4;            // process A/B the variants
   if (var == 2) {
@@ -974,12 +987,18 @@ void launch_flash_dkv(const bf16* Q, const bf16* K, const bf16* V,
                        V, dO, valid, bias, lse, Dterm, dK, dV, B, H, L, scale,
                        causal, p8, seed);
   } else if (var == 4) {
-    lds = 2 * TK * 128 + 64 * 272;  // K,V + pd/ds reused as the pair image
+    // K,V + max(pd/ds region, 64x272 pair image)
+    lds = 2 * TK * 128 + 18432;
     hipLaunchKernelGGL(flash_dkv_kernel<4>, grid, dim3(256), lds, stream, Q, K,
                        V, dO, valid, bias, lse, Dterm, dK, dV, B, H, L, scale,
                        causal, p8, seed);
   } else if (var == 3) {
     hipLaunchKernelGGL(flash_dkv_kernel<3>, grid, dim3(256), lds, stream, Q, K,
+                       V, dO, valid, bias, lse, Dterm, dK, dV, B, H, L, scale,
+                       causal, p8, seed);
+  } else if (var == 5) {
+    lds += 4 * 2 * DKV_T_BYTES;  // bounce staging + pair-reduce epilogue
+    hipLaunchKernelGGL(flash_dkv_kernel<5>, grid, dim3(256), lds, stream, Q, K,
                        V, dO, valid, bias, lse, Dterm, dK, dV, B, H, L, scale,
                        causal, p8, seed);
   } else if (var == 9) {
