@@ -38,6 +38,7 @@ valid = torch.full((B,), L, dtype=torch.int32, device=dev)
 O, lse = ext.flash_attn_fwd(q, k, v, H, valid, None, 0.125, False, 0.0, 0)
 Dt = torch.empty(B, H, L, device=dev)
 xff = torch.randn(N, FF, device=dev, dtype=bf)
+_, lnm, lnr = ext.layernorm_fwd(x, b32, b32, 1e-5)
 
 wf32 = w.float()
 xf32 = x.float()
@@ -57,6 +58,9 @@ variants = {
     "bias_gelu fwd (8192x3072)": lambda: ext.bias_gelu_fwd(xff, bff),
     "bias_gelu bwd": lambda: ext.bias_gelu_bwd(dyff, xff, bff),
     "colsum (8192x768)": lambda: ext.colsum(dy),
+    "layernorm fwd (8192x768)": lambda: ext.layernorm_fwd(x, b32, b32, 1e-5),
+    "layernorm bwd": lambda: ext.layernorm_bwd(dy, x, b32, lnm, lnr),
+    "layernorm wgrad": lambda: ext.layernorm_wgrad(dy, x, lnm, lnr),
     "copy 50MB (roofline ref)": lambda: xff.clone(),
 }
 
