@@ -82,6 +82,8 @@ template <typename T>
 void launch_softmax_mask_bwd(const T*, const T*, T*, long, int, float, float,
                              unsigned long long, hipStream_t);
 void dkv_prof_fetch(unsigned long long*);
+template <typename T>
+void launch_embed_scatter(const T*, const long*, float*, long, int, long, hipStream_t);
 
 #define CHECK_GPU(t) \
   TORCH_CHECK((t).is_cuda() && (t).is_contiguous(), #t " must be contiguous GPU tensor")
@@ -756,6 +758,21 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_wgrad", &rmsnorm_wgrad);
   m.def("flash_attn_fwd", &flash_attn_fwd);
   m.def("flash_attn_bwd", &flash_attn_bwd);
+  m.def("embed_scatter", [](at::Tensor dY, at::Tensor idx, long num_rows, long padding_idx) {
+    CHECK_GPU(dY);
+    TORCH_CHECK(idx.scalar_type() == at::kLong && idx.is_contiguous());
+    const int D = dY.size(-1);
+    const long N = dY.numel() / D;
+    auto dW = at::zeros({num_rows, (long)D},
+                        dY.options().dtype(at::kFloat));
+    dispatch_float_bf16(dY, "embed_scatter", [&](auto tag) {
+      using T = decltype(tag);
+      launch_embed_scatter<T>(ptr<T>(dY), idx.data_ptr<long>(),
+                              dW.data_ptr<float>(), N, D, padding_idx,
+                              cur_stream());
+    });
+    return dW;
+  });
   m.def("dkv_prof", []() {
     // variant-9 instrumentation readout: per-segment cycle totals
     auto t = torch::zeros({8}, torch::dtype(torch::kLong));

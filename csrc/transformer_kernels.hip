@@ -740,3 +740,48 @@ void launch_rmsnorm_wgrad(const T* dy, const T* x, const float* rstd,
 
 INSTANTIATE_TK(float)
 INSTANTIATE_TK(__hip_bfloat16)
+
+// ---------------------------------------------------------------------------
+// Embedding backward: dW[idx[n]] += dY[n] (fp32 accum), padding_idx skipped.
+// Replaces torch's sort + segment-reduce + scatter stack (~0.5 ms/step on
+// the LineVul step for 3 embeddings).
+// ---------------------------------------------------------------------------
+
+template <typename T>
+__global__ void embed_scatter_kernel(const T* __restrict__ dY,
+                                     const long* __restrict__ idx,
+                                     float* __restrict__ dW, long N, int D,
+                                     long padding_idx) {
+  const long row = (long)(blockIdx.x * (blockDim.x / WAVE)) + threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  if (row >= N) return;
+  const long i = idx[row];
+  if (i == padding_idx) return;
+  const T* src = dY + row * D;
+  float* dst = dW + i * D;
+  if (D % (WAVE * 4) == 0) {
+    for (int d = lane * 4; d < D; d += WAVE * 4) {
+      const f4 a = load4(src + d);
+#pragma unroll
+      for (int u = 0; u < 4; ++u) atomicAdd(dst + d + u, a.v[u]);
+    }
+  } else {
+    for (int d = lane; d < D; d += WAVE) atomicAdd(dst + d, tf(src[d]));
+  }
+}
+
+template <typename T>
+void launch_embed_scatter(const T* dY, const long* idx, float* dW, long N,
+                          int D, long padding_idx, hipStream_t stream) {
+  const int grid = (int)((N + ROWS_PER_BLOCK - 1) / ROWS_PER_BLOCK);
+  if (grid)
+    hipLaunchKernelGGL(embed_scatter_kernel<T>, dim3(grid),
+                       dim3(WAVE * ROWS_PER_BLOCK), 0, stream, dY, idx, dW, N,
+                       D, padding_idx);
+}
+
+template void launch_embed_scatter<float>(const float*, const long*, float*,
+                                          long, int, long, hipStream_t);
+template void launch_embed_scatter<__hip_bfloat16>(const __hip_bfloat16*,
+                                                   const long*, float*, long,
+                                                   int, long, hipStream_t);

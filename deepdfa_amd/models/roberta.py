@@ -81,12 +81,17 @@ class RobertaEmbeddings(nn.Module):
         self.padding_idx = cfg.pad_token_id
 
     def forward(self, input_ids):
+        from ..ops.transformer import embedding_lookup
+
         mask = input_ids.ne(self.padding_idx).long()
         position_ids = torch.cumsum(mask, dim=1) * mask + self.padding_idx
+        # token type is always 0 for this model family: broadcasting row 0
+        # gives the identical result with a cheap sum-reduce gradient
+        # instead of an all-rows-collide scatter
         emb = (
-            self.word_embeddings(input_ids)
-            + self.position_embeddings(position_ids)
-            + self.token_type_embeddings(torch.zeros_like(input_ids))
+            embedding_lookup(input_ids, self.word_embeddings.weight, self.padding_idx)
+            + embedding_lookup(position_ids, self.position_embeddings.weight, self.padding_idx)
+            + self.token_type_embeddings.weight[0]
         )
         return self.dropout(self.LayerNorm(emb))
 

@@ -218,7 +218,12 @@ class T5Stack(nn.Module):
         self.dropout = nn.Dropout(cfg.dropout_rate)
 
     def forward(self, input_ids, valid, enc=None, enc_valid=None):
-        x = self.dropout(self.embed_tokens(input_ids))
+        from ..ops.transformer import embedding_lookup
+
+        x = self.dropout(
+            embedding_lookup(input_ids, self.embed_tokens.weight,
+                             self.embed_tokens.padding_idx)
+        )
         if x.is_cuda and torch.is_autocast_enabled():
             x = x.to(torch.bfloat16)
         L = input_ids.shape[1]

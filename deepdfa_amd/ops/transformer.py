@@ -267,3 +267,31 @@ def masked_softmax(S: torch.Tensor, valid: Optional[torch.Tensor], scale: float)
     # index 1 (the post-dropout output) is the differentiable one; with
     # dropout 0 it aliases the probabilities
     return masked_softmax_dropout(S, valid, scale, 0.0)[1]
+
+
+class _EmbeddingLookup(torch.autograd.Function):
+    """nn.Embedding fwd (torch gather) with a custom scatter-add backward
+    (csrc/transformer_kernels.hip embed_scatter): replaces torch's
+    radix-sort + segment-reduce + scatter stack in the grad path."""
+
+    @staticmethod
+    def forward(ctx, indices, weight, padding_idx):
+        ctx.save_for_backward(indices)
+        ctx.num_rows = weight.shape[0]
+        ctx.w_dtype = weight.dtype
+        ctx.padding_idx = padding_idx if padding_idx is not None else -1
+        return torch.nn.functional.embedding(indices, weight, padding_idx)
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = load_ext(required=True)
+        (indices,) = ctx.saved_tensors
+        dw = ext.embed_scatter(dy.contiguous(), indices.reshape(-1).contiguous(),
+                               ctx.num_rows, ctx.padding_idx)
+        return None, dw.to(ctx.w_dtype), None
+
+
+def embedding_lookup(indices, weight, padding_idx=None):
+    if indices.is_cuda:
+        return _EmbeddingLookup.apply(indices, weight, padding_idx)
+    return torch.nn.functional.embedding(indices, weight, padding_idx)
