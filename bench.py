@@ -156,7 +156,9 @@ def build_linevul(args, rank, device, use_cuda, with_ddfa: bool):
             "parallelism": f"dp{world_size()}",
         },
     )
-    return model, opt_fn, make_step, meta, False
+    # capture-safe: fixed shapes, FlatAdamW with device bias correction,
+    # capture-refresh weight casts (see ops.transformer.CAPTURE_REFRESH)
+    return model, opt_fn, make_step, meta, True
 
 
 def build_codet5(args, rank, device, use_cuda, with_ddfa: bool):
@@ -281,6 +283,9 @@ def main():
         step(i)
 
     if use_graphs:
+        from deepdfa_amd.ops import transformer as _tops
+
+        _tops.CAPTURE_REFRESH[0] = True  # weight casts re-run inside the graph
         torch.cuda.synchronize()
         pool = torch.cuda.graph_pool_handle()
         graphs = []

@@ -71,7 +71,7 @@ void launch_flash_dq(const __hip_bfloat16*, const __hip_bfloat16*, const __hip_b
 void launch_wgrad2(const __hip_bfloat16*, const __hip_bfloat16*, float*, int, int, int,
                    hipStream_t);
 void launch_adamw_fused(float*, const float*, float*, float*, long, float, float, float, float,
-                        float, int, int, hipStream_t);
+                        float, const float*, int, hipStream_t);
 void launch_gemm2(const __hip_bfloat16*, const __hip_bfloat16*, const float*,
                   const __hip_bfloat16*, __hip_bfloat16*, int, int, int, hipStream_t);
 void launch_flash_dkv(const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*,
@@ -722,14 +722,17 @@ at::Tensor gemm2(at::Tensor A, at::Tensor W, c10::optional<at::Tensor> bias,
 
 
 void adamw_fused(at::Tensor p, at::Tensor g, at::Tensor m, at::Tensor v, double lr,
-                 double beta1, double beta2, double eps, double weight_decay, int64_t step,
+                 double beta1, double beta2, double eps, double weight_decay, at::Tensor bc,
                  bool l2_mode) {
   CHECK_GPU(p);
   TORCH_CHECK(p.scalar_type() == at::kFloat && g.scalar_type() == at::kFloat);
   TORCH_CHECK(p.numel() == g.numel() && p.numel() == m.numel() && p.numel() == v.numel());
+  TORCH_CHECK(bc.is_cuda() && bc.scalar_type() == at::kFloat && bc.numel() == 2,
+              "bc must be a device float[2] = {1-b1^t, 1-b2^t}");
   launch_adamw_fused(p.data_ptr<float>(), g.data_ptr<float>(), m.data_ptr<float>(),
                      v.data_ptr<float>(), p.numel(), (float)lr, (float)beta1, (float)beta2,
-                     (float)eps, (float)weight_decay, (int)step, l2_mode ? 1 : 0, cur_stream());
+                     (float)eps, (float)weight_decay, bc.data_ptr<float>(), l2_mode ? 1 : 0,
+                     cur_stream());
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {

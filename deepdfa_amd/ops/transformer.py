@@ -68,6 +68,21 @@ def bias_gelu(x, bias):
     return torch.nn.functional.gelu(x.float() + bias.float()).to(x.dtype)
 
 
+# cast-cache invalidation epoch: optimizers that write master weights
+# through raw extension kernels (FlatAdamW) bypass torch's version
+# counters, so they bump this instead
+_weights_epoch = [0]
+
+# hipGraph capture mode: when True, fused_linear re-casts weights INTO its
+# cached buffers every call so the cast is captured and replays against the
+# updated master weights
+CAPTURE_REFRESH = [False]
+
+
+def bump_weights_epoch():
+    _weights_epoch[0] += 1
+
+
 _seed_state = {"base": None, "ctr": 0}
 
 
@@ -144,11 +159,25 @@ class _LinearBf16(torch.autograd.Function):
         # and transpose copies otherwise dominate the elementwise kernel
         # count (~3 kernels x 72 linears per step)
         cache = getattr(weight, "_dfa_cast_cache", None)
-        if cache is None or cache[0] != weight._version:
+        key = (weight._version, _weights_epoch[0])
+        if cache is None:
             w16 = weight.detach().to(torch.bfloat16).contiguous()
             wt16 = w16.t().contiguous()
             b32 = bias.detach().float().contiguous() if bias is not None else None
-            weight._dfa_cast_cache = (weight._version, w16, wt16, b32)
+            weight._dfa_cast_cache = (key, w16, wt16, b32)
+        elif CAPTURE_REFRESH[0] or cache[0] != key:
+            # refresh INTO the cached buffers: under hipGraph capture the
+            # cast must be part of the graph (stable addresses, re-run each
+            # replay); in eager it re-casts only when the key moved
+            _, w16, wt16, b32 = cache
+            w16.copy_(weight.detach())
+            wt16.copy_(w16.t())
+            if bias is not None:
+                if b32 is None:
+                    b32 = bias.detach().float().contiguous()
+                else:
+                    b32.copy_(bias.detach())
+            weight._dfa_cast_cache = (key, w16, wt16, b32)
         else:
             _, w16, wt16, b32 = cache
             if bias is not None and b32 is None:

@@ -59,10 +59,10 @@ class FlatAdamW:
         return torch.linalg.vector_norm(self.flat_g)
 
     def clip_grad_norm_(self, max_norm: float) -> torch.Tensor:
+        # capture-safe: no host readback — clamp the coefficient on-device
         norm = self.grad_norm()
-        scale = max_norm / (norm + 1e-6)
-        if float(scale) < 1.0:
-            self.flat_g.mul_(scale)
+        coef = torch.clamp(max_norm / (norm + 1e-6), max=1.0)
+        self.flat_g.mul_(coef)
         return norm
 
     def allreduce_grads(self):
@@ -79,12 +79,28 @@ class FlatAdamW:
         b1, b2 = self.betas
         if self.flat_p.is_cuda:
             from ..ops import load_ext
+            from ..ops.transformer import bump_weights_epoch
 
             ext = load_ext(required=True)
+            # bias corrections computed ON DEVICE from a device step counter:
+            # correct under hipGraph replay (the captured add_/pow advance
+            # every replay; a python int would freeze at capture time)
+            if not hasattr(self, "_step_t") or not self._step_t.is_cuda:
+                self._step_t = torch.zeros((), device=self.flat_p.device)
+                self._step_t.fill_(float(self.step_count - 1))
+                self._betas_t = torch.tensor([b1, b2], device=self.flat_p.device)
+                self._bc = torch.empty(2, device=self.flat_p.device)
+            self._step_t.add_(1.0)
+            torch.pow(self._betas_t, self._step_t, out=self._bc)
+            torch.neg(self._bc, out=self._bc)
+            self._bc.add_(1.0)
             ext.adamw_fused(
                 self.flat_p, self.flat_g, self.m, self.v, lr, b1, b2, self.eps,
-                self.weight_decay, self.step_count, self.l2_mode,
+                self.weight_decay, self._bc, self.l2_mode,
             )
+            # the raw kernel write bypasses torch version counters —
+            # invalidate the bf16 weight-cast caches explicitly
+            bump_weights_epoch()
             return
         # CPU reference path (same math)
         g = self.flat_g
