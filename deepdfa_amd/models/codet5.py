@@ -45,11 +45,16 @@ class DefectModel(nn.Module):
             output_hidden_only=True,
         )
         eos_mask = source_ids.eq(self.config.eos_token_id)
-        if len(torch.unique(eos_mask.sum(1))) > 1:
+        if not source_ids.is_cuda and len(torch.unique(eos_mask.sum(1))) > 1:
             raise ValueError("All examples must have the same number of <eos> tokens.")
-        b, _, d = hidden.shape
-        vec = hidden[eos_mask, :].view(b, -1, d)[:, -1, :]
-        return vec
+        # last-EOS gather with fixed shapes (the reference's boolean-index
+        # form forces a host sync + data-dependent shape every forward;
+        # equal-eos-count validation stays on the CPU path)
+        b, L, d = hidden.shape
+        idx = torch.where(
+            eos_mask, torch.arange(L, device=source_ids.device), -1
+        ).max(dim=1).values.clamp_(min=0)
+        return hidden.gather(1, idx.view(b, 1, 1).expand(b, 1, d)).squeeze(1)
 
     def forward(self, source_ids, labels=None, graphs=None):
         vec = self.get_t5_vec(source_ids)
@@ -103,10 +108,13 @@ class CloneModel(nn.Module):
         hidden = self.encoder(source_ids, attention_mask=attention_mask, labels=source_ids,
                               output_hidden_only=True)
         eos_mask = source_ids.eq(self.config.eos_token_id)
-        if len(torch.unique(eos_mask.sum(1))) > 1:
+        if not source_ids.is_cuda and len(torch.unique(eos_mask.sum(1))) > 1:
             raise ValueError("All examples must have the same number of <eos> tokens.")
-        b, _, d = hidden.shape
-        return hidden[eos_mask, :].view(b, -1, d)[:, -1, :]
+        b, L, d = hidden.shape
+        idx = torch.where(
+            eos_mask, torch.arange(L, device=source_ids.device), -1
+        ).max(dim=1).values.clamp_(min=0)
+        return hidden.gather(1, idx.view(b, 1, 1).expand(b, 1, d)).squeeze(1)
 
     def forward(self, source_ids, labels=None):
         source_ids = source_ids.view(-1, self.max_source_length)
