@@ -83,6 +83,17 @@ void launch_softmax_mask_bwd(const T*, const T*, T*, long, int, float, float,
                              unsigned long long, hipStream_t);
 void dkv_prof_fetch(unsigned long long*);
 template <typename T>
+void launch_ln_res_dropout_fwd(const T*, const T*, const float*, const float*, T*, float*,
+                               float*, long, int, float, float, unsigned long long, hipStream_t);
+template <typename T>
+void launch_ln_res_dropout_bwd(const T*, const T*, const T*, const float*, const float*,
+                               const float*, T*, T*, long, int, float, unsigned long long,
+                               hipStream_t);
+template <typename T>
+void launch_ln_res_dropout_wgrad(const T*, const T*, const T*, const float*, const float*,
+                                 float*, float*, long, int, float, unsigned long long,
+                                 hipStream_t);
+template <typename T>
 void launch_embed_scatter(const T*, const long*, float*, long, int, long, hipStream_t);
 
 #define CHECK_GPU(t) \
@@ -761,6 +772,60 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_wgrad", &rmsnorm_wgrad);
   m.def("flash_attn_fwd", &flash_attn_fwd);
   m.def("flash_attn_bwd", &flash_attn_bwd);
+  m.def("ln_res_dropout_fwd", [](at::Tensor h, at::Tensor res, at::Tensor gamma,
+                                 at::Tensor beta, double eps, double p, int64_t seed) {
+    CHECK_GPU(h);
+    CHECK_GPU(res);
+    const int D = h.size(-1);
+    TORCH_CHECK(D % 256 == 0, "ln_res_dropout: D must be a multiple of 256");
+    const long N = h.numel() / D;
+    auto y = at::empty_like(h);
+    auto mean = at::empty({N}, h.options().dtype(at::kFloat));
+    auto rstd = at::empty({N}, h.options().dtype(at::kFloat));
+    dispatch_float_bf16(h, "ln_res_dropout_fwd", [&](auto tag) {
+      using T = decltype(tag);
+      launch_ln_res_dropout_fwd<T>(ptr<T>(h), ptr<T>(res), gamma.data_ptr<float>(),
+                                   beta.data_ptr<float>(), mptr<T>(y),
+                                   mean.data_ptr<float>(), rstd.data_ptr<float>(), N, D,
+                                   (float)eps, (float)p, (unsigned long long)seed,
+                                   cur_stream());
+    });
+    return std::vector<at::Tensor>{y, mean, rstd};
+  });
+  m.def("ln_res_dropout_bwd", [](at::Tensor dy, at::Tensor h, at::Tensor res,
+                                 at::Tensor gamma, at::Tensor mean, at::Tensor rstd,
+                                 double p, int64_t seed) {
+    CHECK_GPU(dy);
+    const int D = dy.size(-1);
+    const long N = dy.numel() / D;
+    auto dz = at::empty_like(dy);
+    auto dh = at::empty_like(dy);
+    dispatch_float_bf16(dy, "ln_res_dropout_bwd", [&](auto tag) {
+      using T = decltype(tag);
+      launch_ln_res_dropout_bwd<T>(ptr<T>(dy), ptr<T>(h), ptr<T>(res),
+                                   gamma.data_ptr<float>(), mean.data_ptr<float>(),
+                                   rstd.data_ptr<float>(), mptr<T>(dz), mptr<T>(dh), N, D,
+                                   (float)p, (unsigned long long)seed, cur_stream());
+    });
+    return std::vector<at::Tensor>{dz, dh};
+  });
+  m.def("ln_res_dropout_wgrad", [](at::Tensor dy, at::Tensor h, at::Tensor res,
+                                   at::Tensor mean, at::Tensor rstd, double p,
+                                   int64_t seed) {
+    CHECK_GPU(dy);
+    const int D = dy.size(-1);
+    const long N = dy.numel() / D;
+    auto dgamma = at::zeros({(long)D}, dy.options().dtype(at::kFloat));
+    auto dbeta = at::zeros({(long)D}, dy.options().dtype(at::kFloat));
+    dispatch_float_bf16(dy, "ln_res_dropout_wgrad", [&](auto tag) {
+      using T = decltype(tag);
+      launch_ln_res_dropout_wgrad<T>(ptr<T>(dy), ptr<T>(h), ptr<T>(res),
+                                     mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                                     dgamma.data_ptr<float>(), dbeta.data_ptr<float>(), N,
+                                     D, (float)p, (unsigned long long)seed, cur_stream());
+    });
+    return std::vector<at::Tensor>{dgamma, dbeta};
+  });
   m.def("embed_scatter", [](at::Tensor dY, at::Tensor idx, long num_rows, long padding_idx) {
     CHECK_GPU(dY);
     TORCH_CHECK(idx.scalar_type() == at::kLong && idx.is_contiguous());

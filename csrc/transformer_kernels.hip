@@ -785,3 +785,213 @@ template void launch_embed_scatter<float>(const float*, const long*, float*,
 template void launch_embed_scatter<__hip_bfloat16>(const __hip_bfloat16*,
                                                    const long*, float*, long,
                                                    int, long, hipStream_t);
+
+// ---------------------------------------------------------------------------
+// Fused y = LayerNorm(dropout(h) + residual): the transformer residual
+// pattern (RobertaSelfOutput/RobertaOutput) as ONE kernel per direction.
+// Stateless dropout (hash4 mask, regenerated in backward) — nothing but
+// (h, res) and the seed is stored; z = dropout(h)+res is recomputed.
+// Requires D % 256 == 0 (vector path only; the op wrapper falls back).
+// ---------------------------------------------------------------------------
+
+template <typename T>
+__global__ void ln_res_dropout_fwd_kernel(
+    const T* __restrict__ h, const T* __restrict__ res,
+    const float* __restrict__ gamma, const float* __restrict__ beta,
+    T* __restrict__ y, float* __restrict__ mean, float* __restrict__ rstd,
+    long N, int D, float eps, unsigned p8, unsigned long long seed,
+    float dscale) {
+  const long row = (long)(blockIdx.x * (blockDim.x / WAVE)) + threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  if (row >= N) return;
+  const T* hr = h + row * D;
+  const T* rr = res + row * D;
+  float s = 0.f, s2 = 0.f;
+  for (int d = lane * 4; d < D; d += WAVE * 4) {
+    const f4 a = load4(hr + d);
+    const f4 b = load4(rr + d);
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      const bool keep = (p8 == 0) || keep_mask(row * D + d + u, seed, p8);
+      const float z = (keep ? a.v[u] * dscale : 0.f) + b.v[u];
+      s += z;
+      s2 += z * z;
+    }
+  }
+  s = wave_sum(s);
+  s2 = wave_sum(s2);
+  const float mu = s / D;
+  const float rs = rsqrtf(s2 / D - mu * mu + eps);
+  if (lane == 0) {
+    mean[row] = mu;
+    rstd[row] = rs;
+  }
+  T* yr = y + row * D;
+  for (int d = lane * 4; d < D; d += WAVE * 4) {
+    const f4 a = load4(hr + d);
+    const f4 b = load4(rr + d);
+    const f4 g = load4(gamma + d);
+    const f4 bb = load4(beta + d);
+    f4 o;
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      const bool keep = (p8 == 0) || keep_mask(row * D + d + u, seed, p8);
+      const float z = (keep ? a.v[u] * dscale : 0.f) + b.v[u];
+      o.v[u] = (z - mu) * rs * g.v[u] + bb.v[u];
+    }
+    store4(yr + d, o);
+  }
+}
+
+template <typename T>
+__global__ void ln_res_dropout_bwd_kernel(
+    const T* __restrict__ dy, const T* __restrict__ h,
+    const T* __restrict__ res, const float* __restrict__ gamma,
+    const float* __restrict__ mean, const float* __restrict__ rstd,
+    T* __restrict__ dz, T* __restrict__ dh, long N, int D, unsigned p8,
+    unsigned long long seed, float dscale) {
+  const long row = (long)(blockIdx.x * (blockDim.x / WAVE)) + threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  if (row >= N) return;
+  const T* dyr = dy + row * D;
+  const T* hr = h + row * D;
+  const T* rr = res + row * D;
+  const float mu = mean[row], rs = rstd[row];
+  float c1 = 0.f, c2 = 0.f;
+  for (int d = lane * 4; d < D; d += WAVE * 4) {
+    const f4 a = load4(hr + d);
+    const f4 b = load4(rr + d);
+    const f4 dyv = load4(dyr + d);
+    const f4 g4 = load4(gamma + d);
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      const bool keep = (p8 == 0) || keep_mask(row * D + d + u, seed, p8);
+      const float z = (keep ? a.v[u] * dscale : 0.f) + b.v[u];
+      const float g = dyv.v[u] * g4.v[u];
+      c1 += g;
+      c2 += g * (z - mu) * rs;
+    }
+  }
+  c1 = wave_sum(c1) / D;
+  c2 = wave_sum(c2) / D;
+  T* dzr = dz + row * D;
+  T* dhr = dh + row * D;
+  for (int d = lane * 4; d < D; d += WAVE * 4) {
+    const f4 a = load4(hr + d);
+    const f4 b = load4(rr + d);
+    const f4 dyv = load4(dyr + d);
+    const f4 g4 = load4(gamma + d);
+    f4 oz, oh;
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      const bool keep = (p8 == 0) || keep_mask(row * D + d + u, seed, p8);
+      const float z = (keep ? a.v[u] * dscale : 0.f) + b.v[u];
+      const float xh = (z - mu) * rs;
+      const float dzv = rs * (dyv.v[u] * g4.v[u] - c1 - xh * c2);
+      oz.v[u] = dzv;
+      oh.v[u] = keep ? dzv * dscale : 0.f;
+    }
+    store4(dzr + d, oz);
+    store4(dhr + d, oh);
+  }
+}
+
+template <typename T>
+__global__ void ln_res_dropout_wgrad_kernel(
+    const T* __restrict__ dy, const T* __restrict__ h,
+    const T* __restrict__ res, const float* __restrict__ mean,
+    const float* __restrict__ rstd, float* __restrict__ dgamma,
+    float* __restrict__ dbeta, long N, int D, int rows_per_block, unsigned p8,
+    unsigned long long seed, float dscale) {
+  const int d = blockIdx.x * blockDim.x + threadIdx.x;
+  if (d >= D) return;
+  const long r0 = (long)blockIdx.y * rows_per_block;
+  const long r1 = min((long)(r0 + rows_per_block), N);
+  float dg[4] = {}, db[4] = {};
+  long r = r0;
+  for (; r + 4 <= r1; r += 4) {
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      const long rr_ = r + u;
+      const float g = tf(dy[rr_ * D + d]);
+      const bool keep = (p8 == 0) || keep_mask(rr_ * D + d, seed, p8);
+      const float z = (keep ? tf(h[rr_ * D + d]) * dscale : 0.f) + tf(res[rr_ * D + d]);
+      dg[u] += g * (z - mean[rr_]) * rstd[rr_];
+      db[u] += g;
+    }
+  }
+  for (; r < r1; ++r) {
+    const float g = tf(dy[r * D + d]);
+    const bool keep = (p8 == 0) || keep_mask(r * D + d, seed, p8);
+    const float z = (keep ? tf(h[r * D + d]) * dscale : 0.f) + tf(res[r * D + d]);
+    dg[0] += g * (z - mean[r]) * rstd[r];
+    db[0] += g;
+  }
+  atomicAdd(dgamma + d, dg[0] + dg[1] + dg[2] + dg[3]);
+  atomicAdd(dbeta + d, db[0] + db[1] + db[2] + db[3]);
+}
+
+template <typename T>
+void launch_ln_res_dropout_fwd(const T* h, const T* res, const float* gamma,
+                               const float* beta, T* y, float* mean,
+                               float* rstd, long N, int D, float eps,
+                               float dropout_p, unsigned long long seed,
+                               hipStream_t stream) {
+  const unsigned p8 = (unsigned)(dropout_p * 256.0f);
+  const float dscale = (p8 > 0) ? 256.0f / (256.0f - p8) : 1.0f;
+  const int grid = (int)((N + ROWS_PER_BLOCK - 1) / ROWS_PER_BLOCK);
+  if (grid)
+    hipLaunchKernelGGL(ln_res_dropout_fwd_kernel<T>, dim3(grid),
+                       dim3(WAVE * ROWS_PER_BLOCK), 0, stream, h, res, gamma,
+                       beta, y, mean, rstd, N, D, eps, p8, seed, dscale);
+}
+
+template <typename T>
+void launch_ln_res_dropout_bwd(const T* dy, const T* h, const T* res,
+                               const float* gamma, const float* mean,
+                               const float* rstd, T* dz, T* dh, long N, int D,
+                               float dropout_p, unsigned long long seed,
+                               hipStream_t stream) {
+  const unsigned p8 = (unsigned)(dropout_p * 256.0f);
+  const float dscale = (p8 > 0) ? 256.0f / (256.0f - p8) : 1.0f;
+  const int grid = (int)((N + ROWS_PER_BLOCK - 1) / ROWS_PER_BLOCK);
+  if (grid)
+    hipLaunchKernelGGL(ln_res_dropout_bwd_kernel<T>, dim3(grid),
+                       dim3(WAVE * ROWS_PER_BLOCK), 0, stream, dy, h, res,
+                       gamma, mean, rstd, dz, dh, N, D, p8, seed, dscale);
+}
+
+template <typename T>
+void launch_ln_res_dropout_wgrad(const T* dy, const T* h, const T* res,
+                                 const float* mean, const float* rstd,
+                                 float* dgamma, float* dbeta, long N, int D,
+                                 float dropout_p, unsigned long long seed,
+                                 hipStream_t stream) {
+  const unsigned p8 = (unsigned)(dropout_p * 256.0f);
+  const float dscale = (p8 > 0) ? 256.0f / (256.0f - p8) : 1.0f;
+  const int block = 256;
+  const int colb = (D + block - 1) / block;
+  const int rows_per_block = 64;
+  const int rowb = (int)((N + rows_per_block - 1) / rows_per_block);
+  if (colb && rowb)
+    hipLaunchKernelGGL(ln_res_dropout_wgrad_kernel<T>, dim3(colb, rowb),
+                       dim3(block), 0, stream, dy, h, res, mean, rstd, dgamma,
+                       dbeta, N, D, rows_per_block, p8, seed, dscale);
+}
+
+#define INST_LNRD(T)                                                          \
+  template void launch_ln_res_dropout_fwd<T>(const T*, const T*, const float*,\
+                                             const float*, T*, float*, float*,\
+                                             long, int, float, float,         \
+                                             unsigned long long, hipStream_t);\
+  template void launch_ln_res_dropout_bwd<T>(const T*, const T*, const T*,    \
+                                             const float*, const float*,      \
+                                             const float*, T*, T*, long, int, \
+                                             float, unsigned long long,       \
+                                             hipStream_t);                    \
+  template void launch_ln_res_dropout_wgrad<T>(                               \
+      const T*, const T*, const T*, const float*, const float*, float*,       \
+      float*, long, int, float, unsigned long long, hipStream_t);
+INST_LNRD(float)
+INST_LNRD(__hip_bfloat16)
+#undef INST_LNRD

@@ -35,6 +35,7 @@ from ..ops.transformer import (
     flash_usable,
     fused_linear,
     layer_norm,
+    layer_norm_res_dropout,
     masked_softmax_dropout,
 )
 
@@ -140,7 +141,9 @@ class RobertaSelfOutput(nn.Module):
 
     def forward(self, hidden, residual):
         h = fused_linear(hidden, self.dense.weight, self.dense.bias)
-        return self.LayerNorm(self.dropout(h) + residual)
+        p = self.dropout.p if self.training else 0.0
+        return layer_norm_res_dropout(h, residual, self.LayerNorm.weight,
+                                      self.LayerNorm.bias, p, self.LayerNorm.eps)
 
 
 class RobertaAttention(nn.Module):
@@ -173,7 +176,9 @@ class RobertaOutput(nn.Module):
 
     def forward(self, hidden, residual):
         h = fused_linear(hidden, self.dense.weight, self.dense.bias)
-        return self.LayerNorm(self.dropout(h) + residual)
+        p = self.dropout.p if self.training else 0.0
+        return layer_norm_res_dropout(h, residual, self.LayerNorm.weight,
+                                      self.LayerNorm.bias, p, self.LayerNorm.eps)
 
 
 class RobertaLayer(nn.Module):

@@ -324,3 +324,47 @@ def embedding_lookup(indices, weight, padding_idx=None):
     if indices.is_cuda:
         return _EmbeddingLookup.apply(indices, weight, padding_idx)
     return torch.nn.functional.embedding(indices, weight, padding_idx)
+
+
+class _LayerNormResDropout(torch.autograd.Function):
+    """y = LayerNorm(dropout(h) + residual) in one kernel per direction
+    (csrc/transformer_kernels.hip ln_res_dropout_*): the transformer
+    residual pattern with stateless dropout (mask regenerated from the
+    seed in backward — only (h, res, seed) are kept)."""
+
+    @staticmethod
+    def forward(ctx, h, res, weight, bias, dropout_p, eps):
+        ext = load_ext(required=True)
+        wf = weight.float().contiguous()
+        bf = bias.float().contiguous()
+        h = h.contiguous()
+        res = res.contiguous()
+        seed = _next_seed() if dropout_p > 0 else 0
+        y, mean, rstd = ext.ln_res_dropout_fwd(h, res, wf, bf, eps, dropout_p, seed)
+        ctx.save_for_backward(h, res, wf, mean, rstd)
+        ctx.meta = (dropout_p, seed)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = load_ext(required=True)
+        h, res, wf, mean, rstd = ctx.saved_tensors
+        p, seed = ctx.meta
+        dy = dy.contiguous()
+        dz, dh = ext.ln_res_dropout_bwd(dy, h, res, wf, mean, rstd, p, seed)
+        dgamma, dbeta = ext.ln_res_dropout_wgrad(dy, h, res, mean, rstd, p, seed)
+        return dh, dz, dgamma, dbeta, None, None
+
+
+def layer_norm_res_dropout(h, res, weight, bias, dropout_p=0.0, eps=1e-5):
+    """Fused LayerNorm(dropout(h) + res); falls back to composition on CPU
+    or when D is not a multiple of 256."""
+    if h.is_cuda and h.shape[-1] % 256 == 0:
+        return _LayerNormResDropout.apply(h, res, weight, bias, dropout_p, eps)
+    z = torch.nn.functional.dropout(h, dropout_p) if dropout_p > 0 else h
+    z = z + res
+    return layer_norm(z, weight, bias, eps) if z.is_cuda else (
+        torch.nn.functional.layer_norm(
+            z.float(), (z.shape[-1],), weight.float(), bias.float(), eps
+        ).to(z.dtype)
+    )

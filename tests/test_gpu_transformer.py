@@ -198,3 +198,41 @@ def test_embed_scatter_matches_torch():
     out = embedding_lookup(idx, w2, 1)
     out.backward(dy.float())
     assert torch.allclose(w2.grad, w.grad, atol=1e-3, rtol=1e-3)
+
+
+@pytest.mark.gpu
+def test_ln_res_dropout_matches_composition():
+    from deepdfa_amd.ops.transformer import layer_norm_res_dropout
+
+    torch.manual_seed(0)
+    N, D = 512, 768
+    h = (torch.randn(N, D, device="cuda", dtype=torch.bfloat16) * 0.5).requires_grad_()
+    res = torch.randn(N, D, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    w = (torch.randn(D, device="cuda") * 0.1 + 1).requires_grad_()
+    b = (torch.randn(D, device="cuda") * 0.1).requires_grad_()
+
+    # p=0: must match the unfused composition exactly (same LN math)
+    y = layer_norm_res_dropout(h, res, w, b, 0.0, 1e-5)
+    ref = torch.nn.functional.layer_norm(
+        (h.float() + res.float()), (D,), w, b, 1e-5
+    )
+    assert (y.float() - ref).abs().max() < 2e-2  # bf16 IO
+    g = torch.randn_like(y)
+    y.backward(g, retain_graph=False)
+    hg, rg, wg, bg = h.grad.clone(), res.grad.clone(), w.grad.clone(), b.grad.clone()
+    h.grad = res.grad = w.grad = b.grad = None
+    ref.backward(g.float())
+    assert (hg.float() - h.grad.float()).abs().max() < 2e-2
+    assert (rg.float() - res.grad.float()).abs().max() < 2e-2
+    assert torch.allclose(wg, w.grad, atol=0.5, rtol=0.05)
+    assert torch.allclose(bg, b.grad, atol=0.5, rtol=0.05)
+
+    # p>0: dropout statistics + gradient-mask consistency
+    h2 = h.detach().clone().requires_grad_(True)
+    res2 = res.detach().clone().requires_grad_(True)
+    y2 = layer_norm_res_dropout(h2, res2, w, b, 0.5, 1e-5)
+    assert y2.shape == (N, D) and torch.isfinite(y2.float()).all()
+    y2.sum().backward()
+    frac = (h2.grad.float() == 0).float().mean().item()
+    assert 0.4 < frac < 0.6  # ~half the h-gradient masked
+    assert (res2.grad.float() == 0).float().mean().item() < 0.05
