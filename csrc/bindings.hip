@@ -61,13 +61,13 @@ template <typename T>
 void launch_rmsnorm_wgrad(const T*, const T*, const float*, float*, long, int, hipStream_t);
 void launch_flash_fwd(const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*,
                       const int*, const float*, __hip_bfloat16*, float*, int, int, int, float,
-                      int, unsigned, unsigned long long, hipStream_t);
+                      int, unsigned, unsigned long long, long, long, hipStream_t);
 void launch_flash_dterm(const __hip_bfloat16*, const __hip_bfloat16*, float*, int, int, int,
                         hipStream_t);
 void launch_flash_dq(const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*,
                      const __hip_bfloat16*, const int*, const float*, const float*, const float*,
                      __hip_bfloat16*, float*, int, int, int, float, int, unsigned,
-                     unsigned long long, hipStream_t);
+                     unsigned long long, long, long, hipStream_t);
 void launch_wgrad2(const __hip_bfloat16*, const __hip_bfloat16*, float*, int, int, int,
                    hipStream_t);
 void launch_adamw_fused(float*, const float*, float*, float*, long, float, float, float, float,
@@ -77,7 +77,7 @@ void launch_gemm2(const __hip_bfloat16*, const __hip_bfloat16*, const float*,
 void launch_flash_dkv(const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*,
                       const __hip_bfloat16*, const int*, const float*, const float*, const float*,
                       __hip_bfloat16*, __hip_bfloat16*, int, int, int, float, int, unsigned,
-                      unsigned long long, hipStream_t);
+                      unsigned long long, long, long, hipStream_t);
 template <typename T>
 void launch_softmax_mask_bwd(const T*, const T*, T*, long, int, float, float,
                              unsigned long long, hipStream_t);
@@ -648,18 +648,27 @@ static const int* opt_valid_ptr(const c10::optional<at::Tensor>& valid) {
   return valid->data_ptr<int>();
 }
 
+// Q/K/V may be row-strided views (slices of a fused QKV projection):
+// last dim contiguous, batch stride == L * row stride
+static long fa_ld(const at::Tensor& t, int L) {
+  TORCH_CHECK(t.is_cuda() && t.dim() == 3 && t.stride(2) == 1 &&
+                  t.stride(0) == (long)L * t.stride(1) && t.stride(1) >= t.size(2),
+              "flash tensor must be a (B, L, D) row-strided view");
+  return t.stride(1);
+}
+
 std::vector<at::Tensor> flash_attn_fwd(at::Tensor Q, at::Tensor K, at::Tensor V, int64_t H,
                                        c10::optional<at::Tensor> valid,
                                        c10::optional<at::Tensor> bias, double scale, bool causal,
                                        double dropout_p, int64_t seed) {
-  CHECK_GPU(Q);
-  CHECK_GPU(K);
-  CHECK_GPU(V);
   TORCH_CHECK(Q.scalar_type() == at::kBFloat16, "flash attention is bf16");
   const int B = Q.size(0), L = Q.size(1);
+  const long ldq = fa_ld(Q, L);
+  const long ldkv = fa_ld(K, L);
+  TORCH_CHECK(fa_ld(V, L) == ldkv, "K and V must share a row stride");
   TORCH_CHECK(Q.size(2) == H * 64, "head_dim must be 64");
   TORCH_CHECK(L % 64 == 0, "L must be a multiple of 64");
-  auto O = at::empty_like(Q);
+  auto O = at::empty({B, L, H * 64}, Q.options());
   auto lse = at::empty({B, H, L}, Q.options().dtype(at::kFloat));
   const float* bptr = nullptr;
   if (bias.has_value()) {
@@ -669,7 +678,8 @@ std::vector<at::Tensor> flash_attn_fwd(at::Tensor Q, at::Tensor K, at::Tensor V,
   }
   launch_flash_fwd(ptr<bf16_t>(Q), ptr<bf16_t>(K), ptr<bf16_t>(V), opt_valid_ptr(valid), bptr,
                    mptr<bf16_t>(O), lse.data_ptr<float>(), B, H, L, (float)scale, causal ? 1 : 0,
-                   (unsigned)(dropout_p * 256.0), (unsigned long long)seed, cur_stream());
+                   (unsigned)(dropout_p * 256.0), (unsigned long long)seed, ldq, ldkv,
+                   cur_stream());
   return {O, lse};
 }
 
@@ -680,12 +690,16 @@ std::vector<at::Tensor> flash_attn_bwd(at::Tensor dO, at::Tensor Q, at::Tensor K
                                        double dropout_p, int64_t seed, bool need_dbias) {
   CHECK_GPU(dO);
   const int B = Q.size(0), L = Q.size(1);
+  const long ldq = fa_ld(Q, L);
+  const long ldkv = fa_ld(K, L);
+  TORCH_CHECK(fa_ld(V, L) == ldkv, "K and V must share a row stride");
   auto stream = cur_stream();
   auto Dterm = at::empty({B, H, L}, Q.options().dtype(at::kFloat));
   launch_flash_dterm(ptr<bf16_t>(dO), ptr<bf16_t>(O), Dterm.data_ptr<float>(), B, H, L, stream);
-  auto dQ = at::empty_like(Q);
-  auto dK = at::empty_like(K);
-  auto dV = at::empty_like(V);
+  // grads are allocated CONTIGUOUS regardless of the input views
+  auto dQ = at::empty({B, L, (long)H * 64}, Q.options());
+  auto dK = at::empty({B, L, (long)H * 64}, Q.options());
+  auto dV = at::empty({B, L, (long)H * 64}, Q.options());
   const float* bptr = nullptr;
   if (bias.has_value()) bptr = bias->data_ptr<float>();
   at::Tensor dBias;
@@ -698,11 +712,11 @@ std::vector<at::Tensor> flash_attn_bwd(at::Tensor dO, at::Tensor Q, at::Tensor K
   launch_flash_dq(ptr<bf16_t>(Q), ptr<bf16_t>(K), ptr<bf16_t>(V), ptr<bf16_t>(dO),
                   opt_valid_ptr(valid), bptr, lse.data_ptr<float>(), Dterm.data_ptr<float>(),
                   mptr<bf16_t>(dQ), dbias_ptr, B, H, L, (float)scale, causal ? 1 : 0,
-                  (unsigned)(dropout_p * 256.0), (unsigned long long)seed, stream);
+                  (unsigned)(dropout_p * 256.0), (unsigned long long)seed, ldq, ldkv, stream);
   launch_flash_dkv(ptr<bf16_t>(Q), ptr<bf16_t>(K), ptr<bf16_t>(V), ptr<bf16_t>(dO),
                    opt_valid_ptr(valid), bptr, lse.data_ptr<float>(), Dterm.data_ptr<float>(),
                    mptr<bf16_t>(dK), mptr<bf16_t>(dV), B, H, L, (float)scale, causal ? 1 : 0,
-                   (unsigned)(dropout_p * 256.0), (unsigned long long)seed, stream);
+                   (unsigned)(dropout_p * 256.0), (unsigned long long)seed, ldq, ldkv, stream);
   if (need_dbias) return {dQ, dK, dV, dBias};
   return {dQ, dK, dV};
 }

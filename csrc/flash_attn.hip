@@ -70,7 +70,9 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
     const bf16* __restrict__ V, const int* __restrict__ valid,
     const float* __restrict__ bias, bf16* __restrict__ O,
     float* __restrict__ lse, int B, int H, int L, float scale, int causal,
-    unsigned p8, unsigned long long seed) {
+    unsigned p8, unsigned long long seed, long ldq, long ldkv) {
+  // ldq/ldkv: row strides (elements) of the Q and K/V inputs — H*64 for
+  // contiguous tensors, 3*H*64 when they are slices of a fused QKV buffer
   extern __shared__ __attribute__((aligned(16))) char smem[];
   char* k_lds = smem;                // [TK][64] bf16 swizzled = 8 KiB
   char* v_lds = smem + TK * 128;     // [64][TK] bf16 (transposed) = 8 KiB
@@ -98,10 +100,10 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
       const int d = ks * 32 + (lane >> 4) * 8;
       if (q_full)
         q_frag[fq][ks] = *reinterpret_cast<const bf16x8*>(
-            Q + ((long)b * L + qrow) * HD + (long)h * 64 + d);
+            Q + ((long)b * L + qrow) * ldq + (long)h * 64 + d);
       else if (qrow < L)
         q_frag[fq][ks] = *reinterpret_cast<const bf16x8*>(
-            Q + ((long)b * L + qrow) * HD + (long)h * 64 + d);
+            Q + ((long)b * L + qrow) * ldq + (long)h * 64 + d);
       else
         q_frag[fq][ks] = bf16x8{};
     }
@@ -126,14 +128,14 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
         bf16 vv[8] = {};
         if (kv_full) {
           kv = *reinterpret_cast<const uint4v*>(
-              K + ((long)b * L + key) * HD + (long)h * 64 + off / 2);
+              K + ((long)b * L + key) * ldkv + (long)h * 64 + off / 2);
           *reinterpret_cast<uint4v*>(vv) = *reinterpret_cast<const uint4v*>(
-              V + ((long)b * L + key) * HD + (long)h * 64 + off / 2);
+              V + ((long)b * L + key) * ldkv + (long)h * 64 + off / 2);
         } else if (key < vl) {
           kv = *reinterpret_cast<const uint4v*>(
-              K + ((long)b * L + key) * HD + (long)h * 64 + off / 2);
+              K + ((long)b * L + key) * ldkv + (long)h * 64 + off / 2);
           *reinterpret_cast<uint4v*>(vv) = *reinterpret_cast<const uint4v*>(
-              V + ((long)b * L + key) * HD + (long)h * 64 + off / 2);
+              V + ((long)b * L + key) * ldkv + (long)h * 64 + off / 2);
         }
         *reinterpret_cast<uint4v*>(k_lds + fa_swz(row + rr, off)) = kv;
         const int d0 = off / 2;
@@ -394,7 +396,8 @@ __global__ __launch_bounds__(256) void flash_dq_kernel(
     const int* __restrict__ valid, const float* __restrict__ bias,
     const float* __restrict__ lse, const float* __restrict__ Dterm,
     bf16* __restrict__ dQ, float* __restrict__ dBias, int B, int H, int L,
-    float scale, int causal, unsigned p8, unsigned long long seed) {
+    float scale, int causal, unsigned p8, unsigned long long seed, long ldq,
+    long ldkv) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   char* k_lds = smem;                 // [key][d] swizzled, 8 KiB
   char* kt_lds = smem + TK * 128;     // [d][key] swizzled, 8 KiB
@@ -419,13 +422,14 @@ __global__ __launch_bounds__(256) void flash_dq_kernel(
     for (int ks = 0; ks < 2; ++ks) {
       const int qrow = qw + fq * 16 + (lane & 15);
       const int d = ks * 32 + (lane >> 4) * 8;
-      const long base = ((long)b * L + qrow) * HD + (long)h * 64 + d;
+      const long baseq = ((long)b * L + qrow) * ldq + (long)h * 64 + d;
+      const long baseo = ((long)b * L + qrow) * HD + (long)h * 64 + d;
       if (q_full) {
-        q_frag[fq][ks] = *reinterpret_cast<const bf16x8*>(Q + base);
-        do_frag[fq][ks] = *reinterpret_cast<const bf16x8*>(dO + base);
+        q_frag[fq][ks] = *reinterpret_cast<const bf16x8*>(Q + baseq);
+        do_frag[fq][ks] = *reinterpret_cast<const bf16x8*>(dO + baseo);
       } else if (qrow < L) {
-        q_frag[fq][ks] = *reinterpret_cast<const bf16x8*>(Q + base);
-        do_frag[fq][ks] = *reinterpret_cast<const bf16x8*>(dO + base);
+        q_frag[fq][ks] = *reinterpret_cast<const bf16x8*>(Q + baseq);
+        do_frag[fq][ks] = *reinterpret_cast<const bf16x8*>(dO + baseo);
       } else {
         q_frag[fq][ks] = bf16x8{};
         do_frag[fq][ks] = bf16x8{};
@@ -453,14 +457,14 @@ __global__ __launch_bounds__(256) void flash_dq_kernel(
         uint4v kv = {}, vv4 = {};
         if (kv_full) {
           kv = *reinterpret_cast<const uint4v*>(
-              K + ((long)b * L + key) * HD + (long)h * 64 + off / 2);
+              K + ((long)b * L + key) * ldkv + (long)h * 64 + off / 2);
           vv4 = *reinterpret_cast<const uint4v*>(
-              V + ((long)b * L + key) * HD + (long)h * 64 + off / 2);
+              V + ((long)b * L + key) * ldkv + (long)h * 64 + off / 2);
         } else if (key < vl) {
           kv = *reinterpret_cast<const uint4v*>(
-              K + ((long)b * L + key) * HD + (long)h * 64 + off / 2);
+              K + ((long)b * L + key) * ldkv + (long)h * 64 + off / 2);
           vv4 = *reinterpret_cast<const uint4v*>(
-              V + ((long)b * L + key) * HD + (long)h * 64 + off / 2);
+              V + ((long)b * L + key) * ldkv + (long)h * 64 + off / 2);
         }
         *reinterpret_cast<uint4v*>(k_lds + fa_swz(row + rr, off)) = kv;
         *reinterpret_cast<uint4v*>(v_lds + fa_swz(row + rr, off)) = vv4;
@@ -619,7 +623,8 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
     const int* __restrict__ valid, const float* __restrict__ bias,
     const float* __restrict__ lse, const float* __restrict__ Dterm,
     bf16* __restrict__ dK, bf16* __restrict__ dV, int B, int H, int L,
-    float scale, int causal, unsigned p8, unsigned long long seed) {
+    float scale, int causal, unsigned p8, unsigned long long seed, long ldq,
+    long ldkv) {
   // Wave grid 2 (key halves) x 2 (q interleave): halves the per-wave
   // accumulator footprint (the 64-key variant needed 128 fp32 accumulators
   // on top of ~176 VGPRs -> 1 wave/SIMD on the unified register file).
@@ -666,14 +671,14 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
       uint4v kv = {}, vv = {};
       if (kv_full) {
         kv = *reinterpret_cast<const uint4v*>(
-            K + ((long)b * L + key) * HD + (long)h * 64 + off / 2);
+            K + ((long)b * L + key) * ldkv + (long)h * 64 + off / 2);
         vv = *reinterpret_cast<const uint4v*>(
-            V + ((long)b * L + key) * HD + (long)h * 64 + off / 2);
+            V + ((long)b * L + key) * ldkv + (long)h * 64 + off / 2);
       } else if (key < vl && kv0 < vl) {
         kv = *reinterpret_cast<const uint4v*>(
-            K + ((long)b * L + key) * HD + (long)h * 64 + off / 2);
+            K + ((long)b * L + key) * ldkv + (long)h * 64 + off / 2);
         vv = *reinterpret_cast<const uint4v*>(
-            V + ((long)b * L + key) * HD + (long)h * 64 + off / 2);
+            V + ((long)b * L + key) * ldkv + (long)h * 64 + off / 2);
       }
       *reinterpret_cast<uint4v*>(k_lds + fa_swz(row + rr, off)) = kv;
       *reinterpret_cast<uint4v*>(v_lds + fa_swz(row + rr, off)) = vv;
@@ -709,13 +714,14 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
         for (int ks = 0; ks < 2; ++ks) {
           const int qrow = qw + fq * 16 + (lane & 15);
           const int d = ks * 32 + (lane >> 4) * 8;
-          const long base = ((long)b * L + qrow) * HD + (long)h * 64 + d;
+          const long baseq = ((long)b * L + qrow) * ldq + (long)h * 64 + d;
+          const long baseo = ((long)b * L + qrow) * HD + (long)h * 64 + d;
           if (q_full) {
-            q_frag[fq][ks] = *reinterpret_cast<const bf16x8*>(Q + base);
-            do_frag[fq][ks] = *reinterpret_cast<const bf16x8*>(dO + base);
+            q_frag[fq][ks] = *reinterpret_cast<const bf16x8*>(Q + baseq);
+            do_frag[fq][ks] = *reinterpret_cast<const bf16x8*>(dO + baseo);
           } else if (qrow < L) {
-            q_frag[fq][ks] = *reinterpret_cast<const bf16x8*>(Q + base);
-            do_frag[fq][ks] = *reinterpret_cast<const bf16x8*>(dO + base);
+            q_frag[fq][ks] = *reinterpret_cast<const bf16x8*>(Q + baseq);
+            do_frag[fq][ks] = *reinterpret_cast<const bf16x8*>(dO + baseo);
           } else {
             q_frag[fq][ks] = bf16x8{};
             do_frag[fq][ks] = bf16x8{};
@@ -828,10 +834,11 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
           const int d = fd * 16 + (lane & 15);
 #pragma unroll
           for (int j = 0; j < 8; ++j) {
-            const long base =
-                ((long)b * L + qw + (lane >> 4) * 8 + j) * HD + (long)h * 64 + d;
-            dob[fd][j] = *reinterpret_cast<const __bf16*>(dO + base);
-            qb[fd][j] = *reinterpret_cast<const __bf16*>(Q + base);
+            const long row = (long)b * L + qw + (lane >> 4) * 8 + j;
+            dob[fd][j] = *reinterpret_cast<const __bf16*>(
+                dO + row * HD + (long)h * 64 + d);
+            qb[fd][j] = *reinterpret_cast<const __bf16*>(
+                Q + row * ldq + (long)h * 64 + d);
           }
         }
       } else {
@@ -842,9 +849,11 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
           for (int j = 0; j < 8; ++j) {
             const int qrow = qw + (lane >> 4) * 8 + j;
             if (qrow < L) {
-              const long base = ((long)b * L + qrow) * HD + (long)h * 64 + d;
-              dob[fd][j] = *reinterpret_cast<const __bf16*>(dO + base);
-              qb[fd][j] = *reinterpret_cast<const __bf16*>(Q + base);
+              const long row = (long)b * L + qrow;
+              dob[fd][j] = *reinterpret_cast<const __bf16*>(
+                  dO + row * HD + (long)h * 64 + d);
+              qb[fd][j] = *reinterpret_cast<const __bf16*>(
+                  Q + row * ldq + (long)h * 64 + d);
             } else {
               dob[fd][j] = (__bf16)0.0f;
               qb[fd][j] = (__bf16)0.0f;
@@ -941,12 +950,13 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
 void launch_flash_fwd(const bf16* Q, const bf16* K, const bf16* V,
                       const int* valid, const float* bias, bf16* O, float* lse,
                       int B, int H, int L, float scale, int causal,
-                      unsigned p8, unsigned long long seed,
-                      hipStream_t stream) {
+                      unsigned p8, unsigned long long seed, long ldq,
+                      long ldkv, hipStream_t stream) {
   const dim3 grid((L + NWAVE * TQW - 1) / (NWAVE * TQW), B * H);
   const size_t lds = 2 * TK * 128;
   hipLaunchKernelGGL(flash_fwd_kernel, grid, dim3(256), lds, stream, Q, K, V,
-                     valid, bias, O, lse, B, H, L, scale, causal, p8, seed);
+                     valid, bias, O, lse, B, H, L, scale, causal, p8, seed,
+                     ldq, ldkv);
 }
 
 void launch_flash_dterm(const bf16* dO, const bf16* O, float* Dterm, int B,
@@ -963,20 +973,20 @@ void launch_flash_dq(const bf16* Q, const bf16* K, const bf16* V,
                      const float* lse, const float* Dterm, bf16* dQ,
                      float* dBias, int B, int H, int L, float scale,
                      int causal, unsigned p8, unsigned long long seed,
-                     hipStream_t stream) {
+                     long ldq, long ldkv, hipStream_t stream) {
   const dim3 grid((L + NWAVE * TQW - 1) / (NWAVE * TQW), B * H);
   const size_t lds = 3 * TK * 128;
   hipLaunchKernelGGL(flash_dq_kernel, grid, dim3(256), lds, stream, Q, K, V,
                      dO, valid, bias, lse, Dterm, dQ, dBias, B, H, L, scale,
-                     causal, p8, seed);
+                     causal, p8, seed, ldq, ldkv);
 }
 
 void launch_flash_dkv(const bf16* Q, const bf16* K, const bf16* V,
                       const bf16* dO, const int* valid, const float* bias,
                       const float* lse, const float* Dterm, bf16* dK,
                       bf16* dV, int B, int H, int L, float scale, int causal,
-                      unsigned p8, unsigned long long seed,
-                      hipStream_t stream) {
+                      unsigned p8, unsigned long long seed, long ldq,
+                      long ldkv, hipStream_t stream) {
   const dim3 grid((L + TK - 1) / TK, B * H);
   size_t lds = 2 * TK * 128 + 18432;  // K,V + pd/ds bounces (72-B rows)
   const char* e = getenv("DFA_DKV_VARIANT");  // re-read: lets one probe
@@ -985,30 +995,30 @@ void launch_flash_dkv(const bf16* Q, const bf16* K, const bf16* V,
     lds += 4 * 2 * DKV_T_BYTES;
     hipLaunchKernelGGL(flash_dkv_kernel<2>, grid, dim3(256), lds, stream, Q, K,
                        V, dO, valid, bias, lse, Dterm, dK, dV, B, H, L, scale,
-                       causal, p8, seed);
+                       causal, p8, seed, ldq, ldkv);
   } else if (var == 4) {
     // K,V + max(pd/ds region, 64x272 pair image)
     lds = 2 * TK * 128 + 18432;
     hipLaunchKernelGGL(flash_dkv_kernel<4>, grid, dim3(256), lds, stream, Q, K,
                        V, dO, valid, bias, lse, Dterm, dK, dV, B, H, L, scale,
-                       causal, p8, seed);
+                       causal, p8, seed, ldq, ldkv);
   } else if (var == 3) {
     hipLaunchKernelGGL(flash_dkv_kernel<3>, grid, dim3(256), lds, stream, Q, K,
                        V, dO, valid, bias, lse, Dterm, dK, dV, B, H, L, scale,
-                       causal, p8, seed);
+                       causal, p8, seed, ldq, ldkv);
   } else if (var == 5) {
     lds += 4 * 2 * DKV_T_BYTES;  // bounce staging + pair-reduce epilogue
     hipLaunchKernelGGL(flash_dkv_kernel<5>, grid, dim3(256), lds, stream, Q, K,
                        V, dO, valid, bias, lse, Dterm, dK, dV, B, H, L, scale,
-                       causal, p8, seed);
+                       causal, p8, seed, ldq, ldkv);
   } else if (var == 9) {
     hipLaunchKernelGGL(flash_dkv_kernel<9>, grid, dim3(256), lds, stream, Q, K,
                        V, dO, valid, bias, lse, Dterm, dK, dV, B, H, L, scale,
-                       causal, p8, seed);
+                       causal, p8, seed, ldq, ldkv);
   } else {
     hipLaunchKernelGGL(flash_dkv_kernel<0>, grid, dim3(256), lds, stream, Q, K,
                        V, dO, valid, bias, lse, Dterm, dK, dV, B, H, L, scale,
-                       causal, p8, seed);
+                       causal, p8, seed, ldq, ldkv);
   }
 }
 

@@ -109,12 +109,21 @@ class RobertaSelfAttention(nn.Module):
         self.dropout = nn.Dropout(cfg.attention_probs_dropout_prob)  # CPU path
 
     def forward(self, x, valid: Optional[torch.Tensor], output_attentions: bool = False):
+        from ..ops.transformer import fused_qkv
+
         B, L, D = x.shape
         H, d = self.num_heads, self.head_dim
         p = self.dropout_p if self.training else 0.0
-        q = fused_linear(x, self.query.weight, self.query.bias)
-        k = fused_linear(x, self.key.weight, self.key.bias)
-        v = fused_linear(x, self.value.weight, self.value.bias)
+        qkv = None
+        if d == 64 and not output_attentions and L % 64 == 0:
+            qkv = fused_qkv(x, self.query.weight, self.key.weight, self.value.weight,
+                            self.query.bias, self.key.bias, self.value.bias)
+        if qkv is not None:
+            q, k, v = qkv[..., :D], qkv[..., D:2 * D], qkv[..., 2 * D:]
+        else:
+            q = fused_linear(x, self.query.weight, self.query.bias)
+            k = fused_linear(x, self.key.weight, self.key.bias)
+            v = fused_linear(x, self.value.weight, self.value.bias)
         if d == 64 and not output_attentions and flash_usable(q, L):
             out = flash_attention(q, k, v, H, valid=valid, scale=1.0 / math.sqrt(d),
                                   dropout_p=p)

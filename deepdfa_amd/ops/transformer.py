@@ -227,10 +227,10 @@ class _FlashAttention(torch.autograd.Function):
     def forward(ctx, q, k, v, H, valid, bias, scale, causal, dropout_p):
         ext = load_ext(required=True)
         seed = _next_seed() if dropout_p > 0 else 0
-        O, lse = ext.flash_attn_fwd(
-            q.contiguous(), k.contiguous(), v.contiguous(), H, valid, bias,
-            scale, causal, dropout_p, seed,
-        )
+        # q/k/v may be row-strided slices of a fused QKV buffer — the
+        # kernels take a row stride, so no contiguous() copies here
+        O, lse = ext.flash_attn_fwd(q, k, v, H, valid, bias,
+                                    scale, causal, dropout_p, seed)
         ctx.save_for_backward(q, k, v, O, lse)
         ctx.meta = (H, valid, bias, scale, causal, dropout_p, seed)
         ctx.need_dbias = bias is not None and ctx.needs_input_grad[5]
@@ -242,7 +242,7 @@ class _FlashAttention(torch.autograd.Function):
         q, k, v, O, lse = ctx.saved_tensors
         H, valid, bias, scale, causal, dropout_p, seed = ctx.meta
         outs = ext.flash_attn_bwd(
-            dO.contiguous(), q.contiguous(), k.contiguous(), v.contiguous(), O, lse,
+            dO.contiguous(), q, k, v, O, lse,
             H, valid, bias, scale, causal, dropout_p, seed, ctx.need_dbias,
         )
         dbias = outs[3] if ctx.need_dbias else None
@@ -368,3 +368,59 @@ def layer_norm_res_dropout(h, res, weight, bias, dropout_p=0.0, eps=1e-5):
             z.float(), (z.shape[-1],), weight.float(), bias.float(), eps
         ).to(z.dtype)
     )
+
+
+class _QKVLinear(torch.autograd.Function):
+    """Fused QKV projection: one (N, 3D) gemm2 / wgrad per layer instead of
+    three 768-column calls (a 64x6-block grid underfills 256 CUs; the
+    64x18 fused grid balances, and one K=8192 wgrad replaces three)."""
+
+    @staticmethod
+    def forward(ctx, x, wq, wk, wv, bq, bk, bv):
+        ext = load_ext(required=True)
+        x2d = x.reshape(-1, x.shape[-1]).contiguous()
+        cache = getattr(wq, "_dfa_qkv_cache", None)
+        key = (wq._version, wk._version, wv._version, _weights_epoch[0])
+        if cache is None:
+            wcat = torch.cat([wq.detach(), wk.detach(), wv.detach()])
+            w16 = wcat.to(torch.bfloat16).contiguous()
+            wt16 = w16.t().contiguous()
+            b32 = torch.cat([bq.detach(), bk.detach(), bv.detach()]).float().contiguous()
+            wq._dfa_qkv_cache = (key, w16, wt16, b32)
+        elif CAPTURE_REFRESH[0] or cache[0] != key:
+            _, w16, wt16, b32 = cache
+            D = wq.shape[0]
+            w16[:D].copy_(wq.detach())
+            w16[D:2 * D].copy_(wk.detach())
+            w16[2 * D:].copy_(wv.detach())
+            wt16.copy_(w16.t())
+            b32[:D].copy_(bq.detach())
+            b32[D:2 * D].copy_(bk.detach())
+            b32[2 * D:].copy_(bv.detach())
+            wq._dfa_qkv_cache = (key, w16, wt16, b32)
+        else:
+            _, w16, wt16, b32 = cache
+        out = ext.gemm2(x2d, w16, b32, None)
+        ctx.save_for_backward(x2d, w16, wt16)
+        ctx.x_shape = x.shape
+        return out.view(*x.shape[:-1], w16.shape[0])
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = load_ext(required=True)
+        x2d, w16, wt16 = ctx.saved_tensors
+        dy2d = dy.reshape(-1, dy.shape[-1]).contiguous()
+        dx = ext.gemm2(dy2d, wt16, None, None)
+        dw = ext.wgrad(dy2d, x2d)  # (3D, K) fp32
+        db = ext.colsum(dy2d)
+        D = w16.shape[0] // 3
+        return (dx.view(ctx.x_shape), dw[:D], dw[D:2 * D], dw[2 * D:],
+                db[:D], db[D:2 * D], db[2 * D:])
+
+
+def fused_qkv(x, wq, wk, wv, bq, bk, bv):
+    """Returns the fused (B, L, 3D) projection, or None if the geometry
+    doesn't fit the custom GEMMs (caller falls back to per-projection)."""
+    if linear_usable(x, wq) and wq.shape[0] % 128 == 0:
+        return _QKVLinear.apply(x, wq, wk, wv, bq, bk, bv)
+    return None

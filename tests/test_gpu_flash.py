@@ -144,3 +144,34 @@ def test_roberta_flash_vs_materialized(dev):
     mask = ids.ne(1).unsqueeze(-1)
     err = ((out_flash.float() - out_mat.float()) * mask).abs().max().item()
     assert err < 0.05, err
+
+
+@pytest.mark.gpu
+def test_flash_strided_qkv_views_match_contiguous():
+    """Row-strided Q/K/V (slices of a fused QKV buffer) must give the same
+    forward and backward as contiguous tensors."""
+    from deepdfa_amd.ops.transformer import flash_attention
+
+    torch.manual_seed(0)
+    B, L, H, d = 2, 128, 4, 64
+    D = H * d
+    qkv = (torch.randn(B, L, 3 * D, device="cuda", dtype=torch.bfloat16) * 0.3
+           ).requires_grad_()
+    qc = qkv[..., :D].detach().contiguous().requires_grad_()
+    kc = qkv[..., D:2 * D].detach().contiguous().requires_grad_()
+    vc = qkv[..., 2 * D:].detach().contiguous().requires_grad_()
+    valid = torch.tensor([L, 96], dtype=torch.int32, device="cuda")
+
+    o1 = flash_attention(qkv[..., :D], qkv[..., D:2 * D], qkv[..., 2 * D:],
+                         H, valid=valid, scale=0.125)
+    o2 = flash_attention(qc, kc, vc, H, valid=valid, scale=0.125)
+    assert torch.equal(o1, o2)
+    g = torch.randn_like(o1)
+    o1.backward(g)
+    o2.backward(g)
+    dq = qkv.grad[..., :D]
+    dk = qkv.grad[..., D:2 * D]
+    dv = qkv.grad[..., 2 * D:]
+    assert torch.equal(dq, qc.grad)
+    assert torch.equal(dk, kc.grad)
+    assert torch.equal(dv, vc.grad)
