@@ -142,9 +142,20 @@ class T5Attention(nn.Module):
         def split(t, L):
             return t.view(B, L, H, d).transpose(1, 2)
 
-        qp = fused_linear(x, self.q.weight)
-        kp = fused_linear(src, self.k.weight)
-        vp = fused_linear(src, self.v.weight)
+        qkv = None
+        if kv is None and d == 64 and Lq % 64 == 0:
+            from ..ops.transformer import fused_qkv
+
+            qkv = fused_qkv(x, self.q.weight, self.k.weight, self.v.weight)
+        if qkv is not None:
+            inner = H * d
+            qp = qkv[..., :inner]
+            kp = qkv[..., inner:2 * inner]
+            vp = qkv[..., 2 * inner:]
+        else:
+            qp = fused_linear(x, self.q.weight)
+            kp = fused_linear(src, self.k.weight)
+            vp = fused_linear(src, self.v.weight)
         causal = self.causal and kv is None
         if d == 64 and flash_usable(qp, Lq, Lk):
             bias = None

@@ -381,11 +381,13 @@ class _QKVLinear(torch.autograd.Function):
         x2d = x.reshape(-1, x.shape[-1]).contiguous()
         cache = getattr(wq, "_dfa_qkv_cache", None)
         key = (wq._version, wk._version, wv._version, _weights_epoch[0])
+        has_bias = bq is not None
         if cache is None:
             wcat = torch.cat([wq.detach(), wk.detach(), wv.detach()])
             w16 = wcat.to(torch.bfloat16).contiguous()
             wt16 = w16.t().contiguous()
-            b32 = torch.cat([bq.detach(), bk.detach(), bv.detach()]).float().contiguous()
+            b32 = (torch.cat([bq.detach(), bk.detach(), bv.detach()]).float().contiguous()
+                   if has_bias else None)
             wq._dfa_qkv_cache = (key, w16, wt16, b32)
         elif CAPTURE_REFRESH[0] or cache[0] != key:
             _, w16, wt16, b32 = cache
@@ -394,14 +396,16 @@ class _QKVLinear(torch.autograd.Function):
             w16[D:2 * D].copy_(wk.detach())
             w16[2 * D:].copy_(wv.detach())
             wt16.copy_(w16.t())
-            b32[:D].copy_(bq.detach())
-            b32[D:2 * D].copy_(bk.detach())
-            b32[2 * D:].copy_(bv.detach())
+            if has_bias:
+                b32[:D].copy_(bq.detach())
+                b32[D:2 * D].copy_(bk.detach())
+                b32[2 * D:].copy_(bv.detach())
             wq._dfa_qkv_cache = (key, w16, wt16, b32)
         else:
             _, w16, wt16, b32 = cache
         out = ext.gemm2(x2d, w16, b32, None)
         ctx.save_for_backward(x2d, w16, wt16)
+        ctx.has_bias = has_bias
         ctx.x_shape = x.shape
         return out.view(*x.shape[:-1], w16.shape[0])
 
@@ -412,13 +416,16 @@ class _QKVLinear(torch.autograd.Function):
         dy2d = dy.reshape(-1, dy.shape[-1]).contiguous()
         dx = ext.gemm2(dy2d, wt16, None, None)
         dw = ext.wgrad(dy2d, x2d)  # (3D, K) fp32
-        db = ext.colsum(dy2d)
         D = w16.shape[0] // 3
+        if ctx.has_bias:
+            db = ext.colsum(dy2d)
+            return (dx.view(ctx.x_shape), dw[:D], dw[D:2 * D], dw[2 * D:],
+                    db[:D], db[D:2 * D], db[2 * D:])
         return (dx.view(ctx.x_shape), dw[:D], dw[D:2 * D], dw[2 * D:],
-                db[:D], db[D:2 * D], db[2 * D:])
+                None, None, None)
 
 
-def fused_qkv(x, wq, wk, wv, bq, bk, bv):
+def fused_qkv(x, wq, wk, wv, bq=None, bk=None, bv=None):
     """Returns the fused (B, L, 3D) projection, or None if the geometry
     doesn't fit the custom GEMMs (caller falls back to per-projection)."""
     if linear_usable(x, wq) and wq.shape[0] % 128 == 0:
