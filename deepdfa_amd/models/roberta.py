@@ -69,11 +69,20 @@ class LayerNorm(nn.Module):
         return layer_norm(x, self.weight, self.bias, self.eps)
 
 
+class Embedding(nn.Embedding):
+    # nn.Embedding with the custom scatter-add backward on GPU; keeps
+    # the module-call path so forward hooks (saliency scoring) fire
+    def forward(self, input):
+        from ..ops.transformer import embedding_lookup
+
+        return embedding_lookup(input, self.weight, self.padding_idx)
+
+
 class RobertaEmbeddings(nn.Module):
     def __init__(self, cfg: RobertaConfig):
         super().__init__()
-        self.word_embeddings = nn.Embedding(cfg.vocab_size, cfg.hidden_size, padding_idx=cfg.pad_token_id)
-        self.position_embeddings = nn.Embedding(
+        self.word_embeddings = Embedding(cfg.vocab_size, cfg.hidden_size, padding_idx=cfg.pad_token_id)
+        self.position_embeddings = Embedding(
             cfg.max_position_embeddings, cfg.hidden_size, padding_idx=cfg.pad_token_id
         )
         self.token_type_embeddings = nn.Embedding(cfg.type_vocab_size, cfg.hidden_size)
@@ -82,16 +91,14 @@ class RobertaEmbeddings(nn.Module):
         self.padding_idx = cfg.pad_token_id
 
     def forward(self, input_ids):
-        from ..ops.transformer import embedding_lookup
-
         mask = input_ids.ne(self.padding_idx).long()
         position_ids = torch.cumsum(mask, dim=1) * mask + self.padding_idx
         # token type is always 0 for this model family: broadcasting row 0
         # gives the identical result with a cheap sum-reduce gradient
         # instead of an all-rows-collide scatter
         emb = (
-            embedding_lookup(input_ids, self.word_embeddings.weight, self.padding_idx)
-            + embedding_lookup(position_ids, self.position_embeddings.weight, self.padding_idx)
+            self.word_embeddings(input_ids)
+            + self.position_embeddings(position_ids)
             + self.token_type_embeddings.weight[0]
         )
         return self.dropout(self.LayerNorm(emb))
