@@ -33,17 +33,21 @@ __device__ __forceinline__ int swz(int row, int byte) {
 }
 
 // out(N, COL) = concat_K(A1, A2)(N, K) @ W(COL, K)^T + bias
-// grid = (ceil(N/BM), COL/BN), block = 256 (4 waves, 2x2 of 32x64 tiles)
+// grid = (ceil(N/BMT), COL/BN), block = 256 (4 waves, 2x2)
+// BMT = 64 normally; 32 when the 64-row grid would underfill 256 CUs —
+// these GNN GEMMs are HBM-latency-bound at ~0.7 waves/SIMD (PMC: WAIT_ANY
+// 73% of wave cycles), so doubling the block count is the lever
+template <int BMT>
 __global__ __launch_bounds__(256) void gemm_bias_kernel(
     const bf16* __restrict__ A1, const bf16* __restrict__ A2,
     const bf16* __restrict__ W, const bf16* __restrict__ bias,
     const bf16* __restrict__ addend, bf16* __restrict__ out, int N, int K,
     int K1, int COL) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  char* a_lds = smem;                 // BM x ROWB = 8 KiB
-  char* b_lds = smem + BM * ROWB;     // BN x ROWB = 16 KiB
+  char* a_lds = smem;                 // BMT x ROWB
+  char* b_lds = smem + BMT * ROWB;    // BN x ROWB = 16 KiB
 
-  const int r0 = blockIdx.x * BM;
+  const int r0 = blockIdx.x * BMT;
   const int c0 = blockIdx.y * BN;
   const int tid = threadIdx.x;
   const int lane = tid & (WAVE - 1);
@@ -51,14 +55,15 @@ __global__ __launch_bounds__(256) void gemm_bias_kernel(
   const int wm = wid >> 1;            // 2 row-waves of 32 rows
   const int wn = wid & 1;             // 2 col-waves of 64 cols
 
-  f32x4 acc[2][4] = {};               // per wave: 32x64 = 2x4 fragments
+  constexpr int MFRAG = BMT / 32;     // fragments per row-wave
+  f32x4 acc[MFRAG][4] = {};
 
   for (int kk = 0; kk < K; kk += BK) {
-    // stage A tile: BM rows x 64 k (128 B/row); 8 threads/row, 32 rows/pass
+    // stage A tile: BMT rows x 64 k (128 B/row); 8 threads/row, 32 rows/pass
     {
       const int row = tid >> 3;        // 0..31
       const int off = (tid & 7) * 16;  // byte offset in row
-      for (int rr = row; rr < BM; rr += 32) {
+      for (int rr = row; rr < BMT; rr += 32) {
         const int gr = r0 + rr;
         uint4v v = {};
         if (gr < N) {
@@ -86,10 +91,10 @@ __global__ __launch_bounds__(256) void gemm_bias_kernel(
     for (int ks = 0; ks < BK / 32; ++ks) {
       // fragment k-range: lane kb = lane>>4 (4 groups of 8 elems)
       const int kbyte = ks * 64 + (lane >> 4) * 16;
-      bf16x8 a_frag[2], b_frag[4];
+      bf16x8 a_frag[MFRAG], b_frag[4];
 #pragma unroll
-      for (int m = 0; m < 2; ++m) {
-        const int row = wm * 32 + m * 16 + (lane & 15);
+      for (int m = 0; m < MFRAG; ++m) {
+        const int row = wm * (BMT / 2) + m * 16 + (lane & 15);
         a_frag[m] = *reinterpret_cast<const bf16x8*>(a_lds + swz(row, kbyte));
       }
 #pragma unroll
@@ -98,7 +103,7 @@ __global__ __launch_bounds__(256) void gemm_bias_kernel(
         b_frag[n] = *reinterpret_cast<const bf16x8*>(b_lds + swz(row, kbyte));
       }
 #pragma unroll
-      for (int m = 0; m < 2; ++m)
+      for (int m = 0; m < MFRAG; ++m)
 #pragma unroll
         for (int n = 0; n < 4; ++n)
           acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
@@ -109,8 +114,8 @@ __global__ __launch_bounds__(256) void gemm_bias_kernel(
 
   // epilogue: D mapping col = lane&15, row = (lane>>4)*4 + i
 #pragma unroll
-  for (int m = 0; m < 2; ++m) {
-    const int row_base = r0 + wm * 32 + m * 16 + (lane >> 4) * 4;
+  for (int m = 0; m < MFRAG; ++m) {
+    const int row_base = r0 + wm * (BMT / 2) + m * 16 + (lane >> 4) * 4;
 #pragma unroll
     for (int n = 0; n < 4; ++n) {
       const int col = c0 + wn * 64 + n * 16 + (lane & 15);
@@ -131,8 +136,16 @@ __global__ __launch_bounds__(256) void gemm_bias_kernel(
 void launch_gemm_bias(const bf16* A1, const bf16* A2, const bf16* W,
                       const bf16* bias, const bf16* addend, bf16* out, int N,
                       int K, int K1, int COL, hipStream_t stream) {
-  const dim3 grid((N + BM - 1) / BM, COL / BN);
-  const size_t lds = (BM + BN) * ROWB;
-  hipLaunchKernelGGL(gemm_bias_kernel, grid, dim3(256), lds, stream, A1, A2, W,
-                     bias, addend, out, N, K, K1, COL);
+  const long blocks64 = (long)((N + BM - 1) / BM) * (COL / BN);
+  if (blocks64 < 384) {
+    const dim3 grid((N + 31) / 32, COL / BN);
+    const size_t lds = (32 + BN) * ROWB;
+    hipLaunchKernelGGL(gemm_bias_kernel<32>, grid, dim3(256), lds, stream, A1,
+                       A2, W, bias, addend, out, N, K, K1, COL);
+  } else {
+    const dim3 grid((N + BM - 1) / BM, COL / BN);
+    const size_t lds = (BM + BN) * ROWB;
+    hipLaunchKernelGGL(gemm_bias_kernel<BM>, grid, dim3(256), lds, stream, A1,
+                       A2, W, bias, addend, out, N, K, K1, COL);
+  }
 }

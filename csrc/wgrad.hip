@@ -27,12 +27,19 @@ using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
 using f32x4 = __attribute__((ext_vector_type(4))) float;
 using uint4v = __attribute__((ext_vector_type(4))) unsigned int;
 
+// transposed-fragment bank map: at a 256-B row stride all four (lane>>4)
+// k-groups (8 rows = 512 dwords = 0 mod 32) hit the same 8 banks; XOR the
+// column byte with ((k>>3)&3)<<5 (bits 5-6 only: bijective inside the row,
+// keeps 16-B write alignment, disjoint bank blocks per k-group) — the same
+// fix as csrc/wgrad2.hip, worth 4.07e6 LDS_BANK_CONFLICT/call here
+#define W1_ADDR(k, colbyte) ((k)*256 + ((colbyte) ^ ((((k) >> 3) & 3) << 5)))
+
 __global__ __launch_bounds__(256) void wgrad_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ B1,
     const bf16* __restrict__ B2, float* __restrict__ out, int K, int M,
     int C, int C1, int kchunk) {
-  __shared__ __bf16 lds_a[KS][128];
-  __shared__ __bf16 lds_b[KS][128];
+  __shared__ char lds_a[KS * 256];
+  __shared__ char lds_b[KS * 256];
 
   const int m0 = blockIdx.x * 128;
   const int c0 = blockIdx.y * 128;
@@ -65,8 +72,8 @@ __global__ __launch_bounds__(256) void wgrad_kernel(
             vb = *reinterpret_cast<const uint4v*>(src);
           }
         }
-        *reinterpret_cast<uint4v*>(&lds_a[row + rr][coff]) = va;
-        *reinterpret_cast<uint4v*>(&lds_b[row + rr][coff]) = vb;
+        *reinterpret_cast<uint4v*>(lds_a + W1_ADDR(row + rr, coff * 2)) = va;
+        *reinterpret_cast<uint4v*>(lds_b + W1_ADDR(row + rr, coff * 2)) = vb;
       }
     }
     __syncthreads();
@@ -78,8 +85,9 @@ __global__ __launch_bounds__(256) void wgrad_kernel(
       const int c_loc = wc * 64 + f * 16 + (lane & 15);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        a_frag[f][j] = lds_a[(lane >> 4) * 8 + j][m_loc];
-        b_frag[f][j] = lds_b[(lane >> 4) * 8 + j][c_loc];
+        const int k = (lane >> 4) * 8 + j;
+        a_frag[f][j] = *reinterpret_cast<const __bf16*>(lds_a + W1_ADDR(k, m_loc * 2));
+        b_frag[f][j] = *reinterpret_cast<const __bf16*>(lds_b + W1_ADDR(k, c_loc * 2));
       }
     }
 #pragma unroll
