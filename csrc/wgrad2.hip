@@ -79,17 +79,20 @@ __global__ __launch_bounds__(256) void wgrad2_kernel(
   // scalar LDS loads (~4-way), double-buffered with the T14 load split so
   // the HBM latency hides under the MFMAs.
   extern __shared__ __attribute__((aligned(16))) char smem[];
-// row stride 272 B + k-group XOR: at any 16-B-multiple stride the scalar
-// transposed fragment reads put all four (lane>>4) k-groups (8 k apart,
-// 8*stride = 0 mod 32 dwords) on the SAME 8 banks -> 4-way conflict
-// (SQ_LDS_BANK_CONFLICT = 70% of ACTIVE_INST). XORing the column byte
-// with ((k>>3)&3)<<5 is a bijection inside the 256-B row, keeps the 16-B
-// write alignment (bits 5-6 only), and maps the four k-groups onto the
-// four disjoint 8-dword bank blocks -> full 32-bank coverage.
-#define W2_ROWB 272
-#define W2_ADDR(k, colbyte) ((k)*W2_ROWB + ((colbyte) ^ ((((k) >> 3) & 3) << 5)))
-#define A_LDS(i) (smem + (i)*35840)            // [k 64][m 128] bf16 padded
-#define B_LDS(i) (smem + 17408 + (i)*35840)    // [k 64][c 128] bf16 padded
+// Subtiled image for ds_read_b64_tr_b16 hardware-transpose reads (guide
+// T10; mapping verified by tools/tr_probe): [4 k][16 col] row-major
+// subtiles, padded 128->144 B so the 16 staging b128 writes (subtile
+// stride * 2 lanes/subtile) spread over the 64-dword bank modulus.
+// One tr read delivers 4 k-column elements per lane (lane l&15 = column)
+// — 2 reads per 8-deep MFMA fragment instead of 8 scalar ds_read_u16.
+#define W2_SUBB 144
+#define W2_SUB(kb, cb) (((kb)*8 + (cb)) * W2_SUBB)
+#define W2_TILEB (16 * 8 * W2_SUBB)  // 64 k x 128 col = 18432 B
+#define A_LDS(i) (smem + (i) * (2 * W2_TILEB))
+#define B_LDS(i) (smem + W2_TILEB + (i) * (2 * W2_TILEB))
+
+using bf16x4v = __attribute__((ext_vector_type(4))) __bf16;
+typedef __attribute__((address_space(3))) bf16x4v lds_bf16x4;
   const int m0 = blockIdx.x * WBM;
   const int c0 = blockIdx.y * WBC;
   const int k_begin = blockIdx.z * kchunk;
@@ -102,13 +105,18 @@ __global__ __launch_bounds__(256) void wgrad2_kernel(
 
   f32x4 acc[4][4] = {};
 
-  // natural-layout write: thread covers (k = tid>>4 + 16p, col8 = (tid&15)*8)
+  // subtiled write: thread covers (k = tid>>4 + 16p, cols (tid&15)*8..+7)
+  // -> one 16-B write into half ((tid&15)&1) of subtile (k>>2, (tid&15)>>1)
   auto write_nat = [&](char* lds, const uint4v regs[4]) {
     const int kr = tid >> 4;
-    const int c8b = (tid & 15) * 16;
+    const int cb = (tid & 15) >> 1;
+    const int halfb = ((tid & 15) & 1) * 16;
 #pragma unroll
-    for (int p = 0; p < 4; ++p)
-      *reinterpret_cast<uint4v*>(lds + W2_ADDR(kr + 16 * p, c8b)) = regs[p];
+    for (int p = 0; p < 4; ++p) {
+      const int k = kr + 16 * p;
+      *reinterpret_cast<uint4v*>(lds + W2_SUB(k >> 2, cb) + (k & 3) * 32 +
+                                 halfb) = regs[p];
+    }
   };
 
   uint4v a_regs[4], b_regs[4];
@@ -127,19 +135,23 @@ __global__ __launch_bounds__(256) void wgrad2_kernel(
     }
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
+      const int kb0 = ks * 8 + (lane >> 4) * 2;  // first 4-k subtile row
+      const int slot = (lane & 15) * 8;          // this lane's 8-B column slot
       bf16x8 a_frag[4], b_frag[4];
 #pragma unroll
       for (int f = 0; f < 4; ++f) {
-        const int mloc = wm * 64 + f * 16 + (lane & 15);
-        const int cloc = wc * 64 + f * 16 + (lane & 15);
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          const int k = ks * 32 + (lane >> 4) * 8 + j;
-          a_frag[f][j] = *reinterpret_cast<const __bf16*>(
-              A_LDS(cur) + W2_ADDR(k, mloc * 2));
-          b_frag[f][j] = *reinterpret_cast<const __bf16*>(
-              B_LDS(cur) + W2_ADDR(k, cloc * 2));
-        }
+        const int cba = wm * 4 + f;
+        const int cbb = wc * 4 + f;
+        bf16x4v* af = reinterpret_cast<bf16x4v*>(&a_frag[f]);
+        bf16x4v* bf = reinterpret_cast<bf16x4v*>(&b_frag[f]);
+        af[0] = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+            (lds_bf16x4*)(A_LDS(cur) + W2_SUB(kb0, cba) + slot));
+        af[1] = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+            (lds_bf16x4*)(A_LDS(cur) + W2_SUB(kb0 + 1, cba) + slot));
+        bf[0] = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+            (lds_bf16x4*)(B_LDS(cur) + W2_SUB(kb0, cbb) + slot));
+        bf[1] = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+            (lds_bf16x4*)(B_LDS(cur) + W2_SUB(kb0 + 1, cbb) + slot));
       }
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
@@ -188,6 +200,6 @@ void launch_wgrad2(const bf16* A, const bf16* B, float* out, int K, int M,
   if (kchunk < 512) kchunk = min(((K + WBK - 1) / WBK) * WBK, 512);
   zsplit = (K + kchunk - 1) / kchunk;
   const dim3 grid(M / WBM, C / WBC, zsplit);
-  hipLaunchKernelGGL(wgrad2_kernel, grid, dim3(256), 71680, stream, A, B, out,
+  hipLaunchKernelGGL(wgrad2_kernel, grid, dim3(256), 73728, stream, A, B, out,
                      K, M, C, kchunk, zsplit);
 }
