@@ -613,6 +613,14 @@ __device__ __forceinline__ void recompute_pT32(
 #define DKV_T_STRIDE 72
 #define DKV_T_BYTES (64 * DKV_T_STRIDE)
 
+// variant 6: [4 q][16 d] subtiles (144-B padded) for ds_read_b64_tr_b16
+// B-fragment reads (see csrc/wgrad2.hip / tools/tr_probe): staging is 16-B
+// vector writes straight from the q/dO fragments, reads are 2 tr reads per
+// fragment — replaces the 64 scalar L1 transposed loads per strip
+#define DKV_SUB(kb, cb) (((kb)*4 + (cb)) * 144)
+using bf16x4v = __attribute__((ext_vector_type(4))) __bf16;
+typedef __attribute__((address_space(3))) bf16x4v lds_bf16x4;
+
 // variant-9 instrumentation accumulator (cycles per kernel segment)
 __device__ unsigned long long dfa_dkv_prof[8];
 
@@ -743,6 +751,20 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
                   (bf16)q_frag[fq][ks][j];
             }
       }
+      if (VAR == 6) {
+        // subtiled staging: one 16-B vector write per (fq, ks) fragment
+#pragma unroll
+        for (int fq = 0; fq < 2; ++fq)
+#pragma unroll
+          for (int ks = 0; ks < 2; ++ks) {
+            const int q_loc = fq * 16 + (lane & 15);
+            const int d = ks * 32 + (lane >> 4) * 8;
+            const int off = DKV_SUB(q_loc >> 2, d >> 4) + (q_loc & 3) * 32 +
+                            ((d >> 3) & 1) * 16;
+            *reinterpret_cast<bf16x8*>(dot_lds + off) = do_frag[fq][ks];
+            *reinterpret_cast<bf16x8*>(qt_lds + off) = q_frag[fq][ks];
+          }
+      }
 
       float lse_w[2], dterm_w[2];
 #pragma unroll
@@ -819,7 +841,23 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
       // and dropping the [d][q] LDS bounce removes ~2/3 of the kernel's
       // LDS-issue cost (PMC: WAIT_INST_ANY 55%, LDS_IDX 39% of cycles)
       bf16x8 dob[4], qb[4];
-      if (VAR == 2 || VAR == 5) {
+      if (VAR == 6) {
+        const int kb0 = (lane >> 4) * 2;
+        const int slot = (lane & 15) * 8;
+#pragma unroll
+        for (int fd = 0; fd < 4; ++fd) {
+          bf16x4v* dv4 = reinterpret_cast<bf16x4v*>(&dob[fd]);
+          bf16x4v* qv4 = reinterpret_cast<bf16x4v*>(&qb[fd]);
+          dv4[0] = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+              (lds_bf16x4*)(dot_lds + DKV_SUB(kb0, fd) + slot));
+          dv4[1] = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+              (lds_bf16x4*)(dot_lds + DKV_SUB(kb0 + 1, fd) + slot));
+          qv4[0] = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+              (lds_bf16x4*)(qt_lds + DKV_SUB(kb0, fd) + slot));
+          qv4[1] = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+              (lds_bf16x4*)(qt_lds + DKV_SUB(kb0 + 1, fd) + slot));
+        }
+      } else if (VAR == 2 || VAR == 5) {
         // vector B-fragment reads from the padded [d][q] bounce images
         // (per-wave buffers: the compiler orders same-wave LDS write->read)
 #pragma unroll
@@ -875,7 +913,7 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
     }
   }
 
-  if (VAR == 4 || VAR == 5) {
+  if (VAR == 4 || VAR == 5 || VAR == 6) {
     // pair-wise reduce: the two waves of a key half (wq 0/1) hold the only
     // partials for those 32 keys. wq=1 vector-writes its accs to a [d][key]
     // f32 image with 272-B padded rows ((4d + key) mod 64 distinct within
@@ -1009,6 +1047,11 @@ void launch_flash_dkv(const bf16* Q, const bf16* K, const bf16* V,
   } else if (var == 5) {
     lds += 4 * 2 * DKV_T_BYTES;  // bounce staging + pair-reduce epilogue
     hipLaunchKernelGGL(flash_dkv_kernel<5>, grid, dim3(256), lds, stream, Q, K,
+                       V, dO, valid, bias, lse, Dterm, dK, dV, B, H, L, scale,
+                       causal, p8, seed, ldq, ldkv);
+  } else if (var == 6) {
+    lds += 4 * 2 * DKV_T_BYTES;  // subtiled tr_b16 bounce + pair-reduce
+    hipLaunchKernelGGL(flash_dkv_kernel<6>, grid, dim3(256), lds, stream, Q, K,
                        V, dO, valid, bias, lse, Dterm, dK, dV, B, H, L, scale,
                        causal, p8, seed, ldq, ldkv);
   } else if (var == 9) {
