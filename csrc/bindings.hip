@@ -83,6 +83,11 @@ void launch_softmax_mask_bwd(const T*, const T*, T*, long, int, float, float,
                              unsigned long long, hipStream_t);
 void dkv_prof_fetch(unsigned long long*);
 template <typename T>
+void launch_dropout_add_fwd(const T*, const T*, T*, long, float, unsigned long long,
+                            hipStream_t);
+template <typename T>
+void launch_dropout_add_bwd(const T*, T*, long, float, unsigned long long, hipStream_t);
+template <typename T>
 void launch_ln_res_dropout_fwd(const T*, const T*, const float*, const float*, T*, float*,
                                float*, long, int, float, float, unsigned long long, hipStream_t);
 template <typename T>
@@ -786,6 +791,30 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_wgrad", &rmsnorm_wgrad);
   m.def("flash_attn_fwd", &flash_attn_fwd);
   m.def("flash_attn_bwd", &flash_attn_bwd);
+  m.def("dropout_add_fwd", [](at::Tensor h, at::Tensor res, double p, int64_t seed) {
+    CHECK_GPU(h);
+    CHECK_GPU(res);
+    TORCH_CHECK(h.numel() == res.numel());
+    TORCH_CHECK(h.numel() % (h.scalar_type() == at::kFloat ? 4 : 8) == 0,
+                "dropout_add: numel must be a multiple of the 16-B vector");
+    auto out = at::empty_like(h);
+    dispatch_float_bf16(h, "dropout_add_fwd", [&](auto tag) {
+      using T = decltype(tag);
+      launch_dropout_add_fwd<T>(ptr<T>(h), ptr<T>(res), mptr<T>(out), h.numel(), (float)p,
+                                (unsigned long long)seed, cur_stream());
+    });
+    return out;
+  });
+  m.def("dropout_add_bwd", [](at::Tensor dy, double p, int64_t seed) {
+    CHECK_GPU(dy);
+    auto dh = at::empty_like(dy);
+    dispatch_float_bf16(dy, "dropout_add_bwd", [&](auto tag) {
+      using T = decltype(tag);
+      launch_dropout_add_bwd<T>(ptr<T>(dy), mptr<T>(dh), dy.numel(), (float)p,
+                                (unsigned long long)seed, cur_stream());
+    });
+    return dh;
+  });
   m.def("ln_res_dropout_fwd", [](at::Tensor h, at::Tensor res, at::Tensor gamma,
                                  at::Tensor beta, double eps, double p, int64_t seed) {
     CHECK_GPU(h);

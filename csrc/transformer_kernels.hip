@@ -995,3 +995,94 @@ void launch_ln_res_dropout_wgrad(const T* dy, const T* h, const T* res,
 INST_LNRD(float)
 INST_LNRD(__hip_bfloat16)
 #undef INST_LNRD
+
+// ---------------------------------------------------------------------------
+// Fused out = res + dropout(h): the pre-norm residual pattern (T5 layers).
+// Stateless dropout; backward needs only one kernel (dh = mask * dy, and
+// d(res) = dy is a pass-through the op returns directly).
+// ---------------------------------------------------------------------------
+
+template <typename T>
+__global__ void dropout_add_fwd_kernel(const T* __restrict__ h,
+                                       const T* __restrict__ res,
+                                       T* __restrict__ out, long total,
+                                       unsigned p8, unsigned long long seed,
+                                       float dscale) {
+  constexpr int VEC = 16 / sizeof(T);
+  const long nvec = total / VEC;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long iv = (long)blockIdx.x * blockDim.x + threadIdx.x; iv < nvec;
+       iv += stride) {
+    T vh[VEC], vr[VEC], vo[VEC];
+    *reinterpret_cast<ulonglong2*>(vh) =
+        *reinterpret_cast<const ulonglong2*>(h + iv * VEC);
+    *reinterpret_cast<ulonglong2*>(vr) =
+        *reinterpret_cast<const ulonglong2*>(res + iv * VEC);
+#pragma unroll
+    for (int u = 0; u < VEC; ++u) {
+      const bool keep = (p8 == 0) || keep_mask(iv * VEC + u, seed, p8);
+      vo[u] = ff<T>(tf(vr[u]) + (keep ? tf(vh[u]) * dscale : 0.f));
+    }
+    *reinterpret_cast<ulonglong2*>(out + iv * VEC) =
+        *reinterpret_cast<const ulonglong2*>(vo);
+  }
+}
+
+template <typename T>
+__global__ void dropout_add_bwd_kernel(const T* __restrict__ dy,
+                                       T* __restrict__ dh, long total,
+                                       unsigned p8, unsigned long long seed,
+                                       float dscale) {
+  constexpr int VEC = 16 / sizeof(T);
+  const long nvec = total / VEC;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long iv = (long)blockIdx.x * blockDim.x + threadIdx.x; iv < nvec;
+       iv += stride) {
+    T v[VEC], o[VEC];
+    *reinterpret_cast<ulonglong2*>(v) =
+        *reinterpret_cast<const ulonglong2*>(dy + iv * VEC);
+#pragma unroll
+    for (int u = 0; u < VEC; ++u) {
+      const bool keep = (p8 == 0) || keep_mask(iv * VEC + u, seed, p8);
+      o[u] = ff<T>(keep ? tf(v[u]) * dscale : 0.f);
+    }
+    *reinterpret_cast<ulonglong2*>(dh + iv * VEC) =
+        *reinterpret_cast<const ulonglong2*>(o);
+  }
+}
+
+template <typename T>
+void launch_dropout_add_fwd(const T* h, const T* res, T* out, long total,
+                            float p, unsigned long long seed,
+                            hipStream_t stream) {
+  const unsigned p8 = (unsigned)(p * 256.0f);
+  const float dscale = (p8 > 0) ? 256.0f / (256.0f - p8) : 1.0f;
+  const int block = 256;
+  const long nvec = total / (16 / sizeof(T));
+  const int grid = (int)min((nvec + block - 1) / block, (long)4096);
+  if (grid)
+    hipLaunchKernelGGL(dropout_add_fwd_kernel<T>, dim3(grid), dim3(block), 0,
+                       stream, h, res, out, total, p8, seed, dscale);
+}
+
+template <typename T>
+void launch_dropout_add_bwd(const T* dy, T* dh, long total, float p,
+                            unsigned long long seed, hipStream_t stream) {
+  const unsigned p8 = (unsigned)(p * 256.0f);
+  const float dscale = (p8 > 0) ? 256.0f / (256.0f - p8) : 1.0f;
+  const int block = 256;
+  const long nvec = total / (16 / sizeof(T));
+  const int grid = (int)min((nvec + block - 1) / block, (long)4096);
+  if (grid)
+    hipLaunchKernelGGL(dropout_add_bwd_kernel<T>, dim3(grid), dim3(block), 0,
+                       stream, dy, dh, total, p8, seed, dscale);
+}
+
+#define INST_DA(T)                                                            \
+  template void launch_dropout_add_fwd<T>(const T*, const T*, T*, long, float,\
+                                          unsigned long long, hipStream_t);   \
+  template void launch_dropout_add_bwd<T>(const T*, T*, long, float,          \
+                                          unsigned long long, hipStream_t);
+INST_DA(float)
+INST_DA(__hip_bfloat16)
+#undef INST_DA

@@ -21,7 +21,7 @@ from typing import Optional
 import torch
 from torch import nn
 
-from ..ops.transformer import (flash_attention, flash_usable, fused_linear, masked_softmax_dropout, rms_norm)
+from ..ops.transformer import (dropout_add, flash_attention, flash_usable, fused_linear, masked_softmax_dropout, rms_norm)
 
 
 @dataclass
@@ -186,7 +186,7 @@ class T5LayerSelfAttention(nn.Module):
     def forward(self, x, valid, position_bias, dropout_p):
         y = self.SelfAttention(self.layer_norm(x), valid, position_bias=position_bias,
                                dropout_p=dropout_p)
-        return x + self.dropout(y)
+        return dropout_add(y, x, self.dropout.p, self.training)
 
 
 class T5LayerCrossAttention(nn.Module):
@@ -198,7 +198,7 @@ class T5LayerCrossAttention(nn.Module):
 
     def forward(self, x, enc, enc_valid, dropout_p):
         y = self.EncDecAttention(self.layer_norm(x), enc_valid, kv=enc, dropout_p=dropout_p)
-        return x + self.dropout(y)
+        return dropout_add(y, x, self.dropout.p, self.training)
 
 
 class T5DenseActDense(nn.Module):
@@ -221,7 +221,8 @@ class T5LayerFF(nn.Module):
         self.dropout = nn.Dropout(cfg.dropout_rate)
 
     def forward(self, x):
-        return x + self.dropout(self.DenseReluDense(self.layer_norm(x)))
+        return dropout_add(self.DenseReluDense(self.layer_norm(x)), x,
+                           self.dropout.p, self.training)
 
 
 class T5Block(nn.Module):

@@ -431,3 +431,34 @@ def fused_qkv(x, wq, wk, wv, bq=None, bk=None, bv=None):
     if linear_usable(x, wq) and wq.shape[0] % 128 == 0:
         return _QKVLinear.apply(x, wq, wk, wv, bq, bk, bv)
     return None
+
+
+class _DropoutAdd(torch.autograd.Function):
+    """out = res + dropout(h) in one kernel (T5 pre-norm residuals);
+    d(res) = dy passes through with no kernel."""
+
+    @staticmethod
+    def forward(ctx, h, res, p):
+        ext = load_ext(required=True)
+        seed = _next_seed() if p > 0 else 0
+        ctx.meta = (p, seed)
+        return ext.dropout_add_fwd(h.contiguous(), res.contiguous(), p, seed)
+
+    @staticmethod
+    def backward(ctx, dy):
+        p, seed = ctx.meta
+        if p == 0:
+            return dy, dy, None
+        ext = load_ext(required=True)
+        dy = dy.contiguous()
+        return ext.dropout_add_bwd(dy, p, seed), dy, None
+
+
+def dropout_add(h, res, p=0.0, training=True):
+    """res + dropout(h); fused on GPU, torch composition elsewhere."""
+    if not training:
+        p = 0.0
+    if h.is_cuda and h.numel() % (4 if h.dtype == torch.float32 else 8) == 0:
+        return _DropoutAdd.apply(h, res, p)
+    z = torch.nn.functional.dropout(h, p) if p > 0 else h
+    return res + z

@@ -236,3 +236,30 @@ def test_ln_res_dropout_matches_composition():
     frac = (h2.grad.float() == 0).float().mean().item()
     assert 0.4 < frac < 0.6  # ~half the h-gradient masked
     assert (res2.grad.float() == 0).float().mean().item() < 0.05
+
+
+@pytest.mark.gpu
+def test_dropout_add_semantics():
+    from deepdfa_amd.ops.transformer import dropout_add
+
+    torch.manual_seed(0)
+    h = torch.randn(256, 768, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    res = torch.randn(256, 768, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    # p=0 path: exact add
+    out = dropout_add(h, res, 0.0)
+    assert torch.equal(out, (h + res).detach())
+    out.sum().backward()
+    assert torch.all(h.grad == 1) and torch.all(res.grad == 1)
+    h.grad = res.grad = None
+    # p>0: mask consistency between fwd and bwd (exact-representable
+    # inputs: h=1, res=0, so out==0 iff dropped)
+    h1 = torch.ones_like(h).requires_grad_()
+    r0 = torch.zeros_like(res).requires_grad_()
+    out = dropout_add(h1, r0, 0.5)
+    g = torch.ones_like(out)
+    out.backward(g)
+    dropped = out.detach() == 0
+    assert 0.4 < dropped.float().mean().item() < 0.6
+    assert torch.all(h1.grad[dropped].float() == 0)
+    assert torch.all(h1.grad[~dropped].float() == 2.0)  # 1/(1-p)
+    assert torch.equal(r0.grad, g)
