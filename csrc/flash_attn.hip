@@ -114,39 +114,60 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
   float l_st[2] = {0.f, 0.f};
   f32x4 o_acc[2][4] = {};  // [fq][fd]: O rows (l>>4)*4+i of fq block, col fd*16+(l&15)
 
+  // T14 double-buffered K/V staging: next tile's global loads issue before
+  // this tile's MFMAs; one barrier per tile (write targets the buffer the
+  // NEXT iteration reads — no same-buffer hazard)
   const int kv_end = causal ? min(vl, q0 + NWAVE * TQW) : vl;
-  for (int kv0 = 0; kv0 < kv_end; kv0 += TK) {
-    // ---- stage K tile [key][d] (swizzled) and V tile transposed [d][key] ----
-    {
-      const int row = tid >> 3;          // 0..31
-      const int off = (tid & 7) * 16;    // byte offset (8 bf16)
-      const bool kv_full = (kv0 + TK) <= vl;  // uniform fast path
+  const int srow = tid >> 3;           // staging: 0..31
+  const int soff = (tid & 7) * 16;     // byte offset (8 bf16)
+  auto load_tile = [&](int kv0, uint4v kreg[2], uint4v vreg[2]) {
+    const bool kv_full = (kv0 + TK) <= vl;  // uniform fast path
 #pragma unroll
-      for (int rr = 0; rr < TK; rr += 32) {
-        const int key = kv0 + row + rr;
-        uint4v kv = {};
-        bf16 vv[8] = {};
-        if (kv_full) {
-          kv = *reinterpret_cast<const uint4v*>(
-              K + ((long)b * L + key) * ldkv + (long)h * 64 + off / 2);
-          *reinterpret_cast<uint4v*>(vv) = *reinterpret_cast<const uint4v*>(
-              V + ((long)b * L + key) * ldkv + (long)h * 64 + off / 2);
-        } else if (key < vl) {
-          kv = *reinterpret_cast<const uint4v*>(
-              K + ((long)b * L + key) * ldkv + (long)h * 64 + off / 2);
-          *reinterpret_cast<uint4v*>(vv) = *reinterpret_cast<const uint4v*>(
-              V + ((long)b * L + key) * ldkv + (long)h * 64 + off / 2);
-        }
-        *reinterpret_cast<uint4v*>(k_lds + fa_swz(row + rr, off)) = kv;
-        const int d0 = off / 2;
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          *reinterpret_cast<bf16*>(
-              v_lds + fa_swzT(d0 + j, (row + rr) * 2)) = vv[j];
-        }
+    for (int rr = 0; rr < 2; ++rr) {
+      const int key = kv0 + srow + rr * 32;
+      kreg[rr] = {};
+      vreg[rr] = {};
+      if (kv_full) {
+        kreg[rr] = *reinterpret_cast<const uint4v*>(
+            K + ((long)b * L + key) * ldkv + (long)h * 64 + soff / 2);
+        vreg[rr] = *reinterpret_cast<const uint4v*>(
+            V + ((long)b * L + key) * ldkv + (long)h * 64 + soff / 2);
+      } else if (key < vl) {
+        kreg[rr] = *reinterpret_cast<const uint4v*>(
+            K + ((long)b * L + key) * ldkv + (long)h * 64 + soff / 2);
+        vreg[rr] = *reinterpret_cast<const uint4v*>(
+            V + ((long)b * L + key) * ldkv + (long)h * 64 + soff / 2);
       }
     }
-    __syncthreads();
+  };
+  auto write_tile = [&](char* kb, char* vb, const uint4v kreg[2],
+                        const uint4v vreg[2]) {
+#pragma unroll
+    for (int rr = 0; rr < 2; ++rr) {
+      *reinterpret_cast<uint4v*>(kb + fa_swz(srow + rr * 32, soff)) = kreg[rr];
+      bf16 vv[8];
+      *reinterpret_cast<uint4v*>(vv) = vreg[rr];
+      const int d0 = soff / 2;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        *reinterpret_cast<bf16*>(vb + fa_swzT(d0 + j, (srow + rr * 32) * 2)) =
+            vv[j];
+    }
+  };
+#define K_BUF(i) (k_lds + (i)*2 * TK * 128)
+#define V_BUF(i) (v_lds + (i)*2 * TK * 128)
+  uint4v kreg[2], vreg[2];
+  int cur = 0;
+  if (kv_end > 0) {
+    load_tile(0, kreg, vreg);
+    write_tile(K_BUF(0), V_BUF(0), kreg, vreg);
+  }
+  __syncthreads();
+  for (int kv0 = 0; kv0 < kv_end; kv0 += TK) {
+    const bool has_next = (kv0 + TK) < kv_end;
+    if (has_next) load_tile(kv0 + TK, kreg, vreg);
+    char* k_lds_c = K_BUF(cur);
+    char* v_lds_c = V_BUF(cur);
 
     // ---- S^T = K @ Q^T : D[key][qrow] ----
     f32x4 s_acc[4][2] = {};  // [fkey][fq]
@@ -157,7 +178,7 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
       for (int fk = 0; fk < 4; ++fk) {
         const int key = fk * 16 + (lane & 15);
         const int kbyte = ks * 64 + (lane >> 4) * 16;
-        k_frag[fk] = *reinterpret_cast<const bf16x8*>(k_lds + fa_swz(key, kbyte));
+        k_frag[fk] = *reinterpret_cast<const bf16x8*>(k_lds_c + fa_swz(key, kbyte));
       }
 #pragma unroll
       for (int fk = 0; fk < 4; ++fk)
@@ -268,7 +289,7 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
       for (int fd = 0; fd < 4; ++fd) {
         const int d = fd * 16 + (lane & 15);
         const int keybyte = kp * 64 + (lane >> 4) * 16;
-        v_frag[fd] = *reinterpret_cast<const bf16x8*>(v_lds + fa_swzT(d, keybyte));
+        v_frag[fd] = *reinterpret_cast<const bf16x8*>(v_lds_c + fa_swzT(d, keybyte));
       }
 #pragma unroll
       for (int fq = 0; fq < 2; ++fq)
@@ -277,8 +298,12 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
           o_acc[fq][fd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               pa[fq], v_frag[fd], o_acc[fq][fd], 0, 0, 0);
     }
+    if (has_next) write_tile(K_BUF(cur ^ 1), V_BUF(cur ^ 1), kreg, vreg);
     __syncthreads();
+    cur ^= 1;
   }
+#undef K_BUF
+#undef V_BUF
 
   // ---- epilogue: O /= l, write (B, L, H*64); save lse = m + log(l) ----
 #pragma unroll
@@ -445,41 +470,62 @@ __global__ __launch_bounds__(256) void flash_dq_kernel(
 
   f32x4 dq_acc[2][4] = {};  // [fq][fd]: rows (l>>4)*4+i, col fd*16+(l&15)
 
+  // T14 double-buffered staging of K (nat + transposed) and V (nat)
   const int kv_end = causal ? min(vl, q0 + NWAVE * TQW) : vl;
-  for (int kv0 = 0; kv0 < kv_end; kv0 += TK) {
-    {  // stage K (nat + transposed) and V (nat)
-      const int row = threadIdx.x >> 3;
-      const int off = (threadIdx.x & 7) * 16;
-      const bool kv_full = (kv0 + TK) <= vl;
+  const int srow = threadIdx.x >> 3;
+  const int soff = (threadIdx.x & 7) * 16;
+  auto dq_load = [&](int kv0, uint4v kreg[2], uint4v vreg[2]) {
+    const bool kv_full = (kv0 + TK) <= vl;
 #pragma unroll
-      for (int rr = 0; rr < TK; rr += 32) {
-        const int key = kv0 + row + rr;
-        uint4v kv = {}, vv4 = {};
-        if (kv_full) {
-          kv = *reinterpret_cast<const uint4v*>(
-              K + ((long)b * L + key) * ldkv + (long)h * 64 + off / 2);
-          vv4 = *reinterpret_cast<const uint4v*>(
-              V + ((long)b * L + key) * ldkv + (long)h * 64 + off / 2);
-        } else if (key < vl) {
-          kv = *reinterpret_cast<const uint4v*>(
-              K + ((long)b * L + key) * ldkv + (long)h * 64 + off / 2);
-          vv4 = *reinterpret_cast<const uint4v*>(
-              V + ((long)b * L + key) * ldkv + (long)h * 64 + off / 2);
-        }
-        *reinterpret_cast<uint4v*>(k_lds + fa_swz(row + rr, off)) = kv;
-        *reinterpret_cast<uint4v*>(v_lds + fa_swz(row + rr, off)) = vv4;
-        bf16 kk[8];
-        *reinterpret_cast<uint4v*>(kk) = kv;
-        const int d0 = off / 2;
-#pragma unroll
-        for (int j = 0; j < 8; ++j)
-          *reinterpret_cast<bf16*>(kt_lds + fa_swzT(d0 + j, (row + rr) * 2)) = kk[j];
+    for (int rr = 0; rr < 2; ++rr) {
+      const int key = kv0 + srow + rr * 32;
+      kreg[rr] = {};
+      vreg[rr] = {};
+      if (kv_full) {
+        kreg[rr] = *reinterpret_cast<const uint4v*>(
+            K + ((long)b * L + key) * ldkv + (long)h * 64 + soff / 2);
+        vreg[rr] = *reinterpret_cast<const uint4v*>(
+            V + ((long)b * L + key) * ldkv + (long)h * 64 + soff / 2);
+      } else if (key < vl) {
+        kreg[rr] = *reinterpret_cast<const uint4v*>(
+            K + ((long)b * L + key) * ldkv + (long)h * 64 + soff / 2);
+        vreg[rr] = *reinterpret_cast<const uint4v*>(
+            V + ((long)b * L + key) * ldkv + (long)h * 64 + soff / 2);
       }
     }
-    __syncthreads();
+  };
+  auto dq_write = [&](char* kb, char* ktb, char* vb, const uint4v kreg[2],
+                      const uint4v vreg[2]) {
+#pragma unroll
+    for (int rr = 0; rr < 2; ++rr) {
+      *reinterpret_cast<uint4v*>(kb + fa_swz(srow + rr * 32, soff)) = kreg[rr];
+      *reinterpret_cast<uint4v*>(vb + fa_swz(srow + rr * 32, soff)) = vreg[rr];
+      bf16 kk[8];
+      *reinterpret_cast<uint4v*>(kk) = kreg[rr];
+      const int d0 = soff / 2;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        *reinterpret_cast<bf16*>(ktb + fa_swzT(d0 + j, (srow + rr * 32) * 2)) =
+            kk[j];
+    }
+  };
+#define DQ_BUF(base, i) ((base) + (i)*3 * TK * 128)
+  uint4v kreg[2], vreg[2];
+  int cur = 0;
+  if (kv_end > 0) {
+    dq_load(0, kreg, vreg);
+    dq_write(DQ_BUF(k_lds, 0), DQ_BUF(kt_lds, 0), DQ_BUF(v_lds, 0), kreg, vreg);
+  }
+  __syncthreads();
+  for (int kv0 = 0; kv0 < kv_end; kv0 += TK) {
+    const bool has_next = (kv0 + TK) < kv_end;
+    if (has_next) dq_load(kv0 + TK, kreg, vreg);
+    char* k_lds_c = DQ_BUF(k_lds, cur);
+    char* kt_lds_c = DQ_BUF(kt_lds, cur);
+    char* v_lds_c = DQ_BUF(v_lds, cur);
 
     float p[4][2][4];
-    recompute_pT(k_lds, q_frag, bias, lse_w, lane, h, L, qw, kv0, vl, scale,
+    recompute_pT(k_lds_c, q_frag, bias, lse_w, lane, h, L, qw, kv0, vl, scale,
                  causal, p);
 
     // dPd^T = V dO^T (same structure as S^T), then ds
@@ -491,7 +537,7 @@ __global__ __launch_bounds__(256) void flash_dq_kernel(
       for (int fk = 0; fk < 4; ++fk) {
         const int key = fk * 16 + (lane & 15);
         const int kbyte = ks * 64 + (lane >> 4) * 16;
-        v_frag[fk] = *reinterpret_cast<const bf16x8*>(v_lds + fa_swz(key, kbyte));
+        v_frag[fk] = *reinterpret_cast<const bf16x8*>(v_lds_c + fa_swz(key, kbyte));
       }
 #pragma unroll
       for (int fk = 0; fk < 4; ++fk)
@@ -541,7 +587,7 @@ __global__ __launch_bounds__(256) void flash_dq_kernel(
       for (int fd = 0; fd < 4; ++fd) {
         const int d = fd * 16 + (lane & 15);
         const int keybyte = kp * 64 + (lane >> 4) * 16;
-        kt_frag[fd] = *reinterpret_cast<const bf16x8*>(kt_lds + fa_swzT(d, keybyte));
+        kt_frag[fd] = *reinterpret_cast<const bf16x8*>(kt_lds_c + fa_swzT(d, keybyte));
       }
 #pragma unroll
       for (int fq = 0; fq < 2; ++fq)
@@ -550,8 +596,13 @@ __global__ __launch_bounds__(256) void flash_dq_kernel(
           dq_acc[fq][fd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               dsa[fq], kt_frag[fd], dq_acc[fq][fd], 0, 0, 0);
     }
+    if (has_next)
+      dq_write(DQ_BUF(k_lds, cur ^ 1), DQ_BUF(kt_lds, cur ^ 1),
+               DQ_BUF(v_lds, cur ^ 1), kreg, vreg);
     __syncthreads();
+    cur ^= 1;
   }
+#undef DQ_BUF
 
   // write dQ (B, L, H*64)
 #pragma unroll
@@ -991,7 +1042,7 @@ void launch_flash_fwd(const bf16* Q, const bf16* K, const bf16* V,
                       unsigned p8, unsigned long long seed, long ldq,
                       long ldkv, hipStream_t stream) {
   const dim3 grid((L + NWAVE * TQW - 1) / (NWAVE * TQW), B * H);
-  const size_t lds = 2 * TK * 128;
+  const size_t lds = 2 * 2 * TK * 128;  // double-buffered K+V tiles
   hipLaunchKernelGGL(flash_fwd_kernel, grid, dim3(256), lds, stream, Q, K, V,
                      valid, bias, O, lse, B, H, L, scale, causal, p8, seed,
                      ldq, ldkv);
@@ -1013,7 +1064,7 @@ void launch_flash_dq(const bf16* Q, const bf16* K, const bf16* V,
                      int causal, unsigned p8, unsigned long long seed,
                      long ldq, long ldkv, hipStream_t stream) {
   const dim3 grid((L + NWAVE * TQW - 1) / (NWAVE * TQW), B * H);
-  const size_t lds = 3 * TK * 128;
+  const size_t lds = 2 * 3 * TK * 128;  // double-buffered K/K^T/V tiles
   hipLaunchKernelGGL(flash_dq_kernel, grid, dim3(256), lds, stream, Q, K, V,
                      dO, valid, bias, lse, Dterm, dQ, dBias, B, H, L, scale,
                      causal, p8, seed, ldq, ldkv);
