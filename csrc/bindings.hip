@@ -67,7 +67,7 @@ void launch_flash_dterm(const __hip_bfloat16*, const __hip_bfloat16*, float*, in
 void launch_flash_dq(const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*,
                      const __hip_bfloat16*, const int*, const float*, const float*, const float*,
                      __hip_bfloat16*, float*, int, int, int, float, int, unsigned,
-                     unsigned long long, long, long, hipStream_t);
+                     unsigned long long, long, long, long, hipStream_t);
 void launch_wgrad2(const __hip_bfloat16*, const __hip_bfloat16*, float*, int, int, int,
                    hipStream_t);
 void launch_adamw_fused(float*, const float*, float*, float*, long, float, float, float, float,
@@ -77,7 +77,7 @@ void launch_gemm2(const __hip_bfloat16*, const __hip_bfloat16*, const float*,
 void launch_flash_dkv(const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*,
                       const __hip_bfloat16*, const int*, const float*, const float*, const float*,
                       __hip_bfloat16*, __hip_bfloat16*, int, int, int, float, int, unsigned,
-                      unsigned long long, long, long, hipStream_t);
+                      unsigned long long, long, long, long, hipStream_t);
 template <typename T>
 void launch_softmax_mask_bwd(const T*, const T*, T*, long, int, float, float,
                              unsigned long long, hipStream_t);
@@ -692,7 +692,8 @@ std::vector<at::Tensor> flash_attn_bwd(at::Tensor dO, at::Tensor Q, at::Tensor K
                                        at::Tensor O, at::Tensor lse, int64_t H,
                                        c10::optional<at::Tensor> valid,
                                        c10::optional<at::Tensor> bias, double scale, bool causal,
-                                       double dropout_p, int64_t seed, bool need_dbias) {
+                                       double dropout_p, int64_t seed, bool need_dbias,
+                                       bool fused_grads) {
   CHECK_GPU(dO);
   const int B = Q.size(0), L = Q.size(1);
   const long ldq = fa_ld(Q, L);
@@ -701,10 +702,25 @@ std::vector<at::Tensor> flash_attn_bwd(at::Tensor dO, at::Tensor Q, at::Tensor K
   auto stream = cur_stream();
   auto Dterm = at::empty({B, H, L}, Q.options().dtype(at::kFloat));
   launch_flash_dterm(ptr<bf16_t>(dO), ptr<bf16_t>(O), Dterm.data_ptr<float>(), B, H, L, stream);
-  // grads are allocated CONTIGUOUS regardless of the input views
-  auto dQ = at::empty({B, L, (long)H * 64}, Q.options());
-  auto dK = at::empty({B, L, (long)H * 64}, Q.options());
-  auto dV = at::empty({B, L, (long)H * 64}, Q.options());
+  // fused_grads: one (B, L, 3*H*64) buffer whose [q|k|v] slices the
+  // kernels write directly (ld 3*H*64) — lets the attention op consume a
+  // fused QKV projection with zero slice-backward scatter work
+  const long HD64 = (long)H * 64;
+  at::Tensor dQKV, dQ, dK, dV;
+  long ldout = HD64;
+  if (fused_grads) {
+    TORCH_CHECK(ldq == 3 * HD64 && ldkv == 3 * HD64,
+                "fused_grads expects q/k/v slices of one (B, L, 3D) buffer");
+    dQKV = at::empty({B, L, 3 * HD64}, Q.options());
+    ldout = 3 * HD64;
+    dQ = dQKV.narrow(2, 0, HD64);
+    dK = dQKV.narrow(2, HD64, HD64);
+    dV = dQKV.narrow(2, 2 * HD64, HD64);
+  } else {
+    dQ = at::empty({B, L, HD64}, Q.options());
+    dK = at::empty({B, L, HD64}, Q.options());
+    dV = at::empty({B, L, HD64}, Q.options());
+  }
   const float* bptr = nullptr;
   if (bias.has_value()) bptr = bias->data_ptr<float>();
   at::Tensor dBias;
@@ -717,11 +733,17 @@ std::vector<at::Tensor> flash_attn_bwd(at::Tensor dO, at::Tensor Q, at::Tensor K
   launch_flash_dq(ptr<bf16_t>(Q), ptr<bf16_t>(K), ptr<bf16_t>(V), ptr<bf16_t>(dO),
                   opt_valid_ptr(valid), bptr, lse.data_ptr<float>(), Dterm.data_ptr<float>(),
                   mptr<bf16_t>(dQ), dbias_ptr, B, H, L, (float)scale, causal ? 1 : 0,
-                  (unsigned)(dropout_p * 256.0), (unsigned long long)seed, ldq, ldkv, stream);
+                  (unsigned)(dropout_p * 256.0), (unsigned long long)seed, ldq, ldkv, ldout,
+                  stream);
   launch_flash_dkv(ptr<bf16_t>(Q), ptr<bf16_t>(K), ptr<bf16_t>(V), ptr<bf16_t>(dO),
                    opt_valid_ptr(valid), bptr, lse.data_ptr<float>(), Dterm.data_ptr<float>(),
-                   mptr<bf16_t>(dK), mptr<bf16_t>(dV), B, H, L, (float)scale, causal ? 1 : 0,
-                   (unsigned)(dropout_p * 256.0), (unsigned long long)seed, ldq, ldkv, stream);
+                   mptr<bf16_t>(dK), mptr<bf16_t>(dV), B, H, L, (float)scale,
+                   causal ? 1 : 0, (unsigned)(dropout_p * 256.0), (unsigned long long)seed,
+                   ldq, ldkv, ldout, stream);
+  if (fused_grads) {
+    if (need_dbias) return {dQKV, dBias};
+    return {dQKV};
+  }
   if (need_dbias) return {dQ, dK, dV, dBias};
   return {dQ, dK, dV};
 }

@@ -422,7 +422,7 @@ __global__ __launch_bounds__(256) void flash_dq_kernel(
     const float* __restrict__ lse, const float* __restrict__ Dterm,
     bf16* __restrict__ dQ, float* __restrict__ dBias, int B, int H, int L,
     float scale, int causal, unsigned p8, unsigned long long seed, long ldq,
-    long ldkv) {
+    long ldkv, long ldout) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   char* k_lds = smem;                 // [key][d] swizzled, 8 KiB
   char* kt_lds = smem + TK * 128;     // [d][key] swizzled, 8 KiB
@@ -614,7 +614,7 @@ __global__ __launch_bounds__(256) void flash_dq_kernel(
 #pragma unroll
       for (int fd = 0; fd < 4; ++fd) {
         const int d = fd * 16 + (lane & 15);
-        dQ[((long)b * L + qrow) * HD + (long)h * 64 + d] =
+        dQ[((long)b * L + qrow) * ldout + (long)h * 64 + d] =
             __float2bfloat16(dq_acc[fq][fd][i]);
       }
     }
@@ -683,7 +683,7 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
     const float* __restrict__ lse, const float* __restrict__ Dterm,
     bf16* __restrict__ dK, bf16* __restrict__ dV, int B, int H, int L,
     float scale, int causal, unsigned p8, unsigned long long seed, long ldq,
-    long ldkv) {
+    long ldkv, long ldout) {
   // Wave grid 2 (key halves) x 2 (q interleave): halves the per-wave
   // accumulator footprint (the 64-key variant needed 128 fp32 accumulators
   // on top of ~176 VGPRs -> 1 wave/SIMD on the unified register file).
@@ -993,7 +993,7 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
       _Pragma("unroll") for (int i = 0; i < 4; ++i) {                         \
         const int key = kv0 + key_loc + i;                                    \
         if (key < L)                                                          \
-          OUT[((long)b * L + key) * HD + (long)h * 64 + d] =                  \
+          OUT[((long)b * L + key) * ldout + (long)h * 64 + d] =               \
               __float2bfloat16(ACC[fk][fd][i] + other[i]);                    \
       }                                                                       \
     }                                                                         \
@@ -1022,7 +1022,7 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
     const int key = kv0 + i / 64;                                             \
     const int d = i % 64;                                                     \
     if (key < L)                                                              \
-      OUT[((long)b * L + key) * HD + (long)h * 64 + d] =                      \
+      OUT[((long)b * L + key) * ldout + (long)h * 64 + d] =                   \
           __float2bfloat16(red[i]);                                           \
   }
   FA_REDUCE_STORE(dv_acc, dV)
@@ -1062,12 +1062,12 @@ void launch_flash_dq(const bf16* Q, const bf16* K, const bf16* V,
                      const float* lse, const float* Dterm, bf16* dQ,
                      float* dBias, int B, int H, int L, float scale,
                      int causal, unsigned p8, unsigned long long seed,
-                     long ldq, long ldkv, hipStream_t stream) {
+                     long ldq, long ldkv, long ldout, hipStream_t stream) {
   const dim3 grid((L + NWAVE * TQW - 1) / (NWAVE * TQW), B * H);
   const size_t lds = 2 * 3 * TK * 128;  // double-buffered K/K^T/V tiles
   hipLaunchKernelGGL(flash_dq_kernel, grid, dim3(256), lds, stream, Q, K, V,
                      dO, valid, bias, lse, Dterm, dQ, dBias, B, H, L, scale,
-                     causal, p8, seed, ldq, ldkv);
+                     causal, p8, seed, ldq, ldkv, ldout);
 }
 
 void launch_flash_dkv(const bf16* Q, const bf16* K, const bf16* V,
@@ -1075,7 +1075,7 @@ void launch_flash_dkv(const bf16* Q, const bf16* K, const bf16* V,
                       const float* lse, const float* Dterm, bf16* dK,
                       bf16* dV, int B, int H, int L, float scale, int causal,
                       unsigned p8, unsigned long long seed, long ldq,
-                      long ldkv, hipStream_t stream) {
+                      long ldkv, long ldout, hipStream_t stream) {
   const dim3 grid((L + TK - 1) / TK, B * H);
   size_t lds = 2 * TK * 128 + 18432;  // K,V + pd/ds bounces (72-B rows)
   const char* e = getenv("DFA_DKV_VARIANT");  // re-read: lets one probe
@@ -1084,35 +1084,35 @@ void launch_flash_dkv(const bf16* Q, const bf16* K, const bf16* V,
     lds += 4 * 2 * DKV_T_BYTES;
     hipLaunchKernelGGL(flash_dkv_kernel<2>, grid, dim3(256), lds, stream, Q, K,
                        V, dO, valid, bias, lse, Dterm, dK, dV, B, H, L, scale,
-                       causal, p8, seed, ldq, ldkv);
+                       causal, p8, seed, ldq, ldkv, ldout);
   } else if (var == 4) {
     // K,V + max(pd/ds region, 64x272 pair image)
     lds = 2 * TK * 128 + 18432;
     hipLaunchKernelGGL(flash_dkv_kernel<4>, grid, dim3(256), lds, stream, Q, K,
                        V, dO, valid, bias, lse, Dterm, dK, dV, B, H, L, scale,
-                       causal, p8, seed, ldq, ldkv);
+                       causal, p8, seed, ldq, ldkv, ldout);
   } else if (var == 3) {
     hipLaunchKernelGGL(flash_dkv_kernel<3>, grid, dim3(256), lds, stream, Q, K,
                        V, dO, valid, bias, lse, Dterm, dK, dV, B, H, L, scale,
-                       causal, p8, seed, ldq, ldkv);
+                       causal, p8, seed, ldq, ldkv, ldout);
   } else if (var == 5) {
     lds += 4 * 2 * DKV_T_BYTES;  // bounce staging + pair-reduce epilogue
     hipLaunchKernelGGL(flash_dkv_kernel<5>, grid, dim3(256), lds, stream, Q, K,
                        V, dO, valid, bias, lse, Dterm, dK, dV, B, H, L, scale,
-                       causal, p8, seed, ldq, ldkv);
+                       causal, p8, seed, ldq, ldkv, ldout);
   } else if (var == 6) {
     lds += 4 * 2 * DKV_T_BYTES;  // subtiled tr_b16 bounce + pair-reduce
     hipLaunchKernelGGL(flash_dkv_kernel<6>, grid, dim3(256), lds, stream, Q, K,
                        V, dO, valid, bias, lse, Dterm, dK, dV, B, H, L, scale,
-                       causal, p8, seed, ldq, ldkv);
+                       causal, p8, seed, ldq, ldkv, ldout);
   } else if (var == 9) {
     hipLaunchKernelGGL(flash_dkv_kernel<9>, grid, dim3(256), lds, stream, Q, K,
                        V, dO, valid, bias, lse, Dterm, dK, dV, B, H, L, scale,
-                       causal, p8, seed, ldq, ldkv);
+                       causal, p8, seed, ldq, ldkv, ldout);
   } else {
     hipLaunchKernelGGL(flash_dkv_kernel<0>, grid, dim3(256), lds, stream, Q, K,
                        V, dO, valid, bias, lse, Dterm, dK, dV, B, H, L, scale,
-                       causal, p8, seed, ldq, ldkv);
+                       causal, p8, seed, ldq, ldkv, ldout);
   }
 }
 

@@ -126,11 +126,14 @@ class RobertaSelfAttention(nn.Module):
             qkv = fused_qkv(x, self.query.weight, self.key.weight, self.value.weight,
                             self.query.bias, self.key.bias, self.value.bias)
         if qkv is not None:
-            q, k, v = qkv[..., :D], qkv[..., D:2 * D], qkv[..., 2 * D:]
-        else:
-            q = fused_linear(x, self.query.weight, self.query.bias)
-            k = fused_linear(x, self.key.weight, self.key.bias)
-            v = fused_linear(x, self.value.weight, self.value.bias)
+            from ..ops.transformer import flash_attention_qkv
+
+            out = flash_attention_qkv(qkv, H, valid=valid,
+                                      scale=1.0 / math.sqrt(d), dropout_p=p)
+            return out, None
+        q = fused_linear(x, self.query.weight, self.query.bias)
+        k = fused_linear(x, self.key.weight, self.key.bias)
+        v = fused_linear(x, self.value.weight, self.value.bias)
         if d == 64 and not output_attentions and flash_usable(q, L):
             out = flash_attention(q, k, v, H, valid=valid, scale=1.0 / math.sqrt(d),
                                   dropout_p=p)

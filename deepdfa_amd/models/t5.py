@@ -147,16 +147,19 @@ class T5Attention(nn.Module):
             from ..ops.transformer import fused_qkv
 
             qkv = fused_qkv(x, self.q.weight, self.k.weight, self.v.weight)
-        if qkv is not None:
-            inner = H * d
-            qp = qkv[..., :inner]
-            kp = qkv[..., inner:2 * inner]
-            vp = qkv[..., 2 * inner:]
-        else:
-            qp = fused_linear(x, self.q.weight)
-            kp = fused_linear(src, self.k.weight)
-            vp = fused_linear(src, self.v.weight)
         causal = self.causal and kv is None
+        if qkv is not None:
+            from ..ops.transformer import flash_attention_qkv
+
+            bias = None
+            if position_bias is not None:
+                bias = position_bias.squeeze(0).float().contiguous()
+            out = flash_attention_qkv(qkv, H, valid=valid, bias=bias, scale=1.0,
+                                      causal=causal, dropout_p=dropout_p)
+            return fused_linear(out, self.o.weight)
+        qp = fused_linear(x, self.q.weight)
+        kp = fused_linear(src, self.k.weight)
+        vp = fused_linear(src, self.v.weight)
         if d == 64 and flash_usable(qp, Lq, Lk):
             bias = None
             if position_bias is not None:

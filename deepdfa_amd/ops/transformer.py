@@ -243,7 +243,7 @@ class _FlashAttention(torch.autograd.Function):
         H, valid, bias, scale, causal, dropout_p, seed = ctx.meta
         outs = ext.flash_attn_bwd(
             dO.contiguous(), q, k, v, O, lse,
-            H, valid, bias, scale, causal, dropout_p, seed, ctx.need_dbias,
+            H, valid, bias, scale, causal, dropout_p, seed, ctx.need_dbias, False,
         )
         dbias = outs[3] if ctx.need_dbias else None
         return outs[0], outs[1], outs[2], None, None, dbias, None, None, None
@@ -462,3 +462,43 @@ def dropout_add(h, res, p=0.0, training=True):
         return _DropoutAdd.apply(h, res, p)
     z = torch.nn.functional.dropout(h, p) if p > 0 else h
     return res + z
+
+
+class _FlashAttentionQKV(torch.autograd.Function):
+    """Flash attention taking the FUSED (B, L, 3D) QKV projection as one
+    input: forward slices q/k/v as strided views (the kernels take row
+    strides), backward writes dq/dk/dv into ONE (B, L, 3D) buffer — no
+    slice-backward zero+scatter work in autograd."""
+
+    @staticmethod
+    def forward(ctx, qkv, H, valid, bias, scale, causal, dropout_p):
+        ext = load_ext(required=True)
+        D = qkv.shape[-1] // 3
+        q, k, v = qkv[..., :D], qkv[..., D:2 * D], qkv[..., 2 * D:]
+        seed = _next_seed() if dropout_p > 0 else 0
+        O, lse = ext.flash_attn_fwd(q, k, v, H, valid, bias, scale, causal,
+                                    dropout_p, seed)
+        ctx.save_for_backward(qkv, O, lse)
+        ctx.meta = (H, valid, bias, scale, causal, dropout_p, seed)
+        ctx.need_dbias = bias is not None and ctx.needs_input_grad[3]
+        return O
+
+    @staticmethod
+    def backward(ctx, dO):
+        ext = load_ext(required=True)
+        qkv, O, lse = ctx.saved_tensors
+        H, valid, bias, scale, causal, dropout_p, seed = ctx.meta
+        D = qkv.shape[-1] // 3
+        q, k, v = qkv[..., :D], qkv[..., D:2 * D], qkv[..., 2 * D:]
+        outs = ext.flash_attn_bwd(
+            dO.contiguous(), q, k, v, O, lse,
+            H, valid, bias, scale, causal, dropout_p, seed, ctx.need_dbias, True,
+        )
+        dbias = outs[1] if ctx.need_dbias else None
+        return outs[0], None, None, dbias, None, None, None
+
+
+def flash_attention_qkv(qkv, num_heads, valid=None, bias=None, scale=1.0,
+                        causal=False, dropout_p=0.0):
+    return _FlashAttentionQKV.apply(qkv, num_heads, valid, bias, scale, causal,
+                                    dropout_p)
