@@ -34,7 +34,7 @@ VARS = ["4", "6"]
 
 def run_bwd():
     return ext.flash_attn_bwd(dO, q, k, v, O, lse, H, valid, None, 0.125,
-                              False, 0.0, 0, False)
+                              False, 0.0, 0, False, False)
 
 
 # numerics: each variant vs variant 0

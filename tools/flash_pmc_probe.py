@@ -31,6 +31,6 @@ valid = torch.full((B,), L, dtype=torch.int32, device=dev)
 O, lse = ext.flash_attn_fwd(q, k, v, H, valid, None, 0.125, False, 0.0, 0)
 torch.cuda.synchronize()
 for _ in range(30):
-    ext.flash_attn_bwd(dO, q, k, v, O, lse, H, valid, None, 0.125, False, 0.0, 0, False)
+    ext.flash_attn_bwd(dO, q, k, v, O, lse, H, valid, None, 0.125, False, 0.0, 0, False, False)
 torch.cuda.synchronize()
 print("done")

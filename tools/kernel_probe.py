@@ -54,7 +54,7 @@ variants = {
     "matmul fwd bf16 (hipblaslt)": lambda: torch.matmul(x, w.t()),
     "flash fwd": lambda: ext.flash_attn_fwd(q, k, v, H, valid, None, 0.125, False, 0.0, 0),
     "flash bwd (dterm+dq+dkv)": lambda: ext.flash_attn_bwd(
-        dO, q, k, v, O, lse, H, valid, None, 0.125, False, 0.0, 0, False),
+        dO, q, k, v, O, lse, H, valid, None, 0.125, False, 0.0, 0, False, False),
     "bias_gelu fwd (8192x3072)": lambda: ext.bias_gelu_fwd(xff, bff),
     "bias_gelu bwd": lambda: ext.bias_gelu_bwd(dyff, xff, bff),
     "colsum (8192x768)": lambda: ext.colsum(dy),
