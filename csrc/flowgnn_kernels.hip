@@ -536,9 +536,11 @@ void launch_gru_gates2_bwd(const T* grad_h_new, const T* h, const T* r,
 
 template <typename T>
 void launch_colsum(const T* x, float* out, int N, int C, hipStream_t stream) {
-  const int block = 256;
+  // narrow C (the GNN's 128-column grads): a 256-thread block would idle
+  // half its lanes; shrink the block and the row chunk together
+  const int block = (C < 256) ? ((C < 64) ? 64 : C) : 256;
   const int colb = (C + block - 1) / block;
-  const int rows_per_block = 64;
+  const int rows_per_block = (C < 256) ? 32 : 64;
   const int rowb = (N + rows_per_block - 1) / rows_per_block;
   if (N > 0 && C > 0)
     hipLaunchKernelGGL(colsum_kernel<T>, dim3(colb, rowb), dim3(block), 0,
