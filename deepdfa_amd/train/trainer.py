@@ -45,7 +45,11 @@ class Trainer:
         log_every_n_steps: int = 50,
         seed: int = 0,
         precision: str = "bf16",  # "bf16" (autocast on GPU) or "fp32"
+        detect_anomaly: bool = False,
     ):
+        # reference config_default.yaml:37 trainer.detect_anomaly — autograd
+        # NaN/inf detection; the kernel-level analog is scripts/run_sanitize.sh
+        self.detect_anomaly = detect_anomaly
         self.precision = precision
         self.max_epochs = max_epochs
         self.root = default_root_dir
@@ -104,6 +108,8 @@ class Trainer:
         from ..utils.logging import ScalarLogger
 
         torch.manual_seed(self.seed)
+        if self.detect_anomaly:
+            torch.autograd.set_detect_anomaly(True)
         model = model.to(self.device)
         if optimizer is None:
             optimizer = torch.optim.Adam(model.parameters(), lr=1e-3, weight_decay=1e-2)
