@@ -92,8 +92,8 @@ void launch_ln_res_dropout_fwd(const T*, const T*, const float*, const float*, T
                                float*, long, int, float, float, unsigned long long, hipStream_t);
 template <typename T>
 void launch_ln_res_dropout_bwd(const T*, const T*, const T*, const float*, const float*,
-                               const float*, T*, T*, long, int, float, unsigned long long,
-                               hipStream_t);
+                               const float*, T*, T*, T*, long, int, float,
+                               unsigned long long, hipStream_t);
 template <typename T>
 void launch_ln_res_dropout_wgrad(const T*, const T*, const T*, const float*, const float*,
                                  float*, float*, long, int, float, unsigned long long,
@@ -865,14 +865,16 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     const long N = dy.numel() / D;
     auto dz = at::empty_like(dy);
     auto dh = at::empty_like(dy);
+    auto z = at::empty_like(dy);
     dispatch_float_bf16(dy, "ln_res_dropout_bwd", [&](auto tag) {
       using T = decltype(tag);
       launch_ln_res_dropout_bwd<T>(ptr<T>(dy), ptr<T>(h), ptr<T>(res),
                                    gamma.data_ptr<float>(), mean.data_ptr<float>(),
-                                   rstd.data_ptr<float>(), mptr<T>(dz), mptr<T>(dh), N, D,
-                                   (float)p, (unsigned long long)seed, cur_stream());
+                                   rstd.data_ptr<float>(), mptr<T>(dz), mptr<T>(dh),
+                                   mptr<T>(z), N, D, (float)p, (unsigned long long)seed,
+                                   cur_stream());
     });
-    return std::vector<at::Tensor>{dz, dh};
+    return std::vector<at::Tensor>{dz, dh, z};
   });
   m.def("ln_res_dropout_wgrad", [](at::Tensor dy, at::Tensor h, at::Tensor res,
                                    at::Tensor mean, at::Tensor rstd, double p,

@@ -848,8 +848,10 @@ __global__ void ln_res_dropout_bwd_kernel(
     const T* __restrict__ dy, const T* __restrict__ h,
     const T* __restrict__ res, const float* __restrict__ gamma,
     const float* __restrict__ mean, const float* __restrict__ rstd,
-    T* __restrict__ dz, T* __restrict__ dh, long N, int D, unsigned p8,
-    unsigned long long seed, float dscale) {
+    T* __restrict__ dz, T* __restrict__ dh, T* __restrict__ zout, long N,
+    int D, unsigned p8, unsigned long long seed, float dscale) {
+  // zout: reconstructed z = dropout(h)+res, written once here so the
+  // column-reduction wgrad can run the plain (hash-free) LN wgrad kernel
   const long row = (long)(blockIdx.x * (blockDim.x / WAVE)) + threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
   if (row >= N) return;
@@ -876,12 +878,13 @@ __global__ void ln_res_dropout_bwd_kernel(
   c2 = wave_sum(c2) / D;
   T* dzr = dz + row * D;
   T* dhr = dh + row * D;
+  T* zr = zout + row * D;
   for (int d = lane * 4; d < D; d += WAVE * 4) {
     const f4 a = load4(hr + d);
     const f4 b = load4(rr + d);
     const f4 dyv = load4(dyr + d);
     const f4 g4 = load4(gamma + d);
-    f4 oz, oh;
+    f4 oz, oh, ozv;
 #pragma unroll
     for (int u = 0; u < 4; ++u) {
       const bool keep = (p8 == 0) || keep_mask(row * D + d + u, seed, p8);
@@ -890,9 +893,11 @@ __global__ void ln_res_dropout_bwd_kernel(
       const float dzv = rs * (dyv.v[u] * g4.v[u] - c1 - xh * c2);
       oz.v[u] = dzv;
       oh.v[u] = keep ? dzv * dscale : 0.f;
+      ozv.v[u] = z;
     }
     store4(dzr + d, oz);
     store4(dhr + d, oh);
+    store4(zr + d, ozv);
   }
 }
 
@@ -949,16 +954,17 @@ void launch_ln_res_dropout_fwd(const T* h, const T* res, const float* gamma,
 template <typename T>
 void launch_ln_res_dropout_bwd(const T* dy, const T* h, const T* res,
                                const float* gamma, const float* mean,
-                               const float* rstd, T* dz, T* dh, long N, int D,
-                               float dropout_p, unsigned long long seed,
-                               hipStream_t stream) {
+                               const float* rstd, T* dz, T* dh, T* zout,
+                               long N, int D, float dropout_p,
+                               unsigned long long seed, hipStream_t stream) {
   const unsigned p8 = (unsigned)(dropout_p * 256.0f);
   const float dscale = (p8 > 0) ? 256.0f / (256.0f - p8) : 1.0f;
   const int grid = (int)((N + ROWS_PER_BLOCK - 1) / ROWS_PER_BLOCK);
   if (grid)
     hipLaunchKernelGGL(ln_res_dropout_bwd_kernel<T>, dim3(grid),
                        dim3(WAVE * ROWS_PER_BLOCK), 0, stream, dy, h, res,
-                       gamma, mean, rstd, dz, dh, N, D, p8, seed, dscale);
+                       gamma, mean, rstd, dz, dh, zout, N, D, p8, seed,
+                       dscale);
 }
 
 template <typename T>
@@ -986,8 +992,8 @@ void launch_ln_res_dropout_wgrad(const T* dy, const T* h, const T* res,
                                              unsigned long long, hipStream_t);\
   template void launch_ln_res_dropout_bwd<T>(const T*, const T*, const T*,    \
                                              const float*, const float*,      \
-                                             const float*, T*, T*, long, int, \
-                                             float, unsigned long long,       \
+                                             const float*, T*, T*, T*, long,  \
+                                             int, float, unsigned long long,  \
                                              hipStream_t);                    \
   template void launch_ln_res_dropout_wgrad<T>(                               \
       const T*, const T*, const T*, const float*, const float*, float*,       \

@@ -351,8 +351,10 @@ class _LayerNormResDropout(torch.autograd.Function):
         h, res, wf, mean, rstd = ctx.saved_tensors
         p, seed = ctx.meta
         dy = dy.contiguous()
-        dz, dh = ext.ln_res_dropout_bwd(dy, h, res, wf, mean, rstd, p, seed)
-        dgamma, dbeta = ext.ln_res_dropout_wgrad(dy, h, res, mean, rstd, p, seed)
+        # bwd also reconstructs z = dropout(h)+res once, so the column
+        # reduction runs the plain (hash-free) LN wgrad kernel
+        dz, dh, z = ext.ln_res_dropout_bwd(dy, h, res, wf, mean, rstd, p, seed)
+        dgamma, dbeta = ext.layernorm_wgrad(dy, z, mean, rstd)
         return dh, dz, dgamma, dbeta, None, None
 
 
