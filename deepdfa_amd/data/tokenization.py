@@ -21,7 +21,7 @@ encode(text, max_length) -> List[int].
 from __future__ import annotations
 
 import re
-from typing import List, Optional
+from typing import List
 
 from .. import hashstr
 

@@ -11,7 +11,6 @@ keywords/ directory semantics) with 4x weight on keywords.
 from __future__ import annotations
 
 import math
-from collections import Counter
 from typing import Dict, List, Sequence
 
 from .bleu import _ngrams, bleu

@@ -18,7 +18,6 @@ embedding dim and GGNN width are both 128, matching the reference's
 
 from __future__ import annotations
 
-import math
 from typing import Dict, Optional
 
 import torch

@@ -10,7 +10,7 @@ is a custom kernel.
 
 from __future__ import annotations
 
-from typing import List, Optional
+from typing import Optional
 
 import torch
 

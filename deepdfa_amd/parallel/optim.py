@@ -13,7 +13,6 @@ model here (~224M params => ~3.6 GB of param+grad+state).
 
 from __future__ import annotations
 
-import math
 from typing import Iterable, Optional
 
 import torch

@@ -19,7 +19,6 @@ from __future__ import annotations
 import argparse
 import logging
 import os
-import pickle
 from typing import Dict, Set
 
 from ..data.dclass import ds_partition, synthetic_bigvul_df

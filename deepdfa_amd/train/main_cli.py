@@ -23,7 +23,6 @@ import argparse
 import json
 import logging
 import os
-import sys
 import time
 import traceback
 from typing import Dict, List, Optional

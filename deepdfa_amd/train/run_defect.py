@@ -13,7 +13,6 @@ import argparse
 import logging
 import os
 
-import numpy as np
 import torch
 from torch.utils.data import DataLoader, DistributedSampler, RandomSampler, SequentialSampler
 

@@ -16,7 +16,7 @@ import os
 import torch
 from torch.utils.data import DataLoader, Dataset, RandomSampler
 
-from ..data.text_dataset import TextDataset, synthetic_func_source
+from ..data.text_dataset import synthetic_func_source
 from ..data.tokenization import HashTokenizer
 from ..evaluator import smoothed_bleu4
 from ..models.t5 import T5Config, T5ForConditionalGeneration
