@@ -154,3 +154,33 @@ def drop_lone_nodes(cpg: CPG) -> CPG:
         touched.add(s)
         touched.add(d)
     return CPG({i: p for i, p in cpg.nodes.items() if i in touched}, list(cpg.edges))
+
+
+def group_nodes_by_line(cpg: CPG) -> CPG:
+    """Statement-level grouping (reference sastvd/linevd/utils.py:6-22
+    ne_groupnodes): collapse all CPG nodes sharing a lineNumber into one
+    statement node (the first), re-target edges to line numbers, dedupe,
+    and drop nodes with no line. Node ids in the result ARE line numbers.
+    The real-Joern path runs this before CFG extraction; the synthetic
+    generator already emits one node per statement."""
+    line_of = {}
+    rep: Dict[int, Dict] = {}
+    for nid in sorted(cpg.nodes):
+        props = cpg.nodes[nid]
+        ln = props.get("lineNumber", -1)
+        if ln is None or ln < 0:
+            continue
+        line_of[nid] = ln
+        if ln not in rep:
+            rep[ln] = dict(props)
+    edges = []
+    seen = set()
+    for s, d, t in cpg.edges:
+        if s not in line_of or d not in line_of:
+            continue
+        e = (line_of[s], line_of[d], t)
+        if e not in seen and e[0] != e[1]:
+            seen.add(e)
+            edges.append(e)
+    touched = {x for s, d, _ in edges for x in (s, d)}
+    return CPG({ln: p for ln, p in rep.items() if ln in touched}, edges)

@@ -176,3 +176,21 @@ def test_tokenise_lines():
 
     out = tokenise_lines("int fooBar = 1;\nreturn fooBar;")
     assert out == [["int", "foo", "Bar", "1"], ["return", "foo", "Bar"]]
+
+
+def test_group_nodes_by_line():
+    from deepdfa_amd.analysis.dataflow import CPG
+    from deepdfa_amd.pipeline.cpg import group_nodes_by_line
+
+    nodes = {
+        1: {"_label": "CALL", "lineNumber": 10, "code": "x = a"},
+        2: {"_label": "IDENTIFIER", "lineNumber": 10, "code": "x"},
+        3: {"_label": "CALL", "lineNumber": 11, "code": "y = x"},
+        4: {"_label": "CALL", "lineNumber": -1, "code": "noline"},
+    }
+    edges = [(1, 2, "AST"), (1, 3, "CFG"), (2, 3, "CFG"), (1, 4, "CFG")]
+    g = group_nodes_by_line(CPG(nodes, edges))
+    # lines 10 and 11 remain; intra-line edge dropped; duplicate collapsed
+    assert set(g.nodes) == {10, 11}
+    assert g.edges == [(10, 11, "CFG")]
+    assert g.nodes[10]["code"] == "x = a"  # first node per line wins
