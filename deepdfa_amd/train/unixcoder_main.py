@@ -151,6 +151,25 @@ def recall_at_topk_loc(line_scores: List[Tuple[List[float], List[int]]], k_perce
     return found / max(1, total)
 
 
+def ifa(line_scores: List[Tuple[List[float], List[int]]]) -> float:
+    """Initial False Alarm (reference unixcoder linevul_main.py:676
+    min_clean_lines_inspected): per example, the number of CLEAN lines
+    ranked above the highest-ranked flaw line; mean over examples."""
+    vals = []
+    for scores, flaw in line_scores:
+        if not flaw:
+            continue
+        order = sorted(range(len(scores)), key=lambda i: -scores[i])
+        flaw_set = set(flaw)
+        clean_seen = 0
+        for i in order:
+            if i in flaw_set:
+                break
+            clean_seen += 1
+        vals.append(clean_seen)
+    return float(sum(vals) / len(vals)) if vals else 0.0
+
+
 def top_k_accuracy(line_scores, k: int = 10) -> float:
     """Fraction of examples whose top-k ranked lines hit >= 1 flaw line."""
     hit, n = 0, 0
@@ -280,6 +299,12 @@ def main(argv=None):
         results["effort@topk"] = effort_at_topk(line_results, args.effort_at_top_k)
         results["recall@topk_loc"] = recall_at_topk_loc(line_results, args.top_k_recall_loc)
         results["top_k_accuracy"] = top_k_accuracy(line_results, args.top_k_constant)
+        results["ifa"] = ifa(line_results)
+        # reference :700-701 writes per-method IFA records
+        rec_dir = os.path.join(args.output_dir, "ifa_records")
+        os.makedirs(rec_dir, exist_ok=True)
+        with open(os.path.join(rec_dir, f"ifa_{args.reasoning_method}.txt"), "w") as f:
+            f.write(str([results["ifa"]]))
     return results
 
 

@@ -90,3 +90,14 @@ def test_dbgbench_family_and_holdout_eval(tmp_path):
         "--eval_batch_size", "8", "--output_dir", str(tmp_path / "uxc"),
     ])
     assert "test" in res
+
+
+def test_ifa_metric():
+    from deepdfa_amd.train.unixcoder_main import ifa
+
+    # flaw line ranked first -> 0 clean lines inspected
+    assert ifa([([0.9, 0.1, 0.2], [0])]) == 0.0
+    # two clean lines rank above the only flaw line
+    assert ifa([([0.9, 0.8, 0.1], [2])]) == 2.0
+    # mean over examples; no-flaw examples skipped
+    assert ifa([([0.9, 0.1], [1]), ([0.5, 0.9], [1]), ([0.3], [])]) == 0.5
