@@ -206,6 +206,10 @@ def build_args(argv=None):
     p.add_argument("--time", action="store_true")
     p.add_argument("--n_synthetic", type=int, default=2000)
     p.add_argument("--num_layers", type=int, default=12, help="encoder depth (tests use small)")
+    p.add_argument("--split", default="fixed",
+                   choices=["fixed", "random", "cross_project"],
+                   help="partitioning scheme (cross_project = the reference's "
+                        "cross-project generalization experiments)")
     p.add_argument("--train_data_file", default=None)
     p.add_argument("--eval_data_file", default=None)
     p.add_argument("--test_data_file", default=None)
@@ -297,7 +301,8 @@ def main(argv=None):
 
     datasets = {
         part: TextDataset(
-            tokenizer, args, partition=part, block_size=args.block_size, n_synthetic=n_syn
+            tokenizer, args, partition=part, block_size=args.block_size,
+            n_synthetic=n_syn, split=args.split,
         )
         for part in ("train", "val", "test")
     }

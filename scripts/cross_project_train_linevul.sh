@@ -1,0 +1,12 @@
+#!/bin/bash
+# Cross-project LineVul training (reference scripts/cross_project_train_linevul.sh:
+# train on one project subset, generalize to held-out projects). Here the
+# project split comes from the dataset's project column (--split cross_project).
+seed=${1:-1}
+python -m deepdfa_amd.train.linevul_main \
+  --do_train --do_test --no_flowgnn \
+  --split cross_project \
+  --output_dir saved_models/cross_project \
+  --epochs 10 --block_size 512 --train_batch_size 16 --eval_batch_size 16 \
+  --learning_rate 2e-5 --max_grad_norm 1.0 --seed "$seed" "${@:2}" \
+  2>&1 | tee "train_cross_project_linevul_${seed}.log"
