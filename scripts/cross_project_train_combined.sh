@@ -1,4 +1,5 @@
 #!/bin/bash
+cd "$(dirname "$0")/.."
 # Cross-project DeepDFA+LineVul training (reference
 # scripts/cross_project_train_combined.sh).
 seed=${1:-1}

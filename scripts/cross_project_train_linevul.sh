@@ -1,4 +1,5 @@
 #!/bin/bash
+cd "$(dirname "$0")/.."
 # Cross-project LineVul training (reference scripts/cross_project_train_linevul.sh:
 # train on one project subset, generalize to held-out projects). Here the
 # project split comes from the dataset's project column (--split cross_project).

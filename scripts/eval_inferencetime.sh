@@ -1,4 +1,5 @@
 #!/bin/bash
+cd "$(dirname "$0")/.."
 # Inference-time evaluation (reference eval_inferencetime_{linevul,combined}[_cpu].sh):
 #   bash scripts/eval_inferencetime.sh [combined|linevul] [gpu|cpu]
 variant=${1:-combined}

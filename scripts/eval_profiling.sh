@@ -1,4 +1,5 @@
 #!/bin/bash
+cd "$(dirname "$0")/.."
 # FLOPs-profiling evaluation (reference eval_profiling_{linevul,combined}[_cpu].sh):
 #   bash scripts/eval_profiling.sh [combined|linevul] [gpu|cpu]
 variant=${1:-combined}

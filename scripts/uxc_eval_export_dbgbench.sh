@@ -1,4 +1,5 @@
 #!/bin/bash
+cd "$(dirname "$0")/.."
 # DbgBench prediction export (reference run_all_eval_export_dbgbench[_combined].sh):
 # evaluates on the DbgBench-shaped held-out set and dumps per-example
 # predictions + the CodeT5-format dataset.

@@ -1,4 +1,5 @@
 #!/bin/bash
+cd "$(dirname "$0")/.."
 # Line-level localization metrics sweep (reference rq ... localization runs):
 # Effort@TopK, Recall@TopK%LOC, Top-k accuracy, IFA per reasoning method.
 for method in attention saliency; do

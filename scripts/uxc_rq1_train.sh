@@ -1,4 +1,5 @@
 #!/bin/bash
+cd "$(dirname "$0")/.."
 # UniXcoder RQ1 training (reference LineVul/unixcoder/rq1_train_uxc.sh and
 # the _noflowgnn / _size / _crossproject variants — variant selection via
 # extra flags, e.g. --num_layers 6 for the size ablation).
