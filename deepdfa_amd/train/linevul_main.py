@@ -124,7 +124,7 @@ def train(args, model, train_dataset, eval_dataset, flowgnn_dataset, device):
             torch.nn.utils.clip_grad_norm_(model.parameters(), args.max_grad_norm)
             optimizer.step()
             scheduler.step()
-            losses.append(float(loss))
+            losses.append(float(loss.detach()))
             if step % 100 == 0:
                 logger.info("epoch %d step %d loss %.4f", epoch, step, np.mean(losses[-100:]))
         results = evaluate(args, model, eval_dataset, flowgnn_dataset, device)
