@@ -69,3 +69,31 @@ def test_flat_adamw_training_loss_descends():
     # memorizing a fixed batch must drive the loss down hard; a stale
     # bf16 weight cache would keep it flat
     assert losses[-1] < 0.5 * losses[0], (losses[0], losses[-1])
+
+
+@pytest.mark.gpu
+def test_codet5_defect_training_loss_descends():
+    """CodeT5 DefectModel short fine-tune on a fixed batch (T5 stack +
+    relative-bias grads + fused pre-norm residuals)."""
+    from deepdfa_amd.models.codet5 import DefectModel
+    from deepdfa_amd.models.t5 import T5Config
+    from deepdfa_amd.parallel.optim import FlatAdamW
+
+    torch.manual_seed(0)
+    device = torch.device("cuda:0")
+    cfg = T5Config(num_layers=2, num_decoder_layers=2)
+    model = DefectModel(config=cfg).to(device)
+    opt = FlatAdamW(model.parameters(), lr=1e-4)
+    ids = torch.randint(3, cfg.vocab_size, (8, 128), device=device)
+    ids[:, -1] = cfg.eos_token_id
+    labels = torch.randint(0, 2, (8,), device=device)
+    losses = []
+    for _ in range(25):
+        with torch.autocast(device_type="cuda", dtype=torch.bfloat16):
+            loss, _ = model(ids, labels=labels)
+        opt.zero_grad()
+        loss.backward()
+        opt.clip_grad_norm_(1.0)
+        opt.step()
+        losses.append(float(loss.detach()))
+    assert losses[-1] < 0.7 * losses[0], (losses[0], losses[-1])
