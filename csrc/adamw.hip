@@ -20,10 +20,12 @@ extern "C" __global__ void adamw_fused_kernel(
   const long stride = (long)gridDim.x * blockDim.x;
   for (long iv = (long)blockIdx.x * blockDim.x + threadIdx.x; iv < nvec;
        iv += stride) {
-    f4 pv = reinterpret_cast<f4*>(p)[iv];
-    f4 gv = reinterpret_cast<const f4*>(g)[iv];
-    f4 mv = reinterpret_cast<f4*>(m)[iv];
-    f4 vv = reinterpret_cast<f4*>(v)[iv];
+    // non-temporal: ~1.5 GB of optimizer state per pass can never live in
+    // L2 — keep it from evicting everything else
+    f4 pv = __builtin_nontemporal_load(&reinterpret_cast<f4*>(p)[iv]);
+    f4 gv = __builtin_nontemporal_load(&reinterpret_cast<const f4*>(g)[iv]);
+    f4 mv = __builtin_nontemporal_load(&reinterpret_cast<f4*>(m)[iv]);
+    f4 vv = __builtin_nontemporal_load(&reinterpret_cast<f4*>(v)[iv]);
 #pragma unroll
     for (int u = 0; u < 4; ++u) {
       float gi = gv[u];
@@ -36,9 +38,9 @@ extern "C" __global__ void adamw_fused_kernel(
       if (!l2_mode) pi -= lr * weight_decay * pi;  // decoupled decay (AdamW)
       pv[u] = pi - lr * (mi / bc1) / (sqrtf(vi / bc2) + eps);
     }
-    reinterpret_cast<f4*>(p)[iv] = pv;
-    reinterpret_cast<f4*>(m)[iv] = mv;
-    reinterpret_cast<f4*>(v)[iv] = vv;
+    __builtin_nontemporal_store(pv, &reinterpret_cast<f4*>(p)[iv]);
+    __builtin_nontemporal_store(mv, &reinterpret_cast<f4*>(m)[iv]);
+    __builtin_nontemporal_store(vv, &reinterpret_cast<f4*>(v)[iv]);
   }
   // scalar tail
   for (long i = 4 * nvec + (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
