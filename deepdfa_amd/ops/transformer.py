@@ -182,7 +182,13 @@ class _LinearBf16(torch.autograd.Function):
             _, w16, wt16, b32 = cache
             if bias is not None and b32 is None:
                 b32 = bias.detach().float().contiguous()
-        out = ext.gemm2(x2d, w16, b32, None)
+        # shape dispatch (measured, profiles/optimization_r2.md): our gemm2
+        # beats hipBLASLt on the square 768-ish shapes; the library wins the
+        # fat no-bias ones (K or COL >= ~2k)
+        if b32 is None and (w16.shape[0] > 1024 or w16.shape[1] > 1024):
+            out = torch.matmul(x2d, w16.t())
+        else:
+            out = ext.gemm2(x2d, w16, b32, None)
         ctx.save_for_backward(x2d, w16, wt16)
         ctx.has_bias = bias is not None
         ctx.x_shape = x.shape
@@ -193,7 +199,10 @@ class _LinearBf16(torch.autograd.Function):
         ext = load_ext(required=True)
         x2d, w16, wt16 = ctx.saved_tensors
         dy2d = dy.reshape(-1, dy.shape[-1]).contiguous()
-        dx = ext.gemm2(dy2d, wt16, None, None)  # dY @ W
+        if w16.shape[0] > 1024 or w16.shape[1] > 1024:
+            dx = torch.matmul(dy2d, w16)  # dY @ W, library wins fat shapes
+        else:
+            dx = ext.gemm2(dy2d, wt16, None, None)
         dw = ext.wgrad(dy2d, x2d)  # (COL, K) fp32
         db = ext.colsum(dy2d) if ctx.has_bias else None
         return dx.view(ctx.x_shape), dw, db
@@ -416,7 +425,7 @@ class _QKVLinear(torch.autograd.Function):
         ext = load_ext(required=True)
         x2d, w16, wt16 = ctx.saved_tensors
         dy2d = dy.reshape(-1, dy.shape[-1]).contiguous()
-        dx = ext.gemm2(dy2d, wt16, None, None)
+        dx = torch.matmul(dy2d, w16)  # (N, 3D) @ (3D, D): fat K, library wins
         dw = ext.wgrad(dy2d, x2d)  # (3D, K) fp32
         D = w16.shape[0] // 3
         if ctx.has_bias:
