@@ -82,6 +82,7 @@ template <typename T>
 void launch_softmax_mask_bwd(const T*, const T*, T*, long, int, float, float,
                              unsigned long long, hipStream_t);
 void dkv_prof_fetch(unsigned long long*);
+void fwd_prof_fetch(unsigned long long*);
 template <typename T>
 void launch_dropout_add_fwd(const T*, const T*, T*, long, float, unsigned long long,
                             hipStream_t);
@@ -907,6 +908,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
                               cur_stream());
     });
     return dW;
+  });
+  m.def("fwd_prof", []() {
+    auto t = torch::zeros({8}, torch::dtype(torch::kLong));
+    fwd_prof_fetch(reinterpret_cast<unsigned long long*>(t.data_ptr<long>()));
+    return t;
   });
   m.def("dkv_prof", []() {
     // variant-9 instrumentation readout: per-segment cycle totals

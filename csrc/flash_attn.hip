@@ -65,6 +65,11 @@ __device__ __forceinline__ int fa_swzT(int row, int byte) {
   return row * 128 + (byte ^ (((row >> 3) & 7) << 4));
 }
 
+// fwd segment accumulator (env DFA_FWD_PROF=1 selects the instrumented
+// instantiation)
+__device__ unsigned long long dfa_fwd_prof[8];
+
+template <int PROF>
 __global__ __launch_bounds__(256) void flash_fwd_kernel(
     const bf16* __restrict__ Q, const bf16* __restrict__ K,
     const bf16* __restrict__ V, const int* __restrict__ valid,
@@ -109,6 +114,14 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
     }
   }
 
+  unsigned long long fsegt[8] = {};
+  unsigned long long fseg_last = PROF ? __builtin_readcyclecounter() : 0;
+#define FSEG(i)                                                  \
+  if (PROF) {                                                    \
+    unsigned long long now = __builtin_readcyclecounter();       \
+    fsegt[i] += now - fseg_last;                                 \
+    fseg_last = now;                                             \
+  }
   // softmax state per (fq, its lane-column qrow)
   float m_st[2] = {-3.4e38f, -3.4e38f};
   float l_st[2] = {0.f, 0.f};
@@ -165,9 +178,11 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
   __syncthreads();
   for (int kv0 = 0; kv0 < kv_end; kv0 += TK) {
     const bool has_next = (kv0 + TK) < kv_end;
+    FSEG(0)  // loop head
     if (has_next) load_tile(kv0 + TK, kreg, vreg);
     char* k_lds_c = K_BUF(cur);
     char* v_lds_c = V_BUF(cur);
+    FSEG(1)  // next-tile load issue
 
     // ---- S^T = K @ Q^T : D[key][qrow] ----
     f32x4 s_acc[4][2] = {};  // [fkey][fq]
@@ -187,6 +202,7 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
           s_acc[fk][fq] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               k_frag[fk], q_frag[fq][ks], s_acc[fk][fq], 0, 0, 0);
     }
+    FSEG(2)  // QK^T MFMAs (+k_frag reads)
 
     // ---- masking + bias + tile row-max (over keys, per qrow column) ----
     const int qcol[2] = {qw + (lane & 15), qw + 16 + (lane & 15)};
@@ -233,6 +249,7 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
       sum_p[fq] += __shfl_xor(sum_p[fq], 32);
       l_st[fq] = l_st[fq] * alpha[fq] + sum_p[fq];
     }
+    FSEG(3)  // mask + max + exp + sums (softmax VALU/shfl)
 
     // ---- dropout on P (post-softmax-numerator; scaled at epilogue) ----
     if (p8 > 0) {
@@ -260,6 +277,7 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
 #pragma unroll
         for (int fd = 0; fd < 4; ++fd) o_acc[fq][fd][i] *= a;
       }
+    FSEG(4)  // dropout + O rescale
 
     // ---- P^T (D-layout) -> P A-fragments via lane exchange, then PV ----
     // A-frag for PV k-step kp (keys kp*32..kp*32+31): lane needs
@@ -284,6 +302,7 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
           pa[fq][j] = (__bf16)(hi_half ? v1 : v0);
         }
       }
+      FSEG(5)  // P exchange (shfl)
       bf16x8 v_frag[4];
 #pragma unroll
       for (int fd = 0; fd < 4; ++fd) {
@@ -298,6 +317,7 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
           o_acc[fq][fd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               pa[fq], v_frag[fd], o_acc[fq][fd], 0, 0, 0);
     }
+    FSEG(6)  // v_frag reads + PV MFMAs
     if (has_next) write_tile(K_BUF(cur ^ 1), V_BUF(cur ^ 1), kreg, vreg);
     __syncthreads();
     cur ^= 1;
@@ -305,7 +325,10 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
 #undef K_BUF
 #undef V_BUF
 
-  // ---- epilogue: O /= l, write (B, L, H*64); save lse = m + log(l) ----
+  // ---- epilogue: O /= l via a wave-local LDS bounce (the direct form is
+  // 32 scalar 2-B stores per lane — 24% of the kernel, FSEG profile);
+  // all waves passed the loop's final barrier, so the K/V buffers are free
+  char* obuf = smem + wid * 4096;  // [32 rows][64 d] bf16, fa_swz'd
 #pragma unroll
   for (int fq = 0; fq < 2; ++fq) {
     const float inv_l = (l_st[fq] > 0.f) ? 1.0f / l_st[fq] : 0.f;
@@ -317,18 +340,34 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
     }
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
-      const int qrow16 = (lane >> 4) * 4 + i;
-      const int qrow = qw + fq * 16 + qrow16;
-      const float il = __shfl(inv_l, qrow16);
-      if (qrow >= L) continue;
+      const int row_loc = fq * 16 + (lane >> 4) * 4 + i;
+      const float il = __shfl(inv_l, (lane >> 4) * 4 + i);
 #pragma unroll
       for (int fd = 0; fd < 4; ++fd) {
         const int d = fd * 16 + (lane & 15);
-        O[((long)b * L + qrow) * HD + (long)h * 64 + d] =
+        *reinterpret_cast<bf16*>(obuf + fa_swz(row_loc & 31, d * 2)) =
             __float2bfloat16(o_acc[fq][fd][i] * il);
       }
     }
+    // wave-local buffer: only lgkm ordering needed (same wave reads)
+#pragma unroll
+    for (int pp = 0; pp < 2; ++pp) {
+      const int row_loc = fq * 16 + pp * 8 + (lane >> 3);
+      const int qrow = qw + row_loc;
+      if (qrow < L) {
+        const uint4v vvv = *reinterpret_cast<const uint4v*>(
+            obuf + fa_swz(row_loc & 31, (lane & 7) * 16));
+        *reinterpret_cast<uint4v*>(
+            O + ((long)b * L + qrow) * HD + (long)h * 64 + (lane & 7) * 8) = vvv;
+      }
+    }
   }
+  FSEG(7)  // epilogue
+  if (PROF && lane == 0) {
+#pragma unroll
+    for (int i = 0; i < 8; ++i) atomicAdd(&dfa_fwd_prof[i], fsegt[i]);
+  }
+#undef FSEG
 }
 
 // ---------------------------------------------------------------------------
@@ -1036,6 +1075,13 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
 #undef SEG_MARK
 }
 
+void fwd_prof_fetch(unsigned long long* out) {
+  (void)hipMemcpyFromSymbol(out, HIP_SYMBOL(dfa_fwd_prof),
+                            8 * sizeof(unsigned long long));
+  unsigned long long z[8] = {};
+  (void)hipMemcpyToSymbol(HIP_SYMBOL(dfa_fwd_prof), z, sizeof(z));
+}
+
 void launch_flash_fwd(const bf16* Q, const bf16* K, const bf16* V,
                       const int* valid, const float* bias, bf16* O, float* lse,
                       int B, int H, int L, float scale, int causal,
@@ -1043,9 +1089,15 @@ void launch_flash_fwd(const bf16* Q, const bf16* K, const bf16* V,
                       long ldkv, hipStream_t stream) {
   const dim3 grid((L + NWAVE * TQW - 1) / (NWAVE * TQW), B * H);
   const size_t lds = 2 * 2 * TK * 128;  // double-buffered K+V tiles
-  hipLaunchKernelGGL(flash_fwd_kernel, grid, dim3(256), lds, stream, Q, K, V,
-                     valid, bias, O, lse, B, H, L, scale, causal, p8, seed,
-                     ldq, ldkv);
+  const char* e = getenv("DFA_FWD_PROF");
+  if (e && e[0] == '1')
+    hipLaunchKernelGGL(flash_fwd_kernel<1>, grid, dim3(256), lds, stream, Q, K,
+                       V, valid, bias, O, lse, B, H, L, scale, causal, p8,
+                       seed, ldq, ldkv);
+  else
+    hipLaunchKernelGGL(flash_fwd_kernel<0>, grid, dim3(256), lds, stream, Q, K,
+                       V, valid, bias, O, lse, B, H, L, scale, causal, p8,
+                       seed, ldq, ldkv);
 }
 
 void launch_flash_dterm(const bf16* dO, const bf16* O, float* Dterm, int B,
