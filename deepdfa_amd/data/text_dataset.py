@@ -31,8 +31,12 @@ class InputFeatures:
     index: int
 
 
-def synthetic_func_source(_id: int, n_lines: int = 12) -> str:
-    """Deterministic pseudo-C function for tokenizer-driven paths."""
+def synthetic_func_source(_id: int, n_lines: int = 12, vul: int = 0) -> str:
+    """Deterministic pseudo-C function for tokenizer-driven paths. When
+    `vul` is set the function contains an unchecked strcpy/alloca pattern —
+    a LEARNABLE signal, so end-to-end training demos and regression tests
+    show real F1 movement instead of fitting noise (real Big-Vul data
+    plugs in via the CSV/jsonl loaders)."""
     rng = np.random.RandomState(_id % (2**31))
     names = ["buf", "len", "ptr", "data", "size", "idx", "tmp", "out", "src", "dst"]
     types = ["int", "char *", "size_t", "uint32_t", "void *"]
@@ -46,6 +50,9 @@ def synthetic_func_source(_id: int, n_lines: int = 12) -> str:
             lines.append(f"  memcpy({a}, {b}, {c});")
         else:
             lines.append(f"  {a} = {b} {op} {c};")
+    if vul:
+        pos = 1 + rng.randint(max(1, len(lines) - 1))
+        lines.insert(pos, f"  strcpy({names[rng.randint(10)]}, user_input);")
     lines.append("  return 0;\n}")
     return "\n".join(lines)
 
@@ -94,7 +101,9 @@ class TextDataset(Dataset):
 
                 df = synthetic_bigvul_df(n_synthetic)
                 df = ds_partition(df, partition, split=split)
-                df = df.assign(func=[synthetic_func_source(i) for i in df.id])
+                df = df.assign(
+                    func=[synthetic_func_source(i, vul=v) for i, v in zip(df.id, df.vul)]
+                )
         self.df = df.reset_index(drop=True)
         self.examples = [
             convert_examples_to_features(

@@ -266,7 +266,10 @@ def main(argv=None):
         from ..data.text_dataset import synthetic_func_source
 
         dbg = ds("dbgbench", n=args.n_synthetic, seed=args.seed)
-        dbg = dbg.assign(func=[synthetic_func_source(i + 10**6) for i in dbg.id])
+        dbg = dbg.assign(
+            func=[synthetic_func_source(i + 10**6, vul=v)
+                  for i, v in zip(dbg.id, dbg.vul)]
+        )
         datasets["test"] = TextDataset(tokenizer, args, df=dbg,
                                        block_size=args.block_size)
     results = {}
