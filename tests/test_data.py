@@ -161,3 +161,44 @@ def test_linevul_tokenizer_flag_routing():
     tok = build_tokenizer(args, cfg)
     assert not isinstance(tok, HashTokenizer)
     assert len(tok.encode("int main() { return 0; }", max_length=32)) == 32
+
+
+def test_text_dataset_bigvul_csv_schema(tmp_path):
+    """Real Big-Vul CSVs carry processed_func/target plus extra columns —
+    the loader must accept that schema (roadmap de-risk)."""
+    import pandas as pd
+
+    from deepdfa_amd.data.text_dataset import TextDataset
+    from deepdfa_amd.data.tokenization import HashTokenizer
+
+    df = pd.DataFrame({
+        "index": range(6),
+        "processed_func": [f"int f{i}() {{ return {i}; }}" for i in range(6)],
+        "target": [0, 1, 0, 0, 1, 0],
+        "project": ["qemu"] * 3 + ["ffmpeg"] * 3,
+        "CWE ID": ["CWE-119"] * 6,
+        "id": range(6),
+    })
+    path = tmp_path / "train.csv"
+    df.to_csv(path, index=False)
+    ds = TextDataset(HashTokenizer(), file_path=str(path), block_size=32)
+    assert len(ds) == 6
+    ids, label, index = ds[1]
+    assert label.item() == 1 and ids.shape == (32,)
+
+
+def test_text_dataset_jsonl_schema(tmp_path):
+    """CodeT5-style jsonl (idx/code/target per line, _utils.read_defect_examples)."""
+    import json
+
+    from deepdfa_amd.data.text_dataset import TextDataset
+    from deepdfa_amd.data.tokenization import HashTokenizer
+
+    path = tmp_path / "train.jsonl"
+    with open(path, "w") as f:
+        for i in range(4):
+            f.write(json.dumps({"idx": i, "code": f"void g{i}() {{}}", "target": i % 2}) + "\n")
+    ds = TextDataset(HashTokenizer(), file_path=str(path), block_size=32)
+    assert len(ds) == 4
+    _, label, index = ds[3]
+    assert label.item() == 1 and index.item() == 3
