@@ -524,7 +524,8 @@ __global__ void softmax_mask_fwd_kernel(const T* __restrict__ S,
   sum = wave_sum(sum);
   const float inv = (sum > 0.f) ? 1.0f / sum : 0.f;
   const unsigned p8 = (unsigned)(dropout_p * 256.0f);
-  const float dscale = (dropout_p > 0.f) ? 1.0f / (1.0f - dropout_p) : 1.0f;
+  // match the byte-quantized keep threshold: P(keep) = (256-p8)/256
+  const float dscale = (p8 > 0) ? 256.0f / (256.0f - p8) : 1.0f;
   for (int it = 0; it < niter; ++it) {
     const int j0 = (it * WAVE + lane) * VEC;
     T pv[VEC], pdv[VEC];
@@ -560,7 +561,8 @@ __global__ void softmax_mask_bwd_kernel(const T* __restrict__ dPd,
   constexpr int VEC = 16 / sizeof(T);
   const int niter = (L + WAVE * VEC - 1) / (WAVE * VEC);
   const unsigned p8 = (unsigned)(dropout_p * 256.0f);
-  const float dscale = (dropout_p > 0.f) ? 1.0f / (1.0f - dropout_p) : 1.0f;
+  // match the byte-quantized keep threshold: P(keep) = (256-p8)/256
+  const float dscale = (p8 > 0) ? 256.0f / (256.0f - p8) : 1.0f;
   float dp[4][VEC], p[4][VEC];
   float dot = 0.f;
   for (int it = 0; it < niter; ++it) {

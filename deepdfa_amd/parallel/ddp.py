@@ -22,6 +22,7 @@ SURVEY.md §5.8 "latency-bound" note).
 
 from __future__ import annotations
 
+import contextlib
 import os
 from typing import Dict, List, Optional
 
@@ -60,8 +61,12 @@ class _Bucket:
             self.flat[o : o + n].view(p.shape)
             for p, o, n in zip(params, self.offsets, self.numels)
         ]
-        self.pending = 0
+        self.filled = [False] * len(params)
         self.work = None
+
+    @property
+    def pending(self):
+        return sum(1 for f in self.filled if not f)
 
 
 class DDPEngine:
@@ -76,6 +81,7 @@ class DDPEngine:
         self.model = model
         self.enabled = world_size() > 1
         self.params = [p for p in model.parameters() if p.requires_grad]
+        self._sync = True
         if not self.enabled:
             self.buckets: List[_Bucket] = []
             return
@@ -109,33 +115,81 @@ class DDPEngine:
             self._hooks.append(
                 p.register_post_accumulate_grad_hook(self._on_grad_ready)
             )
+        self._sync = True
         self._reset_pending()
 
     def _reset_pending(self):
         for b in self.buckets:
-            b.pending = len(b.params)
+            b.filled = [False] * len(b.params)
             b.work = None
+        # collectives must be launched in the SAME order on every rank
+        # (RCCL executes them in stream order per communicator) — launch
+        # buckets strictly in index order, like torch DDP's sequence rule
+        self._next_launch = 0
+
+    @contextlib.contextmanager
+    def no_sync(self):
+        """Disable per-backward reduction inside gradient-accumulation
+        micro-batches: gradients just accumulate into p.grad; the final
+        (sync) backward's hooks pick up the accumulated totals. Mirrors
+        torch DDP.no_sync() semantics for drivers like run_defect.py with
+        --gradient_accumulation_steps > 1."""
+        old = self._sync
+        self._sync = False
+        try:
+            yield
+        finally:
+            self._sync = old
 
     def _on_grad_ready(self, p: torch.Tensor):
+        if not self.enabled or not self._sync:
+            return
         b, i = self._param_bucket[id(p)]
+        assert not b.filled[i], (
+            "gradient produced twice before finalize(); wrap accumulation "
+            "micro-batches in engine.no_sync()"
+        )
         b.views[i].copy_(p.grad.detach().to(b.flat.dtype))
-        b.pending -= 1
-        if b.pending == 0:
-            b.flat.div_(world_size())
-            b.work = dist.all_reduce(b.flat, async_op=True)
+        b.filled[i] = True
+        # fire every consecutively-ready bucket starting at _next_launch
+        while self._next_launch < len(self.buckets):
+            nb = self.buckets[self._next_launch]
+            if not all(nb.filled):
+                break
+            nb.flat.div_(world_size())
+            nb.work = dist.all_reduce(nb.flat, async_op=True)
+            self._next_launch += 1
 
     def finalize(self):
         """Wait for all buckets and write averaged grads back. Call between
-        loss.backward() and optimizer.step()."""
+        loss.backward() and optimizer.step().
+
+        Every rank MUST call this at each sync boundary even if its backward
+        was skipped (e.g. run_defect.py's all-graphs-missing batch skip):
+        buckets whose hooks never fired are filled from p.grad (zeros when
+        grad is None) and reduced here, so the collectives stay matched
+        across ranks."""
         if not self.enabled:
             return
+        # launch any buckets the hooks didn't (backward skipped entirely —
+        # dummy participation with zero grads — or a frozen/partial subset):
+        # fill missing views from p.grad / zeros, in index order
+        for b in self.buckets[self._next_launch :]:
+            for i, p in enumerate(b.params):
+                if not b.filled[i]:
+                    if p.grad is None:
+                        b.views[i].zero_()
+                    else:
+                        b.views[i].copy_(p.grad.detach().to(b.flat.dtype))
+            b.flat.div_(world_size())
+            b.work = dist.all_reduce(b.flat, async_op=True)
         for b in self.buckets:
-            if b.work is None and b.pending == len(b.params):
-                continue  # bucket never touched this step (frozen path)
-            assert b.work is not None, "bucket incomplete at finalize"
             b.work.wait()
             for p, v in zip(b.params, b.views):
-                p.grad.detach().copy_(v.to(p.grad.dtype))
+                if p.grad is None:
+                    p.grad = v.to(p.dtype).clone()
+                else:
+                    p.grad.detach().copy_(v.to(p.grad.dtype))
         self._reset_pending()
 
     def all_reduce_scalar(self, t: torch.Tensor, op: str = "max") -> torch.Tensor:

@@ -117,10 +117,11 @@ class FlatAdamW:
     # -- minimal state-dict compat -------------------------------------------
 
     def state_dict(self):
+        # clone: the live buffers keep mutating after a checkpoint snapshot
         return {
-            "flat_p": self.flat_p,
-            "m": self.m,
-            "v": self.v,
+            "flat_p": self.flat_p.detach().clone(),
+            "m": self.m.detach().clone(),
+            "v": self.v.detach().clone(),
             "step_count": self.step_count,
             "lr": self.param_groups[0]["lr"],
         }
@@ -131,3 +132,10 @@ class FlatAdamW:
         self.v.copy_(sd["v"])
         self.step_count = sd["step_count"]
         self.param_groups[0]["lr"] = sd["lr"]
+        # the device-side step counter drives GPU bias correction — resync
+        # it (and drop the derived buffers) so a resume doesn't step with a
+        # stale count
+        if hasattr(self, "_step_t"):
+            del self._step_t
+            del self._betas_t
+            del self._bc

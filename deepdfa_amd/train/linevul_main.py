@@ -299,10 +299,21 @@ def main(argv=None):
         )
     model = Model(config=cfg, flowgnn_encoder=flowgnn_model).to(device)
 
+    # real-data parity (reference linevul_main.py:427-432): explicit
+    # per-partition files override the synthetic generator entirely
+    file_map = {
+        "train": args.train_data_file,
+        "val": args.eval_data_file,
+        "test": args.test_data_file,
+    }
+    flag_names = {"train": "train_data_file", "val": "eval_data_file", "test": "test_data_file"}
+    for part, path in file_map.items():
+        if path is not None and not os.path.exists(path):
+            raise FileNotFoundError(f"--{flag_names[part]}: {path}")
     datasets = {
         part: TextDataset(
-            tokenizer, args, partition=part, block_size=args.block_size,
-            n_synthetic=n_syn, split=args.split,
+            tokenizer, args, file_path=file_map[part], partition=part,
+            block_size=args.block_size, n_synthetic=n_syn, split=args.split,
         )
         for part in ("train", "val", "test")
     }

@@ -10,6 +10,7 @@ eval (:311-413), best checkpoint `checkpoint-best-acc/pytorch_model.bin` +
 from __future__ import annotations
 
 import argparse
+import contextlib
 import logging
 import os
 
@@ -86,6 +87,11 @@ def build_args(argv=None):
     p.add_argument("--n_synthetic", type=int, default=2000)
     p.add_argument("--num_layers", type=int, default=12)
     p.add_argument("--d_model", type=int, default=768)
+    # real-data parity (reference CodeT5/configs.py:42-47): explicit jsonl
+    # files per partition (idx/code/target rows, _utils.py:260-278)
+    p.add_argument("--train_filename", default=None)
+    p.add_argument("--dev_filename", default=None)
+    p.add_argument("--test_filename", default=None)
     return p.parse_args(argv)
 
 
@@ -115,8 +121,14 @@ def main(argv=None):
         fg_ds = BigVulDatasetLineVD(partition="all", n_synthetic=args.n_synthetic,
                                     missing_rate=0.07)
     model = DefectModel(config=cfg, flowgnn_encoder=fg).to(device)
+    file_map = {"train": args.train_filename, "val": args.dev_filename,
+                "test": args.test_filename}
+    flag_names = {"train": "train_filename", "val": "dev_filename", "test": "test_filename"}
+    for part, path in file_map.items():
+        if path is not None and not os.path.exists(path):
+            raise FileNotFoundError(f"--{flag_names[part]}: {path}")
     datasets = {
-        part: TextDataset(tokenizer, args, partition=part,
+        part: TextDataset(tokenizer, args, file_path=file_map[part], partition=part,
                           block_size=args.max_source_length, n_synthetic=args.n_synthetic)
         for part in ("train", "val", "test")
     }
@@ -140,19 +152,26 @@ def main(argv=None):
                 sampler.set_epoch(epoch)
             opt.zero_grad(set_to_none=True)
             for step, (ids, label, index) in enumerate(loader):
+                # boundary is driven by the loader-index counter, which is
+                # rank-synchronized (DistributedSampler pads every rank to
+                # the same length) — a rank that skips a batch still hits
+                # the same boundaries and participates in the collectives
+                is_boundary = (step + 1) % args.gradient_accumulation_steps == 0
                 graphs, keep = join_graphs(fg_ds, index, device)
                 if keep is not None and not keep.all():
                     ids, label = ids[keep], label[keep]
-                if keep is not None and graphs is None:
-                    continue
-                ids = add_eos(ids, cfg.pad_token_id, cfg.eos_token_id).to(device)
-                label = label.to(device)
-                with torch.autocast(device_type="cuda", dtype=torch.bfloat16,
-                                    enabled=device.type == "cuda"):
-                    loss, _prob = model(ids, labels=label, graphs=graphs)
-                (loss / args.gradient_accumulation_steps).backward()
-                if (step + 1) % args.gradient_accumulation_steps == 0:
-                    ddp.finalize()
+                skipped = keep is not None and graphs is None
+                if not skipped:
+                    ids = add_eos(ids, cfg.pad_token_id, cfg.eos_token_id).to(device)
+                    label = label.to(device)
+                    sync_ctx = contextlib.nullcontext() if is_boundary else ddp.no_sync()
+                    with sync_ctx:
+                        with torch.autocast(device_type="cuda", dtype=torch.bfloat16,
+                                            enabled=device.type == "cuda"):
+                            loss, _prob = model(ids, labels=label, graphs=graphs)
+                        (loss / args.gradient_accumulation_steps).backward()
+                if is_boundary:
+                    ddp.finalize()  # dummy-participates (zero grads) on skip
                     torch.nn.utils.clip_grad_norm_(model.parameters(), 1.0)
                     opt.step()
                     sched.step()

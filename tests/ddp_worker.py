@@ -15,6 +15,7 @@ from deepdfa_amd.models import FlowGNNGGNNModule  # noqa: E402
 def main():
     rank, world = int(sys.argv[1]), int(sys.argv[2])
     file_name, out_path = sys.argv[3], sys.argv[4]
+    mode = sys.argv[5] if len(sys.argv) > 5 else "basic"
     dist.init_process_group("gloo", init_method=f"file://{file_name}", rank=rank, world_size=world)
     os.environ["WORLD_SIZE"] = str(world)
     from deepdfa_amd.parallel.ddp import DDPEngine
@@ -23,10 +24,31 @@ def main():
     model = FlowGNNGGNNModule(input_dim=1002, hidden_dim=32, n_steps=2, num_output_layers=3)
     ddp = DDPEngine(model, bucket_cap_mb=0.25)  # force multiple buckets
     assert ddp.enabled and len(ddp.buckets) > 1
-    g = synthetic_cfg_batch(8, seed=rank)  # different data per rank
-    loss = model.training_step((g, {}))
-    loss.backward()
-    ddp.finalize()
+    if mode == "accum":
+        # gradient accumulation: 2 no_sync micro-batches + 1 sync micro-batch;
+        # reduced grad must be mean over ranks of the SUM over micro-batches
+        for micro in range(3):
+            g = synthetic_cfg_batch(8, seed=rank * 10 + micro)
+            loss = model.training_step((g, {}))
+            if micro < 2:
+                with ddp.no_sync():
+                    loss.backward()
+            else:
+                loss.backward()
+        ddp.finalize()
+    elif mode == "skip":
+        # rank 1 skips its backward entirely (all-graphs-missing batch) but
+        # still participates in the collectives via finalize()
+        if rank == 0:
+            g = synthetic_cfg_batch(8, seed=0)
+            loss = model.training_step((g, {}))
+            loss.backward()
+        ddp.finalize()
+    else:
+        g = synthetic_cfg_batch(8, seed=rank)  # different data per rank
+        loss = model.training_step((g, {}))
+        loss.backward()
+        ddp.finalize()
     grad_vec = torch.cat([p.grad.flatten() for p in model.parameters() if p.grad is not None])
     opt = torch.optim.SGD(model.parameters(), lr=0.1)
     opt.step()

@@ -27,27 +27,53 @@ def _single_process_grads(seed_data):
     return torch.cat([p.grad.flatten() for p in model.parameters() if p.grad is not None])
 
 
-def test_ddp_two_ranks(tmp_path):
+def _run_workers(tmp_path, mode):
     f = str(tmp_path / "rendezvous")
     outs_paths = [str(tmp_path / f"out{r}.pt") for r in range(2)]
     procs = [
         subprocess.Popen(
-            [sys.executable, os.path.join(REPO, "tests", "ddp_worker.py"), str(r), "2", f, outs_paths[r]],
+            [sys.executable, os.path.join(REPO, "tests", "ddp_worker.py"),
+             str(r), "2", f, outs_paths[r], mode],
             env={**os.environ, "PYTHONPATH": REPO},
             stdout=subprocess.PIPE,
             stderr=subprocess.STDOUT,
         )
         for r in range(2)
     ]
-    logs = []
     for p in procs:
         out, _ = p.communicate(timeout=240)
-        logs.append(out.decode())
         assert p.returncode == 0, out.decode()[-2000:]
-    outs = [torch.load(p, weights_only=True) for p in outs_paths]
+    return [torch.load(p, weights_only=True) for p in outs_paths]
+
+
+def test_ddp_two_ranks(tmp_path):
+    outs = _run_workers(tmp_path, "basic")
     # 1. both ranks ended with identical params (broadcast + identical update)
     assert torch.allclose(outs[0]["params"], outs[1]["params"], atol=1e-7)
     # 2. grads identical across ranks and equal to the average of per-rank grads
     assert torch.allclose(outs[0]["grad"], outs[1]["grad"], atol=1e-7)
     expected = (_single_process_grads(0) + _single_process_grads(1)) / 2
     assert torch.allclose(outs[0]["grad"], expected, atol=1e-5)
+
+
+def test_ddp_gradient_accumulation(tmp_path):
+    """no_sync() micro-batches accumulate locally; the boundary reduce must
+    average the SUM of all micro-batch grads (advisor round-1 high finding:
+    the hook engine used to reduce after micro-batch 1 and discard the rest)."""
+    outs = _run_workers(tmp_path, "accum")
+    assert torch.allclose(outs[0]["grad"], outs[1]["grad"], atol=1e-7)
+    expected = sum(
+        _single_process_grads(r * 10 + m) for r in range(2) for m in range(3)
+    ) / 2
+    assert torch.allclose(outs[0]["grad"], expected, atol=1e-5)
+
+
+def test_ddp_skip_participation(tmp_path):
+    """A rank whose batch was entirely missing still participates in the
+    collectives via finalize() (zero contribution) — no hang, and the
+    average is rank0's grads / world_size."""
+    outs = _run_workers(tmp_path, "skip")
+    assert torch.allclose(outs[0]["grad"], outs[1]["grad"], atol=1e-7)
+    expected = _single_process_grads(0) / 2
+    assert torch.allclose(outs[0]["grad"], expected, atol=1e-5)
+    assert torch.allclose(outs[0]["params"], outs[1]["params"], atol=1e-7)
