@@ -70,7 +70,19 @@ class BaseModule(nn.Module):
 
     # -- loss -----------------------------------------------------------------
 
-    def loss_fn(self, logits: torch.Tensor, label: torch.Tensor) -> torch.Tensor:
+    def loss_fn(
+        self,
+        logits: torch.Tensor,
+        label: torch.Tensor,
+        weight: Optional[torch.Tensor] = None,
+    ) -> torch.Tensor:
+        """weight (per-graph 0/1) masks out capture-padding dummy graphs
+        (graph/pad.py): the weighted mean equals the unpadded batch mean."""
+        if weight is not None:
+            per = torch.nn.functional.binary_cross_entropy_with_logits(
+                logits, label, pos_weight=self.pos_weight, reduction="none"
+            )
+            return (per * weight).sum() / weight.sum().clamp(min=1.0)
         return torch.nn.functional.binary_cross_entropy_with_logits(
             logits, label, pos_weight=self.pos_weight
         )
@@ -84,6 +96,19 @@ class BaseModule(nn.Module):
         loss = self.loss_fn(logits.float(), label)
         with torch.no_grad():
             self.metrics["train"].to(logits.device).update(torch.sigmoid(logits.float()), label)
+        return loss
+
+    def training_step_masked(self, graph, extrafeats, weight: torch.Tensor) -> torch.Tensor:
+        """Capture-path training step over a padded batch (graph/pad.py):
+        identical math to training_step on the unpadded batch — the dummy
+        graphs contribute zero loss, zero gradient and zero metric counts."""
+        label = self.get_label(graph)
+        logits = self(graph, extrafeats)
+        loss = self.loss_fn(logits.float(), label, weight=weight)
+        with torch.no_grad():
+            self.metrics["train"].to(logits.device).update(
+                torch.sigmoid(logits.float()), label, mask=weight
+            )
         return loss
 
     @torch.no_grad()

@@ -207,9 +207,15 @@ def main(argv: Optional[List[str]] = None) -> Dict:
         if args.subcommand == "fit":
             from ..utils.logging import HPOReporter
 
-            opt = torch.optim.Adam(
-                [p for p in model.parameters() if p.requires_grad], **cfg["optimizer"]
-            )
+            params = [p for p in model.parameters() if p.requires_grad]
+            if getattr(trainer, "graph_capture", False) and torch.cuda.is_available():
+                # capture-safe flat optimizer (device-side bias correction);
+                # l2_mode=True == torch.optim.Adam's weight_decay semantics
+                from ..parallel.optim import FlatAdamW
+
+                opt = FlatAdamW(params, l2_mode=True, **cfg["optimizer"])
+            else:
+                opt = torch.optim.Adam(params, **cfg["optimizer"])
             reporter = HPOReporter(run_dir)
             out = trainer.fit(model, dm, optimizer=opt, hpo_reporter=reporter)
             # post-fit: validate the best checkpoint (main_cli.py:167-184)

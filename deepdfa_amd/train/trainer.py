@@ -13,9 +13,13 @@ Parity targets:
   * .ckpt payload: state_dict + optimizer state + hyper_parameters
     (Lightning layout subset, enough for resume + transfer).
 
-MI355X-native execution: model/graphs on cuda:0 per process, bf16 autocast
-optional, hipGraph step capture for the launch-bound flow-GNN
-(torch.cuda.CUDAGraph == hipGraph on ROCm), DDP via deepdfa_amd.parallel.
+MI355X-native execution: one process per GPU (torchrun) with the RCCL DDP
+engine — rank-consistent per-epoch undersampling via a shared-seed shuffle
++ strided shard (datamodule), per-step bucketed gradient all-reduce
+(parallel/ddp.py), metric-counter + loss all-reduce at epoch boundaries,
+rank-0 checkpoint writes; bf16 autocast; hipGraph step capture for the
+launch-bound flow-GNN (capture.py; torch.cuda.CUDAGraph == hipGraph on
+ROCm) when graph_capture is enabled on a single-GPU run.
 """
 
 from __future__ import annotations
@@ -30,7 +34,29 @@ from typing import Dict, Optional
 
 import torch
 
+from ..parallel.ddp import DDPEngine, init_distributed, world_size
+
 logger = logging.getLogger(__name__)
+
+
+def _dist_rank_ws():
+    import torch.distributed as dist
+
+    if dist.is_available() and dist.is_initialized():
+        return dist.get_rank(), dist.get_world_size()
+    return 0, 1
+
+
+def _allreduce_sums(vals, device):
+    """Sum a list of python floats across ranks (no-op if not distributed)."""
+    import torch.distributed as dist
+
+    if not (dist.is_available() and dist.is_initialized()):
+        return vals
+    t = torch.tensor(vals, dtype=torch.float64,
+                     device=device if device.type == "cuda" else "cpu")
+    dist.all_reduce(t)
+    return t.tolist()
 
 
 class Trainer:
@@ -46,7 +72,9 @@ class Trainer:
         seed: int = 0,
         precision: str = "bf16",  # "bf16" (autocast on GPU) or "fp32"
         detect_anomaly: bool = False,
+        graph_capture: bool = True,  # hipGraph step capture (GPU, ws=1, FlatAdamW)
     ):
+        self.graph_capture = graph_capture
         # reference config_default.yaml:37 trainer.detect_anomaly — autograd
         # NaN/inf detection; the kernel-level analog is scripts/run_sanitize.sh
         self.detect_anomaly = detect_anomaly
@@ -107,55 +135,102 @@ class Trainer:
     def fit(self, model, datamodule, optimizer=None, hpo_reporter=None) -> Dict:
         from ..utils.logging import ScalarLogger
 
+        rank = init_distributed()
+        ws = world_size()
+        if ws > 1 and torch.cuda.is_available():
+            self.device = torch.device("cuda", int(os.environ.get("LOCAL_RANK", 0)))
+            torch.cuda.set_device(self.device)
         torch.manual_seed(self.seed)
         if self.detect_anomaly:
             torch.autograd.set_detect_anomaly(True)
         model = model.to(self.device)
         if optimizer is None:
             optimizer = torch.optim.Adam(model.parameters(), lr=1e-3, weight_decay=1e-2)
+        # gradient averaging: flat optimizers (FlatAdamW) do ONE flat
+        # all-reduce after backward; otherwise the bucketed-overlap engine
+        flat = hasattr(optimizer, "allreduce_grads")
+        ddp = None
+        if flat:
+            if ws > 1:
+                with torch.no_grad():
+                    import torch.distributed as dist
+
+                    dist.broadcast(optimizer.flat_p, src=0)
+        else:
+            ddp = DDPEngine(model)
+        captured = None
+        if (
+            self.graph_capture
+            and ws == 1
+            and self.device.type == "cuda"
+            and self.precision == "bf16"
+            and hasattr(optimizer, "flat_g")
+            and hasattr(model, "training_step_masked")
+        ):
+            from .capture import CapturedTrainStep
+
+            captured = CapturedTrainStep(
+                model, optimizer, batch_size=datamodule.batch_size,
+                grad_clip=self.grad_clip,
+            )
+            logger.info("fit: hipGraph step capture enabled")
         history = []
-        scalars = ScalarLogger(self.root)
+        scalars = ScalarLogger(self.root) if rank == 0 else None
+        # one shared generator, advanced once per epoch on EVERY rank: all
+        # ranks draw the identical epoch permutation, then take their shard
         gen = torch.Generator().manual_seed(self.seed)
         for epoch in range(self.max_epochs):
             model.train()
             t0 = _time.perf_counter()
-            train_loader = datamodule.train_dataloader(generator=gen)
+            train_loader = datamodule.train_dataloader(generator=gen, rank=rank, world=ws)
             n_batches = 0
             loss_sum = 0.0
             for batch in train_loader:
                 batch = self._to_device(batch)
-                with self._autocast():
-                    loss = model.training_step(batch)
-                optimizer.zero_grad(set_to_none=True)
-                loss.backward()
-                if self.grad_clip:
-                    torch.nn.utils.clip_grad_norm_(model.parameters(), self.grad_clip)
-                optimizer.step()
+                if captured is not None:
+                    captured(batch)
+                else:
+                    with self._autocast():
+                        loss = model.training_step(batch)
+                    optimizer.zero_grad(set_to_none=True)
+                    loss.backward()
+                    if flat and ws > 1:
+                        optimizer.allreduce_grads()
+                    elif ddp is not None:
+                        ddp.finalize()
+                    if self.grad_clip:
+                        torch.nn.utils.clip_grad_norm_(model.parameters(), self.grad_clip)
+                    optimizer.step()
+                    loss_sum += float(loss.detach())
                 self.global_step += 1
                 n_batches += 1
-                loss_sum += float(loss.detach())
-            train_metrics = model.epoch_metrics("train")
+            if captured is not None:
+                loss_sum = captured.pop_loss_sum()
+            train_metrics = model.epoch_metrics("train")  # counter all-reduce
+            loss_sum, nb = _allreduce_sums([loss_sum, float(n_batches)], self.device)
             row = {
                 "epoch": epoch,
-                "train_loss": loss_sum / max(1, n_batches),
+                "train_loss": loss_sum / max(1.0, nb),
                 "epoch_time_s": _time.perf_counter() - t0,
                 **train_metrics,
             }
             if (epoch + 1) % self.check_val_every == 0:
                 val = self.validate(model, datamodule, _during_fit=True)
                 row.update(val)
-                val_loss = val["val_loss"]
+                val_loss = val["val_loss"]  # identical on every rank (reduced)
                 if val_loss < self._best_val:
                     self._best_val = val_loss
-                    # remove previous best (save_top_k: 1)
-                    if self._best_path and os.path.exists(self._best_path):
-                        os.unlink(self._best_path)
+                    if rank == 0:
+                        # remove previous best (save_top_k: 1)
+                        if self._best_path and os.path.exists(self._best_path):
+                            os.unlink(self._best_path)
                     name = f"performance-{epoch:02d}-{self.global_step:02d}-{val_loss:02f}.ckpt"
-                    self._best_path = self.save_checkpoint(
-                        model, optimizer, epoch, val_loss, name
-                    )
-                self.save_checkpoint(model, optimizer, epoch, val_loss, "last.ckpt")
-            if (epoch + 1) % self.periodic_every == 0:
+                    self._best_path = os.path.join(self.ckpt_dir, name)
+                    if rank == 0:
+                        self.save_checkpoint(model, optimizer, epoch, val_loss, name)
+                if rank == 0:
+                    self.save_checkpoint(model, optimizer, epoch, val_loss, "last.ckpt")
+            if (epoch + 1) % self.periodic_every == 0 and rank == 0:
                 self.save_checkpoint(
                     model,
                     optimizer,
@@ -164,43 +239,72 @@ class Trainer:
                     f"periodical-{epoch}-{self.global_step}.ckpt",
                 )
             history.append(row)
-            scalars.log({**row, "lr": optimizer.param_groups[0].get("lr", 0.0)}, self.global_step)
+            if scalars is not None:
+                scalars.log({**row, "lr": optimizer.param_groups[0].get("lr", 0.0)}, self.global_step)
             if hpo_reporter is not None and "val_f1" in row:
                 hpo_reporter.report_intermediate(row["val_f1"])
             logger.info("epoch %d: %s", epoch, json.dumps({k: round(v, 5) if isinstance(v, float) else v for k, v in row.items()}))
-        scalars.close()
+        if scalars is not None:
+            scalars.close()
+        if captured is not None:
+            logger.info(
+                "fit: %d steps captured-replayed, %d eager fallbacks, %d shape buckets",
+                captured.steps - captured.eager_steps, captured.eager_steps,
+                len(captured.buckets),
+            )
+        if ws > 1:
+            import torch.distributed as dist
+
+            dist.barrier()  # rank0 finished writing checkpoints
         return {"history": history, "best_checkpoint": self.best_checkpoint()}
 
     @torch.no_grad()
     def validate(self, model, datamodule, _during_fit: bool = False) -> Dict:
+        rank, ws = _dist_rank_ws()
         model = model.to(self.device)
         model.eval()
         loss_sum, n = 0.0, 0
-        for batch in datamodule.val_dataloader():
+        for batch in datamodule.val_dataloader(rank=rank, world=ws):
             batch = self._to_device(batch)
             with self._autocast():
                 loss = model.validation_step(batch)
             loss_sum += float(loss)
             n += 1
-        out = {"val_loss": loss_sum / max(1, n)}
-        out.update(model.epoch_metrics("val"))
+        loss_sum, nf = _allreduce_sums([loss_sum, float(n)], self.device)
+        out = {"val_loss": loss_sum / max(1.0, nf)}
+        out.update(model.epoch_metrics("val"))  # counter all-reduce inside
         if not _during_fit:
             model.train()
         return out
 
     @torch.no_grad()
     def test(self, model, datamodule, out_dir: Optional[str] = None) -> Dict:
+        rank, ws = _dist_rank_ws()
         model = model.to(self.device)
         model.eval()
         loss_sum, n = 0.0, 0
-        for batch in datamodule.test_dataloader():
+        for batch in datamodule.test_dataloader(rank=rank, world=ws):
             batch = self._to_device(batch)
             with self._autocast():
                 loss = model.test_step(batch)
             loss_sum += float(loss)
             n += 1
-        results = {"test_loss": loss_sum / max(1, n)}
-        results.update(model.test_epoch_end(out_dir=out_dir or self.root))
+        loss_sum, nf = _allreduce_sums([loss_sum, float(n)], self.device)
+        if ws > 1:
+            # gather per-example preds/labels (already on CPU) so the PR
+            # curve / classification report cover the whole test split
+            import torch.distributed as dist
+
+            gp: list = [None] * ws
+            gl: list = [None] * ws
+            dist.all_gather_object(gp, model._test_preds)
+            dist.all_gather_object(gl, model._test_labels)
+            model._test_preds = [t for lst in gp for t in lst]
+            model._test_labels = [t for lst in gl for t in lst]
+        results = {"test_loss": loss_sum / max(1.0, nf)}
+        results.update(
+            model.test_epoch_end(out_dir=(out_dir or self.root) if rank == 0 else None)
+        )
         return results
 
     # -- helpers --------------------------------------------------------------

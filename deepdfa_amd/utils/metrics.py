@@ -25,13 +25,27 @@ class BinaryStats:
         self.counts = self.counts.to(device)
         return self
 
-    def update(self, probs: torch.Tensor, labels: torch.Tensor) -> None:
+    def update(
+        self,
+        probs: torch.Tensor,
+        labels: torch.Tensor,
+        mask: Optional[torch.Tensor] = None,
+    ) -> None:
+        """mask (same shape, truthy = count) excludes e.g. the dummy graphs a
+        capture-padded batch appends (graph/pad.py)."""
         pred = (probs >= self.threshold).to(torch.int64)
         lab = labels.to(torch.int64)
-        tp = ((pred == 1) & (lab == 1)).sum()
-        fp = ((pred == 1) & (lab == 0)).sum()
-        tn = ((pred == 0) & (lab == 0)).sum()
-        fn = ((pred == 0) & (lab == 1)).sum()
+        if mask is not None:
+            m = mask.to(torch.bool)
+            tp = ((pred == 1) & (lab == 1) & m).sum()
+            fp = ((pred == 1) & (lab == 0) & m).sum()
+            tn = ((pred == 0) & (lab == 0) & m).sum()
+            fn = ((pred == 0) & (lab == 1) & m).sum()
+        else:
+            tp = ((pred == 1) & (lab == 1)).sum()
+            fp = ((pred == 1) & (lab == 0)).sum()
+            tn = ((pred == 0) & (lab == 0)).sum()
+            fn = ((pred == 0) & (lab == 1)).sum()
         self.counts += torch.stack([tp, fp, tn, fn])
 
     def reset(self) -> None:

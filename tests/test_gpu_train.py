@@ -97,3 +97,37 @@ def test_codet5_defect_training_loss_descends():
         opt.step()
         losses.append(float(loss.detach()))
     assert losses[-1] < 0.7 * losses[0], (losses[0], losses[-1])
+
+
+@pytest.mark.gpu
+def test_trainer_fit_graph_capture_parity(tmp_path):
+    """Production fit with hipGraph capture vs eager on the same seed/data:
+    histories must match closely (capture pads batches with masked dummy
+    graphs — numerically equivalent up to bf16 tiling differences)."""
+    from deepdfa_amd.data.datamodule import BigVulDatasetLineVDDataModule
+    from deepdfa_amd.models import FlowGNNGGNNModule
+    from deepdfa_amd.parallel.optim import FlatAdamW
+    from deepdfa_amd.train.trainer import Trainer
+
+    results = {}
+    for mode in ("eager", "capture"):
+        torch.manual_seed(0)
+        dm = BigVulDatasetLineVDDataModule(batch_size=32, n_synthetic=400,
+                                           undersample="v1.0", seed=0)
+        model = FlowGNNGGNNModule(input_dim=dm.input_dim, hidden_dim=32,
+                                  n_steps=5, num_output_layers=3)
+        trainer = Trainer(max_epochs=2, default_root_dir=str(tmp_path / mode),
+                          precision="bf16", seed=0,
+                          graph_capture=(mode == "capture"))
+        opt = FlatAdamW(model.to(trainer.device).parameters(), lr=1e-3,
+                        weight_decay=1e-2, l2_mode=True)
+        out = trainer.fit(model, dm, optimizer=opt)
+        results[mode] = out["history"]
+    for r_e, r_c in zip(results["eager"], results["capture"]):
+        assert abs(r_e["train_loss"] - r_c["train_loss"]) < 0.05, (r_e, r_c)
+        # identical data order => identical counts up to bf16 threshold flips
+        total = max(1.0, r_e["train_tp"] + r_e["train_fp"] + r_e["train_tn"] + r_e["train_fn"])
+        assert r_e["train_tp"] + r_e["train_fp"] + r_e["train_tn"] + r_e["train_fn"] == \
+               r_c["train_tp"] + r_c["train_fp"] + r_c["train_tn"] + r_c["train_fn"]
+        assert abs(r_e["train_f1"] - r_c["train_f1"]) < 0.15, (r_e, r_c)
+    assert abs(results["eager"][-1]["val_loss"] - results["capture"][-1]["val_loss"]) < 0.05

@@ -1,0 +1,132 @@
+"""DDP through the production Trainer/datamodule path (VERDICT round-1 #1):
+rank-consistent undersampling, sharded loaders, metric/loss all-reduce,
+rank-identical results — verified with 2 gloo processes on CPU, plus
+deterministic sharding checks without any process group.
+"""
+
+import os
+import subprocess
+import sys
+
+import numpy as np
+import torch
+
+from deepdfa_amd.data.datamodule import BigVulDatasetLineVDDataModule
+from deepdfa_amd.graph.pad import bucket_shape, pad_batch
+from deepdfa_amd.graph.synthetic import synthetic_cfg_batch
+from deepdfa_amd.models import FlowGNNGGNNModule
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _shard_indices(dm, seed, rank, world):
+    gen = torch.Generator().manual_seed(seed)
+    loader = dm.train_dataloader(generator=gen, rank=rank, world=world)
+    return list(loader.dataset.indices)
+
+
+def test_datamodule_rank_sharding_partitions_epoch_draw():
+    """Two ranks with identically-seeded generators draw the SAME epoch
+    permutation and take disjoint shards covering the whole (padded) draw."""
+    dms = [
+        BigVulDatasetLineVDDataModule(batch_size=8, n_synthetic=300,
+                                      undersample="v1.0", seed=0)
+        for _ in range(2)
+    ]
+    s0 = _shard_indices(dms[0], 5, rank=0, world=2)
+    s1 = _shard_indices(dms[1], 5, rank=1, world=2)
+    assert len(s0) == len(s1)  # padded to equal length
+    union = sorted(s0 + s1)
+    # reproduce the epoch draw a third identical datamodule would make
+    dm_ref = BigVulDatasetLineVDDataModule(batch_size=8, n_synthetic=300,
+                                           undersample="v1.0", seed=0)
+    draw = np.asarray(dm_ref.train.get_epoch_indices())
+    perm = torch.randperm(len(draw), generator=torch.Generator().manual_seed(5)).numpy()
+    idx = draw[perm]
+    if len(idx) % 2:
+        idx = np.concatenate([idx, idx[:1]])
+    assert union == sorted(idx.tolist())
+    # the unpadded portions are disjoint
+    overlap = set(s0) & set(s1)
+    dup_pad = len(idx) - len(set(idx.tolist()))
+    assert len(overlap) <= dup_pad + 1
+
+
+def test_datamodule_epoch_draws_differ():
+    dm = BigVulDatasetLineVDDataModule(batch_size=8, n_synthetic=300,
+                                       undersample="v1.0", seed=0)
+    gen = torch.Generator().manual_seed(5)
+    a = list(dm.train_dataloader(generator=gen).dataset.indices)
+    b = list(dm.train_dataloader(generator=gen).dataset.indices)
+    assert a != b  # per-epoch resample/reshuffle
+
+
+def test_trainer_fit_two_ranks(tmp_path):
+    f = str(tmp_path / "rdv")
+    outs_paths = [str(tmp_path / f"out{r}.pt") for r in range(2)]
+    procs = [
+        subprocess.Popen(
+            [sys.executable, os.path.join(REPO, "tests", "fit_worker.py"),
+             str(r), "2", f, outs_paths[r], str(tmp_path)],
+            env={**os.environ, "PYTHONPATH": REPO},
+            stdout=subprocess.PIPE, stderr=subprocess.STDOUT,
+        )
+        for r in range(2)
+    ]
+    for p in procs:
+        out, _ = p.communicate(timeout=300)
+        assert p.returncode == 0, out.decode()[-3000:]
+    outs = [torch.load(p, weights_only=False) for p in outs_paths]
+    # identical final params on both ranks (broadcast + averaged grads)
+    assert torch.allclose(outs[0]["params"], outs[1]["params"], atol=1e-6)
+    # identical history rows (losses/metrics are all-reduced)
+    for r0, r1 in zip(outs[0]["history"], outs[1]["history"]):
+        for k in r0:
+            if k == "epoch_time_s":
+                continue
+            assert r0[k] == r1[k], (k, r0[k], r1[k])
+    # a best checkpoint was written (by rank 0) and both ranks agree on it
+    assert outs[0]["best"] == outs[1]["best"]
+    assert outs[0]["best"] and os.path.exists(outs[0]["best"])
+
+
+def test_pad_batch_masked_step_matches_unpadded():
+    """Capture-path equivalence: the padded batch + weight mask gives the
+    same loss, gradients, and metric counts as the raw batch."""
+    torch.manual_seed(3)
+    model = FlowGNNGGNNModule(input_dim=1002, hidden_dim=8, n_steps=2,
+                              num_output_layers=2)
+    g = synthetic_cfg_batch(12, seed=4)
+    loss_ref = model.training_step((g, {}))
+    loss_ref.backward()
+    grads_ref = torch.cat([p.grad.flatten().clone() for p in model.parameters()])
+    counts_ref = model.metrics["train"].counts.clone()
+    model.zero_grad(set_to_none=False)
+    model.metrics["train"].reset()
+
+    shape = bucket_shape(g, b_pad=16, node_q=256, edge_q=512)
+    padded, w = pad_batch(g, *shape)
+    assert (padded.num_nodes, padded.num_edges, padded.num_graphs) == shape
+    loss_pad = model.training_step_masked(padded, {}, w)
+    loss_pad.backward()
+    grads_pad = torch.cat([p.grad.flatten() for p in model.parameters()])
+    assert torch.allclose(loss_ref, loss_pad, atol=1e-6)
+    assert torch.allclose(grads_ref, grads_pad, atol=1e-5)
+    assert torch.equal(counts_ref, model.metrics["train"].counts)
+
+
+def test_pad_batch_structure():
+    g = synthetic_cfg_batch(5, seed=0)
+    padded, w = pad_batch(g, g.num_nodes + 40, g.num_edges + 60, 8)
+    assert padded.num_graphs == 8 and w.sum() == 5
+    # real prefix untouched
+    assert torch.equal(padded.indptr[: g.num_nodes + 1], g.indptr)
+    assert torch.equal(padded.indices[: g.num_edges], g.indices)
+    assert torch.equal(padded.node_offsets[:6], g.node_offsets)
+    # monotone CSR, in == out edge counts
+    d = padded.indptr.to(torch.int64).diff()
+    assert (d >= 0).all() and int(padded.indptr[-1]) == padded.num_edges
+    assert int(padded.t_indptr[-1]) == padded.num_edges
+    # dummy nodes have zero features
+    for k, v in padded.ndata.items():
+        assert (v[g.num_nodes:] == 0).all()
