@@ -774,6 +774,59 @@ at::Tensor gemm2(at::Tensor A, at::Tensor W, c10::optional<at::Tensor> bias,
 }
 
 
+void launch_lmhead_ce_fwd(const __hip_bfloat16*, const __hip_bfloat16*, const int*,
+                          float, long, int, int, int, float*, float*, float*,
+                          float*, float*, int, hipStream_t);
+void launch_lmhead_ce_bwd(const __hip_bfloat16*, const __hip_bfloat16*, const int*,
+                          const float*, const float*, float, long, int, int, int,
+                          __hip_bfloat16*, hipStream_t);
+
+// K21 forward: per-row (loss, lse) of CE(softmax(scale * h @ Wp^T), target)
+// without materializing logits. h (M,K) bf16, Wp (Vp,K) bf16 zero-padded
+// past V, targets (M,) int32 with -1 = ignore. csrc/lmhead_ce.hip.
+std::vector<at::Tensor> lmhead_ce_fwd(at::Tensor h, at::Tensor Wp, at::Tensor targets,
+                                      double scale, int64_t V) {
+  CHECK_GPU(h);
+  CHECK_GPU(Wp);
+  TORCH_CHECK(h.scalar_type() == at::kBFloat16 && Wp.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(targets.scalar_type() == at::kInt && targets.is_contiguous());
+  const long M = h.size(0);
+  const int K = h.size(1), Vp = Wp.size(0);
+  TORCH_CHECK(Wp.size(1) == K && M % 128 == 0 && K % 64 == 0 && Vp % 128 == 0,
+              "lmhead_ce geometry violated: ", M, "x", K, " vocab ", Vp);
+  TORCH_CHECK(targets.numel() == M);
+  const int n_chunks = (Vp / 128 + 7) / 8;
+  auto fopt = h.options().dtype(at::kFloat);
+  auto pmax = at::empty({M, n_chunks}, fopt);
+  auto psum = at::empty({M, n_chunks}, fopt);
+  auto ptgt = at::empty({M, n_chunks}, fopt);
+  auto lse = at::empty({M}, fopt);
+  auto loss = at::empty({M}, fopt);
+  launch_lmhead_ce_fwd(ptr<bf16_t>(h), ptr<bf16_t>(Wp), targets.data_ptr<int>(),
+                       (float)scale, M, K, (int)V, Vp, pmax.data_ptr<float>(),
+                       psum.data_ptr<float>(), ptgt.data_ptr<float>(),
+                       lse.data_ptr<float>(), loss.data_ptr<float>(), n_chunks,
+                       cur_stream());
+  return {loss, lse};
+}
+
+// K21 backward: dlogits (M, Vp) bf16 = (softmax - onehot(target)) * gscale,
+// recomputed tile-by-tile (cols >= V zeroed).
+at::Tensor lmhead_ce_bwd(at::Tensor h, at::Tensor Wp, at::Tensor targets,
+                         at::Tensor lse, at::Tensor gscale, double scale, int64_t V) {
+  CHECK_GPU(h);
+  const long M = h.size(0);
+  const int K = h.size(1), Vp = Wp.size(0);
+  TORCH_CHECK(lse.scalar_type() == at::kFloat && gscale.scalar_type() == at::kFloat);
+  TORCH_CHECK(lse.numel() == M && gscale.numel() == M);
+  auto dlogits = at::empty({M, (long)Vp}, h.options());
+  launch_lmhead_ce_bwd(ptr<bf16_t>(h), ptr<bf16_t>(Wp), targets.data_ptr<int>(),
+                       lse.data_ptr<float>(), gscale.data_ptr<float>(),
+                       (float)scale, M, K, (int)V, Vp,
+                       mptr<bf16_t>(dlogits), cur_stream());
+  return dlogits;
+}
+
 void adamw_fused(at::Tensor p, at::Tensor g, at::Tensor m, at::Tensor v, double lr,
                  double beta1, double beta2, double eps, double weight_decay, at::Tensor bc,
                  bool l2_mode) {
@@ -814,6 +867,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_wgrad", &rmsnorm_wgrad);
   m.def("flash_attn_fwd", &flash_attn_fwd);
   m.def("flash_attn_bwd", &flash_attn_bwd);
+  m.def("lmhead_ce_fwd", &lmhead_ce_fwd);
+  m.def("lmhead_ce_bwd", &lmhead_ce_bwd);
   m.def("dropout_add_fwd", [](at::Tensor h, at::Tensor res, double p, int64_t seed) {
     CHECK_GPU(h);
     CHECK_GPU(res);

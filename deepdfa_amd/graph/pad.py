@@ -56,12 +56,13 @@ def pad_batch(
     assert b_extra >= 1 and n_extra >= b_extra and e_extra >= b_extra, (
         f"padding target too small: ({n_pad},{e_pad},{b_pad}) for ({N},{E},{B})"
     )
+    dev = g.indptr.device
     # dummy graph 0 absorbs the surplus nodes; the rest get 1 node each
     nodes_per_dummy = [n_extra - (b_extra - 1)] + [1] * (b_extra - 1)
     node_offsets = torch.cat(
         [
-            g.node_offsets.to(torch.int64),
-            N + torch.cumsum(torch.tensor(nodes_per_dummy, dtype=torch.int64), 0),
+            g.node_offsets.to(dev).to(torch.int64),
+            N + torch.cumsum(torch.tensor(nodes_per_dummy, dtype=torch.int64, device=dev), 0),
         ]
     ).to(torch.int32)
 
@@ -69,11 +70,11 @@ def pad_batch(
     # the LAST dummy node (so CSR row pointers stay monotone)
     base_loops = n_extra
     surplus = e_extra - base_loops
-    counts = torch.ones(n_extra, dtype=torch.int64)
+    counts = torch.ones(n_extra, dtype=torch.int64, device=dev)
     counts[-1] += surplus
     pad_indptr = g.indptr[-1].to(torch.int64) + torch.cumsum(counts, 0)
     loop_src = torch.repeat_interleave(
-        torch.arange(N, N + n_extra, dtype=torch.int64), counts
+        torch.arange(N, N + n_extra, dtype=torch.int64, device=dev), counts
     ).to(torch.int32)
     indptr = torch.cat([g.indptr.to(torch.int64), pad_indptr]).to(torch.int32)
     indices = torch.cat([g.indices, loop_src])
@@ -87,7 +88,7 @@ def pad_batch(
         pad_shape = (n_extra,) + tuple(v.shape[1:])
         ndata[k] = torch.cat([v, torch.zeros(pad_shape, dtype=v.dtype, device=v.device)])
 
-    weight = torch.zeros(b_pad, dtype=torch.float32)
+    weight = torch.zeros(b_pad, dtype=torch.float32, device=dev)
     weight[:B] = 1.0
     out = BatchedCFG(node_offsets, indptr, indices, t_indptr, t_indices, ndata)
     assert out.num_nodes == n_pad and out.num_edges == e_pad and out.num_graphs == b_pad

@@ -313,9 +313,17 @@ class T5ForConditionalGeneration(nn.Module):
         dec = self.decoder(decoder_input_ids, dec_valid, enc=enc, enc_valid=enc_valid)
         if output_hidden_only:
             return dec
-        h = dec
-        if self.config.tie_word_embeddings:
-            h = h * (self.config.d_model ** -0.5)
+        scale = self.config.d_model ** -0.5 if self.config.tie_word_embeddings else 1.0
+        if labels is not None:
+            # K21 fused LM-head GEMM + CE (csrc/lmhead_ce.hip): logits are
+            # never materialized (the reference path builds b*512*32100
+            # fp32 twice, CodeT5/models.py:140-149); eager fallback inside
+            from ..ops.transformer import lmhead_ce_usable, lmhead_cross_entropy
+
+            if lmhead_ce_usable(dec, self.lm_head.weight):
+                loss = lmhead_cross_entropy(dec, self.lm_head.weight, labels, scale)
+                return loss, None, dec
+        h = dec if scale == 1.0 else dec * scale
         logits = self.lm_head(h)
         loss = None
         if labels is not None:

@@ -513,3 +513,81 @@ def flash_attention_qkv(qkv, num_heads, valid=None, bias=None, scale=1.0,
                         causal=False, dropout_p=0.0):
     return _FlashAttentionQKV.apply(qkv, num_heads, valid, bias, scale, causal,
                                     dropout_p)
+
+
+class _LMHeadCE(torch.autograd.Function):
+    """K21: fused LM-head GEMM + cross-entropy over the 32100-token vocab
+    (csrc/lmhead_ce.hip). Forward streams vocab tiles with online
+    logsumexp — logits are never materialized (the reference call site,
+    CodeT5/models.py:140-149, materializes b*512*32100 fp32 twice).
+    Backward recomputes tiles into scaled bf16 dlogits, then dh/dW are two
+    library GEMMs. Semantics == F.cross_entropy(logits.float(), targets,
+    ignore_index=-100) with logits = scale * h @ W^T (mean over valid)."""
+
+    @staticmethod
+    def forward(ctx, h, weight, targets, scale):
+        ext = load_ext(required=True)
+        K = h.shape[-1]
+        h2d = h.reshape(-1, K).to(torch.bfloat16).contiguous()
+        V = weight.shape[0]
+        Vp = (V + 127) // 128 * 128
+        # padded bf16 vocab-weight cache (same invalidation protocol as
+        # _LinearBf16: parameter version + weights epoch + capture refresh)
+        cache = getattr(weight, "_dfa_vocab_cache", None)
+        key = (weight._version, _weights_epoch[0])
+        if cache is None:
+            Wp = torch.zeros(Vp, K, dtype=torch.bfloat16, device=weight.device)
+            Wp[:V].copy_(weight.detach())
+            weight._dfa_vocab_cache = (key, Wp)
+        elif CAPTURE_REFRESH[0] or cache[0] != key:
+            _, Wp = cache
+            Wp[:V].copy_(weight.detach())
+            weight._dfa_vocab_cache = (key, Wp)
+        else:
+            _, Wp = cache
+        tgt = targets.reshape(-1).to(torch.int32)
+        tgt = torch.where(tgt == -100, tgt.new_full((), -1), tgt).contiguous()
+        loss_rows, lse = ext.lmhead_ce_fwd(h2d, Wp, tgt, float(scale), V)
+        n_valid = (tgt >= 0).sum().to(torch.float32).clamp(min=1.0)
+        loss = loss_rows.sum() / n_valid
+        ctx.save_for_backward(h2d, Wp, tgt, lse, n_valid)
+        ctx.scale = float(scale)
+        ctx.V = V
+        ctx.h_shape = h.shape
+        ctx.h_dtype = h.dtype
+        ctx.w_dtype = weight.dtype
+        return loss
+
+    @staticmethod
+    def backward(ctx, grad):
+        ext = load_ext(required=True)
+        h2d, Wp, tgt, lse, n_valid = ctx.saved_tensors
+        # fold the chain scale into gscale once: it multiplies both dh and dW
+        g = grad.float() * ctx.scale / n_valid
+        gscale = torch.where(tgt >= 0, g, torch.zeros_like(g)).contiguous()
+        dlogits = ext.lmhead_ce_bwd(h2d, Wp, tgt, lse, gscale, ctx.scale, ctx.V)
+        dh = torch.matmul(dlogits, Wp)            # (M, K) bf16
+        dw = torch.matmul(dlogits.t(), h2d)       # (Vp, K) bf16
+        return (
+            dh.view(ctx.h_shape).to(ctx.h_dtype),
+            dw[: ctx.V].to(ctx.w_dtype),
+            None,
+            None,
+        )
+
+
+def lmhead_ce_usable(h, weight) -> bool:
+    if not (h.is_cuda and load_ext() is not None):
+        return False
+    M = h.numel() // h.shape[-1]
+    return M % 128 == 0 and h.shape[-1] % 64 == 0
+
+
+def lmhead_cross_entropy(h, weight, targets, scale: float = 1.0):
+    """Mean CE over rows where target != -100, logits = scale * h @ W^T."""
+    if lmhead_ce_usable(h, weight):
+        return _LMHeadCE.apply(h, weight, targets, scale)
+    logits = torch.nn.functional.linear(h.reshape(-1, h.shape[-1]) * scale, weight)
+    return torch.nn.functional.cross_entropy(
+        logits.float(), targets.reshape(-1), ignore_index=-100
+    )

@@ -186,10 +186,12 @@ class Trainer:
             n_batches = 0
             loss_sum = 0.0
             for batch in train_loader:
-                batch = self._to_device(batch)
                 if captured is not None:
+                    # padding happens on the CPU batch; the captured step
+                    # copies the padded tensors into its static device buffers
                     captured(batch)
                 else:
+                    batch = self._to_device(batch)
                     with self._autocast():
                         loss = model.training_step(batch)
                     optimizer.zero_grad(set_to_none=True)

@@ -28,6 +28,7 @@ ext = CUDAExtension(
         "csrc/gemm_bf16.hip",
         "csrc/wgrad2.hip",
         "csrc/adamw.hip",
+        "csrc/lmhead_ce.hip",
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
