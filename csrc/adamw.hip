@@ -14,8 +14,13 @@ using f4 = __attribute__((ext_vector_type(4))) float;
 extern "C" __global__ void adamw_fused_kernel(
     float* __restrict__ p, const float* __restrict__ g, float* __restrict__ m,
     float* __restrict__ v, long n, float lr, float beta1, float beta2,
-    float eps, float weight_decay, const float* __restrict__ bc, int l2_mode) {
-  const float bc1 = bc[0], bc2 = bc[1];
+    float eps, float weight_decay, const float* __restrict__ step, int l2_mode) {
+  // bias corrections from the device-side step counter (pre-incremented by
+  // the optimizer): correct under hipGraph replay, and folds the former
+  // pow/neg/add elementwise chain (3 graph nodes) into this kernel
+  const float t = step[0];
+  const float bc1 = 1.f - __powf(beta1, t);
+  const float bc2 = 1.f - __powf(beta2, t);
   const long nvec = n / 4;
   const long stride = (long)gridDim.x * blockDim.x;
   for (long iv = (long)blockIdx.x * blockDim.x + threadIdx.x; iv < nvec;
@@ -59,13 +64,11 @@ extern "C" __global__ void adamw_fused_kernel(
 
 void launch_adamw_fused(float* p, const float* g, float* m, float* v, long n,
                         float lr, float beta1, float beta2, float eps,
-                        float weight_decay, const float* bc, int l2_mode,
+                        float weight_decay, const float* step, int l2_mode,
                         hipStream_t stream) {
-  // bc = device pointer to {1-beta1^t, 1-beta2^t}: computed on-device so a
-  // hipGraph replay advances the step count correctly
   const int block = 256;
   const int grid = (int)min((n / 4 + block - 1) / block, (long)4096);
   hipLaunchKernelGGL(adamw_fused_kernel, dim3(grid), dim3(block), 0, stream, p,
-                     g, m, v, n, lr, beta1, beta2, eps, weight_decay, bc,
+                     g, m, v, n, lr, beta1, beta2, eps, weight_decay, step,
                      l2_mode);
 }

@@ -31,6 +31,13 @@ void launch_gru_gates2_bwd(const T*, const T*, const T*, const T*, const T*, con
                            long, int, hipStream_t);
 template <typename T>
 void launch_colsum(const T*, float*, int, int, hipStream_t);
+template <typename T>
+void launch_spmm_sum_strided(const int*, const int*, const T*, T*, int, int, long,
+                             hipStream_t);
+void launch_gemm_bias2(const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*,
+                       const __hip_bfloat16*, const __hip_bfloat16*, long,
+                       const __hip_bfloat16*, long, __hip_bfloat16*, int, int, int, int,
+                       hipStream_t);
 void launch_gemm_bias(const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*,
                       const __hip_bfloat16*, const __hip_bfloat16*, __hip_bfloat16*, int, int,
                       int, int, hipStream_t);
@@ -335,19 +342,40 @@ at::Tensor wgrad(at::Tensor A, at::Tensor B) {
   return out;
 }
 
+void launch_pack_gru_weights(const float*, const float*, const float*, const float*,
+                             const float*, const float*, int, __hip_bfloat16*,
+                             __hip_bfloat16*, __hip_bfloat16*, __hip_bfloat16*,
+                             __hip_bfloat16*, __hip_bfloat16*, hipStream_t);
+
+// one-launch refresh of every derived GGNN weight buffer (bf16 casts,
+// Wcat/WcatT block matrix, merged bias) — see ops/flowgnn.py cache
+void pack_gru_weights(at::Tensor W_e, at::Tensor b_e, at::Tensor W_ih, at::Tensor W_hh,
+                      at::Tensor b_ih, at::Tensor b_hh, at::Tensor w_e16, at::Tensor b_e16,
+                      at::Tensor Wcat, at::Tensor WcatT, at::Tensor b_cat, at::Tensor W_eT) {
+  CHECK_GPU(W_e);
+  const int H = W_e.size(0);
+  TORCH_CHECK(W_e.scalar_type() == at::kFloat && W_ih.scalar_type() == at::kFloat,
+              "pack_gru_weights reads fp32 master weights");
+  TORCH_CHECK(Wcat.size(0) == 4 * H && Wcat.size(1) == 2 * H);
+  launch_pack_gru_weights(W_e.data_ptr<float>(), b_e.data_ptr<float>(),
+                          W_ih.data_ptr<float>(), W_hh.data_ptr<float>(),
+                          b_ih.data_ptr<float>(), b_hh.data_ptr<float>(), H,
+                          mptr<bf16_t>(w_e16), mptr<bf16_t>(b_e16), mptr<bf16_t>(Wcat),
+                          mptr<bf16_t>(WcatT), mptr<bf16_t>(b_cat), mptr<bf16_t>(W_eT),
+                          cur_stream());
+}
+
 std::vector<at::Tensor> ggnn_fused_fwd(at::Tensor indptr, at::Tensor indices, at::Tensor x,
-                                       at::Tensor W_e, at::Tensor b_e, at::Tensor W_ih,
-                                       at::Tensor W_hh, at::Tensor b_ih, at::Tensor b_hh,
-                                       long n_steps) {
+                                       at::Tensor W_e, at::Tensor b_e, at::Tensor Wcat,
+                                       at::Tensor b_cat, long n_steps) {
   CHECK_GPU(x);
   TORCH_CHECK(x.scalar_type() == at::kBFloat16, "fused GGNN path is bf16");
   const long N = x.size(0);
   const long H = x.size(1);
   TORCH_CHECK(H % 64 == 0 && (4 * H) % 128 == 0, "H must suit the MFMA tile");
+  TORCH_CHECK(Wcat.size(0) == 4 * H && Wcat.size(1) == 2 * H && b_cat.numel() == 4 * H,
+              "Wcat/b_cat must be prebuilt (ops/flowgnn.py cache)");
   auto stream = cur_stream();
-  auto Wcat = build_wcat(W_ih, W_hh, H);
-  auto b_cat = at::cat({b_ih.narrow(0, 0, 2 * H) + b_hh.narrow(0, 0, 2 * H),
-                        b_ih.narrow(0, 2 * H, H), b_hh.narrow(0, 2 * H, H)});
   const long S = n_steps;
   auto opts = x.options();
   // HH[s] = hidden state entering step s; HH[S] = final output.
@@ -380,8 +408,8 @@ std::vector<at::Tensor> ggnn_fused_fwd(at::Tensor indptr, at::Tensor indices, at
 }
 
 std::vector<at::Tensor> ggnn_fused_bwd(at::Tensor grad_out, at::Tensor t_indptr,
-                                       at::Tensor t_indices, at::Tensor x, at::Tensor W_e,
-                                       at::Tensor W_ih, at::Tensor W_hh, at::Tensor HH,
+                                       at::Tensor t_indices, at::Tensor x, at::Tensor W_eT,
+                                       at::Tensor WcatT, at::Tensor HH,
                                        at::Tensor M, at::Tensor R, at::Tensor Z, at::Tensor Nn,
                                        at::Tensor HN, long n_steps) {
   const long N = x.size(0);
@@ -389,9 +417,8 @@ std::vector<at::Tensor> ggnn_fused_bwd(at::Tensor grad_out, at::Tensor t_indptr,
   const long S = n_steps;
   const long NH = N * H;
   auto stream = cur_stream();
-  auto Wcat = build_wcat(W_ih, W_hh, H);
-  auto WcatT = Wcat.t().contiguous();      // (2H, 4H)
-  auto W_eT = W_e.t().contiguous();        // (H, H)
+  TORCH_CHECK(WcatT.size(0) == 2 * H && WcatT.size(1) == 4 * H && W_eT.size(0) == H,
+              "WcatT/W_eT must be prebuilt (ops/flowgnn.py cache)");
   auto opts = x.options();
   // per-step gate/message grads, kept for ONE batched K = S*N weight-grad
   // GEMM at the end (hipBLASLt's transpose-A kernels are the pathology the
@@ -400,7 +427,7 @@ std::vector<at::Tensor> ggnn_fused_bwd(at::Tensor grad_out, at::Tensor t_indptr,
   auto Gwh = at::empty({S, N, H}, opts);
   auto grad_h = grad_out.contiguous().clone();
   auto grad_hd = at::empty({N, H}, opts);
-  auto tmp = at::empty({N, H}, opts);
+  auto grad_A = at::empty({N, 2 * H}, opts);
   for (long s = S - 1; s >= 0; --s) {
     const bf16_t* h_in = ptr<bf16_t>(HH) + s * NH;
     auto ggic = Ggicat.select(0, s);
@@ -410,16 +437,17 @@ std::vector<at::Tensor> ggnn_fused_bwd(at::Tensor grad_out, at::Tensor t_indptr,
                                   ptr<bf16_t>(HN) + s * NH, mptr<bf16_t>(ggic),
                                   mptr<bf16_t>(grad_hd), NH, H, stream);
     // grad_A = ggic @ Wcat : (N, 2H); left half = grad_m, right = gh-path
-    auto grad_A = at::empty({N, 2 * H}, opts);
     launch_gemm_bias(ptr<bf16_t>(ggic), nullptr, ptr<bf16_t>(WcatT), nullptr, nullptr,
                      mptr<bf16_t>(grad_A), N, 4 * H, 4 * H, 2 * H, stream);
-    auto grad_m = grad_A.narrow(1, 0, H).contiguous();
-    launch_spmm_sum<bf16_t>(t_indptr.data_ptr<int>(), t_indices.data_ptr<int>(),
-                            ptr<bf16_t>(grad_m), mptr<bf16_t>(gwh), N, H, stream);
-    // grad wrt h_in = direct z-path + gh-path + (grad_wh @ W_e)
-    at::add_out(tmp, grad_hd, grad_A.narrow(1, H, H));
-    launch_gemm_bias(ptr<bf16_t>(gwh), nullptr, ptr<bf16_t>(W_eT), nullptr, ptr<bf16_t>(tmp),
-                     mptr<bf16_t>(grad_h), N, H, H, H, stream);
+    // grad_m read STRIDED out of grad_A's left half (no contiguous copy)
+    launch_spmm_sum_strided<bf16_t>(t_indptr.data_ptr<int>(), t_indices.data_ptr<int>(),
+                                    ptr<bf16_t>(grad_A), mptr<bf16_t>(gwh), N, H,
+                                    2 * H, stream);
+    // grad wrt h_in = (grad_wh @ W_e) + direct z-path + gh-path, the two
+    // additions fused into the GEMM epilogue as strided addends
+    launch_gemm_bias2(ptr<bf16_t>(gwh), nullptr, ptr<bf16_t>(W_eT), nullptr,
+                      ptr<bf16_t>(grad_hd), H, ptr<bf16_t>(grad_A) + H, 2 * H,
+                      mptr<bf16_t>(grad_h), N, H, H, H, stream);
   }
   // batched weight/bias grads over all steps (K = S*N)
   auto A_g = Ggicat.view({S * N, 4 * H});
@@ -434,8 +462,9 @@ std::vector<at::Tensor> ggnn_fused_bwd(at::Tensor grad_out, at::Tensor t_indptr,
   launch_colsum<bf16_t>(ptr<bf16_t>(A_g), cs4.data_ptr<float>(), S * N, 4 * H, stream);
   auto cs_e = at::zeros({H}, opts.dtype(at::kFloat));
   launch_colsum<bf16_t>(ptr<bf16_t>(A_w), cs_e.data_ptr<float>(), S * N, H, stream);
-  // scatter gWcat blocks back to the GRUCell weight layout
-  auto gW_ih = gWcat.narrow(0, 0, 3 * H).narrow(1, 0, H).contiguous();
+  // scatter gWcat blocks back to the GRUCell weight layout (views are fine
+  // as autograd outputs; no contiguous copy)
+  auto gW_ih = gWcat.narrow(0, 0, 3 * H).narrow(1, 0, H);
   auto gW_hh = at::cat({gWcat.narrow(0, 0, 2 * H).narrow(1, H, H),
                         gWcat.narrow(0, 3 * H, H).narrow(1, H, H)});
   auto gb_ih = cs4.narrow(0, 0, 3 * H).contiguous();
@@ -828,16 +857,16 @@ at::Tensor lmhead_ce_bwd(at::Tensor h, at::Tensor Wp, at::Tensor targets,
 }
 
 void adamw_fused(at::Tensor p, at::Tensor g, at::Tensor m, at::Tensor v, double lr,
-                 double beta1, double beta2, double eps, double weight_decay, at::Tensor bc,
+                 double beta1, double beta2, double eps, double weight_decay, at::Tensor step,
                  bool l2_mode) {
   CHECK_GPU(p);
   TORCH_CHECK(p.scalar_type() == at::kFloat && g.scalar_type() == at::kFloat);
   TORCH_CHECK(p.numel() == g.numel() && p.numel() == m.numel() && p.numel() == v.numel());
-  TORCH_CHECK(bc.is_cuda() && bc.scalar_type() == at::kFloat && bc.numel() == 2,
-              "bc must be a device float[2] = {1-b1^t, 1-b2^t}");
+  TORCH_CHECK(step.is_cuda() && step.scalar_type() == at::kFloat && step.numel() == 1,
+              "step must be a device float[1] step counter (pre-incremented)");
   launch_adamw_fused(p.data_ptr<float>(), g.data_ptr<float>(), m.data_ptr<float>(),
                      v.data_ptr<float>(), p.numel(), (float)lr, (float)beta1, (float)beta2,
-                     (float)eps, (float)weight_decay, bc.data_ptr<float>(), l2_mode ? 1 : 0,
+                     (float)eps, (float)weight_decay, step.data_ptr<float>(), l2_mode ? 1 : 0,
                      cur_stream());
 }
 
@@ -856,6 +885,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gru_gates2_fwd", &gru_gates2_fwd);
   m.def("colsum", &colsum);
   m.def("ggnn_fused_fwd", &ggnn_fused_fwd);
+  m.def("pack_gru_weights", &pack_gru_weights);
   m.def("layernorm_fwd", &layernorm_fwd);
   m.def("layernorm_bwd", &layernorm_bwd);
   m.def("layernorm_wgrad", &layernorm_wgrad);

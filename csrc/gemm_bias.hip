@@ -41,7 +41,9 @@ template <int BMT>
 __global__ __launch_bounds__(256) void gemm_bias_kernel(
     const bf16* __restrict__ A1, const bf16* __restrict__ A2,
     const bf16* __restrict__ W, const bf16* __restrict__ bias,
-    const bf16* __restrict__ addend, bf16* __restrict__ out, int N, int K,
+    const bf16* __restrict__ addend, long astride,
+    const bf16* __restrict__ addend2, long astride2,
+    bf16* __restrict__ out, int N, int K,
     int K1, int COL) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   char* a_lds = smem;                 // BMT x ROWB
@@ -125,7 +127,8 @@ __global__ __launch_bounds__(256) void gemm_bias_kernel(
         const int row = row_base + i;
         if (row < N) {
           float v = acc[m][n][i] + b;
-          if (addend) v += __bfloat162float(addend[(long)row * COL + col]);
+          if (addend) v += __bfloat162float(addend[(long)row * astride + col]);
+          if (addend2) v += __bfloat162float(addend2[(long)row * astride2 + col]);
           out[(long)row * COL + col] = __float2bfloat16(v);
         }
       }
@@ -133,19 +136,29 @@ __global__ __launch_bounds__(256) void gemm_bias_kernel(
   }
 }
 
-void launch_gemm_bias(const bf16* A1, const bf16* A2, const bf16* W,
-                      const bf16* bias, const bf16* addend, bf16* out, int N,
-                      int K, int K1, int COL, hipStream_t stream) {
+void launch_gemm_bias2(const bf16* A1, const bf16* A2, const bf16* W,
+                       const bf16* bias, const bf16* addend, long astride,
+                       const bf16* addend2, long astride2, bf16* out, int N,
+                       int K, int K1, int COL, hipStream_t stream) {
   const long blocks64 = (long)((N + BM - 1) / BM) * (COL / BN);
   if (blocks64 < 384) {
     const dim3 grid((N + 31) / 32, COL / BN);
     const size_t lds = (32 + BN) * ROWB;
     hipLaunchKernelGGL(gemm_bias_kernel<32>, grid, dim3(256), lds, stream, A1,
-                       A2, W, bias, addend, out, N, K, K1, COL);
+                       A2, W, bias, addend, astride, addend2, astride2, out, N,
+                       K, K1, COL);
   } else {
     const dim3 grid((N + BM - 1) / BM, COL / BN);
     const size_t lds = (BM + BN) * ROWB;
     hipLaunchKernelGGL(gemm_bias_kernel<BM>, grid, dim3(256), lds, stream, A1,
-                       A2, W, bias, addend, out, N, K, K1, COL);
+                       A2, W, bias, addend, astride, addend2, astride2, out, N,
+                       K, K1, COL);
   }
+}
+
+void launch_gemm_bias(const bf16* A1, const bf16* A2, const bf16* W,
+                      const bf16* bias, const bf16* addend, bf16* out, int N,
+                      int K, int K1, int COL, hipStream_t stream) {
+  launch_gemm_bias2(A1, A2, W, bias, addend, COL, nullptr, 0, out, N, K, K1,
+                    COL, stream);
 }

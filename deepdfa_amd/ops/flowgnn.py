@@ -127,24 +127,72 @@ def gru_cell(
     return _GruGates.apply(gi, gh, h.to(gi.dtype))
 
 
+def _ggnn_pack_cache(linear: torch.nn.Linear, gru: torch.nn.GRUCell):
+    """Derived-weight buffers for the fused GGNN path: bf16 casts of the
+    step linear, the block gate matrix Wcat(4H,2H) + transpose, merged
+    bias, W_e transpose — all refreshed by ONE pack kernel
+    (csrc/flowgnn_kernels.hip pack_gru_weights) under the same
+    version/epoch/capture invalidation protocol as the transformer cast
+    caches. Replaces the per-call chain of ~26 cast/cat/transpose nodes."""
+    from ._ext import load_ext
+    from .transformer import CAPTURE_REFRESH, _weights_epoch
+
+    ext = load_ext(required=True)
+    key = (
+        linear.weight._version, gru.weight_ih._version, gru.weight_hh._version,
+        linear.bias._version, gru.bias_ih._version, gru.bias_hh._version,
+        _weights_epoch[0],
+    )
+    cache = getattr(gru, "_dfa_pack_cache", None)
+    if cache is not None and not CAPTURE_REFRESH[0] and cache[0] == key:
+        return cache[1]
+    H = gru.weight_hh.shape[1]
+    dev = gru.weight_hh.device
+    bf = torch.bfloat16
+    if cache is None:
+        buf = {
+            "w_e16": torch.empty(H, H, dtype=bf, device=dev),
+            "b_e16": torch.empty(H, dtype=bf, device=dev),
+            "Wcat": torch.empty(4 * H, 2 * H, dtype=bf, device=dev),
+            "WcatT": torch.empty(2 * H, 4 * H, dtype=bf, device=dev),
+            "b_cat": torch.empty(4 * H, dtype=bf, device=dev),
+            "W_eT": torch.empty(H, H, dtype=bf, device=dev),
+        }
+    else:
+        buf = cache[1]
+    ext.pack_gru_weights(
+        linear.weight.detach(), linear.bias.detach(), gru.weight_ih.detach(),
+        gru.weight_hh.detach(), gru.bias_ih.detach(), gru.bias_hh.detach(),
+        buf["w_e16"], buf["b_e16"], buf["Wcat"], buf["WcatT"], buf["b_cat"],
+        buf["W_eT"],
+    )
+    gru._dfa_pack_cache = (key, buf)
+    return buf
+
+
 class _GGNNFused(torch.autograd.Function):
     """Whole n_steps GGNN loop as one autograd node, driven from C++
     (bindings ggnn_fused_fwd/bwd): per step one MFMA gemm_bias for W_e h,
     the CSR segment-sum, one split-A MFMA GEMM against the block weight
     matrix Wcat producing all GRU gate pre-activations, and a fused gate
-    kernel. bf16 compute, fp32-master-friendly (weight grads come back
-    fp32)."""
+    kernel. bf16 compute from cached packed weights; weight grads come
+    back fp32 (master dtype) directly — no cast chain in either
+    direction."""
 
     @staticmethod
-    def forward(ctx, x, w_e, b_e, w_ih, w_hh, b_ih, b_hh, indptr, indices, t_indptr, t_indices, n_steps):
+    def forward(ctx, x, w_e, b_e, w_ih, w_hh, b_ih, b_hh, buf, graph, n_steps):
         from ._ext import load_ext
 
         ext = load_ext(required=True)
         h_final, HH, M, R, Z, Nn, HN = ext.ggnn_fused_fwd(
-            indptr, indices, x, w_e, b_e, w_ih, w_hh, b_ih, b_hh, n_steps
+            graph.indptr, graph.indices, x, buf["w_e16"], buf["b_e16"],
+            buf["Wcat"], buf["b_cat"], n_steps,
         )
-        ctx.save_for_backward(x, w_e, w_ih, w_hh, HH, M, R, Z, Nn, HN, t_indptr, t_indices)
+        ctx.save_for_backward(x, buf["W_eT"], buf["WcatT"], HH, M, R, Z, Nn, HN,
+                              graph.t_indptr, graph.t_indices)
         ctx.n_steps = n_steps
+        ctx.grad_dtypes = (w_e.dtype, b_e.dtype, w_ih.dtype, w_hh.dtype,
+                           b_ih.dtype, b_hh.dtype)
         return h_final
 
     @staticmethod
@@ -152,35 +200,25 @@ class _GGNNFused(torch.autograd.Function):
         from ._ext import load_ext
 
         ext = load_ext(required=True)
-        x, w_e, w_ih, w_hh, HH, M, R, Z, Nn, HN, t_indptr, t_indices = ctx.saved_tensors
+        x, W_eT, WcatT, HH, M, R, Z, Nn, HN, t_indptr, t_indices = ctx.saved_tensors
         grad_x, gW_e, gb_e, gW_ih, gW_hh, gb_ih, gb_hh = ext.ggnn_fused_bwd(
-            grad_out.contiguous(), t_indptr, t_indices, x, w_e, w_ih, w_hh,
+            grad_out.contiguous(), t_indptr, t_indices, x, W_eT, WcatT,
             HH, M, R, Z, Nn, HN, ctx.n_steps,
         )
-        dt = x.dtype
-        return (
-            grad_x,
-            gW_e.to(dt), gb_e.to(dt), gW_ih.to(dt), gW_hh.to(dt), gb_ih.to(dt), gb_hh.to(dt),
-            None, None, None, None, None,
-        )
+        dts = ctx.grad_dtypes
+        grads = [gW_e, gb_e, gW_ih, gW_hh, gb_ih, gb_hh]
+        grads = [g if g.dtype == dt else g.to(dt) for g, dt in zip(grads, dts)]
+        return (grad_x, *grads, None, None, None)
 
 
 def ggnn_fused(x, graph, linear: torch.nn.Linear, gru: torch.nn.GRUCell, n_steps: int):
     """bf16 fused GGNN path (GPU only)."""
-    bf = torch.bfloat16
+    buf = _ggnn_pack_cache(linear, gru)
     return _GGNNFused.apply(
-        x.to(bf).contiguous(),
-        linear.weight.to(bf).contiguous(),
-        linear.bias.to(bf).contiguous(),
-        gru.weight_ih.to(bf).contiguous(),
-        gru.weight_hh.to(bf).contiguous(),
-        gru.bias_ih.to(bf).contiguous(),
-        gru.bias_hh.to(bf).contiguous(),
-        graph.indptr,
-        graph.indices,
-        graph.t_indptr,
-        graph.t_indices,
-        n_steps,
+        x.to(torch.bfloat16).contiguous(),
+        linear.weight, linear.bias, gru.weight_ih, gru.weight_hh,
+        gru.bias_ih, gru.bias_hh,
+        buf, graph, n_steps,
     )
 
 

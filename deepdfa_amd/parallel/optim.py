@@ -81,21 +81,17 @@ class FlatAdamW:
             from ..ops.transformer import bump_weights_epoch
 
             ext = load_ext(required=True)
-            # bias corrections computed ON DEVICE from a device step counter:
-            # correct under hipGraph replay (the captured add_/pow advance
-            # every replay; a python int would freeze at capture time)
+            # bias corrections computed IN the adamw kernel from a device
+            # step counter: correct under hipGraph replay (the captured
+            # add_ advances every replay; a python int would freeze at
+            # capture time), and only ONE tiny elementwise node remains
             if not hasattr(self, "_step_t") or not self._step_t.is_cuda:
-                self._step_t = torch.zeros((), device=self.flat_p.device)
+                self._step_t = torch.zeros(1, device=self.flat_p.device)
                 self._step_t.fill_(float(self.step_count - 1))
-                self._betas_t = torch.tensor([b1, b2], device=self.flat_p.device)
-                self._bc = torch.empty(2, device=self.flat_p.device)
             self._step_t.add_(1.0)
-            torch.pow(self._betas_t, self._step_t, out=self._bc)
-            torch.neg(self._bc, out=self._bc)
-            self._bc.add_(1.0)
             ext.adamw_fused(
                 self.flat_p, self.flat_g, self.m, self.v, lr, b1, b2, self.eps,
-                self.weight_decay, self._bc, self.l2_mode,
+                self.weight_decay, self._step_t, self.l2_mode,
             )
             # the raw kernel write bypasses torch version counters —
             # invalidate the bf16 weight-cast caches explicitly
@@ -133,9 +129,6 @@ class FlatAdamW:
         self.step_count = sd["step_count"]
         self.param_groups[0]["lr"] = sd["lr"]
         # the device-side step counter drives GPU bias correction — resync
-        # it (and drop the derived buffers) so a resume doesn't step with a
-        # stale count
+        # it so a resume doesn't step with a stale count
         if hasattr(self, "_step_t"):
             del self._step_t
-            del self._betas_t
-            del self._bc
