@@ -345,13 +345,19 @@ at::Tensor wgrad(at::Tensor A, at::Tensor B) {
 void launch_pack_gru_weights(const float*, const float*, const float*, const float*,
                              const float*, const float*, int, __hip_bfloat16*,
                              __hip_bfloat16*, __hip_bfloat16*, __hip_bfloat16*,
-                             __hip_bfloat16*, __hip_bfloat16*, hipStream_t);
+                             __hip_bfloat16*, __hip_bfloat16*, __hip_bfloat16*,
+                             __hip_bfloat16*, hipStream_t);
+void launch_gemm_gru(const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*,
+                     const __hip_bfloat16*, const __hip_bfloat16*, __hip_bfloat16*,
+                     __hip_bfloat16*, __hip_bfloat16*, __hip_bfloat16*, __hip_bfloat16*,
+                     int, int, int, int, hipStream_t);
 
 // one-launch refresh of every derived GGNN weight buffer (bf16 casts,
 // Wcat/WcatT block matrix, merged bias) — see ops/flowgnn.py cache
 void pack_gru_weights(at::Tensor W_e, at::Tensor b_e, at::Tensor W_ih, at::Tensor W_hh,
                       at::Tensor b_ih, at::Tensor b_hh, at::Tensor w_e16, at::Tensor b_e16,
-                      at::Tensor Wcat, at::Tensor WcatT, at::Tensor b_cat, at::Tensor W_eT) {
+                      at::Tensor Wcat, at::Tensor WcatT, at::Tensor b_cat, at::Tensor W_eT,
+                      at::Tensor Wcat_perm, at::Tensor b_perm) {
   CHECK_GPU(W_e);
   const int H = W_e.size(0);
   TORCH_CHECK(W_e.scalar_type() == at::kFloat && W_ih.scalar_type() == at::kFloat,
@@ -362,19 +368,21 @@ void pack_gru_weights(at::Tensor W_e, at::Tensor b_e, at::Tensor W_ih, at::Tenso
                           b_ih.data_ptr<float>(), b_hh.data_ptr<float>(), H,
                           mptr<bf16_t>(w_e16), mptr<bf16_t>(b_e16), mptr<bf16_t>(Wcat),
                           mptr<bf16_t>(WcatT), mptr<bf16_t>(b_cat), mptr<bf16_t>(W_eT),
+                          mptr<bf16_t>(Wcat_perm), mptr<bf16_t>(b_perm),
                           cur_stream());
 }
 
 std::vector<at::Tensor> ggnn_fused_fwd(at::Tensor indptr, at::Tensor indices, at::Tensor x,
-                                       at::Tensor W_e, at::Tensor b_e, at::Tensor Wcat,
-                                       at::Tensor b_cat, long n_steps) {
+                                       at::Tensor W_e, at::Tensor b_e, at::Tensor Wcat_perm,
+                                       at::Tensor b_perm, long n_steps) {
   CHECK_GPU(x);
   TORCH_CHECK(x.scalar_type() == at::kBFloat16, "fused GGNN path is bf16");
   const long N = x.size(0);
   const long H = x.size(1);
   TORCH_CHECK(H % 64 == 0 && (4 * H) % 128 == 0, "H must suit the MFMA tile");
-  TORCH_CHECK(Wcat.size(0) == 4 * H && Wcat.size(1) == 2 * H && b_cat.numel() == 4 * H,
-              "Wcat/b_cat must be prebuilt (ops/flowgnn.py cache)");
+  TORCH_CHECK(Wcat_perm.size(0) == 4 * H && Wcat_perm.size(1) == 2 * H &&
+              b_perm.numel() == 4 * H,
+              "Wcat_perm/b_perm must be prebuilt (ops/flowgnn.py cache)");
   auto stream = cur_stream();
   const long S = n_steps;
   auto opts = x.options();
@@ -387,7 +395,6 @@ std::vector<at::Tensor> ggnn_fused_fwd(at::Tensor indptr, at::Tensor indices, at
   auto Nn = at::empty({S, N, H}, opts);
   auto HN = at::empty({S, N, H}, opts);
   auto wh = at::empty({N, H}, opts);
-  auto gicat = at::empty({N, 4 * H}, opts);
   const long NH = N * H;
   for (long s = 0; s < S; ++s) {
     const bf16_t* h = ptr<bf16_t>(HH) + s * NH;
@@ -396,12 +403,12 @@ std::vector<at::Tensor> ggnn_fused_fwd(at::Tensor indptr, at::Tensor indices, at
     bf16_t* m = mptr<bf16_t>(M) + s * NH;
     launch_spmm_sum<bf16_t>(indptr.data_ptr<int>(), indices.data_ptr<int>(), ptr<bf16_t>(wh), m,
                             N, H, stream);
-    launch_gemm_bias(m, h, ptr<bf16_t>(Wcat), ptr<bf16_t>(b_cat), nullptr, mptr<bf16_t>(gicat),
-                     N, 2 * H, H, 4 * H, stream);
-    launch_gru_gates2_fwd<bf16_t>(ptr<bf16_t>(gicat), h, mptr<bf16_t>(HH) + (s + 1) * NH,
-                                  mptr<bf16_t>(R) + s * NH, mptr<bf16_t>(Z) + s * NH,
-                                  mptr<bf16_t>(Nn) + s * NH, mptr<bf16_t>(HN) + s * NH, NH, H,
-                                  stream);
+    // fused gate GEMM + GRU cell (gate-interleaved Wcat layout): replaces
+    // the gicat GEMM + separate gru_gates2_fwd, with fp32 pre-activations
+    launch_gemm_gru(m, h, ptr<bf16_t>(Wcat_perm), ptr<bf16_t>(b_perm), h,
+                    mptr<bf16_t>(HH) + (s + 1) * NH, mptr<bf16_t>(R) + s * NH,
+                    mptr<bf16_t>(Z) + s * NH, mptr<bf16_t>(Nn) + s * NH,
+                    mptr<bf16_t>(HN) + s * NH, N, 2 * H, H, H, stream);
   }
   auto h_final = HH.select(0, S);
   return {h_final, HH, M, R, Z, Nn, HN};

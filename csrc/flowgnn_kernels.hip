@@ -603,7 +603,8 @@ __global__ void pack_gru_weights_kernel(
     const float* __restrict__ b_ih, const float* __restrict__ b_hh, int H,
     __hip_bfloat16* __restrict__ w_e16, __hip_bfloat16* __restrict__ b_e16,
     __hip_bfloat16* __restrict__ Wcat, __hip_bfloat16* __restrict__ WcatT,
-    __hip_bfloat16* __restrict__ b_cat, __hip_bfloat16* __restrict__ W_eT) {
+    __hip_bfloat16* __restrict__ b_cat, __hip_bfloat16* __restrict__ W_eT,
+    __hip_bfloat16* __restrict__ Wcat_perm, __hip_bfloat16* __restrict__ b_perm) {
   const long HH_ = (long)H * H;
   const long WCAT = 8L * HH_;  // (4H, 2H)
   // index space: HH_ covers w_e16+W_eT, WCAT covers Wcat+WcatT (each index
@@ -636,6 +637,8 @@ __global__ void pack_gru_weights_kernel(
       const __hip_bfloat16 b = __float2bfloat16(v);
       Wcat[o] = b;
       WcatT[c * (4L * H) + r] = b;
+      // gate-interleaved layout for the fused GEMM+GRU kernel: row j*4+g
+      Wcat_perm[(rr * 4 + gate) * (2L * H) + c] = b;
       continue;
     }
     o -= WCAT;
@@ -644,7 +647,9 @@ __global__ void pack_gru_weights_kernel(
       if (o < 2 * H) v = b_ih[o] + b_hh[o];
       else if (o < 3 * H) v = b_ih[o];
       else v = b_hh[o - H];
-      b_cat[o] = __float2bfloat16(v);
+      const __hip_bfloat16 bb = __float2bfloat16(v);
+      b_cat[o] = bb;
+      b_perm[(o % H) * 4 + (o / H)] = bb;
       continue;
     }
     o -= 4L * H;
@@ -658,11 +663,12 @@ void launch_pack_gru_weights(const float* W_e, const float* b_e,
                              __hip_bfloat16* w_e16, __hip_bfloat16* b_e16,
                              __hip_bfloat16* Wcat, __hip_bfloat16* WcatT,
                              __hip_bfloat16* b_cat, __hip_bfloat16* W_eT,
+                             __hip_bfloat16* Wcat_perm, __hip_bfloat16* b_perm,
                              hipStream_t stream) {
   const long total = (long)H * H + 8L * H * H + 5L * H;
   const int block = 256;
   const int grid = (int)min((total + block - 1) / block, (long)1024);
   hipLaunchKernelGGL(pack_gru_weights_kernel, dim3(grid), dim3(block), 0,
                      stream, W_e, b_e, W_ih, W_hh, b_ih, b_hh, H, w_e16, b_e16,
-                     Wcat, WcatT, b_cat, W_eT);
+                     Wcat, WcatT, b_cat, W_eT, Wcat_perm, b_perm);
 }

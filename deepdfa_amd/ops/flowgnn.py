@@ -157,6 +157,8 @@ def _ggnn_pack_cache(linear: torch.nn.Linear, gru: torch.nn.GRUCell):
             "WcatT": torch.empty(2 * H, 4 * H, dtype=bf, device=dev),
             "b_cat": torch.empty(4 * H, dtype=bf, device=dev),
             "W_eT": torch.empty(H, H, dtype=bf, device=dev),
+            "Wcat_perm": torch.empty(4 * H, 2 * H, dtype=bf, device=dev),
+            "b_perm": torch.empty(4 * H, dtype=bf, device=dev),
         }
     else:
         buf = cache[1]
@@ -164,7 +166,7 @@ def _ggnn_pack_cache(linear: torch.nn.Linear, gru: torch.nn.GRUCell):
         linear.weight.detach(), linear.bias.detach(), gru.weight_ih.detach(),
         gru.weight_hh.detach(), gru.bias_ih.detach(), gru.bias_hh.detach(),
         buf["w_e16"], buf["b_e16"], buf["Wcat"], buf["WcatT"], buf["b_cat"],
-        buf["W_eT"],
+        buf["W_eT"], buf["Wcat_perm"], buf["b_perm"],
     )
     gru._dfa_pack_cache = (key, buf)
     return buf
@@ -186,7 +188,7 @@ class _GGNNFused(torch.autograd.Function):
         ext = load_ext(required=True)
         h_final, HH, M, R, Z, Nn, HN = ext.ggnn_fused_fwd(
             graph.indptr, graph.indices, x, buf["w_e16"], buf["b_e16"],
-            buf["Wcat"], buf["b_cat"], n_steps,
+            buf["Wcat_perm"], buf["b_perm"], n_steps,
         )
         ctx.save_for_backward(x, buf["W_eT"], buf["WcatT"], HH, M, R, Z, Nn, HN,
                               graph.t_indptr, graph.t_indices)
