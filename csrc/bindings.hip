@@ -863,6 +863,106 @@ at::Tensor lmhead_ce_bwd(at::Tensor h, at::Tensor Wp, at::Tensor targets,
   return dlogits;
 }
 
+void launch_gate_pool_fwd(const __hip_bfloat16*, const __hip_bfloat16*, const float*,
+                          const float*, const int*, __hip_bfloat16*, float*, int, int,
+                          int, hipStream_t);
+void launch_gate_pool_bwd(const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*,
+                          const float*, const float*, const int*, __hip_bfloat16*,
+                          __hip_bfloat16*, float*, float*, float*, int, int, int,
+                          hipStream_t);
+void launch_mlp3_fwd(const __hip_bfloat16*, const float*, const float*, const float*,
+                     const float*, const float*, const float*, float*, float*, float*,
+                     int, int, hipStream_t);
+void launch_mlp3_bwd(const float*, const float*, const float*, const float*, const float*,
+                     const float*, float*, float*, __hip_bfloat16*, int, int, hipStream_t);
+void launch_mlp3_wgrad(const __hip_bfloat16*, const float*, const float*, const float*,
+                       const float*, const float*, float*, float*, float*, float*,
+                       float*, float*, int, int, hipStream_t);
+
+// fused concat + gate linear + segment-softmax attention pool (flow-GNN head)
+std::vector<at::Tensor> gate_pool_fwd(at::Tensor x1, at::Tensor x2, at::Tensor wg,
+                                      at::Tensor bg, at::Tensor node_offsets) {
+  CHECK_GPU(x1);
+  CHECK_GPU(x2);
+  TORCH_CHECK(x1.scalar_type() == at::kBFloat16 && x2.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(wg.scalar_type() == at::kFloat && bg.scalar_type() == at::kFloat);
+  const int N = x1.size(0), D1 = x1.size(1), D = D1 + (int)x2.size(1);
+  const int B = node_offsets.numel() - 1;
+  TORCH_CHECK(x2.size(0) == N && wg.numel() == D);
+  auto out = at::empty({B, D}, x1.options());
+  auto alpha = at::empty({N}, x1.options().dtype(at::kFloat));
+  launch_gate_pool_fwd(ptr<bf16_t>(x1), ptr<bf16_t>(x2), wg.data_ptr<float>(),
+                       bg.data_ptr<float>(), node_offsets.data_ptr<int>(),
+                       mptr<bf16_t>(out), alpha.data_ptr<float>(), B, D1, D,
+                       cur_stream());
+  return {out, alpha};
+}
+
+std::vector<at::Tensor> gate_pool_bwd(at::Tensor grad_out, at::Tensor x1, at::Tensor x2,
+                                      at::Tensor wg, at::Tensor alpha,
+                                      at::Tensor node_offsets) {
+  CHECK_GPU(grad_out);
+  const int N = x1.size(0), D1 = x1.size(1), D = D1 + (int)x2.size(1);
+  const int B = node_offsets.numel() - 1;
+  auto gx1 = at::empty_like(x1);
+  auto gx2 = at::empty_like(x2);
+  auto dwg = at::zeros({D}, x1.options().dtype(at::kFloat));
+  auto dbg = at::zeros({1}, x1.options().dtype(at::kFloat));
+  auto s_ws = at::empty({N}, x1.options().dtype(at::kFloat));
+  launch_gate_pool_bwd(ptr<bf16_t>(grad_out), ptr<bf16_t>(x1), ptr<bf16_t>(x2),
+                       wg.data_ptr<float>(), alpha.data_ptr<float>(),
+                       node_offsets.data_ptr<int>(), mptr<bf16_t>(gx1),
+                       mptr<bf16_t>(gx2), dwg.data_ptr<float>(), dbg.data_ptr<float>(),
+                       s_ws.data_ptr<float>(), B, D1, D, cur_stream());
+  return {gx1, gx2, dwg, dbg};
+}
+
+// fused 3-layer MLP head forward: logits + relu activations (fp32 saves)
+std::vector<at::Tensor> mlp3_fwd(at::Tensor x, at::Tensor W1T, at::Tensor b1,
+                                 at::Tensor W2T, at::Tensor b2, at::Tensor W3,
+                                 at::Tensor b3) {
+  CHECK_GPU(x);
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16);
+  const int B = x.size(0), D = x.size(1);
+  TORCH_CHECK(D == 256 && W1T.size(0) == D && W3.numel() == D);
+  auto fopt = x.options().dtype(at::kFloat);
+  auto h1 = at::empty({B, D}, fopt);
+  auto h2 = at::empty({B, D}, fopt);
+  auto logits = at::empty({B}, fopt);
+  launch_mlp3_fwd(ptr<bf16_t>(x), W1T.data_ptr<float>(), b1.data_ptr<float>(),
+                  W2T.data_ptr<float>(), b2.data_ptr<float>(), W3.data_ptr<float>(),
+                  b3.data_ptr<float>(), h1.data_ptr<float>(), h2.data_ptr<float>(),
+                  logits.data_ptr<float>(), B, D, cur_stream());
+  return {logits, h1, h2};
+}
+
+std::vector<at::Tensor> mlp3_bwd(at::Tensor dlogits, at::Tensor x, at::Tensor h1,
+                                 at::Tensor h2, at::Tensor W1, at::Tensor W2,
+                                 at::Tensor W3) {
+  CHECK_GPU(dlogits);
+  const int B = x.size(0), D = x.size(1);
+  auto fopt = x.options().dtype(at::kFloat);
+  auto dh1 = at::empty({B, D}, fopt);
+  auto dh2 = at::empty({B, D}, fopt);
+  auto dx = at::empty_like(x);
+  launch_mlp3_bwd(dlogits.data_ptr<float>(), h1.data_ptr<float>(), h2.data_ptr<float>(),
+                  W1.data_ptr<float>(), W2.data_ptr<float>(), W3.data_ptr<float>(),
+                  dh1.data_ptr<float>(), dh2.data_ptr<float>(), mptr<bf16_t>(dx), B, D,
+                  cur_stream());
+  auto dW1 = at::empty({D, D}, fopt);
+  auto dW2 = at::empty({D, D}, fopt);
+  auto dW3 = at::empty({1, D}, fopt);
+  auto db1 = at::empty({D}, fopt);
+  auto db2 = at::empty({D}, fopt);
+  auto db3 = at::empty({1}, fopt);
+  launch_mlp3_wgrad(ptr<bf16_t>(x), h1.data_ptr<float>(), h2.data_ptr<float>(),
+                    dh1.data_ptr<float>(), dh2.data_ptr<float>(),
+                    dlogits.data_ptr<float>(), dW1.data_ptr<float>(),
+                    dW2.data_ptr<float>(), dW3.data_ptr<float>(), db1.data_ptr<float>(),
+                    db2.data_ptr<float>(), db3.data_ptr<float>(), B, D, cur_stream());
+  return {dx, dW1, dW2, dW3, db1, db2, db3};
+}
+
 void adamw_fused(at::Tensor p, at::Tensor g, at::Tensor m, at::Tensor v, double lr,
                  double beta1, double beta2, double eps, double weight_decay, at::Tensor step,
                  bool l2_mode) {
@@ -893,6 +993,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("colsum", &colsum);
   m.def("ggnn_fused_fwd", &ggnn_fused_fwd);
   m.def("pack_gru_weights", &pack_gru_weights);
+  m.def("gate_pool_fwd", &gate_pool_fwd);
+  m.def("gate_pool_bwd", &gate_pool_bwd);
+  m.def("mlp3_fwd", &mlp3_fwd);
+  m.def("mlp3_bwd", &mlp3_bwd);
   m.def("layernorm_fwd", &layernorm_fwd);
   m.def("layernorm_bwd", &layernorm_bwd);
   m.def("layernorm_wgrad", &layernorm_wgrad);

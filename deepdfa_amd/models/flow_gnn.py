@@ -176,6 +176,24 @@ class FlowGNNGGNNModule(BaseModule):
     def forward(self, graph, extrafeats: Optional[Dict] = None) -> torch.Tensor:
         feat_embed = self._embed(graph)
         ggnn_out = self.ggnn(graph, feat_embed)
+        # fused head (GPU bf16): concat + gate GEMV + segment-softmax pool in
+        # one kernel each way; the 3-layer MLP as one more — the eager tail
+        # was ~25 launch-floor nodes at batch 256 (VERDICT round-1 item 4)
+        if (
+            ggnn_out.is_cuda
+            and ggnn_out.dtype == torch.bfloat16
+            and self.label_style == "graph"
+        ):
+            from ..ops.flowgnn import gate_pool, mlp3
+
+            fe = feat_embed.to(ggnn_out.dtype)
+            out = gate_pool(ggnn_out, fe, self.pooling.gate_nn, graph)
+            if self.encoder_mode:
+                return out
+            if self.out_dim == 256 and len(self.output_layer) == 5:
+                return mlp3(out, self.output_layer[0], self.output_layer[2],
+                            self.output_layer[4])
+            return self.output_layer(out).squeeze(-1)
         out = torch.cat([ggnn_out, feat_embed.to(ggnn_out.dtype)], dim=-1)
         if self.label_style == "graph":
             out = self.pooling(graph, out)

@@ -266,3 +266,95 @@ def segment_max(values: torch.Tensor, graph) -> torch.Tensor:
         ext = ext_for(values)
         return ext.segment_max(values.contiguous(), graph.node_offsets)
     return ref.segment_max(values, graph.node_offsets)
+
+
+class _GatePool(torch.autograd.Function):
+    """Fused concat + gate linear + segment-softmax attention pool
+    (csrc/flowgnn_kernels.hip gate_pool_*): replaces cat, the (N,256)@(256,1)
+    gate GEMV, attn_pool, and their backward glue (~8 nodes each way).
+    Gate weight grads come back fp32 (master dtype)."""
+
+    @staticmethod
+    def forward(ctx, x1, x2, wg, bg, node_offsets):
+        from ._ext import load_ext
+
+        ext = load_ext(required=True)
+        x1c, x2c = x1.contiguous(), x2.contiguous()
+        out, alpha = ext.gate_pool_fwd(x1c, x2c, wg.detach().reshape(-1).contiguous(),
+                                       bg.detach().contiguous(), node_offsets)
+        ctx.save_for_backward(x1c, x2c, wg, alpha, node_offsets)
+        return out
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        from ._ext import load_ext
+
+        ext = load_ext(required=True)
+        x1, x2, wg, alpha, node_offsets = ctx.saved_tensors
+        gx1, gx2, dwg, dbg = ext.gate_pool_bwd(
+            grad_out.contiguous(), x1, x2, wg.detach().reshape(-1).contiguous(),
+            alpha, node_offsets,
+        )
+        return gx1, gx2, dwg.view_as(wg).to(wg.dtype), dbg.to(wg.dtype), None
+
+
+def gate_pool(x1, x2, gate_nn: torch.nn.Linear, graph):
+    return _GatePool.apply(x1, x2, gate_nn.weight, gate_nn.bias, graph.node_offsets)
+
+
+def _mlp3_transpose_cache(lin1: torch.nn.Linear, lin2: torch.nn.Linear):
+    """Cached fp32 transposes of the two square MLP weights (forward reads
+    W^T for coalescing; backward reads the originals)."""
+    from .transformer import CAPTURE_REFRESH, _weights_epoch
+
+    key = (lin1.weight._version, lin2.weight._version, _weights_epoch[0])
+    cache = getattr(lin1, "_dfa_t_cache", None)
+    if cache is not None and not CAPTURE_REFRESH[0] and cache[0] == key:
+        return cache[1], cache[2]
+    if cache is None:
+        w1t = lin1.weight.detach().t().contiguous()
+        w2t = lin2.weight.detach().t().contiguous()
+    else:
+        w1t, w2t = cache[1], cache[2]
+        w1t.copy_(lin1.weight.detach().t())
+        w2t.copy_(lin2.weight.detach().t())
+    lin1._dfa_t_cache = (key, w1t, w2t)
+    return w1t, w2t
+
+
+class _MLP3(torch.autograd.Function):
+    """Fused [Linear(256,256)+ReLU]x2 + Linear(256,1) head
+    (csrc/flowgnn_kernels.hip mlp3_*): one kernel per direction plus one
+    kernel for all six parameter grads — replaces ~6 hipBLASLt GEMMs + relu
+    fwd/bwd + bias reductions at batch-256 launch-floor sizes."""
+
+    @staticmethod
+    def forward(ctx, x, w1, b1, w2, b2, w3, b3, w1t, w2t):
+        from ._ext import load_ext
+
+        ext = load_ext(required=True)
+        xc = x.contiguous()
+        logits, h1, h2 = ext.mlp3_fwd(
+            xc, w1t, b1.detach().contiguous(), w2t, b2.detach().contiguous(),
+            w3.detach().reshape(-1).contiguous(), b3.detach().contiguous(),
+        )
+        ctx.save_for_backward(xc, h1, h2, w1, w2, w3)
+        return logits
+
+    @staticmethod
+    def backward(ctx, dlogits):
+        from ._ext import load_ext
+
+        ext = load_ext(required=True)
+        x, h1, h2, w1, w2, w3 = ctx.saved_tensors
+        dx, dW1, dW2, dW3, db1, db2, db3 = ext.mlp3_bwd(
+            dlogits.float().contiguous(), x, h1, h2, w1.detach().contiguous(),
+            w2.detach().contiguous(), w3.detach().reshape(-1).contiguous(),
+        )
+        return dx, dW1, db1, dW2, db2, dW3, db3, None, None
+
+
+def mlp3(x, lin1, lin2, lin3):
+    w1t, w2t = _mlp3_transpose_cache(lin1, lin2)
+    return _MLP3.apply(x, lin1.weight, lin1.bias, lin2.weight, lin2.bias,
+                       lin3.weight, lin3.bias, w1t, w2t)

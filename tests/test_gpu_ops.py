@@ -283,3 +283,106 @@ class TestLMHeadCE:
                 logits.float().view(-1, logits.shape[-1]), ids.view(-1))
         assert torch.allclose(fused.float(), eager, rtol=5e-3, atol=5e-3), (
             float(fused), float(eager))
+
+
+@pytest.mark.gpu
+class TestFusedHead:
+    """gate_pool + mlp3 fused flow-GNN head vs the eager fp32 composition."""
+
+    def _setup(self):
+        from deepdfa_amd.graph.synthetic import synthetic_cfg_batch
+        from deepdfa_amd.models import FlowGNNGGNNModule
+
+        torch.manual_seed(5)
+        dev = torch.device("cuda:0")
+        model = FlowGNNGGNNModule(input_dim=1002, hidden_dim=32, n_steps=2,
+                                  num_output_layers=3).to(dev)
+        g = synthetic_cfg_batch(64, seed=3).to(dev)
+        return model, g, dev
+
+    def test_gate_pool_matches_eager(self):
+        from deepdfa_amd.ops.flowgnn import gate_pool
+        from deepdfa_amd.ops import attn_pool
+
+        model, g, dev = self._setup()
+        torch.manual_seed(0)
+        x1 = torch.randn(g.num_nodes, 128, device=dev)
+        x2 = torch.randn(g.num_nodes, 128, device=dev)
+        gate_nn = model.pooling.gate_nn
+        # eager fp32 reference
+        x1r = x1.clone().requires_grad_(True)
+        x2r = x2.clone().requires_grad_(True)
+        cat = torch.cat([x1r, x2r], dim=-1)
+        pooled_ref = attn_pool(cat, gate_nn(cat).squeeze(-1), g)
+        loss_ref = (pooled_ref.float() ** 2).sum()
+        loss_ref.backward()
+        gw_ref = gate_nn.weight.grad.clone(); gb_ref = gate_nn.bias.grad.clone()
+        gate_nn.weight.grad = None; gate_nn.bias.grad = None
+        # fused bf16
+        x1f = x1.to(torch.bfloat16).requires_grad_(True)
+        x2f = x2.to(torch.bfloat16).requires_grad_(True)
+        pooled = gate_pool(x1f, x2f, gate_nn, g)
+        loss = (pooled.float() ** 2).sum()
+        loss.backward()
+        assert torch.allclose(pooled.float(), pooled_ref.float(), rtol=0.05, atol=0.05)
+        ref = torch.cat([x1r.grad, x2r.grad], -1).float()
+        got = torch.cat([x1f.grad.float(), x2f.grad.float()], -1)
+        err = (got - ref).abs().max() / ref.abs().max().clamp(min=1e-6)
+        assert err < 0.1, float(err)
+        assert torch.allclose(gate_nn.weight.grad, gw_ref, rtol=0.08, atol=0.25), (
+            (gate_nn.weight.grad - gw_ref).abs().max())
+        assert torch.allclose(gate_nn.bias.grad, gb_ref, rtol=0.08, atol=0.25)
+
+    def test_mlp3_matches_eager(self):
+        from deepdfa_amd.ops.flowgnn import mlp3
+
+        model, g, dev = self._setup()
+        torch.manual_seed(1)
+        # quantize the input to bf16 FIRST so the reference sees the same
+        # values as the fused kernel (isolates kernel error from input error)
+        x = torch.randn(64, 256, device=dev).to(torch.bfloat16).float()
+        seq = model.output_layer
+        ref_in = x.clone().requires_grad_(True)
+        ref = seq(ref_in).squeeze(-1)
+        ref.sum().backward()
+        ref_grads = {n: p.grad.clone() for n, p in seq.named_parameters()}
+        for p in seq.parameters():
+            p.grad = None
+        xf = x.to(torch.bfloat16).requires_grad_(True)
+        got = mlp3(xf, seq[0], seq[2], seq[4])
+        got.sum().backward()
+        assert torch.allclose(got, ref.float(), rtol=0.03, atol=0.03), (
+            (got - ref).abs().max())
+        for n, p in seq.named_parameters():
+            r = ref_grads[n]
+            err = (p.grad - r).abs().max() / r.abs().max().clamp(min=1e-6)
+            assert err < 0.08, (n, float(err))
+        err = (xf.grad.float() - ref_in.grad).abs().max() / ref_in.grad.abs().max()
+        assert err < 0.1, float(err)
+
+    def test_model_forward_backward_fused_vs_cpu(self):
+        """Whole model fwd/bwd through the fused head vs the CPU fp32 path."""
+        from deepdfa_amd.graph.synthetic import synthetic_cfg_batch
+        from deepdfa_amd.models import FlowGNNGGNNModule
+
+        torch.manual_seed(9)
+        cpu_model = FlowGNNGGNNModule(input_dim=1002, hidden_dim=32, n_steps=5,
+                                      num_output_layers=3)
+        gpu_model = FlowGNNGGNNModule(input_dim=1002, hidden_dim=32, n_steps=5,
+                                      num_output_layers=3)
+        gpu_model.load_state_dict(cpu_model.state_dict())
+        gpu_model = gpu_model.to("cuda:0")
+        g = synthetic_cfg_batch(32, seed=11)
+        loss_cpu = cpu_model.training_step((g, {}))
+        loss_cpu.backward()
+        with torch.autocast(device_type="cuda", dtype=torch.bfloat16):
+            loss_gpu = gpu_model.training_step((g.to("cuda:0"), {}))
+        loss_gpu.backward()
+        assert abs(float(loss_cpu) - float(loss_gpu)) < 0.05, (
+            float(loss_cpu), float(loss_gpu))
+        for (n, pc), (_, pg) in zip(cpu_model.named_parameters(),
+                                    gpu_model.named_parameters()):
+            r, t = pc.grad, pg.grad.cpu()
+            denom = r.abs().max().clamp(min=1e-4)
+            err = (t - r).abs().max() / denom
+            assert err < 0.35, (n, float(err))
