@@ -42,7 +42,7 @@ void launch_gemm_bias(const __hip_bfloat16*, const __hip_bfloat16*, const __hip_
                       const __hip_bfloat16*, const __hip_bfloat16*, __hip_bfloat16*, int, int,
                       int, int, hipStream_t);
 void launch_wgrad(const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*, float*,
-                  int, int, int, int, hipStream_t);
+                  float*, int, int, int, int, hipStream_t);
 template <typename T>
 void launch_layernorm_fwd(const T*, const float*, const float*, T*, float*, float*, long, int,
                           float, hipStream_t);
@@ -337,8 +337,8 @@ at::Tensor wgrad(at::Tensor A, at::Tensor B) {
     return out;
   }
   auto out = at::zeros({M, C}, A.options().dtype(at::kFloat));
-  launch_wgrad(ptr<bf16_t>(A), ptr<bf16_t>(B), nullptr, out.data_ptr<float>(), K, M, C, C,
-               cur_stream());
+  launch_wgrad(ptr<bf16_t>(A), ptr<bf16_t>(B), nullptr, out.data_ptr<float>(), nullptr,
+               K, M, C, C, cur_stream());
   return out;
 }
 
@@ -459,16 +459,16 @@ std::vector<at::Tensor> ggnn_fused_bwd(at::Tensor grad_out, at::Tensor t_indptr,
   // batched weight/bias grads over all steps (K = S*N)
   auto A_g = Ggicat.view({S * N, 4 * H});
   auto gWcat = at::zeros({4 * H, 2 * H}, opts.dtype(at::kFloat));
-  launch_wgrad(ptr<bf16_t>(A_g), ptr<bf16_t>(M), ptr<bf16_t>(HH), gWcat.data_ptr<float>(), S * N,
-               4 * H, 2 * H, H, stream);  // split-B form (no [m|h] concat)
+  auto cs4 = at::zeros({4 * H}, opts.dtype(at::kFloat));
+  // bias grads (column sums of the gate/message grads) ride along in the
+  // wgrad kernels' A-tile staging — no separate colsum pass
+  launch_wgrad(ptr<bf16_t>(A_g), ptr<bf16_t>(M), ptr<bf16_t>(HH), gWcat.data_ptr<float>(),
+               cs4.data_ptr<float>(), S * N, 4 * H, 2 * H, H, stream);
   auto A_w = Gwh.view({S * N, H});
   auto gW_e = at::zeros({H, H}, opts.dtype(at::kFloat));
-  launch_wgrad(ptr<bf16_t>(A_w), ptr<bf16_t>(HH), nullptr, gW_e.data_ptr<float>(), S * N, H, H,
-               H, stream);
-  auto cs4 = at::zeros({4 * H}, opts.dtype(at::kFloat));
-  launch_colsum<bf16_t>(ptr<bf16_t>(A_g), cs4.data_ptr<float>(), S * N, 4 * H, stream);
   auto cs_e = at::zeros({H}, opts.dtype(at::kFloat));
-  launch_colsum<bf16_t>(ptr<bf16_t>(A_w), cs_e.data_ptr<float>(), S * N, H, stream);
+  launch_wgrad(ptr<bf16_t>(A_w), ptr<bf16_t>(HH), nullptr, gW_e.data_ptr<float>(),
+               cs_e.data_ptr<float>(), S * N, H, H, H, stream);
   // scatter gWcat blocks back to the GRUCell weight layout (views are fine
   // as autograd outputs; no contiguous copy)
   auto gW_ih = gWcat.narrow(0, 0, 3 * H).narrow(1, 0, H);

@@ -34,10 +34,13 @@ using uint4v = __attribute__((ext_vector_type(4))) unsigned int;
 // fix as csrc/wgrad2.hip, worth 4.07e6 LDS_BANK_CONFLICT/call here
 #define W1_ADDR(k, colbyte) ((k)*256 + ((colbyte) ^ ((((k) >> 3) & 3) << 5)))
 
+// csum (optional): column sums of A (the bias gradients) accumulated by
+// the blockIdx.y == 0 blocks from the already-staged LDS tiles — removes
+// the separate colsum pass over the same 60 MB of activation grads
 __global__ __launch_bounds__(256) void wgrad_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ B1,
-    const bf16* __restrict__ B2, float* __restrict__ out, int K, int M,
-    int C, int C1, int kchunk) {
+    const bf16* __restrict__ B2, float* __restrict__ out,
+    float* __restrict__ csum, int K, int M, int C, int C1, int kchunk) {
   __shared__ char lds_a[KS * 256];
   __shared__ char lds_b[KS * 256];
 
@@ -52,6 +55,8 @@ __global__ __launch_bounds__(256) void wgrad_kernel(
   const int wc = wid & 1;
 
   f32x4 acc[4][4] = {};
+  float cs_acc = 0.f;
+  const bool do_csum = (csum != nullptr) && (blockIdx.y == 0) && (tid < 128);
 
   for (int k0 = k_begin; k0 < k_end; k0 += KS) {
     // stage [KS][128] tiles (row-major, coalesced 16B; 16 threads/row)
@@ -78,6 +83,14 @@ __global__ __launch_bounds__(256) void wgrad_kernel(
     }
     __syncthreads();
 
+    if (do_csum) {
+      const int klim = min(KS, k_end - k0);
+#pragma unroll 8
+      for (int k = 0; k < klim; ++k)
+        cs_acc += __bfloat162float(
+            *reinterpret_cast<const __bf16*>(lds_a + W1_ADDR(k, tid * 2)));
+    }
+
     bf16x8 a_frag[4], b_frag[4];
 #pragma unroll
     for (int f = 0; f < 4; ++f) {
@@ -99,6 +112,7 @@ __global__ __launch_bounds__(256) void wgrad_kernel(
     __syncthreads();
   }
 
+  if (do_csum && m0 + tid < M) atomicAdd(csum + m0 + tid, cs_acc);
   // epilogue: D col = lane&15, row = (lane>>4)*4 + i; atomic fp32 accumulate
 #pragma unroll
   for (int fm = 0; fm < 4; ++fm) {
@@ -117,7 +131,8 @@ __global__ __launch_bounds__(256) void wgrad_kernel(
 }
 
 void launch_wgrad(const bf16* A, const bf16* B1, const bf16* B2, float* out,
-                  int K, int M, int C, int C1, hipStream_t stream) {
+                  float* csum, int K, int M, int C, int C1,
+                  hipStream_t stream) {
   // size the K-split so the grid comfortably fills 256 CUs
   const int tiles = ((M + 127) / 128) * ((C + 127) / 128);
   int zsplit = max(1, 512 / tiles);
@@ -129,5 +144,5 @@ void launch_wgrad(const bf16* A, const bf16* B1, const bf16* B2, float* out,
   zsplit = (K + kchunk - 1) / kchunk;
   const dim3 grid((M + 127) / 128, (C + 127) / 128, zsplit);
   hipLaunchKernelGGL(wgrad_kernel, grid, dim3(256), 0, stream, A, B1, B2, out,
-                     K, M, C, C1, kchunk);
+                     csum, K, M, C, C1, kchunk);
 }
