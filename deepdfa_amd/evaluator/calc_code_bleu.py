@@ -1,11 +1,11 @@
 """CodeBLEU: weighted combination of ngram / weighted-ngram / AST / dataflow
 match (reference CodeT5/evaluator/CodeBLEU/calc_code_bleu.py:1-81).
 
-The AST and dataflow components need tree_sitter language parsers; this
-environment has none, so those components are gated — when unavailable the
-remaining weights are renormalized (and the report says which components
-ran). The weighted-ngram component uses the C keyword list (reference
-keywords/ directory semantics) with 4x weight on keywords.
+All four components run natively: the AST and dataflow components use the
+self-contained C parser / DFG extractor (cparser.py, dfg_c.py) instead of
+the reference's tree-sitter build (parser/build.sh — unavailable offline).
+The weighted-ngram component uses the C keyword list (reference keywords/
+directory semantics) with 4x weight on keywords.
 """
 
 from __future__ import annotations
@@ -45,40 +45,38 @@ def weighted_ngram_match(references: List[Sequence[str]],
     return math.exp(log_p / 4.0)
 
 
-def _tree_sitter_available() -> bool:
-    try:
-        import tree_sitter  # noqa: F401
+def syntax_match(references: List[str], candidates: List[str]) -> float:
+    """AST subtree match over the native C parser (reference
+    syntax_match.py:26-75 algorithm; no tree-sitter dependency)."""
+    from .dfg_c import corpus_syntax_match
 
-        return True
-    except ImportError:
-        return False
-
-
-def syntax_match(references, candidates) -> float:  # pragma: no cover - gated
-    raise RuntimeError("syntax_match needs tree_sitter (not installed)")
+    return corpus_syntax_match([[r] for r in references], candidates)
 
 
-def dataflow_match(references, candidates) -> float:  # pragma: no cover - gated
-    raise RuntimeError("dataflow_match needs tree_sitter (not installed)")
+def dataflow_match(references: List[str], candidates: List[str]) -> float:
+    """Normalized DFG-triple match (reference dataflow_match.py:28-147
+    algorithm over the native C DFG extractor)."""
+    from .dfg_c import corpus_dataflow_match
+
+    return corpus_dataflow_match([[r] for r in references], candidates)
 
 
 def calc_code_bleu(
     references: List[str], candidates: List[str],
     weights=(0.25, 0.25, 0.25, 0.25),
 ) -> Dict[str, float]:
+    """All four CodeBLEU components with the reference weighting
+    (calc_code_bleu.py:60-64): alpha*ngram + beta*weighted_ngram +
+    gamma*syntax + theta*dataflow."""
     refs = [r.split() for r in references]
     cands = [c.split() for c in candidates]
     comps = {
         "ngram_match": bleu(refs, cands),
         "weighted_ngram_match": weighted_ngram_match(refs, cands),
+        "syntax_match": syntax_match(references, candidates),
+        "dataflow_match": dataflow_match(references, candidates),
     }
-    used_w = [weights[0], weights[1]]
-    if _tree_sitter_available():  # pragma: no cover
-        comps["syntax_match"] = syntax_match(references, candidates)
-        comps["dataflow_match"] = dataflow_match(references, candidates)
-        used_w += [weights[2], weights[3]]
-    total_w = sum(used_w)
-    score = sum(w / total_w * v for w, v in zip(used_w, comps.values()))
+    score = sum(w * v for w, v in zip(weights, comps.values()))
     comps["code_bleu"] = score
-    comps["components_used"] = len(used_w)
+    comps["components_used"] = 4
     return comps

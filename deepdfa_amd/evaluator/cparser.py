@@ -1,0 +1,544 @@
+"""Self-contained C-subset lexer + parser for CodeBLEU's AST and dataflow
+components.
+
+The reference evaluator (CodeT5/evaluator/CodeBLEU/syntax_match.py,
+dataflow_match.py, parser/DFG.py) drives a tree-sitter grammar compiled at
+install time (parser/build.sh); this environment has no tree-sitter, so
+this module provides the same capability natively: a tolerant
+recursive-descent parser over a C subset producing typed AST nodes with
+token positions, plus s-expressions matching tree-sitter's subtree-shape
+idea. Robustness contract: parsing NEVER raises — unparseable stretches
+degrade to flat `error` nodes so partial credit still accrues (generated
+code is frequently ill-formed).
+"""
+
+from __future__ import annotations
+
+import re
+from dataclasses import dataclass, field
+from typing import List, Optional, Tuple
+
+C_KEYWORDS = {
+    "auto", "break", "case", "char", "const", "continue", "default", "do",
+    "double", "else", "enum", "extern", "float", "for", "goto", "if", "int",
+    "long", "register", "return", "short", "signed", "sizeof", "static",
+    "struct", "switch", "typedef", "union", "unsigned", "void", "volatile",
+    "while", "inline", "restrict", "_Bool",
+}
+
+_TYPE_KEYWORDS = {
+    "void", "char", "short", "int", "long", "float", "double", "signed",
+    "unsigned", "struct", "union", "enum", "const", "static", "extern",
+    "auto", "register", "volatile", "inline", "restrict", "_Bool",
+}
+
+# longest-first operator list
+_OPS = [
+    ">>=", "<<=", "...", "->", "++", "--", "<<", ">>", "<=", ">=", "==",
+    "!=", "&&", "||", "+=", "-=", "*=", "/=", "%=", "&=", "|=", "^=",
+    "+", "-", "*", "/", "%", "=", "<", ">", "!", "~", "&", "|", "^", "?",
+    ":", ";", ",", ".", "(", ")", "[", "]", "{", "}",
+]
+
+_TOKEN_RE = re.compile(
+    r"""
+    (?P<ws>\s+)
+  | (?P<comment>//[^\n]*|/\*.*?\*/)
+  | (?P<string>"(?:\\.|[^"\\])*")
+  | (?P<char>'(?:\\.|[^'\\])*')
+  | (?P<number>(?:0[xX][0-9a-fA-F]+|\d+\.\d*(?:[eE][-+]?\d+)?|\.\d+|\d+)[uUlLfF]*)
+  | (?P<ident>[A-Za-z_]\w*)
+  | (?P<op>""" + "|".join(re.escape(o) for o in _OPS) + r""")
+  | (?P<other>.)
+    """,
+    re.VERBOSE | re.DOTALL,
+)
+
+
+@dataclass
+class Token:
+    kind: str  # ident/number/string/char/op/keyword
+    text: str
+    idx: int  # token index in the stream
+
+
+def remove_comments(code: str) -> str:
+    out = []
+    for m in _TOKEN_RE.finditer(code):
+        if m.lastgroup == "comment":
+            out.append(" ")
+        else:
+            out.append(m.group())
+    return "".join(out)
+
+
+def tokenize(code: str) -> List[Token]:
+    toks: List[Token] = []
+    for m in _TOKEN_RE.finditer(code):
+        kind = m.lastgroup
+        if kind in ("ws", "comment"):
+            continue
+        text = m.group()
+        if kind == "ident" and text in C_KEYWORDS:
+            kind = "keyword"
+        if kind == "other":
+            kind = "op"
+        toks.append(Token(kind, text, len(toks)))
+    return toks
+
+
+@dataclass
+class Node:
+    type: str
+    children: List["Node"] = field(default_factory=list)
+    token: Optional[Token] = None  # leaves only
+
+    def sexp(self) -> str:
+        if not self.children:
+            return f"({self.type})"
+        return f"({self.type} " + " ".join(c.sexp() for c in self.children) + ")"
+
+    def walk(self):
+        yield self
+        for c in self.children:
+            yield from c.walk()
+
+    def leaves(self):
+        if self.token is not None and not self.children:
+            yield self
+        for c in self.children:
+            yield from c.leaves()
+
+
+def _leaf(tok: Token) -> Node:
+    type_map = {
+        "ident": "identifier",
+        "number": "number_literal",
+        "string": "string_literal",
+        "char": "char_literal",
+        "keyword": tok.text,
+    }
+    return Node(type_map.get(tok.kind, tok.text), token=tok)
+
+
+_ASSIGN_OPS = {"=", "+=", "-=", "*=", "/=", "%=", "&=", "|=", "^=", "<<=", ">>="}
+_BINARY_LEVELS = [
+    {"||"}, {"&&"}, {"|"}, {"^"}, {"&"}, {"==", "!="},
+    {"<", ">", "<=", ">="}, {"<<", ">>"}, {"+", "-"}, {"*", "/", "%"},
+]
+_UNARY_OPS = {"+", "-", "!", "~", "*", "&", "++", "--"}
+
+
+class _Parser:
+    def __init__(self, toks: List[Token]):
+        self.toks = toks
+        self.i = 0
+
+    # -- token helpers --------------------------------------------------------
+
+    def peek(self, off=0) -> Optional[Token]:
+        j = self.i + off
+        return self.toks[j] if j < len(self.toks) else None
+
+    def at(self, *texts) -> bool:
+        t = self.peek()
+        return t is not None and t.text in texts
+
+    def take(self) -> Token:
+        t = self.toks[self.i]
+        self.i += 1
+        return t
+
+    def expect(self, text) -> Optional[Node]:
+        if self.at(text):
+            return _leaf(self.take())
+        return None
+
+    # -- declarations / top level ---------------------------------------------
+
+    def parse(self) -> Node:
+        items = []
+        guard = -1
+        while self.i < len(self.toks):
+            if self.i == guard:  # no progress: consume as error leaf
+                items.append(Node("error", [_leaf(self.take())]))
+            guard = self.i
+            n = self.external_decl()
+            if n is not None:
+                items.append(n)
+        return Node("translation_unit", items)
+
+    def _looks_like_type(self) -> bool:
+        t = self.peek()
+        if t is None:
+            return False
+        if t.kind == "keyword" and t.text in _TYPE_KEYWORDS:
+            return True
+        # `Foo *x` / `Foo x` heuristic: ident ident / ident * ident
+        if t.kind == "ident":
+            t1, t2 = self.peek(1), self.peek(2)
+            if t1 is not None and t1.kind == "ident":
+                return True
+            if (t1 is not None and t1.text == "*" and t2 is not None
+                    and t2.kind == "ident"):
+                return True
+        return False
+
+    def external_decl(self) -> Optional[Node]:
+        if self.at("{"):
+            return self.compound()
+        if self._looks_like_type():
+            mark = self.i
+            spec = self.type_spec()
+            decl = self.declarator()
+            if decl is not None and self.at("("):
+                params = self.param_list()
+                if self.at("{"):
+                    body = self.compound()
+                    return Node("function_definition", [spec, decl, params, body])
+                self.expect(";")
+                return Node("declaration", [spec, decl, params])
+            self.i = mark
+            return self.declaration()
+        return self.statement()
+
+    def type_spec(self) -> Node:
+        parts = []
+        while True:
+            t = self.peek()
+            if t is None:
+                break
+            if t.kind == "keyword" and t.text in _TYPE_KEYWORDS:
+                parts.append(_leaf(self.take()))
+                if parts[-1].type in ("struct", "union", "enum") and self.peek() is not None \
+                        and self.peek().kind == "ident":
+                    parts.append(_leaf(self.take()))
+                continue
+            if t.kind == "ident" and not parts:
+                parts.append(_leaf(self.take()))
+                continue
+            break
+        while self.at("*"):
+            parts.append(_leaf(self.take()))
+        return Node("type_specifier", parts)
+
+    def declarator(self) -> Optional[Node]:
+        while self.at("*"):
+            self.take()
+        t = self.peek()
+        if t is not None and t.kind == "ident":
+            n = _leaf(self.take())
+            while self.at("["):
+                self.take()
+                if not self.at("]"):
+                    self.expr_assign()
+                self.expect("]")
+                n = Node("array_declarator", [n])
+            return n
+        return None
+
+    def param_list(self) -> Node:
+        params = []
+        self.expect("(")
+        guard = -1
+        while not self.at(")") and self.peek() is not None:
+            if self.i == guard:
+                self.take()
+            guard = self.i
+            if self.at(","):
+                self.take()
+                continue
+            spec = self.type_spec()
+            decl = self.declarator()
+            kids = [spec] + ([decl] if decl else [])
+            params.append(Node("parameter_declaration", kids))
+        self.expect(")")
+        return Node("parameter_list", params)
+
+    def declaration(self) -> Node:
+        spec = self.type_spec()
+        kids: List[Node] = [spec]
+        guard = -1
+        while not self.at(";") and self.peek() is not None:
+            if self.i == guard:
+                kids.append(Node("error", [_leaf(self.take())]))
+            guard = self.i
+            decl = self.declarator()
+            if decl is None:
+                continue
+            if self.at("="):
+                self.take()
+                init = self.expr_assign()
+                kids.append(Node("init_declarator", [decl, init]))
+            else:
+                kids.append(decl)
+            if self.at(","):
+                self.take()
+        self.expect(";")
+        return Node("declaration", kids)
+
+    # -- statements -----------------------------------------------------------
+
+    def compound(self) -> Node:
+        self.expect("{")
+        items = []
+        guard = -1
+        while not self.at("}") and self.peek() is not None:
+            if self.i == guard:
+                items.append(Node("error", [_leaf(self.take())]))
+            guard = self.i
+            n = self.block_item()
+            if n is not None:
+                items.append(n)
+        self.expect("}")
+        return Node("compound_statement", items)
+
+    def block_item(self) -> Optional[Node]:
+        if self._looks_like_type():
+            return self.declaration()
+        return self.statement()
+
+    def statement(self) -> Optional[Node]:
+        t = self.peek()
+        if t is None:
+            return None
+        if t.text == "{":
+            return self.compound()
+        if t.text == ";":
+            self.take()
+            return Node("empty_statement")
+        if t.text == "if":
+            self.take()
+            self.expect("(")
+            cond = self.expression()
+            self.expect(")")
+            then = self.statement()
+            kids = [cond] + ([then] if then else [])
+            if self.at("else"):
+                self.take()
+                els = self.statement()
+                if els:
+                    kids.append(els)
+            return Node("if_statement", kids)
+        if t.text == "while":
+            self.take()
+            self.expect("(")
+            cond = self.expression()
+            self.expect(")")
+            body = self.statement()
+            return Node("while_statement", [cond] + ([body] if body else []))
+        if t.text == "do":
+            self.take()
+            body = self.statement()
+            self.expect("while")
+            self.expect("(")
+            cond = self.expression()
+            self.expect(")")
+            self.expect(";")
+            return Node("do_statement", ([body] if body else []) + [cond])
+        if t.text == "for":
+            self.take()
+            self.expect("(")
+            init = None
+            if not self.at(";"):
+                init = self.declaration() if self._looks_like_type() else Node(
+                    "expression_statement", [self.expression()])
+                if not isinstance(init, Node) or init.type != "declaration":
+                    self.expect(";")
+            else:
+                self.take()
+            cond = None if self.at(";") else self.expression()
+            self.expect(";")
+            step = None if self.at(")") else self.expression()
+            self.expect(")")
+            body = self.statement()
+            kids = [k for k in (init, cond, step, body) if k is not None]
+            return Node("for_statement", kids)
+        if t.text == "return":
+            self.take()
+            val = None if self.at(";") else self.expression()
+            self.expect(";")
+            return Node("return_statement", [val] if val else [])
+        if t.text in ("break", "continue"):
+            self.take()
+            self.expect(";")
+            return Node(f"{t.text}_statement")
+        if t.text == "goto":
+            self.take()
+            lbl = self.peek()
+            if lbl is not None and lbl.kind == "ident":
+                self.take()
+            self.expect(";")
+            return Node("goto_statement")
+        if t.text == "switch":
+            self.take()
+            self.expect("(")
+            cond = self.expression()
+            self.expect(")")
+            body = self.statement()
+            return Node("switch_statement", [cond] + ([body] if body else []))
+        if t.text in ("case", "default"):
+            self.take()
+            if t.text == "case":
+                self.expr_assign()
+            self.expect(":")
+            return Node("case_label")
+        # label?
+        if t.kind == "ident" and self.peek(1) is not None and self.peek(1).text == ":":
+            self.take()
+            self.take()
+            return Node("labeled_statement")
+        expr = self.expression()
+        self.expect(";")
+        return Node("expression_statement", [expr])
+
+    # -- expressions (precedence climbing) ------------------------------------
+
+    def expression(self) -> Node:
+        n = self.expr_assign()
+        while self.at(","):
+            self.take()
+            rhs = self.expr_assign()
+            n = Node("comma_expression", [n, rhs])
+        return n
+
+    def expr_assign(self) -> Node:
+        lhs = self.expr_ternary()
+        t = self.peek()
+        if t is not None and t.text in _ASSIGN_OPS:
+            op = self.take()
+            rhs = self.expr_assign()
+            typ = "assignment_expression" if op.text == "=" else "augmented_assignment"
+            return Node(typ, [lhs, Node("operator", token=op), rhs])
+        return lhs
+
+    def expr_ternary(self) -> Node:
+        cond = self.expr_binary(0)
+        if self.at("?"):
+            self.take()
+            a = self.expr_assign()
+            self.expect(":")
+            b = self.expr_assign()
+            return Node("conditional_expression", [cond, a, b])
+        return cond
+
+    def expr_binary(self, level: int) -> Node:
+        if level >= len(_BINARY_LEVELS):
+            return self.expr_unary()
+        n = self.expr_binary(level + 1)
+        while self.at(*_BINARY_LEVELS[level]):
+            op = self.take()
+            rhs = self.expr_binary(level + 1)
+            n = Node("binary_expression", [n, Node("operator", token=op), rhs])
+        return n
+
+    def expr_unary(self) -> Node:
+        t = self.peek()
+        if t is not None and t.text in _UNARY_OPS:
+            op = self.take()
+            operand = self.expr_unary()
+            typ = "update_expression" if op.text in ("++", "--") else "unary_expression"
+            return Node(typ, [Node("operator", token=op), operand])
+        if t is not None and t.text == "sizeof":
+            self.take()
+            if self.at("("):
+                self.take()
+                inner = self.type_spec() if self._looks_like_type() else self.expression()
+                self.expect(")")
+                return Node("sizeof_expression", [inner])
+            return Node("sizeof_expression", [self.expr_unary()])
+        # cast: ( type ) expr
+        if t is not None and t.text == "(":
+            mark = self.i
+            self.take()
+            if self._looks_like_type():
+                spec = self.type_spec()
+                if self.at(")"):
+                    self.take()
+                    if not self.at(";", ",", ")", "]", "}") and self.peek() is not None:
+                        return Node("cast_expression", [spec, self.expr_unary()])
+            self.i = mark
+        return self.expr_postfix()
+
+    def expr_postfix(self) -> Node:
+        n = self.expr_primary()
+        while True:
+            if self.at("("):
+                self.take()
+                args = []
+                guard = -1
+                while not self.at(")") and self.peek() is not None:
+                    if self.i == guard:
+                        self.take()
+                    guard = self.i
+                    args.append(self.expr_assign())
+                    if self.at(","):
+                        self.take()
+                self.expect(")")
+                n = Node("call_expression", [n, Node("argument_list", args)])
+            elif self.at("["):
+                self.take()
+                idx = self.expression() if not self.at("]") else Node("error")
+                self.expect("]")
+                n = Node("subscript_expression", [n, idx])
+            elif self.at(".", "->"):
+                self.take()
+                t = self.peek()
+                if t is not None and t.kind == "ident":
+                    n = Node("field_expression", [n, _leaf(self.take())])
+                else:
+                    break
+            elif self.at("++", "--"):
+                op = self.take()
+                n = Node("update_expression", [n, Node("operator", token=op)])
+            else:
+                break
+        return n
+
+    def expr_primary(self) -> Node:
+        t = self.peek()
+        if t is None:
+            return Node("error")
+        if t.text == "(":
+            self.take()
+            inner = self.expression()
+            self.expect(")")
+            return Node("parenthesized_expression", [inner])
+        if t.kind in ("ident", "number", "string", "char"):
+            return _leaf(self.take())
+        if t.text == "{":  # initializer list
+            self.take()
+            items = []
+            guard = -1
+            while not self.at("}") and self.peek() is not None:
+                if self.i == guard:
+                    self.take()
+                guard = self.i
+                items.append(self.expr_assign())
+                if self.at(","):
+                    self.take()
+            self.expect("}")
+            return Node("initializer_list", items)
+        return Node("error", [_leaf(self.take())])
+
+
+def parse_c(code: str) -> Node:
+    """Parse (a fragment of) C source; never raises."""
+    try:
+        return _Parser(tokenize(code)).parse()
+    except RecursionError:
+        return Node("translation_unit", [Node("error")])
+
+
+def all_subtree_sexps(root: Node) -> List[str]:
+    """All internal-node s-expressions (tree-sitter get_all_sub_trees
+    semantics: every node with children, plus the root)."""
+    out = []
+    stack = [root]
+    while stack:
+        n = stack.pop()
+        out.append(n.sexp())
+        for c in n.children:
+            if c.children:
+                stack.append(c)
+    return out

@@ -45,7 +45,7 @@ def test_bleu_properties():
     assert bleu([ref], [ref]) > 0.95
     r = calc_code_bleu(["int x = 1 ;"], ["int x = 1 ;"])
     assert r["code_bleu"] > 0.9
-    assert r["components_used"] == 2  # tree_sitter gated off here
+    assert r["components_used"] == 4  # native AST/DFG components
     # keyword weighting: a keyword MATCH outweighs an identifier match at
     # the same number of token mismatches
     kw_match = calc_code_bleu(["return x ;"], ["return y ;"])
@@ -71,3 +71,74 @@ def test_run_clone_end_to_end(tmp_path):
         "--max_source_length", "32", "--output_dir", str(tmp_path / "clone"),
     ])
     assert "train_loss" in res and "test_acc" in res
+
+
+class TestCodeBLEUFull:
+    """AST + dataflow CodeBLEU components over the native C parser
+    (reference CodeT5/evaluator/CodeBLEU parity, VERDICT round-1 item 5)."""
+
+    REF = "int f ( int n ) { int a = n + 1 ; if ( a > 2 ) { a = a * 2 ; } return a ; }"
+
+    def test_identical_scores_one(self):
+        from deepdfa_amd.evaluator.calc_code_bleu import calc_code_bleu
+
+        out = calc_code_bleu([self.REF], [self.REF])
+        assert out["components_used"] == 4
+        for k in ("ngram_match", "weighted_ngram_match", "syntax_match",
+                  "dataflow_match", "code_bleu"):
+            assert abs(out[k] - 1.0) < 1e-9, (k, out[k])
+
+    def test_alpha_renaming_invariance(self):
+        """Renamed variables keep syntax and dataflow at 1.0 while ngram
+        drops — the defining CodeBLEU property."""
+        from deepdfa_amd.evaluator.calc_code_bleu import calc_code_bleu
+
+        cand = self.REF.replace(" a ", " z ").replace(" n ", " m ")
+        out = calc_code_bleu([self.REF], [cand])
+        assert out["syntax_match"] == 1.0
+        assert out["dataflow_match"] == 1.0
+        assert out["ngram_match"] < 1.0
+        assert out["code_bleu"] < 1.0
+
+    def test_structural_difference_detected(self):
+        from deepdfa_amd.evaluator.calc_code_bleu import calc_code_bleu
+
+        out = calc_code_bleu([self.REF], ["void g ( ) { while ( x ) y ++ ; }"])
+        assert out["syntax_match"] < 0.3
+        assert out["code_bleu"] < 0.3
+
+    def test_parser_never_raises_on_garbage(self):
+        from deepdfa_amd.evaluator.cparser import parse_c
+        from deepdfa_amd.evaluator.dfg_c import get_data_flow
+
+        for junk in ["", "@@@ %% ((", "int int int", "if ( { ) }", "a +", "}{"]:
+            root = parse_c(junk)
+            assert root.type == "translation_unit"
+            get_data_flow(junk)  # must not raise
+
+    def test_dfg_extraction_semantics(self):
+        from deepdfa_amd.evaluator.dfg_c import get_data_flow
+
+        dfg = get_data_flow("int a = b ; a = a + c ;")
+        # b read (new), a comesFrom b, a/c reads, a computedFrom a,c
+        rels = [(v, r, tuple(ps)) for v, _i, r, ps, _pi in dfg]
+        assert ("a", "comesFrom", ("b",)) in rels
+        assert any(v == "a" and r == "computedFrom" and "c" in ps
+                   for v, r, ps in rels)
+
+    def test_dfg_loop_backedge(self):
+        from deepdfa_amd.evaluator.dfg_c import get_data_flow
+
+        dfg = get_data_flow("while ( i < n ) { i = i + 1 ; }")
+        # the second pass sees i defined inside the body (back-edge flow)
+        computed = [x for x in dfg if x[2] == "computedFrom" and x[0] == "i"]
+        assert computed
+
+    def test_syntax_match_partial(self):
+        from deepdfa_amd.evaluator.dfg_c import corpus_syntax_match
+
+        score = corpus_syntax_match(
+            [["int f ( ) { return a + b ; }"]],
+            ["int g ( ) { int x ; return a + b ; }"],
+        )
+        assert 0.0 < score < 1.0
