@@ -65,12 +65,26 @@ __device__ __forceinline__ int fa_swzT(int row, int byte) {
   return row * 128 + (byte ^ (((row >> 3) & 7) << 4));
 }
 
+// fwd V image: [4 key][16 d] row-major bf16 subtiles padded 128->144 B for
+// ds_read_b64_tr_b16 hardware-transpose reads (same verified layout as
+// csrc/wgrad2.hip / tools/tr_probe): staging V is then a 16-B vector write
+// per 8 elements (the old transposed image needed 8 scalar 2-B writes —
+// the fwd loop-head's dominant cost per the FSEG profile), and each PV
+// B-fragment is 2 tr reads.
+#define FWD_VSUB(kb, db) (((kb) * 4 + (db)) * 144)
+#define FWD_VBYTES ((TK / 4) * 4 * 144)
+using fwd_bf16x4v = __attribute__((ext_vector_type(4))) __bf16;
+typedef __attribute__((address_space(3))) fwd_bf16x4v fwd_lds_bf16x4;
+
 // fwd segment accumulator (env DFA_FWD_PROF=1 selects the instrumented
 // instantiation)
 __device__ unsigned long long dfa_fwd_prof[8];
 
-template <int PROF>
-__global__ __launch_bounds__(256) void flash_fwd_kernel(
+// NW = waves per block (8 preferred for L >= 256: K/V staging, barriers
+// and tile loads amortize over 2x the query rows; 4-wave fallback keeps
+// short-sequence grids filled)
+template <int PROF, int NW>
+__global__ __launch_bounds__(NW * 64, 12 / NW) void flash_fwd_kernel(
     const bf16* __restrict__ Q, const bf16* __restrict__ K,
     const bf16* __restrict__ V, const int* __restrict__ valid,
     const float* __restrict__ bias, bf16* __restrict__ O,
@@ -80,12 +94,12 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
   // contiguous tensors, 3*H*64 when they are slices of a fused QKV buffer
   extern __shared__ __attribute__((aligned(16))) char smem[];
   char* k_lds = smem;                // [TK][64] bf16 swizzled = 8 KiB
-  char* v_lds = smem + TK * 128;     // [64][TK] bf16 (transposed) = 8 KiB
+  char* v_lds = smem + TK * 128;     // [TK/4][4 key x 16 d] tr16 subtiles
 
   const int bh = blockIdx.y;          // (b, h)
   const int b = bh / H;
   const int h = bh % H;
-  const int q0 = blockIdx.x * (NWAVE * TQW);
+  const int q0 = blockIdx.x * (NW * TQW);
   const int tid = threadIdx.x;
   const int lane = tid & (WAVE - 1);
   const int wid = tid >> 6;
@@ -130,14 +144,16 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
   // T14 double-buffered K/V staging: next tile's global loads issue before
   // this tile's MFMAs; one barrier per tile (write targets the buffer the
   // NEXT iteration reads — no same-buffer hazard)
-  const int kv_end = causal ? min(vl, q0 + NWAVE * TQW) : vl;
-  const int srow = tid >> 3;           // staging: 0..31
+  const int kv_end = causal ? min(vl, q0 + NW * TQW) : vl;
+  constexpr int SROWS = NW * 8;        // staging rows per pass
+  constexpr int NPASS = TK / SROWS;
+  const int srow = tid >> 3;           // staging: 0..SROWS-1
   const int soff = (tid & 7) * 16;     // byte offset (8 bf16)
-  auto load_tile = [&](int kv0, uint4v kreg[2], uint4v vreg[2]) {
+  auto load_tile = [&](int kv0, uint4v kreg[NPASS], uint4v vreg[NPASS]) {
     const bool kv_full = (kv0 + TK) <= vl;  // uniform fast path
 #pragma unroll
-    for (int rr = 0; rr < 2; ++rr) {
-      const int key = kv0 + srow + rr * 32;
+    for (int rr = 0; rr < NPASS; ++rr) {
+      const int key = kv0 + srow + rr * SROWS;
       kreg[rr] = {};
       vreg[rr] = {};
       if (kv_full) {
@@ -153,23 +169,20 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
       }
     }
   };
-  auto write_tile = [&](char* kb, char* vb, const uint4v kreg[2],
-                        const uint4v vreg[2]) {
+  auto write_tile = [&](char* kb, char* vb, const uint4v kreg[NPASS],
+                        const uint4v vreg[NPASS]) {
 #pragma unroll
-    for (int rr = 0; rr < 2; ++rr) {
-      *reinterpret_cast<uint4v*>(kb + fa_swz(srow + rr * 32, soff)) = kreg[rr];
-      bf16 vv[8];
-      *reinterpret_cast<uint4v*>(vv) = vreg[rr];
+    for (int rr = 0; rr < NPASS; ++rr) {
+      *reinterpret_cast<uint4v*>(kb + fa_swz(srow + rr * SROWS, soff)) = kreg[rr];
+      const int key = srow + rr * SROWS;
       const int d0 = soff / 2;
-#pragma unroll
-      for (int j = 0; j < 8; ++j)
-        *reinterpret_cast<bf16*>(vb + fa_swzT(d0 + j, (srow + rr * 32) * 2)) =
-            vv[j];
+      *reinterpret_cast<uint4v*>(vb + FWD_VSUB(key >> 2, d0 >> 4) +
+                                 (key & 3) * 32 + (d0 & 15) * 2) = vreg[rr];
     }
   };
-#define K_BUF(i) (k_lds + (i)*2 * TK * 128)
-#define V_BUF(i) (v_lds + (i)*2 * TK * 128)
-  uint4v kreg[2], vreg[2];
+#define K_BUF(i) (k_lds + (i) * (TK * 128 + FWD_VBYTES))
+#define V_BUF(i) (v_lds + (i) * (TK * 128 + FWD_VBYTES))
+  uint4v kreg[NPASS], vreg[NPASS];
   int cur = 0;
   if (kv_end > 0) {
     load_tile(0, kreg, vreg);
@@ -183,6 +196,7 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
     char* k_lds_c = K_BUF(cur);
     char* v_lds_c = V_BUF(cur);
     FSEG(1)  // next-tile load issue
+    if (has_next) write_tile(K_BUF(cur ^ 1), V_BUF(cur ^ 1), kreg, vreg);
 
     // ---- S^T = K @ Q^T : D[key][qrow] ----
     f32x4 s_acc[4][2] = {};  // [fkey][fq]
@@ -195,18 +209,22 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
         const int kbyte = ks * 64 + (lane >> 4) * 16;
         k_frag[fk] = *reinterpret_cast<const bf16x8*>(k_lds_c + fa_swz(key, kbyte));
       }
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int fk = 0; fk < 4; ++fk)
 #pragma unroll
         for (int fq = 0; fq < 2; ++fq)
           s_acc[fk][fq] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               k_frag[fk], q_frag[fq][ks], s_acc[fk][fq], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
     }
     FSEG(2)  // QK^T MFMAs (+k_frag reads)
 
     // ---- masking + bias + tile row-max (over keys, per qrow column) ----
+    // scores -> probabilities transform IN s_acc: a separate p[4][2][4]
+    // copy cost 32 VGPRs (232 total = 2 waves/SIMD; the kernel is
+    // latency-bound at that occupancy, PMC WAIT_ANY 35%)
     const int qcol[2] = {qw + (lane & 15), qw + 16 + (lane & 15)};
-    float p[4][2][4];  // [fkey][fq][i] raw scores -> probabilities
     float tile_max[2] = {-3.4e38f, -3.4e38f};
 #pragma unroll
     for (int fk = 0; fk < 4; ++fk)
@@ -219,7 +237,7 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
           if (bias) s += bias[((long)h * L + qcol[fq]) * L + key];
           const bool masked = key >= vl || (causal && key > qcol[fq]) || qcol[fq] >= L;
           s = masked ? -3.4e38f : s;
-          p[fk][fq][i] = s;
+          s_acc[fk][fq][i] = s;
           tile_max[fq] = fmaxf(tile_max[fq], s);
         }
 #pragma unroll
@@ -239,10 +257,10 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
       for (int fk = 0; fk < 4; ++fk)
 #pragma unroll
         for (int i = 0; i < 4; ++i) {
-          const float e = (p[fk][fq][i] > -3.0e38f && m_new > -3.0e38f)
-                              ? __expf(p[fk][fq][i] - m_new)
+          const float e = (s_acc[fk][fq][i] > -3.0e38f && m_new > -3.0e38f)
+                              ? __expf(s_acc[fk][fq][i] - m_new)
                               : 0.f;
-          p[fk][fq][i] = e;
+          s_acc[fk][fq][i] = e;
           sum_p[fq] += e;
         }
       sum_p[fq] += __shfl_xor(sum_p[fq], 16);
@@ -263,7 +281,7 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
             const int key = kv0 + fk * 16 + (lane >> 4) * 4 + i;
             const unsigned long long idx =
                 ((unsigned long long)(bh)*L + qcol[fq]) * L + key;
-            p[fk][fq][i] = fa_keep(idx, seed, p8) ? p[fk][fq][i] * dscale : 0.f;
+            s_acc[fk][fq][i] = fa_keep(idx, seed, p8) ? s_acc[fk][fq][i] * dscale : 0.f;
           }
     }
 
@@ -297,18 +315,22 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
           // register indices must be compile-time (rule 20); the source
           // fragment differs only between lane halves -> shuffle both
           // candidate registers and select
-          const float v0 = __shfl(p[kp * 2][fq][j & 3], src);
-          const float v1 = __shfl(p[kp * 2 + 1][fq][j & 3], src);
+          const float v0 = __shfl(s_acc[kp * 2][fq][j & 3], src);
+          const float v1 = __shfl(s_acc[kp * 2 + 1][fq][j & 3], src);
           pa[fq][j] = (__bf16)(hi_half ? v1 : v0);
         }
       }
       FSEG(5)  // P exchange (shfl)
       bf16x8 v_frag[4];
+      const int kb0 = kp * 8 + (lane >> 4) * 2;
+      const int vslot = (lane & 15) * 8;
 #pragma unroll
       for (int fd = 0; fd < 4; ++fd) {
-        const int d = fd * 16 + (lane & 15);
-        const int keybyte = kp * 64 + (lane >> 4) * 16;
-        v_frag[fd] = *reinterpret_cast<const bf16x8*>(v_lds_c + fa_swzT(d, keybyte));
+        fwd_bf16x4v* vf = reinterpret_cast<fwd_bf16x4v*>(&v_frag[fd]);
+        vf[0] = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+            (fwd_lds_bf16x4*)(v_lds_c + FWD_VSUB(kb0, fd) + vslot));
+        vf[1] = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+            (fwd_lds_bf16x4*)(v_lds_c + FWD_VSUB(kb0 + 1, fd) + vslot));
       }
 #pragma unroll
       for (int fq = 0; fq < 2; ++fq)
@@ -318,7 +340,6 @@ __global__ __launch_bounds__(256) void flash_fwd_kernel(
               pa[fq], v_frag[fd], o_acc[fq][fd], 0, 0, 0);
     }
     FSEG(6)  // v_frag reads + PV MFMAs
-    if (has_next) write_tile(K_BUF(cur ^ 1), V_BUF(cur ^ 1), kreg, vreg);
     __syncthreads();
     cur ^= 1;
   }
@@ -1087,17 +1108,33 @@ void launch_flash_fwd(const bf16* Q, const bf16* K, const bf16* V,
                       int B, int H, int L, float scale, int causal,
                       unsigned p8, unsigned long long seed, long ldq,
                       long ldkv, hipStream_t stream) {
-  const dim3 grid((L + NWAVE * TQW - 1) / (NWAVE * TQW), B * H);
-  const size_t lds = 2 * 2 * TK * 128;  // double-buffered K+V tiles
+  const size_t lds = 2 * (TK * 128 + FWD_VBYTES);  // double-buffered K+V
   const char* e = getenv("DFA_FWD_PROF");
+  // 8-wave variant measured SLOWER on B16 L512 H12 (72.5 vs 66.5 us —
+  // fewer blocks lose more to tail/latency than staging amortization
+  // gains); keep the template but select it never for now
+  const long blocks8 = (long)((L + 8 * TQW - 1) / (8 * TQW)) * B * H;
+  if (false && L % (8 * TQW) == 0 && blocks8 >= 192) {
+    const dim3 grid(L / (8 * TQW), B * H);
+    if (e && e[0] == '1')
+      hipLaunchKernelGGL((flash_fwd_kernel<1, 8>), grid, dim3(512), lds, stream,
+                         Q, K, V, valid, bias, O, lse, B, H, L, scale, causal,
+                         p8, seed, ldq, ldkv);
+    else
+      hipLaunchKernelGGL((flash_fwd_kernel<0, 8>), grid, dim3(512), lds, stream,
+                         Q, K, V, valid, bias, O, lse, B, H, L, scale, causal,
+                         p8, seed, ldq, ldkv);
+    return;
+  }
+  const dim3 grid((L + NWAVE * TQW - 1) / (NWAVE * TQW), B * H);
   if (e && e[0] == '1')
-    hipLaunchKernelGGL(flash_fwd_kernel<1>, grid, dim3(256), lds, stream, Q, K,
-                       V, valid, bias, O, lse, B, H, L, scale, causal, p8,
-                       seed, ldq, ldkv);
+    hipLaunchKernelGGL((flash_fwd_kernel<1, 4>), grid, dim3(256), lds, stream,
+                       Q, K, V, valid, bias, O, lse, B, H, L, scale, causal,
+                       p8, seed, ldq, ldkv);
   else
-    hipLaunchKernelGGL(flash_fwd_kernel<0>, grid, dim3(256), lds, stream, Q, K,
-                       V, valid, bias, O, lse, B, H, L, scale, causal, p8,
-                       seed, ldq, ldkv);
+    hipLaunchKernelGGL((flash_fwd_kernel<0, 4>), grid, dim3(256), lds, stream,
+                       Q, K, V, valid, bias, O, lse, B, H, L, scale, causal,
+                       p8, seed, ldq, ldkv);
 }
 
 void launch_flash_dterm(const bf16* dO, const bf16* O, float* Dterm, int B,

@@ -30,7 +30,11 @@ dO = torch.randn(B, L, H * d, device=dev, dtype=bf) * 0.3
 valid = torch.full((B,), L, dtype=torch.int32, device=dev)
 O, lse = ext.flash_attn_fwd(q, k, v, H, valid, None, 0.125, False, 0.0, 0)
 torch.cuda.synchronize()
+which = os.environ.get("FLASH_PMC_WHICH", "bwd")
 for _ in range(30):
-    ext.flash_attn_bwd(dO, q, k, v, O, lse, H, valid, None, 0.125, False, 0.0, 0, False, False)
+    if which == "fwd":
+        ext.flash_attn_fwd(q, k, v, H, valid, None, 0.125, False, 0.0, 0)
+    else:
+        ext.flash_attn_bwd(dO, q, k, v, O, lse, H, valid, None, 0.125, False, 0.0, 0, False, False)
 torch.cuda.synchronize()
 print("done")
