@@ -132,6 +132,43 @@ class T5Attention(nn.Module):
         buckets, onehot = self._bias_cache[key]
         return _RelBias.apply(self.relative_attention_bias.weight, buckets, onehot)
 
+    def decode_step(self, x, kv=None, cache=None, position_bias=None, valid=None):
+        """Incremental decoding (eval-only KV cache): x is the NEW token
+        slice (B, 1, D). Self-attention (kv=None) appends this step's K/V
+        to cache['k']/['v'] and attends over all cached positions;
+        cross-attention computes K/V once into the cache. Replaces the
+        reference-style full re-run per generated token (quadratic in the
+        generated length; HF generate's past_key_values capability).
+        position_bias: (1, H, Lq, t) slice for the new query rows (shared
+        from the stack's first block, as in training)."""
+        B, Lq, _ = x.shape
+        H, d = self.cfg.num_heads, self.cfg.d_kv
+        q = fused_linear(x, self.q.weight).view(B, Lq, H, d).transpose(1, 2)
+        if kv is None:  # self-attention: extend the cache
+            k_new = fused_linear(x, self.k.weight).view(B, Lq, H, d).transpose(1, 2)
+            v_new = fused_linear(x, self.v.weight).view(B, Lq, H, d).transpose(1, 2)
+            if cache.get("k") is None:
+                cache["k"], cache["v"] = k_new, v_new
+            else:
+                cache["k"] = torch.cat([cache["k"], k_new], dim=2)
+                cache["v"] = torch.cat([cache["v"], v_new], dim=2)
+        elif cache.get("k") is None:  # cross-attention: compute once
+            Lk = kv.shape[1]
+            cache["k"] = fused_linear(kv, self.k.weight).view(B, Lk, H, d).transpose(1, 2)
+            cache["v"] = fused_linear(kv, self.v.weight).view(B, Lk, H, d).transpose(1, 2)
+        k, v = cache["k"], cache["v"]
+        scores = torch.matmul(q, k.transpose(-1, -2)).float()  # T5: no scale
+        if position_bias is not None:
+            scores = scores + position_bias
+        if valid is not None:  # encoder padding mask (cross-attention)
+            t = k.shape[2]
+            key_mask = torch.arange(t, device=x.device).view(1, 1, 1, t) >= \
+                valid.view(-1, 1, 1, 1)
+            scores = scores.masked_fill(key_mask, float("-inf"))
+        probs = torch.softmax(scores, dim=-1).to(q.dtype)
+        ctx = torch.matmul(probs, v).transpose(1, 2).reshape(B, Lq, H * d)
+        return fused_linear(ctx, self.o.weight)
+
     def forward(self, x, valid, kv=None, position_bias=None, dropout_p=0.0):
         B, Lq, _ = x.shape
         src = kv if kv is not None else x
@@ -274,6 +311,47 @@ class T5Stack(nn.Module):
             x = blk(x, valid, position_bias, enc=enc, enc_valid=enc_valid, dropout_p=p)
         return self.dropout(self.final_layer_norm(x))
 
+    @torch.no_grad()
+    def decode_step(self, ids_step, enc, enc_valid, state: "_DecodeState"):
+        """One incremental decoder step over the NEW token column
+        (B, 1) using per-block KV caches (eval-only; dropout off)."""
+        from ..ops.transformer import embedding_lookup
+
+        assert self.is_decoder
+        x = embedding_lookup(ids_step, self.embed_tokens.weight,
+                             self.embed_tokens.padding_idx)
+        if x.is_cuda:
+            x = x.to(torch.bfloat16)
+        t_new = state.t + 1
+        attn0 = self.block[0].layer[0].SelfAttention
+        bias = attn0.compute_bias(t_new, t_new, ids_step.device)[:, :, -1:, :]
+        for blk, cache in zip(self.block, state.caches):
+            h = blk.layer[0].layer_norm(x)
+            x = x + blk.layer[0].SelfAttention.decode_step(
+                h, cache=cache["self"], position_bias=bias)
+            h = blk.layer[1].layer_norm(x)
+            x = x + blk.layer[1].EncDecAttention.decode_step(
+                h, kv=enc, cache=cache["cross"], valid=enc_valid)
+            x = blk.layer[2](x)
+        state.t = t_new
+        return self.final_layer_norm(x)
+
+
+class _DecodeState:
+    """Per-generation KV caches: one {self, cross} pair per decoder block."""
+
+    def __init__(self, n_blocks: int):
+        self.caches = [{"self": {}, "cross": {}} for _ in range(n_blocks)]
+        self.t = 0  # decoded length so far
+
+    def reorder(self, idx: torch.Tensor):
+        """Beam search: reindex every cached tensor along batch*beam."""
+        for c in self.caches:
+            for part in ("self", "cross"):
+                for key in ("k", "v"):
+                    if c[part].get(key) is not None:
+                        c[part][key] = c[part][key].index_select(0, idx)
+
 
 class T5ForConditionalGeneration(nn.Module):
     def __init__(self, cfg: T5Config):
@@ -337,8 +415,10 @@ class T5ForConditionalGeneration(nn.Module):
     def generate(self, input_ids, attention_mask=None, max_length: int = 64,
                  num_beams: int = 1):
         """Greedy / beam-search generation (reference CodeT5 Beam,
-        models.py:298-408 capability). The decoder is re-run per step
-        (no KV cache yet); beams are batched through the decoder."""
+        models.py:298-408 capability) with per-layer KV caches
+        (T5Stack.decode_step): each step runs ONE decoder column instead of
+        re-running the whole prefix — linear, not quadratic, in the
+        generated length. Beam search reorders the caches by beam index."""
         if attention_mask is None:
             attention_mask = input_ids.ne(self.config.pad_token_id)
         enc_valid = attention_mask.sum(1).to(torch.int32)
@@ -347,24 +427,23 @@ class T5ForConditionalGeneration(nn.Module):
         device = input_ids.device
         eos, pad, start = (self.config.eos_token_id, self.config.pad_token_id,
                            self.config.decoder_start_token_id)
+        scale = self.config.d_model ** -0.5 if self.config.tie_word_embeddings else 1.0
 
-        def logits_for(dec_in, enc_rep, valid_rep):
-            dec_valid = torch.full((dec_in.shape[0],), dec_in.shape[1],
-                                   dtype=torch.int32, device=device)
-            dec = self.decoder(dec_in, dec_valid, enc=enc_rep, enc_valid=valid_rep)
-            h = dec[:, -1]
-            if self.config.tie_word_embeddings:
-                h = h * (self.config.d_model ** -0.5)
-            return self.lm_head(h).float()
+        def step_logits(cur, enc_rep, valid_rep, state):
+            h = self.decoder.decode_step(cur, enc_rep, valid_rep, state)[:, -1]
+            return self.lm_head(h.float() * scale if scale != 1.0 else h.float())
 
         if num_beams <= 1:
+            state = _DecodeState(len(self.decoder.block))
             seq = torch.full((B, 1), start, dtype=torch.long, device=device)
+            cur = seq
             done = torch.zeros(B, dtype=torch.bool, device=device)
             for _ in range(max_length - 1):
-                nxt = logits_for(seq, enc, enc_valid).argmax(-1)
+                nxt = step_logits(cur, enc, enc_valid, state).argmax(-1)
                 nxt = torch.where(done, torch.full_like(nxt, pad), nxt)
                 seq = torch.cat([seq, nxt.unsqueeze(1)], dim=1)
                 done |= nxt == eos
+                cur = nxt.unsqueeze(1)
                 if bool(done.all()):
                     break
             return seq
@@ -372,12 +451,14 @@ class T5ForConditionalGeneration(nn.Module):
         K = num_beams
         enc_rep = enc.repeat_interleave(K, dim=0)
         valid_rep = enc_valid.repeat_interleave(K)
+        state = _DecodeState(len(self.decoder.block))
         seq = torch.full((B * K, 1), start, dtype=torch.long, device=device)
+        cur = seq
         beam_scores = torch.full((B, K), -1e9, device=device)
         beam_scores[:, 0] = 0.0
         done = torch.zeros(B * K, dtype=torch.bool, device=device)
         for _ in range(max_length - 1):
-            logp = torch.log_softmax(logits_for(seq, enc_rep, valid_rep), -1)
+            logp = torch.log_softmax(step_logits(cur, enc_rep, valid_rep, state), -1)
             logp = logp.masked_fill(done.unsqueeze(1), 0.0)
             V = logp.shape[-1]
             total = (beam_scores.view(-1, 1) + logp).view(B, K * V)
@@ -385,10 +466,11 @@ class T5ForConditionalGeneration(nn.Module):
             beam_idx = top_idx // V
             tok_idx = top_idx % V
             flat_src = (torch.arange(B, device=device).unsqueeze(1) * K + beam_idx).view(-1)
-            seq = torch.cat([seq[flat_src],
-                             tok_idx.view(-1, 1)], dim=1)
+            seq = torch.cat([seq[flat_src], tok_idx.view(-1, 1)], dim=1)
+            state.reorder(flat_src)
             done = done[flat_src] | (tok_idx.view(-1) == eos)
             beam_scores = top_scores
+            cur = tok_idx.view(-1, 1)
             if bool(done.all()):
                 break
         return seq.view(B, K, -1)[:, 0]

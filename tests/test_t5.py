@@ -127,3 +127,54 @@ def test_eos_count_check():
     ids[0, 5] = 2  # extra EOS in row 0 only
     with pytest.raises(ValueError):
         model(ids, labels=torch.tensor([0, 1, 0]))
+
+
+class TestKVCachedGeneration:
+    """generate() now decodes incrementally with per-layer KV caches; it
+    must produce exactly the tokens of a full-prefix re-run per step."""
+
+    def _model(self):
+        torch.manual_seed(3)
+        cfg = T5Config(num_layers=2, num_decoder_layers=2, d_model=64,
+                       d_ff=128, num_heads=2, vocab_size=200)
+        m = T5ForConditionalGeneration(cfg).eval()
+        return m, cfg
+
+    def _reference_greedy(self, m, cfg, ids, max_length):
+        # the old quadratic scheme: full decoder re-run per step
+        attention_mask = ids.ne(cfg.pad_token_id)
+        enc_valid = attention_mask.sum(1).to(torch.int32)
+        enc = m.encoder(ids, enc_valid)
+        B = ids.shape[0]
+        seq = torch.full((B, 1), cfg.decoder_start_token_id, dtype=torch.long)
+        done = torch.zeros(B, dtype=torch.bool)
+        for _ in range(max_length - 1):
+            dec_valid = torch.full((B,), seq.shape[1], dtype=torch.int32)
+            dec = m.decoder(seq, dec_valid, enc=enc, enc_valid=enc_valid)
+            h = dec[:, -1] * (cfg.d_model ** -0.5)
+            nxt = m.lm_head(h.float()).argmax(-1)
+            nxt = torch.where(done, torch.full_like(nxt, cfg.pad_token_id), nxt)
+            seq = torch.cat([seq, nxt.unsqueeze(1)], dim=1)
+            done |= nxt == cfg.eos_token_id
+            if bool(done.all()):
+                break
+        return seq
+
+    @torch.no_grad()
+    def test_greedy_matches_full_rerun(self):
+        m, cfg = self._model()
+        ids = torch.randint(3, cfg.vocab_size, (3, 24))
+        ids[:, -1] = cfg.eos_token_id
+        ids[0, 12:] = cfg.pad_token_id  # ragged source (cross-attn mask)
+        got = m.generate(ids, max_length=12)
+        want = self._reference_greedy(m, cfg, ids, 12)
+        L = min(got.shape[1], want.shape[1])
+        assert torch.equal(got[:, :L], want[:, :L]), (got, want)
+
+    @torch.no_grad()
+    def test_beam_runs_and_respects_eos(self):
+        m, cfg = self._model()
+        ids = torch.randint(3, cfg.vocab_size, (2, 16))
+        ids[:, -1] = cfg.eos_token_id
+        out = m.generate(ids, max_length=10, num_beams=3)
+        assert out.shape[0] == 2 and out.shape[1] <= 10
