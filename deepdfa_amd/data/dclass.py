@@ -113,6 +113,22 @@ def ds_partition(
         if partition == "all":
             return df
         return df[df.partition == partition]
+    if split.startswith("linevul"):
+        # the reference's SAVED split files (datasets.py:449-452
+        # get_linevul_splits: linevul_splits.csv with id-indexed "split"
+        # column) — the split the headline Big-Vul F1 numbers are quoted
+        # on. "linevul:<path>" points at the csv; ids missing from the
+        # file are dropped (the reference's splits cover every kept id).
+        path = split.split(":", 1)[1] if ":" in split else "linevul_splits.csv"
+        sp = pd.read_csv(path, index_col=0)
+        col = "split" if "split" in sp.columns else sp.columns[0]
+        mapping = sp[col].to_dict()
+        part = df["id"].map(mapping)
+        df = df.assign(partition=part.replace({"valid": "val"}))
+        df = df[df.partition.notna()]
+        if partition == "all":
+            return df
+        return df[df.partition == partition]
     if split.startswith("random"):
         s = int(split.split("_")[1]) if "_" in split else seed
         rng = np.random.RandomState(s)

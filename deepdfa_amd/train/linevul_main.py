@@ -207,9 +207,9 @@ def build_args(argv=None):
     p.add_argument("--n_synthetic", type=int, default=2000)
     p.add_argument("--num_layers", type=int, default=12, help="encoder depth (tests use small)")
     p.add_argument("--split", default="fixed",
-                   choices=["fixed", "random", "cross_project"],
-                   help="partitioning scheme (cross_project = the reference's "
-                        "cross-project generalization experiments)")
+                   help="partitioning scheme: fixed | random | cross_project "
+                        "| linevul:<saved_splits.csv> (the reference's saved "
+                        "LineVul split files, datasets.py:449)")
     p.add_argument("--train_data_file", default=None)
     p.add_argument("--eval_data_file", default=None)
     p.add_argument("--test_data_file", default=None)

@@ -202,3 +202,27 @@ def test_text_dataset_jsonl_schema(tmp_path):
     assert len(ds) == 4
     _, label, index = ds[3]
     assert label.item() == 1 and index.item() == 3
+
+
+def test_linevul_saved_split_files(tmp_path):
+    """split="linevul:<csv>" loads the reference's saved split file
+    (datasets.py:449-452) — the split the headline F1 is quoted on."""
+    import pandas as pd
+
+    from deepdfa_amd.data.dclass import ds_partition, synthetic_bigvul_df
+
+    df = synthetic_bigvul_df(50)
+    splits = pd.DataFrame(
+        {"split": ["train"] * 30 + ["valid"] * 10 + ["test"] * 8},
+        index=pd.Index(range(48), name="id"),
+    )
+    path = str(tmp_path / "linevul_splits.csv")
+    splits.to_csv(path)
+    tr = ds_partition(df, "train", split=f"linevul:{path}")
+    va = ds_partition(df, "val", split=f"linevul:{path}")
+    te = ds_partition(df, "test", split=f"linevul:{path}")
+    assert len(tr) == 30 and len(va) == 10 and len(te) == 8
+    assert set(tr.id) == set(range(30))
+    # ids 48, 49 are absent from the split file -> dropped entirely
+    al = ds_partition(df, "all", split=f"linevul:{path}")
+    assert 48 not in set(al.id) and 49 not in set(al.id)
