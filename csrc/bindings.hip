@@ -889,6 +889,8 @@ std::vector<at::Tensor> gate_pool_fwd(at::Tensor x1, at::Tensor x2, at::Tensor w
   const int N = x1.size(0), D1 = x1.size(1), D = D1 + (int)x2.size(1);
   const int B = node_offsets.numel() - 1;
   TORCH_CHECK(x2.size(0) == N && wg.numel() == D);
+  TORCH_CHECK(D == 256 && D1 == 128, "fused gate_pool is built for the "
+              "128+128 concat geometry (vectorized lane mapping)");
   auto out = at::empty({B, D}, x1.options());
   auto alpha = at::empty({N}, x1.options().dtype(at::kFloat));
   launch_gate_pool_fwd(ptr<bf16_t>(x1), ptr<bf16_t>(x2), wg.data_ptr<float>(),

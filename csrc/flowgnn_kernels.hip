@@ -709,18 +709,27 @@ __global__ void gate_pool_fwd_kernel(
   const int wid = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
   const int nw = blockDim.x / WAVE;
+  // vectorized gate dot: lane owns elements [4*lane, 4*lane+4) so lanes
+  // 0..31 read x1 and 32..63 read x2 as one 8-B load each (the scalar
+  // 2-B form was the kernel's dominant cost)
+  const int dv = 4 * lane;
+  const __hip_bfloat16* xs_half = (dv < D1) ? x1 : x2;
+  const int Ds_half = (dv < D1) ? D1 : D2;
+  const int dd_half = (dv < D1) ? dv : dv - D1;
   for (int v = wid * 2; v < n; v += nw * 2) {
     float s0 = 0.f, s1 = 0.f;
     const bool has1 = (v + 1) < n;
-    for (int d = lane; d < D; d += WAVE) {
-      const float w = ws[d];
-      const float a0 = (d < D1) ? to_f(x1[(long)(lo + v) * D1 + d])
-                                : to_f(x2[(long)(lo + v) * D2 + d - D1]);
-      s0 += w * a0;
+    {
+      __hip_bfloat16 a4[4];
+      *reinterpret_cast<uint2*>(a4) = *reinterpret_cast<const uint2*>(
+          xs_half + (long)(lo + v) * Ds_half + dd_half);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) s0 += ws[dv + j] * to_f(a4[j]);
       if (has1) {
-        const float a1 = (d < D1) ? to_f(x1[(long)(lo + v + 1) * D1 + d])
-                                  : to_f(x2[(long)(lo + v + 1) * D2 + d - D1]);
-        s1 += w * a1;
+        *reinterpret_cast<uint2*>(a4) = *reinterpret_cast<const uint2*>(
+            xs_half + (long)(lo + v + 1) * Ds_half + dd_half);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) s1 += ws[dv + j] * to_f(a4[j]);
       }
     }
     for (int off = WAVE / 2; off > 0; off >>= 1) {
@@ -797,20 +806,29 @@ __global__ void gate_pool_bwd_kernel(
   const int wid = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
   const int nw = blockDim.x / WAVE;
+  // lane owns elements [4*lane, 4*lane+4): one 8-B load per node per pass
+  // (the scalar strided form was the kernel's dominant cost at 97 us),
+  // lanes 0..31 on the x1 half, 32..63 on x2
+  const int dv = 4 * lane;
+  const __hip_bfloat16* xs_half = (dv < D1) ? x1 : x2;
+  __hip_bfloat16* gxs_half = (dv < D1) ? gx1 : gx2;
+  const int Ds_half = (dv < D1) ? D1 : D2;
+  const int dd_half = (dv < D1) ? dv : dv - D1;
   // pass 1: s_v = <go, x_v>
   float local_dot = 0.f;
   for (int v = wid * 2; v < n; v += nw * 2) {
     float sv0 = 0.f, sv1 = 0.f;
     const bool has1 = (v + 1) < n;
-    for (int d = lane; d < D; d += WAVE) {
-      const float a0 = (d < D1) ? to_f(x1[(long)(lo + v) * D1 + d])
-                                : to_f(x2[(long)(lo + v) * D2 + d - D1]);
-      sv0 += go[d] * a0;
-      if (has1) {
-        const float a1 = (d < D1) ? to_f(x1[(long)(lo + v + 1) * D1 + d])
-                                  : to_f(x2[(long)(lo + v + 1) * D2 + d - D1]);
-        sv1 += go[d] * a1;
-      }
+    __hip_bfloat16 a4[4];
+    *reinterpret_cast<uint2*>(a4) = *reinterpret_cast<const uint2*>(
+        xs_half + (long)(lo + v) * Ds_half + dd_half);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) sv0 += go[dv + j] * to_f(a4[j]);
+    if (has1) {
+      *reinterpret_cast<uint2*>(a4) = *reinterpret_cast<const uint2*>(
+          xs_half + (long)(lo + v + 1) * Ds_half + dd_half);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) sv1 += go[dv + j] * to_f(a4[j]);
     }
     for (int off = WAVE / 2; off > 0; off >>= 1) {
       sv0 += __shfl_down(sv0, off);
@@ -827,23 +845,29 @@ __global__ void gate_pool_bwd_kernel(
   }
   __syncthreads();
   const float dot = block_reduce(lane == 0 ? local_dot : 0.f, scratch, 1);
-  // pass 2: dx (pool path + gate path), dgate, local dwg
+  // pass 2: dx (pool + gate paths) as one 8-B store per node, dwg
+  // accumulated in registers (the per-(node,d) LDS atomics were serial)
   float local_dbg = 0.f;
+  float ldw[4] = {};
   for (int v = wid; v < n; v += nw) {
     const float a = alpha[lo + v];
     const float dgate = a * (s_ws[lo + v] - dot);
     if (lane == 0) local_dbg += dgate;
-    for (int d = lane; d < D; d += WAVE) {
-      const float xv = (d < D1) ? to_f(x1[(long)(lo + v) * D1 + d])
-                                : to_f(x2[(long)(lo + v) * D2 + d - D1]);
-      const float gxv = a * go[d] + dgate * ws[d];
-      if (d < D1)
-        gx1[(long)(lo + v) * D1 + d] = from_f<__hip_bfloat16>(gxv);
-      else
-        gx2[(long)(lo + v) * D2 + d - D1] = from_f<__hip_bfloat16>(gxv);
-      atomicAdd(&dw[d], dgate * xv);
+    __hip_bfloat16 a4[4], o4[4];
+    *reinterpret_cast<uint2*>(a4) = *reinterpret_cast<const uint2*>(
+        xs_half + (long)(lo + v) * Ds_half + dd_half);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const float xv = to_f(a4[j]);
+      o4[j] = from_f<__hip_bfloat16>(a * go[dv + j] + dgate * ws[dv + j]);
+      ldw[j] += dgate * xv;
     }
+    *reinterpret_cast<uint2*>(gxs_half + (long)(lo + v) * Ds_half + dd_half) =
+        *reinterpret_cast<const uint2*>(o4);
   }
+#pragma unroll
+  for (int j = 0; j < 4; ++j)
+    if (ldw[j] != 0.f) atomicAdd(&dw[dv + j], ldw[j]);
   __syncthreads();
   for (int d = threadIdx.x; d < D; d += blockDim.x)
     if (dw[d] != 0.f) atomicAdd(&dwg[d], dw[d]);
@@ -985,16 +1009,31 @@ __global__ void mlp3_wgrad_kernel(
        o += (long)gridDim.x * blockDim.x) {
     if (o < DD) {  // dW1[i][j] = sum_r dh1[r][i] * x[r][j]
       const int i = (int)(o / D), jj = (int)(o % D);
-      float s = 0.f;
-      for (int r = 0; r < B; ++r)
-        s += dh1[(long)r * D + i] * to_f(x[(long)r * D + jj]);
-      dW1[o] = s;
+      // 4 independent chains: the single-accumulator form serializes 257
+      // FMA+L2-load latencies (72.8 us measured; ~8 expected)
+      float s0 = 0.f, s1 = 0.f, s2 = 0.f, s3 = 0.f;
+      int r = 0;
+      for (; r + 4 <= B; r += 4) {
+        s0 += dh1[(long)r * D + i] * to_f(x[(long)r * D + jj]);
+        s1 += dh1[(long)(r + 1) * D + i] * to_f(x[(long)(r + 1) * D + jj]);
+        s2 += dh1[(long)(r + 2) * D + i] * to_f(x[(long)(r + 2) * D + jj]);
+        s3 += dh1[(long)(r + 3) * D + i] * to_f(x[(long)(r + 3) * D + jj]);
+      }
+      for (; r < B; ++r) s0 += dh1[(long)r * D + i] * to_f(x[(long)r * D + jj]);
+      dW1[o] = (s0 + s1) + (s2 + s3);
     } else if (o < 2 * DD) {
       const long oo = o - DD;
       const int i = (int)(oo / D), jj = (int)(oo % D);
-      float s = 0.f;
-      for (int r = 0; r < B; ++r) s += dh2[(long)r * D + i] * h1[(long)r * D + jj];
-      dW2[oo] = s;
+      float s0 = 0.f, s1 = 0.f, s2 = 0.f, s3 = 0.f;
+      int r = 0;
+      for (; r + 4 <= B; r += 4) {
+        s0 += dh2[(long)r * D + i] * h1[(long)r * D + jj];
+        s1 += dh2[(long)(r + 1) * D + i] * h1[(long)(r + 1) * D + jj];
+        s2 += dh2[(long)(r + 2) * D + i] * h1[(long)(r + 2) * D + jj];
+        s3 += dh2[(long)(r + 3) * D + i] * h1[(long)(r + 3) * D + jj];
+      }
+      for (; r < B; ++r) s0 += dh2[(long)r * D + i] * h1[(long)r * D + jj];
+      dW2[oo] = (s0 + s1) + (s2 + s3);
     } else if (o < 2 * DD + D) {
       const int jj = (int)(o - 2 * DD);
       float s = 0.f;
