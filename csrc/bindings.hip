@@ -9,6 +9,8 @@
 // the launchers are defined (at global scope) in flowgnn_kernels.hip
 template <typename T>
 void launch_embed4_fwd(const T*, const long*, T*, int, int, hipStream_t);
+template <typename T, typename TO>
+void launch_embed4_fwd2(const T*, const long*, TO*, int, int, hipStream_t);
 template <typename T>
 void launch_embed4_bwd(const T*, const long*, float*, long, int, hipStream_t);
 template <typename T>
@@ -136,7 +138,7 @@ T* mptr(at::Tensor& t) {
   return reinterpret_cast<T*>(t.data_ptr());
 }
 
-at::Tensor embed4_fwd(at::Tensor tables, at::Tensor idx) {
+at::Tensor embed4_fwd(at::Tensor tables, at::Tensor idx, bool out_bf16 = false) {
   CHECK_GPU(tables);
   CHECK_GPU(idx);
   TORCH_CHECK(tables.dim() == 3 && tables.size(0) == 4 && tables.size(2) == 32,
@@ -144,6 +146,14 @@ at::Tensor embed4_fwd(at::Tensor tables, at::Tensor idx) {
   TORCH_CHECK(idx.dim() == 2 && idx.size(1) == 4 && idx.scalar_type() == at::kLong);
   const int N = idx.size(0);
   const int V = tables.size(1);
+  if (out_bf16 && tables.scalar_type() == at::kFloat) {
+    // autocast path: gather fp32 master rows, emit bf16 directly
+    auto out = at::empty({N, 128}, tables.options().dtype(at::kBFloat16));
+    launch_embed4_fwd2<float, __hip_bfloat16>(
+        tables.data_ptr<float>(), idx.data_ptr<long>(),
+        reinterpret_cast<__hip_bfloat16*>(out.data_ptr()), N, V, cur_stream());
+    return out;
+  }
   auto out = at::empty({N, 128}, tables.options());
   dispatch_float_bf16(tables, "embed4_fwd", [&](auto tag) {
     using T = decltype(tag);
@@ -981,7 +991,7 @@ void adamw_fused(at::Tensor p, at::Tensor g, at::Tensor m, at::Tensor v, double 
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "deepdfa_amd MI355X (gfx950) kernels";
-  m.def("embed4_fwd", &embed4_fwd);
+  m.def("embed4_fwd", &embed4_fwd, pybind11::arg("tables"), pybind11::arg("idx"), pybind11::arg("out_bf16") = false);
   m.def("embed4_bwd", &embed4_bwd);
   m.def("spmm_sum", &spmm_sum);
   m.def("gru_gates_fwd", &gru_gates_fwd);

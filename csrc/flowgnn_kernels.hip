@@ -44,10 +44,13 @@ __device__ __forceinline__ float sigmoidf_(float x) { return 1.0f / (1.0f + __ex
 // One wave per node; lane l writes out[n][2l..2l+1]; feature f = l/16.
 // ---------------------------------------------------------------------------
 
-template <typename T>
+template <typename T, typename TO>
 __global__ void embed4_fwd_kernel(const T* __restrict__ tables,
                                   const long* __restrict__ idx,
-                                  T* __restrict__ out, int N, int V) {
+                                  TO* __restrict__ out, int N, int V) {
+  // TO may differ from T: under bf16 autocast the gather emits bf16
+  // directly from the fp32 master tables (removes the .to(bf16) cast
+  // kernels + their backward casts on both consumers of the embedding)
   const int wid = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
   if (wid >= N) return;
@@ -56,9 +59,9 @@ __global__ void embed4_fwd_kernel(const T* __restrict__ tables,
   const int off = c & 31;          // offset within the 32-wide embedding
   const long row = idx[(long)wid * 4 + f];
   const T* src = tables + ((long)f * V + row) * 32 + off;
-  T* dst = out + (long)wid * 128 + c;
-  dst[0] = src[0];
-  dst[1] = src[1];
+  TO* dst = out + (long)wid * 128 + c;
+  dst[0] = from_f<TO>(to_f(src[0]));
+  dst[1] = from_f<TO>(to_f(src[1]));
 }
 
 template <typename T>
@@ -436,15 +439,23 @@ __global__ void segment_max_kernel(const float* __restrict__ values,
 // Launchers (called from bindings.cpp)
 // ---------------------------------------------------------------------------
 
-template <typename T>
-void launch_embed4_fwd(const T* tables, const long* idx, T* out, int N, int V,
-                       hipStream_t stream) {
+template <typename T, typename TO>
+void launch_embed4_fwd2(const T* tables, const long* idx, TO* out, int N, int V,
+                        hipStream_t stream) {
   const int waves_per_block = 4;
   const int block = WAVE * waves_per_block;
   const int grid = (N + waves_per_block - 1) / waves_per_block;
   if (grid > 0)
-    hipLaunchKernelGGL(embed4_fwd_kernel<T>, dim3(grid), dim3(block), 0, stream,
-                       tables, idx, out, N, V);
+    hipLaunchKernelGGL((embed4_fwd_kernel<T, TO>), dim3(grid), dim3(block), 0,
+                       stream, tables, idx, out, N, V);
+}
+template void launch_embed4_fwd2<float, __hip_bfloat16>(
+    const float*, const long*, __hip_bfloat16*, int, int, hipStream_t);
+
+template <typename T>
+void launch_embed4_fwd(const T* tables, const long* idx, T* out, int N, int V,
+                       hipStream_t stream) {
+  launch_embed4_fwd2<T, T>(tables, idx, out, N, V, stream);
 }
 
 template <typename T>

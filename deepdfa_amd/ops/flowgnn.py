@@ -27,7 +27,11 @@ class _Embed4(torch.autograd.Function):
         ctx.tables_shape = tables.shape
         ext = ext_for(tables)
         if ext is not None:
-            return ext.embed4_fwd(tables, idx)
+            # under autocast emit bf16 straight out of the gather: both
+            # consumers (GGNN input, fused head) want bf16, so this removes
+            # two cast kernels + their backward casts per step
+            out_bf16 = tables.dtype == torch.float32 and torch.is_autocast_enabled()
+            return ext.embed4_fwd(tables, idx, out_bf16)
         return ref.embed4_fwd(list(tables), idx)
 
     @staticmethod
