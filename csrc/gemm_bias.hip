@@ -60,34 +60,54 @@ __global__ __launch_bounds__(256) void gemm_bias_kernel(
   constexpr int MFRAG = BMT / 32;     // fragments per row-wave
   f32x4 acc[MFRAG][4] = {};
 
+  // T14 double-buffered staging: next K-tile's global loads issue before
+  // this tile's MFMAs (these N x 128-ish GEMMs are HBM-latency-bound at
+  // ~1.4 blocks/CU; single-buffered staging serialized load latency with
+  // the MFMA phase)
+  const int srow = tid >> 3;         // 0..31
+  const int soff = (tid & 7) * 16;   // byte offset in row
+  constexpr int NAR = BMT / 32;      // A rows per thread
+  auto load_regs = [&](int kk, uint4v ar[NAR], uint4v br[4]) {
+#pragma unroll
+    for (int p = 0; p < NAR; ++p) {
+      const int rr = srow + p * 32;
+      const int gr = r0 + rr;
+      ar[p] = {};
+      if (gr < N) {
+        const int gk = kk + soff / 2;  // element index in K
+        const bf16* src = (gk < K1) ? (A1 + (long)gr * K1 + gk)
+                                    : (A2 + (long)gr * (K - K1) + (gk - K1));
+        ar[p] = *reinterpret_cast<const uint4v*>(src);
+      }
+    }
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      const int rr = srow + p * 32;
+      br[p] = *reinterpret_cast<const uint4v*>(W + (long)(c0 + rr) * K + kk +
+                                               soff / 2);
+    }
+  };
+  auto write_regs = [&](char* ab, char* bb, const uint4v ar[NAR],
+                        const uint4v br[4]) {
+#pragma unroll
+    for (int p = 0; p < NAR; ++p)
+      *reinterpret_cast<uint4v*>(ab + swz(srow + p * 32, soff)) = ar[p];
+#pragma unroll
+    for (int p = 0; p < 4; ++p)
+      *reinterpret_cast<uint4v*>(bb + swz(srow + p * 32, soff)) = br[p];
+  };
+#define GB_ABUF(i) (a_lds + (i) * (BMT + BN) * ROWB)
+#define GB_BBUF(i) (b_lds + (i) * (BMT + BN) * ROWB)
+  uint4v areg[NAR], breg[4];
+  load_regs(0, areg, breg);
+  write_regs(GB_ABUF(0), GB_BBUF(0), areg, breg);
+  __syncthreads();
+  int cur = 0;
   for (int kk = 0; kk < K; kk += BK) {
-    // stage A tile: BMT rows x 64 k (128 B/row); 8 threads/row, 32 rows/pass
-    {
-      const int row = tid >> 3;        // 0..31
-      const int off = (tid & 7) * 16;  // byte offset in row
-      for (int rr = row; rr < BMT; rr += 32) {
-        const int gr = r0 + rr;
-        uint4v v = {};
-        if (gr < N) {
-          const int gk = kk + off / 2;  // element index in K
-          const bf16* src = (gk < K1) ? (A1 + (long)gr * K1 + gk)
-                                      : (A2 + (long)gr * (K - K1) + (gk - K1));
-          v = *reinterpret_cast<const uint4v*>(src);
-        }
-        *reinterpret_cast<uint4v*>(a_lds + swz(rr, off)) = v;
-      }
-    }
-    // stage B tile: BN rows (output cols) x 64 k
-    {
-      const int row = tid >> 3;
-      const int off = (tid & 7) * 16;
-      for (int rr = row; rr < BN; rr += 32) {
-        const bf16* src = W + (long)(c0 + rr) * K + kk + off / 2;
-        *reinterpret_cast<uint4v*>(b_lds + swz(rr, off)) =
-            *reinterpret_cast<const uint4v*>(src);
-      }
-    }
-    __syncthreads();
+    const bool has_next = (kk + BK) < K;
+    if (has_next) load_regs(kk + BK, areg, breg);
+    char* a_lds_c = GB_ABUF(cur);
+    char* b_lds_c = GB_BBUF(cur);
 
 #pragma unroll
     for (int ks = 0; ks < BK / 32; ++ks) {
@@ -97,22 +117,28 @@ __global__ __launch_bounds__(256) void gemm_bias_kernel(
 #pragma unroll
       for (int m = 0; m < MFRAG; ++m) {
         const int row = wm * (BMT / 2) + m * 16 + (lane & 15);
-        a_frag[m] = *reinterpret_cast<const bf16x8*>(a_lds + swz(row, kbyte));
+        a_frag[m] = *reinterpret_cast<const bf16x8*>(a_lds_c + swz(row, kbyte));
       }
 #pragma unroll
       for (int n = 0; n < 4; ++n) {
         const int row = wn * 64 + n * 16 + (lane & 15);
-        b_frag[n] = *reinterpret_cast<const bf16x8*>(b_lds + swz(row, kbyte));
+        b_frag[n] = *reinterpret_cast<const bf16x8*>(b_lds_c + swz(row, kbyte));
       }
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int m = 0; m < MFRAG; ++m)
 #pragma unroll
         for (int n = 0; n < 4; ++n)
           acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               a_frag[m], b_frag[n], acc[m][n], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
     }
+    if (has_next) write_regs(GB_ABUF(cur ^ 1), GB_BBUF(cur ^ 1), areg, breg);
     __syncthreads();
+    cur ^= 1;
   }
+#undef GB_ABUF
+#undef GB_BBUF
 
   // epilogue: D mapping col = lane&15, row = (lane>>4)*4 + i
 #pragma unroll
@@ -143,13 +169,13 @@ void launch_gemm_bias2(const bf16* A1, const bf16* A2, const bf16* W,
   const long blocks64 = (long)((N + BM - 1) / BM) * (COL / BN);
   if (blocks64 < 384) {
     const dim3 grid((N + 31) / 32, COL / BN);
-    const size_t lds = (32 + BN) * ROWB;
+    const size_t lds = 2 * (32 + BN) * ROWB;
     hipLaunchKernelGGL(gemm_bias_kernel<32>, grid, dim3(256), lds, stream, A1,
                        A2, W, bias, addend, astride, addend2, astride2, out, N,
                        K, K1, COL);
   } else {
     const dim3 grid((N + BM - 1) / BM, COL / BN);
-    const size_t lds = (BM + BN) * ROWB;
+    const size_t lds = 2 * (BM + BN) * ROWB;
     hipLaunchKernelGGL(gemm_bias_kernel<BM>, grid, dim3(256), lds, stream, A1,
                        A2, W, bias, addend, astride, addend2, astride2, out, N,
                        K, K1, COL);
@@ -198,32 +224,51 @@ __global__ __launch_bounds__(256) void gemm_gru_kernel(
   constexpr int MFRAG = BMT / 32;
   f32x4 acc[MFRAG][4] = {};
 
+  // same T14 register double-buffering as gemm_bias_kernel above
+  const int srow = tid >> 3;
+  const int soff = (tid & 7) * 16;
+  constexpr int NAR = BMT / 32;
+  auto load_regs = [&](int kk, uint4v ar[NAR], uint4v br[4]) {
+#pragma unroll
+    for (int p = 0; p < NAR; ++p) {
+      const int rr = srow + p * 32;
+      const int gr = r0 + rr;
+      ar[p] = {};
+      if (gr < N) {
+        const int gk = kk + soff / 2;
+        const bf16* src = (gk < K1) ? (A1 + (long)gr * K1 + gk)
+                                    : (A2 + (long)gr * (K - K1) + (gk - K1));
+        ar[p] = *reinterpret_cast<const uint4v*>(src);
+      }
+    }
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      const int rr = srow + p * 32;
+      br[p] = *reinterpret_cast<const uint4v*>(Wperm + (long)(c0 + rr) * K +
+                                               kk + soff / 2);
+    }
+  };
+  auto write_regs = [&](char* ab, char* bb, const uint4v ar[NAR],
+                        const uint4v br[4]) {
+#pragma unroll
+    for (int p = 0; p < NAR; ++p)
+      *reinterpret_cast<uint4v*>(ab + swz(srow + p * 32, soff)) = ar[p];
+#pragma unroll
+    for (int p = 0; p < 4; ++p)
+      *reinterpret_cast<uint4v*>(bb + swz(srow + p * 32, soff)) = br[p];
+  };
+#define GB_ABUF(i) (a_lds + (i) * (BMT + BN) * ROWB)
+#define GB_BBUF(i) (b_lds + (i) * (BMT + BN) * ROWB)
+  uint4v areg[NAR], breg[4];
+  load_regs(0, areg, breg);
+  write_regs(GB_ABUF(0), GB_BBUF(0), areg, breg);
+  __syncthreads();
+  int cur = 0;
   for (int kk = 0; kk < K; kk += BK) {
-    {
-      const int row = tid >> 3;
-      const int off = (tid & 7) * 16;
-      for (int rr = row; rr < BMT; rr += 32) {
-        const int gr = r0 + rr;
-        uint4v v = {};
-        if (gr < N) {
-          const int gk = kk + off / 2;
-          const bf16* src = (gk < K1) ? (A1 + (long)gr * K1 + gk)
-                                      : (A2 + (long)gr * (K - K1) + (gk - K1));
-          v = *reinterpret_cast<const uint4v*>(src);
-        }
-        *reinterpret_cast<uint4v*>(a_lds + swz(rr, off)) = v;
-      }
-    }
-    {
-      const int row = tid >> 3;
-      const int off = (tid & 7) * 16;
-      for (int rr = row; rr < BN; rr += 32) {
-        const bf16* src = Wperm + (long)(c0 + rr) * K + kk + off / 2;
-        *reinterpret_cast<uint4v*>(b_lds + swz(rr, off)) =
-            *reinterpret_cast<const uint4v*>(src);
-      }
-    }
-    __syncthreads();
+    const bool has_next = (kk + BK) < K;
+    if (has_next) load_regs(kk + BK, areg, breg);
+    char* a_lds_c = GB_ABUF(cur);
+    char* b_lds_c = GB_BBUF(cur);
 #pragma unroll
     for (int ks = 0; ks < BK / 32; ++ks) {
       const int kbyte = ks * 64 + (lane >> 4) * 16;
@@ -231,22 +276,28 @@ __global__ __launch_bounds__(256) void gemm_gru_kernel(
 #pragma unroll
       for (int m = 0; m < MFRAG; ++m) {
         const int row = wm * (BMT / 2) + m * 16 + (lane & 15);
-        a_frag[m] = *reinterpret_cast<const bf16x8*>(a_lds + swz(row, kbyte));
+        a_frag[m] = *reinterpret_cast<const bf16x8*>(a_lds_c + swz(row, kbyte));
       }
 #pragma unroll
       for (int n = 0; n < 4; ++n) {
         const int row = wn * 64 + n * 16 + (lane & 15);
-        b_frag[n] = *reinterpret_cast<const bf16x8*>(b_lds + swz(row, kbyte));
+        b_frag[n] = *reinterpret_cast<const bf16x8*>(b_lds_c + swz(row, kbyte));
       }
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int m = 0; m < MFRAG; ++m)
 #pragma unroll
         for (int n = 0; n < 4; ++n)
           acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               a_frag[m], b_frag[n], acc[m][n], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
     }
+    if (has_next) write_regs(GB_ABUF(cur ^ 1), GB_BBUF(cur ^ 1), areg, breg);
     __syncthreads();
+    cur ^= 1;
   }
+#undef GB_ABUF
+#undef GB_BBUF
 
   // bounce fp32 gate pre-activations (+bias) through LDS, row stride 132
   float* tile = reinterpret_cast<float*>(smem);
@@ -294,12 +345,12 @@ void launch_gemm_gru(const bf16* A1, const bf16* A2, const bf16* Wperm,
   const long blocks64 = (long)((N + BM - 1) / BM) * (COL / BN);
   if (blocks64 < 384) {
     const dim3 grid((N + 31) / 32, COL / BN);
-    const size_t lds = max((size_t)(32 + BN) * ROWB, (size_t)32 * 132 * 4);
+    const size_t lds = max((size_t)2 * (32 + BN) * ROWB, (size_t)32 * 132 * 4);
     hipLaunchKernelGGL(gemm_gru_kernel<32>, grid, dim3(256), lds, stream, A1,
                        A2, Wperm, bperm, h_in, h_new, R, Z, Nn, HN, N, K, K1, H);
   } else {
     const dim3 grid((N + BM - 1) / BM, COL / BN);
-    const size_t lds = max((size_t)(BM + BN) * ROWB, (size_t)BM * 132 * 4);
+    const size_t lds = max((size_t)2 * (BM + BN) * ROWB, (size_t)BM * 132 * 4);
     hipLaunchKernelGGL(gemm_gru_kernel<BM>, grid, dim3(256), lds, stream, A1,
                        A2, Wperm, bperm, h_in, h_new, R, Z, Nn, HN, N, K, K1, H);
   }
