@@ -961,8 +961,8 @@ std::vector<at::Tensor> mlp3_bwd(at::Tensor dlogits, at::Tensor x, at::Tensor h1
                   W1.data_ptr<float>(), W2.data_ptr<float>(), W3.data_ptr<float>(),
                   dh1.data_ptr<float>(), dh2.data_ptr<float>(), mptr<bf16_t>(dx), B, D,
                   cur_stream());
-  auto dW1 = at::empty({D, D}, fopt);
-  auto dW2 = at::empty({D, D}, fopt);
+  auto dW1 = at::zeros({D, D}, fopt);  // atomic r-chunk epilogue
+  auto dW2 = at::zeros({D, D}, fopt);
   auto dW3 = at::empty({1, D}, fopt);
   auto db1 = at::empty({D}, fopt);
   auto db2 = at::empty({D}, fopt);

@@ -190,7 +190,8 @@ class FlowGNNGGNNModule(BaseModule):
             out = gate_pool(ggnn_out, fe, self.pooling.gate_nn, graph)
             if self.encoder_mode:
                 return out
-            if self.out_dim == 256 and len(self.output_layer) == 5:
+            if (self.out_dim == 256 and len(self.output_layer) == 5
+                    and out.shape[0] <= 1024):  # mlp3_wgrad LDS bound
                 return mlp3(out, self.output_layer[0], self.output_layer[2],
                             self.output_layer[4])
             return self.output_layer(out).squeeze(-1)

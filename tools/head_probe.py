@@ -49,3 +49,10 @@ def eager_fb():
     logits = seq(out.float()).squeeze(-1)
     logits.sum().backward()
 print("eager head f+b:", bench(eager_fb), "us")
+
+# isolate mlp3 backward (wgrad included)
+pooled_g = pooled.detach().requires_grad_(True)
+def mlp3_bwd_only():
+    logits = mlp3(pooled_g, seq[0], seq[2], seq[4])
+    logits.sum().backward()
+print("mlp3 f+b:", bench(mlp3_bwd_only), "us")
