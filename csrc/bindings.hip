@@ -77,8 +77,8 @@ void launch_flash_dq(const __hip_bfloat16*, const __hip_bfloat16*, const __hip_b
                      const __hip_bfloat16*, const int*, const float*, const float*, const float*,
                      __hip_bfloat16*, float*, int, int, int, float, int, unsigned,
                      unsigned long long, long, long, long, hipStream_t);
-void launch_wgrad2(const __hip_bfloat16*, const __hip_bfloat16*, float*, int, int, int,
-                   hipStream_t);
+void launch_wgrad2(const __hip_bfloat16*, const __hip_bfloat16*, float*, float*,
+                   int, int, int, hipStream_t);
 void launch_adamw_fused(float*, const float*, float*, float*, long, float, float, float, float,
                         float, const float*, int, hipStream_t);
 void launch_gemm2(const __hip_bfloat16*, const __hip_bfloat16*, const float*,
@@ -333,7 +333,7 @@ at::Tensor wgrad(at::Tensor A, at::Tensor B) {
   CHECK_GPU(B);
   const int K = A.size(0), M = A.size(1), C = B.size(1);
   TORCH_CHECK(B.size(0) == K && M % 64 == 0 && C % 8 == 0);
-  if (M % 128 == 0 && C % 128 == 0 && K % 64 == 0) {
+  if (M % 128 == 0 && C % 128 == 0) {  // wgrad2 zero-fills K tails
     const int tiles = (M / 128) * (C / 128);
     int zsplit = std::max(1, 512 / tiles);
     int kchunk = (K + zsplit - 1) / zsplit;
@@ -342,8 +342,8 @@ at::Tensor wgrad(at::Tensor A, at::Tensor B) {
     zsplit = (K + kchunk - 1) / kchunk;
     auto out = zsplit == 1 ? at::empty({M, C}, A.options().dtype(at::kFloat))
                            : at::zeros({M, C}, A.options().dtype(at::kFloat));
-    launch_wgrad2(ptr<bf16_t>(A), ptr<bf16_t>(B), out.data_ptr<float>(), K, M, C,
-                  cur_stream());
+    launch_wgrad2(ptr<bf16_t>(A), ptr<bf16_t>(B), out.data_ptr<float>(), nullptr,
+                  K, M, C, cur_stream());
     return out;
   }
   auto out = at::zeros({M, C}, A.options().dtype(at::kFloat));
@@ -471,14 +471,17 @@ std::vector<at::Tensor> ggnn_fused_bwd(at::Tensor grad_out, at::Tensor t_indptr,
   auto gWcat = at::zeros({4 * H, 2 * H}, opts.dtype(at::kFloat));
   auto cs4 = at::zeros({4 * H}, opts.dtype(at::kFloat));
   // bias grads (column sums of the gate/message grads) ride along in the
-  // wgrad kernels' A-tile staging — no separate colsum pass
+  // wgrad kernels' A-tile staging — no separate colsum pass. The gate
+  // wgrad is a SPLIT-B product (no [m|h] concat) so it stays on the
+  // generic split-K kernel; the W_e grad (plain A^T B, 128x128) takes the
+  // tr16-subtiled wgrad2 (K tails zero-filled in staging).
   launch_wgrad(ptr<bf16_t>(A_g), ptr<bf16_t>(M), ptr<bf16_t>(HH), gWcat.data_ptr<float>(),
                cs4.data_ptr<float>(), S * N, 4 * H, 2 * H, H, stream);
   auto A_w = Gwh.view({S * N, H});
   auto gW_e = at::zeros({H, H}, opts.dtype(at::kFloat));
   auto cs_e = at::zeros({H}, opts.dtype(at::kFloat));
-  launch_wgrad(ptr<bf16_t>(A_w), ptr<bf16_t>(HH), nullptr, gW_e.data_ptr<float>(),
-               cs_e.data_ptr<float>(), S * N, H, H, H, stream);
+  launch_wgrad2(ptr<bf16_t>(A_w), ptr<bf16_t>(HH), gW_e.data_ptr<float>(),
+                cs_e.data_ptr<float>(), S * N, H, H, stream);
   // scatter gWcat blocks back to the GRUCell weight layout (views are fine
   // as autograd outputs; no contiguous copy)
   auto gW_ih = gWcat.narrow(0, 0, 3 * H).narrow(1, 0, H);

@@ -71,9 +71,14 @@ __device__ __forceinline__ void w2_write(char* lds, const uint4v regs[4],
   }
 }
 
+// csum (optional): column sums of A (bias grads) accumulated by the
+// blockIdx.y == 0 blocks from the staged subtiled image — lets the GGNN
+// batched weight grads (K = steps*nodes, any K) use this kernel without a
+// separate 60 MB colsum pass (same fold as csrc/wgrad.hip)
 __global__ __launch_bounds__(256) void wgrad2_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ B,
-    float* __restrict__ out, int K, int M, int C, int kchunk, int zsplit) {
+    float* __restrict__ out, float* __restrict__ csum, int K, int M, int C,
+    int kchunk, int zsplit) {
   // Natural [k][col] staging (coalesced 16-B LDS writes; the transposed
   // write variant serialized on 2 banks), transposed FRAGMENT READS as
   // scalar LDS loads (~4-way), double-buffered with the T14 load split so
@@ -104,6 +109,8 @@ typedef __attribute__((address_space(3))) bf16x4v lds_bf16x4;
   const int wc = wid & 1;
 
   f32x4 acc[4][4] = {};
+  float cs_acc = 0.f;
+  const bool do_csum = (csum != nullptr) && (blockIdx.y == 0) && (tid < 128);
 
   // subtiled write: thread covers (k = tid>>4 + 16p, cols (tid&15)*8..+7)
   // -> one 16-B write into half ((tid&15)&1) of subtile (k>>2, (tid&15)>>1)
@@ -132,6 +139,14 @@ typedef __attribute__((address_space(3))) bf16x4v lds_bf16x4;
     if (has_next) {  // T14: issue next tile's loads before this tile's MFMAs
       w2_load(A, M, k0 + WBK, m0, tid, a_regs, k_end);
       w2_load(B, C, k0 + WBK, c0, tid, b_regs, k_end);
+    }
+    if (do_csum) {
+      const int klim = min(WBK, k_end - k0);
+      const int cb = tid >> 4, cc = (tid & 15) * 2;
+#pragma unroll 8
+      for (int k = 0; k < klim; ++k)
+        cs_acc += __bfloat162float(*reinterpret_cast<const __bf16*>(
+            A_LDS(cur) + W2_SUB(k >> 2, cb) + (k & 3) * 32 + cc));
     }
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
@@ -173,6 +188,10 @@ typedef __attribute__((address_space(3))) bf16x4v lds_bf16x4;
 
 #undef A_LDS
 #undef B_LDS
+  if (do_csum) {
+    const int m = m0 + (tid >> 4) * 16 + (tid & 15);
+    if (m < M) atomicAdd(csum + m, cs_acc);
+  }
   // epilogue: D col = lane&15 (C dim), row = (lane>>4)*4 + i (M dim)
 #pragma unroll
   for (int fm = 0; fm < 4; ++fm) {
@@ -191,8 +210,8 @@ typedef __attribute__((address_space(3))) bf16x4v lds_bf16x4;
   }
 }
 
-void launch_wgrad2(const bf16* A, const bf16* B, float* out, int K, int M,
-                   int C, hipStream_t stream) {
+void launch_wgrad2(const bf16* A, const bf16* B, float* out, float* csum,
+                   int K, int M, int C, hipStream_t stream) {
   const int tiles = (M / WBM) * (C / WBC);
   int zsplit = max(1, 512 / tiles);
   int kchunk = (K + zsplit - 1) / zsplit;
@@ -201,5 +220,5 @@ void launch_wgrad2(const bf16* A, const bf16* B, float* out, int K, int M,
   zsplit = (K + kchunk - 1) / kchunk;
   const dim3 grid(M / WBM, C / WBC, zsplit);
   hipLaunchKernelGGL(wgrad2_kernel, grid, dim3(256), 73728, stream, A, B, out,
-                     K, M, C, kchunk, zsplit);
+                     csum, K, M, C, kchunk, zsplit);
 }
