@@ -168,10 +168,33 @@ class FlowGNNGGNNModule(BaseModule):
             idx = torch.stack(
                 [graph.ndata[f"_ABS_DATAFLOW_{of}"] for of in ALL_FEATS], dim=1
             )
-            tables = torch.stack([self.all_embeddings[of].weight for of in ALL_FEATS])
+            tables = self._stacked_tables()
             return embed4(tables, idx)
         feat = graph.ndata[self.feature_keys["feature"]]
         return self.embedding(feat)
+
+    def _stacked_tables(self) -> torch.Tensor:
+        """(4, V, 32) stacked embedding tables, cached per weight version
+        (the per-forward torch.stack was one more graph node per step);
+        non-leaf during training so grads still flow to each table."""
+        ws = [self.all_embeddings[of].weight for of in ALL_FEATS]
+        if any(w.requires_grad for w in ws) and torch.is_grad_enabled():
+            return torch.stack(ws)  # autograd path: must stay in the graph
+        from ..ops.transformer import CAPTURE_REFRESH, _weights_epoch
+
+        key = tuple(w._version for w in ws) + (_weights_epoch[0],)
+        cache = getattr(self, "_dfa_tables_cache", None)
+        if cache is None:
+            buf = torch.stack([w.detach() for w in ws])
+            self._dfa_tables_cache = (key, buf)
+        elif CAPTURE_REFRESH[0] or cache[0] != key:
+            buf = cache[1]
+            for i, w in enumerate(ws):
+                buf[i].copy_(w.detach())
+            self._dfa_tables_cache = (key, buf)
+        else:
+            buf = cache[1]
+        return buf
 
     def forward(self, graph, extrafeats: Optional[Dict] = None) -> torch.Tensor:
         feat_embed = self._embed(graph)
