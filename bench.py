@@ -54,7 +54,7 @@ def build_ddfa(args, rank, device, use_cuda):
     def make_step(opt, ddp, set_to_none):
         flat = hasattr(opt, "allreduce_grads")
 
-        def step(i):
+        def fwd_bwd(i):
             g = batches[i % len(batches)]
             label = model.get_label(g)
             if autocast is not None:
@@ -65,13 +65,22 @@ def build_ddfa(args, rank, device, use_cuda):
             loss = model.loss_fn(logits.float(), label)
             opt.zero_grad(set_to_none=set_to_none and not flat)
             loss.backward()
+            return loss
+
+        def finish():
             if flat:
                 opt.allreduce_grads()
             elif ddp is not None:
                 ddp.finalize()
             opt.step()
+
+        def step(i):
+            loss = fwd_bwd(i)
+            finish()
             return loss
 
+        step.fwd_bwd = fwd_bwd
+        step.finish = finish
         return step
 
     meta = dict(
@@ -123,7 +132,7 @@ def build_linevul(args, rank, device, use_cuda, with_ddfa: bool):
     autocast = torch.autocast(device_type="cuda", dtype=torch.bfloat16) if use_cuda else None
 
     def make_step(opt, ddp, set_to_none):
-        def step(i):
+        def fwd_bwd(i):
             item = batches[i % len(batches)]
             ids, labels = item[0], item[1]
             g = item[2] if with_ddfa else None
@@ -134,11 +143,20 @@ def build_linevul(args, rank, device, use_cuda, with_ddfa: bool):
                 loss, _prob = model(ids, labels=labels, graphs=g)
             opt.zero_grad(set_to_none=set_to_none)
             loss.backward()
+            return loss
+
+        def finish():
             opt.allreduce_grads()
             opt.clip_grad_norm_(1.0)
             opt.step()
+
+        def step(i):
+            loss = fwd_bwd(i)
+            finish()
             return loss
 
+        step.fwd_bwd = fwd_bwd
+        step.finish = finish
         return step
 
     name = "LineVul+DeepDFA(CodeBERT-base+FlowGNN)" if with_ddfa else "LineVul(CodeBERT-base)"
@@ -192,7 +210,7 @@ def build_codet5(args, rank, device, use_cuda, with_ddfa: bool):
     autocast = torch.autocast(device_type="cuda", dtype=torch.bfloat16) if use_cuda else None
 
     def make_step(opt, ddp, set_to_none):
-        def step(i):
+        def fwd_bwd(i):
             item = batches[i % len(batches)]
             ids, labels = item[0], item[1]
             g = item[2] if with_ddfa else None
@@ -203,11 +221,20 @@ def build_codet5(args, rank, device, use_cuda, with_ddfa: bool):
                 loss, _prob = model(ids, labels=labels, graphs=g)
             opt.zero_grad(set_to_none=set_to_none)
             loss.backward()
+            return loss
+
+        def finish():
             opt.allreduce_grads()
             opt.clip_grad_norm_(1.0)
             opt.step()
+
+        def step(i):
+            loss = fwd_bwd(i)
+            finish()
             return loss
 
+        step.fwd_bwd = fwd_bwd
+        step.finish = finish
         return step
 
     name = "CodeT5+DeepDFA(codet5-base+FlowGNN)" if with_ddfa else "CodeT5(codet5-base)"
@@ -270,7 +297,7 @@ def main():
             args, rank, device, use_cuda, with_ddfa=(args.model == "codet5_ddfa")
         )
 
-    use_graphs = use_cuda and capture_ok and not args.no_graph_capture and ws == 1
+    use_graphs = use_cuda and capture_ok and not args.no_graph_capture
     opt = opt_fn(capturable=use_graphs)
     if args.model == "ddfa":
         if hasattr(opt, "flat_p"):
@@ -297,15 +324,28 @@ def main():
         torch.cuda.synchronize()
         pool = torch.cuda.graph_pool_handle()
         graphs = []
-        for i in range(args.n_batches):
-            cg = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(cg, pool=pool):
-                step(i)
-            graphs.append(cg)
-        torch.cuda.synchronize()
+        # multi-rank: capture the compute only (fwd+bwd); the RCCL
+        # all-reduce + optimizer run eagerly between replays so no
+        # collective sits inside a hipGraph (and N>1 keeps the captured
+        # step rate instead of dropping to ~100 eager launches/step)
+        capture_fn = step if ws == 1 else step.fwd_bwd
+        finish_fn = None if ws == 1 else step.finish
+        try:
+            for i in range(args.n_batches):
+                cg = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(cg, pool=pool):
+                    capture_fn(i)
+                graphs.append(cg)
+            torch.cuda.synchronize()
 
-        def step(i):  # noqa: F811 — replay path
-            graphs[i % len(graphs)].replay()
+            def step(i):  # noqa: F811 — replay path
+                graphs[i % len(graphs)].replay()
+                if finish_fn is not None:
+                    finish_fn()
+        except RuntimeError as e:  # capture unsupported: stay eager
+            print(f"[bench] graph capture failed ({e}); running eager",
+                  file=sys.stderr)
+            torch.cuda.synchronize()
 
     def barrier_sync():
         if ws > 1:

@@ -148,7 +148,9 @@ typedef __attribute__((address_space(3))) bf16x4v lds_bf16x4;
         cs_acc += __bfloat162float(*reinterpret_cast<const __bf16*>(
             A_LDS(cur) + W2_SUB(k >> 2, cb) + (k & 3) * 32 + cc));
     }
-#pragma unroll
+    // ks stays a ROLLED loop: fully unrolling doubled the inner body and
+    // the kernel ran at 34% SQ_WAIT_INST_ANY (instruction-fetch starved)
+#pragma unroll 1
     for (int ks = 0; ks < 2; ++ks) {
       const int kb0 = ks * 8 + (lane >> 4) * 2;  // first 4-k subtile row
       const int slot = (lane & 15) * 8;          // this lane's 8-B column slot
