@@ -80,7 +80,7 @@ void launch_flash_dq(const __hip_bfloat16*, const __hip_bfloat16*, const __hip_b
 void launch_wgrad2(const __hip_bfloat16*, const __hip_bfloat16*, float*, float*,
                    int, int, int, hipStream_t);
 void launch_adamw_fused(float*, const float*, float*, float*, long, float, float, float, float,
-                        float, const float*, int, hipStream_t);
+                        float, const float*, const float*, int, hipStream_t);
 void launch_gemm2(const __hip_bfloat16*, const __hip_bfloat16*, const float*,
                   const __hip_bfloat16*, __hip_bfloat16*, int, int, int, hipStream_t);
 void launch_flash_dkv(const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*,
@@ -980,16 +980,21 @@ std::vector<at::Tensor> mlp3_bwd(at::Tensor dlogits, at::Tensor x, at::Tensor h1
 
 void adamw_fused(at::Tensor p, at::Tensor g, at::Tensor m, at::Tensor v, double lr,
                  double beta1, double beta2, double eps, double weight_decay, at::Tensor step,
-                 bool l2_mode) {
+                 c10::optional<at::Tensor> gclip, bool l2_mode) {
   CHECK_GPU(p);
   TORCH_CHECK(p.scalar_type() == at::kFloat && g.scalar_type() == at::kFloat);
   TORCH_CHECK(p.numel() == g.numel() && p.numel() == m.numel() && p.numel() == v.numel());
   TORCH_CHECK(step.is_cuda() && step.scalar_type() == at::kFloat && step.numel() == 1,
               "step must be a device float[1] step counter (pre-incremented)");
+  const float* gc = nullptr;
+  if (gclip.has_value()) {
+    TORCH_CHECK(gclip->is_cuda() && gclip->scalar_type() == at::kFloat);
+    gc = gclip->data_ptr<float>();
+  }
   launch_adamw_fused(p.data_ptr<float>(), g.data_ptr<float>(), m.data_ptr<float>(),
                      v.data_ptr<float>(), p.numel(), (float)lr, (float)beta1, (float)beta2,
-                     (float)eps, (float)weight_decay, step.data_ptr<float>(), l2_mode ? 1 : 0,
-                     cur_stream());
+                     (float)eps, (float)weight_decay, step.data_ptr<float>(), gc,
+                     l2_mode ? 1 : 0, cur_stream());
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {

@@ -182,8 +182,10 @@ class FlowGNNGGNNModule(BaseModule):
             return torch.stack(ws)  # autograd path: must stay in the graph
         from ..ops.transformer import CAPTURE_REFRESH, _weights_epoch
 
-        key = tuple(w._version for w in ws) + (_weights_epoch[0],)
+        key = tuple(w._version for w in ws) + (_weights_epoch[0], str(ws[0].device))
         cache = getattr(self, "_dfa_tables_cache", None)
+        if cache is not None and cache[1].device != ws[0].device:
+            cache = None  # model moved after caching
         if cache is None:
             buf = torch.stack([w.detach() for w in ws])
             self._dfa_tables_cache = (key, buf)
