@@ -324,7 +324,18 @@ class T5Stack(nn.Module):
             x = x.to(torch.bfloat16)
         t_new = state.t + 1
         attn0 = self.block[0].layer[0].SelfAttention
-        bias = attn0.compute_bias(t_new, t_new, ids_step.device)[:, :, -1:, :]
+        # last-query-row bias only: compute_bias builds the (buckets, t*t)
+        # one-hot for training grads — per generated token that is O(t^2)
+        # host work; the decode path needs just the (H, 1, t) slice
+        rel = torch.arange(t_new, device=ids_step.device) - (t_new - 1)
+        buckets = t5_relative_position_bucket(
+            rel, bidirectional=False,
+            num_buckets=self.cfg.relative_attention_num_buckets,
+            max_distance=self.cfg.relative_attention_max_distance,
+        )
+        with torch.no_grad():
+            bias = attn0.relative_attention_bias.weight[buckets].t().reshape(
+                1, self.cfg.num_heads, 1, t_new).float()
         for blk, cache in zip(self.block, state.caches):
             h = blk.layer[0].layer_norm(x)
             x = x + blk.layer[0].SelfAttention.decode_step(
