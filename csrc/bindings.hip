@@ -110,6 +110,11 @@ void launch_ln_res_dropout_wgrad(const T*, const T*, const T*, const float*, con
                                  hipStream_t);
 template <typename T>
 void launch_embed_scatter(const T*, const long*, float*, long, int, long, hipStream_t);
+template <typename T>
+void launch_relu_dropout_fwd(const T*, T*, long, float, unsigned long long, hipStream_t);
+template <typename T>
+void launch_relu_dropout_bwd(const T*, const T*, T*, long, float, unsigned long long,
+                             hipStream_t);
 
 #define CHECK_GPU(t) \
   TORCH_CHECK((t).is_cuda() && (t).is_contiguous(), #t " must be contiguous GPU tensor")
@@ -743,7 +748,8 @@ std::vector<at::Tensor> flash_attn_bwd(at::Tensor dO, at::Tensor Q, at::Tensor K
                                        c10::optional<at::Tensor> valid,
                                        c10::optional<at::Tensor> bias, double scale, bool causal,
                                        double dropout_p, int64_t seed, bool need_dbias,
-                                       bool fused_grads) {
+                                       bool fused_grads,
+                                       c10::optional<at::Tensor> dbias_accum = c10::nullopt) {
   CHECK_GPU(dO);
   const int B = Q.size(0), L = Q.size(1);
   const long ldq = fa_ld(Q, L);
@@ -775,7 +781,14 @@ std::vector<at::Tensor> flash_attn_bwd(at::Tensor dO, at::Tensor Q, at::Tensor K
   if (bias.has_value()) bptr = bias->data_ptr<float>();
   at::Tensor dBias;
   float* dbias_ptr = nullptr;
-  if (need_dbias) {
+  if (dbias_accum.has_value()) {
+    // shared accumulation buffer (T5: 24 layers share one position bias;
+    // the dq kernel's atomics land in ONE tensor instead of 24 separate
+    // dBias allocations + the autograd fan-in adds)
+    TORCH_CHECK(dbias_accum->scalar_type() == at::kFloat &&
+                dbias_accum->numel() == (long)H * L * L);
+    dbias_ptr = dbias_accum->data_ptr<float>();
+  } else if (need_dbias) {
     TORCH_CHECK(bias.has_value());
     dBias = at::zeros({H, L, L}, Q.options().dtype(at::kFloat));
     dbias_ptr = dBias.data_ptr<float>();
@@ -791,10 +804,10 @@ std::vector<at::Tensor> flash_attn_bwd(at::Tensor dO, at::Tensor Q, at::Tensor K
                    causal ? 1 : 0, (unsigned)(dropout_p * 256.0), (unsigned long long)seed,
                    ldq, ldkv, ldout, stream);
   if (fused_grads) {
-    if (need_dbias) return {dQKV, dBias};
+    if (need_dbias && !dbias_accum.has_value()) return {dQKV, dBias};
     return {dQKV};
   }
-  if (need_dbias) return {dQ, dK, dV, dBias};
+  if (need_dbias && !dbias_accum.has_value()) return {dQ, dK, dV, dBias};
   return {dQ, dK, dV};
 }
 
@@ -1027,9 +1040,34 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_bwd", &rmsnorm_bwd);
   m.def("rmsnorm_wgrad", &rmsnorm_wgrad);
   m.def("flash_attn_fwd", &flash_attn_fwd);
-  m.def("flash_attn_bwd", &flash_attn_bwd);
+  m.def("flash_attn_bwd", &flash_attn_bwd, pybind11::arg("dO"), pybind11::arg("Q"),
+        pybind11::arg("K"), pybind11::arg("V"), pybind11::arg("O"), pybind11::arg("lse"),
+        pybind11::arg("H"), pybind11::arg("valid"), pybind11::arg("bias"),
+        pybind11::arg("scale"), pybind11::arg("causal"), pybind11::arg("dropout_p"),
+        pybind11::arg("seed"), pybind11::arg("need_dbias"), pybind11::arg("fused_grads"),
+        pybind11::arg("dbias_accum") = pybind11::none());
   m.def("lmhead_ce_fwd", &lmhead_ce_fwd);
   m.def("lmhead_ce_bwd", &lmhead_ce_bwd);
+  m.def("relu_dropout_fwd", [](at::Tensor x, double p, int64_t seed) {
+    CHECK_GPU(x);
+    auto out = at::empty_like(x);
+    dispatch_float_bf16(x, "relu_dropout_fwd", [&](auto tag) {
+      using T = decltype(tag);
+      launch_relu_dropout_fwd<T>(ptr<T>(x), mptr<T>(out), x.numel(), (float)p,
+                                 (unsigned long long)seed, cur_stream());
+    });
+    return out;
+  });
+  m.def("relu_dropout_bwd", [](at::Tensor dy, at::Tensor x, double p, int64_t seed) {
+    CHECK_GPU(dy);
+    auto dx = at::empty_like(dy);
+    dispatch_float_bf16(dy, "relu_dropout_bwd", [&](auto tag) {
+      using T = decltype(tag);
+      launch_relu_dropout_bwd<T>(ptr<T>(dy), ptr<T>(x), mptr<T>(dx), dy.numel(),
+                                 (float)p, (unsigned long long)seed, cur_stream());
+    });
+    return dx;
+  });
   m.def("dropout_add_fwd", [](at::Tensor h, at::Tensor res, double p, int64_t seed) {
     CHECK_GPU(h);
     CHECK_GPU(res);

@@ -1094,3 +1094,95 @@ void launch_dropout_add_bwd(const T* dy, T* dh, long total, float p,
 INST_DA(float)
 INST_DA(__hip_bfloat16)
 #undef INST_DA
+
+// Fused ReLU + dropout (T5 FFN: wo(dropout(relu(wi x))) — torch ran relu,
+// a bernoulli mask, a mul and a masked_scale backward as 4 separate
+// passes over the (N, 3072) activation). Stateless: the keep mask is
+// regenerated from the seed in backward; only the pre-ReLU input is saved.
+template <typename T>
+__global__ void relu_dropout_fwd_kernel(const T* __restrict__ x,
+                                        T* __restrict__ out, long total,
+                                        unsigned p8, unsigned long long seed,
+                                        float dscale) {
+  constexpr int VEC = 16 / sizeof(T);
+  const long nvec = total / VEC;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long iv = (long)blockIdx.x * blockDim.x + threadIdx.x; iv < nvec;
+       iv += stride) {
+    T vx[VEC], vo[VEC];
+    *reinterpret_cast<ulonglong2*>(vx) =
+        *reinterpret_cast<const ulonglong2*>(x + iv * VEC);
+#pragma unroll
+    for (int u = 0; u < VEC; ++u) {
+      const bool keep = (p8 == 0) || keep_mask(iv * VEC + u, seed, p8);
+      const float r = fmaxf(tf(vx[u]), 0.f);
+      vo[u] = ff<T>(keep ? r * dscale : 0.f);
+    }
+    *reinterpret_cast<ulonglong2*>(out + iv * VEC) =
+        *reinterpret_cast<const ulonglong2*>(vo);
+  }
+}
+
+template <typename T>
+__global__ void relu_dropout_bwd_kernel(const T* __restrict__ dy,
+                                        const T* __restrict__ x,
+                                        T* __restrict__ dx, long total,
+                                        unsigned p8, unsigned long long seed,
+                                        float dscale) {
+  constexpr int VEC = 16 / sizeof(T);
+  const long nvec = total / VEC;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long iv = (long)blockIdx.x * blockDim.x + threadIdx.x; iv < nvec;
+       iv += stride) {
+    T vd[VEC], vx[VEC], vo[VEC];
+    *reinterpret_cast<ulonglong2*>(vd) =
+        *reinterpret_cast<const ulonglong2*>(dy + iv * VEC);
+    *reinterpret_cast<ulonglong2*>(vx) =
+        *reinterpret_cast<const ulonglong2*>(x + iv * VEC);
+#pragma unroll
+    for (int u = 0; u < VEC; ++u) {
+      const bool keep = (p8 == 0) || keep_mask(iv * VEC + u, seed, p8);
+      const bool pos = tf(vx[u]) > 0.f;
+      vo[u] = ff<T>((keep && pos) ? tf(vd[u]) * dscale : 0.f);
+    }
+    *reinterpret_cast<ulonglong2*>(dx + iv * VEC) =
+        *reinterpret_cast<const ulonglong2*>(vo);
+  }
+}
+
+template <typename T>
+void launch_relu_dropout_fwd(const T* x, T* out, long total, float dropout_p,
+                             unsigned long long seed, hipStream_t stream) {
+  constexpr int VEC = 16 / sizeof(T);
+  const unsigned p8 = (unsigned)(dropout_p * 256.0f);
+  const float dscale = (p8 > 0) ? 256.0f / (256.0f - p8) : 1.0f;
+  const int block = 256;
+  const int grid = (int)min((total / VEC + block - 1) / block, (long)4096);
+  hipLaunchKernelGGL(relu_dropout_fwd_kernel<T>, dim3(grid), dim3(block), 0,
+                     stream, x, out, total, p8, seed, dscale);
+}
+
+template <typename T>
+void launch_relu_dropout_bwd(const T* dy, const T* x, T* dx, long total,
+                             float dropout_p, unsigned long long seed,
+                             hipStream_t stream) {
+  constexpr int VEC = 16 / sizeof(T);
+  const unsigned p8 = (unsigned)(dropout_p * 256.0f);
+  const float dscale = (p8 > 0) ? 256.0f / (256.0f - p8) : 1.0f;
+  const int block = 256;
+  const int grid = (int)min((total / VEC + block - 1) / block, (long)4096);
+  hipLaunchKernelGGL(relu_dropout_bwd_kernel<T>, dim3(grid), dim3(block), 0,
+                     stream, dy, x, dx, total, p8, seed, dscale);
+}
+
+template void launch_relu_dropout_fwd<__hip_bfloat16>(
+    const __hip_bfloat16*, __hip_bfloat16*, long, float, unsigned long long,
+    hipStream_t);
+template void launch_relu_dropout_bwd<__hip_bfloat16>(
+    const __hip_bfloat16*, const __hip_bfloat16*, __hip_bfloat16*, long, float,
+    unsigned long long, hipStream_t);
+template void launch_relu_dropout_fwd<float>(
+    const float*, float*, long, float, unsigned long long, hipStream_t);
+template void launch_relu_dropout_bwd<float>(
+    const float*, const float*, float*, long, float, unsigned long long,
+    hipStream_t);

@@ -169,7 +169,8 @@ class T5Attention(nn.Module):
         ctx = torch.matmul(probs, v).transpose(1, 2).reshape(B, Lq, H * d)
         return fused_linear(ctx, self.o.weight)
 
-    def forward(self, x, valid, kv=None, position_bias=None, dropout_p=0.0):
+    def forward(self, x, valid, kv=None, position_bias=None, dropout_p=0.0,
+                bias_accum=None):
         B, Lq, _ = x.shape
         src = kv if kv is not None else x
         Lk = src.shape[1]
@@ -189,9 +190,11 @@ class T5Attention(nn.Module):
 
             bias = None
             if position_bias is not None:
-                bias = position_bias.squeeze(0).float().contiguous()
+                bsrc = position_bias.detach() if bias_accum is not None else position_bias
+                bias = bsrc.squeeze(0).float().contiguous()
             out = flash_attention_qkv(qkv, H, valid=valid, bias=bias, scale=1.0,
-                                      causal=causal, dropout_p=dropout_p)
+                                      causal=causal, dropout_p=dropout_p,
+                                      bias_accum=bias_accum)
             return fused_linear(out, self.o.weight)
         qp = fused_linear(x, self.q.weight)
         kp = fused_linear(src, self.k.weight)
@@ -199,9 +202,11 @@ class T5Attention(nn.Module):
         if d == 64 and flash_usable(qp, Lq, Lk):
             bias = None
             if position_bias is not None:
-                bias = position_bias.squeeze(0).float().contiguous()
+                bsrc = position_bias.detach() if bias_accum is not None else position_bias
+                bias = bsrc.squeeze(0).float().contiguous()
             out = flash_attention(qp, kp, vp, H, valid=valid, bias=bias, scale=1.0,
-                                  causal=causal, dropout_p=dropout_p)
+                                  causal=causal, dropout_p=dropout_p,
+                                  bias_accum=bias_accum)
             return fused_linear(out, self.o.weight)
         q = split(qp, Lq)
         k = split(kp, Lk)
@@ -222,9 +227,9 @@ class T5LayerSelfAttention(nn.Module):
         self.layer_norm = T5LayerNorm(cfg.d_model, cfg.layer_norm_epsilon)
         self.dropout = nn.Dropout(cfg.dropout_rate)
 
-    def forward(self, x, valid, position_bias, dropout_p):
+    def forward(self, x, valid, position_bias, dropout_p, bias_accum=None):
         y = self.SelfAttention(self.layer_norm(x), valid, position_bias=position_bias,
-                               dropout_p=dropout_p)
+                               dropout_p=dropout_p, bias_accum=bias_accum)
         return dropout_add(y, x, self.dropout.p, self.training)
 
 
@@ -248,8 +253,11 @@ class T5DenseActDense(nn.Module):
         self.dropout = nn.Dropout(cfg.dropout_rate)
 
     def forward(self, x):
-        h = torch.relu(fused_linear(x, self.wi.weight))
-        return fused_linear(self.dropout(h), self.wo.weight)
+        from ..ops.transformer import relu_dropout
+
+        h = relu_dropout(fused_linear(x, self.wi.weight), self.dropout.p,
+                         self.training)
+        return fused_linear(h, self.wo.weight)
 
 
 class T5LayerFF(nn.Module):
@@ -274,8 +282,9 @@ class T5Block(nn.Module):
         layers.append(T5LayerFF(cfg))
         self.layer = nn.ModuleList(layers)
 
-    def forward(self, x, valid, position_bias, enc=None, enc_valid=None, dropout_p=0.0):
-        x = self.layer[0](x, valid, position_bias, dropout_p)
+    def forward(self, x, valid, position_bias, enc=None, enc_valid=None,
+                dropout_p=0.0, bias_accum=None):
+        x = self.layer[0](x, valid, position_bias, dropout_p, bias_accum)
         if self.is_decoder:
             x = self.layer[1](x, enc, enc_valid, dropout_p)
         return self.layer[-1](x)
@@ -306,9 +315,21 @@ class T5Stack(nn.Module):
         L = input_ids.shape[1]
         attn0 = self.block[0].layer[0].SelfAttention
         position_bias = attn0.compute_bias(L, L, input_ids.device)
+        bias_accum = None
+        if (x.is_cuda and torch.is_grad_enabled()
+                and position_bias.requires_grad):
+            bias_accum = torch.zeros(
+                self.cfg.num_heads, L, L, dtype=torch.float32, device=x.device
+            )
+            x = _BiasGradSink.apply(x, position_bias, bias_accum)
+            # position_bias stays differentiable: the flash paths detach it
+            # (their dBias atomics land in bias_accum); any layer that falls
+            # back to the torch path keeps the autograd route, and the two
+            # contributions sum correctly
         p = self.cfg.dropout_rate if self.training else 0.0
         for blk in self.block:
-            x = blk(x, valid, position_bias, enc=enc, enc_valid=enc_valid, dropout_p=p)
+            x = blk(x, valid, position_bias, enc=enc, enc_valid=enc_valid,
+                    dropout_p=p, bias_accum=bias_accum)
         return self.dropout(self.final_layer_norm(x))
 
     @torch.no_grad()
@@ -346,6 +367,28 @@ class T5Stack(nn.Module):
             x = blk.layer[2](x)
         state.t = t_new
         return self.final_layer_norm(x)
+
+
+class _BiasGradSink(torch.autograd.Function):
+    """Applied to the stack INPUT so its backward runs AFTER every layer's:
+    returns the shared atomically-accumulated position-bias gradient as the
+    grad of the (differentiable) bias — the 24 layers then consume a
+    DETACHED bias and the flash dq kernels' dBias atomics all land in one
+    buffer (replacing 24 per-layer dBias tensors + the autograd fan-in
+    adds)."""
+
+    @staticmethod
+    def forward(ctx, x, bias, accum):
+        ctx.accum = accum
+        ctx.bias_shape = bias.shape
+        # view_as: returning the input unchanged would bypass this node
+        # (autograd keeps the original grad_fn) and the bias grad would
+        # never be emitted
+        return x.view_as(x)
+
+    @staticmethod
+    def backward(ctx, g):
+        return g, ctx.accum.view(ctx.bias_shape), None
 
 
 class _DecodeState:

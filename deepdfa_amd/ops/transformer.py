@@ -233,7 +233,8 @@ class _FlashAttention(torch.autograd.Function):
     `bias` is the T5 additive position bias (fp32, (H, L, L))."""
 
     @staticmethod
-    def forward(ctx, q, k, v, H, valid, bias, scale, causal, dropout_p):
+    def forward(ctx, q, k, v, H, valid, bias, scale, causal, dropout_p,
+                bias_accum=None):
         ext = load_ext(required=True)
         seed = _next_seed() if dropout_p > 0 else 0
         # q/k/v may be row-strided slices of a fused QKV buffer — the
@@ -241,7 +242,7 @@ class _FlashAttention(torch.autograd.Function):
         O, lse = ext.flash_attn_fwd(q, k, v, H, valid, bias,
                                     scale, causal, dropout_p, seed)
         ctx.save_for_backward(q, k, v, O, lse)
-        ctx.meta = (H, valid, bias, scale, causal, dropout_p, seed)
+        ctx.meta = (H, valid, bias, scale, causal, dropout_p, seed, bias_accum)
         ctx.need_dbias = bias is not None and ctx.needs_input_grad[5]
         return O
 
@@ -249,18 +250,21 @@ class _FlashAttention(torch.autograd.Function):
     def backward(ctx, dO):
         ext = load_ext(required=True)
         q, k, v, O, lse = ctx.saved_tensors
-        H, valid, bias, scale, causal, dropout_p, seed = ctx.meta
+        H, valid, bias, scale, causal, dropout_p, seed, bias_accum = ctx.meta
         outs = ext.flash_attn_bwd(
             dO.contiguous(), q, k, v, O, lse,
             H, valid, bias, scale, causal, dropout_p, seed, ctx.need_dbias, False,
+            bias_accum,
         )
-        dbias = outs[3] if ctx.need_dbias else None
-        return outs[0], outs[1], outs[2], None, None, dbias, None, None, None
+        dbias = outs[3] if (ctx.need_dbias and bias_accum is None) else None
+        return (outs[0], outs[1], outs[2], None, None, dbias, None, None, None,
+                None)
 
 
 def flash_attention(q, k, v, num_heads, valid=None, bias=None, scale=1.0,
-                    causal=False, dropout_p=0.0):
-    return _FlashAttention.apply(q, k, v, num_heads, valid, bias, scale, causal, dropout_p)
+                    causal=False, dropout_p=0.0, bias_accum=None):
+    return _FlashAttention.apply(q, k, v, num_heads, valid, bias, scale, causal,
+                                 dropout_p, bias_accum)
 
 
 def flash_usable(x, L, Lk=None) -> bool:
@@ -482,7 +486,8 @@ class _FlashAttentionQKV(torch.autograd.Function):
     slice-backward zero+scatter work in autograd."""
 
     @staticmethod
-    def forward(ctx, qkv, H, valid, bias, scale, causal, dropout_p):
+    def forward(ctx, qkv, H, valid, bias, scale, causal, dropout_p,
+                bias_accum=None):
         ext = load_ext(required=True)
         D = qkv.shape[-1] // 3
         q, k, v = qkv[..., :D], qkv[..., D:2 * D], qkv[..., 2 * D:]
@@ -490,7 +495,7 @@ class _FlashAttentionQKV(torch.autograd.Function):
         O, lse = ext.flash_attn_fwd(q, k, v, H, valid, bias, scale, causal,
                                     dropout_p, seed)
         ctx.save_for_backward(qkv, O, lse)
-        ctx.meta = (H, valid, bias, scale, causal, dropout_p, seed)
+        ctx.meta = (H, valid, bias, scale, causal, dropout_p, seed, bias_accum)
         ctx.need_dbias = bias is not None and ctx.needs_input_grad[3]
         return O
 
@@ -498,21 +503,24 @@ class _FlashAttentionQKV(torch.autograd.Function):
     def backward(ctx, dO):
         ext = load_ext(required=True)
         qkv, O, lse = ctx.saved_tensors
-        H, valid, bias, scale, causal, dropout_p, seed = ctx.meta
+        H, valid, bias, scale, causal, dropout_p, seed, bias_accum = ctx.meta
         D = qkv.shape[-1] // 3
         q, k, v = qkv[..., :D], qkv[..., D:2 * D], qkv[..., 2 * D:]
         outs = ext.flash_attn_bwd(
             dO.contiguous(), q, k, v, O, lse,
             H, valid, bias, scale, causal, dropout_p, seed, ctx.need_dbias, True,
+            bias_accum,
         )
-        dbias = outs[1] if ctx.need_dbias else None
-        return outs[0], None, None, dbias, None, None, None
+        # bias_accum set: the dq kernel atomics landed in the SHARED buffer
+        # (T5's 24 layers share one position bias) — no per-layer dBias
+        dbias = outs[1] if (ctx.need_dbias and bias_accum is None) else None
+        return outs[0], None, None, dbias, None, None, None, None
 
 
 def flash_attention_qkv(qkv, num_heads, valid=None, bias=None, scale=1.0,
-                        causal=False, dropout_p=0.0):
+                        causal=False, dropout_p=0.0, bias_accum=None):
     return _FlashAttentionQKV.apply(qkv, num_heads, valid, bias, scale, causal,
-                                    dropout_p)
+                                    dropout_p, bias_accum)
 
 
 class _LMHeadCE(torch.autograd.Function):
@@ -591,3 +599,34 @@ def lmhead_cross_entropy(h, weight, targets, scale: float = 1.0):
     return torch.nn.functional.cross_entropy(
         logits.float(), targets.reshape(-1), ignore_index=-100
     )
+
+
+class _ReluDropout(torch.autograd.Function):
+    """Fused ReLU + dropout (T5 FFN inner activation): stateless mask from
+    the seed, only the pre-ReLU input saved."""
+
+    @staticmethod
+    def forward(ctx, x, p, training):
+        ext = load_ext(required=True)
+        seed = _next_seed() if (training and p > 0) else 0
+        pp = p if training else 0.0
+        xc = x.contiguous()
+        out = ext.relu_dropout_fwd(xc, pp, seed)
+        ctx.save_for_backward(xc)
+        ctx.p = pp
+        ctx.seed = seed
+        return out
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = load_ext(required=True)
+        (x,) = ctx.saved_tensors
+        dx = ext.relu_dropout_bwd(dy.contiguous(), x, ctx.p, ctx.seed)
+        return dx, None, None
+
+
+def relu_dropout(x, p=0.0, training=True):
+    if x.is_cuda:
+        return _ReluDropout.apply(x, p, training)
+    h = torch.relu(x)
+    return torch.nn.functional.dropout(h, p) if (training and p > 0) else h

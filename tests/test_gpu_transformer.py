@@ -263,3 +263,32 @@ def test_dropout_add_semantics():
     assert torch.all(h1.grad[dropped].float() == 0)
     assert torch.all(h1.grad[~dropped].float() == 2.0)  # 1/(1-p)
     assert torch.equal(r0.grad, g)
+
+
+@pytest.mark.gpu
+def test_t5_shared_bias_grad_matches_cpu():
+    """The 24-layer position-bias gradient now accumulates atomically in
+    ONE shared buffer (_BiasGradSink); the relative_attention_bias weight
+    grad must match the CPU autograd fan-in reference."""
+    from deepdfa_amd.models.t5 import T5Config, T5ForConditionalGeneration
+
+    torch.manual_seed(0)
+    cfg = T5Config(num_layers=3, num_decoder_layers=3, d_model=128, d_ff=256,
+                   num_heads=2, vocab_size=500, dropout_rate=0.0)
+    cpu = T5ForConditionalGeneration(cfg)
+    gpu = T5ForConditionalGeneration(cfg)
+    gpu.load_state_dict(cpu.state_dict())
+    gpu = gpu.to("cuda:0")
+    ids = torch.randint(3, cfg.vocab_size, (2, 64))
+    for model, dev in ((cpu, "cpu"), (gpu, "cuda:0")):
+        model.train()
+        dec = model(ids.to(dev), labels=ids.to(dev), output_hidden_only=True)
+        dec.float().pow(2).mean().backward()
+    for name in ("encoder", "decoder"):
+        wc = getattr(cpu, name).block[0].layer[0].SelfAttention.relative_attention_bias.weight
+        wg = getattr(gpu, name).block[0].layer[0].SelfAttention.relative_attention_bias.weight
+        assert wg.grad is not None and wg.grad.abs().sum() > 0, name
+        ref = wc.grad
+        got = wg.grad.cpu()
+        err = (got - ref).abs().max() / ref.abs().max().clamp(min=1e-8)
+        assert err < 0.08, (name, float(err))
