@@ -38,8 +38,10 @@ def bucket_shape(
     b_extra = b_pad - g.num_graphs
     assert b_extra >= 1, "b_pad must exceed the real graph count"
     n_min = g.num_nodes + b_extra
-    e_min = g.num_edges + b_extra
     n_pad = ((n_min + node_q - 1) // node_q) * node_q
+    # every padded node carries >= 1 self-loop, so the edge budget must
+    # cover the NODE padding (not just the dummy-graph minimum)
+    e_min = g.num_edges + (n_pad - g.num_nodes)
     e_pad = ((e_min + edge_q - 1) // edge_q) * edge_q
     return n_pad, e_pad, b_pad
 
@@ -53,7 +55,7 @@ def pad_batch(
     b_extra = b_pad - B
     n_extra = n_pad - N
     e_extra = e_pad - E
-    assert b_extra >= 1 and n_extra >= b_extra and e_extra >= b_extra, (
+    assert b_extra >= 1 and n_extra >= b_extra and e_extra >= n_extra, (
         f"padding target too small: ({n_pad},{e_pad},{b_pad}) for ({N},{E},{B})"
     )
     dev = g.indptr.device

@@ -207,6 +207,9 @@ def main(argv: Optional[List[str]] = None) -> Dict:
         if args.subcommand == "fit":
             from ..utils.logging import HPOReporter
 
+            # the flat optimizer re-views parameters into its master
+            # buffer — the model must already live on the training device
+            model = model.to(trainer.device)
             params = [p for p in model.parameters() if p.requires_grad]
             if getattr(trainer, "graph_capture", False) and torch.cuda.is_available():
                 # capture-safe flat optimizer (device-side bias correction);

@@ -146,6 +146,17 @@ class Trainer:
         model = model.to(self.device)
         if optimizer is None:
             optimizer = torch.optim.Adam(model.parameters(), lr=1e-3, weight_decay=1e-2)
+        if hasattr(optimizer, "flat_p") and (
+            optimizer.flat_p.device.type != self.device.type
+            or (self.device.index is not None
+                and optimizer.flat_p.device.index != self.device.index)
+        ):
+            raise RuntimeError(
+                "FlatAdamW was built before the model moved to "
+                f"{self.device}: its flat master buffer lives on "
+                f"{optimizer.flat_p.device} and the parameter views are "
+                "broken. Construct it AFTER model.to(device)."
+            )
         # gradient averaging: flat optimizers (FlatAdamW) do ONE flat
         # all-reduce after backward; otherwise the bucketed-overlap engine
         flat = hasattr(optimizer, "allreduce_grads")

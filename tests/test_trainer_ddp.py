@@ -130,3 +130,18 @@ def test_pad_batch_structure():
     # dummy nodes have zero features
     for k, v in padded.ndata.items():
         assert (v[g.num_nodes:] == 0).all()
+
+
+def test_bucket_shape_covers_node_heavy_padding():
+    """Regression: when node padding exceeds the edge quantum's slack the
+    edge budget must still give every padded node its self-loop (a real
+    batch shape hit 'repeats can not be negative' in pad_batch)."""
+    from deepdfa_amd.graph.pad import bucket_shape, pad_batch
+    from deepdfa_amd.graph.synthetic import synthetic_cfg_batch
+
+    g = synthetic_cfg_batch(48, seed=2)
+    # adversarial quanta: huge node quantum, tiny edge quantum
+    shape = bucket_shape(g, b_pad=64, node_q=4096, edge_q=64)
+    padded, w = pad_batch(g, *shape)
+    assert padded.num_nodes == shape[0] and padded.num_edges == shape[1]
+    assert int(w.sum()) == 48
