@@ -889,6 +889,11 @@ at::Tensor lmhead_ce_bwd(at::Tensor h, at::Tensor Wp, at::Tensor targets,
   return dlogits;
 }
 
+void launch_bce_logits_fwd(const float*, const float*, const float*, const float*,
+                           float*, int, hipStream_t);
+void launch_bce_logits_bwd(const float*, const float*, const float*, const float*,
+                           const float*, const float*, float*, int, hipStream_t);
+
 void launch_gate_pool_fwd(const __hip_bfloat16*, const __hip_bfloat16*, const float*,
                           const float*, const int*, __hip_bfloat16*, float*, int, int,
                           int, hipStream_t);
@@ -1026,6 +1031,31 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("colsum", &colsum);
   m.def("ggnn_fused_fwd", &ggnn_fused_fwd);
   m.def("pack_gru_weights", &pack_gru_weights);
+  m.def("bce_logits_fwd", [](at::Tensor logits, at::Tensor labels,
+                             c10::optional<at::Tensor> weight,
+                             c10::optional<at::Tensor> pos_weight) {
+    CHECK_GPU(logits);
+    auto out2 = at::empty({2}, logits.options());
+    launch_bce_logits_fwd(
+        logits.data_ptr<float>(), labels.data_ptr<float>(),
+        weight.has_value() ? weight->data_ptr<float>() : nullptr,
+        pos_weight.has_value() ? pos_weight->data_ptr<float>() : nullptr,
+        out2.data_ptr<float>(), logits.numel(), cur_stream());
+    return out2;
+  });
+  m.def("bce_logits_bwd", [](at::Tensor logits, at::Tensor labels,
+                             c10::optional<at::Tensor> weight,
+                             c10::optional<at::Tensor> pos_weight,
+                             at::Tensor grad, at::Tensor out2) {
+    auto dlogits = at::empty_like(logits);
+    launch_bce_logits_bwd(
+        logits.data_ptr<float>(), labels.data_ptr<float>(),
+        weight.has_value() ? weight->data_ptr<float>() : nullptr,
+        pos_weight.has_value() ? pos_weight->data_ptr<float>() : nullptr,
+        grad.data_ptr<float>(), out2.data_ptr<float>(),
+        dlogits.data_ptr<float>(), logits.numel(), cur_stream());
+    return dlogits;
+  });
   m.def("gate_pool_fwd", &gate_pool_fwd);
   m.def("gate_pool_bwd", &gate_pool_bwd);
   m.def("mlp3_fwd", &mlp3_fwd);

@@ -77,15 +77,13 @@ class BaseModule(nn.Module):
         weight: Optional[torch.Tensor] = None,
     ) -> torch.Tensor:
         """weight (per-graph 0/1) masks out capture-padding dummy graphs
-        (graph/pad.py): the weighted mean equals the unpadded batch mean."""
-        if weight is not None:
-            per = torch.nn.functional.binary_cross_entropy_with_logits(
-                logits, label, pos_weight=self.pos_weight, reduction="none"
-            )
-            return (per * weight).sum() / weight.sum().clamp(min=1.0)
-        return torch.nn.functional.binary_cross_entropy_with_logits(
-            logits, label, pos_weight=self.pos_weight
-        )
+        (graph/pad.py): the weighted mean equals the unpadded batch mean.
+        GPU path = ONE fused kernel each way (ops.flowgnn.bce_with_logits);
+        torch's chain was ~6 launch-floor nodes at batch 256."""
+        from ..ops.flowgnn import bce_with_logits
+
+        return bce_with_logits(logits, label, weight=weight,
+                               pos_weight=self.pos_weight)
 
     # -- steps (called by the trainer) ---------------------------------------
 

@@ -386,3 +386,32 @@ class TestFusedHead:
             denom = r.abs().max().clamp(min=1e-4)
             err = (t - r).abs().max() / denom
             assert err < 0.35, (n, float(err))
+
+
+@pytest.mark.gpu
+def test_bce_with_logits_fused_matches_torch():
+    from deepdfa_amd.ops.flowgnn import bce_with_logits
+
+    torch.manual_seed(0)
+    dev = torch.device("cuda:0")
+    for B, pw, use_w in [(256, None, False), (257, 3.5, True), (64, 1.7, False)]:
+        logits = torch.randn(B, device=dev, requires_grad=True)
+        labels = (torch.rand(B, device=dev) < 0.4).float()
+        weight = None
+        if use_w:
+            weight = torch.ones(B, device=dev)
+            weight[-5:] = 0.0
+        pos_weight = torch.tensor([pw], device=dev) if pw else None
+        loss = bce_with_logits(logits, labels, weight=weight, pos_weight=pos_weight)
+        loss.backward()
+        ref_in = logits.detach().clone().requires_grad_(True)
+        if weight is not None:
+            per = torch.nn.functional.binary_cross_entropy_with_logits(
+                ref_in, labels, pos_weight=pos_weight, reduction="none")
+            ref = (per * weight).sum() / weight.sum()
+        else:
+            ref = torch.nn.functional.binary_cross_entropy_with_logits(
+                ref_in, labels, pos_weight=pos_weight)
+        ref.backward()
+        assert torch.allclose(loss, ref, rtol=1e-5, atol=1e-6), (float(loss), float(ref))
+        assert torch.allclose(logits.grad, ref_in.grad, rtol=1e-4, atol=1e-7)
