@@ -110,6 +110,8 @@ void launch_ln_res_dropout_wgrad(const T*, const T*, const T*, const float*, con
                                  hipStream_t);
 template <typename T>
 void launch_embed_scatter(const T*, const long*, float*, long, int, long, hipStream_t);
+void launch_relbias_wgrad(const float*, const int*, float*, long, int, int,
+                          hipStream_t);
 template <typename T>
 void launch_relu_dropout_fwd(const T*, T*, long, float, unsigned long long, hipStream_t);
 template <typename T>
@@ -1078,6 +1080,18 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         pybind11::arg("dbias_accum") = pybind11::none());
   m.def("lmhead_ce_fwd", &lmhead_ce_fwd);
   m.def("lmhead_ce_bwd", &lmhead_ce_bwd);
+  m.def("relbias_wgrad", [](at::Tensor dbias, at::Tensor buckets, int64_t nb) {
+    CHECK_GPU(dbias);
+    TORCH_CHECK(dbias.scalar_type() == at::kFloat &&
+                buckets.scalar_type() == at::kInt && buckets.is_contiguous());
+    const int H = dbias.size(0);
+    const long QK = dbias.numel() / H;
+    TORCH_CHECK(buckets.numel() == QK);
+    auto dw = at::zeros({nb, (long)H}, dbias.options());
+    launch_relbias_wgrad(dbias.data_ptr<float>(), buckets.data_ptr<int>(),
+                         dw.data_ptr<float>(), QK, H, (int)nb, cur_stream());
+    return dw;
+  });
   m.def("relu_dropout_fwd", [](at::Tensor x, double p, int64_t seed) {
     CHECK_GPU(x);
     auto out = at::empty_like(x);

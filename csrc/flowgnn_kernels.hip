@@ -916,6 +916,7 @@ void launch_gate_pool_bwd(const __hip_bfloat16* grad_out, const __hip_bfloat16* 
 // (ILP breaks the serial accumulation latency; 4x fewer weight reads per
 // row) and the grid covers ceil(B/RB) blocks for occupancy.
 #define MLP_RB 4
+template <int RB>
 __global__ void mlp3_fwd_kernel(
     const __hip_bfloat16* __restrict__ x, const float* __restrict__ W1T,
     const float* __restrict__ b1, const float* __restrict__ W2T,
@@ -924,20 +925,20 @@ __global__ void mlp3_fwd_kernel(
     float* __restrict__ h2, float* __restrict__ logits, int B, int D) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   float* xs = reinterpret_cast<float*>(smem);  // RB * D
-  float* scratch = xs + MLP_RB * D;            // 8
+  float* scratch = xs + RB * D;                // 8
   const int j = threadIdx.x;
-  const int r0 = blockIdx.x * MLP_RB;
-  const int nr = min(MLP_RB, B - r0);
+  const int r0 = blockIdx.x * RB;
+  const int nr = min(RB, B - r0);
   if (nr <= 0) return;
   for (int r = 0; r < nr; ++r) xs[r * D + j] = to_f(x[(long)(r0 + r) * D + j]);
   __syncthreads();
-  float a[MLP_RB];
+  float a[RB];
 #pragma unroll
-  for (int r = 0; r < MLP_RB; ++r) a[r] = b1[j];
+  for (int r = 0; r < RB; ++r) a[r] = b1[j];
   for (int d = 0; d < D; ++d) {
     const float w = W1T[(long)d * D + j];
 #pragma unroll
-    for (int r = 0; r < MLP_RB; ++r) a[r] += w * xs[r * D + d];
+    for (int r = 0; r < RB; ++r) a[r] += w * xs[r * D + d];
   }
   __syncthreads();
   for (int r = 0; r < nr; ++r) {
@@ -947,11 +948,11 @@ __global__ void mlp3_fwd_kernel(
   }
   __syncthreads();
 #pragma unroll
-  for (int r = 0; r < MLP_RB; ++r) a[r] = b2[j];
+  for (int r = 0; r < RB; ++r) a[r] = b2[j];
   for (int d = 0; d < D; ++d) {
     const float w = W2T[(long)d * D + j];
 #pragma unroll
-    for (int r = 0; r < MLP_RB; ++r) a[r] += w * xs[r * D + d];
+    for (int r = 0; r < RB; ++r) a[r] += w * xs[r * D + d];
   }
   const float w3 = W3[j];
   for (int r = 0; r < nr; ++r) {
@@ -1110,11 +1111,14 @@ void launch_mlp3_fwd(const __hip_bfloat16* x, const float* W1T, const float* b1,
                      const float* W2T, const float* b2, const float* W3,
                      const float* b3, float* h1, float* h2, float* logits,
                      int B, int D, hipStream_t stream) {
-  const int blocks = (B + MLP_RB - 1) / MLP_RB;
-  const size_t lds = (MLP_RB * D + 8) * sizeof(float);
+  // RB=2 doubles the block count (B=257 -> 129): in a captured stream the
+  // kernel runs alone, so grid width beats per-thread load reuse
+  constexpr int RB = 2;
+  const int blocks = (B + RB - 1) / RB;
+  const size_t lds = (RB * D + 8) * sizeof(float);
   if (B > 0)
-    hipLaunchKernelGGL(mlp3_fwd_kernel, dim3(blocks), dim3(D), lds, stream, x,
-                       W1T, b1, W2T, b2, W3, b3, h1, h2, logits, B, D);
+    hipLaunchKernelGGL(mlp3_fwd_kernel<RB>, dim3(blocks), dim3(D), lds, stream,
+                       x, W1T, b1, W2T, b2, W3, b3, h1, h2, logits, B, D);
 }
 
 void launch_mlp3_bwd(const float* dlogits, const float* h1, const float* h2,

@@ -75,19 +75,33 @@ def t5_relative_position_bucket(relative_position, bidirectional, num_buckets, m
 
 class _RelBias(torch.autograd.Function):
     """Relative-position bias gather with a static-structure backward:
-    fwd (1, H, Lq, Lk) = weight.T[:, buckets]; bwd dW = onehot @ dBias^T."""
+    fwd (1, H, Lq, Lk) = weight.T[:, buckets]. Backward on GPU is the
+    bucket scatter-reduce kernel (relbias_wgrad — the onehot-matmul form
+    hit hipBLASLt's tall-skinny pathology, 381 us); CPU keeps the onehot
+    matmul."""
 
     @staticmethod
     def forward(ctx, weight, buckets, onehot):
-        ctx.save_for_backward(onehot)
+        ctx.save_for_backward(onehot, buckets)
         ctx.w_dtype = weight.dtype
+        ctx.nb = weight.shape[0]
         with torch.no_grad():
             return weight.t()[:, buckets].unsqueeze(0).float().contiguous()
 
     @staticmethod
     def backward(ctx, grad):
-        (onehot,) = ctx.saved_tensors
+        onehot, buckets = ctx.saved_tensors
         H = grad.shape[1]
+        if grad.is_cuda:
+            from ..ops import load_ext
+
+            ext = load_ext()
+            if ext is not None:
+                bk32 = buckets.reshape(-1).to(torch.int32).contiguous()
+                dw = ext.relbias_wgrad(
+                    grad.reshape(H, -1).float().contiguous(), bk32, ctx.nb
+                )
+                return dw.to(ctx.w_dtype), None, None
         g = grad.reshape(H, -1).t().float()  # (Lq*Lk, H)
         return (onehot @ g).to(ctx.w_dtype), None, None
 
