@@ -80,7 +80,7 @@ void launch_flash_dq(const __hip_bfloat16*, const __hip_bfloat16*, const __hip_b
 void launch_wgrad2(const __hip_bfloat16*, const __hip_bfloat16*, float*, float*,
                    int, int, int, hipStream_t);
 void launch_adamw_fused(float*, const float*, float*, float*, long, float, float, float, float,
-                        float, const float*, const float*, int, hipStream_t);
+                        float, const float*, const float*, int, void*, hipStream_t);
 void launch_gemm2(const __hip_bfloat16*, const __hip_bfloat16*, const float*,
                   const __hip_bfloat16*, __hip_bfloat16*, int, int, int, hipStream_t);
 void launch_flash_dkv(const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*,
@@ -1000,7 +1000,8 @@ std::vector<at::Tensor> mlp3_bwd(at::Tensor dlogits, at::Tensor x, at::Tensor h1
 
 void adamw_fused(at::Tensor p, at::Tensor g, at::Tensor m, at::Tensor v, double lr,
                  double beta1, double beta2, double eps, double weight_decay, at::Tensor step,
-                 c10::optional<at::Tensor> gclip, bool l2_mode) {
+                 c10::optional<at::Tensor> gclip, bool l2_mode,
+                 c10::optional<at::Tensor> p16) {
   CHECK_GPU(p);
   TORCH_CHECK(p.scalar_type() == at::kFloat && g.scalar_type() == at::kFloat);
   TORCH_CHECK(p.numel() == g.numel() && p.numel() == m.numel() && p.numel() == v.numel());
@@ -1011,10 +1012,17 @@ void adamw_fused(at::Tensor p, at::Tensor g, at::Tensor m, at::Tensor v, double 
     TORCH_CHECK(gclip->is_cuda() && gclip->scalar_type() == at::kFloat);
     gc = gclip->data_ptr<float>();
   }
+  void* s16 = nullptr;
+  if (p16.has_value()) {
+    TORCH_CHECK(p16->is_cuda() && p16->scalar_type() == at::kBFloat16 &&
+                p16->numel() == p.numel() && p16->is_contiguous(),
+                "p16 must be a contiguous bf16 shadow of p");
+    s16 = p16->data_ptr();
+  }
   launch_adamw_fused(p.data_ptr<float>(), g.data_ptr<float>(), m.data_ptr<float>(),
                      v.data_ptr<float>(), p.numel(), (float)lr, (float)beta1, (float)beta2,
                      (float)eps, (float)weight_decay, step.data_ptr<float>(), gc,
-                     l2_mode ? 1 : 0, cur_stream());
+                     l2_mode ? 1 : 0, s16, cur_stream());
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -1219,7 +1227,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     return t;
   });
   m.def("gemm2", &gemm2);
-  m.def("adamw_fused", &adamw_fused);
+  m.def("adamw_fused", &adamw_fused, pybind11::arg("p"), pybind11::arg("g"),
+        pybind11::arg("m"), pybind11::arg("v"), pybind11::arg("lr"),
+        pybind11::arg("beta1"), pybind11::arg("beta2"), pybind11::arg("eps"),
+        pybind11::arg("weight_decay"), pybind11::arg("step"),
+        pybind11::arg("gclip") = pybind11::none(),
+        pybind11::arg("l2_mode") = false,
+        pybind11::arg("p16") = pybind11::none());
   m.def("softmax_mask_bwd", &softmax_mask_bwd);
   m.def("ggnn_fused_bwd", &ggnn_fused_bwd);
 }

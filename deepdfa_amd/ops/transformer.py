@@ -83,6 +83,31 @@ def bump_weights_epoch():
     _weights_epoch[0] += 1
 
 
+def _shadow_w16(weight):
+    """The optimizer-maintained flat bf16 shadow view of `weight`
+    (parallel/optim.py FlatAdamW), re-synced if a torch-level write
+    (checkpoint load, broadcast) moved the parameter's version counter
+    since the last sync. None when no flat optimizer owns the weight."""
+    w16 = getattr(weight, "_dfa_w16", None)
+    if w16 is None:
+        return None
+    if weight._version != weight._dfa_w16_ver:
+        w16.copy_(weight.detach())
+        weight._dfa_w16_ver = weight._version
+    return w16
+
+
+def _bias_f32(bias):
+    """Master-fp32 biases are used in place (no copy); anything else is
+    cast-copied (rare: only models trained without the flat optimizer)."""
+    if bias is None:
+        return None
+    b = bias.detach()
+    if b.dtype == torch.float32 and b.is_contiguous():
+        return b
+    return b.float().contiguous()
+
+
 _seed_state = {"base": None, "ctr": 0}
 
 
@@ -155,37 +180,64 @@ class _LinearBf16(torch.autograd.Function):
     def forward(ctx, x, weight, bias):
         ext = load_ext(required=True)
         x2d = x.reshape(-1, x.shape[-1]).contiguous()
-        # cache the bf16 / transposed weight per parameter VERSION: the cast
-        # and transpose copies otherwise dominate the elementwise kernel
-        # count (~3 kernels x 72 linears per step)
-        cache = getattr(weight, "_dfa_cast_cache", None)
-        key = (weight._version, _weights_epoch[0])
-        if cache is None:
-            w16 = weight.detach().to(torch.bfloat16).contiguous()
-            wt16 = w16.t().contiguous()
-            b32 = bias.detach().float().contiguous() if bias is not None else None
-            weight._dfa_cast_cache = (key, w16, wt16, b32)
-        elif CAPTURE_REFRESH[0] or cache[0] != key:
-            # refresh INTO the cached buffers: under hipGraph capture the
-            # cast must be part of the graph (stable addresses, re-run each
-            # replay); in eager it re-casts only when the key moved
-            _, w16, wt16, b32 = cache
-            w16.copy_(weight.detach())
-            wt16.copy_(w16.t())
-            if bias is not None:
-                if b32 is None:
-                    b32 = bias.detach().float().contiguous()
+        # fat shapes route to the library GEMMs and never read wt16 (backward
+        # uses w16 directly), so the transpose copy is skipped for them
+        fat = weight.shape[0] > 1024 or weight.shape[1] > 1024
+        w16 = _shadow_w16(weight)
+        if w16 is not None:
+            # optimizer-maintained bf16 shadow (parallel/optim.py): no cast
+            # kernel — the adamw kernel wrote this view at the last step.
+            # Only the transposed copy (square shapes) needs per-step refresh.
+            b32 = _bias_f32(bias)
+            if fat:
+                wt16 = None
+            else:
+                cache = getattr(weight, "_dfa_wt_cache", None)
+                key = (weight._version, _weights_epoch[0])
+                if cache is None:
+                    wt16 = w16.t().contiguous()
+                    weight._dfa_wt_cache = (key, wt16)
+                elif CAPTURE_REFRESH[0] or cache[0] != key:
+                    _, wt16 = cache
+                    wt16.copy_(w16.t())
+                    weight._dfa_wt_cache = (key, wt16)
                 else:
-                    b32.copy_(bias.detach())
-            weight._dfa_cast_cache = (key, w16, wt16, b32)
+                    _, wt16 = cache
         else:
-            _, w16, wt16, b32 = cache
-            if bias is not None and b32 is None:
-                b32 = bias.detach().float().contiguous()
+            # no flat optimizer: cache the bf16 / transposed weight per
+            # parameter VERSION — the cast and transpose copies otherwise
+            # dominate the elementwise kernel count (~3 x 72 linears/step)
+            cache = getattr(weight, "_dfa_cast_cache", None)
+            key = (weight._version, _weights_epoch[0])
+            if cache is None:
+                w16 = weight.detach().to(torch.bfloat16).contiguous()
+                wt16 = None if fat else w16.t().contiguous()
+                b32 = bias.detach().float().contiguous() if bias is not None else None
+                weight._dfa_cast_cache = (key, w16, wt16, b32)
+            elif CAPTURE_REFRESH[0] or cache[0] != key:
+                # refresh INTO the cached buffers: under hipGraph capture the
+                # cast must be part of the graph (stable addresses, re-run
+                # each replay); in eager it re-casts only when the key moved
+                _, w16, wt16, b32 = cache
+                w16.copy_(weight.detach())
+                if wt16 is not None:
+                    wt16.copy_(w16.t())
+                if bias is not None:
+                    if b32 is None:
+                        b32 = bias.detach().float().contiguous()
+                    else:
+                        b32.copy_(bias.detach())
+                weight._dfa_cast_cache = (key, w16, wt16, b32)
+            else:
+                _, w16, wt16, b32 = cache
+                if bias is not None and b32 is None:
+                    b32 = bias.detach().float().contiguous()
+        if bias is None:
+            b32 = None
         # shape dispatch (measured, profiles/optimization_r2.md): our gemm2
         # beats hipBLASLt on the square 768-ish shapes; the library wins the
         # fat no-bias ones (K or COL >= ~2k)
-        if b32 is None and (w16.shape[0] > 1024 or w16.shape[1] > 1024):
+        if b32 is None and fat:
             out = torch.matmul(x2d, w16.t())
         else:
             out = ext.gemm2(x2d, w16, b32, None)
@@ -394,32 +446,52 @@ class _QKVLinear(torch.autograd.Function):
     def forward(ctx, x, wq, wk, wv, bq, bk, bv):
         ext = load_ext(required=True)
         x2d = x.reshape(-1, x.shape[-1]).contiguous()
-        cache = getattr(wq, "_dfa_qkv_cache", None)
-        key = (wq._version, wk._version, wv._version, _weights_epoch[0])
         has_bias = bq is not None
-        if cache is None:
-            wcat = torch.cat([wq.detach(), wk.detach(), wv.detach()])
-            w16 = wcat.to(torch.bfloat16).contiguous()
-            wt16 = w16.t().contiguous()
-            b32 = (torch.cat([bq.detach(), bk.detach(), bv.detach()]).float().contiguous()
-                   if has_bias else None)
-            wq._dfa_qkv_cache = (key, w16, wt16, b32)
-        elif CAPTURE_REFRESH[0] or cache[0] != key:
-            _, w16, wt16, b32 = cache
-            D = wq.shape[0]
-            w16[:D].copy_(wq.detach())
-            w16[D:2 * D].copy_(wk.detach())
-            w16[2 * D:].copy_(wv.detach())
-            wt16.copy_(w16.t())
+        # bias-free q/k/v defined back-to-back (T5) sit ADJACENT in the flat
+        # optimizer's bf16 shadow: the packed (3D, K) weight is a zero-copy
+        # view of the shadow buffer — no per-step packing or cast kernels
+        sq, sk, sv = _shadow_w16(wq), _shadow_w16(wk), _shadow_w16(wv)
+        if (
+            sq is not None and sk is not None and sv is not None
+            and sk.data_ptr() == sq.data_ptr() + 2 * sq.numel()
+            and sv.data_ptr() == sk.data_ptr() + 2 * sk.numel()
+        ):
+            base, off = wq._dfa_w16_base, wq._dfa_w16_off
+            n3 = sq.numel() + sk.numel() + sv.numel()
+            w16 = base[off : off + n3].view(3 * wq.shape[0], wq.shape[1])
+            b32 = None
             if has_bias:
-                b32[:D].copy_(bq.detach())
-                b32[D:2 * D].copy_(bk.detach())
-                b32[2 * D:].copy_(bv.detach())
-            wq._dfa_qkv_cache = (key, w16, wt16, b32)
+                cache = getattr(wq, "_dfa_qkv_cache", None)
+                key = (bq._version, bk._version, bv._version, _weights_epoch[0])
+                if cache is None or CAPTURE_REFRESH[0] or cache[0] != key:
+                    b32 = torch.cat([bq.detach(), bk.detach(), bv.detach()]).float()
+                    wq._dfa_qkv_cache = (key, b32)
+                else:
+                    b32 = cache[1]
         else:
-            _, w16, wt16, b32 = cache
+            cache = getattr(wq, "_dfa_qkv_cache", None)
+            key = (wq._version, wk._version, wv._version, _weights_epoch[0])
+            if cache is None:
+                wcat = torch.cat([wq.detach(), wk.detach(), wv.detach()])
+                w16 = wcat.to(torch.bfloat16).contiguous()
+                b32 = (torch.cat([bq.detach(), bk.detach(), bv.detach()]).float().contiguous()
+                       if has_bias else None)
+                wq._dfa_qkv_cache = (key, w16, b32)
+            elif CAPTURE_REFRESH[0] or cache[0] != key:
+                _, w16, b32 = cache
+                D = wq.shape[0]
+                w16[:D].copy_(wq.detach())
+                w16[D:2 * D].copy_(wk.detach())
+                w16[2 * D:].copy_(wv.detach())
+                if has_bias:
+                    b32[:D].copy_(bq.detach())
+                    b32[D:2 * D].copy_(bk.detach())
+                    b32[2 * D:].copy_(bv.detach())
+                wq._dfa_qkv_cache = (key, w16, b32)
+            else:
+                _, w16, b32 = cache
         out = ext.gemm2(x2d, w16, b32, None)
-        ctx.save_for_backward(x2d, w16, wt16)
+        ctx.save_for_backward(x2d, w16)
         ctx.has_bias = has_bias
         ctx.x_shape = x.shape
         return out.view(*x.shape[:-1], w16.shape[0])
@@ -427,7 +499,7 @@ class _QKVLinear(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         ext = load_ext(required=True)
-        x2d, w16, wt16 = ctx.saved_tensors
+        x2d, w16 = ctx.saved_tensors
         dy2d = dy.reshape(-1, dy.shape[-1]).contiguous()
         dx = torch.matmul(dy2d, w16)  # (N, 3D) @ (3D, D): fat K, library wins
         dw = ext.wgrad(dy2d, x2d)  # (3D, K) fp32
@@ -545,11 +617,15 @@ class _LMHeadCE(torch.autograd.Function):
         key = (weight._version, _weights_epoch[0])
         if cache is None:
             Wp = torch.zeros(Vp, K, dtype=torch.bfloat16, device=weight.device)
-            Wp[:V].copy_(weight.detach())
+            src = _shadow_w16(weight)
+            Wp[:V].copy_(src if src is not None else weight.detach())
             weight._dfa_vocab_cache = (key, Wp)
         elif CAPTURE_REFRESH[0] or cache[0] != key:
             _, Wp = cache
-            Wp[:V].copy_(weight.detach())
+            # refresh from the optimizer's bf16 shadow when present: a
+            # bf16->bf16 copy is 2/3 the traffic of the fp32 cast at V=32100
+            src = _shadow_w16(weight)
+            Wp[:V].copy_(src if src is not None else weight.detach())
             weight._dfa_vocab_cache = (key, Wp)
         else:
             _, Wp = cache

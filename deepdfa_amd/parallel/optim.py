@@ -43,6 +43,23 @@ class FlatAdamW:
             p.data = self.flat_p[off : off + n].view(p.shape)
             p.grad = self.flat_g[off : off + n].view(p.shape)
             off += n
+        # bf16 shadow of the flat master params, written by the adamw kernel
+        # itself each step (+2 B/param on a 32 B/param pass): the fused-linear
+        # weight views (ops/transformer.py) read from it, replacing per-weight
+        # fp32->bf16 cast kernels every step. `_dfa_w16_ver` records the param
+        # version the shadow was last synced at so torch-level writes
+        # (checkpoint loads) are detected and re-cast by the consumer.
+        self.flat_p16 = None
+        if device.type == "cuda":
+            self.flat_p16 = self.flat_p.to(torch.bfloat16)
+            off = 0
+            for p in self.params:
+                n = p.numel()
+                p._dfa_w16 = self.flat_p16[off : off + n].view(p.shape)
+                p._dfa_w16_ver = p._version
+                p._dfa_w16_base = self.flat_p16
+                p._dfa_w16_off = off
+                off += n
         self.lr = lr
         self.betas = betas
         self.eps = eps
@@ -99,6 +116,7 @@ class FlatAdamW:
             ext.adamw_fused(
                 self.flat_p, self.flat_g, self.m, self.v, lr, b1, b2, self.eps,
                 self.weight_decay, self._step_t, gclip, self.l2_mode,
+                self.flat_p16,
             )
             self._clip_coef = None
             # the raw kernel write bypasses torch version counters —
