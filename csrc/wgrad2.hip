@@ -213,7 +213,7 @@ typedef __attribute__((address_space(3))) bf16x4v lds_bf16x4;
 }
 
 void launch_wgrad2(const bf16* A, const bf16* B, float* out, float* csum,
-                   int K, int M, int C, hipStream_t stream) {
+                   int K, int M, int C, int accumulate, hipStream_t stream) {
   const int tiles = (M / WBM) * (C / WBC);
   int zsplit = max(1, 512 / tiles);
   int kchunk = (K + zsplit - 1) / zsplit;
@@ -221,6 +221,9 @@ void launch_wgrad2(const bf16* A, const bf16* B, float* out, float* csum,
   if (kchunk < 512) kchunk = min(((K + WBK - 1) / WBK) * WBK, 512);
   zsplit = (K + kchunk - 1) / kchunk;
   const dim3 grid(M / WBM, C / WBC, zsplit);
+  // accumulate: out is a live accumulator (the flat .grad view) — force the
+  // atomic epilogue even when a single K chunk would have plain-stored
+  const int zarg = accumulate ? max(zsplit, 2) : zsplit;
   hipLaunchKernelGGL(wgrad2_kernel, grid, dim3(256), 73728, stream, A, B, out,
-                     csum, K, M, C, kchunk, zsplit);
+                     csum, K, M, C, kchunk, zarg);
 }

@@ -244,6 +244,7 @@ class _LinearBf16(torch.autograd.Function):
         ctx.save_for_backward(x2d, w16, wt16)
         ctx.has_bias = bias is not None
         ctx.x_shape = x.shape
+        ctx.wp, ctx.bp = weight, bias
         return out.view(*x.shape[:-1], w16.shape[0])
 
     @staticmethod
@@ -255,8 +256,21 @@ class _LinearBf16(torch.autograd.Function):
             dx = torch.matmul(dy2d, w16)  # dY @ W, library wins fat shapes
         else:
             dx = ext.gemm2(dy2d, wt16, None, None)
-        dw = ext.wgrad(dy2d, x2d)  # (COL, K) fp32
-        db = ext.colsum(dy2d) if ctx.has_bias else None
+        # direct accumulation (parallel/optim.py): under the flat optimizer
+        # the kernels atomically add into the pre-zeroed flat .grad views and
+        # autograd sees None — no zeros() fill, no AccumulateGrad add
+        dw = db = None
+        wp, bp = ctx.wp, ctx.bp
+        if ctx.needs_input_grad[1]:
+            if getattr(wp, "_dfa_w16", None) is not None and wp.grad is not None:
+                ext.wgrad(dy2d, x2d, out=wp.grad)
+            else:
+                dw = ext.wgrad(dy2d, x2d)  # (COL, K) fp32
+        if ctx.has_bias and ctx.needs_input_grad[2]:
+            if getattr(bp, "_dfa_w16", None) is not None and bp.grad is not None:
+                ext.colsum(dy2d, out=bp.grad)
+            else:
+                db = ext.colsum(dy2d)
         return dx.view(ctx.x_shape), dw, db
 
 
@@ -336,6 +350,7 @@ class _RMSNorm(torch.autograd.Function):
         x = x.contiguous()
         y, rstd = ext.rmsnorm_fwd(x, wf, eps)
         ctx.save_for_backward(x, wf, rstd)
+        ctx.wp = weight
         return y
 
     @staticmethod
@@ -344,6 +359,12 @@ class _RMSNorm(torch.autograd.Function):
         x, wf, rstd = ctx.saved_tensors
         dy = dy.contiguous()
         dx = ext.rmsnorm_bwd(dy, x, wf, rstd)
+        wp = ctx.wp
+        if (getattr(wp, "_dfa_w16", None) is not None and wp.grad is not None
+                and ctx.needs_input_grad[1]):
+            # direct accumulation into the flat .grad view (parallel/optim.py)
+            ext.rmsnorm_wgrad(dy, x, rstd, out=wp.grad)
+            return dx, None, None
         dgamma = ext.rmsnorm_wgrad(dy, x, rstd)
         return dx, dgamma, None
 
@@ -374,12 +395,21 @@ class _EmbeddingLookup(torch.autograd.Function):
         ctx.num_rows = weight.shape[0]
         ctx.w_dtype = weight.dtype
         ctx.padding_idx = padding_idx if padding_idx is not None else -1
+        ctx.wp = weight
         return torch.nn.functional.embedding(indices, weight, padding_idx)
 
     @staticmethod
     def backward(ctx, dy):
         ext = load_ext(required=True)
         (indices,) = ctx.saved_tensors
+        wp = ctx.wp
+        if (getattr(wp, "_dfa_w16", None) is not None and wp.grad is not None
+                and ctx.w_dtype == torch.float32):
+            # scatter straight into the flat fp32 .grad view (pre-zeroed):
+            # skips a (V, D) zeros + full-table AccumulateGrad add
+            ext.embed_scatter(dy.contiguous(), indices.reshape(-1).contiguous(),
+                              ctx.num_rows, ctx.padding_idx, out=wp.grad)
+            return None, None, None
         dw = ext.embed_scatter(dy.contiguous(), indices.reshape(-1).contiguous(),
                                ctx.num_rows, ctx.padding_idx)
         return None, dw.to(ctx.w_dtype), None
@@ -494,6 +524,7 @@ class _QKVLinear(torch.autograd.Function):
         ctx.save_for_backward(x2d, w16)
         ctx.has_bias = has_bias
         ctx.x_shape = x.shape
+        ctx.wparams = (wq, wk, wv)
         return out.view(*x.shape[:-1], w16.shape[0])
 
     @staticmethod
@@ -502,8 +533,23 @@ class _QKVLinear(torch.autograd.Function):
         x2d, w16 = ctx.saved_tensors
         dy2d = dy.reshape(-1, dy.shape[-1]).contiguous()
         dx = torch.matmul(dy2d, w16)  # (N, 3D) @ (3D, D): fat K, library wins
-        dw = ext.wgrad(dy2d, x2d)  # (3D, K) fp32
         D = w16.shape[0] // 3
+        wq, wk, wv = ctx.wparams
+        # q/k/v grads adjacent in the flat grad buffer (T5, bias-free):
+        # ONE wgrad accumulates the packed (3D, K) grad straight into the
+        # flat .grad region — no zeros, no slicing, no AccumulateGrad adds
+        if (
+            not ctx.has_bias
+            and getattr(wq, "_dfa_w16", None) is not None and wq.grad is not None
+            and wk.grad is not None and wv.grad is not None
+            and wk.grad.data_ptr() == wq.grad.data_ptr() + 4 * wq.numel()
+            and wv.grad.data_ptr() == wk.grad.data_ptr() + 4 * wk.numel()
+        ):
+            n3 = wq.numel() + wk.numel() + wv.numel()
+            g3 = wq._dfa_gbase[wq._dfa_goff : wq._dfa_goff + n3].view(3 * D, -1)
+            ext.wgrad(dy2d, x2d, out=g3)
+            return (dx.view(ctx.x_shape), None, None, None, None, None, None)
+        dw = ext.wgrad(dy2d, x2d)  # (3D, K) fp32
         if ctx.has_bias:
             db = ext.colsum(dy2d)
             return (dx.view(ctx.x_shape), dw[:D], dw[D:2 * D], dw[2 * D:],

@@ -59,6 +59,12 @@ class FlatAdamW:
                 p._dfa_w16_ver = p._version
                 p._dfa_w16_base = self.flat_p16
                 p._dfa_w16_off = off
+                # flat-grad view bookkeeping for DIRECT gradient accumulation:
+                # weight-grad kernels atomically add into .grad (pre-zeroed by
+                # zero_grad's single fill) and return None to autograd,
+                # skipping the per-param zeros() + AccumulateGrad add pair
+                p._dfa_gbase = self.flat_g
+                p._dfa_goff = off
                 off += n
         self.lr = lr
         self.betas = betas

@@ -131,3 +131,60 @@ def test_trainer_fit_graph_capture_parity(tmp_path):
                r_c["train_tp"] + r_c["train_fp"] + r_c["train_tn"] + r_c["train_fn"]
         assert abs(r_e["train_f1"] - r_c["train_f1"]) < 0.15, (r_e, r_c)
     assert abs(results["eager"][-1]["val_loss"] - results["capture"][-1]["val_loss"]) < 0.05
+
+
+@pytest.mark.gpu
+def test_direct_grad_accumulation_matches_autograd():
+    """With FlatAdamW attached, weight-grad kernels accumulate DIRECTLY into
+    the flat .grad views (ops/transformer.py backward direct paths) instead
+    of returning tensors for AccumulateGrad. Same module, same input, grads
+    must match the plain-autograd route across every op family: square
+    linear (wgrad2 + colsum), fat linear, adjacent q/k/v (packed wgrad),
+    rms_norm (atomic dgamma), embedding (scatter-add)."""
+    import torch
+    from torch import nn
+    from deepdfa_amd.ops.transformer import (embedding_lookup, fused_linear,
+                                             fused_qkv, rms_norm)
+    from deepdfa_amd.parallel.optim import FlatAdamW
+
+    torch.manual_seed(0)
+    dev = "cuda"
+
+    class Tiny(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.emb = nn.Embedding(512, 768)
+            self.gamma = nn.Parameter(torch.ones(768))
+            self.wq = nn.Linear(768, 768, bias=False)
+            self.wk = nn.Linear(768, 768, bias=False)
+            self.wv = nn.Linear(768, 768, bias=False)
+            self.o = nn.Linear(768, 768)
+            self.wi = nn.Linear(768, 3072, bias=False)
+
+        def forward(self, ids):
+            e = embedding_lookup(ids, self.emb.weight)
+            h = rms_norm(e, self.gamma).to(torch.bfloat16)
+            qkv = fused_qkv(h, self.wq.weight, self.wk.weight, self.wv.weight)
+            y = fused_linear(h, self.o.weight, self.o.bias)
+            z = fused_linear(y, self.wi.weight, None)
+            return qkv.float().square().mean() + z.float().square().mean()
+
+    model = Tiny().to(dev)
+    ids = torch.randint(0, 512, (4, 64), device=dev)
+
+    loss_ref = model(ids)
+    loss_ref.backward()
+    ref = {n: p.grad.detach().clone() for n, p in model.named_parameters()}
+    for p in model.parameters():
+        p.grad = None
+
+    opt = FlatAdamW(model.parameters(), lr=1e-3)
+    opt.zero_grad()
+    loss = model(ids)
+    loss.backward()
+    assert torch.allclose(loss.float(), loss_ref.float(), atol=1e-5)
+    for n, p in model.named_parameters():
+        r = ref[n].float()
+        tol = 1e-4 + 1e-4 * r.abs().max()
+        assert torch.allclose(p.grad.float(), r, atol=float(tol)), (
+            n, (p.grad.float() - r).abs().max())
