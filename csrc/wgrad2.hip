@@ -15,6 +15,7 @@
 
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
+#include <cstdlib>
 
 #define WAVE 64
 #define WBM 128  // out rows (M) per block
@@ -215,10 +216,16 @@ typedef __attribute__((address_space(3))) bf16x4v lds_bf16x4;
 void launch_wgrad2(const bf16* A, const bf16* B, float* out, float* csum,
                    int K, int M, int C, int accumulate, hipStream_t stream) {
   const int tiles = (M / WBM) * (C / WBC);
+  // K-chunk floor trades chip fill (more z-chunks) against fp32 atomic
+  // output traffic (each chunk adds one full M x C atomic pass); 768 won
+  // the sweep at the square 768 shapes (27.2 -> 23.7 us at K=4096) and at
+  // K=8192, fat shapes indifferent (tools/wg2_probe.py)
+  int kfloor = 768;
+  if (const char* e = getenv("DFA_WG2_KFLOOR")) kfloor = atoi(e);
   int zsplit = max(1, 512 / tiles);
   int kchunk = (K + zsplit - 1) / zsplit;
   kchunk = ((kchunk + WBK - 1) / WBK) * WBK;
-  if (kchunk < 512) kchunk = min(((K + WBK - 1) / WBK) * WBK, 512);
+  if (kchunk < kfloor) kchunk = min(((K + WBK - 1) / WBK) * WBK, kfloor);
   zsplit = (K + kchunk - 1) / kchunk;
   const dim3 grid(M / WBM, C / WBC, zsplit);
   // accumulate: out is a live accumulator (the flat .grad view) — force the
