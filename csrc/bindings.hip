@@ -780,7 +780,8 @@ std::vector<at::Tensor> flash_attn_bwd(at::Tensor dO, at::Tensor Q, at::Tensor K
                                        c10::optional<at::Tensor> bias, double scale, bool causal,
                                        double dropout_p, int64_t seed, bool need_dbias,
                                        bool fused_grads,
-                                       c10::optional<at::Tensor> dbias_accum = c10::nullopt) {
+                                       c10::optional<at::Tensor> dbias_accum = c10::nullopt,
+                                       bool kv_fused = false) {
   CHECK_GPU(dO);
   const int B = Q.size(0), L = Q.size(1);
   const long ldq = fa_ld(Q, L);
@@ -793,16 +794,26 @@ std::vector<at::Tensor> flash_attn_bwd(at::Tensor dO, at::Tensor Q, at::Tensor K
   // kernels write directly (ld 3*H*64) — lets the attention op consume a
   // fused QKV projection with zero slice-backward scatter work
   const long HD64 = (long)H * 64;
-  at::Tensor dQKV, dQ, dK, dV;
-  long ldout = HD64;
+  at::Tensor dQKV, dKV, dQ, dK, dV;
+  long ld_dq = HD64, ld_dkv = HD64;
   if (fused_grads) {
     TORCH_CHECK(ldq == 3 * HD64 && ldkv == 3 * HD64,
                 "fused_grads expects q/k/v slices of one (B, L, 3D) buffer");
     dQKV = at::empty({B, L, 3 * HD64}, Q.options());
-    ldout = 3 * HD64;
+    ld_dq = ld_dkv = 3 * HD64;
     dQ = dQKV.narrow(2, 0, HD64);
     dK = dQKV.narrow(2, HD64, HD64);
     dV = dQKV.narrow(2, 2 * HD64, HD64);
+  } else if (kv_fused) {
+    // cross-attention: dense dQ, dK/dV written as the [k|v] slices of one
+    // (B, L, 2D) buffer matching a fused K/V projection of the encoder
+    TORCH_CHECK(ldkv == 2 * HD64,
+                "kv_fused expects k/v slices of one (B, L, 2D) buffer");
+    dQ = at::empty({B, L, HD64}, Q.options());
+    dKV = at::empty({B, L, 2 * HD64}, Q.options());
+    ld_dkv = 2 * HD64;
+    dK = dKV.narrow(2, 0, HD64);
+    dV = dKV.narrow(2, HD64, HD64);
   } else {
     dQ = at::empty({B, L, HD64}, Q.options());
     dK = at::empty({B, L, HD64}, Q.options());
@@ -827,16 +838,20 @@ std::vector<at::Tensor> flash_attn_bwd(at::Tensor dO, at::Tensor Q, at::Tensor K
   launch_flash_dq(ptr<bf16_t>(Q), ptr<bf16_t>(K), ptr<bf16_t>(V), ptr<bf16_t>(dO),
                   opt_valid_ptr(valid), bptr, lse.data_ptr<float>(), Dterm.data_ptr<float>(),
                   mptr<bf16_t>(dQ), dbias_ptr, B, H, L, (float)scale, causal ? 1 : 0,
-                  (unsigned)(dropout_p * 256.0), (unsigned long long)seed, ldq, ldkv, ldout,
+                  (unsigned)(dropout_p * 256.0), (unsigned long long)seed, ldq, ldkv, ld_dq,
                   stream);
   launch_flash_dkv(ptr<bf16_t>(Q), ptr<bf16_t>(K), ptr<bf16_t>(V), ptr<bf16_t>(dO),
                    opt_valid_ptr(valid), bptr, lse.data_ptr<float>(), Dterm.data_ptr<float>(),
                    mptr<bf16_t>(dK), mptr<bf16_t>(dV), B, H, L, (float)scale,
                    causal ? 1 : 0, (unsigned)(dropout_p * 256.0), (unsigned long long)seed,
-                   ldq, ldkv, ldout, stream);
+                   ldq, ldkv, ld_dkv, stream);
   if (fused_grads) {
     if (need_dbias && !dbias_accum.has_value()) return {dQKV, dBias};
     return {dQKV};
+  }
+  if (kv_fused) {
+    if (need_dbias && !dbias_accum.has_value()) return {dQ, dKV, dBias};
+    return {dQ, dKV};
   }
   if (need_dbias && !dbias_accum.has_value()) return {dQ, dK, dV, dBias};
   return {dQ, dK, dV};
@@ -1116,7 +1131,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         pybind11::arg("H"), pybind11::arg("valid"), pybind11::arg("bias"),
         pybind11::arg("scale"), pybind11::arg("causal"), pybind11::arg("dropout_p"),
         pybind11::arg("seed"), pybind11::arg("need_dbias"), pybind11::arg("fused_grads"),
-        pybind11::arg("dbias_accum") = pybind11::none());
+        pybind11::arg("dbias_accum") = pybind11::none(),
+        pybind11::arg("kv_fused") = false);
   m.def("lmhead_ce_fwd", &lmhead_ce_fwd);
   m.def("lmhead_ce_bwd", &lmhead_ce_bwd);
   m.def("relbias_wgrad", [](at::Tensor dbias, at::Tensor buckets, int64_t nb) {

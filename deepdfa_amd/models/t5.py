@@ -211,6 +211,18 @@ class T5Attention(nn.Module):
                                       bias_accum=bias_accum)
             return fused_linear(out, self.o.weight)
         qp = fused_linear(x, self.q.weight)
+        if (kv is not None and d == 64 and position_bias is None
+                and flash_usable(qp, Lq, Lk)):
+            # cross-attention: ONE fused K/V projection of the encoder
+            # states (k/v weights adjacent in the flat optimizer buffer)
+            # + flash with strided k/v slices and a packed dKV backward
+            from ..ops.transformer import flash_attention_kv, fused_kv
+
+            kvp = fused_kv(src, self.k.weight, self.v.weight)
+            if kvp is not None:
+                out = flash_attention_kv(qp, kvp, H, valid=valid, scale=1.0,
+                                         dropout_p=dropout_p)
+                return fused_linear(out, self.o.weight)
         kp = fused_linear(src, self.k.weight)
         vp = fused_linear(src, self.v.weight)
         if d == 64 and flash_usable(qp, Lq, Lk):

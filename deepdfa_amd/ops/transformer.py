@@ -566,6 +566,109 @@ def fused_qkv(x, wq, wk, wv, bq=None, bk=None, bv=None):
     return None
 
 
+class _KVLinear(torch.autograd.Function):
+    """Fused bias-free K/V projection for cross-attention (both read the
+    SAME encoder states): one (N, 2D) gemm2/wgrad per layer instead of two
+    768-column calls. With the flat optimizer the packed (2D, K) weight is a
+    zero-copy view (k/v adjacent), and the packed grad accumulates directly
+    into the flat .grad region (one wgrad, no AccumulateGrad adds)."""
+
+    @staticmethod
+    def forward(ctx, x, wk, wv):
+        ext = load_ext(required=True)
+        x2d = x.reshape(-1, x.shape[-1]).contiguous()
+        sk, sv = _shadow_w16(wk), _shadow_w16(wv)
+        if (
+            sk is not None and sv is not None
+            and sv.data_ptr() == sk.data_ptr() + 2 * sk.numel()
+        ):
+            base, off = wk._dfa_w16_base, wk._dfa_w16_off
+            n2 = sk.numel() + sv.numel()
+            w16 = base[off : off + n2].view(2 * wk.shape[0], wk.shape[1])
+        else:
+            cache = getattr(wk, "_dfa_kv_cache", None)
+            key = (wk._version, wv._version, _weights_epoch[0])
+            if cache is None:
+                w16 = torch.cat([wk.detach(), wv.detach()]).to(torch.bfloat16).contiguous()
+                wk._dfa_kv_cache = (key, w16)
+            elif CAPTURE_REFRESH[0] or cache[0] != key:
+                _, w16 = cache
+                D = wk.shape[0]
+                w16[:D].copy_(wk.detach())
+                w16[D:].copy_(wv.detach())
+                wk._dfa_kv_cache = (key, w16)
+            else:
+                _, w16 = cache
+        out = ext.gemm2(x2d, w16, None, None)
+        ctx.save_for_backward(x2d, w16)
+        ctx.x_shape = x.shape
+        ctx.wparams = (wk, wv)
+        return out.view(*x.shape[:-1], w16.shape[0])
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = load_ext(required=True)
+        x2d, w16 = ctx.saved_tensors
+        dy2d = dy.reshape(-1, dy.shape[-1]).contiguous()
+        dx = torch.matmul(dy2d, w16)  # (N, 2D) @ (2D, D): fat K, library wins
+        D = w16.shape[0] // 2
+        wk, wv = ctx.wparams
+        if (
+            getattr(wk, "_dfa_w16", None) is not None and wk.grad is not None
+            and wv.grad is not None
+            and wv.grad.data_ptr() == wk.grad.data_ptr() + 4 * wk.numel()
+        ):
+            n2 = wk.numel() + wv.numel()
+            g2 = wk._dfa_gbase[wk._dfa_goff : wk._dfa_goff + n2].view(2 * D, -1)
+            ext.wgrad(dy2d, x2d, out=g2)
+            return dx.view(ctx.x_shape), None, None
+        dw = ext.wgrad(dy2d, x2d)
+        return dx.view(ctx.x_shape), dw[:D], dw[D:]
+
+
+def fused_kv(x, wk, wv):
+    """Fused (B, L, 2D) cross-attention K/V projection, or None when the
+    geometry doesn't fit the custom GEMMs."""
+    if linear_usable(x, wk) and wk.shape[0] % 128 == 0:
+        return _KVLinear.apply(x, wk, wv)
+    return None
+
+
+class _FlashAttentionKV(torch.autograd.Function):
+    """Flash attention with a dense Q and the FUSED (B, L, 2D) K/V
+    projection (cross-attention): k/v are strided slices forward, backward
+    writes dk/dv into one (B, L, 2D) buffer (flash_attn_bwd kv_fused)."""
+
+    @staticmethod
+    def forward(ctx, q, kv, H, valid, scale, dropout_p):
+        ext = load_ext(required=True)
+        D = kv.shape[-1] // 2
+        k, v = kv[..., :D], kv[..., D:]
+        seed = _next_seed() if dropout_p > 0 else 0
+        O, lse = ext.flash_attn_fwd(q, k, v, H, valid, None, scale, False,
+                                    dropout_p, seed)
+        ctx.save_for_backward(q, kv, O, lse)
+        ctx.meta = (H, valid, scale, dropout_p, seed)
+        return O
+
+    @staticmethod
+    def backward(ctx, dO):
+        ext = load_ext(required=True)
+        q, kv, O, lse = ctx.saved_tensors
+        H, valid, scale, dropout_p, seed = ctx.meta
+        D = kv.shape[-1] // 2
+        k, v = kv[..., :D], kv[..., D:]
+        dq, dkv = ext.flash_attn_bwd(
+            dO.contiguous(), q, k, v, O, lse, H, valid, None, scale, False,
+            dropout_p, seed, False, False, None, kv_fused=True,
+        )
+        return dq, dkv, None, None, None, None
+
+
+def flash_attention_kv(q, kv, num_heads, valid=None, scale=1.0, dropout_p=0.0):
+    return _FlashAttentionKV.apply(q, kv, num_heads, valid, scale, dropout_p)
+
+
 class _DropoutAdd(torch.autograd.Function):
     """out = res + dropout(h) in one kernel (T5 pre-norm residuals);
     d(res) = dy passes through with no kernel."""
