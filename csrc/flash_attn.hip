@@ -96,10 +96,19 @@ __global__ __launch_bounds__(NW * 64, 12 / NW) void flash_fwd_kernel(
   char* k_lds = smem;                // [TK][64] bf16 swizzled = 8 KiB
   char* v_lds = smem + TK * 128;     // [TK/4][4 key x 16 d] tr16 subtiles
 
-  const int bh = blockIdx.y;          // (b, h)
-  const int b = bh / H;
-  const int h = bh % H;
-  const int q0 = blockIdx.x * (NW * TQW);
+  // XCD-affine remap: with the natural (x=tile, y=b*H+h) order the B
+  // blocks that re-read one bias slice bias[h, tile rows, :] scatter
+  // across XCDs and the ~B-fold (H, L, L) fp32 bias re-read misses L2.
+  // Deriving (b, h, tile) from the linear dispatch id as id = b*S + slice
+  // (S = gridDim.x*H) makes same-slice blocks id-congruent mod S; since
+  // workgroups round-robin across the 8 XCDs, S % 8 == 0 pins each slice's
+  // B blocks to ONE XCD whose L2 then serves the repeats.
+  const int lid = blockIdx.y * gridDim.x + blockIdx.x;
+  const int S = gridDim.x * H;
+  const int b = lid / S;
+  const int h = (lid % S) / gridDim.x;
+  const int q0 = ((lid % S) % gridDim.x) * (NW * TQW);
+  const int bh = b * H + h;
   const int tid = threadIdx.x;
   const int lane = tid & (WAVE - 1);
   const int wid = tid >> 6;
@@ -488,10 +497,12 @@ __global__ __launch_bounds__(256) void flash_dq_kernel(
   char* kt_lds = smem + TK * 128;     // [d][key] swizzled, 8 KiB
   char* v_lds = smem + 2 * TK * 128;  // [key][d] swizzled, 8 KiB
 
-  const int bh = blockIdx.y;
-  const int b = bh / H;
-  const int h = bh % H;
-  const int q0 = blockIdx.x * (NWAVE * TQW);
+  const int lid = blockIdx.y * gridDim.x + blockIdx.x;  // XCD-affine (see fwd)
+  const int S = gridDim.x * H;
+  const int b = lid / S;
+  const int h = (lid % S) / gridDim.x;
+  const int q0 = ((lid % S) % gridDim.x) * (NWAVE * TQW);
+  const int bh = b * H + h;
   const int lane = threadIdx.x & (WAVE - 1);
   const int wid = threadIdx.x >> 6;
   const int qw = q0 + wid * TQW;
@@ -758,10 +769,12 @@ __global__ __launch_bounds__(256) void flash_dkv_kernel(
   char* v_lds = smem + TK * 128;
   char* wave_base = smem + 2 * TK * 128;
 
-  const int bh = blockIdx.y;
-  const int b = bh / H;
-  const int h = bh % H;
-  const int kv0 = blockIdx.x * TK;
+  const int lid = blockIdx.y * gridDim.x + blockIdx.x;  // XCD-affine (see fwd)
+  const int S = gridDim.x * H;
+  const int b = lid / S;
+  const int h = (lid % S) / gridDim.x;
+  const int kv0 = ((lid % S) % gridDim.x) * TK;
+  const int bh = b * H + h;
   const int lane = threadIdx.x & (WAVE - 1);
   const int wid = threadIdx.x >> 6;
   const int wk = wid >> 1;   // key half: keys wk*32 .. wk*32+31
