@@ -169,18 +169,27 @@ at::Tensor embed4_fwd(at::Tensor tables, at::Tensor idx, bool out_bf16 = false) 
   return out;
 }
 
-at::Tensor embed4_bwd(at::Tensor grad_out, at::Tensor idx, long V, long Demb) {
+at::Tensor embed4_bwd(at::Tensor grad_out, at::Tensor idx, long V, long Demb,
+                      c10::optional<at::Tensor> out_opt) {
   CHECK_GPU(grad_out);
   CHECK_GPU(idx);
   TORCH_CHECK(Demb == 32, "Demb must be 32");
-  auto grad = at::zeros({4, V, Demb}, grad_out.options().dtype(at::kFloat));
+  at::Tensor grad;
+  const bool acc = out_opt.has_value();
+  if (acc) {  // accumulate into the flat .grad region of the 4 tables
+    grad = *out_opt;
+    TORCH_CHECK(grad.is_cuda() && grad.scalar_type() == at::kFloat &&
+                grad.is_contiguous() && grad.numel() == 4 * V * Demb);
+  } else {
+    grad = at::zeros({4, V, Demb}, grad_out.options().dtype(at::kFloat));
+  }
   const long total = grad_out.numel();
   dispatch_float_bf16(grad_out, "embed4_bwd", [&](auto tag) {
     using T = decltype(tag);
     launch_embed4_bwd<T>(ptr<T>(grad_out), idx.data_ptr<long>(), grad.data_ptr<float>(), total,
                          (int)V, cur_stream());
   });
-  return grad.to(grad_out.scalar_type());
+  return acc ? grad : grad.to(grad_out.scalar_type());
 }
 
 at::Tensor spmm_sum(at::Tensor indptr, at::Tensor indices, at::Tensor x) {
@@ -449,7 +458,9 @@ std::vector<at::Tensor> ggnn_fused_bwd(at::Tensor grad_out, at::Tensor t_indptr,
                                        at::Tensor t_indices, at::Tensor x, at::Tensor W_eT,
                                        at::Tensor WcatT, at::Tensor HH,
                                        at::Tensor M, at::Tensor R, at::Tensor Z, at::Tensor Nn,
-                                       at::Tensor HN, long n_steps) {
+                                       at::Tensor HN, long n_steps,
+                                       c10::optional<at::Tensor> out_we,
+                                       c10::optional<at::Tensor> out_be) {
   const long N = x.size(0);
   const long H = x.size(1);
   const long S = n_steps;
@@ -499,10 +510,21 @@ std::vector<at::Tensor> ggnn_fused_bwd(at::Tensor grad_out, at::Tensor t_indptr,
   launch_wgrad(ptr<bf16_t>(A_g), ptr<bf16_t>(M), ptr<bf16_t>(HH), gWcat.data_ptr<float>(),
                cs4.data_ptr<float>(), S * N, 4 * H, 2 * H, H, stream);
   auto A_w = Gwh.view({S * N, H});
-  auto gW_e = at::zeros({H, H}, opts.dtype(at::kFloat));
-  auto cs_e = at::zeros({H}, opts.dtype(at::kFloat));
+  at::Tensor gW_e, cs_e;
+  const bool acc_we = out_we.has_value();
+  if (acc_we) {  // accumulate W_e/b_e grads into the flat .grad views
+    gW_e = *out_we;
+    cs_e = *out_be;
+    TORCH_CHECK(gW_e.is_cuda() && gW_e.scalar_type() == at::kFloat &&
+                gW_e.is_contiguous() && gW_e.numel() == H * H &&
+                cs_e.is_cuda() && cs_e.scalar_type() == at::kFloat &&
+                cs_e.numel() == H);
+  } else {
+    gW_e = at::zeros({H, H}, opts.dtype(at::kFloat));
+    cs_e = at::zeros({H}, opts.dtype(at::kFloat));
+  }
   launch_wgrad2(ptr<bf16_t>(A_w), ptr<bf16_t>(HH), gW_e.data_ptr<float>(),
-                cs_e.data_ptr<float>(), S * N, H, H, 0, stream);
+                cs_e.data_ptr<float>(), S * N, H, H, acc_we ? 1 : 0, stream);
   // scatter gWcat blocks back to the GRUCell weight layout (views are fine
   // as autograd outputs; no contiguous copy)
   auto gW_ih = gWcat.narrow(0, 0, 3 * H).narrow(1, 0, H);
@@ -979,14 +1001,24 @@ std::vector<at::Tensor> gate_pool_fwd(at::Tensor x1, at::Tensor x2, at::Tensor w
 
 std::vector<at::Tensor> gate_pool_bwd(at::Tensor grad_out, at::Tensor x1, at::Tensor x2,
                                       at::Tensor wg, at::Tensor alpha,
-                                      at::Tensor node_offsets) {
+                                      at::Tensor node_offsets,
+                                      c10::optional<at::Tensor> out_wg,
+                                      c10::optional<at::Tensor> out_bg) {
   CHECK_GPU(grad_out);
   const int N = x1.size(0), D1 = x1.size(1), D = D1 + (int)x2.size(1);
   const int B = node_offsets.numel() - 1;
   auto gx1 = at::empty_like(x1);
   auto gx2 = at::empty_like(x2);
-  auto dwg = at::zeros({D}, x1.options().dtype(at::kFloat));
-  auto dbg = at::zeros({1}, x1.options().dtype(at::kFloat));
+  at::Tensor dwg, dbg;
+  if (out_wg.has_value()) {  // accumulate into the flat .grad views (atomic)
+    dwg = *out_wg;
+    dbg = *out_bg;
+    TORCH_CHECK(dwg.is_cuda() && dwg.scalar_type() == at::kFloat && dwg.numel() == D &&
+                dbg.is_cuda() && dbg.scalar_type() == at::kFloat && dbg.numel() == 1);
+  } else {
+    dwg = at::zeros({D}, x1.options().dtype(at::kFloat));
+    dbg = at::zeros({1}, x1.options().dtype(at::kFloat));
+  }
   auto s_ws = at::empty({N}, x1.options().dtype(at::kFloat));
   launch_gate_pool_bwd(ptr<bf16_t>(grad_out), ptr<bf16_t>(x1), ptr<bf16_t>(x2),
                        wg.data_ptr<float>(), alpha.data_ptr<float>(),
@@ -1017,7 +1049,8 @@ std::vector<at::Tensor> mlp3_fwd(at::Tensor x, at::Tensor W1T, at::Tensor b1,
 
 std::vector<at::Tensor> mlp3_bwd(at::Tensor dlogits, at::Tensor x, at::Tensor h1,
                                  at::Tensor h2, at::Tensor W1, at::Tensor W2,
-                                 at::Tensor W3) {
+                                 at::Tensor W3,
+                                 c10::optional<std::vector<at::Tensor>> outs) {
   CHECK_GPU(dlogits);
   const int B = x.size(0), D = x.size(1);
   auto fopt = x.options().dtype(at::kFloat);
@@ -1028,17 +1061,31 @@ std::vector<at::Tensor> mlp3_bwd(at::Tensor dlogits, at::Tensor x, at::Tensor h1
                   W1.data_ptr<float>(), W2.data_ptr<float>(), W3.data_ptr<float>(),
                   dh1.data_ptr<float>(), dh2.data_ptr<float>(), mptr<bf16_t>(dx), B, D,
                   cur_stream());
-  auto dW1 = at::zeros({D, D}, fopt);  // atomic r-chunk epilogue
-  auto dW2 = at::zeros({D, D}, fopt);
-  auto dW3 = at::empty({1, D}, fopt);
-  auto db1 = at::empty({D}, fopt);
-  auto db2 = at::empty({D}, fopt);
-  auto db3 = at::empty({1}, fopt);
+  at::Tensor dW1, dW2, dW3, db1, db2, db3;
+  if (outs.has_value()) {  // accumulate into the flat .grad views
+    TORCH_CHECK(outs->size() == 6, "outs = [dW1, db1, dW2, db2, dW3, db3]");
+    dW1 = (*outs)[0]; db1 = (*outs)[1]; dW2 = (*outs)[2];
+    db2 = (*outs)[3]; dW3 = (*outs)[4]; db3 = (*outs)[5];
+    for (auto& t : *outs)
+      TORCH_CHECK(t.is_cuda() && t.scalar_type() == at::kFloat && t.is_contiguous());
+    TORCH_CHECK(dW1.numel() == (long)D * D && dW2.numel() == (long)D * D &&
+                dW3.numel() == D && db1.numel() == D && db2.numel() == D &&
+                db3.numel() == 1);
+  } else {
+    // the wgrad tail accumulates (+=) so every output starts zeroed
+    dW1 = at::zeros({D, D}, fopt);
+    dW2 = at::zeros({D, D}, fopt);
+    dW3 = at::zeros({1, D}, fopt);
+    db1 = at::zeros({D}, fopt);
+    db2 = at::zeros({D}, fopt);
+    db3 = at::zeros({1}, fopt);
+  }
   launch_mlp3_wgrad(ptr<bf16_t>(x), h1.data_ptr<float>(), h2.data_ptr<float>(),
                     dh1.data_ptr<float>(), dh2.data_ptr<float>(),
                     dlogits.data_ptr<float>(), dW1.data_ptr<float>(),
                     dW2.data_ptr<float>(), dW3.data_ptr<float>(), db1.data_ptr<float>(),
                     db2.data_ptr<float>(), db3.data_ptr<float>(), B, D, cur_stream());
+  if (outs.has_value()) return {dx};
   return {dx, dW1, dW2, dW3, db1, db2, db3};
 }
 
@@ -1072,7 +1119,9 @@ void adamw_fused(at::Tensor p, at::Tensor g, at::Tensor m, at::Tensor v, double 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "deepdfa_amd MI355X (gfx950) kernels";
   m.def("embed4_fwd", &embed4_fwd, pybind11::arg("tables"), pybind11::arg("idx"), pybind11::arg("out_bf16") = false);
-  m.def("embed4_bwd", &embed4_bwd);
+  m.def("embed4_bwd", &embed4_bwd, pybind11::arg("grad_out"), pybind11::arg("idx"),
+        pybind11::arg("V"), pybind11::arg("Demb"),
+        pybind11::arg("out") = pybind11::none());
   m.def("spmm_sum", &spmm_sum);
   m.def("gru_gates_fwd", &gru_gates_fwd);
   m.def("gru_gates_bwd", &gru_gates_bwd);
@@ -1112,9 +1161,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     return dlogits;
   });
   m.def("gate_pool_fwd", &gate_pool_fwd);
-  m.def("gate_pool_bwd", &gate_pool_bwd);
+  m.def("gate_pool_bwd", &gate_pool_bwd, pybind11::arg("grad_out"), pybind11::arg("x1"),
+        pybind11::arg("x2"), pybind11::arg("wg"), pybind11::arg("alpha"),
+        pybind11::arg("node_offsets"), pybind11::arg("out_wg") = pybind11::none(),
+        pybind11::arg("out_bg") = pybind11::none());
   m.def("mlp3_fwd", &mlp3_fwd);
-  m.def("mlp3_bwd", &mlp3_bwd);
+  m.def("mlp3_bwd", &mlp3_bwd, pybind11::arg("dlogits"), pybind11::arg("x"),
+        pybind11::arg("h1"), pybind11::arg("h2"), pybind11::arg("W1"),
+        pybind11::arg("W2"), pybind11::arg("W3"),
+        pybind11::arg("outs") = pybind11::none());
   m.def("layernorm_fwd", &layernorm_fwd);
   m.def("layernorm_bwd", &layernorm_bwd);
   m.def("layernorm_wgrad", &layernorm_wgrad);
@@ -1290,5 +1345,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         pybind11::arg("l2_mode") = false,
         pybind11::arg("p16") = pybind11::none());
   m.def("softmax_mask_bwd", &softmax_mask_bwd);
-  m.def("ggnn_fused_bwd", &ggnn_fused_bwd);
+  m.def("ggnn_fused_bwd", &ggnn_fused_bwd, pybind11::arg("grad_out"),
+        pybind11::arg("t_indptr"), pybind11::arg("t_indices"), pybind11::arg("x"),
+        pybind11::arg("W_eT"), pybind11::arg("WcatT"), pybind11::arg("HH"),
+        pybind11::arg("M"), pybind11::arg("R"), pybind11::arg("Z"), pybind11::arg("Nn"),
+        pybind11::arg("HN"), pybind11::arg("n_steps"),
+        pybind11::arg("out_we") = pybind11::none(),
+        pybind11::arg("out_be") = pybind11::none());
 }

@@ -1086,7 +1086,7 @@ __global__ void mlp3_wgrad_kernel(
       s3 += dlogits[r + 3] * h2[(long)(r + 3) * D + o];
     }
     for (; r < B; ++r) s0 += dlogits[r] * h2[(long)r * D + o];
-    dW3[o] = (s0 + s1) + (s2 + s3);
+    dW3[o] += (s0 + s1) + (s2 + s3);  // accumulate: out may be a live .grad
   } else if (o < 3 * D) {
     const float* src = srcs[o / D];
     const int jj = o % D;
@@ -1099,10 +1099,10 @@ __global__ void mlp3_wgrad_kernel(
     }
     for (; r < B; ++r) s0 += src[(long)r * D + jj];
     float* dst = (o < 2 * D) ? db1 : db2;
-    dst[jj] = (s0 + s1) + (s2 + s3);
+    dst[jj] += (s0 + s1) + (s2 + s3);
   } else {
     for (int r = 0; r < B; ++r) s0 += dlogits[r];
-    db3[0] = s0;
+    db3[0] += s0;
   }
 }
 

@@ -168,6 +168,15 @@ class FlowGNNGGNNModule(BaseModule):
             idx = torch.stack(
                 [graph.ndata[f"_ABS_DATAFLOW_{of}"] for of in ALL_FEATS], dim=1
             )
+            if torch.is_grad_enabled() and idx.is_cuda:
+                # flat-optimizer fast path: zero-copy bf16 stack view of the
+                # adjacent tables + direct scatter into the flat .grad region
+                from ..ops.flowgnn import embed4_direct
+
+                ws = [self.all_embeddings[of].weight for of in ALL_FEATS]
+                out = embed4_direct(ws, idx)
+                if out is not None:
+                    return out
             tables = self._stacked_tables()
             return embed4(tables, idx)
         feat = graph.ndata[self.feature_keys["feature"]]

@@ -52,6 +52,57 @@ def embed4(tables: torch.Tensor, idx: torch.Tensor) -> torch.Tensor:
     return _Embed4.apply(tables, idx)
 
 
+class _Embed4Direct(torch.autograd.Function):
+    """K1 under the flat optimizer: the (4, V, 32) table stack is a
+    ZERO-COPY bf16 view of the optimizer's shadow buffer (the 4 tables are
+    adjacent), and the backward scatter-adds straight into the flat fp32
+    .grad region — no per-step stack copy, no (4, V, 32) zeros, no unbind
+    adds, no dtype casts. `w0` is only the autograd hook that makes the
+    output differentiable; its grad is returned as None."""
+
+    @staticmethod
+    def forward(ctx, w0, tables16, idx, wrefs):
+        ctx.save_for_backward(idx)
+        ctx.wrefs = wrefs
+        ctx.V = tables16.shape[1]
+        ext = ext_for(tables16)
+        return ext.embed4_fwd(tables16, idx, False)
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        (idx,) = ctx.saved_tensors
+        ws = ctx.wrefs
+        w0 = ws[0]
+        n4 = sum(w.numel() for w in ws)
+        g4 = w0._dfa_gbase[w0._dfa_goff : w0._dfa_goff + n4]
+        ext = ext_for(grad_out)
+        ext.embed4_bwd(grad_out.contiguous(), idx, ctx.V, 32, out=g4)
+        return None, None, None, None
+
+
+def embed4_direct_view(ws):
+    """Returns the zero-copy (4, V, 32) bf16 stack view of the 4 table
+    weights' optimizer shadow, or None when unavailable (no flat optimizer,
+    non-adjacent tables, or CPU)."""
+    if any(getattr(w, "_dfa_w16", None) is None or w.grad is None for w in ws):
+        return None
+    base = ws[0]._dfa_w16_base
+    off = ws[0]._dfa_w16_off
+    n = ws[0].numel()
+    for i, w in enumerate(ws):
+        if (w._dfa_w16_base is not base or w._dfa_w16_off != off + i * n
+                or w.numel() != n or w.shape[-1] != 32):
+            return None
+    return base[off : off + 4 * n].view(4, ws[0].shape[0], 32)
+
+
+def embed4_direct(ws, idx):
+    tables16 = embed4_direct_view(ws)
+    if tables16 is None:
+        return None
+    return _Embed4Direct.apply(ws[0], tables16, idx, ws)
+
+
 class _SpmmSum(torch.autograd.Function):
     """K2 aggregation: m[v] = sum_{u->v} x[u] over the block-diagonal CSR.
     Backward gathers through the transpose (CSC)."""
@@ -199,6 +250,7 @@ class _GGNNFused(torch.autograd.Function):
         ctx.n_steps = n_steps
         ctx.grad_dtypes = (w_e.dtype, b_e.dtype, w_ih.dtype, w_hh.dtype,
                            b_ih.dtype, b_hh.dtype)
+        ctx.we_refs = (w_e, b_e)
         return h_final
 
     @staticmethod
@@ -207,13 +259,20 @@ class _GGNNFused(torch.autograd.Function):
 
         ext = load_ext(required=True)
         x, W_eT, WcatT, HH, M, R, Z, Nn, HN, t_indptr, t_indices = ctx.saved_tensors
+        w_e, b_e = ctx.we_refs
+        direct = (getattr(w_e, "_dfa_w16", None) is not None
+                  and w_e.grad is not None and b_e.grad is not None)
         grad_x, gW_e, gb_e, gW_ih, gW_hh, gb_ih, gb_hh = ext.ggnn_fused_bwd(
             grad_out.contiguous(), t_indptr, t_indices, x, W_eT, WcatT,
             HH, M, R, Z, Nn, HN, ctx.n_steps,
+            out_we=w_e.grad if direct else None,
+            out_be=b_e.grad if direct else None,
         )
         dts = ctx.grad_dtypes
         grads = [gW_e, gb_e, gW_ih, gW_hh, gb_ih, gb_hh]
         grads = [g if g.dtype == dt else g.to(dt) for g, dt in zip(grads, dts)]
+        if direct:
+            grads[0] = grads[1] = None  # accumulated in-kernel (flat .grad)
         return (grad_x, *grads, None, None, None)
 
 
@@ -287,6 +346,7 @@ class _GatePool(torch.autograd.Function):
         out, alpha = ext.gate_pool_fwd(x1c, x2c, wg.detach().reshape(-1).contiguous(),
                                        bg.detach().contiguous(), node_offsets)
         ctx.save_for_backward(x1c, x2c, wg, alpha, node_offsets)
+        ctx.gparams = (wg, bg)
         return out
 
     @staticmethod
@@ -295,6 +355,14 @@ class _GatePool(torch.autograd.Function):
 
         ext = load_ext(required=True)
         x1, x2, wg, alpha, node_offsets = ctx.saved_tensors
+        wgp, bgp = ctx.gparams
+        if (getattr(wgp, "_dfa_w16", None) is not None and wgp.grad is not None
+                and bgp is not None and bgp.grad is not None):
+            gx1, gx2, _, _ = ext.gate_pool_bwd(
+                grad_out.contiguous(), x1, x2, wg.detach().reshape(-1).contiguous(),
+                alpha, node_offsets, out_wg=wgp.grad, out_bg=bgp.grad,
+            )
+            return gx1, gx2, None, None, None
         gx1, gx2, dwg, dbg = ext.gate_pool_bwd(
             grad_out.contiguous(), x1, x2, wg.detach().reshape(-1).contiguous(),
             alpha, node_offsets,
@@ -343,6 +411,7 @@ class _MLP3(torch.autograd.Function):
             w3.detach().reshape(-1).contiguous(), b3.detach().contiguous(),
         )
         ctx.save_for_backward(xc, h1, h2, w1, w2, w3)
+        ctx.mparams = (w1, b1, w2, b2, w3, b3)
         return logits
 
     @staticmethod
@@ -351,6 +420,16 @@ class _MLP3(torch.autograd.Function):
 
         ext = load_ext(required=True)
         x, h1, h2, w1, w2, w3 = ctx.saved_tensors
+        ps = ctx.mparams
+        if all(getattr(q, "_dfa_w16", None) is not None and q.grad is not None
+               for q in ps):
+            (dx,) = ext.mlp3_bwd(
+                dlogits.float().contiguous(), x, h1, h2, w1.detach().contiguous(),
+                w2.detach().contiguous(), w3.detach().reshape(-1).contiguous(),
+                outs=[ps[0].grad, ps[1].grad, ps[2].grad, ps[3].grad,
+                      ps[4].grad, ps[5].grad],
+            )
+            return dx, None, None, None, None, None, None, None, None
         dx, dW1, dW2, dW3, db1, db2, db3 = ext.mlp3_bwd(
             dlogits.float().contiguous(), x, h1, h2, w1.detach().contiguous(),
             w2.detach().contiguous(), w3.detach().reshape(-1).contiguous(),

@@ -188,3 +188,42 @@ def test_direct_grad_accumulation_matches_autograd():
         tol = 1e-4 + 1e-4 * r.abs().max()
         assert torch.allclose(p.grad.float(), r, atol=float(tol)), (
             n, (p.grad.float() - r).abs().max())
+
+
+@pytest.mark.gpu
+def test_ddfa_direct_grad_accumulation_matches_autograd():
+    """FlowGNN direct-grad paths (embed4_direct, ggnn W_e/b_e, gate_pool,
+    mlp3 — ops/flowgnn.py): grads with FlatAdamW attached must match the
+    plain autograd route on the same batch."""
+    import torch
+    from deepdfa_amd.graph.synthetic import synthetic_cfg_batch
+    from deepdfa_amd.models import FlowGNNGGNNModule
+    from deepdfa_amd.parallel.optim import FlatAdamW
+
+    torch.manual_seed(0)
+    model = FlowGNNGGNNModule(input_dim=1002).to("cuda")
+    g = synthetic_cfg_batch(64, seed=1).to("cuda")
+
+    def step():
+        with torch.autocast(device_type="cuda", dtype=torch.bfloat16):
+            return model.training_step((g, {}))
+
+    loss_ref = step()
+    loss_ref.backward()
+    ref = {n: p.grad.detach().clone() for n, p in model.named_parameters()}
+    for p in model.parameters():
+        p.grad = None
+    model.metrics["train"].reset()
+
+    opt = FlatAdamW(model.parameters(), lr=1e-3)
+    opt.zero_grad()
+    loss = step()
+    loss.backward()
+    assert torch.allclose(loss.float(), loss_ref.float(), atol=1e-4, rtol=1e-4)
+    bad = []
+    for n, p in model.named_parameters():
+        r = ref[n].float()
+        tol = 2e-4 + 2e-4 * float(r.abs().max())
+        if not torch.allclose(p.grad.float(), r, atol=tol):
+            bad.append((n, float((p.grad.float() - r).abs().max())))
+    assert not bad, bad
