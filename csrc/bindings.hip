@@ -44,7 +44,8 @@ void launch_gemm_bias(const __hip_bfloat16*, const __hip_bfloat16*, const __hip_
                       const __hip_bfloat16*, const __hip_bfloat16*, __hip_bfloat16*, int, int,
                       int, int, hipStream_t);
 void launch_wgrad(const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*, float*,
-                  float*, int, int, int, int, hipStream_t);
+                  float*, int, int, int, int, hipStream_t, float* = nullptr,
+                  float* = nullptr, float* = nullptr, float* = nullptr);
 template <typename T>
 void launch_layernorm_fwd(const T*, const float*, const float*, T*, float*, float*, long, int,
                           float, hipStream_t);
@@ -460,7 +461,8 @@ std::vector<at::Tensor> ggnn_fused_bwd(at::Tensor grad_out, at::Tensor t_indptr,
                                        at::Tensor M, at::Tensor R, at::Tensor Z, at::Tensor Nn,
                                        at::Tensor HN, long n_steps,
                                        c10::optional<at::Tensor> out_we,
-                                       c10::optional<at::Tensor> out_be) {
+                                       c10::optional<at::Tensor> out_be,
+                                       c10::optional<std::vector<at::Tensor>> gru_outs) {
   const long N = x.size(0);
   const long H = x.size(1);
   const long S = n_steps;
@@ -500,15 +502,32 @@ std::vector<at::Tensor> ggnn_fused_bwd(at::Tensor grad_out, at::Tensor t_indptr,
   }
   // batched weight/bias grads over all steps (K = S*N)
   auto A_g = Ggicat.view({S * N, 4 * H});
-  auto gWcat = at::zeros({4 * H, 2 * H}, opts.dtype(at::kFloat));
-  auto cs4 = at::zeros({4 * H}, opts.dtype(at::kFloat));
+  const bool gru_direct = gru_outs.has_value();
+  at::Tensor gWcat, cs4;
+  if (gru_direct) {  // scatter epilogue straight into the 4 flat .grad views
+    TORCH_CHECK(gru_outs->size() == 4, "gru_outs = [gW_ih, gW_hh, gb_ih, gb_hh]");
+    for (auto& t : *gru_outs)
+      TORCH_CHECK(t.is_cuda() && t.scalar_type() == at::kFloat && t.is_contiguous());
+    TORCH_CHECK((*gru_outs)[0].numel() == 3 * H * H &&
+                (*gru_outs)[1].numel() == 3 * H * H &&
+                (*gru_outs)[2].numel() == 3 * H && (*gru_outs)[3].numel() == 3 * H);
+  } else {
+    gWcat = at::zeros({4 * H, 2 * H}, opts.dtype(at::kFloat));
+    cs4 = at::zeros({4 * H}, opts.dtype(at::kFloat));
+  }
   // bias grads (column sums of the gate/message grads) ride along in the
   // wgrad kernels' A-tile staging — no separate colsum pass. The gate
   // wgrad is a SPLIT-B product (no [m|h] concat) so it stays on the
   // generic split-K kernel; the W_e grad (plain A^T B, 128x128) takes the
   // tr16-subtiled wgrad2 (K tails zero-filled in staging).
-  launch_wgrad(ptr<bf16_t>(A_g), ptr<bf16_t>(M), ptr<bf16_t>(HH), gWcat.data_ptr<float>(),
-               cs4.data_ptr<float>(), S * N, 4 * H, 2 * H, H, stream);
+  if (gru_direct)
+    launch_wgrad(ptr<bf16_t>(A_g), ptr<bf16_t>(M), ptr<bf16_t>(HH), nullptr, nullptr,
+                 S * N, 4 * H, 2 * H, H, stream, (*gru_outs)[0].data_ptr<float>(),
+                 (*gru_outs)[1].data_ptr<float>(), (*gru_outs)[2].data_ptr<float>(),
+                 (*gru_outs)[3].data_ptr<float>());
+  else
+    launch_wgrad(ptr<bf16_t>(A_g), ptr<bf16_t>(M), ptr<bf16_t>(HH), gWcat.data_ptr<float>(),
+                 cs4.data_ptr<float>(), S * N, 4 * H, 2 * H, H, stream);
   auto A_w = Gwh.view({S * N, H});
   at::Tensor gW_e, cs_e;
   const bool acc_we = out_we.has_value();
@@ -525,6 +544,10 @@ std::vector<at::Tensor> ggnn_fused_bwd(at::Tensor grad_out, at::Tensor t_indptr,
   }
   launch_wgrad2(ptr<bf16_t>(A_w), ptr<bf16_t>(HH), gW_e.data_ptr<float>(),
                 cs_e.data_ptr<float>(), S * N, H, H, acc_we ? 1 : 0, stream);
+  if (gru_direct) {  // grads already accumulated in the flat views
+    auto none = at::empty({0}, opts.dtype(at::kFloat));
+    return {grad_h, gW_e, cs_e, none, none, none, none};
+  }
   // scatter gWcat blocks back to the GRUCell weight layout (views are fine
   // as autograd outputs; no contiguous copy)
   auto gW_ih = gWcat.narrow(0, 0, 3 * H).narrow(1, 0, H);
@@ -1351,5 +1374,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         pybind11::arg("M"), pybind11::arg("R"), pybind11::arg("Z"), pybind11::arg("Nn"),
         pybind11::arg("HN"), pybind11::arg("n_steps"),
         pybind11::arg("out_we") = pybind11::none(),
-        pybind11::arg("out_be") = pybind11::none());
+        pybind11::arg("out_be") = pybind11::none(),
+        pybind11::arg("gru_outs") = pybind11::none());
 }

@@ -37,10 +37,20 @@ using uint4v = __attribute__((ext_vector_type(4))) unsigned int;
 // csum (optional): column sums of A (the bias gradients) accumulated by
 // the blockIdx.y == 0 blocks from the already-staged LDS tiles — removes
 // the separate colsum pass over the same 60 MB of activation grads
+// GRU scatter epilogue (optional): when out_ih != nullptr the (4H, 2H)
+// gWcat tile output is scattered straight into the torch-GRUCell grad
+// layouts instead of one packed buffer — gW_ih (3H, H) = rows [0,3H) x
+// m-cols, gW_hh (3H, H) = rows [0,2H) + [3H,4H) x h-cols (the [2H,3H)
+// h-block is structurally zero), and the bias column-sums go to b_ih
+// (rows [0,3H)) and b_hh (rows [0,2H) + [3H,4H)), rows [0,2H) feeding
+// BOTH biases (r/z gates add b_ih and b_hh). Replaces the narrow/cat/
+// AccumulateGrad chain in the flat-optimizer path (ops/flowgnn.py).
 __global__ __launch_bounds__(256) void wgrad_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ B1,
     const bf16* __restrict__ B2, float* __restrict__ out,
-    float* __restrict__ csum, int K, int M, int C, int C1, int kchunk) {
+    float* __restrict__ csum, int K, int M, int C, int C1, int kchunk,
+    float* __restrict__ out_ih, float* __restrict__ out_hh,
+    float* __restrict__ cs_ih, float* __restrict__ cs_hh) {
   __shared__ char lds_a[KS * 256];
   __shared__ char lds_b[KS * 256];
 
@@ -56,7 +66,8 @@ __global__ __launch_bounds__(256) void wgrad_kernel(
 
   f32x4 acc[4][4] = {};
   float cs_acc = 0.f;
-  const bool do_csum = (csum != nullptr) && (blockIdx.y == 0) && (tid < 128);
+  const bool do_csum = (csum != nullptr || cs_ih != nullptr) &&
+                       (blockIdx.y == 0) && (tid < 128);
 
   for (int k0 = k_begin; k0 < k_end; k0 += KS) {
     // stage [KS][128] tiles (row-major, coalesced 16B; 16 threads/row)
@@ -112,7 +123,17 @@ __global__ __launch_bounds__(256) void wgrad_kernel(
     __syncthreads();
   }
 
-  if (do_csum && m0 + tid < M) atomicAdd(csum + m0 + tid, cs_acc);
+  if (do_csum && m0 + tid < M) {
+    const int m = m0 + tid;
+    if (cs_ih != nullptr) {
+      const int Hg = C1;  // gate width: C1 == H for the GGNN gWcat call
+      if (m < 3 * Hg) atomicAdd(cs_ih + m, cs_acc);
+      if (m < 2 * Hg) atomicAdd(cs_hh + m, cs_acc);
+      else if (m >= 3 * Hg) atomicAdd(cs_hh + m - Hg, cs_acc);
+    } else {
+      atomicAdd(csum + m, cs_acc);
+    }
+  }
   // epilogue: D col = lane&15, row = (lane>>4)*4 + i; atomic fp32 accumulate
 #pragma unroll
   for (int fm = 0; fm < 4; ++fm) {
@@ -124,15 +145,28 @@ __global__ __launch_bounds__(256) void wgrad_kernel(
 #pragma unroll
       for (int i = 0; i < 4; ++i) {
         const int m = m_base + i;
-        if (m < M) atomicAdd(out + (long)m * C + c, acc[fm][fc][i]);
+        if (m >= M) continue;
+        if (out_ih != nullptr) {
+          const int Hg = C1;
+          if (c < Hg) {
+            if (m < 3 * Hg) atomicAdd(out_ih + (long)m * Hg + c, acc[fm][fc][i]);
+          } else {
+            const int ch = c - Hg;
+            if (m < 2 * Hg) atomicAdd(out_hh + (long)m * Hg + ch, acc[fm][fc][i]);
+            else if (m >= 3 * Hg)
+              atomicAdd(out_hh + (long)(m - Hg) * Hg + ch, acc[fm][fc][i]);
+          }
+        } else {
+          atomicAdd(out + (long)m * C + c, acc[fm][fc][i]);
+        }
       }
     }
   }
 }
 
 void launch_wgrad(const bf16* A, const bf16* B1, const bf16* B2, float* out,
-                  float* csum, int K, int M, int C, int C1,
-                  hipStream_t stream) {
+                  float* csum, int K, int M, int C, int C1, hipStream_t stream,
+                  float* out_ih, float* out_hh, float* cs_ih, float* cs_hh) {
   // size the K-split so the grid comfortably fills 256 CUs
   const int tiles = ((M + 127) / 128) * ((C + 127) / 128);
   int zsplit = max(1, 512 / tiles);
@@ -144,5 +178,5 @@ void launch_wgrad(const bf16* A, const bf16* B1, const bf16* B2, float* out,
   zsplit = (K + kchunk - 1) / kchunk;
   const dim3 grid((M + 127) / 128, (C + 127) / 128, zsplit);
   hipLaunchKernelGGL(wgrad_kernel, grid, dim3(256), 0, stream, A, B1, B2, out,
-                     csum, K, M, C, C1, kchunk);
+                     csum, K, M, C, C1, kchunk, out_ih, out_hh, cs_ih, cs_hh);
 }

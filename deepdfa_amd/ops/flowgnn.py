@@ -250,7 +250,7 @@ class _GGNNFused(torch.autograd.Function):
         ctx.n_steps = n_steps
         ctx.grad_dtypes = (w_e.dtype, b_e.dtype, w_ih.dtype, w_hh.dtype,
                            b_ih.dtype, b_hh.dtype)
-        ctx.we_refs = (w_e, b_e)
+        ctx.we_refs = (w_e, b_e, w_ih, w_hh, b_ih, b_hh)
         return h_final
 
     @staticmethod
@@ -259,20 +259,26 @@ class _GGNNFused(torch.autograd.Function):
 
         ext = load_ext(required=True)
         x, W_eT, WcatT, HH, M, R, Z, Nn, HN, t_indptr, t_indices = ctx.saved_tensors
-        w_e, b_e = ctx.we_refs
-        direct = (getattr(w_e, "_dfa_w16", None) is not None
-                  and w_e.grad is not None and b_e.grad is not None)
+        params = ctx.we_refs
+        direct = all(getattr(q, "_dfa_w16", None) is not None and q.grad is not None
+                     for q in params)
+        if direct:
+            w_e, b_e, w_ih, w_hh, b_ih, b_hh = params
+            grad_x, *_ = ext.ggnn_fused_bwd(
+                grad_out.contiguous(), t_indptr, t_indices, x, W_eT, WcatT,
+                HH, M, R, Z, Nn, HN, ctx.n_steps,
+                out_we=w_e.grad, out_be=b_e.grad,
+                gru_outs=[w_ih.grad, w_hh.grad, b_ih.grad, b_hh.grad],
+            )
+            # every parameter grad accumulated in-kernel (flat .grad views)
+            return (grad_x, None, None, None, None, None, None, None, None, None)
         grad_x, gW_e, gb_e, gW_ih, gW_hh, gb_ih, gb_hh = ext.ggnn_fused_bwd(
             grad_out.contiguous(), t_indptr, t_indices, x, W_eT, WcatT,
             HH, M, R, Z, Nn, HN, ctx.n_steps,
-            out_we=w_e.grad if direct else None,
-            out_be=b_e.grad if direct else None,
         )
         dts = ctx.grad_dtypes
         grads = [gW_e, gb_e, gW_ih, gW_hh, gb_ih, gb_hh]
         grads = [g if g.dtype == dt else g.to(dt) for g, dt in zip(grads, dts)]
-        if direct:
-            grads[0] = grads[1] = None  # accumulated in-kernel (flat .grad)
         return (grad_x, *grads, None, None, None)
 
 
