@@ -243,7 +243,11 @@ __global__ __launch_bounds__(NW * 64, 12 / NW) void flash_fwd_kernel(
         for (int i = 0; i < 4; ++i) {
           const int key = kv0 + fk * 16 + (lane >> 4) * 4 + i;
           float s = s_acc[fk][fq][i] * scale;
-          if (bias) s += bias[((long)h * L + qcol[fq]) * L + key];
+          // bias is stored TRANSPOSED (H, key, qcol): the 16 qcol lanes of
+        // one unrolled load hit one 64-B line (row-major (H, q, key) made
+        // every lane fetch its own line: measured ~half the biased
+        // kernel's time, tools/flash_bias_probe.py)
+        if (bias) s += bias[((long)h * L + key) * L + qcol[fq]];
           const bool masked = key >= vl || (causal && key > qcol[fq]) || qcol[fq] >= L;
           s = masked ? -3.4e38f : s;
           s_acc[fk][fq][i] = s;
@@ -477,7 +481,7 @@ __device__ __forceinline__ void recompute_pT(
       for (int i = 0; i < 4; ++i) {
         const int key = kv0 + fk * 16 + (lane >> 4) * 4 + i;
         float s = s_acc[fk][fq][i] * scale;
-        if (bias) s += bias[((long)h * L + qcol) * L + key];
+        if (bias) s += bias[((long)h * L + key) * L + qcol];  // (H, key, qcol) layout
         const bool masked = key >= vl || (causal && key > qcol) || qcol >= L;
         p[fk][fq][i] = masked ? 0.f : __expf(s - lse_w[fq]);
       }
@@ -635,7 +639,7 @@ __global__ __launch_bounds__(256) void flash_dq_kernel(
           }
           ds[fk][fq][i] = p[fk][fq][i] * (dp - dterm_w[fq]);
           if (dBias && ds[fk][fq][i] != 0.f)
-            atomicAdd(dBias + ((long)h * L + qcol) * L + key, ds[fk][fq][i]);
+            atomicAdd(dBias + ((long)h * L + key) * L + qcol, ds[fk][fq][i]);
         }
 
     // dQ += scale * ds @ K : A = ds (q, key) via lane exchange, B = K^T
@@ -722,7 +726,7 @@ __device__ __forceinline__ void recompute_pT32(
       for (int i = 0; i < 4; ++i) {
         const int key = kv0 + key_off + fk * 16 + (lane >> 4) * 4 + i;
         float sc = s_acc[fk][fq][i] * scale;
-        if (bias) sc += bias[((long)h * L + qcol) * L + key];
+        if (bias) sc += bias[((long)h * L + key) * L + qcol];  // (H, key, qcol) layout
         const bool masked = key >= vl || (causal && key > qcol) || qcol >= L;
         p[fk][fq][i] = masked ? 0.f : __expf(sc - lse_w[fq]);
       }

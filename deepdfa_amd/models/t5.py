@@ -204,8 +204,7 @@ class T5Attention(nn.Module):
 
             bias = None
             if position_bias is not None:
-                bsrc = position_bias.detach() if bias_accum is not None else position_bias
-                bias = bsrc.squeeze(0).float().contiguous()
+                bias = _flash_bias_T(position_bias, bias_accum)
             out = flash_attention_qkv(qkv, H, valid=valid, bias=bias, scale=1.0,
                                       causal=causal, dropout_p=dropout_p,
                                       bias_accum=bias_accum)
@@ -228,8 +227,7 @@ class T5Attention(nn.Module):
         if d == 64 and flash_usable(qp, Lq, Lk):
             bias = None
             if position_bias is not None:
-                bsrc = position_bias.detach() if bias_accum is not None else position_bias
-                bias = bsrc.squeeze(0).float().contiguous()
+                bias = _flash_bias_T(position_bias, bias_accum)
             out = flash_attention(qp, kp, vp, H, valid=valid, bias=bias, scale=1.0,
                                   causal=causal, dropout_p=dropout_p,
                                   bias_accum=bias_accum)
@@ -395,6 +393,22 @@ class T5Stack(nn.Module):
         return self.final_layer_norm(x)
 
 
+def _flash_bias_T(position_bias, bias_accum):
+    """(H, Lk, Lq) TRANSPOSED fp32 position bias for the flash kernels
+    (csrc/flash_attn.hip reads bias key-major so the 16 q-column lanes of
+    each unrolled load share one 64-B line; the row-major layout made every
+    lane fetch its own line — measured ~half the biased kernel time,
+    tools/flash_bias_probe.py). Cached on the per-step position_bias tensor
+    so all layers share ONE transpose kernel."""
+    cached = getattr(position_bias, "_dfa_biasT", None)
+    if cached is not None:
+        return cached
+    bsrc = position_bias.detach() if bias_accum is not None else position_bias
+    bT = bsrc.squeeze(0).float().transpose(-1, -2).contiguous()
+    position_bias._dfa_biasT = bT
+    return bT
+
+
 class _BiasGradSink(torch.autograd.Function):
     """Applied to the stack INPUT so its backward runs AFTER every layer's:
     returns the shared atomically-accumulated position-bias gradient as the
@@ -414,7 +428,9 @@ class _BiasGradSink(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, g):
-        return g, ctx.accum.view(ctx.bias_shape), None
+        # accum holds the TRANSPOSED (H, Lk, Lq) atomics from the dq
+        # kernels (_flash_bias_T layout); emit the grad row-major
+        return g, ctx.accum.transpose(-1, -2).reshape(ctx.bias_shape), None
 
 
 class _DecodeState:

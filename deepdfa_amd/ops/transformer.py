@@ -296,7 +296,8 @@ def fused_linear(x, weight, bias=None):
 class _FlashAttention(torch.autograd.Function):
     """Fused attention, head_dim 64, (B, L, H*64) layout (csrc/flash_attn.hip):
     masked softmax + dropout + PV in one kernel, FA2-style two-pass backward.
-    `bias` is the T5 additive position bias (fp32, (H, L, L))."""
+    `bias` is the T5 additive position bias, fp32, TRANSPOSED key-major
+    (H, Lk, Lq) — see models/t5.py _flash_bias_T."""
 
     @staticmethod
     def forward(ctx, q, k, v, H, valid, bias, scale, causal, dropout_p,

@@ -67,7 +67,8 @@ def test_flash_fwd_bias(dev):
 
     q, k, v = rand_qkv(dev, seed=2)
     bias = torch.randn(4, 128, 128, device=dev)
-    out = flash_attention(q, k, v, 4, bias=bias, scale=1.0)
+    out = flash_attention(q, k, v, 4, bias=bias.transpose(-1, -2).contiguous(),
+                          scale=1.0)
     ref_o = materialized(q, k, v, 4, None, bias, 1.0, False)
     err = (out.float() - ref_o).abs().max().item()
     assert err < 5e-2, err
@@ -102,7 +103,8 @@ def test_flash_bwd_bias_grad(dev):
 
     q, k, v = rand_qkv(dev, B=2, seed=4)
     bias = torch.randn(4, 128, 128, device=dev, requires_grad=True)
-    out = flash_attention(q, k, v, 4, bias=bias, scale=1.0, causal=True)
+    out = flash_attention(q, k, v, 4, bias=bias.transpose(-1, -2).contiguous(),
+                          scale=1.0, causal=True)
     go = torch.randn_like(out)
     out.backward(go)
     b2 = bias.detach().clone().requires_grad_(True)
