@@ -227,3 +227,34 @@ def test_ddfa_direct_grad_accumulation_matches_autograd():
         if not torch.allclose(p.grad.float(), r, atol=tol):
             bad.append((n, float((p.grad.float() - r).abs().max())))
     assert not bad, bad
+
+
+@pytest.mark.gpu
+def test_direct_grad_accumulation_sums_over_microbatches():
+    """Gradient accumulation contract: with FlatAdamW, every direct-grad
+    kernel must ACCUMULATE into .grad (never store) — two backwards before
+    a step must yield exactly 2x the single-backward grads. Covers the
+    mlp3 tail (+=), wgrad2 forced-atomic, gate_pool/embed4/ggnn atomics."""
+    import torch
+    from deepdfa_amd.graph.synthetic import synthetic_cfg_batch
+    from deepdfa_amd.models import FlowGNNGGNNModule
+    from deepdfa_amd.parallel.optim import FlatAdamW
+
+    torch.manual_seed(0)
+    model = FlowGNNGGNNModule(input_dim=1002).to("cuda")
+    g = synthetic_cfg_batch(48, seed=7).to("cuda")
+    opt = FlatAdamW(model.parameters(), lr=1e-3)
+
+    def loss_fn():
+        with torch.autocast(device_type="cuda", dtype=torch.bfloat16):
+            return model.training_step((g, {}))
+
+    opt.zero_grad()
+    loss_fn().backward()
+    once = opt.flat_g.detach().clone()
+    opt.zero_grad()
+    loss_fn().backward()
+    loss_fn().backward()
+    twice = opt.flat_g.detach().clone()
+    err = (twice - 2 * once).abs().max() / once.abs().max().clamp(min=1e-8)
+    assert float(err) < 1e-3, float(err)
