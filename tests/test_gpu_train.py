@@ -258,3 +258,34 @@ def test_direct_grad_accumulation_sums_over_microbatches():
     twice = opt.flat_g.detach().clone()
     err = (twice - 2 * once).abs().max() / once.abs().max().clamp(min=1e-8)
     assert float(err) < 1e-3, float(err)
+
+
+@pytest.mark.gpu
+def test_transformer_direct_grad_sums_over_microbatches():
+    """Same accumulation contract for the transformer direct paths
+    (wgrad2 forced-atomic, colsum, rmsnorm_wgrad, embed_scatter, packed
+    q/k/v wgrad)."""
+    import torch
+    from deepdfa_amd.models.t5 import T5Config, T5ForConditionalGeneration
+    from deepdfa_amd.parallel.optim import FlatAdamW
+
+    torch.manual_seed(0)
+    cfg = T5Config(num_layers=2, num_decoder_layers=2, vocab_size=1024)
+    model = T5ForConditionalGeneration(cfg).to("cuda")
+    ids = torch.randint(3, cfg.vocab_size, (2, 64), device="cuda")
+    opt = FlatAdamW(model.parameters(), lr=1e-3)
+
+    def loss_fn():
+        with torch.autocast(device_type="cuda", dtype=torch.bfloat16):
+            dec = model(ids, labels=ids, output_hidden_only=True)
+        return dec.float().square().mean()
+
+    opt.zero_grad()
+    loss_fn().backward()
+    once = opt.flat_g.detach().clone()
+    opt.zero_grad()
+    loss_fn().backward()
+    loss_fn().backward()
+    twice = opt.flat_g.detach().clone()
+    err = (twice - 2 * once).abs().max() / once.abs().max().clamp(min=1e-8)
+    assert float(err) < 2e-3, float(err)
