@@ -1059,14 +1059,12 @@ __global__ void mlp3_wgrad_kernel(
 #pragma unroll
       for (int i = 0; i < MW_IB; ++i) acc[i] += dr[i] * xv;
     }
-    if (nrch == 1) {
+    // ALWAYS accumulate: dW may be a live flat .grad view (direct-grad
+    // mode) carrying an earlier micro-batch's contribution — a plain
+    // store at nrch==1 dropped it (caught by the 2x-accumulation test)
 #pragma unroll
-      for (int i = 0; i < MW_IB; ++i) dW[(long)(i0 + i) * D + jj] = acc[i];
-    } else {
-#pragma unroll
-      for (int i = 0; i < MW_IB; ++i)
-        atomicAdd(&dW[(long)(i0 + i) * D + jj], acc[i]);
-    }
+    for (int i = 0; i < MW_IB; ++i)
+      atomicAdd(&dW[(long)(i0 + i) * D + jj], acc[i]);
     return;
   }
   // tail blocks: dW3 + biases, one output per thread across 4 blocks,
