@@ -270,7 +270,8 @@ def test_transformer_direct_grad_sums_over_microbatches():
     from deepdfa_amd.parallel.optim import FlatAdamW
 
     torch.manual_seed(0)
-    cfg = T5Config(num_layers=2, num_decoder_layers=2, vocab_size=1024)
+    cfg = T5Config(num_layers=2, num_decoder_layers=2, vocab_size=1024,
+                   dropout_rate=0.0)  # dropout redraws per microbatch
     model = T5ForConditionalGeneration(cfg).to("cuda")
     ids = torch.randint(3, cfg.vocab_size, (2, 64), device="cuda")
     opt = FlatAdamW(model.parameters(), lr=1e-3)
