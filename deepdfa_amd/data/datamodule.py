@@ -19,6 +19,17 @@ from .dataset import BigVulDatasetLineVD, collate_graphs
 from .features import parse_limits
 
 
+def _collate_bytes(items):
+    # serialize to plain bytes: returning thousands of small tensors per
+    # batch through worker queues exhausts the fd-passing shared-memory
+    # strategy ("received 0 items of ancdata")
+    import io
+
+    buf = io.BytesIO()
+    torch.save(items, buf)
+    return buf.getvalue()
+
+
 class BigVulDatasetLineVDDataModule:
     def __init__(
         self,
@@ -77,7 +88,7 @@ class BigVulDatasetLineVDDataModule:
         return (n_neg / n_pos) if n_pos else 1.0
 
     def _loader(self, ds, shuffle: bool, epoch_subset: bool = False, generator=None,
-                rank: int = 0, world: int = 1):
+                rank: int = 0, world: int = 1, workers: Optional[int] = None):
         if epoch_subset and (ds.undersample is not None or ds.oversample is not None):
             idx = np.asarray(ds.get_epoch_indices())
         else:
@@ -105,10 +116,36 @@ class BigVulDatasetLineVDDataModule:
             dataset,
             batch_size=self.batch_size,
             shuffle=False,  # order fixed by the shared permutation above
-            num_workers=self.train_workers,
+            # train: fresh undersample draw each epoch -> parallel workers
+            # win (cold generation). val/test: FIXED id sets -> workers=0 so
+            # the in-process graph cache serves every epoch after the first
+            # (worker processes are respawned per epoch and lose theirs).
+            num_workers=self.train_workers if workers is None else workers,
             collate_fn=collate_graphs,
             drop_last=False,
         )
+
+    def _prewarm(self, ds):
+        """Fill a fixed split's in-process graph cache ONCE with parallel
+        workers (synthetic generation is CPU-bound); the split's loaders
+        then run workers=0 so the cache serves every later epoch — worker
+        processes are respawned per epoch and would lose theirs."""
+        if (self.train_workers <= 0 or ds.graph_dir is not None
+                or ds.missing_rate > 0 or len(ds) < 4096):
+            return
+        missing = [i for i in range(len(ds)) if ds.idx2id[i] not in ds._cache]
+        if len(missing) < 4096:
+            return
+        import io
+
+        tmp = DataLoader(Subset(ds, missing), batch_size=64,
+                         num_workers=self.train_workers,
+                         collate_fn=_collate_bytes)
+        pos = 0
+        for blob in tmp:  # DataLoader preserves sample order across workers
+            for g, _extra in torch.load(io.BytesIO(blob), weights_only=False):
+                ds._cache[ds.idx2id[missing[pos]]] = g
+                pos += 1
 
     def train_dataloader(self, generator: Optional[torch.Generator] = None,
                          rank: int = 0, world: int = 1):
@@ -118,7 +155,11 @@ class BigVulDatasetLineVDDataModule:
                             generator=generator, rank=rank, world=world)
 
     def val_dataloader(self, rank: int = 0, world: int = 1):
-        return self._loader(self.val, shuffle=False, rank=rank, world=world)
+        self._prewarm(self.val)
+        return self._loader(self.val, shuffle=False, rank=rank, world=world,
+                            workers=0)
 
     def test_dataloader(self, rank: int = 0, world: int = 1):
-        return self._loader(self.test, shuffle=False, rank=rank, world=world)
+        self._prewarm(self.test)
+        return self._loader(self.test, shuffle=False, rank=rank, world=world,
+                            workers=0)
