@@ -27,10 +27,6 @@ using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
 using f32x4 = __attribute__((ext_vector_type(4))) float;
 using uint4v = __attribute__((ext_vector_type(4))) unsigned int;
 
-__device__ __forceinline__ int w2_swz(int row, int byte) {
-  return row * 128 + (byte ^ ((row & 7) << 4));
-}
-
 // each thread loads 4 x 16 B of one operand tile (64 K-rows x 128 cols):
 // pass p: k = (t>>4) + 16p, col8 = (t&15)*8
 __device__ __forceinline__ void w2_load(const bf16* __restrict__ src, long ld,
@@ -53,22 +49,6 @@ __device__ __forceinline__ void w2_load(const bf16* __restrict__ src, long ld,
       if (k < k_lim)
         regs[p] = *reinterpret_cast<const uint4v*>(src + (long)k * ld + col0 + c8);
     }
-  }
-}
-
-// transposed write: element (k, col8 + j) -> lds[col][k*2 bytes], swizzled
-__device__ __forceinline__ void w2_write(char* lds, const uint4v regs[4],
-                                         int tid) {
-  const int kr = tid >> 4;
-  const int c8 = (tid & 15) * 8;
-#pragma unroll
-  for (int p = 0; p < 4; ++p) {
-    const int k = kr + 16 * p;
-    bf16 vals[8];
-    *reinterpret_cast<uint4v*>(vals) = regs[p];
-#pragma unroll
-    for (int j = 0; j < 8; ++j)
-      *reinterpret_cast<bf16*>(lds + w2_swz(c8 + j, k * 2)) = vals[j];
   }
 }
 

@@ -202,7 +202,7 @@ def main(argv: Optional[List[str]] = None) -> Dict:
             return {"coverage": cov}
         if args.freeze_graph:
             freeze_graph_weights(model, args.freeze_graph)
-        if cfg.get("ckpt_path"):
+        if cfg.get("ckpt_path") and args.subcommand != "fit":
             trainer.load_checkpoint(model, cfg["ckpt_path"])
         if args.subcommand == "fit":
             from ..utils.logging import HPOReporter
@@ -220,7 +220,10 @@ def main(argv: Optional[List[str]] = None) -> Dict:
             else:
                 opt = torch.optim.Adam(params, **cfg["optimizer"])
             reporter = HPOReporter(run_dir)
-            out = trainer.fit(model, dm, optimizer=opt, hpo_reporter=reporter)
+            # fit(ckpt_path=...) = mid-training RESUME: weights + optimizer
+            # state + epoch/step counters (Lightning semantics)
+            out = trainer.fit(model, dm, optimizer=opt, hpo_reporter=reporter,
+                              ckpt_path=cfg.get("ckpt_path"))
             # post-fit: validate the best checkpoint (main_cli.py:167-184)
             best = out["best_checkpoint"]
             if best:

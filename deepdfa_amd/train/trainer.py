@@ -132,7 +132,8 @@ class Trainer:
 
     # -- loops ----------------------------------------------------------------
 
-    def fit(self, model, datamodule, optimizer=None, hpo_reporter=None) -> Dict:
+    def fit(self, model, datamodule, optimizer=None, hpo_reporter=None,
+            ckpt_path: Optional[str] = None) -> Dict:
         from ..utils.logging import ScalarLogger
 
         rank = init_distributed()
@@ -187,10 +188,26 @@ class Trainer:
             logger.info("fit: hipGraph step capture enabled")
         history = []
         scalars = ScalarLogger(self.root) if rank == 0 else None
+        # mid-training resume (Lightning fit(ckpt_path=...) semantics,
+        # SURVEY §5.4): restore weights + optimizer state + epoch/step
+        # counters and continue from the NEXT epoch
+        start_epoch = 0
+        if ckpt_path:
+            payload = self.load_checkpoint(model, ckpt_path, optimizer)
+            start_epoch = int(payload.get("epoch", -1)) + 1
+            self.global_step = int(payload.get("global_step", 0))
+            if rank == 0:
+                logger.info("fit: resumed from %s at epoch %d (step %d)",
+                            ckpt_path, start_epoch, self.global_step)
         # one shared generator, advanced once per epoch on EVERY rank: all
-        # ranks draw the identical epoch permutation, then take their shard
+        # ranks draw the identical epoch permutation, then take their shard.
+        # On resume, replay the completed epochs' loader draws (index-only,
+        # no graph loading) so the undersample state and permutation stream
+        # match an uninterrupted run.
         gen = torch.Generator().manual_seed(self.seed)
-        for epoch in range(self.max_epochs):
+        for _ in range(start_epoch):
+            datamodule.train_dataloader(generator=gen, rank=rank, world=ws)
+        for epoch in range(start_epoch, self.max_epochs):
             model.train()
             t0 = _time.perf_counter()
             train_loader = datamodule.train_dataloader(generator=gen, rank=rank, world=ws)
