@@ -83,8 +83,11 @@ class _Embed4Direct(torch.autograd.Function):
 def embed4_direct_view(ws):
     """Returns the zero-copy (4, V, 32) bf16 stack view of the 4 table
     weights' optimizer shadow, or None when unavailable (no flat optimizer,
-    non-adjacent tables, or CPU)."""
-    if any(getattr(w, "_dfa_w16", None) is None or w.grad is None for w in ws):
+    non-adjacent tables, or CPU). Syncs each table's shadow first so a
+    torch-level write (checkpoint load) is reflected (_shadow_w16)."""
+    from .transformer import _shadow_w16
+
+    if any(_shadow_w16(w) is None or w.grad is None for w in ws):
         return None
     base = ws[0]._dfa_w16_base
     off = ws[0]._dfa_w16_off
