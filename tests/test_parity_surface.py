@@ -53,10 +53,19 @@ SURFACES = {
     "deepdfa_amd.evaluator.calc_code_bleu": ["calc_code_bleu"],
     "deepdfa_amd.ops.transformer": [
         "layer_norm", "rms_norm", "bias_gelu", "masked_softmax_dropout",
-        "fused_linear", "fused_qkv", "flash_attention", "flash_attention_qkv",
-        "layer_norm_res_dropout", "dropout_add", "embedding_lookup",
+        "fused_linear", "fused_qkv", "fused_kv", "flash_attention",
+        "flash_attention_qkv", "flash_attention_kv", "relu_dropout",
+        "lmhead_cross_entropy", "layer_norm_res_dropout", "dropout_add",
+        "embedding_lookup",
     ],
-    "deepdfa_amd.ops.flowgnn": ["embed4", "spmm_sum", "ggnn_fused", "attn_pool", "segment_max"],
+    "deepdfa_amd.ops.flowgnn": [
+        "embed4", "embed4_direct", "spmm_sum", "ggnn_fused", "attn_pool",
+        "segment_max", "gate_pool", "mlp3", "bce_with_logits",
+    ],
+    "deepdfa_amd.train.capture": ["CapturedTrainStep"],
+    "deepdfa_amd.graph.pad": ["bucket_shape", "pad_batch"],
+    "deepdfa_amd.evaluator.cparser": ["parse_c", "remove_comments"],
+    "deepdfa_amd.evaluator.dfg_c": ["get_data_flow", "normalize_dataflow", "corpus_dataflow_match", "corpus_syntax_match"],
 }
 
 SCRIPTS = [
