@@ -230,6 +230,19 @@ def build_tokenizer(args, cfg, corpus=None):
 
         return _wrap_hf(load_pretrained_tokenizer(args.tokenizer_name), cfg)
     if args.use_word_level_tokenizer or args.use_non_pretrained_tokenizer:
+        # shipped tokenizer assets (reference parity: LineVul checks in
+        # bpe_tokenizer/ + word_level_tokenizer/ JSONs and the flags LOAD
+        # them, linevul_main.py:608-613); retrain from the corpus only when
+        # the asset is absent
+        kind = "word_level_tokenizer" if args.use_word_level_tokenizer else "bpe_tokenizer"
+        asset = os.path.join(
+            os.path.dirname(os.path.dirname(os.path.dirname(os.path.abspath(__file__)))),
+            "assets", kind, "tokenizer.json",
+        )
+        if corpus is None and os.path.exists(asset):
+            from tokenizers import Tokenizer
+
+            return _wrap_hf(Tokenizer.from_file(asset), cfg)
         from ..data.tokenization import (
             synthetic_corpus,
             train_bpe_tokenizer,

@@ -100,3 +100,22 @@ def test_end_to_end_learning_reaches_high_f1(tmp_path):
     best = linevul_main.train(args, model, dss["train"], dss["val"], None,
                               torch.device("cpu"))
     assert best > 0.8, best
+
+
+def test_shipped_tokenizer_assets_load():
+    """The checked-in bpe/word-level tokenizer JSONs (assets/, reference
+    LineVul ships the same two) load and round-trip C-ish source."""
+    import os
+
+    from deepdfa_amd.train.linevul_main import build_args, build_tokenizer
+    from deepdfa_amd.models.roberta import RobertaConfig
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    for flag, kind in (("--use_non_pretrained_tokenizer", "bpe_tokenizer"),
+                       ("--use_word_level_tokenizer", "word_level_tokenizer")):
+        assert os.path.exists(os.path.join(repo, "assets", kind, "tokenizer.json"))
+        args = build_args([flag])
+        cfg = RobertaConfig(vocab_size=50265)
+        tok = build_tokenizer(args, cfg)
+        ids = tok.encode("int main(void) { return strcpy(dst, src); }", 64)
+        assert len(ids) == 64 and max(ids) < 50265
