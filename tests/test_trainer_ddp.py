@@ -145,3 +145,29 @@ def test_bucket_shape_covers_node_heavy_padding():
     padded, w = pad_batch(g, *shape)
     assert padded.num_nodes == shape[0] and padded.num_edges == shape[1]
     assert int(w.sum()) == 48
+
+
+def test_val_loader_prewarm_fills_cache_once():
+    """Fixed eval splits: the datamodule prewarm fills the in-process graph
+    cache with parallel workers on the FIRST val_dataloader() call; later
+    epochs serve from cache (no regeneration) with workers=0 loaders."""
+    dm = BigVulDatasetLineVDDataModule(batch_size=64, n_synthetic=60000,
+                                       undersample="v1.0", seed=0,
+                                       train_workers=4)
+    before = len(dm.val._cache)
+    loader = dm.val_dataloader()
+    assert loader.num_workers == 0
+    after = len(dm.val._cache)
+    assert before < 4096 <= after  # prewarm populated the cache
+    assert after >= len(dm.val)
+    # second call: nothing new to warm, loader unchanged
+    dm.val_dataloader()
+    assert len(dm.val._cache) == after
+
+
+def test_small_split_skips_prewarm():
+    dm = BigVulDatasetLineVDDataModule(batch_size=8, n_synthetic=300,
+                                       undersample="v1.0", seed=0,
+                                       train_workers=4)
+    dm.val_dataloader()
+    assert len(dm.val._cache) < 4096  # below threshold: lazy loading only
