@@ -98,3 +98,36 @@ def test_synthetic_batch_deterministic():
     assert torch.equal(a.ndata["_ABS_DATAFLOW_api"], b.ndata["_ABS_DATAFLOW_api"])
     c = synthetic_cfg_batch(8, seed=43)
     assert not torch.equal(a.ndata["_ABS_DATAFLOW_api"], c.ndata["_ABS_DATAFLOW_api"])
+
+
+def test_pad_batch_fuzz_invariants():
+    """Property fuzz over random batches x adversarial quanta: pad_batch
+    must always produce exactly the bucket shape, a monotone CSR whose
+    in/out edge counts agree, zeroed dummy features, and a weight mask
+    selecting exactly the real graphs."""
+    import torch
+
+    from deepdfa_amd.graph.pad import bucket_shape, pad_batch
+    from deepdfa_amd.graph.synthetic import synthetic_cfg_batch
+
+    for seed in range(12):
+        bs = 3 + (seed * 7) % 40
+        g = synthetic_cfg_batch(bs, seed=seed)
+        node_q = [64, 256, 1024, 4096][seed % 4]
+        edge_q = [32, 64, 512, 4096][(seed + 1) % 4]
+        shape = bucket_shape(g, b_pad=bs + 1 + seed % 3, node_q=node_q, edge_q=edge_q)
+        padded, w = pad_batch(g, *shape)
+        assert (padded.num_nodes, padded.num_edges, padded.num_graphs) == shape
+        assert int(w.sum()) == bs and w.numel() == shape[2]
+        d = padded.indptr.to(torch.int64).diff()
+        assert (d >= 0).all() and int(padded.indptr[-1]) == padded.num_edges
+        assert int(padded.t_indptr[-1]) == padded.num_edges
+        assert (padded.indices < padded.num_nodes).all()
+        # real prefix untouched; dummy features zero
+        assert torch.equal(padded.indices[: g.num_edges], g.indices)
+        for k, v in padded.ndata.items():
+            assert (v[g.num_nodes:] == 0).all(), k
+        # per-graph node offsets cover [0, n_pad] monotonically
+        off = padded.node_offsets.to(torch.int64)
+        assert int(off[0]) == 0 and int(off[-1]) == padded.num_nodes
+        assert (off.diff() >= 0).all()
