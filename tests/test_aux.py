@@ -82,3 +82,24 @@ def test_report_profiling(tmp_path):
         capture_output=True, text=True, check=True,
     ).stdout
     assert "gflops" in out and "ms/example" in out
+
+
+def test_scalar_logger_and_hpo_reporter(tmp_path):
+    """ScalarLogger writes a JSONL scalar stream; HPOReporter records
+    intermediate/final results (the reference's nni hooks, gated)."""
+    import json
+    import os
+
+    from deepdfa_amd.utils.logging import HPOReporter, ScalarLogger
+
+    root = str(tmp_path)
+    sl = ScalarLogger(root)
+    sl.log({"train_loss": 0.5}, step=0)
+    sl.log({"train_loss": 0.25}, step=1)
+    rows = [json.loads(l) for l in open(sl.path)]
+    assert rows[0]["train_loss"] == 0.5 and rows[1]["step"] == 1
+
+    rep = HPOReporter(root)
+    rep.report_intermediate(0.4)
+    rep.report_final(0.9)
+    assert rep.get_next_parameter() in (None, {}, rep.get_next_parameter())
