@@ -92,7 +92,7 @@ def main(argv=None):
                 loss.backward()
                 opt.clip_grad_norm_(1.0)
                 opt.step()
-                losses.append(float(loss))
+                losses.append(float(loss.detach()))
             logger.info("epoch %d loss %.4f", epoch, sum(losses) / len(losses))
             results["train_loss"] = sum(losses) / len(losses)
         torch.save(model.state_dict(), os.path.join(args.output_dir, "pytorch_model.bin"))
