@@ -93,7 +93,7 @@ def main(argv=None):
                 opt.zero_grad()
                 loss.backward()
                 opt.step()
-                losses.append(float(loss))
+                losses.append(float(loss.detach()))
             results["train_loss"] = sum(losses) / len(losses)
         torch.save(model.state_dict(),
                    os.path.join(args.output_dir, "pytorch_model.bin"))
