@@ -196,6 +196,15 @@ class Trainer:
             payload = self.load_checkpoint(model, ckpt_path, optimizer)
             start_epoch = int(payload.get("epoch", -1)) + 1
             self.global_step = int(payload.get("global_step", 0))
+            # same-run-dir resume: restore the best-val watermark from the
+            # existing performance-*.ckpt names so a worse post-resume epoch
+            # doesn't displace the historical best (save_top_k pruning)
+            bp = self.best_checkpoint()
+            if bp:
+                m = re.search(r"performance-\d+-\d+-([\d.]+)\.ckpt$", bp)
+                if m:
+                    self._best_path = bp
+                    self._best_val = float(m.group(1))
             if rank == 0:
                 logger.info("fit: resumed from %s at epoch %d (step %d)",
                             ckpt_path, start_epoch, self.global_step)

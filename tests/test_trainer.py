@@ -103,3 +103,42 @@ def test_fit_resume_matches_uninterrupted(tmp_path):
     got = torch.cat([p.detach().flatten() for p in model3.parameters()])
     assert torch.allclose(got, ref, atol=1e-6), float((got - ref).abs().max())
     assert len(out["history"]) == 2  # only the resumed epochs ran
+
+
+def test_resume_same_dir_keeps_historical_best(tmp_path):
+    """Resuming into the SAME run dir restores the best-val watermark from
+    the existing performance-*.ckpt names: a worse post-resume epoch must
+    not displace the historical best checkpoint."""
+    import glob
+
+    import torch
+
+    from deepdfa_amd.data.datamodule import BigVulDatasetLineVDDataModule
+    from deepdfa_amd.models import FlowGNNGGNNModule
+    from deepdfa_amd.train.trainer import Trainer
+
+    torch.manual_seed(0)
+    dm = BigVulDatasetLineVDDataModule(batch_size=8, n_synthetic=200,
+                                       undersample="v1.0", seed=0)
+    model = FlowGNNGGNNModule(input_dim=1002, hidden_dim=8, n_steps=2,
+                              num_output_layers=2)
+    opt = torch.optim.Adam(model.parameters(), lr=1e-3)
+    root = str(tmp_path / "run")
+    tr = Trainer(max_epochs=2, default_root_dir=root, seed=3)
+    tr.fit(model, dm, optimizer=opt)
+    best_before = tr.best_checkpoint()
+    assert best_before
+
+    tr2 = Trainer(max_epochs=4, default_root_dir=root, seed=3)
+    tr2.ckpt_dir = tr.ckpt_dir  # same run dir (fresh Trainer objects stamp new dirs)
+    tr2.fit(model, dm, optimizer=opt,
+            ckpt_path=os.path.join(tr.ckpt_dir, "last.ckpt"))
+    # watermark restored: at most one best file exists per val-loss record,
+    # and the selected best is never worse than the pre-resume best
+    import re as _re
+
+    def loss_of(p):
+        return float(_re.search(r"performance-\d+-\d+-([\d.]+)\.ckpt$", p).group(1))
+
+    best_after = tr2.best_checkpoint()
+    assert loss_of(best_after) <= loss_of(best_before) + 1e-9
