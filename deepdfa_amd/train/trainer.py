@@ -73,6 +73,7 @@ class Trainer:
         precision: str = "bf16",  # "bf16" (autocast on GPU) or "fp32"
         detect_anomaly: bool = False,
         graph_capture: bool = True,  # hipGraph step capture (GPU, ws=1, FlatAdamW)
+        accumulate_grad_batches: int = 1,  # Lightning-parity micro-batching
     ):
         self.graph_capture = graph_capture
         # reference config_default.yaml:37 trainer.detect_anomaly — autograd
@@ -84,6 +85,7 @@ class Trainer:
         self.check_val_every = check_val_every_n_epoch
         self.periodic_every = periodic_every
         self.grad_clip = grad_clip
+        self.accumulate_grad_batches = max(1, int(accumulate_grad_batches))
         self.log_every = log_every_n_steps
         self.seed = seed
         if device is None:
@@ -173,6 +175,7 @@ class Trainer:
         captured = None
         if (
             self.graph_capture
+            and self.accumulate_grad_batches == 1
             and ws == 1
             and self.device.type == "cuda"
             and self.precision == "bf16"
@@ -222,24 +225,45 @@ class Trainer:
             train_loader = datamodule.train_dataloader(generator=gen, rank=rank, world=ws)
             n_batches = 0
             loss_sum = 0.0
-            for batch in train_loader:
+            accum = self.accumulate_grad_batches
+            loader_iter = list(train_loader) if accum > 1 else train_loader
+            total_b = len(loader_iter) if accum > 1 else None
+            for bi, batch in enumerate(loader_iter):
                 if captured is not None:
                     # padding happens on the CPU batch; the captured step
                     # copies the padded tensors into its static device buffers
                     captured(batch)
                 else:
+                    # accumulation (Lightning accumulate_grad_batches
+                    # semantics): zero at group start, loss/accum per
+                    # micro-batch, one averaging/clip/step at group end
+                    # (also handles a short trailing group)
+                    group_start = bi % accum == 0
+                    group_end = (bi + 1) % accum == 0 or (
+                        total_b is not None and bi + 1 == total_b
+                    )
                     batch = self._to_device(batch)
+                    if group_start:
+                        optimizer.zero_grad(set_to_none=False)
                     with self._autocast():
                         loss = model.training_step(batch)
-                    optimizer.zero_grad(set_to_none=True)
-                    loss.backward()
-                    if flat and ws > 1:
-                        optimizer.allreduce_grads()
-                    elif ddp is not None:
-                        ddp.finalize()
-                    if self.grad_clip:
-                        torch.nn.utils.clip_grad_norm_(model.parameters(), self.grad_clip)
-                    optimizer.step()
+                    scaled = loss / accum if accum > 1 else loss
+                    if ddp is not None and not group_end:
+                        # hook engine: suppress per-micro-batch bucket
+                        # all-reduces; grads accumulate locally until the
+                        # group-end finalize (parallel/ddp.py no_sync)
+                        with ddp.no_sync():
+                            scaled.backward()
+                    else:
+                        scaled.backward()
+                    if group_end:
+                        if flat and ws > 1:
+                            optimizer.allreduce_grads()
+                        elif ddp is not None:
+                            ddp.finalize()
+                        if self.grad_clip:
+                            torch.nn.utils.clip_grad_norm_(model.parameters(), self.grad_clip)
+                        optimizer.step()
                     loss_sum += float(loss.detach())
                 self.global_step += 1
                 n_batches += 1

@@ -142,3 +142,48 @@ def test_resume_same_dir_keeps_historical_best(tmp_path):
 
     best_after = tr2.best_checkpoint()
     assert loss_of(best_after) <= loss_of(best_before) + 1e-9
+
+
+def test_accumulate_grad_batches_matches_joint_step(tmp_path):
+    """Lightning-parity accumulate_grad_batches: two equal micro-batches at
+    accum=2 produce EXACTLY the same update as one step on the averaged
+    loss (loss/accum per micro-batch, zero at group start, step at end)."""
+    import copy
+
+    import torch
+
+    from deepdfa_amd.graph.synthetic import synthetic_cfg_batch
+    from deepdfa_amd.models import FlowGNNGGNNModule
+    from deepdfa_amd.train.trainer import Trainer
+
+    torch.manual_seed(0)
+    model = FlowGNNGGNNModule(input_dim=1002, hidden_dim=8, n_steps=2,
+                              num_output_layers=2)
+    ref = copy.deepcopy(model)
+    b1 = synthetic_cfg_batch(8, seed=1)
+    b2 = synthetic_cfg_batch(8, seed=2)
+
+    class StubDM:
+        batch_size = 8
+
+        def train_dataloader(self, generator=None, rank=0, world=1):
+            return [(b1, {}), (b2, {})]
+
+        def val_dataloader(self, rank=0, world=1):
+            return []
+
+    tr = Trainer(max_epochs=1, default_root_dir=str(tmp_path), seed=0,
+                 precision="fp32", accumulate_grad_batches=2,
+                 check_val_every_n_epoch=100, periodic_every=100)
+    opt = torch.optim.SGD(model.parameters(), lr=0.1)
+    tr.fit(model, StubDM(), optimizer=opt)
+
+    # manual reference: one step on the mean of the two batch losses
+    opt_ref = torch.optim.SGD(ref.parameters(), lr=0.1)
+    opt_ref.zero_grad()
+    loss = (ref.training_step((b1, {})) + ref.training_step((b2, {}))) / 2
+    loss.backward()
+    opt_ref.step()
+
+    for (n, p), (_, q) in zip(model.named_parameters(), ref.named_parameters()):
+        assert torch.allclose(p, q, atol=1e-7), n
