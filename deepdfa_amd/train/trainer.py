@@ -226,9 +226,10 @@ class Trainer:
             n_batches = 0
             loss_sum = 0.0
             accum = self.accumulate_grad_batches
-            loader_iter = list(train_loader) if accum > 1 else train_loader
-            total_b = len(loader_iter) if accum > 1 else None
-            for bi, batch in enumerate(loader_iter):
+            # trailing-group detection needs the batch count; DataLoader
+            # knows it without materializing the epoch
+            total_b = len(train_loader) if accum > 1 else None
+            for bi, batch in enumerate(train_loader):
                 if captured is not None:
                     # padding happens on the CPU batch; the captured step
                     # copies the padded tensors into its static device buffers
