@@ -99,7 +99,7 @@ def train(args, model, train_dataset, eval_dataset, flowgnn_dataset, device):
         sampler = RandomSampler(train_dataset, generator=torch.Generator().manual_seed(args.seed))
     loader = DataLoader(train_dataset, sampler=sampler, batch_size=args.train_batch_size)
     max_steps = args.epochs * len(loader)
-    optimizer = torch.optim.AdamW(model.parameters(), lr=args.learning_rate, eps=1e-8)
+    optimizer = torch.optim.AdamW(model.parameters(), lr=args.learning_rate, eps=1e-8, weight_decay=0.0)  # HF AdamW default (reference parity; torch defaults to 0.01)
     scheduler = linear_warmup_decay(optimizer, int(max_steps * 0.2), max_steps)
     ddp = DDPEngine(model)
     best_f1 = -1.0  # first epoch always checkpoints

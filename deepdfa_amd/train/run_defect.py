@@ -142,7 +142,7 @@ def main(argv=None):
         loader = DataLoader(train_ds, sampler=sampler, batch_size=args.train_batch_size)
         steps_per_epoch = max(1, len(loader) // args.gradient_accumulation_steps)
         max_steps = args.num_train_epochs * steps_per_epoch
-        opt = torch.optim.AdamW(model.parameters(), lr=args.learning_rate, eps=1e-8)
+        opt = torch.optim.AdamW(model.parameters(), lr=args.learning_rate, eps=1e-8, weight_decay=0.0)  # HF AdamW default (reference parity; torch defaults to 0.01)
         sched = linear_warmup_decay(opt, int(max_steps * 0.1), max_steps)
         ddp = DDPEngine(model)
         best_f1, not_improved = -1.0, 0
