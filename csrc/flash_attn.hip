@@ -7,8 +7,10 @@
 // Shapes: Q/K/V/O are (B, L, H*64) — the projections' natural output — so
 // no transpose/contiguous copies exist anywhere in the attention path.
 // Supports: suffix-padding valid lengths per batch row, causal masking
-// (T5 decoder), additive position bias (T5 relative attention, fp32
-// (H, L, L)), fused stateless dropout (mask regenerated in backward),
+// (T5 decoder), additive position bias (T5 relative attention, fp32,
+// stored TRANSPOSED key-major (H, Lk, Lq) so the 16 q-column lanes of an
+// unrolled load share one 64-B line — models/t5.py _flash_bias_T),
+// fused stateless dropout (mask regenerated in backward),
 // softmax scale, and the log-sum-exp save for the backward pass.
 //
 // Tiling (cdna_hip_programming.md §B attention ladder, adapted to d=64):
