@@ -72,3 +72,34 @@ def test_scalar_and_hpo_logging(tmp_path, monkeypatch):
     hpo = [json.loads(l) for l in open(os.path.join(root, "hpo_metrics.jsonl"))]
     assert any("intermediate" in r for r in hpo)
     assert any("final" in r for r in hpo)
+
+
+def test_fit_resume_through_cli(tmp_path, monkeypatch):
+    """`fit --ckpt_path last.ckpt` resumes through main_cli: the resumed
+    run trains only the remaining epochs (history length) and reuses the
+    epoch counter from the checkpoint."""
+    monkeypatch.chdir(os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    out1, root1 = run(tmp_path, ["--trainer.max_epochs", "2",
+                                 "--data.train_workers", "0"])
+    last = None
+    for dirpath, _dirs, files in os.walk(root1):
+        if "last.ckpt" in files:
+            last = os.path.join(dirpath, "last.ckpt")
+    assert last, "last.ckpt not written"
+    payload = torch.load(last, map_location="cpu", weights_only=False)
+    assert payload["epoch"] == 1  # two epochs ran (0, 1)
+
+    root2 = str(tmp_path / "resumed")
+    out2 = main_cli.main([
+        "fit",
+        "--config", "configs/config_default.yaml",
+        "--config", "configs/config_bigvul.yaml",
+        "--trainer.max_epochs", "4",
+        "--trainer.default_root_dir", root2,
+        "--data.n_synthetic", "200",
+        "--data.batch_size", "64",
+        "--data.train_workers", "0",
+        "--ckpt_path", last,
+    ])
+    assert len(out2["history"]) == 2  # epochs 2 and 3 only
+    assert out2["history"][0]["epoch"] == 2
