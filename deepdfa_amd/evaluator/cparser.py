@@ -94,9 +94,15 @@ class Node:
     token: Optional[Token] = None  # leaves only
 
     def sexp(self) -> str:
+        # error-leaf types carry the raw token text; literal parentheses
+        # would unbalance the s-expression STRING (the subtree match
+        # compares sexps as strings) — emit named placeholders instead
+        t = self.type
+        if "(" in t or ")" in t:
+            t = t.replace("(", "<lparen>").replace(")", "<rparen>")
         if not self.children:
-            return f"({self.type})"
-        return f"({self.type} " + " ".join(c.sexp() for c in self.children) + ")"
+            return f"({t})"
+        return f"({t} " + " ".join(c.sexp() for c in self.children) + ")"
 
     def walk(self):
         yield self

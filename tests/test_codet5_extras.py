@@ -142,3 +142,47 @@ class TestCodeBLEUFull:
             ["int g ( ) { int x ; return a + b ; }"],
         )
         assert 0.0 < score < 1.0
+
+
+def test_cparser_never_raises_fuzz():
+    """The tolerant parser must survive arbitrary input (reference behavior:
+    tree-sitter always yields a tree; calc_code_bleu feeds it raw model
+    output). Random printable garbage, truncated C, unbalanced brackets."""
+    import random
+    import string
+
+    from deepdfa_amd.evaluator.cparser import all_subtree_sexps, parse_c
+
+    rng = random.Random(0)
+    samples = [
+        "", "{", "}}}}", "int f( {", "a+++++b;", '"unterminated',
+        "/* unclosed", "int x = ;;;", "\x00\x01\x02",
+    ]
+    for _ in range(40):
+        n = rng.randint(1, 120)
+        samples.append("".join(rng.choice(string.printable) for _ in range(n)))
+    chunks = ["int f(int a){", "return a+1;", "}", "for(;;)", "while(1)",
+              "x[3] =", "struct s {", "((", "))", "case 1:", "->y", "*p",
+              '"str"', "'c'", "0x1f", "e?a:b", ";;"]
+    for _ in range(40):
+        samples.append(" ".join(rng.choice(chunks) for _ in range(rng.randint(1, 12))))
+    for code in samples:
+        root = parse_c(code)  # must not raise
+        sexps = all_subtree_sexps(root)
+        assert isinstance(sexps, list)
+        for sx in sexps[:50]:
+            assert sx.count("(") == sx.count(")")  # well-formed s-exprs
+
+
+def test_dfg_extractor_never_raises_fuzz():
+    import random
+    import string
+
+    from deepdfa_amd.evaluator.dfg_c import get_data_flow, normalize_dataflow
+
+    rng = random.Random(1)
+    for _ in range(40):
+        n = rng.randint(1, 100)
+        code = "".join(rng.choice(string.printable) for _ in range(n))
+        df = get_data_flow(code)  # must not raise
+        normalize_dataflow(df)
