@@ -186,3 +186,19 @@ def test_dfg_extractor_never_raises_fuzz():
         code = "".join(rng.choice(string.printable) for _ in range(n))
         df = get_data_flow(code)  # must not raise
         normalize_dataflow(df)
+
+
+def test_codebleu_degenerate_inputs():
+    """calc_code_bleu must handle empty/degenerate candidates (models emit
+    empty strings early in training) without raising or NaN."""
+    from deepdfa_amd.evaluator.calc_code_bleu import calc_code_bleu
+
+    ref = "int f(int a){return a+1;}"
+    for cand in ["", ";", "(((", "int", ref]:
+        out = calc_code_bleu([ref], [cand])
+        score = out["code_bleu"]
+        assert 0.0 <= score <= 1.0 and score == score  # finite, in range
+        for k in ("ngram_match", "weighted_ngram_match", "syntax_match",
+                  "dataflow_match"):
+            assert k in out and out[k] == out[k]
+    assert calc_code_bleu([ref], [ref])["code_bleu"] > 0.95
