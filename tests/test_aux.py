@@ -103,3 +103,22 @@ def test_scalar_logger_and_hpo_reporter(tmp_path):
     rep.report_intermediate(0.4)
     rep.report_final(0.9)
     assert rep.get_next_parameter() in (None, {}, rep.get_next_parameter())
+
+
+def test_tokenise_fuzz_never_raises():
+    """IVDetect subtoken splitter must survive arbitrary input (it runs on
+    raw C from the dataset)."""
+    import random
+    import string
+
+    from deepdfa_amd.data.tokenization import tokenise, tokenise_lines
+
+    rng = random.Random(0)
+    for _ in range(60):
+        n = rng.randint(0, 160)
+        s = "".join(rng.choice(string.printable) for _ in range(n))
+        out = tokenise(s)
+        assert isinstance(out, str)
+        tokenise_lines(s)
+    # known behavior: camelCase + specials split
+    assert "camel" in tokenise("camelCase").lower().split()
