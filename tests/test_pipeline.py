@@ -276,3 +276,36 @@ class TestJoernGraphHelpers:
         # round-trips through the parser (native shape)
         flat = parse_dataflow_json(path)
         assert flat == json.load(open(path))["nodes"]
+
+
+def test_rd_straightline_monotonicity():
+    """Reference property test (dataflow.py:253-317 style): on a
+    straight-line chain of fresh-variable assignments, the reaching-def
+    set size at each statement grows monotonically with line number (no
+    kills, every def survives)."""
+    n_stmts = 8
+    nodes = {0: {"_label": "METHOD", "name": "f", "lineNumber": 1}}
+    edges = []
+    nid = 1
+    prev = None
+    for i in range(n_stmts):
+        call = nid
+        nodes[call] = {"_label": "CALL", "name": "<operator>.assignment",
+                       "code": f"v{i} = {i}", "lineNumber": i + 2}
+        nodes[nid + 1] = {"_label": "IDENTIFIER", "name": f"v{i}", "order": 1,
+                          "typeFullName": "int"}
+        nodes[nid + 2] = {"_label": "LITERAL", "name": str(i), "code": str(i),
+                          "order": 2}
+        edges += [(call, nid + 1, "AST"), (call, nid + 2, "AST")]
+        if prev is not None:
+            edges.append((prev, call, "CFG"))
+        prev = call
+        nid += 3
+    cpg = CPG(nodes=nodes, edges=edges)
+    rd = ReachingDefinitions(cpg)
+    IN, OUT = rd.solve()
+    calls = sorted(
+        (v["lineNumber"], k) for k, v in nodes.items() if v["_label"] == "CALL"
+    )
+    sizes = [len(OUT[k]) for _, k in calls]
+    assert sizes == list(range(1, n_stmts + 1))  # strictly growing by one
