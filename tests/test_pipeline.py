@@ -309,3 +309,27 @@ def test_rd_straightline_monotonicity():
     )
     sizes = [len(OUT[k]) for _, k in calls]
     assert sizes == list(range(1, n_stmts + 1))  # strictly growing by one
+
+
+def test_absdf_vocab_determinism_and_bounds():
+    """Vocab build (datasets.py:587-692 semantics): deterministic for the
+    same train features, capped at limit_all, indices start at 2
+    (0 = not-a-definition, 1 = UNKNOWN), most-frequent first."""
+    import pandas as pd
+
+    from deepdfa_amd.pipeline.absdf import SUBKEYS, build_vocab, to_hash
+
+    rows = []
+    for i in range(60):
+        rows.append({k: f"{k}{i % 7}" for k in SUBKEYS})
+    df = pd.DataFrame(rows)
+    v1 = build_vocab(df, SUBKEYS, limit_all=5)
+    v2 = build_vocab(df.copy(), SUBKEYS, limit_all=5)
+    assert v1 == v2
+    assert len(v1) == 5 and min(v1.values()) == 2 and max(v1.values()) == 6
+    # the most frequent hash gets index 2
+    top = to_hash(df.iloc[0], SUBKEYS)  # i%7==0 appears most (ceil(60/7)=9x)
+    assert v1[top] == 2
+    # unlimited: all 7 distinct hashes present
+    v_all = build_vocab(df, SUBKEYS, limit_all=1000)
+    assert len(v_all) == 7
