@@ -363,16 +363,11 @@ at::Tensor wgrad(at::Tensor A, at::Tensor B, c10::optional<at::Tensor> out_opt) 
                 "wgrad out must be a contiguous fp32 (M, C) accumulator");
   }
   if (M % 128 == 0 && C % 128 == 0) {  // wgrad2 zero-fills K tails
-    if (!acc) {
-      const int tiles = (M / 128) * (C / 128);
-      int zsplit = std::max(1, 512 / tiles);
-      int kchunk = (K + zsplit - 1) / zsplit;
-      kchunk = ((kchunk + 63) / 64) * 64;
-      if (kchunk < 512) kchunk = std::min(((K + 63) / 64) * 64, 512);
-      zsplit = (K + kchunk - 1) / kchunk;
-      out = zsplit == 1 ? at::empty({M, C}, A.options().dtype(at::kFloat))
-                        : at::zeros({M, C}, A.options().dtype(at::kFloat));
-    }
+    // always zero-init: whether the launcher plain-stores (one K chunk) or
+    // atomically accumulates (split K) is ITS decision — duplicating the
+    // chunking math here to pick empty-vs-zeros risked an uninitialized
+    // accumulator if the two ever diverged (they did differ by K-floor)
+    if (!acc) out = at::zeros({M, C}, A.options().dtype(at::kFloat));
     launch_wgrad2(ptr<bf16_t>(A), ptr<bf16_t>(B), out.data_ptr<float>(), nullptr,
                   K, M, C, acc ? 1 : 0, cur_stream());
     return out;
